@@ -1,0 +1,112 @@
+// Torch bindings for greptimedb_amd._hip_ops (see kernels.hip).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+#include <vector>
+
+namespace gdb_hip {
+void launch_ts_bucket_agg(
+    const int64_t*, const int32_t*, const double*, int64_t, const int32_t*, int,
+    const int32_t*, int, int64_t, int64_t, int64_t, int64_t, int, int, int64_t,
+    double*, unsigned long long*, unsigned long long*, unsigned long long*,
+    hipStream_t);
+void launch_decode_minmax(
+    const unsigned long long*, const unsigned long long*, const unsigned long long*,
+    double*, double*, int64_t, hipStream_t);
+void launch_filter_series_time(
+    const int64_t*, const int32_t*, const int32_t*, int, int64_t, int64_t,
+    int64_t, bool*, hipStream_t);
+void launch_dedup_mark_last(const int32_t*, const int64_t*, int64_t, bool*, hipStream_t);
+}  // namespace gdb_hip
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+// Fused filter + time-bucket aggregate (K1+K2+K5).
+// ts: i64[n] (ms), series: i32[n], fields: f64[nf_total, field_stride]
+// field_idx: i32[nf] rows of `fields` to aggregate, slot_lut: i32[lut_size].
+// Returns (sum f64, count i64, min f64, max f64), each [nf, n_slots, n_buckets].
+std::vector<torch::Tensor> ts_bucket_agg(
+    torch::Tensor ts, torch::Tensor series, torch::Tensor fields,
+    torch::Tensor field_idx, torch::Tensor slot_lut,
+    int64_t ts_lo, int64_t ts_hi, int64_t origin, int64_t bucket_ms,
+    int64_t n_slots, int64_t n_buckets) {
+  CHECK_GPU(ts); CHECK_GPU(series); CHECK_GPU(fields);
+  CHECK_GPU(field_idx); CHECK_GPU(slot_lut);
+  CHECK_CONTIG(ts); CHECK_CONTIG(series); CHECK_CONTIG(fields);
+  CHECK_CONTIG(field_idx); CHECK_CONTIG(slot_lut);
+  TORCH_CHECK(ts.scalar_type() == torch::kInt64);
+  TORCH_CHECK(series.scalar_type() == torch::kInt32);
+  TORCH_CHECK(fields.scalar_type() == torch::kFloat64);
+  TORCH_CHECK(fields.dim() == 2);
+  const int64_t n = ts.numel();
+  TORCH_CHECK(series.numel() == n);
+  TORCH_CHECK(fields.size(1) >= n, "fields stride shorter than rows");
+  const int nf = (int)field_idx.numel();
+  auto opts_f64 = ts.options().dtype(torch::kFloat64);
+  auto opts_i64 = ts.options().dtype(torch::kInt64);
+  const int64_t cells = nf * n_slots * n_buckets;
+  auto sum = torch::zeros({nf, n_slots, n_buckets}, opts_f64);
+  auto cnt = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+  // min keys init to u64 max, max keys to 0
+  auto minmax_init_min = torch::full({nf, n_slots, n_buckets}, -1, opts_i64);  // 0xFFFF...
+  auto minmax_init_max = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_ts_bucket_agg(
+      ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(), fields.data_ptr<double>(),
+      fields.size(1), field_idx.data_ptr<int32_t>(), nf,
+      slot_lut.data_ptr<int32_t>(), (int)slot_lut.numel(),
+      ts_lo, ts_hi, origin, bucket_ms, (int)n_slots, (int)n_buckets, n,
+      sum.data_ptr<double>(),
+      reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(minmax_init_min.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
+      stream);
+  auto minv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
+  auto maxv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
+  gdb_hip::launch_decode_minmax(
+      reinterpret_cast<unsigned long long*>(minmax_init_min.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
+      minv.data_ptr<double>(), maxv.data_ptr<double>(), cells, stream);
+  return {sum, cnt, minv, maxv};
+}
+
+torch::Tensor filter_series_time(
+    torch::Tensor ts, torch::Tensor series, torch::Tensor slot_lut,
+    int64_t ts_lo, int64_t ts_hi) {
+  CHECK_GPU(ts); CHECK_CONTIG(ts);
+  const int64_t n = ts.numel();
+  auto keep = torch::empty({n}, ts.options().dtype(torch::kBool));
+  const int32_t* lut = nullptr;
+  int lut_size = 0;
+  if (slot_lut.numel() > 0) {
+    CHECK_GPU(slot_lut); CHECK_CONTIG(slot_lut);
+    lut = slot_lut.data_ptr<int32_t>();
+    lut_size = (int)slot_lut.numel();
+  }
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_filter_series_time(
+      ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(), lut, lut_size,
+      ts_lo, ts_hi, n, keep.data_ptr<bool>(), stream);
+  return keep;
+}
+
+torch::Tensor dedup_mark_last(torch::Tensor series, torch::Tensor ts) {
+  CHECK_GPU(series); CHECK_GPU(ts); CHECK_CONTIG(series); CHECK_CONTIG(ts);
+  const int64_t n = ts.numel();
+  auto keep = torch::empty({n}, ts.options().dtype(torch::kBool));
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_dedup_mark_last(
+      series.data_ptr<int32_t>(), ts.data_ptr<int64_t>(), n,
+      keep.data_ptr<bool>(), stream);
+  return keep;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate");
+  m.def("filter_series_time", &filter_series_time, "series/time filter mask");
+  m.def("dedup_mark_last", &dedup_mark_last, "last-row dedup marker");
+}

@@ -1,0 +1,177 @@
+// CDNA4 (gfx950 / MI355X) kernels for the mito-hip scan path.
+//
+// Kernel inventory (SURVEY.md §2.7 numbering):
+//  - ts_bucket_agg_kernel : fused K1 (predicate filter via series LUT) +
+//    K2 (time-range visibility) + K5 (time-bucket aggregate). One pass over
+//    the (ts, series, fields) columns computing sum/count/min/max for every
+//    requested field into [nf, n_slots, n_buckets] accumulators.
+//    Memory-bound by design: ts+series loaded once per row and amortized
+//    over all selected fields.
+//  - filter_series_time_kernel : K1/K2 producing a keep-mask for raw scans.
+//  - dedup_mark_last_kernel : K4 last_row marker on (series, ts)-sorted data.
+//
+// Design notes (cdna_hip_programming.md):
+//  * 256-thread blocks (4 waves of 64), grid-stride, grid capped at
+//    2048 blocks (G11: memory-bound ops).
+//  * f64 min/max accumulate through the monotonic u64 key mapping so we can
+//    use hardware atomicMin/Max on unsigned long long (no f64 min/max atomic).
+//  * NaN field values are nulls (influx missing fields) and are skipped.
+//  * Small aggregate tables (slots*buckets) stay hot in L2/LLC; per-LDS
+//    privatization is a planned optimization once rocprof shows atomic
+//    contention (expected only for >100k groups).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#define DEV_INLINE __device__ __forceinline__
+
+namespace gdb_hip {
+
+DEV_INLINE uint64_t f64_to_key(double d) {
+  uint64_t bits = __double_as_longlong(d);
+  return (int64_t)bits < 0 ? ~bits : (bits | 0x8000000000000000ULL);
+}
+
+DEV_INLINE double key_to_f64(uint64_t key) {
+  uint64_t bits = (key & 0x8000000000000000ULL) ? (key & 0x7FFFFFFFFFFFFFFFULL) : ~key;
+  return __longlong_as_double(bits);
+}
+
+// ---------------------------------------------------------------- K1+K2+K5
+// fields: column-major per field: fields[f * field_stride + row]
+// out arrays: [nf, n_slots, n_buckets]
+__global__ void ts_bucket_agg_kernel(
+    const int64_t* __restrict__ ts,
+    const int32_t* __restrict__ series,
+    const double* __restrict__ fields,
+    int64_t field_stride,
+    const int32_t* __restrict__ field_idx, int nf,
+    const int32_t* __restrict__ slot_lut, int lut_size,
+    int64_t ts_lo, int64_t ts_hi, int64_t origin, int64_t bucket_ms,
+    int n_slots, int n_buckets, int64_t n,
+    double* __restrict__ out_sum,
+    unsigned long long* __restrict__ out_cnt,
+    unsigned long long* __restrict__ out_min,
+    unsigned long long* __restrict__ out_max) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const int64_t t = ts[i];
+    if (t < ts_lo || t >= ts_hi) continue;
+    const int32_t s = series[i];
+    if (s < 0 || s >= lut_size) continue;
+    const int32_t slot = slot_lut[s];
+    if (slot < 0) continue;
+    int64_t b = (t - origin) / bucket_ms;
+    if (b < 0 || b >= n_buckets) continue;
+    const int64_t cell0 = (int64_t)slot * n_buckets + b;
+    for (int f = 0; f < nf; f++) {
+      const double v = fields[(int64_t)field_idx[f] * field_stride + i];
+      if (isnan(v)) continue;
+      const int64_t cell = (int64_t)f * n_slots * n_buckets + cell0;
+      atomicAdd(&out_sum[cell], v);
+      atomicAdd(&out_cnt[cell], 1ULL);
+      const uint64_t k = f64_to_key(v);
+      atomicMin(&out_min[cell], (unsigned long long)k);
+      atomicMax(&out_max[cell], (unsigned long long)k);
+    }
+  }
+}
+
+// decode min/max key arrays into f64 (count==0 → NaN)
+__global__ void decode_minmax_kernel(
+    const unsigned long long* __restrict__ minkey,
+    const unsigned long long* __restrict__ maxkey,
+    const unsigned long long* __restrict__ cnt,
+    double* __restrict__ out_min, double* __restrict__ out_max, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    if (cnt[i] == 0) {
+      out_min[i] = __longlong_as_double(0x7FF8000000000000LL);  // NaN
+      out_max[i] = __longlong_as_double(0x7FF8000000000000LL);
+    } else {
+      out_min[i] = key_to_f64(minkey[i]);
+      out_max[i] = key_to_f64(maxkey[i]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------- K1/K2 mask
+__global__ void filter_series_time_kernel(
+    const int64_t* __restrict__ ts,
+    const int32_t* __restrict__ series,
+    const int32_t* __restrict__ slot_lut, int lut_size,  // lut==nullptr → all series
+    int64_t ts_lo, int64_t ts_hi, int64_t n,
+    bool* __restrict__ keep) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const int64_t t = ts[i];
+    bool k = (t >= ts_lo) & (t < ts_hi);
+    if (k && slot_lut != nullptr) {
+      const int32_t s = series[i];
+      k = (s >= 0) & (s < lut_size) && (slot_lut[s] >= 0);
+    }
+    keep[i] = k;
+  }
+}
+
+// ---------------------------------------------------------------- K4 dedup
+// Input sorted ascending by (series, ts, seq). Keep the LAST row of each
+// (series, ts) group (reference read/dedup.rs LastRow semantics).
+__global__ void dedup_mark_last_kernel(
+    const int32_t* __restrict__ series,
+    const int64_t* __restrict__ ts,
+    int64_t n, bool* __restrict__ keep) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    keep[i] = (i == n - 1) || (series[i] != series[i + 1]) || (ts[i] != ts[i + 1]);
+  }
+}
+
+// ---------------------------------------------------------------- launchers
+
+static inline int grid_for(int64_t n, int block) {
+  int64_t g = (n + block - 1) / block;
+  if (g > 2048) g = 2048;  // G11: cap + grid-stride
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void launch_ts_bucket_agg(
+    const int64_t* ts, const int32_t* series, const double* fields,
+    int64_t field_stride, const int32_t* field_idx, int nf,
+    const int32_t* slot_lut, int lut_size,
+    int64_t ts_lo, int64_t ts_hi, int64_t origin, int64_t bucket_ms,
+    int n_slots, int n_buckets, int64_t n,
+    double* out_sum, unsigned long long* out_cnt,
+    unsigned long long* out_min, unsigned long long* out_max,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(ts_bucket_agg_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      ts, series, fields, field_stride, field_idx, nf, slot_lut, lut_size,
+      ts_lo, ts_hi, origin, bucket_ms, n_slots, n_buckets, n,
+      out_sum, out_cnt, out_min, out_max);
+}
+
+void launch_decode_minmax(
+    const unsigned long long* minkey, const unsigned long long* maxkey,
+    const unsigned long long* cnt, double* out_min, double* out_max,
+    int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(decode_minmax_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      minkey, maxkey, cnt, out_min, out_max, n);
+}
+
+void launch_filter_series_time(
+    const int64_t* ts, const int32_t* series, const int32_t* slot_lut,
+    int lut_size, int64_t ts_lo, int64_t ts_hi, int64_t n, bool* keep,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(filter_series_time_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      ts, series, slot_lut, lut_size, ts_lo, ts_hi, n, keep);
+}
+
+void launch_dedup_mark_last(
+    const int32_t* series, const int64_t* ts, int64_t n, bool* keep,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(dedup_mark_last_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      series, ts, n, keep);
+}
+
+}  // namespace gdb_hip

@@ -1,0 +1,220 @@
+"""MitoEngine: tables → regions on one device, WAL, flush scheduling.
+
+Reference parity: src/mito2/src/engine.rs + worker.rs (WorkerGroup /
+RegionWorker event loops) + src/catalog. MI355X redesign: one engine
+instance per GPU (one process per GPU under torch.distributed); regions of
+a table are sharded across engines by the partition rule (parallel/
+partition.py). Background flush runs on a worker thread; WAL commits group
+all regions of a write batch (reference handle_write.rs:611 batched
+write_wal).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import queue
+import threading
+from dataclasses import dataclass, field
+
+import numpy as np
+
+from greptimedb_amd.engine.region import Region
+from greptimedb_amd.engine.wal import Wal, decode_batch, encode_batch
+from greptimedb_amd.models.schema import TableSchema, region_id as make_region_id
+from greptimedb_amd.utils.errors import TableAlreadyExists, TableNotFound
+
+
+@dataclass
+class EngineConfig:
+    data_dir: str = "./greptime_data"
+    device: str = "cpu"
+    flush_bytes: int = 512 << 20        # per-region memtable flush threshold
+    wal_sync: bool = False              # fdatasync per commit
+    wal_segment_bytes: int = 256 << 20
+    default_regions: int = 4
+    background_flush: bool = True
+
+
+@dataclass
+class TableState:
+    schema: TableSchema
+    regions: list = field(default_factory=list)
+    append_mode: bool = False
+
+
+class MitoEngine:
+    def __init__(self, config: EngineConfig):
+        self.config = config
+        os.makedirs(config.data_dir, exist_ok=True)
+        self.tables: dict[str, TableState] = {}
+        self.next_table_id = 1024
+        self._catalog_path = os.path.join(config.data_dir, "catalog.json")
+        self.wal = Wal(os.path.join(config.data_dir, "wal"),
+                       segment_bytes=config.wal_segment_bytes,
+                       sync_on_commit=config.wal_sync)
+        self._flush_q: queue.Queue = queue.Queue()
+        self._flusher = None
+        self._load_catalog()
+        self._replay_wal()
+        if config.background_flush:
+            self._flusher = threading.Thread(target=self._flush_loop, daemon=True)
+            self._flusher.start()
+
+    # ------------------------------------------------------------- catalog
+
+    def _load_catalog(self):
+        if not os.path.exists(self._catalog_path):
+            return
+        with open(self._catalog_path) as f:
+            cat = json.load(f)
+        self.next_table_id = cat["next_table_id"]
+        for td in cat["tables"]:
+            schema = TableSchema.from_dict(td["schema"])
+            st = TableState(schema=schema, append_mode=td["append_mode"])
+            for rn in range(td["n_regions"]):
+                rid = make_region_id(schema.table_id, rn)
+                rdir = os.path.join(self.config.data_dir, "region", str(rid))
+                st.regions.append(Region(rid, schema, rdir, device=self.config.device,
+                                         append_mode=td["append_mode"]))
+            self.tables[schema.name] = st
+
+    def _save_catalog(self):
+        cat = {
+            "next_table_id": self.next_table_id,
+            "tables": [
+                {
+                    "schema": st.schema.to_dict(),
+                    "n_regions": len(st.regions),
+                    "append_mode": st.append_mode,
+                }
+                for st in self.tables.values()
+            ],
+        }
+        tmp = self._catalog_path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(cat, f)
+        os.rename(tmp, self._catalog_path)
+
+    def create_table(self, schema: TableSchema, n_regions: int | None = None,
+                     append_mode: bool = False, if_not_exists: bool = False) -> TableState:
+        if schema.name in self.tables:
+            if if_not_exists:
+                return self.tables[schema.name]
+            raise TableAlreadyExists(schema.name)
+        if schema.table_id == 0:
+            schema.table_id = self.next_table_id
+        self.next_table_id = max(self.next_table_id, schema.table_id) + 1
+        n_regions = n_regions or self.config.default_regions
+        st = TableState(schema=schema, append_mode=append_mode)
+        for rn in range(n_regions):
+            rid = make_region_id(schema.table_id, rn)
+            rdir = os.path.join(self.config.data_dir, "region", str(rid))
+            st.regions.append(Region(rid, schema, rdir, device=self.config.device,
+                                     append_mode=append_mode))
+        self.tables[schema.name] = st
+        self._save_catalog()
+        return st
+
+    def drop_table(self, name: str):
+        st = self.tables.pop(name, None)
+        if st is None:
+            raise TableNotFound(name)
+        self._save_catalog()
+
+    def table(self, name: str) -> TableState:
+        try:
+            return self.tables[name]
+        except KeyError:
+            raise TableNotFound(name) from None
+
+    # ------------------------------------------------------------- writes
+
+    def write_region(self, table: TableState, region_idx: int,
+                     series_codes: np.ndarray, ts_ms: np.ndarray,
+                     fields: np.ndarray, new_series: list[tuple[int, bytes]],
+                     durable: bool = True) -> int:
+        """WAL-append + memtable-append one region's slice of a write batch.
+        Caller must call `commit_wal()` after all regions of the batch
+        (group commit — durability boundary)."""
+        region: Region = table.regions[region_idx]
+        seq = 0
+        if durable:
+            payload = encode_batch(series_codes, ts_ms, fields,
+                                   region.field_names, new_series)
+            seq = self.wal.append(region.region_id, payload)
+        region.append(series_codes, ts_ms, fields, seq)
+        return seq
+
+    def commit_wal(self):
+        self.wal.commit()
+
+    def maybe_flush(self):
+        for st in self.tables.values():
+            for region in st.regions:
+                if region.should_flush(self.config.flush_bytes):
+                    if self._flusher is not None:
+                        self._flush_q.put(region)
+                    else:
+                        self._flush_region(region)
+
+    def flush_all(self, wait: bool = True):
+        for st in self.tables.values():
+            for region in st.regions:
+                self._flush_region(region)
+
+    def _flush_region(self, region: Region):
+        region.flush()
+        self._purge_wal()
+
+    def _flush_loop(self):
+        while True:
+            region = self._flush_q.get()
+            if region is None:
+                return
+            try:
+                self._flush_region(region)
+            except Exception:  # pragma: no cover
+                import traceback
+                traceback.print_exc()
+
+    def _purge_wal(self):
+        flushed = [r.flushed_seq for st in self.tables.values() for r in st.regions]
+        if flushed:
+            self.wal.purge_before(min(flushed))
+
+    # ------------------------------------------------------------- recovery
+
+    def _replay_wal(self):
+        """Re-apply WAL entries above each region's flushed seq (reference:
+        region/opener.rs:483 replay_memtable)."""
+        regions = {r.region_id: (st, i)
+                   for st in self.tables.values()
+                   for i, r in enumerate(st.regions)}
+        for _seg, rid, seq, payload in self.wal.replay():
+            hit = regions.get(rid)
+            if hit is None:
+                continue
+            st, idx = hit
+            region: Region = st.regions[idx]
+            if seq <= region.flushed_seq:
+                continue
+            series, ts, fields, fnames, new_series = decode_batch(payload)
+            # series codes are stable via the region series log (loaded at
+            # open); payload new_series is belt-and-braces for a lost log tail
+            for _code, pk in new_series:
+                region.series.add_encoded(pk)
+            # field order may differ from region's if schema evolved; map names
+            if fnames != region.field_names:
+                fmap = {fn: i for i, fn in enumerate(fnames)}
+                out = np.full((len(region.field_names), fields.shape[1]), np.nan)
+                for i, fn in enumerate(region.field_names):
+                    if fn in fmap:
+                        out[i] = fields[fmap[fn]]
+                fields = out
+            region.append(series, ts, fields, seq)
+
+    def close(self):
+        if self._flusher is not None:
+            self._flush_q.put(None)
+        self.wal.close()

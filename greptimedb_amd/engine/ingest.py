@@ -1,0 +1,172 @@
+"""Ingestor: wire bytes → parsed columnar batch → partition → WAL → memtable.
+
+Reference parity: src/operator/src/insert.rs (Inserter::handle_row_inserts:
+auto-create tables on demand :562, partition split :389-496) + the influx
+decode in src/servers. The hot path is vectorized: the C++ parser interns
+tagsets to dense sids; routing is two numpy LUT gathers (sid → flat region,
+sid → local code); per-region slices go to WAL then GPU memtable with ONE
+group commit per ingest batch (durability boundary, handle_write.rs:611).
+
+Partitioning: hash(encoded pk) % n_regions (the default when no PARTITION ON
+clause — reference uses MultiDimPartitionRule; explicit partition rules live
+in parallel/partition.py).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from greptimedb_amd import _native
+from greptimedb_amd.engine.engine import MitoEngine, TableState
+from greptimedb_amd.engine.series import tsid_hash
+from greptimedb_amd.models.schema import ColumnSchema, DataType, SemanticType, TableSchema
+
+
+def parse_tagset(key: bytes) -> tuple[str, list[tuple[str, str]]]:
+    """'cpu,hostname=h1,region=r' → ('cpu', [(tag, value)...])."""
+    parts = key.decode().split(",")
+    measurement = parts[0]
+    tags = []
+    for p in parts[1:]:
+        if "=" in p:
+            k, v = p.split("=", 1)
+            tags.append((k, v))
+    return measurement, tags
+
+
+class Ingestor:
+    def __init__(self, engine: MitoEngine, default_regions: int | None = None,
+                 append_mode: bool = True, durable: bool = True):
+        self.engine = engine
+        self.parser = _native.LineParser()
+        self.append_mode = append_mode
+        self.durable = durable
+        self.default_regions = default_regions or engine.config.default_regions
+        # flat routing state, indexed by parser sid
+        self._cap = 1024
+        self.sid_region = np.full(self._cap, -1, dtype=np.int32)   # flat region idx
+        self.sid_local = np.full(self._cap, -1, dtype=np.int32)    # region-local code
+        self.flat_regions: list = []      # (TableState, region_idx)
+        self._region_key: dict = {}       # (table_name, region_idx) -> flat idx
+        self._table_field_map: dict = {}  # table name -> cached np map + src len
+        self.rows_ingested = 0
+        # re-register any series known to existing tables (restart path):
+        # parser starts empty; sids are assigned fresh per process, routing
+        # fills lazily as tagsets arrive.
+
+    def _flat_region(self, st: TableState, region_idx: int) -> int:
+        key = (st.schema.name, region_idx)
+        idx = self._region_key.get(key)
+        if idx is None:
+            idx = len(self.flat_regions)
+            self.flat_regions.append((st, region_idx))
+            self._region_key[key] = idx
+        return idx
+
+    def _grow(self, need: int):
+        cap = self._cap
+        while cap <= need:
+            cap *= 2
+        for name in ("sid_region", "sid_local"):
+            a = getattr(self, name)
+            na = np.full(cap, -1, dtype=np.int32)
+            na[: len(a)] = a
+            setattr(self, name, na)
+        self._cap = cap
+
+    def _register_tagset(self, sid: int, key: bytes):
+        """Cold path: new series — resolve table (auto-create), partition,
+        register the series in its region."""
+        measurement, tags = parse_tagset(key)
+        try:
+            st = self.engine.table(measurement)
+        except Exception:
+            st = self._auto_create_table(measurement, tags)
+        # unify tag order with table primary key; missing tags = None
+        tag_map = dict(tags)
+        tag_tuple = tuple(tag_map.get(t.name) for t in st.schema.tag_columns)
+        # partition by pk hash
+        from greptimedb_amd.engine import pk_codec
+        pk = pk_codec.encode_pk(tag_tuple)
+        region_idx = tsid_hash(pk) % len(st.regions)
+        region = st.regions[region_idx]
+        local = region.register_series(tag_tuple)
+        if sid >= self._cap:
+            self._grow(sid)
+        self.sid_region[sid] = self._flat_region(st, region_idx)
+        self.sid_local[sid] = local
+
+    def _auto_create_table(self, measurement: str, tags: list[tuple[str, str]]) -> TableState:
+        cols = []
+        cid = 0
+        for t, _ in tags:
+            cols.append(ColumnSchema(t, DataType.STRING, SemanticType.TAG, cid)); cid += 1
+        cols.append(ColumnSchema("ts", DataType.TIMESTAMP_MS, SemanticType.TIMESTAMP, cid))
+        schema = TableSchema(name=measurement, columns=cols,
+                             primary_key=[t for t, _ in tags])
+        return self.engine.create_table(schema, n_regions=self.default_regions,
+                                        append_mode=self.append_mode, if_not_exists=True)
+
+    def _field_map(self, st: TableState, parser_fields: list[str],
+                   fields_mat: np.ndarray, rows: np.ndarray) -> np.ndarray:
+        """Map table field order → parser field row index (-1 = absent),
+        auto-adding table fields that now have data."""
+        region0 = st.regions[0]
+        name = st.schema.name
+        cached = self._table_field_map.get(name)
+        if cached is not None and cached[1] == len(parser_fields) and \
+                len(cached[0]) == len(region0.field_names):
+            return cached[0]
+        # find parser fields with any non-NaN data for this table's rows
+        present = [fn for i, fn in enumerate(parser_fields)
+                   if not np.isnan(fields_mat[i][rows]).all()]
+        new = [fn for fn in present if fn not in region0.field_names]
+        if new:
+            for r in st.regions:
+                r.ensure_fields(new)
+        pmap = {fn: i for i, fn in enumerate(parser_fields)}
+        m = np.array([pmap.get(fn, -1) for fn in region0.field_names], dtype=np.int64)
+        self._table_field_map[name] = (m, len(parser_fields))
+        return m
+
+    def ingest_lines(self, data: bytes) -> int:
+        """Parse + route + WAL + memtable-append one wire batch. Returns rows."""
+        series, ts_ns, fields, new_tagsets = self.parser.parse(data)
+        n = len(series)
+        if n == 0:
+            return 0
+        for sid, key in new_tagsets:
+            self._register_tagset(sid, key)
+        parser_fields = self.parser.field_names()
+        fields_mat = np.stack([fields[fn] for fn in parser_fields]) if parser_fields \
+            else np.zeros((0, n))
+        ts_ms = ts_ns // 1_000_000
+
+        region_of = self.sid_region[series]
+        local = self.sid_local[series]
+        order = np.argsort(region_of, kind="stable")
+        region_sorted = region_of[order]
+        bounds = np.flatnonzero(np.diff(region_sorted)) + 1
+        starts = np.concatenate(([0], bounds))
+        ends = np.concatenate((bounds, [n]))
+        engine = self.engine
+        for s, e in zip(starts, ends):
+            flat = int(region_sorted[s])
+            st, region_idx = self.flat_regions[flat]
+            rows = order[s:e]
+            fmap = self._field_map(st, parser_fields, fields_mat, rows)
+            nf = len(fmap)
+            out = np.empty((nf, e - s), dtype=np.float64)
+            for i, src in enumerate(fmap):
+                if src >= 0:
+                    out[i] = fields_mat[src][rows]
+                else:
+                    out[i] = np.nan
+            engine.write_region(st, region_idx,
+                                local[rows].astype(np.int32),
+                                ts_ms[rows], out, [], durable=self.durable)
+        if self.durable:
+            engine.commit_wal()
+        engine.maybe_flush()
+        self.rows_ingested += n
+        return n

@@ -1,0 +1,209 @@
+"""Region: one shard of a table — GPU memtable + SSTs + manifest.
+
+Reference parity: src/mito2 region (region/version.rs VersionControl,
+flush.rs, read/scan_region.rs). MI355X redesign: all row data for a region
+lives on ONE GPU (the region↔GPU mapping is the P1 sharding axis,
+SURVEY.md §2.8); flush sorts/dedups on device, writes the parquet SST from
+a host copy, and keeps the sorted device tensors as the scan cache.
+"""
+
+from __future__ import annotations
+
+import os
+import struct
+import threading
+
+import numpy as np
+import torch
+
+from greptimedb_amd.engine import sst as sst_mod
+from greptimedb_amd.engine.manifest import Manifest
+from greptimedb_amd.engine.memtable import Memtable
+from greptimedb_amd.engine.series import SeriesIndex
+from greptimedb_amd.models.schema import TableSchema
+from greptimedb_amd.ops import dedup_mark_last
+
+
+class ScanSource:
+    """One scan input: device column set + field-name → row mapping."""
+
+    __slots__ = ("ts", "series", "fields", "n", "field_pos", "sorted")
+
+    def __init__(self, ts, series, fields, n, field_pos, sorted):
+        self.ts = ts
+        self.series = series
+        self.fields = fields
+        self.n = n
+        self.field_pos = field_pos
+        self.sorted = sorted
+
+
+class Region:
+    def __init__(self, region_id: int, schema: TableSchema, dir: str,
+                 device: str = "cpu", append_mode: bool = False):
+        self.region_id = region_id
+        self.schema = schema
+        self.dir = dir
+        self.device = device
+        self.append_mode = append_mode
+        self.field_names = [c.name for c in schema.field_columns]
+        self.series = SeriesIndex([c.name for c in schema.tag_columns])
+        os.makedirs(os.path.join(dir, "sst"), exist_ok=True)
+        self.manifest = Manifest(os.path.join(dir, "manifest"))
+        self.memtable = Memtable(len(self.field_names), device=device)
+        self.sst_cache: dict[str, sst_mod.SstBatch] = {}
+        self.flushed_seq = self.manifest.flushed_seq
+        self.last_seq = self.flushed_seq
+        self.lock = threading.Lock()
+        # Series code assignment must be stable across restarts (GPU columns
+        # and SST caches store dense codes): an append-only series log is the
+        # authoritative code order. Loaded BEFORE SSTs.
+        self._series_log_path = os.path.join(dir, "series.log")
+        self._load_series_log()
+        self._series_log = open(self._series_log_path, "ab")
+        self._load_ssts()
+
+    def _load_series_log(self):
+        if not os.path.exists(self._series_log_path):
+            return
+        with open(self._series_log_path, "rb") as f:
+            buf = f.read()
+        off = 0
+        while off + 4 <= len(buf):
+            (ln,) = struct.unpack_from("<I", buf, off)
+            if off + 4 + ln > len(buf):
+                break  # torn tail
+            self.series.add_encoded(buf[off + 4: off + 4 + ln])
+            off += 4 + ln
+
+    def register_series(self, tags: tuple) -> int:
+        """Get-or-create a local series code; new codes are logged durably."""
+        prev = len(self.series)
+        code = self.series.get_or_create(tags)
+        if code >= prev:
+            pk = self.series.pks[code]
+            self._series_log.write(struct.pack("<I", len(pk)) + pk)
+            self._series_log.flush()
+        return code
+
+    # ---------------------------------------------------------------- open
+
+    def _load_ssts(self):
+        """Open path: load manifest SSTs into the device cache."""
+        for fid, meta in self.manifest.files.items():
+            path = os.path.join(self.dir, "sst", f"{fid}.parquet")
+            if not os.path.exists(path):
+                continue
+            dict_values, indices, ts, fields, seq = sst_mod.read_sst(
+                path, self.schema, self.field_names)
+            remap = np.array([self.series.add_encoded(pk) for pk in dict_values],
+                             dtype=np.int32)
+            codes = remap[indices]
+            dev = self.device
+            t_ts = torch.as_tensor(ts).to(dev)
+            t_se = torch.as_tensor(codes).to(dev)
+            t_f = torch.as_tensor(np.ascontiguousarray(fields)).to(dev)
+            t_seq = torch.as_tensor(seq).to(dev)
+            # re-sort by (series, ts, seq): codes may differ from write-time order
+            ord1 = torch.argsort(t_seq, stable=True)
+            ord2 = torch.argsort(t_ts[ord1], stable=True)
+            perm = ord1[ord2]
+            ord3 = torch.argsort(t_se[perm], stable=True)
+            perm = perm[ord3]
+            self.sst_cache[fid] = sst_mod.SstBatch(
+                t_ts[perm].contiguous(), t_se[perm].contiguous(),
+                t_f[:, perm].contiguous(), t_seq[perm].contiguous(),
+                int(meta["min_ts"]), int(meta["max_ts"]),
+                list(self.field_names))
+
+    # ---------------------------------------------------------------- write
+
+    def append(self, series_codes: np.ndarray, ts_ms: np.ndarray,
+               fields: np.ndarray, last_seq: int):
+        with self.lock:
+            self.memtable.append(series_codes, ts_ms, fields)
+            self.last_seq = max(self.last_seq, last_seq)
+
+    def should_flush(self, limit_bytes: int) -> bool:
+        return self.memtable.bytes_used >= limit_bytes
+
+    def flush(self):
+        """Sort + dedup the memtable on device, write an SST, swap memtable."""
+        with self.lock:
+            mem = self.memtable
+            if mem.len == 0:
+                return None
+            n = mem.len
+            flush_seq = self.last_seq
+            prev_flushed = self.flushed_seq
+            flush_field_names = list(self.field_names[: mem.nf])
+            self.memtable = Memtable(len(self.field_names), device=self.device,
+                                     cap=max(mem.cap, 1 << 16))
+        ts, se, fields, perm = mem.sorted_view()
+        if not self.append_mode:
+            keep = dedup_mark_last(se, ts)
+            idx = keep.nonzero(as_tuple=True)[0]
+            ts, se, fields, perm = ts[idx], se[idx], fields[:, idx], perm[idx]
+        ts_h = ts.cpu().numpy()
+        se_h = se.cpu().numpy()
+        f_h = fields.cpu().numpy()
+        # per-row sequence ≈ prev flushed seq + arrival index in this memtable
+        # (row order is the recency order LastRow dedup relies on)
+        seq_h = perm.cpu().numpy().astype(np.int64) + prev_flushed
+        fid = sst_mod.new_file_id()
+        path = os.path.join(self.dir, "sst", f"{fid}.parquet")
+        meta = sst_mod.write_sst(path, self.schema, self.series.pks,
+                                 se_h, ts_h, f_h, seq_h, flush_field_names)
+        meta.seq_max = flush_seq
+        self.manifest.commit({
+            "kind": "edit",
+            "files_to_add": [meta.to_dict()],
+            "files_to_remove": [],
+            "flushed_seq": flush_seq,
+        })
+        self.sst_cache[fid] = sst_mod.SstBatch(
+            ts.contiguous(), se.contiguous(), fields.contiguous(), None,
+            meta.min_ts, meta.max_ts, flush_field_names)
+        self.flushed_seq = flush_seq
+        return meta
+
+    # ---------------------------------------------------------------- schema
+
+    def ensure_fields(self, names: list[str]):
+        """Auto-ALTER: add new field columns (reference insert.rs:562
+        create_or_alter tables on demand)."""
+        new = [n for n in names if n not in self.field_names]
+        if not new:
+            return
+        with self.lock:
+            self.field_names.extend(new)
+            self.memtable.add_fields(len(new))
+
+    # ---------------------------------------------------------------- scan
+
+    def scan_sources(self, ts_lo: int | None = None, ts_hi: int | None = None):
+        """ScanSource list overlapping the time range: SST cache batches
+        (time-pruned, reference scan_region.rs:887) + the active memtable."""
+        out = []
+        for batch in self.sst_cache.values():
+            if ts_lo is not None and batch.max_ts < ts_lo:
+                continue
+            if ts_hi is not None and batch.min_ts >= ts_hi:
+                continue
+            out.append(ScanSource(batch.ts, batch.series, batch.fields, batch.n,
+                                  {fn: i for i, fn in enumerate(batch.field_names)},
+                                  sorted=True))
+        with self.lock:
+            mem = self.memtable
+            n = mem.len
+        if n > 0:
+            if not (ts_lo is not None and mem.max_ts is not None and mem.max_ts < ts_lo) and \
+               not (ts_hi is not None and mem.min_ts is not None and mem.min_ts >= ts_hi):
+                out.append(ScanSource(mem.ts[:n], mem.series[:n], mem.fields, n,
+                                      {fn: i for i, fn in enumerate(self.field_names)},
+                                      sorted=False))
+        return out
+
+    @property
+    def num_rows(self) -> int:
+        return self.memtable.len + sum(b.n for b in self.sst_cache.values())

@@ -1,0 +1,81 @@
+"""Per-region series index: local dense codes ↔ tag values ↔ encoded pk.
+
+Reference parity: mito2 memtable/time_series.rs keys its BTreeMap by encoded
+primary key and src/mito2/src/series_index.rs tracks series; the metric
+engine hashes labels to __tsid (row_modifier.rs:98). Here every region keeps
+a dense local code per series (what the GPU columns store) plus:
+  - encoded pk bytes (SST __primary_key dictionary values),
+  - decoded tag value tuple (for group-by output),
+  - per-tag inverted map value → [codes] (host-side tag predicate probe —
+    the K13 analog for tag equality/IN filters).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from greptimedb_amd.engine import pk_codec
+
+
+def tsid_hash(pk: bytes) -> int:
+    """FNV-1a 64 of the encoded pk (stable __tsid; reference uses a label
+    hash in metric-engine row_modifier.rs:98)."""
+    h = 0xCBF29CE484222325
+    for b in pk:
+        h = ((h ^ b) * 0x100000001B3) & 0xFFFFFFFFFFFFFFFF
+    return h
+
+
+class SeriesIndex:
+    def __init__(self, tag_names: list[str]):
+        self.tag_names = list(tag_names)
+        self.pk_to_code: dict[bytes, int] = {}
+        self.pks: list[bytes] = []
+        self.tag_values: list[tuple] = []
+        # tag name -> value -> list[int] codes
+        self.inverted: dict[str, dict[str, list[int]]] = {t: {} for t in tag_names}
+
+    def __len__(self) -> int:
+        return len(self.pks)
+
+    def get_or_create(self, tags: tuple) -> int:
+        pk = pk_codec.encode_pk(tags)
+        code = self.pk_to_code.get(pk)
+        if code is None:
+            code = self.add(pk, tags)
+        return code
+
+    def add(self, pk: bytes, tags: tuple) -> int:
+        code = len(self.pks)
+        self.pk_to_code[pk] = code
+        self.pks.append(pk)
+        self.tag_values.append(tags)
+        for t, v in zip(self.tag_names, tags):
+            if v is not None:
+                self.inverted[t].setdefault(v, []).append(code)
+        return code
+
+    def add_encoded(self, pk: bytes) -> int:
+        """Register a series seen only as encoded pk (SST load path)."""
+        code = self.pk_to_code.get(pk)
+        if code is not None:
+            return code
+        tags = pk_codec.decode_pk(pk, len(self.tag_names))
+        return self.add(pk, tags)
+
+    # ---------------- tag predicate → slot LUT helpers ----------------
+
+    def codes_for_eq(self, tag: str, value: str) -> list[int]:
+        return self.inverted.get(tag, {}).get(value, [])
+
+    def codes_for_in(self, tag: str, values: list[str]) -> list[int]:
+        out: list[int] = []
+        inv = self.inverted.get(tag, {})
+        for v in values:
+            out.extend(inv.get(v, []))
+        return out
+
+    def tag_array(self, tag: str) -> np.ndarray:
+        """Object array of this tag's value per code (group-by output)."""
+        i = self.tag_names.index(tag)
+        return np.array([t[i] for t in self.tag_values], dtype=object)

@@ -1,0 +1,129 @@
+"""SST layer: mito2-format Parquet files + GPU-resident batch cache.
+
+Reference parity: src/mito2/src/sst/parquet/ — primary-key format
+`field..., time index, __primary_key (dict<u32,binary>), __sequence (u64),
+__op_type (u8)` (format.rs:15-27), row group 102400 rows (parquet.rs:56),
+ZSTD compression (parquet.rs:607). Files are written sorted by
+(__primary_key, ts, seq) like the reference flush/compaction output.
+
+MI355X design: the flush path already has sorted device tensors in HBM, so
+every flushed SST keeps a device-resident `SstBatch` (288 GB per GPU makes
+"keep everything hot" the default caching policy; the parquet file is the
+durable/spill copy). Scans therefore hit HBM, not parquet decode — the
+reference's page-cache/row-group-cache (mito2 cache.rs) collapses into this
+one structure.
+"""
+
+from __future__ import annotations
+
+import os
+import uuid
+from dataclasses import dataclass
+
+import numpy as np
+import pyarrow as pa
+import pyarrow.parquet as pq
+import torch
+
+from greptimedb_amd.models.schema import TableSchema
+
+ROW_GROUP_SIZE = 102400
+OP_PUT = 1
+
+
+@dataclass
+class SstMeta:
+    file_id: str
+    level: int
+    min_ts: int
+    max_ts: int
+    num_rows: int
+    file_size: int
+    seq_max: int
+
+    def to_dict(self):
+        return self.__dict__.copy()
+
+    @staticmethod
+    def from_dict(d):
+        return SstMeta(**d)
+
+
+class SstBatch:
+    """Device-resident, (series, ts, seq)-sorted columns of one SST."""
+
+    def __init__(self, ts: torch.Tensor, series: torch.Tensor, fields: torch.Tensor,
+                 seq: torch.Tensor | None, min_ts: int, max_ts: int,
+                 field_names: list[str]):
+        self.ts = ts
+        self.series = series
+        self.fields = fields  # [nf, n]
+        self.seq = seq
+        self.min_ts = min_ts
+        self.max_ts = max_ts
+        self.field_names = list(field_names)
+
+    @property
+    def n(self) -> int:
+        return self.ts.numel()
+
+
+def write_sst(path: str, schema: TableSchema, pks: list[bytes],
+              series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
+              seq: np.ndarray, field_names: list[str]) -> SstMeta:
+    """Write one mito2-format parquet SST. Inputs are host arrays sorted by
+    (series, ts); `pks[code]` gives the encoded primary key per local code."""
+    n = len(ts_ms)
+    # compact dictionary: unique codes in appearance order
+    uniq, inv = np.unique(series, return_inverse=True)
+    dict_values = pa.array([pks[int(c)] for c in uniq], type=pa.binary())
+    pk_col = pa.DictionaryArray.from_arrays(pa.array(inv.astype(np.uint32), type=pa.uint32()), dict_values)
+
+    cols, names = [], []
+    for i, fn in enumerate(field_names):
+        cols.append(pa.array(fields[i], type=pa.float64()))
+        names.append(fn)
+    cols.append(pa.array(ts_ms, type=pa.timestamp("ms")))
+    names.append(schema.time_index.name)
+    cols.append(pk_col)
+    names.append("__primary_key")
+    cols.append(pa.array(seq.astype(np.uint64), type=pa.uint64()))
+    names.append("__sequence")
+    cols.append(pa.array(np.full(n, OP_PUT, dtype=np.uint8), type=pa.uint8()))
+    names.append("__op_type")
+
+    table = pa.Table.from_arrays(cols, names=names)
+    pq.write_table(table, path, row_group_size=ROW_GROUP_SIZE, compression="zstd")
+    return SstMeta(
+        file_id=os.path.basename(path).replace(".parquet", ""),
+        level=0,
+        min_ts=int(ts_ms.min()) if n else 0,
+        max_ts=int(ts_ms.max()) if n else 0,
+        num_rows=n,
+        file_size=os.path.getsize(path),
+        seq_max=int(seq.max()) if n else 0,
+    )
+
+
+def new_file_id() -> str:
+    return uuid.uuid4().hex
+
+
+def read_sst(path: str, schema: TableSchema, field_names: list[str]):
+    """Read an SST back to host arrays: (pk_list per row-code, series codes
+    i32 (dictionary indices), ts_ms i64, fields f64[nf, n], seq u64).
+    The caller remaps dictionary indices into region-local codes."""
+    t = pq.read_table(path)
+    pk = t.column("__primary_key").combine_chunks()
+    if isinstance(pk, pa.ChunkedArray):
+        pk = pk.chunk(0)
+    dict_values = [v.as_py() for v in pk.dictionary]
+    indices = pk.indices.to_numpy(zero_copy_only=False).astype(np.int32)
+    ts = t.column(schema.time_index.name).cast(pa.int64()).to_numpy(zero_copy_only=False)
+    fields = np.stack([
+        t.column(fn).to_numpy(zero_copy_only=False) if fn in t.column_names
+        else np.full(len(ts), np.nan)
+        for fn in field_names
+    ]) if field_names else np.zeros((0, len(ts)))
+    seq = t.column("__sequence").to_numpy(zero_copy_only=False).astype(np.int64)
+    return dict_values, indices, ts, fields, seq

@@ -1,0 +1,72 @@
+"""CPU reference implementations of the HIP kernels (plain PyTorch, fp64).
+
+These define the semantics the gfx950 kernels in csrc/kernels.hip must match;
+GPU numerics tests (tests/test_ops_gpu.py) compare the two. They also serve
+as the execution path on CPU-only hosts (tests, CI).
+"""
+
+from __future__ import annotations
+
+import torch
+
+
+def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
+                  origin, bucket_ms, n_slots, n_buckets):
+    """Fused filter + time-bucket aggregate.
+
+    ts: i64[n] ms, series: i32[n], fields: f64[nf_total, >=n],
+    field_idx: i32[nf], slot_lut: i32[lut]. Returns (sum, count, min, max),
+    each [nf, n_slots, n_buckets]; min/max are NaN where count==0.
+    NaN field values count as nulls and are skipped.
+    """
+    n = ts.numel()
+    keep = (ts >= ts_lo) & (ts < ts_hi)
+    s = series.long().clamp(0, max(slot_lut.numel() - 1, 0))
+    in_lut = (series >= 0) & (series < slot_lut.numel())
+    slot = torch.where(in_lut, slot_lut[s].long(), torch.full_like(s, -1))
+    keep &= slot >= 0
+    bucket = torch.div(ts - origin, bucket_ms, rounding_mode="floor")
+    keep &= (bucket >= 0) & (bucket < n_buckets)
+
+    nf = field_idx.numel()
+    out_sum = torch.zeros(nf, n_slots, n_buckets, dtype=torch.float64)
+    out_cnt = torch.zeros(nf, n_slots, n_buckets, dtype=torch.int64)
+    out_min = torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64)
+    out_max = torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64)
+
+    idx = keep.nonzero(as_tuple=True)[0]
+    if idx.numel() == 0:
+        return out_sum, out_cnt, out_min, out_max
+    cell = slot[idx] * n_buckets + bucket[idx]
+    for f in range(nf):
+        v = fields[int(field_idx[f])][:n][idx]
+        ok = ~torch.isnan(v)
+        c = cell[ok]
+        vv = v[ok]
+        out_sum[f].view(-1).index_add_(0, c, vv)
+        out_cnt[f].view(-1).index_add_(0, c, torch.ones_like(c))
+        mn = out_min[f].view(-1)
+        mx = out_max[f].view(-1)
+        mn.index_reduce_(0, c, vv, "amin", include_self=False)
+        mx.index_reduce_(0, c, vv, "amax", include_self=False)
+    # index_reduce with include_self=False leaves untouched cells at init NaN
+    return out_sum, out_cnt, out_min, out_max
+
+
+def filter_series_time(ts, series, slot_lut, ts_lo, ts_hi):
+    keep = (ts >= ts_lo) & (ts < ts_hi)
+    if slot_lut is not None and slot_lut.numel() > 0:
+        in_lut = (series >= 0) & (series < slot_lut.numel())
+        s = series.long().clamp(0, slot_lut.numel() - 1)
+        keep &= in_lut & (slot_lut[s] >= 0)
+    return keep
+
+
+def dedup_mark_last(series, ts):
+    """keep[i] ⇔ row i is the last of its (series, ts) group (sorted input)."""
+    n = ts.numel()
+    if n == 0:
+        return torch.zeros(0, dtype=torch.bool)
+    keep = torch.ones(n, dtype=torch.bool)
+    keep[:-1] = (series[:-1] != series[1:]) | (ts[:-1] != ts[1:])
+    return keep

@@ -1,0 +1,58 @@
+"""Dispatch layer: GPU tensors → _hip_ops (gfx950 kernels), CPU → cpu_ref."""
+
+from __future__ import annotations
+
+import torch
+
+from greptimedb_amd.ops import cpu_ref
+from greptimedb_amd.utils.errors import NativeExtensionMissing
+
+try:
+    from greptimedb_amd import _hip_ops  # built by setup.py (in-tree .so)
+
+    _HIP_OPS = _hip_ops
+    _HIP_IMPORT_ERROR = None
+except Exception as e:  # pragma: no cover - exercised only when build broken
+    _HIP_OPS = None
+    _HIP_IMPORT_ERROR = e
+
+
+def hip_ops_available() -> bool:
+    return _HIP_OPS is not None
+
+
+def _require_hip():
+    if _HIP_OPS is None:
+        raise NativeExtensionMissing(
+            "greptimedb_amd._hip_ops is not built; refusing to run the GPU path "
+            "on eager fallback. Run `python setup.py build_ext --inplace`. "
+            f"(import error: {_HIP_IMPORT_ERROR})"
+        )
+    return _HIP_OPS
+
+
+def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
+                  origin, bucket_ms, n_slots, n_buckets):
+    if ts.is_cuda:
+        ops = _require_hip()
+        s, c, mn, mx = ops.ts_bucket_agg(
+            ts, series, fields, field_idx, slot_lut,
+            int(ts_lo), int(ts_hi), int(origin), int(bucket_ms),
+            int(n_slots), int(n_buckets))
+        return s, c, mn, mx
+    return cpu_ref.ts_bucket_agg(ts, series, fields, field_idx, slot_lut,
+                                 ts_lo, ts_hi, origin, bucket_ms, n_slots, n_buckets)
+
+
+def filter_series_time(ts, series, slot_lut, ts_lo, ts_hi):
+    if ts.is_cuda:
+        ops = _require_hip()
+        lut = slot_lut if slot_lut is not None else torch.empty(0, dtype=torch.int32, device=ts.device)
+        return ops.filter_series_time(ts, series, lut, int(ts_lo), int(ts_hi))
+    return cpu_ref.filter_series_time(ts, series, slot_lut, ts_lo, ts_hi)
+
+
+def dedup_mark_last(series, ts):
+    if ts.is_cuda:
+        return _require_hip().dedup_mark_last(series, ts)
+    return cpu_ref.dedup_mark_last(series, ts)

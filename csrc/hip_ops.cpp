@@ -11,7 +11,7 @@ void launch_ts_bucket_agg(
     const int64_t*, const int32_t*, const double*, int64_t, const int32_t*, int,
     const int32_t*, int, int64_t, int64_t, int64_t, int64_t, int, int, int64_t,
     double*, unsigned long long*, unsigned long long*, unsigned long long*,
-    hipStream_t);
+    unsigned long long*, hipStream_t);
 void launch_decode_minmax(
     const unsigned long long*, const unsigned long long*, const unsigned long long*,
     double*, double*, int64_t, hipStream_t);
@@ -50,6 +50,7 @@ std::vector<torch::Tensor> ts_bucket_agg(
   const int64_t cells = nf * n_slots * n_buckets;
   auto sum = torch::zeros({nf, n_slots, n_buckets}, opts_f64);
   auto cnt = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+  auto rows = torch::zeros({n_slots, n_buckets}, opts_i64);
   // min keys init to u64 max, max keys to 0
   auto minmax_init_min = torch::full({nf, n_slots, n_buckets}, -1, opts_i64);  // 0xFFFF...
   auto minmax_init_max = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
@@ -63,6 +64,7 @@ std::vector<torch::Tensor> ts_bucket_agg(
       reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(minmax_init_min.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(rows.data_ptr<int64_t>()),
       stream);
   auto minv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
   auto maxv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
@@ -71,7 +73,7 @@ std::vector<torch::Tensor> ts_bucket_agg(
       reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
       minv.data_ptr<double>(), maxv.data_ptr<double>(), cells, stream);
-  return {sum, cnt, minv, maxv};
+  return {sum, cnt, minv, maxv, rows};
 }
 
 torch::Tensor filter_series_time(

@@ -52,7 +52,8 @@ __global__ void ts_bucket_agg_kernel(
     double* __restrict__ out_sum,
     unsigned long long* __restrict__ out_cnt,
     unsigned long long* __restrict__ out_min,
-    unsigned long long* __restrict__ out_max) {
+    unsigned long long* __restrict__ out_max,
+    unsigned long long* __restrict__ out_rows) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const int64_t t = ts[i];
@@ -64,6 +65,7 @@ __global__ void ts_bucket_agg_kernel(
     int64_t b = (t - origin) / bucket_ms;
     if (b < 0 || b >= n_buckets) continue;
     const int64_t cell0 = (int64_t)slot * n_buckets + b;
+    atomicAdd(&out_rows[cell0], 1ULL);
     for (int f = 0; f < nf; f++) {
       const double v = fields[(int64_t)field_idx[f] * field_stride + i];
       if (isnan(v)) continue;
@@ -144,11 +146,12 @@ void launch_ts_bucket_agg(
     int n_slots, int n_buckets, int64_t n,
     double* out_sum, unsigned long long* out_cnt,
     unsigned long long* out_min, unsigned long long* out_max,
+    unsigned long long* out_rows,
     hipStream_t stream) {
   hipLaunchKernelGGL(ts_bucket_agg_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
       ts, series, fields, field_stride, field_idx, nf, slot_lut, lut_size,
       ts_lo, ts_hi, origin, bucket_ms, n_slots, n_buckets, n,
-      out_sum, out_cnt, out_min, out_max);
+      out_sum, out_cnt, out_min, out_max, out_rows);
 }
 
 void launch_decode_minmax(

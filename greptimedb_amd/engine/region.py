@@ -207,3 +207,18 @@ class Region:
     @property
     def num_rows(self) -> int:
         return self.memtable.len + sum(b.n for b in self.sst_cache.values())
+
+    def time_range(self) -> tuple[int, int] | None:
+        """(min_ts, max_ts) over memtable + SSTs, None if empty."""
+        lo = hi = None
+        for b in self.sst_cache.values():
+            lo = b.min_ts if lo is None else min(lo, b.min_ts)
+            hi = b.max_ts if hi is None else max(hi, b.max_ts)
+        with self.lock:
+            m_lo, m_hi = self.memtable.min_ts, self.memtable.max_ts
+        if m_lo is not None:
+            lo = m_lo if lo is None else min(lo, m_lo)
+            hi = m_hi if hi is None else max(hi, m_hi)
+        if lo is None:
+            return None
+        return lo, hi

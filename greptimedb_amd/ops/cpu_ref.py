@@ -15,9 +15,10 @@ def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
     """Fused filter + time-bucket aggregate.
 
     ts: i64[n] ms, series: i32[n], fields: f64[nf_total, >=n],
-    field_idx: i32[nf], slot_lut: i32[lut]. Returns (sum, count, min, max),
-    each [nf, n_slots, n_buckets]; min/max are NaN where count==0.
-    NaN field values count as nulls and are skipped.
+    field_idx: i32[nf], slot_lut: i32[lut]. Returns (sum, count, min, max,
+    rows) — the first four [nf, n_slots, n_buckets] (min/max NaN where
+    count==0), rows [n_slots, n_buckets] = matching row count per cell
+    (for count(*)/group existence). NaN field values are nulls, skipped.
     """
     n = ts.numel()
     keep = (ts >= ts_lo) & (ts < ts_hi)
@@ -33,11 +34,13 @@ def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
     out_cnt = torch.zeros(nf, n_slots, n_buckets, dtype=torch.int64)
     out_min = torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64)
     out_max = torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64)
+    out_rows = torch.zeros(n_slots, n_buckets, dtype=torch.int64)
 
     idx = keep.nonzero(as_tuple=True)[0]
     if idx.numel() == 0:
-        return out_sum, out_cnt, out_min, out_max
+        return out_sum, out_cnt, out_min, out_max, out_rows
     cell = slot[idx] * n_buckets + bucket[idx]
+    out_rows.view(-1).index_add_(0, cell, torch.ones_like(cell))
     for f in range(nf):
         v = fields[int(field_idx[f])][:n][idx]
         ok = ~torch.isnan(v)
@@ -50,7 +53,7 @@ def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
         mn.index_reduce_(0, c, vv, "amin", include_self=False)
         mx.index_reduce_(0, c, vv, "amax", include_self=False)
     # index_reduce with include_self=False leaves untouched cells at init NaN
-    return out_sum, out_cnt, out_min, out_max
+    return out_sum, out_cnt, out_min, out_max, out_rows
 
 
 def filter_series_time(ts, series, slot_lut, ts_lo, ts_hi):

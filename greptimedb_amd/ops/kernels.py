@@ -35,11 +35,10 @@ def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
                   origin, bucket_ms, n_slots, n_buckets):
     if ts.is_cuda:
         ops = _require_hip()
-        s, c, mn, mx = ops.ts_bucket_agg(
+        return tuple(ops.ts_bucket_agg(
             ts, series, fields, field_idx, slot_lut,
             int(ts_lo), int(ts_hi), int(origin), int(bucket_ms),
-            int(n_slots), int(n_buckets))
-        return s, c, mn, mx
+            int(n_slots), int(n_buckets)))
     return cpu_ref.ts_bucket_agg(ts, series, fields, field_idx, slot_lut,
                                  ts_lo, ts_hi, origin, bucket_ms, n_slots, n_buckets)
 

@@ -1,0 +1,112 @@
+"""Distributed query/ingest context: RCCL (or gloo) collectives.
+
+Reference parity: the scatter-gather in src/query/src/dist_plan
+(MergeScanExec fans out per-region streams, partial aggregates pushed down
+via commutativity analysis and combined at the frontend). MI355X redesign:
+one process per GPU (torch.distributed, backend "nccl" == RCCL over xGMI);
+each rank owns a shard of every table's regions; partial aggregates are
+combined with all-reduce (sum/count: SUM, min: MIN, max: MAX) after group
+keys are unified — small tensors, so a single fused all-reduce per query
+beats streaming Arrow batches through a frontend by orders of magnitude.
+
+Works on CPU with the gloo backend (multi-process tests, no GPU needed).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+class DistContext:
+    def __init__(self, device: str = "cpu"):
+        assert dist.is_initialized(), "torch.distributed must be initialized"
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        self.device = device
+        # collectives run on the compute device for nccl, cpu for gloo
+        self.coll_device = device if dist.get_backend() == "nccl" else "cpu"
+
+    # ------------------------------------------------------------ scalars
+
+    def minmax_ts(self, lo: int | None, hi: int | None):
+        t = torch.tensor([lo if lo is not None else (1 << 62),
+                          -(hi if hi is not None else -(1 << 62))],
+                         dtype=torch.int64, device=self.coll_device)
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        lo2 = int(t[0].item())
+        hi2 = -int(t[1].item())
+        if lo2 == (1 << 62):
+            return None, None
+        return lo2, hi2
+
+    def all_sum(self, value: float) -> float:
+        t = torch.tensor([value], dtype=torch.float64, device=self.coll_device)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return float(t.item())
+
+    def all_max(self, value: float) -> float:
+        t = torch.tensor([value], dtype=torch.float64, device=self.coll_device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return float(t.item())
+
+    def barrier(self):
+        dist.barrier()
+
+    # ------------------------------------------------------------ groups
+
+    def merge_groups(self, group_keys: dict, sums, cnts, mins, maxs, rowcnt):
+        """Unify group keys across ranks and all-reduce the partial
+        aggregate planes. Arrays are numpy [nf, n_slots, n_buckets] (+rowcnt
+        [n_slots, n_buckets]); returns (global_keys_dict, arrays)."""
+        all_keys: list = [None] * self.world
+        dist.all_gather_object(all_keys, list(group_keys.keys()))
+        merged = sorted({k for ks in all_keys for k in ks})
+        gmap = {k: i for i, k in enumerate(merged)}
+        ng = max(len(merged), 1)
+        nf, _, nb = sums.shape
+
+        g_sums = np.zeros((nf, ng, nb))
+        g_cnts = np.zeros((nf, ng, nb), dtype=np.int64)
+        g_mins = np.full((nf, ng, nb), np.inf)
+        g_maxs = np.full((nf, ng, nb), -np.inf)
+        g_rows = np.zeros((ng, nb), dtype=np.int64)
+        if group_keys:
+            local_slots = np.array([gmap[k] for k in group_keys.keys()])
+            src_slots = np.array(list(group_keys.values()))
+            g_sums[:, local_slots] = sums[:, src_slots]
+            g_cnts[:, local_slots] = cnts[:, src_slots]
+            g_mins[:, local_slots] = np.nan_to_num(mins[:, src_slots], nan=np.inf)
+            g_maxs[:, local_slots] = np.nan_to_num(maxs[:, src_slots], nan=-np.inf)
+            g_rows[local_slots] = rowcnt[src_slots]
+
+        dev = self.coll_device
+        t_sum = torch.as_tensor(g_sums, device=dev)
+        t_cnt = torch.as_tensor(g_cnts, device=dev)
+        t_min = torch.as_tensor(g_mins, device=dev)
+        t_max = torch.as_tensor(g_maxs, device=dev)
+        t_row = torch.as_tensor(g_rows, device=dev)
+        dist.all_reduce(t_sum, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t_cnt, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t_min, op=dist.ReduceOp.MIN)
+        dist.all_reduce(t_max, op=dist.ReduceOp.MAX)
+        dist.all_reduce(t_row, op=dist.ReduceOp.SUM)
+        g_sums = t_sum.cpu().numpy()
+        g_cnts = t_cnt.cpu().numpy()
+        g_mins = t_min.cpu().numpy()
+        g_maxs = t_max.cpu().numpy()
+        g_rows = t_row.cpu().numpy()
+        # restore NaN for empty cells
+        g_mins[g_cnts == 0] = np.nan
+        g_maxs[g_cnts == 0] = np.nan
+        return gmap, (g_sums, g_cnts, g_mins, g_maxs, g_rows)
+
+    # ------------------------------------------------------------ raw rows
+
+    def gather_columns(self, col_data: dict[str, np.ndarray]) -> dict[str, np.ndarray]:
+        """Concatenate per-rank row columns (raw scans) on every rank."""
+        gathered: list = [None] * self.world
+        dist.all_gather_object(gathered, {k: np.asarray(v) for k, v in col_data.items()})
+        keys = col_data.keys()
+        return {k: np.concatenate([g[k] for g in gathered]) for k in keys}

@@ -1,0 +1,128 @@
+"""SQL AST node types (reference: src/sql statement types, trimmed to the
+query surface implemented so far)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+
+@dataclass
+class Expr:
+    pass
+
+
+@dataclass
+class Col(Expr):
+    name: str
+
+
+@dataclass
+class Lit(Expr):
+    value: object  # int | float | str | bool | None
+
+
+@dataclass
+class Interval(Expr):
+    ms: int
+    text: str
+
+
+@dataclass
+class Func(Expr):
+    name: str
+    args: list[Expr]
+    distinct: bool = False
+
+
+@dataclass
+class Star(Expr):
+    pass
+
+
+@dataclass
+class BinOp(Expr):
+    op: str  # + - * / % = != < <= > >= and or like
+    left: Expr
+    right: Expr
+
+
+@dataclass
+class UnaryOp(Expr):
+    op: str  # - not
+    operand: Expr
+
+
+@dataclass
+class InList(Expr):
+    expr: Expr
+    items: list[Expr]
+    negated: bool = False
+
+
+@dataclass
+class Between(Expr):
+    expr: Expr
+    low: Expr
+    high: Expr
+    negated: bool = False
+
+
+@dataclass
+class IsNull(Expr):
+    expr: Expr
+    negated: bool = False
+
+
+@dataclass
+class Select:
+    projections: list[tuple[Expr, str | None]]  # (expr, alias)
+    table: str | None
+    where: Expr | None = None
+    group_by: list[Expr] = field(default_factory=list)
+    having: Expr | None = None
+    order_by: list[tuple[Expr, bool]] = field(default_factory=list)  # (expr, desc)
+    limit: int | None = None
+    offset: int | None = None
+
+
+@dataclass
+class CreateTable:
+    name: str
+    columns: list[tuple[str, str, dict]]   # (name, type, {primary/time_index/null})
+    primary_key: list[str]
+    time_index: str | None
+    if_not_exists: bool = False
+    options: dict = field(default_factory=dict)
+    partitions: int | None = None
+
+
+@dataclass
+class DropTable:
+    name: str
+    if_exists: bool = False
+
+
+@dataclass
+class ShowTables:
+    pass
+
+
+@dataclass
+class DescribeTable:
+    name: str
+
+
+@dataclass
+class InsertValues:
+    table: str
+    columns: list[str]
+    rows: list[list[object]]
+
+
+@dataclass
+class Tql:
+    """TQL EVAL (start, end, step) promql_expr  (reference: src/sql TQL)."""
+    start: float
+    end: float
+    step: float
+    query: str

@@ -102,6 +102,34 @@ class DistContext:
         g_maxs[g_cnts == 0] = np.nan
         return gmap, (g_sums, g_cnts, g_mins, g_maxs, g_rows)
 
+    def merge_lastpoint(self, group_keys: dict, best_ts, best_val):
+        """Combine per-rank lastpoint partials: per group, the row with the
+        globally newest ts wins (ties: max value)."""
+        all_keys: list = [None] * self.world
+        dist.all_gather_object(all_keys, list(group_keys.keys()))
+        merged = sorted({k for ks in all_keys for k in ks})
+        gmap = {k: i for i, k in enumerate(merged)}
+        ng = max(len(merged), 1)
+        nf = best_val.shape[0]
+        g_ts = np.full(ng, -(1 << 62), dtype=np.int64)
+        g_val = np.full((nf, ng), -np.inf)
+        if group_keys:
+            l = np.array([gmap[k] for k in group_keys.keys()])
+            s = np.array(list(group_keys.values()))
+            g_ts[l] = best_ts[s]
+            g_val[:, l] = np.nan_to_num(best_val[:, s], nan=-np.inf)
+        dev = self.coll_device
+        t_ts = torch.as_tensor(g_ts, device=dev)
+        dist.all_reduce(t_ts, op=dist.ReduceOp.MAX)
+        g_ts_glob = t_ts.cpu().numpy()
+        mine = g_ts == g_ts_glob
+        g_val[:, ~mine] = -np.inf
+        t_val = torch.as_tensor(g_val, device=dev)
+        dist.all_reduce(t_val, op=dist.ReduceOp.MAX)
+        g_val = t_val.cpu().numpy()
+        g_val[g_val == -np.inf] = np.nan
+        return gmap, g_ts_glob, g_val
+
     # ------------------------------------------------------------ raw rows
 
     def gather_columns(self, col_data: dict[str, np.ndarray]) -> dict[str, np.ndarray]:

@@ -30,7 +30,7 @@ from greptimedb_amd.utils.errors import InvalidArguments, PlanQuery
 from greptimedb_amd.utils.timeutil import parse_ts_ms, trunc_unit_ms
 from greptimedb_amd.ops import ts_bucket_agg, dedup_mark_last
 
-AGG_FUNCS = {"count", "sum", "min", "max", "avg", "mean"}
+AGG_FUNCS = {"count", "sum", "min", "max", "avg", "mean", "last_value"}
 BUCKET_FUNCS = {"date_trunc", "date_bin", "time_bucket"}
 MAX_BUCKETS = 8_000_000
 
@@ -402,6 +402,8 @@ class Executor:
         return sorted(codes)
 
     def _exec_aggregate(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
+        if any(a.func == "last_value" for a in plan.aggs):
+            return self._exec_lastpoint(sel, plan)
         st = plan.table
         device = self.engine.config.device
 
@@ -497,6 +499,148 @@ class Executor:
 
         return self._finalize_agg(sel, plan, group_keys, origin, bucket_ms,
                                   agg_fields, sums, cnts, mins, maxs, rowcnt)
+
+    def _exec_lastpoint(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
+        """TSBS `lastpoint` shape: last_value(field) per group (latest-ts row).
+
+        Reference parity: DISTINCT ON / last_value window in the reference's
+        TSBS adapter. Per region: per-slot argmax over ts, then value-at-max;
+        regions/ranks combine by ts comparison (groups are series-aligned so
+        at most one shard owns a group in the common case)."""
+        st = plan.table
+        device = self.engine.config.device
+        if plan.bucket is not None:
+            raise PlanQuery("last_value with time buckets unsupported")
+        for a in plan.aggs:
+            if a.func != "last_value":
+                raise PlanQuery("cannot mix last_value with other aggregates")
+        lv_fields = sorted({a.arg for a in plan.aggs})
+        ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
+        ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
+
+        group_keys: dict[tuple, int] = {}
+        gt = plan.group_tags
+        best_ts: np.ndarray | None = None
+        best_val: np.ndarray | None = None
+
+        for region in st.regions:
+            cand = self._candidate_codes(region, plan)
+            nser = len(region.series)
+            lut = np.full(nser, -1, dtype=np.int32)
+            it = range(nser) if cand is None else cand
+            if gt:
+                tag_idx = [region.series.tag_names.index(t) for t in gt]
+                for code in it:
+                    tv = region.series.tag_values[code]
+                    key = tuple(tv[i] for i in tag_idx)
+                    lut[code] = group_keys.setdefault(key, len(group_keys))
+            else:
+                group_keys.setdefault((), 0)
+                for code in it:
+                    lut[code] = 0
+            lut_t = torch.as_tensor(lut, device=device)
+            # gather rows
+            chunks = []
+            for src in region.scan_sources(ts_lo, ts_hi):
+                from greptimedb_amd.ops import filter_series_time
+                mask = filter_series_time(src.ts, src.series, lut_t if cand is not None else None,
+                                          ts_lo, ts_hi)
+                if plan.residual is not None:
+                    mask &= self._eval_mask(plan.residual, src, region, device)
+                idx = mask.nonzero(as_tuple=True)[0]
+                if idx.numel() == 0:
+                    continue
+                f_rows = [src.fields[src.field_pos[fn]][idx]
+                          if fn in src.field_pos else
+                          torch.full((idx.numel(),), float("nan"),
+                                     dtype=torch.float64, device=device)
+                          for fn in lv_fields]
+                chunks.append((src.ts[idx], src.series[idx],
+                               torch.stack(f_rows) if f_rows else
+                               torch.zeros((0, idx.numel()), device=device)))
+            if not chunks:
+                continue
+            ts_t = torch.cat([c[0] for c in chunks])
+            se_t = torch.cat([c[1] for c in chunks])
+            f_t = torch.cat([c[2] for c in chunks], dim=1)
+            slots = lut_t[se_t.long()].long()
+            ok = slots >= 0
+            ts_t, slots, f_t = ts_t[ok], slots[ok], f_t[:, ok]
+            if ts_t.numel() == 0:
+                continue
+            ng = len(group_keys) if group_keys else 1
+            # per-slot max ts
+            mx = torch.full((ng,), -(1 << 62), dtype=torch.int64, device=device)
+            mx.index_reduce_(0, slots, ts_t, "amax", include_self=True)
+            # rows at the max; pick last arrival among ties
+            at_max = ts_t == mx[slots]
+            ridx = at_max.nonzero(as_tuple=True)[0]
+            vals = torch.full((len(lv_fields), ng), float("nan"),
+                              dtype=torch.float64, device=device)
+            # scatter in arrival order → later rows win
+            vals[:, slots[ridx]] = f_t[:, ridx]
+            mx_h = mx.cpu().numpy()
+            vals_h = vals.cpu().numpy()
+            if best_ts is None:
+                best_ts, best_val = mx_h, vals_h
+            else:
+                if len(mx_h) > len(best_ts):
+                    pad = len(mx_h) - len(best_ts)
+                    best_ts = np.concatenate([best_ts, np.full(pad, -(1 << 62), dtype=np.int64)])
+                    best_val = np.concatenate(
+                        [best_val, np.full((len(lv_fields), pad), np.nan)], axis=1)
+                newer = mx_h > best_ts[: len(mx_h)]
+                best_ts[: len(mx_h)][newer] = mx_h[newer]
+                best_val[:, : len(mx_h)][:, newer] = vals_h[:, newer]
+        ng = max(len(group_keys), 1)
+        if best_ts is None:
+            best_ts = np.full(ng, -(1 << 62), dtype=np.int64)
+            best_val = np.full((len(lv_fields), ng), np.nan)
+        elif len(best_ts) < ng:
+            pad = ng - len(best_ts)
+            best_ts = np.concatenate([best_ts, np.full(pad, -(1 << 62), dtype=np.int64)])
+            best_val = np.concatenate([best_val, np.full((len(lv_fields), pad), np.nan)], axis=1)
+
+        if self.dist is not None:
+            group_keys, best_ts, best_val = self.dist.merge_lastpoint(
+                group_keys, best_ts, best_val)
+            ng = max(len(group_keys), 1)
+
+        keys_by_slot = [None] * ng
+        for k, s in group_keys.items():
+            keys_by_slot[s] = k
+        valid = best_ts > -(1 << 62)
+        slot_idx = np.flatnonzero(valid)
+        fpos = {fn: i for i, fn in enumerate(lv_fields)}
+
+        names, cols, kinds = [], [], []
+        for e, alias in plan.projections:
+            if isinstance(e, ast.Col) and e.name in gt:
+                i = gt.index(e.name)
+                arr = np.array([keys_by_slot[s][i] for s in slot_idx], dtype=object)
+                names.append(alias or e.name); cols.append(arr); kinds.append("")
+            elif isinstance(e, ast.Func) and e.name == "last_value":
+                arr = best_val[fpos[e.args[0].name]][slot_idx]
+                names.append(alias or _expr_name(e)); cols.append(arr); kinds.append("")
+            else:
+                raise PlanQuery(f"unsupported lastpoint projection {e}")
+        n_out = len(slot_idx)
+        # ORDER BY group tags only
+        idx = np.arange(n_out)
+        for e, desc in reversed(plan.order_by):
+            if isinstance(e, ast.Col):
+                if e.name in gt:
+                    i = gt.index(e.name)
+                    a = np.array([keys_by_slot[s][i] for s in slot_idx], dtype=object)[idx]
+                elif e.name in dict(zip(names, cols)):
+                    a = np.asarray(dict(zip(names, cols))[e.name])[idx]
+                else:
+                    raise PlanQuery("lastpoint ORDER BY must use group tags")
+                o = np.argsort(a, kind="stable")
+                idx = idx[o[::-1] if desc else o]
+        if plan.limit is not None:
+            idx = idx[: plan.limit]
+        return QueryResult(names, [np.asarray(c, dtype=object)[idx] for c in cols], kinds)
 
     def _region_agg_inputs(self, region, plan, device, agg_fields, ts_lo, ts_hi):
         """Yield (ts, series, fields[nf,n], field_idx) kernel inputs for one

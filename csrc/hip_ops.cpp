@@ -19,6 +19,18 @@ void launch_filter_series_time(
     const int64_t*, const int32_t*, const int32_t*, int, int64_t, int64_t,
     int64_t, bool*, hipStream_t);
 void launch_dedup_mark_last(const int32_t*, const int64_t*, int64_t, bool*, hipStream_t);
+void launch_bucket_agg2(
+    const int64_t*, const int32_t*, const double*, int64_t, const int32_t*, int,
+    const int32_t*, int, int64_t, int64_t, int64_t, int64_t, int, int, int64_t,
+    int32_t*, double*, unsigned long long*, unsigned long long*,
+    unsigned long long*, unsigned long long*, hipStream_t);
+void launch_series_last(
+    const int64_t*, const int32_t*, const int32_t*, int, int64_t, int64_t,
+    int64_t, unsigned long long*, hipStream_t);
+void launch_series_last_row(
+    const int64_t*, const int32_t*, const int32_t*, int, int64_t, int64_t,
+    int64_t, const unsigned long long*, unsigned long long,
+    unsigned long long*, hipStream_t);
 }  // namespace gdb_hip
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -55,11 +67,13 @@ std::vector<torch::Tensor> ts_bucket_agg(
   auto minmax_init_min = torch::full({nf, n_slots, n_buckets}, -1, opts_i64);  // 0xFFFF...
   auto minmax_init_max = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
   auto stream = at::cuda::getCurrentHIPStream().stream();
-  gdb_hip::launch_ts_bucket_agg(
+  auto cell = torch::empty({n}, ts.options().dtype(torch::kInt32));
+  gdb_hip::launch_bucket_agg2(
       ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(), fields.data_ptr<double>(),
       fields.size(1), field_idx.data_ptr<int32_t>(), nf,
       slot_lut.data_ptr<int32_t>(), (int)slot_lut.numel(),
       ts_lo, ts_hi, origin, bucket_ms, (int)n_slots, (int)n_buckets, n,
+      cell.data_ptr<int32_t>(),
       sum.data_ptr<double>(),
       reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(minmax_init_min.data_ptr<int64_t>()),
@@ -107,8 +121,39 @@ torch::Tensor dedup_mark_last(torch::Tensor series, torch::Tensor ts) {
   return keep;
 }
 
+// Accumulate per-slot max-ts keys across sources (call once per source with
+// a shared best_key tensor, int64 viewed as u64 keys, init 0).
+void series_last_ts(torch::Tensor ts, torch::Tensor series, torch::Tensor slot_lut,
+                    int64_t ts_lo, int64_t ts_hi, torch::Tensor best_key) {
+  CHECK_GPU(ts); CHECK_CONTIG(ts); CHECK_GPU(best_key); CHECK_CONTIG(best_key);
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_series_last(
+      ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(),
+      slot_lut.data_ptr<int32_t>(), (int)slot_lut.numel(), ts_lo, ts_hi,
+      ts.numel(),
+      reinterpret_cast<unsigned long long*>(best_key.data_ptr<int64_t>()),
+      stream);
+}
+
+void series_last_row(torch::Tensor ts, torch::Tensor series, torch::Tensor slot_lut,
+                     int64_t ts_lo, int64_t ts_hi, torch::Tensor best_key,
+                     int64_t src_tag, torch::Tensor best_row) {
+  CHECK_GPU(ts); CHECK_CONTIG(ts);
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_series_last_row(
+      ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(),
+      slot_lut.data_ptr<int32_t>(), (int)slot_lut.numel(), ts_lo, ts_hi,
+      ts.numel(),
+      reinterpret_cast<const unsigned long long*>(best_key.data_ptr<int64_t>()),
+      (unsigned long long)src_tag,
+      reinterpret_cast<unsigned long long*>(best_row.data_ptr<int64_t>()),
+      stream);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate");
   m.def("filter_series_time", &filter_series_time, "series/time filter mask");
   m.def("dedup_mark_last", &dedup_mark_last, "last-row dedup marker");
+  m.def("series_last_ts", &series_last_ts, "per-slot max-ts accumulate (lastpoint)");
+  m.def("series_last_row", &series_last_row, "per-slot winner row (lastpoint)");
 }

@@ -11,5 +11,6 @@ from greptimedb_amd.ops.kernels import (  # noqa: F401
     dedup_mark_last,
     filter_series_time,
     hip_ops_available,
+    series_last,
     ts_bucket_agg,
 )

@@ -65,6 +65,44 @@ def filter_series_time(ts, series, slot_lut, ts_lo, ts_hi):
     return keep
 
 
+def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
+    """Lastpoint primitive: over sources [(ts, series), ...] (in recency
+    order), find per slot the newest ts and its (source, row); later sources
+    / later rows win ties. Returns (best_ts i64[n_slots], best_src i64,
+    best_row i64), src/row = -1 where empty."""
+    best_ts = torch.full((n_slots,), -(1 << 63) + 1, dtype=torch.int64)
+    best_src = torch.full((n_slots,), -1, dtype=torch.int64)
+    best_row = torch.full((n_slots,), -1, dtype=torch.int64)
+    lut = slot_lut.long()
+    for ts, series in sources:
+        keep = (ts >= ts_lo) & (ts < ts_hi) & (series >= 0) & (series < lut.numel())
+        idx = keep.nonzero(as_tuple=True)[0]
+        if idx.numel() == 0:
+            continue
+        slots = lut[series.long()[idx]]
+        ok = slots >= 0
+        idx, slots = idx[ok], slots[ok]
+        t = ts[idx]
+        mx = torch.full((n_slots,), -(1 << 63) + 1, dtype=torch.int64)
+        mx.index_reduce_(0, slots, t, "amax", include_self=True)
+        best_ts = torch.maximum(best_ts, mx)
+    for si, (ts, series) in enumerate(sources):
+        keep = (ts >= ts_lo) & (ts < ts_hi) & (series >= 0) & (series < lut.numel())
+        idx = keep.nonzero(as_tuple=True)[0]
+        if idx.numel() == 0:
+            continue
+        slots = lut[series.long()[idx]]
+        ok = slots >= 0
+        idx, slots = idx[ok], slots[ok]
+        at = best_ts[slots] == ts[idx]
+        ridx = at.nonzero(as_tuple=True)[0]
+        # sequential scatter: later rows win (arrival order); later sources
+        # override earlier ones entirely (processed in order)
+        best_src[slots[ridx]] = si
+        best_row[slots[ridx]] = idx[ridx]
+    return best_ts, best_src, best_row
+
+
 def dedup_mark_last(series, ts):
     """keep[i] ⇔ row i is the last of its (series, ts) group (sorted input)."""
     n = ts.numel()

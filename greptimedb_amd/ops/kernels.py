@@ -55,3 +55,26 @@ def dedup_mark_last(series, ts):
     if ts.is_cuda:
         return _require_hip().dedup_mark_last(series, ts)
     return cpu_ref.dedup_mark_last(series, ts)
+
+
+def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
+    """See cpu_ref.series_last. GPU path: two atomic-max passes per source."""
+    if not sources or not sources[0][0].is_cuda:
+        return cpu_ref.series_last(sources, slot_lut, ts_lo, ts_hi, n_slots)
+    ops = _require_hip()
+    dev = sources[0][0].device
+    best_key = torch.zeros(n_slots, dtype=torch.int64, device=dev)
+    for ts, series in sources:
+        ops.series_last_ts(ts.contiguous(), series.contiguous(), slot_lut,
+                           int(ts_lo), int(ts_hi), best_key)
+    best_pack = torch.zeros(n_slots, dtype=torch.int64, device=dev)
+    for si, (ts, series) in enumerate(sources):
+        ops.series_last_row(ts.contiguous(), series.contiguous(), slot_lut,
+                            int(ts_lo), int(ts_hi), best_key, si, best_pack)
+    best_ts = torch.bitwise_xor(best_key, torch.tensor(-(1 << 63), device=dev))
+    none = best_pack == 0
+    best_src = torch.where(none, torch.full_like(best_pack, -1), best_pack >> 40)
+    best_row = torch.where(none, torch.full_like(best_pack, -1),
+                           (best_pack & ((1 << 40) - 1)) - 1)
+    best_ts = torch.where(none, torch.full_like(best_ts, -(1 << 63) + 1), best_ts)
+    return best_ts.cpu(), best_src.cpu(), best_row.cpu()

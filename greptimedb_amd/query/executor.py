@@ -539,48 +539,45 @@ class Executor:
                 for code in it:
                     lut[code] = 0
             lut_t = torch.as_tensor(lut, device=device)
-            # gather rows
-            chunks = []
-            for src in region.scan_sources(ts_lo, ts_hi):
-                from greptimedb_amd.ops import filter_series_time
-                mask = filter_series_time(src.ts, src.series, lut_t if cand is not None else None,
-                                          ts_lo, ts_hi)
-                if plan.residual is not None:
-                    mask &= self._eval_mask(plan.residual, src, region, device)
-                idx = mask.nonzero(as_tuple=True)[0]
-                if idx.numel() == 0:
-                    continue
-                f_rows = [src.fields[src.field_pos[fn]][idx]
-                          if fn in src.field_pos else
-                          torch.full((idx.numel(),), float("nan"),
-                                     dtype=torch.float64, device=device)
-                          for fn in lv_fields]
-                chunks.append((src.ts[idx], src.series[idx],
-                               torch.stack(f_rows) if f_rows else
-                               torch.zeros((0, idx.numel()), device=device)))
-            if not chunks:
-                continue
-            ts_t = torch.cat([c[0] for c in chunks])
-            se_t = torch.cat([c[1] for c in chunks])
-            f_t = torch.cat([c[2] for c in chunks], dim=1)
-            slots = lut_t[se_t.long()].long()
-            ok = slots >= 0
-            ts_t, slots, f_t = ts_t[ok], slots[ok], f_t[:, ok]
-            if ts_t.numel() == 0:
+            sources = region.scan_sources(ts_lo, ts_hi)
+            if plan.residual is not None:
+                # rare: pre-filter sources through the residual mask
+                filt = []
+                for src in sources:
+                    mask = self._eval_mask(plan.residual, src, region, device)
+                    idx = mask.nonzero(as_tuple=True)[0]
+                    if idx.numel():
+                        filt.append((src, idx))
+                pairs = [(src.ts[idx].contiguous(), src.series[idx].contiguous())
+                         for src, idx in filt]
+                src_objs = [(src, idx) for src, idx in filt]
+            else:
+                pairs = [(src.ts, src.series) for src in sources]
+                src_objs = [(src, None) for src in sources]
+            if not pairs:
                 continue
             ng = len(group_keys) if group_keys else 1
-            # per-slot max ts
-            mx = torch.full((ng,), -(1 << 62), dtype=torch.int64, device=device)
-            mx.index_reduce_(0, slots, ts_t, "amax", include_self=True)
-            # rows at the max; pick last arrival among ties
-            at_max = ts_t == mx[slots]
-            ridx = at_max.nonzero(as_tuple=True)[0]
+            from greptimedb_amd.ops import series_last
+            b_ts, b_src, b_row = series_last(pairs, lut_t, ts_lo, ts_hi, ng)
+            mx_h = b_ts.numpy().astype(np.int64)
+            # gather winner values per source
             vals = torch.full((len(lv_fields), ng), float("nan"),
                               dtype=torch.float64, device=device)
-            # scatter in arrival order → later rows win
-            vals[:, slots[ridx]] = f_t[:, ridx]
-            mx_h = mx.cpu().numpy()
+            for si, (src, idx) in enumerate(src_objs):
+                slots_here = (b_src == si).nonzero(as_tuple=True)[0]
+                if slots_here.numel() == 0:
+                    continue
+                rows = b_row[slots_here]
+                if idx is not None:
+                    rows = idx.cpu()[rows]
+                rows_d = rows.to(device)
+                slots_d = slots_here.to(device)
+                for fi, fn in enumerate(lv_fields):
+                    p = src.field_pos.get(fn)
+                    if p is not None:
+                        vals[fi, slots_d] = src.fields[p][rows_d]
             vals_h = vals.cpu().numpy()
+            mx_h = np.where(b_src.numpy() >= 0, mx_h, -(1 << 62))
             if best_ts is None:
                 best_ts, best_val = mx_h, vals_h
             else:

@@ -92,3 +92,18 @@ def test_dedup_mark_last():
     assert got.tolist() == [False, True, True, False, True, True]
     assert cpu_ref.dedup_mark_last(torch.zeros(0, dtype=torch.int32),
                                    torch.zeros(0, dtype=torch.int64)).numel() == 0
+
+
+def test_series_last():
+    ts1 = torch.tensor([10, 20, 30, 5], dtype=torch.int64)
+    se1 = torch.tensor([0, 0, 1, 2], dtype=torch.int32)
+    ts2 = torch.tensor([20, 25], dtype=torch.int64)
+    se2 = torch.tensor([0, 1], dtype=torch.int32)
+    lut = torch.tensor([0, 1, -1], dtype=torch.int32)
+    b_ts, b_src, b_row = cpu_ref.series_last(
+        [(ts1, se1), (ts2, se2)], lut, 0, 100, 2)
+    # slot0 (series 0): max ts 20 appears in src0 row1 and src1 row0 → src1 wins tie
+    assert b_ts.tolist() == [20, 30]
+    assert b_src.tolist() == [1, 0]
+    assert b_row.tolist() == [0, 2]
+    # series 2 excluded by lut

@@ -84,3 +84,21 @@ def test_extreme_values():
         torch.as_tensor(fi).to(dev), torch.as_tensor(lut).to(dev), *args)
     assert mn.item() == -np.inf and mx.item() == np.inf
     assert c.item() == n and rows.item() == n
+
+
+def test_series_last_matches_cpu():
+    rng = np.random.RandomState(11)
+    sources = []
+    for _ in range(4):
+        n = 50_000
+        ts = torch.as_tensor(rng.randint(0, 10_000, n).astype(np.int64))
+        se = torch.as_tensor(rng.randint(0, 200, n).astype(np.int32))
+        sources.append((ts, se))
+    lut = torch.as_tensor(rng.randint(-1, 50, 200).astype(np.int32))
+    exp = cpu_ref.series_last(sources, lut, 100, 9_000, 50)
+    got = kernels.series_last([(t.cuda(), s.cuda()) for t, s in sources],
+                              lut.cuda(), 100, 9_000, 50)
+    # ts must match exactly; src/row may differ only when ts ties across
+    # sources... no: tie-break is deterministic (later src, later row)
+    for e, g, name in zip(exp, got, ["ts", "src", "row"]):
+        np.testing.assert_array_equal(e.numpy(), g.numpy(), err_msg=name)

@@ -86,6 +86,7 @@ def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
         mx = torch.full((n_slots,), -(1 << 63) + 1, dtype=torch.int64)
         mx.index_reduce_(0, slots, t, "amax", include_self=True)
         best_ts = torch.maximum(best_ts, mx)
+    best_pack = torch.full((n_slots,), -1, dtype=torch.int64)
     for si, (ts, series) in enumerate(sources):
         keep = (ts >= ts_lo) & (ts < ts_hi) & (series >= 0) & (series < lut.numel())
         idx = keep.nonzero(as_tuple=True)[0]
@@ -96,10 +97,14 @@ def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
         idx, slots = idx[ok], slots[ok]
         at = best_ts[slots] == ts[idx]
         ridx = at.nonzero(as_tuple=True)[0]
-        # sequential scatter: later rows win (arrival order); later sources
-        # override earlier ones entirely (processed in order)
-        best_src[slots[ridx]] = si
-        best_row[slots[ridx]] = idx[ridx]
+        # deterministic tie-break: max (source, row) wins (LastRow recency)
+        pack = (si << 40) | idx[ridx]
+        mx2 = torch.full((n_slots,), -1, dtype=torch.int64)
+        mx2.index_reduce_(0, slots[ridx], pack, "amax", include_self=True)
+        best_pack = torch.maximum(best_pack, mx2)
+    found = best_pack >= 0
+    best_src = torch.where(found, best_pack >> 40, best_src)
+    best_row = torch.where(found, best_pack & ((1 << 40) - 1), best_row)
     return best_ts, best_src, best_row
 
 

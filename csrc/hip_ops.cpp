@@ -31,6 +31,9 @@ void launch_series_last_row(
     const int64_t*, const int32_t*, const int32_t*, int, int64_t, int64_t,
     int64_t, const unsigned long long*, unsigned long long,
     unsigned long long*, hipStream_t);
+void launch_prom_range_eval(
+    const int64_t*, const double*, const int64_t*, const int64_t*, int, int,
+    int64_t, int64_t, int64_t, int64_t, double, int, double*, hipStream_t);
 }  // namespace gdb_hip
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -150,10 +153,29 @@ void series_last_row(torch::Tensor ts, torch::Tensor series, torch::Tensor slot_
       stream);
 }
 
+// PromQL range/instant evaluation over (slot, ts)-sorted samples.
+torch::Tensor prom_range_eval(
+    torch::Tensor ts, torch::Tensor vals, torch::Tensor seg_lo, torch::Tensor seg_hi,
+    int64_t T, int64_t t0, int64_t step_ms, int64_t range_ms, int64_t offset_ms,
+    double param, int64_t mode) {
+  CHECK_GPU(ts); CHECK_CONTIG(ts); CHECK_GPU(vals); CHECK_CONTIG(vals);
+  CHECK_GPU(seg_lo); CHECK_CONTIG(seg_lo); CHECK_GPU(seg_hi); CHECK_CONTIG(seg_hi);
+  const int S = (int)seg_lo.numel();
+  auto out = torch::empty({S, T}, vals.options());
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_prom_range_eval(
+      ts.data_ptr<int64_t>(), vals.data_ptr<double>(),
+      seg_lo.data_ptr<int64_t>(), seg_hi.data_ptr<int64_t>(),
+      S, (int)T, t0, step_ms, range_ms, offset_ms, param, (int)mode,
+      out.data_ptr<double>(), stream);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate");
   m.def("filter_series_time", &filter_series_time, "series/time filter mask");
   m.def("dedup_mark_last", &dedup_mark_last, "last-row dedup marker");
   m.def("series_last_ts", &series_last_ts, "per-slot max-ts accumulate (lastpoint)");
   m.def("series_last_row", &series_last_row, "per-slot winner row (lastpoint)");
+  m.def("prom_range_eval", &prom_range_eval, "PromQL range-vector evaluator");
 }

@@ -287,6 +287,150 @@ __global__ void series_last_row_kernel(
   }
 }
 
+// ------------------------------------------------- K7/K8 PromQL evaluators
+//
+// One kernel evaluates a PromQL selector/range-function over a
+// (slot, ts)-sorted sample array onto the query step grid. Thread =
+// (slot, step): binary-search the window bounds inside the slot's segment,
+// then reduce the window per `mode`. Rate/increase/delta follow Prometheus'
+// extrapolation rules exactly (reference: promql/functions/extrapolate_rate.rs,
+// itself a port of Prometheus' extrapolatedRate).
+
+enum PromMode {
+  PM_INSTANT = 0, PM_RATE = 1, PM_INCREASE = 2, PM_DELTA = 3,
+  PM_AVG = 4, PM_SUM = 5, PM_MIN = 6, PM_MAX = 7, PM_COUNT = 8, PM_LAST = 9,
+  PM_IDELTA = 10, PM_IRATE = 11, PM_DERIV = 12, PM_PREDICT = 13,
+  PM_RESETS = 14, PM_CHANGES = 15, PM_STDDEV = 16, PM_STDVAR = 17,
+  PM_ABSENT_OT = 18,
+};
+
+DEV_INLINE double prom_nan() { return __longlong_as_double(0x7FF8000000000000LL); }
+
+__global__ void prom_range_eval_kernel(
+    const int64_t* __restrict__ ts,
+    const double* __restrict__ vals,
+    const int64_t* __restrict__ seg_lo,   // [S] first row of slot
+    const int64_t* __restrict__ seg_hi,   // [S] one past last row
+    int S, int T,
+    int64_t t0, int64_t step_ms, int64_t range_ms, int64_t offset_ms,
+    double param, int mode,
+    double* __restrict__ out) {           // [S, T]
+  const int64_t total = (int64_t)S * T;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    const int s = (int)(i / T);
+    const int t = (int)(i % T);
+    const int64_t te = t0 + (int64_t)t * step_ms - offset_ms;   // window end (inclusive)
+    const int64_t tb = te - range_ms;                           // window begin (exclusive)
+    const int64_t lo = seg_lo[s], hi = seg_hi[s];
+    // lower: first idx with ts > tb ; upper: last idx with ts <= te
+    int64_t a = lo, b = hi;
+    while (a < b) { int64_t m = (a + b) >> 1; if (ts[m] <= tb) a = m + 1; else b = m; }
+    const int64_t w_lo = a;
+    a = w_lo; b = hi;
+    while (a < b) { int64_t m = (a + b) >> 1; if (ts[m] <= te) a = m + 1; else b = m; }
+    const int64_t w_hi = a;                                     // exclusive
+    const int64_t cnt = w_hi - w_lo;
+    double r = prom_nan();
+    if (mode == PM_INSTANT) {
+      if (cnt > 0) r = vals[w_hi - 1];
+    } else if (mode == PM_LAST) {
+      if (cnt > 0) r = vals[w_hi - 1];
+    } else if (mode == PM_COUNT) {
+      if (cnt > 0) r = (double)cnt;
+    } else if (mode == PM_ABSENT_OT) {
+      r = (cnt > 0) ? prom_nan() : 1.0;
+    } else if (cnt > 0 && (mode == PM_SUM || mode == PM_AVG || mode == PM_MIN ||
+                           mode == PM_MAX || mode == PM_STDDEV || mode == PM_STDVAR)) {
+      double sum = 0, mn = vals[w_lo], mx = vals[w_lo];
+      for (int64_t j = w_lo; j < w_hi; j++) {
+        const double v = vals[j];
+        sum += v; mn = fmin(mn, v); mx = fmax(mx, v);
+      }
+      if (mode == PM_SUM) r = sum;
+      else if (mode == PM_AVG) r = sum / cnt;
+      else if (mode == PM_MIN) r = mn;
+      else if (mode == PM_MAX) r = mx;
+      else {
+        const double mean = sum / cnt;
+        double aux = 0;
+        for (int64_t j = w_lo; j < w_hi; j++) {
+          const double d = vals[j] - mean;
+          aux += d * d;
+        }
+        const double var = aux / cnt;
+        r = (mode == PM_STDVAR) ? var : sqrt(var);
+      }
+    } else if (cnt >= 2 && (mode == PM_RATE || mode == PM_INCREASE || mode == PM_DELTA)) {
+      const bool is_counter = (mode != PM_DELTA);
+      const double first_v = vals[w_lo], last_v = vals[w_hi - 1];
+      const int64_t first_t = ts[w_lo], last_t = ts[w_hi - 1];
+      double total_incr = last_v - first_v;
+      if (is_counter) {
+        double prev = first_v;
+        for (int64_t j = w_lo + 1; j < w_hi; j++) {
+          const double v = vals[j];
+          if (v < prev) total_incr += prev;   // counter reset correction
+          prev = v;
+        }
+      }
+      const double sampled = (double)(last_t - first_t) / 1000.0;
+      const double range_s = (double)range_ms / 1000.0;
+      const double avg_dur = sampled / (cnt - 1);
+      double dur_to_start = (double)(first_t - tb) / 1000.0;
+      const double dur_to_end = (double)(te - last_t) / 1000.0;
+      if (is_counter && total_incr > 0 && first_v >= 0) {
+        const double dur_to_zero = sampled * (first_v / total_incr);
+        if (dur_to_zero < dur_to_start) dur_to_start = dur_to_zero;
+      }
+      const double thresh = avg_dur * 1.1;
+      double ext = sampled;
+      ext += (dur_to_start < thresh) ? dur_to_start : avg_dur / 2;
+      ext += (dur_to_end < thresh) ? dur_to_end : avg_dur / 2;
+      const double factor = (sampled > 0) ? ext / sampled : 1.0;
+      double res = total_incr * factor;
+      if (mode == PM_RATE) res /= range_s;
+      r = res;
+    } else if (cnt >= 2 && (mode == PM_IDELTA || mode == PM_IRATE)) {
+      const double dv = vals[w_hi - 1] - vals[w_hi - 2];
+      const double dt = (double)(ts[w_hi - 1] - ts[w_hi - 2]) / 1000.0;
+      if (mode == PM_IDELTA) {
+        r = dv;
+      } else {
+        double v2 = vals[w_hi - 1], v1 = vals[w_hi - 2];
+        double d = (v2 < v1) ? v2 : dv;   // reset → use raw value
+        r = (dt > 0) ? d / dt : prom_nan();
+      }
+    } else if (cnt >= 2 && (mode == PM_DERIV || mode == PM_PREDICT)) {
+      // least-squares slope/intercept, x relative to window end (Prometheus
+      // uses intercept at `te`)
+      double sx = 0, sy = 0, sxx = 0, sxy = 0;
+      for (int64_t j = w_lo; j < w_hi; j++) {
+        const double x = (double)(ts[j] - te) / 1000.0;
+        const double y = vals[j];
+        sx += x; sy += y; sxx += x * x; sxy += x * y;
+      }
+      const double nn = (double)cnt;
+      const double den = nn * sxx - sx * sx;
+      if (den != 0) {
+        const double slope = (nn * sxy - sx * sy) / den;
+        const double intercept = (sy - slope * sx) / nn;
+        r = (mode == PM_DERIV) ? slope : intercept + slope * param;
+      }
+    } else if (cnt >= 1 && (mode == PM_RESETS || mode == PM_CHANGES)) {
+      double prev = vals[w_lo];
+      int64_t k = 0;
+      for (int64_t j = w_lo + 1; j < w_hi; j++) {
+        const double v = vals[j];
+        if (mode == PM_RESETS ? (v < prev) : (v != prev)) k++;
+        prev = v;
+      }
+      r = (double)k;
+    }
+    out[i] = r;
+  }
+}
+
 // ---------------------------------------------------------------- launchers
 
 static inline int grid_for(int64_t n, int block) {
@@ -370,6 +514,17 @@ void launch_series_last(
     unsigned long long* best_key, hipStream_t stream) {
   hipLaunchKernelGGL(series_last_ts_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
       ts, series, slot_lut, lut_size, ts_lo, ts_hi, n, best_key);
+}
+
+void launch_prom_range_eval(
+    const int64_t* ts, const double* vals, const int64_t* seg_lo,
+    const int64_t* seg_hi, int S, int T, int64_t t0, int64_t step_ms,
+    int64_t range_ms, int64_t offset_ms, double param, int mode, double* out,
+    hipStream_t stream) {
+  const int64_t total = (int64_t)S * T;
+  hipLaunchKernelGGL(prom_range_eval_kernel, dim3(grid_for(total, 256)), dim3(256), 0, stream,
+      ts, vals, seg_lo, seg_hi, S, T, t0, step_ms, range_ms, offset_ms,
+      param, mode, out);
 }
 
 void launch_series_last_row(

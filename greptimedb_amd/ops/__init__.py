@@ -11,6 +11,7 @@ from greptimedb_amd.ops.kernels import (  # noqa: F401
     dedup_mark_last,
     filter_series_time,
     hip_ops_available,
+    prom_range_eval,
     series_last,
     ts_bucket_agg,
 )

@@ -108,6 +108,109 @@ def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
     return best_ts, best_src, best_row
 
 
+PROM_MODES = {
+    "instant": 0, "rate": 1, "increase": 2, "delta": 3,
+    "avg_over_time": 4, "sum_over_time": 5, "min_over_time": 6,
+    "max_over_time": 7, "count_over_time": 8, "last_over_time": 9,
+    "idelta": 10, "irate": 11, "deriv": 12, "predict_linear": 13,
+    "resets": 14, "changes": 15, "stddev_over_time": 16, "stdvar_over_time": 17,
+    "absent_over_time": 18,
+}
+
+
+def prom_range_eval(ts, vals, seg_lo, seg_hi, T, t0, step_ms, range_ms,
+                    offset_ms, param, mode):
+    """Reference implementation of the PromQL window evaluator (see
+    csrc/kernels.hip prom_range_eval_kernel; semantics follow Prometheus'
+    extrapolatedRate & friends)."""
+    import numpy as np
+    ts_h = ts.numpy() if torch.is_tensor(ts) else ts
+    v_h = vals.numpy() if torch.is_tensor(vals) else vals
+    lo_h = seg_lo.numpy() if torch.is_tensor(seg_lo) else seg_lo
+    hi_h = seg_hi.numpy() if torch.is_tensor(seg_hi) else seg_hi
+    S = len(lo_h)
+    out = np.full((S, T), np.nan)
+    for s in range(S):
+        a0, b0 = int(lo_h[s]), int(hi_h[s])
+        tseg = ts_h[a0:b0]
+        vseg = v_h[a0:b0]
+        for t in range(T):
+            te = t0 + t * step_ms - offset_ms
+            tb = te - range_ms
+            w_lo = int(np.searchsorted(tseg, tb, "right"))
+            w_hi = int(np.searchsorted(tseg, te, "right"))
+            cnt = w_hi - w_lo
+            w = vseg[w_lo:w_hi]
+            wt = tseg[w_lo:w_hi]
+            r = np.nan
+            if mode in (0, 9):
+                if cnt:
+                    r = w[-1]
+            elif mode == 8:
+                if cnt:
+                    r = float(cnt)
+            elif mode == 18:
+                r = np.nan if cnt else 1.0
+            elif mode in (4, 5, 6, 7, 16, 17) and cnt:
+                if mode == 5:
+                    r = w.sum()
+                elif mode == 4:
+                    r = w.mean()
+                elif mode == 6:
+                    r = w.min()
+                elif mode == 7:
+                    r = w.max()
+                else:
+                    var = ((w - w.mean()) ** 2).mean()
+                    r = var if mode == 17 else np.sqrt(var)
+            elif mode in (1, 2, 3) and cnt >= 2:
+                is_counter = mode != 3
+                total = w[-1] - w[0]
+                if is_counter:
+                    drops = np.diff(w)
+                    if (drops < 0).any():
+                        total += w[:-1][drops < 0].sum()
+                sampled = (wt[-1] - wt[0]) / 1000.0
+                range_s = range_ms / 1000.0
+                avg_dur = sampled / (cnt - 1)
+                dur_start = (wt[0] - tb) / 1000.0
+                dur_end = (te - wt[-1]) / 1000.0
+                if is_counter and total > 0 and w[0] >= 0:
+                    dz = sampled * (w[0] / total)
+                    dur_start = min(dur_start, dz)
+                thresh = avg_dur * 1.1
+                ext = sampled
+                ext += dur_start if dur_start < thresh else avg_dur / 2
+                ext += dur_end if dur_end < thresh else avg_dur / 2
+                factor = ext / sampled if sampled > 0 else 1.0
+                r = total * factor
+                if mode == 1:
+                    r /= range_s
+            elif mode in (10, 11) and cnt >= 2:
+                dv = w[-1] - w[-2]
+                dt = (wt[-1] - wt[-2]) / 1000.0
+                if mode == 10:
+                    r = dv
+                else:
+                    d = w[-1] if w[-1] < w[-2] else dv
+                    r = d / dt if dt > 0 else np.nan
+            elif mode in (12, 13) and cnt >= 2:
+                x = (wt - te) / 1000.0
+                n = float(cnt)
+                sx, sy = x.sum(), w.sum()
+                sxx, sxy = (x * x).sum(), (x * w).sum()
+                den = n * sxx - sx * sx
+                if den != 0:
+                    slope = (n * sxy - sx * sy) / den
+                    intercept = (sy - slope * sx) / n
+                    r = slope if mode == 12 else intercept + slope * param
+            elif mode in (14, 15) and cnt >= 1:
+                d = np.diff(w)
+                r = float((d < 0).sum()) if mode == 14 else float((d != 0).sum())
+            out[s, t] = r
+    return torch.as_tensor(out)
+
+
 def dedup_mark_last(series, ts):
     """keep[i] ⇔ row i is the last of its (series, ts) group (sorted input)."""
     n = ts.numel()

@@ -57,6 +57,18 @@ def dedup_mark_last(series, ts):
     return cpu_ref.dedup_mark_last(series, ts)
 
 
+def prom_range_eval(ts, vals, seg_lo, seg_hi, T, t0, step_ms, range_ms,
+                    offset_ms, param, mode):
+    """PromQL window evaluator over (slot, ts)-sorted samples → [S, T]."""
+    if torch.is_tensor(ts) and ts.is_cuda:
+        return _require_hip().prom_range_eval(
+            ts.contiguous(), vals.contiguous(), seg_lo.contiguous(),
+            seg_hi.contiguous(), int(T), int(t0), int(step_ms), int(range_ms),
+            int(offset_ms), float(param), int(mode))
+    return cpu_ref.prom_range_eval(ts, vals, seg_lo, seg_hi, T, t0, step_ms,
+                                   range_ms, offset_ms, param, mode)
+
+
 def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
     """See cpu_ref.series_last. GPU path: two atomic-max passes per source."""
     if not sources or not sources[0][0].is_cuda:

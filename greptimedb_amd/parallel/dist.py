@@ -130,6 +130,56 @@ class DistContext:
         g_val[g_val == -np.inf] = np.nan
         return gmap, g_ts_glob, g_val
 
+    def merge_prom_planes(self, keys: list, cnt, s, sq, mn, mx):
+        """Merge PromQL aggregation partial planes [G, T] across ranks.
+        keys are hashable label tuples; cnt/s/sq all-reduce SUM, mn MIN,
+        mx MAX. Returns (global_keys, cnt, s, sq, mn, mx)."""
+        all_keys: list = [None] * self.world
+        dist.all_gather_object(all_keys, keys)
+        merged = sorted({k for ks in all_keys for k in ks})
+        gmap = {k: i for i, k in enumerate(merged)}
+        G = max(len(merged), 1)
+        T = cnt.shape[1]
+        dev = self.coll_device
+
+        def scatter(t, fill=0.0):
+            if t is None:
+                return None
+            out = torch.full((G, T), fill, dtype=torch.float64, device=dev)
+            if keys:
+                idx = torch.as_tensor([gmap[k] for k in keys], device=dev)
+                out[idx] = t.to(dev)
+            return out
+
+        cnt2 = scatter(cnt)
+        s2 = scatter(s)
+        sq2 = scatter(sq)
+        mn2 = scatter(mn, float("inf"))
+        mx2 = scatter(mx, float("-inf"))
+        dist.all_reduce(cnt2, op=dist.ReduceOp.SUM)
+        dist.all_reduce(s2, op=dist.ReduceOp.SUM)
+        if sq2 is not None:
+            dist.all_reduce(sq2, op=dist.ReduceOp.SUM)
+        if mn2 is not None:
+            dist.all_reduce(mn2, op=dist.ReduceOp.MIN)
+        if mx2 is not None:
+            dist.all_reduce(mx2, op=dist.ReduceOp.MAX)
+        return merged, cnt2, s2, sq2, mn2, mx2
+
+    def gather_matrix(self, labels: list, values):
+        """Gather per-rank series matrices (labels + [S, T]) on all ranks."""
+        payload = (labels, values.cpu().numpy())
+        all_p: list = [None] * self.world
+        dist.all_gather_object(all_p, payload)
+        out_labels = []
+        mats = []
+        for ls, m in all_p:
+            out_labels.extend(ls)
+            mats.append(m)
+        import numpy as _np
+        vals = _np.concatenate(mats) if mats else _np.zeros((0, values.shape[1]))
+        return out_labels, torch.as_tensor(vals)
+
     # ------------------------------------------------------------ raw rows
 
     def gather_columns(self, col_data: dict[str, np.ndarray]) -> dict[str, np.ndarray]:

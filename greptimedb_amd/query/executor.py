@@ -168,7 +168,33 @@ class Executor:
             return QueryResult(["Column", "Type", "Semantic"], [names, types, sem])
         if isinstance(stmt, ast.InsertValues):
             return self._exec_insert(stmt)
+        if isinstance(stmt, ast.Tql):
+            return self._exec_tql(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _exec_tql(self, t: ast.Tql) -> QueryResult:
+        """TQL EVAL (start, end, step) expr — PromQL through SQL (reference:
+        src/query/src/promql + sql TQL statement)."""
+        from greptimedb_amd.query.promql.eval import PromEvaluator
+        ev = PromEvaluator(self.engine, dist=self.dist)
+        m = ev.query_range(t.query, t.start, t.end, t.step)
+        label_keys = sorted({k for l in m.labels for k in l if k != "__name__"})
+        ts_col, val_col = [], []
+        label_cols = {k: [] for k in label_keys}
+        vals = m.values.cpu().numpy()
+        for s, labels in enumerate(m.labels):
+            for ti, g in enumerate(m.grid):
+                v = vals[s, ti]
+                if np.isnan(v):
+                    continue
+                ts_col.append(int(g))
+                val_col.append(float(v))
+                for k in label_keys:
+                    label_cols[k].append(labels.get(k))
+        names = ["ts"] + label_keys + ["value"]
+        cols = [np.asarray(ts_col)] + [np.asarray(label_cols[k], dtype=object)
+                                       for k in label_keys] + [np.asarray(val_col)]
+        return QueryResult(names, cols, ["ts"] + [""] * (len(names) - 1))
 
     # ---------------------------------------------------------- DDL / DML
 

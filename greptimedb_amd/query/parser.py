@@ -21,7 +21,7 @@ _TOKEN_RE = re.compile(r"""
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"(?:[^"]|"")*")
   | (?P<id>[A-Za-z_][A-Za-z0-9_.]*)
-  | (?P<op><>|!=|<=|>=|=|<|>|\(|\)|,|\*|\+|-|/|%|;)
+  | (?P<op>=~|!~|<>|!=|<=|>=|=|<|>|\(|\)|\[|\]|\{|\}|,|\*|\+|-|/|%|;)
 """, re.VERBOSE)
 
 _UNITS_MS = {
@@ -51,11 +51,12 @@ def parse_interval_text(text: str) -> int:
 
 
 class Token:
-    __slots__ = ("kind", "value")
+    __slots__ = ("kind", "value", "pos")
 
-    def __init__(self, kind, value):
+    def __init__(self, kind, value, pos=0):
         self.kind = kind
         self.value = value
+        self.pos = pos
 
     def __repr__(self):
         return f"{self.kind}:{self.value}"
@@ -68,21 +69,22 @@ def tokenize(sql: str) -> list[Token]:
         m = _TOKEN_RE.match(sql, pos)
         if not m:
             raise InvalidSyntax(f"bad token at {sql[pos:pos+20]!r}")
+        start = m.start()
         pos = m.end()
         if m.lastgroup is None or m.lastgroup == "comment":
             continue
         kind = m.lastgroup
         v = m.group()
         if kind == "str":
-            out.append(Token("str", v[1:-1].replace("''", "'")))
+            out.append(Token("str", v[1:-1].replace("''", "'"), start))
         elif kind == "qid":
-            out.append(Token("id", v[1:-1].replace('""', '"')))
+            out.append(Token("id", v[1:-1].replace('""', '"'), start))
         elif kind == "num":
-            out.append(Token("num", float(v) if ("." in v or "e" in v or "E" in v) else int(v)))
+            out.append(Token("num", float(v) if ("." in v or "e" in v or "E" in v) else int(v), start))
         elif kind == "id":
-            out.append(Token("id", v))
+            out.append(Token("id", v, start))
         else:
-            out.append(Token("op", v))
+            out.append(Token("op", v, start))
     return out
 
 
@@ -96,6 +98,7 @@ _PRECEDENCE = {
 
 class Parser:
     def __init__(self, sql: str):
+        self.sql = sql
         self.toks = tokenize(sql)
         self.i = 0
 
@@ -349,12 +352,15 @@ class Parser:
         step = parse_interval_text(step_tok.value) / 1000 if isinstance(step_tok.value, str) \
             else float(step_tok.value)
         self.expect_op(")")
-        # rest of token stream is the raw PromQL text — reconstruct naively
-        parts = []
-        while self.peek() is not None and not self.at_op(";"):
-            t = self.next()
-            parts.append(str(t.value) if t.kind != "str" else f'"{t.value}"')
-        return ast.Tql(start, end, step, " ".join(parts))
+        # the rest of the ORIGINAL string is the raw PromQL text (the SQL
+        # lexer cannot tokenize PromQL selectors)
+        if self.peek() is not None:
+            query = self.sql[self.peek().pos:]
+        else:
+            query = ""
+        query = query.rstrip().rstrip(";")
+        self.i = len(self.toks)  # consume everything
+        return ast.Tql(start, end, step, query)
 
     # ---------------- expressions (Pratt) ----------------
     def parse_expr(self, min_prec: int = 0) -> ast.Expr:

@@ -1,0 +1,672 @@
+"""PromQL evaluator: AST → GPU window-evaluation over region shards.
+
+Reference parity: src/promql extension plans (SeriesNormalize /
+InstantManipulate / RangeManipulate / SeriesDivide — K7/K8/K9 in
+SURVEY.md §2.7) + src/query/src/promql/planner.rs. MI355X redesign: instead
+of streaming per-series manipulate operators, a selector is evaluated in ONE
+kernel launch — window rows are gathered (slot, ts)-sorted per region,
+concatenated (regions own disjoint slot ranges), and
+ops.prom_range_eval computes the [S, T] sample matrix (instant lookback or
+range function, Prometheus extrapolation semantics) in one pass.
+Aggregations reduce the matrix on device; cross-rank combine all-reduces
+group planes (parallel/dist.py).
+"""
+
+from __future__ import annotations
+
+import math
+import re as _re
+
+import numpy as np
+import torch
+
+from greptimedb_amd.ops import dedup_mark_last, prom_range_eval
+from greptimedb_amd.ops.cpu_ref import PROM_MODES
+from greptimedb_amd.query.promql import ast
+from greptimedb_amd.query.promql.parser import parse_promql
+from greptimedb_amd.utils.errors import PlanQuery, TableNotFound
+
+DEFAULT_LOOKBACK_S = 300
+
+RANGE_FUNCS = {
+    "rate": "rate", "increase": "increase", "delta": "delta",
+    "idelta": "idelta", "irate": "irate", "deriv": "deriv",
+    "avg_over_time": "avg_over_time", "sum_over_time": "sum_over_time",
+    "min_over_time": "min_over_time", "max_over_time": "max_over_time",
+    "count_over_time": "count_over_time", "last_over_time": "last_over_time",
+    "stddev_over_time": "stddev_over_time", "stdvar_over_time": "stdvar_over_time",
+    "present_over_time": "count_over_time",  # then >0 → 1
+    "changes": "changes", "resets": "resets",
+    "absent_over_time": "absent_over_time",
+}
+
+ELEMENTWISE = {
+    "abs": torch.abs, "ceil": torch.ceil, "floor": torch.floor,
+    "exp": torch.exp, "ln": torch.log, "log2": torch.log2,
+    "log10": torch.log10, "sqrt": torch.sqrt, "sgn": torch.sgn,
+}
+
+
+class PromMatrix:
+    """Evaluated vector: per-series labels + [S, T] value matrix (NaN =
+    no sample) on the query grid."""
+
+    def __init__(self, labels: list[dict], values: torch.Tensor, grid: np.ndarray):
+        self.labels = labels
+        self.values = values
+        self.grid = grid
+
+    @property
+    def S(self):
+        return len(self.labels)
+
+
+class PromScalar:
+    def __init__(self, value, grid):
+        self.value = value  # float
+        self.grid = grid
+
+
+class PromEvaluator:
+    def __init__(self, engine, dist=None, lookback_s: int = DEFAULT_LOOKBACK_S):
+        self.engine = engine
+        self.dist = dist
+        self.lookback_ms = lookback_s * 1000
+
+    # ------------------------------------------------------------ entry
+
+    def query_range(self, q: str, start_s: float, end_s: float, step_s: float) -> PromMatrix:
+        expr = parse_promql(q) if isinstance(q, str) else q
+        t0 = int(start_s * 1000)
+        step = max(int(step_s * 1000), 1)
+        T = int((int(end_s * 1000) - t0) // step) + 1
+        grid = t0 + np.arange(T, dtype=np.int64) * step
+        r = self._eval(expr, t0, step, T, grid)
+        if isinstance(r, PromScalar):
+            vals = torch.full((1, T), float(r.value), dtype=torch.float64)
+            return PromMatrix([{}], vals, grid)
+        return r
+
+    def query_instant(self, q: str, time_s: float) -> PromMatrix:
+        return self.query_range(q, time_s, time_s, 1)
+
+    # ------------------------------------------------------------ eval
+
+    def _eval(self, e, t0, step, T, grid):
+        if isinstance(e, ast.NumberLit):
+            return PromScalar(e.value, grid)
+        if isinstance(e, ast.StringLit):
+            return PromScalar(e.value, grid)
+        if isinstance(e, ast.Unary):
+            r = self._eval(e.expr, t0, step, T, grid)
+            if isinstance(r, PromScalar):
+                return PromScalar(-r.value, grid)
+            return PromMatrix(r.labels, -r.values, grid)
+        if isinstance(e, ast.Selector):
+            if e.range_s:
+                raise PlanQuery("range vector must be wrapped in a function")
+            return self._eval_selector(e, "instant", t0, step, T, grid,
+                                       self.lookback_ms, 0.0)
+        if isinstance(e, ast.Call):
+            return self._eval_call(e, t0, step, T, grid)
+        if isinstance(e, ast.Aggregate):
+            return self._eval_aggregate(e, t0, step, T, grid)
+        if isinstance(e, ast.BinOp):
+            return self._eval_binop(e, t0, step, T, grid)
+        raise PlanQuery(f"promql: unsupported node {type(e).__name__}")
+
+    def _eval_call(self, e: ast.Call, t0, step, T, grid):
+        f = e.func
+        if f in RANGE_FUNCS:
+            param = 0.0
+            sel_idx = 0
+            if f == "quantile_over_time":
+                raise PlanQuery("quantile_over_time not yet supported")
+            if f == "predict_linear":
+                if len(e.args) != 2 or not isinstance(e.args[1], ast.NumberLit):
+                    raise PlanQuery("predict_linear(v[r], t)")
+                param = e.args[1].value
+            sel = e.args[sel_idx]
+            if not isinstance(sel, ast.Selector) or sel.range_s is None:
+                raise PlanQuery(f"{f} needs a range vector argument")
+            m = self._eval_selector(sel, RANGE_FUNCS[f], t0, step, T, grid,
+                                    int(sel.range_s * 1000), param)
+            if f == "present_over_time":
+                v = m.values
+                m = PromMatrix(m.labels, torch.where(v > 0, torch.ones_like(v),
+                                                     torch.full_like(v, float("nan"))), grid)
+            if f in ("rate", "increase", "delta", "idelta", "irate", "deriv",
+                     "predict_linear", "changes", "resets") or "_over_time" in f:
+                m = PromMatrix([_drop_name(l) for l in m.labels], m.values, grid)
+            return m
+        if f == "predict_linear":  # handled above
+            raise PlanQuery("unreachable")
+        if f in ELEMENTWISE:
+            m = self._eval(e.args[0], t0, step, T, grid)
+            if isinstance(m, PromScalar):
+                return PromScalar(float(ELEMENTWISE[f](torch.tensor(m.value))), grid)
+            return PromMatrix([_drop_name(l) for l in m.labels],
+                              ELEMENTWISE[f](m.values), grid)
+        if f == "round":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            to = e.args[1].value if len(e.args) > 1 else 1.0
+            v = torch.round(m.values / to) * to
+            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+        if f in ("clamp", "clamp_min", "clamp_max"):
+            m = self._eval(e.args[0], t0, step, T, grid)
+            v = m.values
+            if f == "clamp":
+                v = v.clamp(e.args[1].value, e.args[2].value)
+            elif f == "clamp_min":
+                v = v.clamp_min(e.args[1].value)
+            else:
+                v = v.clamp_max(e.args[1].value)
+            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+        if f == "scalar":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            if isinstance(m, PromScalar):
+                return m
+            if m.S == 1:
+                return PromMatrix([{}], m.values, grid)
+            return PromMatrix([{}], torch.full((1, T), float("nan"),
+                                               dtype=torch.float64), grid)
+        if f == "vector":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            if isinstance(m, PromScalar):
+                return PromMatrix([{}], torch.full((1, T), float(m.value),
+                                                   dtype=torch.float64), grid)
+            return m
+        if f == "time":
+            return PromMatrix([{}], torch.as_tensor(grid[None, :] / 1000.0,
+                                                    dtype=torch.float64), grid)
+        if f == "timestamp":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            v = torch.where(torch.isnan(m.values),
+                            m.values,
+                            torch.as_tensor(grid[None, :] / 1000.0,
+                                            dtype=torch.float64,
+                                            device=m.values.device))
+            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+        if f == "absent":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            present = (~torch.isnan(m.values)).any(dim=0) if m.S else \
+                torch.zeros(T, dtype=torch.bool)
+            v = torch.where(present, torch.full((T,), float("nan"), dtype=torch.float64),
+                            torch.ones(T, dtype=torch.float64))
+            return PromMatrix([{}], v[None, :], grid)
+        if f == "label_replace":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            dst, repl, src, regex = (a.value for a in e.args[1:5])
+            pat = _re.compile(regex)
+            out = []
+            for l in m.labels:
+                l = dict(l)
+                mm = pat.fullmatch(l.get(src, ""))
+                if mm:
+                    val = mm.expand(repl.replace("$", "\\"))
+                    if val:
+                        l[dst] = val
+                    else:
+                        l.pop(dst, None)
+                out.append(l)
+            return PromMatrix(out, m.values, grid)
+        if f == "label_join":
+            m = self._eval(e.args[0], t0, step, T, grid)
+            dst = e.args[1].value
+            sep = e.args[2].value
+            srcs = [a.value for a in e.args[3:]]
+            out = []
+            for l in m.labels:
+                l = dict(l)
+                l[dst] = sep.join(l.get(s, "") for s in srcs)
+                out.append(l)
+            return PromMatrix(out, m.values, grid)
+        if f == "histogram_quantile":
+            return self._histogram_quantile(e, t0, step, T, grid)
+        raise PlanQuery(f"promql: unsupported function {f}")
+
+    def _histogram_quantile(self, e, t0, step, T, grid):
+        q = e.args[0].value if isinstance(e.args[0], ast.NumberLit) else \
+            self._eval(e.args[0], t0, step, T, grid).value
+        m = self._eval(e.args[1], t0, step, T, grid)
+        # group by labels minus 'le'
+        groups: dict[tuple, list[tuple[float, int]]] = {}
+        keys: dict[tuple, dict] = {}
+        for i, l in enumerate(m.labels):
+            le = l.get("le")
+            if le is None:
+                continue
+            rest = tuple(sorted((k, v) for k, v in l.items() if k not in ("le", "__name__")))
+            groups.setdefault(rest, []).append((float(le) if le != "+Inf" else math.inf, i))
+            keys[rest] = {k: v for k, v in l.items() if k not in ("le", "__name__")}
+        out_labels, rows = [], []
+        vals = m.values
+        for rest, buckets in groups.items():
+            buckets.sort()
+            les = [b[0] for b in buckets]
+            idx = [b[1] for b in buckets]
+            B = vals[idx]  # [nb, T]
+            B = torch.cummax(B, dim=0).values  # enforce monotone
+            total = B[-1]
+            rank = q * total
+            # first bucket with cum >= rank
+            ge = (B >= rank[None, :])
+            first = ge.float().argmax(dim=0)
+            res = torch.full((T,), float("nan"), dtype=torch.float64)
+            les_t = torch.tensor(les, dtype=torch.float64)
+            for t in range(T):  # small T loops acceptable here
+                tot = float(total[t])
+                if not np.isfinite(tot) or tot == 0 or np.isnan(tot):
+                    continue
+                b = int(first[t])
+                hi = les[b]
+                lo = les[b - 1] if b > 0 else 0.0
+                chi = float(B[b, t])
+                clo = float(B[b - 1, t]) if b > 0 else 0.0
+                if math.isinf(hi):
+                    res[t] = les[b - 1] if b > 0 else float("nan")
+                    continue
+                r = float(rank[t])
+                res[t] = lo + (hi - lo) * ((r - clo) / max(chi - clo, 1e-300))
+            out_labels.append(keys[rest])
+            rows.append(res)
+        values = torch.stack(rows) if rows else torch.zeros((0, T), dtype=torch.float64)
+        return PromMatrix(out_labels, values, grid)
+
+    # ------------------------------------------------------------ selector
+
+    def _resolve_table(self, sel: ast.Selector):
+        name = sel.metric
+        field = None
+        for m in sel.matchers:
+            if m.name == "__name__" and m.op == "=":
+                name = m.value
+            if m.name == "__field__" and m.op == "=":
+                field = m.value
+        if name is None:
+            raise PlanQuery("promql: metric name required")
+        try:
+            st = self.engine.table(name)
+        except TableNotFound:
+            # metric "table_field" convention (remote-write flat naming)
+            for tname in self.engine.tables:
+                if name.startswith(tname + "_"):
+                    cand = name[len(tname) + 1:]
+                    if cand in self.engine.tables[tname].regions[0].field_names:
+                        return self.engine.tables[tname], cand
+            return None, None
+        if field is None:
+            fns = st.regions[0].field_names
+            preferred = [f for f in fns if f == "greptime_value"] or fns
+            if len(preferred) != 1:
+                raise PlanQuery(
+                    f"promql: table {name} has {len(fns)} fields; add __field__ matcher")
+            field = preferred[0]
+        return st, field
+
+    def _match_codes(self, region, sel: ast.Selector):
+        """Codes matching all label matchers (None → all)."""
+        tag_names = region.series.tag_names
+        codes = None
+        for m in sel.matchers:
+            if m.name in ("__name__", "__field__"):
+                continue
+            if m.name not in tag_names:
+                # matcher on absent label: = "" / !~ matches-empty keep all,
+                # otherwise empty result
+                if (m.op == "=" and m.value == "") or \
+                   (m.op == "=~" and _re.fullmatch(m.value, "")) or \
+                   (m.op == "!=" and m.value != "") or \
+                   (m.op == "!~" and not _re.fullmatch(m.value, "")):
+                    continue
+                return []
+            inv = region.series.inverted.get(m.name, {})
+            if m.op == "=":
+                got = set(inv.get(m.value, []))
+            elif m.op == "!=":
+                got = set(range(len(region.series))) - set(inv.get(m.value, []))
+            else:
+                pat = _re.compile(m.value)
+                sel_vals = [v for v in inv if pat.fullmatch(v)]
+                got = set()
+                for v in sel_vals:
+                    got.update(inv[v])
+                if m.op == "!~":
+                    got = set(range(len(region.series))) - got
+            codes = got if codes is None else codes & got
+        return None if codes is None else sorted(codes)
+
+    def _eval_selector(self, sel: ast.Selector, func: str, t0, step, T, grid,
+                       range_ms, param) -> PromMatrix:
+        st, field = self._resolve_table(sel)
+        device = self.engine.config.device
+        if st is None:
+            return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
+                                              device=device), grid)
+        offset_ms = int(sel.offset_s * 1000)
+        lo = t0 - offset_ms - range_ms - self.lookback_ms
+        hi = t0 + (T - 1) * step - offset_ms + 1
+
+        labels: list[dict] = []
+        parts = []          # per-region (ts, slots) sorted chunks
+        vparts = []
+        seg_counts = []
+        for region in st.regions:
+            codes = self._match_codes(region, sel)
+            n_codes = len(region.series) if codes is None else len(codes)
+            if n_codes == 0:
+                continue
+            base = len(labels)
+            lut = np.full(len(region.series), -1, dtype=np.int32)
+            it = range(len(region.series)) if codes is None else codes
+            for j, code in enumerate(it):
+                lut[code] = base + j
+                tv = region.series.tag_values[code]
+                l = {t: v for t, v in zip(region.series.tag_names, tv) if v is not None}
+                l["__name__"] = sel.metric or st.schema.name
+                labels.append(l)
+            lut_t = torch.as_tensor(lut, device=device)
+            chunks = []
+            for src in region.scan_sources(lo, hi):
+                p = src.field_pos.get(field)
+                if p is None:
+                    continue
+                from greptimedb_amd.ops import filter_series_time
+                mask = filter_series_time(src.ts, src.series,
+                                          lut_t if codes is not None else None, lo, hi)
+                idx = mask.nonzero(as_tuple=True)[0]
+                if idx.numel() == 0:
+                    continue
+                chunks.append((src.ts[idx], src.series[idx], src.fields[p][idx]))
+            if not chunks:
+                seg_counts.append((base, n_codes, 0))
+                continue
+            ts_t = torch.cat([c[0] for c in chunks])
+            se_t = torch.cat([c[1] for c in chunks])
+            v_t = torch.cat([c[2] for c in chunks])
+            slots = lut_t[se_t.long()]
+            ok = slots >= 0
+            if not bool(ok.all()):
+                ts_t, slots, v_t = ts_t[ok], slots[ok], v_t[ok]
+            # sort by (slot, ts, arrival) then dedup last-wins per ts
+            o = torch.argsort(ts_t, stable=True)
+            perm = o[torch.argsort(slots[o], stable=True)]
+            ts_t, slots, v_t = ts_t[perm], slots[perm], v_t[perm]
+            keep = dedup_mark_last(slots.int().contiguous(), ts_t.contiguous())
+            kidx = keep.nonzero(as_tuple=True)[0]
+            ts_t, slots, v_t = ts_t[kidx], slots[kidx], v_t[kidx]
+            parts.append((ts_t, slots))
+            vparts.append(v_t)
+            seg_counts.append((base, n_codes, ts_t.numel()))
+
+        S = len(labels)
+        if S == 0:
+            return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
+                                              device=device), grid)
+        ts_all = torch.cat([p[0] for p in parts]) if parts else \
+            torch.zeros(0, dtype=torch.int64, device=device)
+        slots_all = torch.cat([p[1] for p in parts]).long() if parts else \
+            torch.zeros(0, dtype=torch.int64, device=device)
+        vals_all = torch.cat(vparts) if vparts else \
+            torch.zeros(0, dtype=torch.float64, device=device)
+        # segments per slot (slots_all ascending because regions own
+        # consecutive slot ranges and each part is slot-sorted)
+        counts = torch.bincount(slots_all, minlength=S)
+        seg_hi = torch.cumsum(counts, 0)
+        seg_lo = seg_hi - counts
+        mode = PROM_MODES["instant"] if func == "instant" else PROM_MODES[func]
+        rng = self.lookback_ms if func == "instant" else range_ms
+        out = prom_range_eval(ts_all.contiguous(), vals_all.contiguous(),
+                              seg_lo.contiguous(), seg_hi.contiguous(),
+                              T, t0, step, rng, offset_ms, param, mode)
+        return PromMatrix(labels, out, grid)
+
+    # ------------------------------------------------------------ aggregate
+
+    def _eval_aggregate(self, e: ast.Aggregate, t0, step, T, grid) -> PromMatrix:
+        m = self._eval(e.expr, t0, step, T, grid)
+        if isinstance(m, PromScalar):
+            raise PlanQuery("aggregate over scalar")
+        op = e.op
+        if op in ("topk", "bottomk"):
+            return self._topk(e, m, grid)
+        group_keys: dict[tuple, int] = {}
+        gidx = []
+        out_labels = []
+        for l in m.labels:
+            key_d = _group_labels(l, e.by, e.without)
+            key = tuple(sorted(key_d.items()))
+            if key not in group_keys:
+                group_keys[key] = len(group_keys)
+                out_labels.append(key_d)
+            gidx.append(group_keys[key])
+        G = max(len(group_keys), 1)
+        dev = m.values.device
+        gi = torch.as_tensor(gidx, dtype=torch.int64, device=dev)
+        v = m.values
+        present = ~torch.isnan(v)
+        v0 = torch.where(present, v, torch.zeros_like(v))
+        cnt = torch.zeros((G, T), dtype=torch.float64, device=dev)
+        if m.S:
+            cnt.index_add_(0, gi, present.double())
+
+        if op == "quantile":
+            if self.dist is not None:
+                raise PlanQuery("distributed quantile aggregation not yet supported")
+            q = e.param.value if isinstance(e.param, ast.NumberLit) else 0.5
+            out = torch.full((G, T), float("nan"), dtype=torch.float64, device=dev)
+            for g in range(G):
+                rows = [i for i, x in enumerate(gidx) if x == g]
+                out[g] = torch.nanquantile(v[rows].float(), q, dim=0).double()
+            out = torch.where(cnt > 0, out, torch.full_like(out, float("nan")))
+            return PromMatrix(out_labels, out, grid)
+
+        # partial planes (mergeable across ranks)
+        s = torch.zeros((G, T), dtype=torch.float64, device=dev)
+        sq = torch.zeros((G, T), dtype=torch.float64, device=dev) \
+            if op in ("stddev", "stdvar") else None
+        mn = mx = None
+        if m.S:
+            s.index_add_(0, gi, v0)
+            if sq is not None:
+                sq.index_add_(0, gi, v0 * v0)
+        if op in ("min", "max"):
+            fill = float("inf") if op == "min" else float("-inf")
+            t = torch.full((G, T), fill, dtype=torch.float64, device=dev)
+            if m.S:
+                vm = torch.where(present, v, torch.full_like(v, fill))
+                t.index_reduce_(0, gi, vm, "amin" if op == "min" else "amax",
+                                include_self=True)
+            if op == "min":
+                mn = t
+            else:
+                mx = t
+
+        if self.dist is not None:
+            out_labels, cnt, s, sq, mn, mx = self.dist.merge_prom_planes(
+                [tuple(sorted(l.items())) for l in out_labels], cnt, s, sq, mn, mx)
+            out_labels = [dict(k) for k in out_labels]
+            G = max(len(out_labels), 1)
+
+        nan = float("nan")
+        if op == "count":
+            out = torch.where(cnt > 0, cnt, torch.full_like(cnt, nan))
+        elif op == "sum":
+            out = torch.where(cnt > 0, s, torch.full_like(s, nan))
+        elif op == "avg":
+            out = torch.where(cnt > 0, s / cnt, torch.full_like(s, nan))
+        elif op in ("stddev", "stdvar"):
+            mean = s / cnt.clamp_min(1)
+            var = (sq / cnt.clamp_min(1) - mean * mean).clamp_min(0)
+            out = torch.where(cnt > 0, var if op == "stdvar" else var.sqrt(),
+                              torch.full_like(var, nan))
+        elif op == "group":
+            out = torch.where(cnt > 0, torch.ones_like(cnt), torch.full_like(cnt, nan))
+        elif op == "min":
+            out = torch.where(cnt > 0, mn, torch.full_like(mn, nan))
+        elif op == "max":
+            out = torch.where(cnt > 0, mx, torch.full_like(mx, nan))
+        else:
+            raise PlanQuery(f"promql: unsupported aggregation {op}")
+        return PromMatrix(out_labels, out, grid)
+
+    def _topk(self, e: ast.Aggregate, m: PromMatrix, grid) -> PromMatrix:
+        k = int(e.param.value) if isinstance(e.param, ast.NumberLit) else 1
+        v = m.values
+        if m.S == 0 or k <= 0:
+            return m
+        desc = e.op == "topk"
+        # rank per timestep; keep sample only when within top/bottom k
+        filled = torch.where(torch.isnan(v),
+                             torch.full_like(v, float("-inf") if desc else float("inf")), v)
+        order = torch.argsort(filled, dim=0, descending=desc)
+        rank = torch.empty_like(order)
+        ar = torch.arange(m.S, device=v.device)[:, None].expand_as(order)
+        rank.scatter_(0, order, ar)
+        keep = rank < k
+        out = torch.where(keep & ~torch.isnan(v), v, torch.full_like(v, float("nan")))
+        used = (~torch.isnan(out)).any(dim=1)
+        idx = used.nonzero(as_tuple=True)[0]
+        return PromMatrix([m.labels[int(i)] for i in idx], out[idx], grid)
+
+    # ------------------------------------------------------------ binop
+
+    def _eval_binop(self, e: ast.BinOp, t0, step, T, grid):
+        l = self._eval(e.left, t0, step, T, grid)
+        r = self._eval(e.right, t0, step, T, grid)
+        if isinstance(l, PromScalar) and isinstance(r, PromScalar):
+            return PromScalar(_scalar_op(e.op, l.value, r.value), grid)
+        if isinstance(l, PromScalar) or isinstance(r, PromScalar):
+            mat, sc, flipped = (r, l, True) if isinstance(l, PromScalar) else (l, r, False)
+            a = mat.values
+            b = torch.as_tensor(float(sc.value), dtype=torch.float64, device=a.device)
+            if flipped:
+                res, keep = _vector_op(e.op, b.expand_as(a), a, e.bool_modifier)
+            else:
+                res, keep = _vector_op(e.op, a, b, e.bool_modifier)
+            labels = mat.labels if e.op not in _CMP_OPS or e.bool_modifier else mat.labels
+            labels = [_drop_name(x) for x in labels]
+            if e.op in _CMP_OPS and not e.bool_modifier:
+                res = torch.where(keep, mat.values, torch.full_like(res, float("nan")))
+            return PromMatrix(labels, res, grid)
+        # vector-vector: one-to-one on matching label sets
+        if e.op in ("and", "or", "unless"):
+            return self._set_op(e.op, l, r, grid)
+        lk = {_match_key(x, e.on, e.ignoring): i for i, x in enumerate(l.labels)}
+        rk = {_match_key(x, e.on, e.ignoring): i for i, x in enumerate(r.labels)}
+        common = [k for k in lk if k in rk]
+        if not common:
+            return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
+                                              device=l.values.device), grid)
+        li = torch.as_tensor([lk[k] for k in common], device=l.values.device)
+        ri = torch.as_tensor([rk[k] for k in common], device=l.values.device)
+        a = l.values[li]
+        b = r.values[ri]
+        res, keep = _vector_op(e.op, a, b, e.bool_modifier)
+        if e.op in _CMP_OPS and not e.bool_modifier:
+            res = torch.where(keep, a, torch.full_like(res, float("nan")))
+        labels = [_drop_name(l.labels[int(i)]) for i in li]
+        return PromMatrix(labels, res, grid)
+
+    def _set_op(self, op, l: PromMatrix, r: PromMatrix, grid):
+        rk = {_match_key(x, None, None) for x in r.labels}
+        if op == "and":
+            idx = [i for i, x in enumerate(l.labels) if _match_key(x, None, None) in rk]
+            rpresent = torch.zeros_like(l.values[0], dtype=torch.bool) if r.S == 0 else None
+            out_rows = []
+            for i in idx:
+                out_rows.append(l.values[i])
+            vals = torch.stack(out_rows) if out_rows else \
+                torch.zeros((0, l.values.shape[1]), dtype=torch.float64,
+                            device=l.values.device)
+            return PromMatrix([l.labels[i] for i in idx], vals, grid)
+        if op == "unless":
+            idx = [i for i, x in enumerate(l.labels) if _match_key(x, None, None) not in rk]
+            vals = l.values[idx] if idx else torch.zeros(
+                (0, l.values.shape[1]), dtype=torch.float64, device=l.values.device)
+            return PromMatrix([l.labels[i] for i in idx], vals, grid)
+        # or: left series + right series not in left
+        lkeys = {_match_key(x, None, None) for x in l.labels}
+        extra = [i for i, x in enumerate(r.labels) if _match_key(x, None, None) not in lkeys]
+        labels = list(l.labels) + [r.labels[i] for i in extra]
+        vals = torch.cat([l.values, r.values[extra]]) if extra else l.values
+        return PromMatrix(labels, vals, grid)
+
+
+_CMP_OPS = {"==", "!=", "<", "<=", ">", ">="}
+
+
+def _scalar_op(op, a, b):
+    if op == "+":
+        return a + b
+    if op == "-":
+        return a - b
+    if op == "*":
+        return a * b
+    if op == "/":
+        return a / b if b != 0 else math.inf if a > 0 else -math.inf if a < 0 else math.nan
+    if op == "%":
+        return math.fmod(a, b) if b != 0 else math.nan
+    if op == "^":
+        return a ** b
+    if op == "==":
+        return 1.0 if a == b else 0.0
+    if op == "!=":
+        return 1.0 if a != b else 0.0
+    if op == "<":
+        return 1.0 if a < b else 0.0
+    if op == "<=":
+        return 1.0 if a <= b else 0.0
+    if op == ">":
+        return 1.0 if a > b else 0.0
+    if op == ">=":
+        return 1.0 if a >= b else 0.0
+    raise PlanQuery(f"promql: op {op}")
+
+
+def _vector_op(op, a, b, bool_mod):
+    if op == "+":
+        return a + b, None
+    if op == "-":
+        return a - b, None
+    if op == "*":
+        return a * b, None
+    if op == "/":
+        return a / b, None
+    if op == "%":
+        return torch.fmod(a, b), None
+    if op == "^":
+        return a ** b, None
+    if op in _CMP_OPS:
+        keep = {"==": a == b, "!=": a != b, "<": a < b,
+                "<=": a <= b, ">": a > b, ">=": a >= b}[op]
+        keep &= ~torch.isnan(a)
+        if torch.is_tensor(b):
+            keep &= ~torch.isnan(b)
+        if bool_mod:
+            nanmask = torch.isnan(a)
+            res = keep.double()
+            res = torch.where(nanmask, torch.full_like(res, float("nan")), res)
+            return res, keep
+        return a, keep
+    raise PlanQuery(f"promql: op {op}")
+
+
+def _drop_name(l: dict) -> dict:
+    return {k: v for k, v in l.items() if k != "__name__"}
+
+
+def _group_labels(l: dict, by, without) -> dict:
+    if by is not None:
+        return {k: l[k] for k in by if k in l}
+    if without is None:
+        return {}  # plain sum(...) collapses all labels
+    drop = set(without) | {"__name__"}
+    return {k: v for k, v in l.items() if k not in drop}
+
+
+def _match_key(l: dict, on, ignoring):
+    if on is not None:
+        return tuple(sorted((k, v) for k, v in l.items() if k in on))
+    drop = set(ignoring or []) | {"__name__"}
+    return tuple(sorted((k, v) for k, v in l.items() if k not in drop))

@@ -1,0 +1,138 @@
+"""PromQL parser + evaluator (CPU path; GPU numerics in test_ops_gpu)."""
+
+import numpy as np
+import pytest
+import torch
+
+from greptimedb_amd.query.executor import Executor
+from greptimedb_amd.query.promql.ast import Aggregate, BinOp, Call, Selector
+from greptimedb_amd.query.promql.eval import PromEvaluator
+from greptimedb_amd.query.promql.parser import parse_duration_s, parse_promql
+
+
+def test_parse_duration():
+    assert parse_duration_s("5m") == 300
+    assert parse_duration_s("1h30m") == 5400
+    assert parse_duration_s("500ms") == 0.5
+
+
+def test_parse_shapes():
+    e = parse_promql('rate(http_requests_total{job="api",code=~"5.."}[5m])')
+    assert isinstance(e, Call) and e.func == "rate"
+    sel = e.args[0]
+    assert isinstance(sel, Selector) and sel.range_s == 300
+    assert {m.name for m in sel.matchers} == {"job", "code"}
+
+    e = parse_promql('sum by (job) (rate(x[1m]))')
+    assert isinstance(e, Aggregate) and e.by == ["job"]
+
+    e = parse_promql('a + b * c')
+    assert isinstance(e, BinOp) and e.op == "+"
+    assert isinstance(e.right, BinOp) and e.right.op == "*"
+
+    e = parse_promql('x offset 5m')
+    assert e.offset_s == 300
+
+
+@pytest.fixture
+def prom_env(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE http_requests_total (job STRING, instance STRING, "
+               "ts TIMESTAMP TIME INDEX, greptime_value DOUBLE, PRIMARY KEY (job, instance))")
+    rows = []
+    for t in range(0, 600, 15):
+        for inst in ("a", "b"):
+            v = t * (1 if inst == "a" else 2)
+            rows.append(f"('api', '{inst}', {t*1000}, {v})")
+    ex.execute("INSERT INTO http_requests_total (job, instance, ts, greptime_value) "
+               "VALUES " + ",".join(rows))
+    return tmp_engine, PromEvaluator(tmp_engine)
+
+
+def test_instant_selector(prom_env):
+    _, ev = prom_env
+    m = ev.query_range("http_requests_total", 0, 600, 60)
+    assert m.S == 2 and m.values.shape == (2, 11)
+    # lookback: value at t is the latest sample ≤ t
+    i_a = [i for i, l in enumerate(m.labels) if l["instance"] == "a"][0]
+    assert float(m.values[i_a][5]) == 300.0
+
+
+def test_rate_linear_counter(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('rate(http_requests_total{instance="a"}[1m])', 120, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 1.0, rtol=1e-12)
+    m = ev.query_range('rate(http_requests_total{instance="b"}[1m])', 120, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 2.0, rtol=1e-12)
+
+
+def test_rate_counter_reset():
+    from greptimedb_amd.ops import cpu_ref
+    ts = torch.tensor([0, 15000, 30000, 45000, 60000], dtype=torch.int64)
+    vals = torch.tensor([100.0, 110.0, 5.0, 15.0, 25.0])  # reset at 30s
+    lo = torch.tensor([0], dtype=torch.int64)
+    hi = torch.tensor([5], dtype=torch.int64)
+    out = cpu_ref.prom_range_eval(ts, vals.double(), lo, hi, 1, 60000, 1, 60000,
+                                  0, 0.0, cpu_ref.PROM_MODES["increase"])
+    # window (0, 60] excludes the t=0 sample: samples 110,5,15,25 over 45s;
+    # raw increase = 25-110 + 110 (reset) = 25; extrapolated ×60/45 = 33.33
+    v = float(out[0][0])
+    assert abs(v - 25.0 * (60.0 / 45.0)) < 1e-9, v
+
+
+def test_sum_and_by(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('sum(rate(http_requests_total[1m]))', 120, 540, 60)
+    assert m.S == 1 and m.labels[0] == {}
+    np.testing.assert_allclose(m.values.numpy(), 3.0, rtol=1e-12)
+    m = ev.query_range('avg by (instance) (http_requests_total)', 300, 300, 1)
+    assert m.S == 2 and all(set(l) == {"instance"} for l in m.labels)
+
+
+def test_binops(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('http_requests_total{instance="a"} * 2 + 1', 300, 300, 1)
+    assert float(m.values[0][0]) == 601
+    m = ev.query_range(
+        'http_requests_total - http_requests_total{instance="a"}', 300, 300, 1)
+    assert m.S == 1 and float(m.values[0][0]) == 0.0
+    m = ev.query_range('http_requests_total > 500', 300, 300, 1)
+    # filter semantics: only instance b (600) survives with original value
+    vals = m.values[:, 0].numpy()
+    assert np.nansum(vals) == 600.0
+
+
+def test_over_time_functions(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('max_over_time(http_requests_total{instance="a"}[2m])',
+                       240, 240, 1)
+    assert float(m.values[0][0]) == 240.0
+    m = ev.query_range('count_over_time(http_requests_total{instance="a"}[1m])',
+                       240, 240, 1)
+    assert float(m.values[0][0]) == 4.0
+
+
+def test_offset(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('http_requests_total{instance="a"} offset 1m', 300, 300, 1)
+    assert float(m.values[0][0]) == 240.0
+
+
+def test_topk(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('topk(1, http_requests_total)', 300, 300, 1)
+    assert m.S == 1 and m.labels[0]["instance"] == "b"
+
+
+def test_absent(prom_env):
+    _, ev = prom_env
+    m = ev.query_range('absent(http_requests_total{instance="zzz"})', 300, 300, 1)
+    assert m.S == 1 and float(m.values[0][0]) == 1.0
+
+
+def test_tql_eval(prom_env):
+    eng, _ = prom_env
+    ex = Executor(eng)
+    r = ex.execute("TQL EVAL (120, 240, '60s') sum(rate(http_requests_total[1m]))")
+    assert len(r) == 3  # 3 grid steps, one series
+    assert abs(r.columns[-1][0] - 3.0) < 1e-9

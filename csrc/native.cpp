@@ -348,6 +348,243 @@ static py::list wal_read_segment(const std::string& path) {
   return out;
 }
 
+// ---------------------------------------------------------------- snappy
+// Snappy block-format decompressor (format: google/snappy format_description.txt).
+// Used for Prometheus remote write request bodies.
+
+static bool snappy_uncompress(const uint8_t* in, size_t n, std::vector<uint8_t>& out) {
+  size_t ip = 0;
+  // preamble: uncompressed length varint
+  uint64_t ulen = 0;
+  int shift = 0;
+  while (ip < n) {
+    uint8_t b = in[ip++];
+    ulen |= (uint64_t)(b & 0x7F) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+    if (shift > 35) return false;
+  }
+  out.clear();
+  out.reserve(ulen);
+  while (ip < n) {
+    const uint8_t tag = in[ip++];
+    const int type = tag & 3;
+    if (type == 0) {  // literal
+      size_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        const int extra = (int)len - 60;
+        if (ip + extra > n) return false;
+        len = 0;
+        for (int i = 0; i < extra; i++) len |= (size_t)in[ip + i] << (8 * i);
+        len += 1;
+        ip += extra;
+      }
+      if (ip + len > n) return false;
+      out.insert(out.end(), in + ip, in + ip + len);
+      ip += len;
+    } else {
+      size_t len, off;
+      if (type == 1) {
+        if (ip >= n) return false;
+        len = ((tag >> 2) & 7) + 4;
+        off = ((size_t)(tag >> 5) << 8) | in[ip++];
+      } else if (type == 2) {
+        if (ip + 2 > n) return false;
+        len = (tag >> 2) + 1;
+        off = in[ip] | ((size_t)in[ip + 1] << 8);
+        ip += 2;
+      } else {
+        if (ip + 4 > n) return false;
+        len = (tag >> 2) + 1;
+        off = in[ip] | ((size_t)in[ip + 1] << 8) |
+              ((size_t)in[ip + 2] << 16) | ((size_t)in[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > out.size()) return false;
+      size_t src = out.size() - off;
+      for (size_t i = 0; i < len; i++) out.push_back(out[src + i]);  // may overlap
+    }
+  }
+  return out.size() == ulen;
+}
+
+static py::bytes py_snappy_uncompress(py::bytes data) {
+  char* buf; Py_ssize_t len;
+  if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+    throw std::runtime_error("expected bytes");
+  std::vector<uint8_t> out;
+  if (!snappy_uncompress(reinterpret_cast<const uint8_t*>(buf), len, out))
+    throw std::runtime_error("snappy: corrupt input");
+  return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
+}
+
+// ------------------------------------------------- prometheus remote write
+// Minimal protobuf wire parser for prometheus.WriteRequest:
+//   WriteRequest{ repeated TimeSeries timeseries = 1 }
+//   TimeSeries{ repeated Label labels = 1; repeated Sample samples = 2 }
+//   Label{ string name = 1; string value = 2 }
+//   Sample{ double value = 1; int64 timestamp = 2 }
+// Label sets are interned (prometheus senders emit sorted labels) to dense
+// series refs like the influx LineParser.
+
+struct PwSlice { const uint8_t* p; size_t n; };
+
+static inline uint64_t pw_varint(const uint8_t*& p, const uint8_t* end) {
+  uint64_t v = 0;
+  int shift = 0;
+  while (p < end) {
+    uint8_t b = *p++;
+    v |= (uint64_t)(b & 0x7F) << shift;
+    if (!(b & 0x80)) return v;
+    shift += 7;
+  }
+  return v;
+}
+
+class PromWriteParser {
+ public:
+  // Returns (series i32[n], ts i64[n] ms, value f64[n],
+  //          new_series [(id, metric, [(name,value)...])])
+  py::tuple parse(py::bytes compressed, bool is_snappy) {
+    char* buf; Py_ssize_t len;
+    if (PyBytes_AsStringAndSize(compressed.ptr(), &buf, &len) != 0)
+      throw std::runtime_error("expected bytes");
+    std::vector<uint8_t> plain;
+    const uint8_t* data;
+    size_t n;
+    if (is_snappy) {
+      if (!snappy_uncompress(reinterpret_cast<const uint8_t*>(buf), len, plain))
+        throw std::runtime_error("snappy: corrupt input");
+      data = plain.data(); n = plain.size();
+    } else {
+      data = reinterpret_cast<const uint8_t*>(buf); n = len;
+    }
+    std::vector<int32_t> series;
+    std::vector<int64_t> ts;
+    std::vector<double> vals;
+    py::list new_series;
+    {
+      const uint8_t* p = data;
+      const uint8_t* end = data + n;
+      while (p < end) {
+        uint64_t key = pw_varint(p, end);
+        if ((key >> 3) == 1 && (key & 7) == 2) {
+          uint64_t tlen = pw_varint(p, end);
+          parse_timeseries(p, p + tlen, series, ts, vals, new_series);
+          p += tlen;
+        } else {
+          skip_field(key & 7, p, end);
+        }
+      }
+    }
+    py::array_t<int32_t> s(series.size());
+    py::array_t<int64_t> t(ts.size());
+    py::array_t<double> v(vals.size());
+    std::memcpy(s.mutable_data(), series.data(), series.size() * 4);
+    std::memcpy(t.mutable_data(), ts.data(), ts.size() * 8);
+    std::memcpy(v.mutable_data(), vals.data(), vals.size() * 8);
+    return py::make_tuple(std::move(s), std::move(t), std::move(v), std::move(new_series));
+  }
+
+  size_t num_series() const { return next_id_; }
+
+ private:
+  static void skip_field(int wt, const uint8_t*& p, const uint8_t* end) {
+    if (wt == 0) { pw_varint(p, end); }
+    else if (wt == 1) { p += 8; }
+    else if (wt == 2) { uint64_t l = pw_varint(p, end); p += l; }
+    else if (wt == 5) { p += 4; }
+    else { p = end; }
+  }
+
+  void parse_timeseries(const uint8_t* p, const uint8_t* end,
+                        std::vector<int32_t>& series, std::vector<int64_t>& ts,
+                        std::vector<double>& vals, py::list& new_series) {
+    // first pass: find label region bounds to build the intern key
+    const uint8_t* q = p;
+    std::string key;  // concatenated raw Label messages — canonical per sender
+    std::vector<std::pair<std::string, std::string>> labels;
+    std::vector<std::pair<double, int64_t>> samples;
+    while (q < end) {
+      uint64_t k = pw_varint(q, end);
+      const int fnum = (int)(k >> 3), wt = (int)(k & 7);
+      if (fnum == 1 && wt == 2) {          // Label
+        uint64_t l = pw_varint(q, end);
+        key.append(reinterpret_cast<const char*>(q), l);
+        key.push_back('\xff');
+        const uint8_t* lp = q;
+        const uint8_t* lend = q + l;
+        std::string lname, lval;
+        while (lp < lend) {
+          uint64_t lk = pw_varint(lp, lend);
+          if ((lk >> 3) == 1 && (lk & 7) == 2) {
+            uint64_t s = pw_varint(lp, lend);
+            lname.assign(reinterpret_cast<const char*>(lp), s);
+            lp += s;
+          } else if ((lk >> 3) == 2 && (lk & 7) == 2) {
+            uint64_t s = pw_varint(lp, lend);
+            lval.assign(reinterpret_cast<const char*>(lp), s);
+            lp += s;
+          } else {
+            skip_field(lk & 7, lp, lend);
+          }
+        }
+        labels.emplace_back(std::move(lname), std::move(lval));
+        q += l;
+      } else if (fnum == 2 && wt == 2) {   // Sample
+        uint64_t l = pw_varint(q, end);
+        const uint8_t* sp = q;
+        const uint8_t* send = q + l;
+        double v = std::nan("");
+        int64_t t = 0;
+        while (sp < send) {
+          uint64_t sk = pw_varint(sp, send);
+          if ((sk >> 3) == 1 && (sk & 7) == 1) {
+            std::memcpy(&v, sp, 8); sp += 8;
+          } else if ((sk >> 3) == 2 && (sk & 7) == 0) {
+            t = (int64_t)pw_varint(sp, send);
+          } else {
+            skip_field(sk & 7, sp, send);
+          }
+        }
+        samples.emplace_back(v, t);
+        q += l;
+      } else {
+        skip_field(wt, q, end);
+      }
+    }
+    if (samples.empty()) return;
+    const uint64_t h = fnv1a(key.data(), key.size());
+    int32_t sid = -1;
+    auto range = ids_.equal_range(h);
+    for (auto it = range.first; it != range.second; ++it) {
+      if (keys_[it->second] == key) { sid = it->second; break; }
+    }
+    if (sid < 0) {
+      sid = next_id_++;
+      ids_.emplace(h, sid);
+      keys_.resize(std::max<size_t>(keys_.size(), sid + 1));
+      keys_[sid] = key;
+      std::string metric;
+      py::list ls;
+      for (auto& [ln, lv] : labels) {
+        if (ln == "__name__") metric = lv;
+        else ls.append(py::make_tuple(ln, lv));
+      }
+      new_series.append(py::make_tuple(sid, metric, std::move(ls)));
+    }
+    for (auto& [v, t] : samples) {
+      series.push_back(sid);
+      ts.push_back(t);
+      vals.push_back(v);
+    }
+  }
+
+  std::unordered_multimap<uint64_t, int32_t> ids_;
+  std::vector<std::string> keys_;
+  int32_t next_id_ = 0;
+};
+
 PYBIND11_MODULE(_native, m) {
   m.doc() = "greptimedb_amd host-native ingest path (line parser + WAL)";
   py::class_<LineParser>(m, "LineParser")
@@ -365,4 +602,9 @@ PYBIND11_MODULE(_native, m) {
       .def("commit", &WalWriter::commit, py::arg("sync") = true)
       .def("segment_bytes", &WalWriter::segment_bytes);
   m.def("wal_read_segment", &wal_read_segment);
+  m.def("snappy_uncompress", &py_snappy_uncompress);
+  py::class_<PromWriteParser>(m, "PromWriteParser")
+      .def(py::init<>())
+      .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)
+      .def("num_series", &PromWriteParser::num_series);
 }

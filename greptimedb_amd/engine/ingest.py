@@ -129,8 +129,9 @@ class Ingestor:
         self._table_field_map[name] = (m, len(parser_fields))
         return m
 
-    def ingest_lines(self, data: bytes) -> int:
-        """Parse + route + WAL + memtable-append one wire batch. Returns rows."""
+    def ingest_lines(self, data: bytes, ts_scale_to_ns: int = 1) -> int:
+        """Parse + route + WAL + memtable-append one wire batch. Returns rows.
+        ts_scale_to_ns: multiplier for non-ns influx `precision` values."""
         series, ts_ns, fields, new_tagsets = self.parser.parse(data)
         n = len(series)
         if n == 0:
@@ -140,6 +141,8 @@ class Ingestor:
         parser_fields = self.parser.field_names()
         fields_mat = np.stack([fields[fn] for fn in parser_fields]) if parser_fields \
             else np.zeros((0, n))
+        if ts_scale_to_ns != 1:
+            ts_ns = ts_ns * ts_scale_to_ns
         ts_ms = ts_ns // 1_000_000
 
         region_of = self.sid_region[series]

@@ -64,6 +64,38 @@ def encode_pk(values: tuple[str, ...]) -> bytes:
     return bytes(out)
 
 
+SPARSE_MARKER = 0xFF
+
+
+def encode_sparse(labels: dict) -> bytes:
+    """Sparse pk (reference: mito-codec SparsePrimaryKeyCodec, used by the
+    metric engine): self-describing sorted (name, value) pairs, prefixed
+    with a marker byte so dense and sparse keys are distinguishable."""
+    out = bytearray([SPARSE_MARKER])
+    for name in sorted(labels):
+        v = labels[name]
+        if v is None:
+            continue
+        out += encode_string(name.encode())
+        out += encode_string(v.encode() if isinstance(v, str) else v)
+    return bytes(out)
+
+
+def decode_sparse(buf: bytes) -> dict:
+    assert buf[0] == SPARSE_MARKER
+    off = 1
+    out = {}
+    while off < len(buf):
+        name, off = decode_string(buf, off)
+        val, off = decode_string(buf, off)
+        out[name.decode()] = val.decode()
+    return out
+
+
+def is_sparse(buf: bytes) -> bool:
+    return len(buf) > 0 and buf[0] == SPARSE_MARKER
+
+
 def decode_pk(buf: bytes, n_cols: int) -> tuple:
     out = []
     off = 0

@@ -86,6 +86,16 @@ class Region:
             self._series_log.flush()
         return code
 
+    def register_series_labels(self, labels: dict) -> int:
+        """Sparse/metric-engine mode: dynamic label-set series."""
+        prev = len(self.series)
+        code = self.series.get_or_create_labels(labels)
+        if code >= prev:
+            pk = self.series.pks[code]
+            self._series_log.write(struct.pack("<I", len(pk)) + pk)
+            self._series_log.flush()
+        return code
+
     # ---------------------------------------------------------------- open
 
     def _load_ssts(self):

@@ -56,12 +56,37 @@ class SeriesIndex:
         return code
 
     def add_encoded(self, pk: bytes) -> int:
-        """Register a series seen only as encoded pk (SST load path)."""
+        """Register a series seen only as encoded pk (SST load / series log)."""
         code = self.pk_to_code.get(pk)
         if code is not None:
             return code
+        if pk_codec.is_sparse(pk):
+            return self.add_labels(pk, pk_codec.decode_sparse(pk))
         tags = pk_codec.decode_pk(pk, len(self.tag_names))
         return self.add(pk, tags)
+
+    # -------- sparse / metric-engine mode: dynamic label sets --------
+
+    def get_or_create_labels(self, labels: dict) -> int:
+        pk = pk_codec.encode_sparse(labels)
+        code = self.pk_to_code.get(pk)
+        if code is None:
+            code = self.add_labels(pk, labels)
+        return code
+
+    def add_labels(self, pk: bytes, labels: dict) -> int:
+        for name in labels:
+            if name not in self.inverted:
+                self.tag_names.append(name)
+                self.inverted[name] = {}
+        code = len(self.pks)
+        self.pk_to_code[pk] = code
+        self.pks.append(pk)
+        self.tag_values.append(tuple(labels.get(n) for n in self.tag_names))
+        for n, v in labels.items():
+            if v is not None:
+                self.inverted[n].setdefault(v, []).append(code)
+        return code
 
     # ---------------- tag predicate → slot LUT helpers ----------------
 
@@ -76,6 +101,13 @@ class SeriesIndex:
         return out
 
     def tag_array(self, tag: str) -> np.ndarray:
-        """Object array of this tag's value per code (group-by output)."""
+        """Object array of this tag's value per code (group-by output).
+        Sparse mode: rows registered before a label first appeared have
+        shorter tuples → None."""
         i = self.tag_names.index(tag)
-        return np.array([t[i] for t in self.tag_values], dtype=object)
+        return np.array([t[i] if i < len(t) else None for t in self.tag_values],
+                        dtype=object)
+
+    def labels_of(self, code: int) -> dict:
+        t = self.tag_values[code]
+        return {n: v for n, v in zip(self.tag_names, t) if v is not None}

@@ -288,7 +288,13 @@ class PromEvaluator:
         try:
             st = self.engine.table(name)
         except TableNotFound:
-            # metric "table_field" convention (remote-write flat naming)
+            # metric-engine physical table (remote-write metrics live there,
+            # multiplexed by the __name__ label — engine/promstore.py)
+            from greptimedb_amd.engine.promstore import PHYSICAL_TABLE, VALUE_FIELD
+            phys = self.engine.tables.get(PHYSICAL_TABLE)
+            if phys is not None:
+                return phys, field or VALUE_FIELD
+            # metric "table_field" convention (flat naming)
             for tname in self.engine.tables:
                 if name.startswith(tname + "_"):
                     cand = name[len(tname) + 1:]
@@ -308,8 +314,15 @@ class PromEvaluator:
         """Codes matching all label matchers (None → all)."""
         tag_names = region.series.tag_names
         codes = None
-        for m in sel.matchers:
-            if m.name in ("__name__", "__field__"):
+        matchers = list(sel.matchers)
+        # metric-engine regions carry __name__ as an ordinary label
+        if "__name__" in region.series.inverted:
+            if sel.metric:
+                matchers.append(ast.Matcher("__name__", "=", sel.metric))
+        else:
+            matchers = [m for m in matchers if m.name != "__name__"]
+        for m in matchers:
+            if m.name == "__field__":
                 continue
             if m.name not in tag_names:
                 # matcher on absent label: = "" / !~ matches-empty keep all,

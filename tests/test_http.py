@@ -1,0 +1,118 @@
+"""HTTP server endpoints (fastapi TestClient; reference: servers/src/http.rs)."""
+
+import struct
+
+import pytest
+
+pytest.importorskip("fastapi")
+from fastapi.testclient import TestClient  # noqa: E402
+
+from greptimedb_amd.servers.http import ServerContext, build_app  # noqa: E402
+
+
+@pytest.fixture
+def client(tmp_engine):
+    ctx = ServerContext(tmp_engine)
+    return TestClient(build_app(ctx))
+
+
+def _pb_varint(v):
+    out = b""
+    while True:
+        b7 = v & 0x7F
+        v >>= 7
+        out += bytes([b7 | (0x80 if v else 0)])
+        if not v:
+            return out
+
+
+def _pb_str(fnum, s):
+    s = s.encode() if isinstance(s, str) else s
+    return _pb_varint((fnum << 3) | 2) + _pb_varint(len(s)) + s
+
+
+def _label(n, v):
+    return _pb_str(1, _pb_str(1, n) + _pb_str(2, v))
+
+
+def _sample(val, ts):
+    return _pb_str(2, _pb_varint((1 << 3) | 1) + struct.pack("<d", val) +
+                   _pb_varint(2 << 3) + _pb_varint(ts))
+
+
+def _ts_msg(labels, samples):
+    return _pb_str(1, b"".join(_label(n, v) for n, v in labels) +
+                   b"".join(_sample(v, t) for v, t in samples))
+
+
+def test_health(client):
+    assert client.get("/health").status_code == 200
+
+
+def test_sql_roundtrip(client):
+    r = client.post("/v1/sql", params={"sql":
+        "CREATE TABLE t (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, PRIMARY KEY (h))"})
+    assert r.status_code == 200 and "output" in r.json()
+    client.post("/v1/sql", params={"sql":
+        "INSERT INTO t (h, ts, v) VALUES ('a', 1000, 1.5)"})
+    r = client.get("/v1/sql", params={"sql": "SELECT h, ts, v FROM t"})
+    body = r.json()
+    rows = body["output"][0]["records"]["rows"]
+    assert rows == [["a", 1000, 1.5]]
+    cols = [c["name"] for c in body["output"][0]["records"]["schema"]["column_schemas"]]
+    assert cols == ["h", "ts", "v"]
+
+
+def test_sql_error(client):
+    r = client.get("/v1/sql", params={"sql": "SELEKT 1"})
+    assert "error" in r.json()
+
+
+def test_influx_write_and_query(client):
+    lines = b"weather,city=sf temp=13.5 1000000000\nweather,city=la temp=22.0 1000000000\n"
+    r = client.post("/v1/influxdb/write", content=lines)
+    assert r.status_code == 204
+    r = client.get("/v1/sql", params={"sql": "SELECT city, temp FROM weather ORDER BY city"})
+    rows = r.json()["output"][0]["records"]["rows"]
+    assert rows == [["la", 22.0], ["sf", 13.5]]
+
+
+def test_influx_precision_ms(client):
+    client.post("/v1/influxdb/write", params={"precision": "ms"},
+                content=b"m1,h=x v=1 1500\n")
+    r = client.get("/v1/sql", params={"sql": "SELECT ts FROM m1"})
+    assert r.json()["output"][0]["records"]["rows"] == [[1500]]
+
+
+def test_remote_write_and_promql(client):
+    req = _ts_msg([("__name__", "up"), ("job", "api")], [(1.0, 1000), (1.0, 61000)]) + \
+          _ts_msg([("__name__", "up"), ("job", "db")], [(0.0, 61000)])
+    r = client.post("/v1/prometheus/write", content=req,
+                    headers={"content-encoding": "identity"})
+    assert r.status_code == 204
+    r = client.get("/v1/prometheus/api/v1/query",
+                   params={"query": "up", "time": "61"})
+    data = r.json()
+    assert data["status"] == "success"
+    assert len(data["data"]["result"]) == 2
+    r = client.get("/v1/prometheus/api/v1/query_range",
+                   params={"query": "sum(up)", "start": "0", "end": "120", "step": "60"})
+    res = r.json()["data"]["result"]
+    assert res and res[0]["metric"] == {}
+    r = client.get("/v1/prometheus/api/v1/label/__name__/values")
+    assert "up" in r.json()["data"]
+    r = client.get("/v1/prometheus/api/v1/labels")
+    assert "job" in r.json()["data"]
+    r = client.get("/v1/prometheus/api/v1/series", params={"match[]": 'up{job="api"}'})
+    assert r.json()["data"] == [{"job": "api", "__name__": "up"}]
+
+
+def test_metrics_endpoint(client):
+    client.get("/v1/sql", params={"sql": "SELECT 1"})
+    r = client.get("/metrics")
+    assert "greptime_http_sql_requests" in r.text
+
+
+def test_status(client):
+    r = client.get("/status")
+    assert "tables" in r.json()

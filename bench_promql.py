@@ -1,0 +1,153 @@
+#!/usr/bin/env python3
+"""PromQL / metric-engine benchmark (BASELINE config 3 shape).
+
+Bulk-loads an N-series metric-engine fixture (labels __name__/job/instance,
+one sample per 15s) directly into device memory, then times the PromQL
+shapes the config names: rate() + sum by() over the sharded series set.
+Distributed: hosts shard by series hash; aggregation planes all-reduce.
+
+  python bench_promql.py --series 1000000 --metrics 100 --minutes 60
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import tempfile
+import time
+
+import numpy as np
+import torch
+
+START_MS = 1451606400000
+INTERVAL_MS = 15_000
+
+
+def load_metric_fixture(engine, n_series: int, n_metrics: int, minutes: int,
+                        rank: int = 0, world: int = 1, seed: int = 3):
+    from greptimedb_amd.engine import sst as sst_mod
+    from greptimedb_amd.engine.promstore import PromStore, VALUE_FIELD
+
+    store = PromStore(engine, durable=False)
+    st = store.table
+    device = engine.config.device
+    gen_dev = device if str(device).startswith("cuda") else "cpu"
+
+    my = [k for k in range(n_series) if world == 1 or (k % world) == rank]
+    n_jobs = max(n_series // 1000, 4)
+    # register per region
+    regions = len(st.regions)
+    per_region: dict[int, list[int]] = {}
+    for k in my:
+        per_region.setdefault(k % regions, []).append(k)
+    T = minutes * 60_000 // INTERVAL_MS
+    ts_e = torch.arange(T, dtype=torch.int64, device=gen_dev) * INTERVAL_MS + START_MS
+    g = torch.Generator(device=gen_dev).manual_seed(seed + rank)
+    total = 0
+    for ridx, ks in per_region.items():
+        region = st.regions[ridx]
+        labels_list = [
+            {"__name__": f"metric_{k % n_metrics}",
+             "job": f"job_{(k // n_metrics) % n_jobs}",
+             "instance": f"inst_{k}"}
+            for k in ks
+        ]
+        codes = region.register_series_bulk(labels_list)
+        order = np.argsort(codes)
+        codes_sorted = torch.as_tensor(codes[order].astype(np.int32), device=gen_dev)
+        H = len(ks)
+        # counter-style values: monotonically increasing per series
+        incr = torch.rand((H, T), generator=g, dtype=torch.float64, device=gen_dev)
+        vals = incr.cumsum(dim=1)[order]
+        ts_flat = ts_e.repeat(H)
+        se_flat = codes_sorted.repeat_interleave(T)
+        fields = vals.reshape(1, H * T)
+        batch = sst_mod.SstBatch(
+            ts_flat.to(device), se_flat.to(device), fields.to(device), None,
+            int(START_MS), int(START_MS + (T - 1) * INTERVAL_MS), [VALUE_FIELD])
+        region.sst_cache[f"promfix_{ridx}"] = batch
+        total += H * T
+    return store, total, n_jobs
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--series", type=int, default=1_000_000)
+    ap.add_argument("--metrics", type=int, default=100)
+    ap.add_argument("--minutes", type=int, default=60)
+    ap.add_argument("--iters", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    have_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+    if world > 1:
+        import torch.distributed as dist
+        dist.init_process_group("nccl" if have_gpu else "gloo")
+
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.parallel.dist import DistContext
+    from greptimedb_amd.query.promql.eval import PromEvaluator
+
+    base = tempfile.mkdtemp(prefix="gdb_prom_")
+    eng = MitoEngine(EngineConfig(data_dir=os.path.join(base, f"r{rank}"),
+                                  device=device, background_flush=False))
+    t0 = time.perf_counter()
+    store, n_samples, n_jobs = load_metric_fixture(
+        eng, args.series, args.metrics, args.minutes, rank, world)
+    dist_ctx = DistContext(device=device) if world > 1 else None
+    tot = n_samples if dist_ctx is None else int(dist_ctx.all_sum(n_samples))
+    if rank == 0:
+        print(f"# fixture: {tot} samples, {args.series} series, "
+              f"{args.metrics} metrics in {time.perf_counter()-t0:.1f}s", flush=True)
+
+    ev = PromEvaluator(eng, dist=dist_ctx)
+    end_s = (START_MS + args.minutes * 60_000) / 1000 - 60
+    start_s = end_s - 1800  # 30m range
+    queries = {
+        "instant-one-metric": ("metric_0", end_s, end_s, 1),
+        "rate-sum-one-metric": (f"sum(rate(metric_1[5m]))", start_s, end_s, 60),
+        "rate-sum-by-job-one-metric": ("sum by (job) (rate(metric_2[5m]))",
+                                       start_s, end_s, 60),
+        "rate-sum-by-name-all": ('sum by (__name__) (rate({__name__=~"metric_.*"}[5m]))',
+                                 start_s, end_s, 60),
+        "sum-by-job-all-series": ('sum by (job) ({__name__=~"metric_.*"})',
+                                  start_s, end_s, 60),
+    }
+    results = {}
+    for name, (q, s, e, stp) in queries.items():
+        times = []
+        S = 0
+        for i in range(args.warmup + args.iters):
+            if dist_ctx:
+                dist_ctx.barrier()
+            if have_gpu:
+                torch.cuda.synchronize()
+            qt0 = time.perf_counter()
+            m = ev.query_range(q, s, e, stp)
+            if have_gpu:
+                torch.cuda.synchronize()
+            dt = (time.perf_counter() - qt0) * 1000
+            if i >= args.warmup:
+                times.append(dt)
+            S = m.S
+        results[name] = {"p50_ms": round(float(np.median(times)), 2), "series_out": S}
+        if rank == 0:
+            print(f"# {name}: p50 {np.median(times):.1f} ms → {S} series", flush=True)
+    if rank == 0:
+        print(json.dumps({"bench": "promql-metric-engine", "series": args.series * 1,
+                          "n_gpus": world, "samples": tot, "queries": results}), flush=True)
+    eng.close()
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

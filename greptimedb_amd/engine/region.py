@@ -96,6 +96,22 @@ class Region:
             self._series_log.flush()
         return code
 
+    def register_series_bulk(self, labels_list) -> np.ndarray:
+        """Bulk sparse registration (fixture/bulk-ingest path): one series-log
+        flush for the whole batch."""
+        prev = len(self.series)
+        codes = np.empty(len(labels_list), dtype=np.int32)
+        for i, labels in enumerate(labels_list):
+            codes[i] = self.series.get_or_create_labels(labels)
+        buf = bytearray()
+        for code in range(prev, len(self.series)):
+            pk = self.series.pks[code]
+            buf += struct.pack("<I", len(pk)) + pk
+        if buf:
+            self._series_log.write(bytes(buf))
+            self._series_log.flush()
+        return codes
+
     # ---------------------------------------------------------------- open
 
     def _load_ssts(self):

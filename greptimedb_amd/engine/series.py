@@ -34,6 +34,7 @@ class SeriesIndex:
         self.tag_values: list[tuple] = []
         # tag name -> value -> list[int] codes
         self.inverted: dict[str, dict[str, list[int]]] = {t: {} for t in tag_names}
+        self._codes_cache: dict = {}
 
     def __len__(self) -> int:
         return len(self.pks)
@@ -111,3 +112,20 @@ class SeriesIndex:
     def labels_of(self, code: int) -> dict:
         t = self.tag_values[code]
         return {n: v for n, v in zip(self.tag_names, t) if v is not None}
+
+    def tag_codes(self, tag: str):
+        """Factorized tag column: (value_id i32[n_series] (-1 = absent),
+        values list). Built from the inverted index (vectorized over unique
+        values, not series) and cached until new series arrive."""
+        cached = self._codes_cache.get(tag)
+        n = len(self.pks)
+        if cached is not None and cached[2] == n:
+            return cached[0], cached[1]
+        arr = np.full(n, -1, dtype=np.int32)
+        values = []
+        inv = self.inverted.get(tag, {})
+        for i, (v, codes) in enumerate(inv.items()):
+            values.append(v)
+            arr[np.asarray(codes, dtype=np.int64)] = i
+        self._codes_cache[tag] = (arr, values, n)
+        return arr, values

@@ -49,16 +49,61 @@ ELEMENTWISE = {
 
 class PromMatrix:
     """Evaluated vector: per-series labels + [S, T] value matrix (NaN =
-    no sample) on the query grid."""
+    no sample) on the query grid.
 
-    def __init__(self, labels: list[dict], values: torch.Tensor, grid: np.ndarray):
-        self.labels = labels
+    Labels are stored COLUMNAR (label name → np object array of length S) so
+    10M-series selectors never build 10M Python dicts; `labels` materializes
+    dicts lazily (cheap for post-aggregation results)."""
+
+    def __init__(self, labels, values: torch.Tensor, grid: np.ndarray,
+                 label_cols: dict | None = None, n_series: int | None = None):
+        self._labels = labels          # list[dict] | None
+        self.label_cols = label_cols   # {name: np object array [S]} | None
+        self._n = n_series
         self.values = values
         self.grid = grid
 
     @property
     def S(self):
-        return len(self.labels)
+        if self._labels is not None:
+            return len(self._labels)
+        if self._n is not None:
+            return self._n
+        return int(self.values.shape[0])
+
+    @property
+    def labels(self) -> list:
+        if self._labels is None:
+            cols = self.label_cols or {}
+            out = [{} for _ in range(self.S)]
+            for n, (codes, values) in cols.items():
+                for i in range(len(codes)):
+                    c = codes[i]
+                    if c >= 0:
+                        out[i][n] = values[c]
+            self._labels = out
+        return self._labels
+
+    @labels.setter
+    def labels(self, v):
+        self._labels = v
+        self.label_cols = None
+
+    def get_label_cols(self) -> dict:
+        """Factorized columnar labels: {name: (value_id i32[S], values)}."""
+        if self.label_cols is not None:
+            return self.label_cols
+        names = sorted({k for l in self.labels for k in l})
+        out = {}
+        for n in names:
+            vmap: dict = {}
+            codes = np.full(self.S, -1, dtype=np.int32)
+            for i, l in enumerate(self.labels):
+                v = l.get(n)
+                if v is not None:
+                    codes[i] = vmap.setdefault(v, len(vmap))
+            out[n] = (codes, list(vmap))
+        return out
 
 
 class PromScalar:
@@ -133,11 +178,11 @@ class PromEvaluator:
                                     int(sel.range_s * 1000), param)
             if f == "present_over_time":
                 v = m.values
-                m = PromMatrix(m.labels, torch.where(v > 0, torch.ones_like(v),
-                                                     torch.full_like(v, float("nan"))), grid)
+                m = _matrix_map(m, torch.where(v > 0, torch.ones_like(v),
+                                               torch.full_like(v, float("nan"))))
             if f in ("rate", "increase", "delta", "idelta", "irate", "deriv",
                      "predict_linear", "changes", "resets") or "_over_time" in f:
-                m = PromMatrix([_drop_name(l) for l in m.labels], m.values, grid)
+                m = _matrix_map(m, drop_name=True)
             return m
         if f == "predict_linear":  # handled above
             raise PlanQuery("unreachable")
@@ -145,13 +190,12 @@ class PromEvaluator:
             m = self._eval(e.args[0], t0, step, T, grid)
             if isinstance(m, PromScalar):
                 return PromScalar(float(ELEMENTWISE[f](torch.tensor(m.value))), grid)
-            return PromMatrix([_drop_name(l) for l in m.labels],
-                              ELEMENTWISE[f](m.values), grid)
+            return _matrix_map(m, ELEMENTWISE[f](m.values), drop_name=True)
         if f == "round":
             m = self._eval(e.args[0], t0, step, T, grid)
             to = e.args[1].value if len(e.args) > 1 else 1.0
             v = torch.round(m.values / to) * to
-            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+            return _matrix_map(m, v, drop_name=True)
         if f in ("clamp", "clamp_min", "clamp_max"):
             m = self._eval(e.args[0], t0, step, T, grid)
             v = m.values
@@ -161,7 +205,7 @@ class PromEvaluator:
                 v = v.clamp_min(e.args[1].value)
             else:
                 v = v.clamp_max(e.args[1].value)
-            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+            return _matrix_map(m, v, drop_name=True)
         if f == "scalar":
             m = self._eval(e.args[0], t0, step, T, grid)
             if isinstance(m, PromScalar):
@@ -186,7 +230,7 @@ class PromEvaluator:
                             torch.as_tensor(grid[None, :] / 1000.0,
                                             dtype=torch.float64,
                                             device=m.values.device))
-            return PromMatrix([_drop_name(l) for l in m.labels], v, grid)
+            return _matrix_map(m, v, drop_name=True)
         if f == "absent":
             m = self._eval(e.args[0], t0, step, T, grid)
             present = (~torch.isnan(m.values)).any(dim=0) if m.S else \
@@ -284,6 +328,12 @@ class PromEvaluator:
             if m.name == "__field__" and m.op == "=":
                 field = m.value
         if name is None:
+            # bare label-matcher selector: evaluate against the metric-engine
+            # physical table when present (all metrics share it)
+            from greptimedb_amd.engine.promstore import PHYSICAL_TABLE, VALUE_FIELD
+            phys = self.engine.tables.get(PHYSICAL_TABLE)
+            if phys is not None:
+                return phys, field or VALUE_FIELD
             raise PlanQuery("promql: metric name required")
         try:
             st = self.engine.table(name)
@@ -360,7 +410,10 @@ class PromEvaluator:
         lo = t0 - offset_ms - range_ms - self.lookback_ms
         hi = t0 + (T - 1) * step - offset_ms + 1
 
-        labels: list[dict] = []
+        S_total = 0
+        col_parts: dict[str, list] = {}   # tag -> [(codes slice, values)...] per region
+        region_sizes: list[int] = []
+        all_tags: list[str] = []
         parts = []          # per-region (ts, slots) sorted chunks
         vparts = []
         seg_counts = []
@@ -369,15 +422,24 @@ class PromEvaluator:
             n_codes = len(region.series) if codes is None else len(codes)
             if n_codes == 0:
                 continue
-            base = len(labels)
+            base = S_total
             lut = np.full(len(region.series), -1, dtype=np.int32)
-            it = range(len(region.series)) if codes is None else codes
-            for j, code in enumerate(it):
-                lut[code] = base + j
-                tv = region.series.tag_values[code]
-                l = {t: v for t, v in zip(region.series.tag_names, tv) if v is not None}
-                l["__name__"] = sel.metric or st.schema.name
-                labels.append(l)
+            if codes is None:
+                lut[:] = base + np.arange(len(region.series), dtype=np.int32)
+            else:
+                ca = np.asarray(codes, dtype=np.int64)
+                lut[ca] = base + np.arange(len(ca), dtype=np.int32)
+            # factorized label columns for the matched codes (vectorized)
+            sel_idx = np.arange(len(region.series)) if codes is None else \
+                np.asarray(codes, dtype=np.int64)
+            for t in region.series.tag_names:
+                tcodes, tvalues = region.series.tag_codes(t)
+                if t not in col_parts:
+                    col_parts[t] = []
+                    all_tags.append(t)
+                col_parts[t].append((len(region_sizes), tcodes[sel_idx], tvalues))
+            region_sizes.append(n_codes)
+            S_total += n_codes
             lut_t = torch.as_tensor(lut, device=device)
             chunks = []
             for src in region.scan_sources(lo, hi):
@@ -412,10 +474,34 @@ class PromEvaluator:
             vparts.append(v_t)
             seg_counts.append((base, n_codes, ts_t.numel()))
 
-        S = len(labels)
+        S = S_total
         if S == 0:
             return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
                                               device=device), grid)
+        # merge per-region factor columns (remap value ids into a shared
+        # vocabulary; vectorized over unique values, not series)
+        label_cols: dict = {}
+        for t in all_tags:
+            vocab: dict = {}
+            out_codes = np.full(S, -1, dtype=np.int32)
+            off = 0
+            by_region = {ri: (c, v) for ri, c, v in col_parts[t]}
+            for ri, sz in enumerate(region_sizes):
+                got = by_region.get(ri)
+                if got is not None:
+                    c, vals = got
+                    remap = np.fromiter((vocab.setdefault(v, len(vocab)) for v in vals),
+                                        dtype=np.int32, count=len(vals))
+                    cc = c.astype(np.int64)
+                    res = np.full(sz, -1, dtype=np.int32)
+                    has = cc >= 0
+                    res[has] = remap[cc[has]]
+                    out_codes[off:off + sz] = res
+                off += sz
+            label_cols[t] = (out_codes, list(vocab))
+        if "__name__" not in label_cols:
+            label_cols["__name__"] = (np.zeros(S, dtype=np.int32),
+                                      [sel.metric or st.schema.name])
         ts_all = torch.cat([p[0] for p in parts]) if parts else \
             torch.zeros(0, dtype=torch.int64, device=device)
         slots_all = torch.cat([p[1] for p in parts]).long() if parts else \
@@ -432,7 +518,7 @@ class PromEvaluator:
         out = prom_range_eval(ts_all.contiguous(), vals_all.contiguous(),
                               seg_lo.contiguous(), seg_hi.contiguous(),
                               T, t0, step, rng, offset_ms, param, mode)
-        return PromMatrix(labels, out, grid)
+        return PromMatrix(None, out, grid, label_cols=label_cols, n_series=S)
 
     # ------------------------------------------------------------ aggregate
 
@@ -443,19 +529,48 @@ class PromEvaluator:
         op = e.op
         if op in ("topk", "bottomk"):
             return self._topk(e, m, grid)
-        group_keys: dict[tuple, int] = {}
-        gidx = []
-        out_labels = []
-        for l in m.labels:
-            key_d = _group_labels(l, e.by, e.without)
-            key = tuple(sorted(key_d.items()))
-            if key not in group_keys:
-                group_keys[key] = len(group_keys)
-                out_labels.append(key_d)
-            gidx.append(group_keys[key])
-        G = max(len(group_keys), 1)
+        # vectorized grouping over factorized label columns (no per-series
+        # Python loop — required for 10M-series metric-engine queries)
+        S = m.S
+        cols = m.get_label_cols()
+        if e.by is not None:
+            gnames = list(e.by)
+        elif e.without is not None:
+            drop = set(e.without) | {"__name__"}
+            gnames = [n for n in cols if n not in drop]
+        else:
+            gnames = []
+        if not gnames or S == 0:
+            gidx_arr = np.zeros(S, dtype=np.int64)
+            out_labels = [{}]
+        else:
+            key = np.zeros(S, dtype=np.int64)
+            mult = 1
+            for n in gnames:
+                codes, values = cols.get(n, (np.full(S, -1, dtype=np.int32), []))
+                card = len(values) + 1
+                if mult > (1 << 62) // max(card, 1):
+                    _, key = np.unique(key, return_inverse=True)
+                    mult = int(key.max()) + 1 if len(key) else 1
+                key = key + (codes.astype(np.int64) + 1) * mult
+                mult *= card
+            uniq, first_idx, gidx_arr = np.unique(key, return_index=True,
+                                                  return_inverse=True)
+            out_labels = []
+            for idx in first_idx:
+                l = {}
+                for n in gnames:
+                    got = cols.get(n)
+                    if got is None:
+                        continue
+                    c = got[0][idx]
+                    if c >= 0:
+                        l[n] = got[1][c]
+                out_labels.append(l)
+        G = max(len(out_labels), 1)
         dev = m.values.device
-        gi = torch.as_tensor(gidx, dtype=torch.int64, device=dev)
+        gi = torch.as_tensor(gidx_arr, dtype=torch.int64, device=dev)
+        gidx = gidx_arr  # for the quantile path below
         v = m.values
         present = ~torch.isnan(v)
         v0 = torch.where(present, v, torch.zeros_like(v))
@@ -557,11 +672,9 @@ class PromEvaluator:
                 res, keep = _vector_op(e.op, b.expand_as(a), a, e.bool_modifier)
             else:
                 res, keep = _vector_op(e.op, a, b, e.bool_modifier)
-            labels = mat.labels if e.op not in _CMP_OPS or e.bool_modifier else mat.labels
-            labels = [_drop_name(x) for x in labels]
             if e.op in _CMP_OPS and not e.bool_modifier:
                 res = torch.where(keep, mat.values, torch.full_like(res, float("nan")))
-            return PromMatrix(labels, res, grid)
+            return _matrix_map(mat, res, drop_name=True)
         # vector-vector: one-to-one on matching label sets
         if e.op in ("and", "or", "unless"):
             return self._set_op(e.op, l, r, grid)
@@ -663,6 +776,19 @@ def _vector_op(op, a, b, bool_mod):
             return res, keep
         return a, keep
     raise PlanQuery(f"promql: op {op}")
+
+
+
+def _matrix_map(m: "PromMatrix", values=None, drop_name=False) -> "PromMatrix":
+    """New matrix with transformed values, keeping labels columnar/lazy."""
+    v = m.values if values is None else values
+    if m.label_cols is not None:
+        cols = m.label_cols
+        if drop_name:
+            cols = {k: c for k, c in cols.items() if k != "__name__"}
+        return PromMatrix(None, v, m.grid, label_cols=cols, n_series=m.S)
+    labels = [_drop_name(l) for l in m.labels] if drop_name else m.labels
+    return PromMatrix(labels, v, m.grid)
 
 
 def _drop_name(l: dict) -> dict:

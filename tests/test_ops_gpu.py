@@ -102,3 +102,34 @@ def test_series_last_matches_cpu():
     # sources... no: tie-break is deterministic (later src, later row)
     for e, g, name in zip(exp, got, ["ts", "src", "row"]):
         np.testing.assert_array_equal(e.numpy(), g.numpy(), err_msg=name)
+
+
+def test_prom_range_eval_matches_cpu():
+    rng = np.random.RandomState(5)
+    S = 40
+    rows = []
+    seg_lo, seg_hi = [], []
+    ts_all, vals_all = [], []
+    off = 0
+    for s in range(S):
+        n = rng.randint(0, 50)
+        ts = np.sort(rng.randint(0, 600_000, n)).astype(np.int64)
+        v = rng.uniform(-100, 100, n)
+        if s % 3 == 0:
+            v = np.abs(np.cumsum(np.abs(v)))  # counter-like
+        seg_lo.append(off); seg_hi.append(off + n); off += n
+        ts_all.append(ts); vals_all.append(v)
+    ts_t = torch.as_tensor(np.concatenate(ts_all))
+    v_t = torch.as_tensor(np.concatenate(vals_all))
+    lo_t = torch.as_tensor(np.array(seg_lo, dtype=np.int64))
+    hi_t = torch.as_tensor(np.array(seg_hi, dtype=np.int64))
+    T = 13
+    for mode_name, mode in cpu_ref.PROM_MODES.items():
+        param = 600.0 if mode_name == "predict_linear" else 0.0
+        exp = cpu_ref.prom_range_eval(ts_t, v_t, lo_t, hi_t, T, 100_000, 40_000,
+                                      120_000, 5_000, param, mode)
+        got = kernels.prom_range_eval(ts_t.cuda(), v_t.cuda(), lo_t.cuda(),
+                                      hi_t.cuda(), T, 100_000, 40_000, 120_000,
+                                      5_000, param, mode)
+        np.testing.assert_allclose(exp.numpy(), got.cpu().numpy(), rtol=1e-10,
+                                   equal_nan=True, err_msg=mode_name)

@@ -361,9 +361,13 @@ class PromEvaluator:
         return st, field
 
     def _match_codes(self, region, sel: ast.Selector):
-        """Codes matching all label matchers (None → all)."""
+        """Codes matching all label matchers. Returns None (= all codes) or a
+        sorted int64 array. Vectorized: regex/eq run over the label VALUE
+        dictionary (small), then a boolean LUT gathers per-series — no
+        per-series Python work (K13's host-side analog for labels)."""
         tag_names = region.series.tag_names
-        codes = None
+        n = len(region.series)
+        mask = None  # np bool [n]
         matchers = list(sel.matchers)
         # metric-engine regions carry __name__ as an ordinary label
         if "__name__" in region.series.inverted:
@@ -375,29 +379,41 @@ class PromEvaluator:
             if m.name == "__field__":
                 continue
             if m.name not in tag_names:
-                # matcher on absent label: = "" / !~ matches-empty keep all,
-                # otherwise empty result
                 if (m.op == "=" and m.value == "") or \
                    (m.op == "=~" and _re.fullmatch(m.value, "")) or \
                    (m.op == "!=" and m.value != "") or \
                    (m.op == "!~" and not _re.fullmatch(m.value, "")):
                     continue
-                return []
-            inv = region.series.inverted.get(m.name, {})
+                return np.zeros(0, dtype=np.int64)
+            codes_arr, values = region.series.tag_codes(m.name)
+            nv = len(values)
+            vmask = np.zeros(nv + 1, dtype=bool)  # +1 slot for absent (-1)
             if m.op == "=":
-                got = set(inv.get(m.value, []))
-            elif m.op == "!=":
-                got = set(range(len(region.series))) - set(inv.get(m.value, []))
+                # exact: dictionary lookup instead of scanning values
+                inv = region.series.inverted.get(m.name, {})
+                got = np.zeros(n, dtype=bool)
+                lst = inv.get(m.value)
+                if lst:
+                    got[np.asarray(lst, dtype=np.int64)] = True
             else:
-                pat = _re.compile(m.value)
-                sel_vals = [v for v in inv if pat.fullmatch(v)]
-                got = set()
-                for v in sel_vals:
-                    got.update(inv[v])
+                if m.op in ("=~", "!~"):
+                    pat = _re.compile(m.value)
+                    for i, v in enumerate(values):
+                        vmask[i] = bool(pat.fullmatch(v))
+                    vmask[nv] = bool(pat.fullmatch(""))  # absent label = ""
+                else:  # !=
+                    for i, v in enumerate(values):
+                        vmask[i] = v != m.value
+                    vmask[nv] = m.value != ""
+                sel_ids = codes_arr.astype(np.int64)
+                sel_ids[sel_ids < 0] = nv
+                got = vmask[sel_ids]
                 if m.op == "!~":
-                    got = set(range(len(region.series))) - got
-            codes = got if codes is None else codes & got
-        return None if codes is None else sorted(codes)
+                    got = ~got
+            mask = got if mask is None else (mask & got)
+        if mask is None:
+            return None
+        return np.flatnonzero(mask)
 
     def _eval_selector(self, sel: ast.Selector, func: str, t0, step, T, grid,
                        range_ms, param) -> PromMatrix:
@@ -442,6 +458,7 @@ class PromEvaluator:
             S_total += n_codes
             lut_t = torch.as_tensor(lut, device=device)
             chunks = []
+            all_sorted = True
             for src in region.scan_sources(lo, hi):
                 p = src.field_pos.get(field)
                 if p is None:
@@ -453,6 +470,7 @@ class PromEvaluator:
                 if idx.numel() == 0:
                     continue
                 chunks.append((src.ts[idx], src.series[idx], src.fields[p][idx]))
+                all_sorted = all_sorted and src.sorted
             if not chunks:
                 seg_counts.append((base, n_codes, 0))
                 continue
@@ -463,13 +481,17 @@ class PromEvaluator:
             ok = slots >= 0
             if not bool(ok.all()):
                 ts_t, slots, v_t = ts_t[ok], slots[ok], v_t[ok]
-            # sort by (slot, ts, arrival) then dedup last-wins per ts
-            o = torch.argsort(ts_t, stable=True)
-            perm = o[torch.argsort(slots[o], stable=True)]
-            ts_t, slots, v_t = ts_t[perm], slots[perm], v_t[perm]
+            # (slot, ts) order required by the window kernel. The slot LUT is
+            # monotone in code, so a single (series, ts)-sorted source is
+            # already slot-sorted — skip the sort (the common SST-cache case).
+            if not (len(chunks) == 1 and all_sorted):
+                o = torch.argsort(ts_t, stable=True)
+                perm = o[torch.argsort(slots[o], stable=True)]
+                ts_t, slots, v_t = ts_t[perm], slots[perm], v_t[perm]
             keep = dedup_mark_last(slots.int().contiguous(), ts_t.contiguous())
             kidx = keep.nonzero(as_tuple=True)[0]
-            ts_t, slots, v_t = ts_t[kidx], slots[kidx], v_t[kidx]
+            if kidx.numel() != ts_t.numel():
+                ts_t, slots, v_t = ts_t[kidx], slots[kidx], v_t[kidx]
             parts.append((ts_t, slots))
             vparts.append(v_t)
             seg_counts.append((base, n_codes, ts_t.numel()))

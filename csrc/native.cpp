@@ -585,6 +585,88 @@ class PromWriteParser {
   int32_t next_id_ = 0;
 };
 
+// ---------------------------------------------------------------- tokenizer
+// Fulltext tokenizer for log columns (reference: src/index fulltext_index —
+// tantivy's default tokenizer ≈ lowercase alphanumeric runs). Terms are
+// interned region-wide; output is CSR (doc offsets + term ids) ready to
+// build GPU posting lists.
+
+class Tokenizer {
+ public:
+  // docs: list[str|bytes]. Returns (offsets u64[n+1], term_ids i32[total],
+  // new_terms [(id, str)...]).
+  py::tuple tokenize(py::list docs) {
+    std::vector<uint64_t> offsets;
+    std::vector<int32_t> ids;
+    py::list new_terms;
+    offsets.reserve(docs.size() + 1);
+    offsets.push_back(0);
+    std::vector<int32_t> doc_terms;
+    for (auto& d : docs) {
+      char* buf = nullptr;
+      Py_ssize_t blen = 0;
+      std::string tmp;
+      if (PyBytes_Check(d.ptr())) {
+        PyBytes_AsStringAndSize(d.ptr(), &buf, &blen);
+      } else if (PyUnicode_Check(d.ptr())) {
+        tmp = py::cast<std::string>(d);
+        buf = tmp.data();
+        blen = tmp.size();
+      }
+      doc_terms.clear();
+      size_t i = 0;
+      while (i < (size_t)blen) {
+        while (i < (size_t)blen && !isalnum((unsigned char)buf[i])) i++;
+        size_t s = i;
+        while (i < (size_t)blen && isalnum((unsigned char)buf[i])) i++;
+        if (i > s) {
+          std::string term(buf + s, i - s);
+          for (auto& c : term) c = tolower((unsigned char)c);
+          const uint64_t h = fnv1a(term.data(), term.size());
+          int32_t tid = -1;
+          auto range = tmap_.equal_range(h);
+          for (auto it = range.first; it != range.second; ++it)
+            if (terms_[it->second] == term) { tid = it->second; break; }
+          if (tid < 0) {
+            tid = (int32_t)terms_.size();
+            tmap_.emplace(h, tid);
+            terms_.push_back(term);
+            new_terms.append(py::make_tuple(tid, py::str(term)));
+          }
+          // dedupe within doc (posting lists store docs, not positions)
+          bool seen = false;
+          for (int32_t t : doc_terms) if (t == tid) { seen = true; break; }
+          if (!seen) doc_terms.push_back(tid);
+        }
+      }
+      ids.insert(ids.end(), doc_terms.begin(), doc_terms.end());
+      offsets.push_back(ids.size());
+    }
+    py::array_t<uint64_t> off(offsets.size());
+    py::array_t<int32_t> tid(ids.size());
+    std::memcpy(off.mutable_data(), offsets.data(), offsets.size() * 8);
+    if (!ids.empty()) std::memcpy(tid.mutable_data(), ids.data(), ids.size() * 4);
+    return py::make_tuple(std::move(off), std::move(tid), std::move(new_terms));
+  }
+
+  // term → id (-1 if unknown); query-side probe
+  int32_t term_id(const std::string& term_in) const {
+    std::string term = term_in;
+    for (auto& c : term) c = tolower((unsigned char)c);
+    const uint64_t h = fnv1a(term.data(), term.size());
+    auto range = tmap_.equal_range(h);
+    for (auto it = range.first; it != range.second; ++it)
+      if (terms_[it->second] == term) return it->second;
+    return -1;
+  }
+
+  size_t num_terms() const { return terms_.size(); }
+
+ private:
+  std::unordered_multimap<uint64_t, int32_t> tmap_;
+  std::vector<std::string> terms_;
+};
+
 PYBIND11_MODULE(_native, m) {
   m.doc() = "greptimedb_amd host-native ingest path (line parser + WAL)";
   py::class_<LineParser>(m, "LineParser")
@@ -607,4 +689,9 @@ PYBIND11_MODULE(_native, m) {
       .def(py::init<>())
       .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)
       .def("num_series", &PromWriteParser::num_series);
+  py::class_<Tokenizer>(m, "Tokenizer")
+      .def(py::init<>())
+      .def("tokenize", &Tokenizer::tokenize)
+      .def("term_id", &Tokenizer::term_id)
+      .def("num_terms", &Tokenizer::num_terms);
 }

@@ -133,7 +133,8 @@ class MitoEngine:
     def write_region(self, table: TableState, region_idx: int,
                      series_codes: np.ndarray, ts_ms: np.ndarray,
                      fields: np.ndarray, new_series: list[tuple[int, bytes]],
-                     durable: bool = True) -> int:
+                     durable: bool = True,
+                     str_fields: dict[str, list] | None = None) -> int:
         """WAL-append + memtable-append one region's slice of a write batch.
         Caller must call `commit_wal()` after all regions of the batch
         (group commit — durability boundary)."""
@@ -141,9 +142,9 @@ class MitoEngine:
         seq = 0
         if durable:
             payload = encode_batch(series_codes, ts_ms, fields,
-                                   region.field_names, new_series)
+                                   region.field_names, new_series, str_fields)
             seq = self.wal.append(region.region_id, payload)
-        region.append(series_codes, ts_ms, fields, seq)
+        region.append(series_codes, ts_ms, fields, seq, str_fields)
         return seq
 
     def commit_wal(self):
@@ -199,7 +200,7 @@ class MitoEngine:
             region: Region = st.regions[idx]
             if seq <= region.flushed_seq:
                 continue
-            series, ts, fields, fnames, new_series = decode_batch(payload)
+            series, ts, fields, fnames, new_series, str_cols = decode_batch(payload)
             # series codes are stable via the region series log (loaded at
             # open); payload new_series is belt-and-braces for a lost log tail
             for _code, pk in new_series:
@@ -212,7 +213,7 @@ class MitoEngine:
                     if fn in fmap:
                         out[i] = fields[fmap[fn]]
                 fields = out
-            region.append(series, ts, fields, seq)
+            region.append(series, ts, fields, seq, str_cols or None)
 
     def close(self):
         if self._flusher is not None:

@@ -27,6 +27,7 @@ class Memtable:
         self.ts = torch.empty(cap, dtype=torch.int64, device=device)
         self.series = torch.empty(cap, dtype=torch.int32, device=device)
         self.fields = torch.empty((n_fields, cap), dtype=torch.float64, device=device)
+        self.str_cols: dict[str, list] = {}   # string fields stay host-side
         self.min_ts: int | None = None
         self.max_ts: int | None = None
 
@@ -48,13 +49,23 @@ class Memtable:
         self.fields = nf_t
         self.cap = new_cap
 
-    def append(self, series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray):
-        """series i32[n], ts_ms i64[n], fields f64[nf, n] (host arrays or tensors)."""
+    def append(self, series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
+               str_fields: dict[str, list] | None = None):
+        """series i32[n], ts_ms i64[n], fields f64[nf, n] (host arrays or
+        tensors); str_fields: {name: list[str|None] of length n}."""
         n = len(ts_ms)
         if n == 0:
             return
         if self.len + n > self.cap:
             self._grow(self.len + n)
+        sf = str_fields or {}
+        for name in set(self.str_cols) | set(sf):
+            col = self.str_cols.get(name)
+            if col is None:
+                col = [None] * self.len
+                self.str_cols[name] = col
+            vals = sf.get(name)
+            col.extend(vals if vals is not None else [None] * n)
         s = torch.as_tensor(series)
         t = torch.as_tensor(ts_ms)
         f = torch.as_tensor(fields)

@@ -25,17 +25,23 @@ from greptimedb_amd.ops import dedup_mark_last
 
 
 class ScanSource:
-    """One scan input: device column set + field-name → row mapping."""
+    """One scan input: device column set + field-name → row mapping.
+    str_cols: host string columns ({name: list|ndarray}), text_probe:
+    callable(col, terms) → bool mask (fulltext MATCHES)."""
 
-    __slots__ = ("ts", "series", "fields", "n", "field_pos", "sorted")
+    __slots__ = ("ts", "series", "fields", "n", "field_pos", "sorted",
+                 "str_cols", "text_probe")
 
-    def __init__(self, ts, series, fields, n, field_pos, sorted):
+    def __init__(self, ts, series, fields, n, field_pos, sorted,
+                 str_cols=None, text_probe=None):
         self.ts = ts
         self.series = series
         self.fields = fields
         self.n = n
         self.field_pos = field_pos
         self.sorted = sorted
+        self.str_cols = str_cols or {}
+        self.text_probe = text_probe
 
 
 class Region:
@@ -46,7 +52,14 @@ class Region:
         self.dir = dir
         self.device = device
         self.append_mode = append_mode
-        self.field_names = [c.name for c in schema.field_columns]
+        self.field_names = [c.name for c in schema.field_columns
+                            if not c.dtype.is_string_like]
+        # string fields (log columns): host-side values + fulltext index
+        self.str_field_names = [c.name for c in schema.field_columns
+                                if c.dtype.is_string_like]
+        from greptimedb_amd.engine.fulltext import FulltextColumn
+        self.text_cols: dict[str, FulltextColumn] = {
+            n: FulltextColumn() for n in self.str_field_names}
         self.series = SeriesIndex([c.name for c in schema.tag_columns])
         os.makedirs(os.path.join(dir, "sst"), exist_ok=True)
         self.manifest = Manifest(os.path.join(dir, "manifest"))
@@ -120,7 +133,7 @@ class Region:
             path = os.path.join(self.dir, "sst", f"{fid}.parquet")
             if not os.path.exists(path):
                 continue
-            dict_values, indices, ts, fields, seq = sst_mod.read_sst(
+            dict_values, indices, ts, fields, seq, str_cols = sst_mod.read_sst(
                 path, self.schema, self.field_names)
             remap = np.array([self.series.add_encoded(pk) for pk in dict_values],
                              dtype=np.int32)
@@ -136,18 +149,47 @@ class Region:
             perm = ord1[ord2]
             ord3 = torch.argsort(t_se[perm], stable=True)
             perm = perm[ord3]
-            self.sst_cache[fid] = sst_mod.SstBatch(
+            batch = sst_mod.SstBatch(
                 t_ts[perm].contiguous(), t_se[perm].contiguous(),
                 t_f[:, perm].contiguous(), t_seq[perm].contiguous(),
                 int(meta["min_ts"]), int(meta["max_ts"]),
                 list(self.field_names))
+            if str_cols:
+                perm_h = perm.cpu().numpy()
+                for name, vals in str_cols.items():
+                    if name not in self.str_field_names:
+                        self.str_field_names.append(name)
+                    if name not in self.text_cols:
+                        from greptimedb_amd.engine.fulltext import FulltextColumn
+                        self.text_cols[name] = FulltextColumn()
+                    arr = vals[perm_h]
+                    batch.str_cols[name] = arr
+                    batch.text_index[name] = self.text_cols[name].build_segment(
+                        list(arr), self.device)
+            self.sst_cache[fid] = batch
 
     # ---------------------------------------------------------------- write
 
     def append(self, series_codes: np.ndarray, ts_ms: np.ndarray,
-               fields: np.ndarray, last_seq: int):
+               fields: np.ndarray, last_seq: int,
+               str_fields: dict[str, list] | None = None):
         with self.lock:
-            self.memtable.append(series_codes, ts_ms, fields)
+            pre_len = self.memtable.len
+            n = len(ts_ms)
+            if str_fields:
+                for name in str_fields:
+                    if name not in self.text_cols:
+                        from greptimedb_amd.engine.fulltext import FulltextColumn
+                        ft = FulltextColumn()
+                        ft.mem.n_rows = pre_len  # rows before this column appeared
+                        self.text_cols[name] = ft
+                        if name not in self.str_field_names:
+                            self.str_field_names.append(name)
+            self.memtable.append(series_codes, ts_ms, fields, str_fields)
+            # keep every text column's row numbering aligned with the memtable
+            for name, ft in self.text_cols.items():
+                vals = (str_fields or {}).get(name)
+                ft.index_batch(vals if vals is not None else [None] * n)
             self.last_seq = max(self.last_seq, last_seq)
 
     def should_flush(self, limit_bytes: int) -> bool:
@@ -165,6 +207,8 @@ class Region:
             flush_field_names = list(self.field_names[: mem.nf])
             self.memtable = Memtable(len(self.field_names), device=self.device,
                                      cap=max(mem.cap, 1 << 16))
+            for ft in self.text_cols.values():
+                ft.reset_mem()  # new memtable rows start at 0
         ts, se, fields, perm = mem.sorted_view()
         if not self.append_mode:
             keep = dedup_mark_last(se, ts)
@@ -173,13 +217,24 @@ class Region:
         ts_h = ts.cpu().numpy()
         se_h = se.cpu().numpy()
         f_h = fields.cpu().numpy()
+        perm_h = perm.cpu().numpy()
+        # string columns: reorder host-side, build device posting segments
+        str_cols_sorted: dict[str, np.ndarray] = {}
+        text_index = {}
+        for name, col in mem.str_cols.items():
+            arr = np.array(col[: n], dtype=object)[perm_h]
+            str_cols_sorted[name] = arr
+            ft = self.text_cols.get(name)
+            if ft is not None:
+                text_index[name] = ft.build_segment(list(arr), self.device)
         # per-row sequence ≈ prev flushed seq + arrival index in this memtable
         # (row order is the recency order LastRow dedup relies on)
-        seq_h = perm.cpu().numpy().astype(np.int64) + prev_flushed
+        seq_h = perm_h.astype(np.int64) + prev_flushed
         fid = sst_mod.new_file_id()
         path = os.path.join(self.dir, "sst", f"{fid}.parquet")
         meta = sst_mod.write_sst(path, self.schema, self.series.pks,
-                                 se_h, ts_h, f_h, seq_h, flush_field_names)
+                                 se_h, ts_h, f_h, seq_h, flush_field_names,
+                                 str_cols=str_cols_sorted)
         meta.seq_max = flush_seq
         self.manifest.commit({
             "kind": "edit",
@@ -187,13 +242,28 @@ class Region:
             "files_to_remove": [],
             "flushed_seq": flush_seq,
         })
-        self.sst_cache[fid] = sst_mod.SstBatch(
+        batch = sst_mod.SstBatch(
             ts.contiguous(), se.contiguous(), fields.contiguous(), None,
             meta.min_ts, meta.max_ts, flush_field_names)
+        batch.str_cols = str_cols_sorted
+        batch.text_index = text_index
+        self.sst_cache[fid] = batch
         self.flushed_seq = flush_seq
         return meta
 
     # ---------------------------------------------------------------- schema
+
+    def ensure_str_fields(self, names: list[str]):
+        """Register string (fulltext) columns on this region."""
+        from greptimedb_amd.engine.fulltext import FulltextColumn
+        with self.lock:
+            for n in names:
+                if n not in self.str_field_names:
+                    self.str_field_names.append(n)
+                if n not in self.text_cols:
+                    ft = FulltextColumn()
+                    ft.mem.n_rows = self.memtable.len
+                    self.text_cols[n] = ft
 
     def ensure_fields(self, names: list[str]):
         """Auto-ALTER: add new field columns (reference insert.rs:562
@@ -211,23 +281,53 @@ class Region:
         """ScanSource list overlapping the time range: SST cache batches
         (time-pruned, reference scan_region.rs:887) + the active memtable."""
         out = []
+        device = self.device
         for batch in self.sst_cache.values():
             if ts_lo is not None and batch.max_ts < ts_lo:
                 continue
             if ts_hi is not None and batch.min_ts >= ts_hi:
                 continue
+
+            def make_probe(b):
+                def probe(col, terms):
+                    ft = self.text_cols.get(col)
+                    if ft is None:
+                        return None
+                    tids = ft.query_tids(terms)
+                    if tids is None:
+                        return torch.zeros(b.n, dtype=torch.bool, device=device)
+                    seg = getattr(b, "text_index", {}).get(col)
+                    if seg is None:
+                        return torch.zeros(b.n, dtype=torch.bool, device=device)
+                    return seg.probe(tids, device)
+                return probe
+
             out.append(ScanSource(batch.ts, batch.series, batch.fields, batch.n,
                                   {fn: i for i, fn in enumerate(batch.field_names)},
-                                  sorted=True))
+                                  sorted=True,
+                                  str_cols=getattr(batch, "str_cols", {}),
+                                  text_probe=make_probe(batch)))
         with self.lock:
             mem = self.memtable
             n = mem.len
         if n > 0:
             if not (ts_lo is not None and mem.max_ts is not None and mem.max_ts < ts_lo) and \
                not (ts_hi is not None and mem.min_ts is not None and mem.min_ts >= ts_hi):
+
+                def mem_probe(col, terms, _n=n, _mem=mem):
+                    ft = self.text_cols.get(col)
+                    if ft is None:
+                        return None
+                    tids = ft.query_tids(terms)
+                    if tids is None:
+                        return torch.zeros(_n, dtype=torch.bool, device=device)
+                    return ft.mem.probe(tids, _n, device)
+
                 out.append(ScanSource(mem.ts[:n], mem.series[:n], mem.fields, n,
                                       {fn: i for i, fn in enumerate(self.field_names)},
-                                      sorted=False))
+                                      sorted=False,
+                                      str_cols={k: v for k, v in mem.str_cols.items()},
+                                      text_probe=mem_probe))
         return out
 
     @property

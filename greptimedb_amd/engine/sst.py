@@ -62,6 +62,8 @@ class SstBatch:
         self.min_ts = min_ts
         self.max_ts = max_ts
         self.field_names = list(field_names)
+        self.str_cols: dict = {}     # host string columns (log fields)
+        self.text_index: dict = {}   # fulltext SegmentPostings per column
 
     @property
     def n(self) -> int:
@@ -70,7 +72,8 @@ class SstBatch:
 
 def write_sst(path: str, schema: TableSchema, pks: list[bytes],
               series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
-              seq: np.ndarray, field_names: list[str]) -> SstMeta:
+              seq: np.ndarray, field_names: list[str],
+              str_cols: dict[str, np.ndarray] | None = None) -> SstMeta:
     """Write one mito2-format parquet SST. Inputs are host arrays sorted by
     (series, ts); `pks[code]` gives the encoded primary key per local code."""
     n = len(ts_ms)
@@ -82,6 +85,9 @@ def write_sst(path: str, schema: TableSchema, pks: list[bytes],
     cols, names = [], []
     for i, fn in enumerate(field_names):
         cols.append(pa.array(fields[i], type=pa.float64()))
+        names.append(fn)
+    for fn, vals in (str_cols or {}).items():
+        cols.append(pa.array(list(vals), type=pa.string()))
         names.append(fn)
     cols.append(pa.array(ts_ms, type=pa.timestamp("ms")))
     names.append(schema.time_index.name)
@@ -111,8 +117,9 @@ def new_file_id() -> str:
 
 def read_sst(path: str, schema: TableSchema, field_names: list[str]):
     """Read an SST back to host arrays: (pk_list per row-code, series codes
-    i32 (dictionary indices), ts_ms i64, fields f64[nf, n], seq u64).
-    The caller remaps dictionary indices into region-local codes."""
+    i32 (dictionary indices), ts_ms i64, fields f64[nf, n], seq u64,
+    str_cols {name: object ndarray}). The caller remaps dictionary indices
+    into region-local codes."""
     t = pq.read_table(path)
     pk = t.column("__primary_key").combine_chunks()
     if isinstance(pk, pa.ChunkedArray):
@@ -126,4 +133,12 @@ def read_sst(path: str, schema: TableSchema, field_names: list[str]):
         for fn in field_names
     ]) if field_names else np.zeros((0, len(ts)))
     seq = t.column("__sequence").to_numpy(zero_copy_only=False).astype(np.int64)
-    return dict_values, indices, ts, fields, seq
+    internal = {schema.time_index.name, "__primary_key", "__sequence", "__op_type"}
+    str_cols = {}
+    for cn in t.column_names:
+        if cn in internal or cn in field_names:
+            continue
+        if pa.types.is_string(t.schema.field(cn).type) or \
+                pa.types.is_large_string(t.schema.field(cn).type):
+            str_cols[cn] = t.column(cn).to_numpy(zero_copy_only=False)
+    return dict_values, indices, ts, fields, seq, str_cols

@@ -20,17 +20,28 @@ from greptimedb_amd import _native
 
 
 def encode_batch(series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
-                 field_names: list[str], new_series: list[tuple[int, bytes]]) -> bytes:
-    """[u32 hdr_len][hdr json][series i32][ts i64][fields f64 nf*n]"""
+                 field_names: list[str], new_series: list[tuple[int, bytes]],
+                 str_cols: dict[str, list] | None = None) -> bytes:
+    """[u32 hdr_len][hdr json][series i32][ts i64][fields f64 nf*n]
+    [per str col: lengths i32[n] (-1=None) + utf8 blob]"""
+    str_cols = str_cols or {}
     hdr = json.dumps({
         "n": int(len(ts_ms)),
         "fields": field_names,
+        "strs": list(str_cols),
         "new_series": [[c, pk.hex()] for c, pk in new_series],
     }).encode()
     parts = [struct.pack("<I", len(hdr)), hdr,
              np.ascontiguousarray(series, dtype=np.int32).tobytes(),
              np.ascontiguousarray(ts_ms, dtype=np.int64).tobytes(),
              np.ascontiguousarray(fields, dtype=np.float64).tobytes()]
+    for name in str_cols:
+        vals = str_cols[name]
+        encs = [v.encode() if isinstance(v, str) else (v or b"") for v in vals]
+        lens = np.array([len(e) if vals[i] is not None else -1
+                         for i, e in enumerate(encs)], dtype=np.int32)
+        parts.append(lens.tobytes())
+        parts.append(b"".join(encs))
     return b"".join(parts)
 
 
@@ -43,8 +54,20 @@ def decode_batch(buf: bytes):
     series = np.frombuffer(buf, dtype=np.int32, count=n, offset=off); off += 4 * n
     ts = np.frombuffer(buf, dtype=np.int64, count=n, offset=off); off += 8 * n
     fields = np.frombuffer(buf, dtype=np.float64, count=nf * n, offset=off).reshape(nf, n)
+    off += 8 * nf * n
+    str_cols = {}
+    for name in hdr.get("strs", []):
+        lens = np.frombuffer(buf, dtype=np.int32, count=n, offset=off); off += 4 * n
+        vals = []
+        for ln in lens:
+            if ln < 0:
+                vals.append(None)
+            else:
+                vals.append(buf[off:off + ln].decode())
+                off += ln
+        str_cols[name] = vals
     new_series = [(c, bytes.fromhex(h)) for c, h in hdr["new_series"]]
-    return series, ts, fields, hdr["fields"], new_series
+    return series, ts, fields, hdr["fields"], new_series, str_cols
 
 
 class Wal:
@@ -102,5 +125,10 @@ class Wal:
                 os.unlink(os.path.join(self.dir, seg))
 
     def close(self):
-        self.writer.commit(self.sync_on_commit)
-        self.writer.close_segment()
+        if self.writer is not None:
+            try:
+                self.writer.commit(self.sync_on_commit)
+            except RuntimeError:
+                pass  # already closed
+            self.writer.close_segment()
+            self.writer = None

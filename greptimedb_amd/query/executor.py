@@ -242,8 +242,18 @@ class Executor:
             _lit_ts_ms(v, True) for v in by_col[ts_name]], dtype=np.int64)
         tag_names = [c.name for c in schema.tag_columns]
         field_names = st.regions[0].field_names
+        str_names = st.regions[0].str_field_names
         new_fields = [c for c in cols
-                      if c not in tag_names and c != ts_name and c not in field_names]
+                      if c not in tag_names and c != ts_name
+                      and c not in field_names and c not in str_names]
+        for c in list(new_fields):
+            # declared string fields, or undeclared columns with str values
+            declared = schema.column(c).dtype.is_string_like if schema.has_column(c) else None
+            sample = next((v for v in by_col[c] if v is not None), None)
+            if declared or (declared is None and isinstance(sample, str)
+                            and c not in field_names):
+                str_names = str_names + [c] if c not in str_names else str_names
+                new_fields.remove(c)
         if new_fields:
             for r in st.regions:
                 r.ensure_fields(new_fields)
@@ -269,7 +279,10 @@ class Executor:
                 if fn in by_col:
                     fmat[j] = [float(by_col[fn][r]) if by_col[fn][r] is not None else np.nan
                                for r in rows]
-            self.engine.write_region(st, ridx, codes[rows], ts_ms[rows], fmat, [])
+            str_fields = {sn: [by_col[sn][r] for r in rows]
+                          for sn in str_names if sn in by_col} or None
+            self.engine.write_region(st, ridx, codes[rows], ts_ms[rows], fmat, [],
+                                     str_fields=str_fields)
         self.engine.commit_wal()
         return QueryResult(["status"], [[f"inserted {n}"]])
 
@@ -739,6 +752,10 @@ class Executor:
                       agg_fields, sums, cnts, mins, maxs, rowcnt) -> QueryResult:
         # cells with data
         slot_idx, bucket_idx = np.nonzero(rowcnt)
+        if len(slot_idx) == 0 and not plan.group_tags and plan.bucket is None:
+            # aggregate without GROUP BY always yields one row (count=0)
+            slot_idx = np.array([0])
+            bucket_idx = np.array([0])
         order = np.argsort(bucket_idx * max(len(group_keys), 1) + slot_idx, kind="stable") \
             if len(slot_idx) else np.array([], dtype=np.int64)
         slot_idx, bucket_idx = slot_idx[order], bucket_idx[order]
@@ -852,17 +869,25 @@ class Executor:
         ts_name = schema.time_index.name
         tag_names = [c.name for c in schema.tag_columns]
         field_names = st.regions[0].field_names
+        str_field_names = []
+        for region in st.regions:
+            for sn in region.str_field_names:
+                if sn not in str_field_names:
+                    str_field_names.append(sn)
 
         # output columns
         out_cols: list[str] = []
         for e, alias in sel.projections:
             if isinstance(e, ast.Star):
-                out_cols.extend(tag_names + [ts_name] + field_names)
+                out_cols.extend(tag_names + [ts_name] + field_names + str_field_names)
             elif isinstance(e, ast.Col):
                 out_cols.append(e.name)
             else:
                 raise PlanQuery("raw SELECT supports columns and * only (no exprs yet)")
         needed_fields = [c for c in out_cols if c in field_names]
+        needed_strs = [c for c in out_cols if c in str_field_names]
+        if needed_strs and not st.append_mode:
+            raise PlanQuery("string columns require append_mode tables")
         order_cols = [e.name for e, _ in plan.order_by if isinstance(e, ast.Col)]
         for c in order_cols:
             if c in field_names and c not in needed_fields:
@@ -899,13 +924,24 @@ class Executor:
                     else:
                         f_rows.append(src.fields[p][idx])
                 f_t = torch.stack(f_rows) if f_rows else torch.zeros((0, idx.numel()), device=device)
-                chunks.append((ts_t, se_t, f_t))
+                s_vals = {}
+                if needed_strs:
+                    hidx = idx.cpu().numpy()
+                    for sn in needed_strs:
+                        col = src.str_cols.get(sn)
+                        if col is None:
+                            s_vals[sn] = np.full(len(hidx), None, dtype=object)
+                        else:
+                            s_vals[sn] = np.asarray(col, dtype=object)[hidx]
+                chunks.append((ts_t, se_t, f_t, s_vals))
             if not chunks:
                 continue
             ts_t = torch.cat([c[0] for c in chunks])
             se_t = torch.cat([c[1] for c in chunks])
             f_t = torch.cat([c[2] for c in chunks], dim=1) if needed_fields else \
                 torch.zeros((0, ts_t.numel()), device=device)
+            s_cols = {sn: np.concatenate([c[3][sn] for c in chunks])
+                      for sn in needed_strs}
             if not st.append_mode and len(chunks) >= 1:
                 # sort by (series, ts, arrival) then keep last
                 arrival = torch.arange(ts_t.numel(), device=device)
@@ -920,11 +956,11 @@ class Executor:
                 kidx = keep.nonzero(as_tuple=True)[0]
                 ts_t, se_t, f_t = ts_t[kidx], se_t[kidx], f_t[:, kidx]
             parts.append((ts_t.cpu().numpy(), se_t.cpu().numpy(), region,
-                          f_t.cpu().numpy()))
+                          f_t.cpu().numpy(), s_cols))
 
         # materialize host rows
         col_data = {c: [] for c in set(out_cols) | set(order_cols)}
-        for ts_h, se_h, region, f_h in parts:
+        for ts_h, se_h, region, f_h, s_cols in parts:
             for c in col_data:
                 if c == ts_name:
                     col_data[c].append(ts_h)
@@ -932,6 +968,8 @@ class Executor:
                     col_data[c].append(region.series.tag_array(c)[se_h])
                 elif c in needed_fields:
                     col_data[c].append(f_h[needed_fields.index(c)])
+                elif c in needed_strs:
+                    col_data[c].append(s_cols[c])
                 else:
                     raise PlanQuery(f"unknown column {c}")
         if parts:
@@ -1003,6 +1041,24 @@ class Executor:
             return lut[src.series.long()]
 
         def ev(x) -> torch.Tensor:
+            if isinstance(x, ast.Func) and x.name in ("matches", "matches_term"):
+                # fulltext probe (reference: matches/matches_term UDFs over the
+                # tantivy index; here: GPU posting-list probe, engine/fulltext.py)
+                if len(x.args) != 2 or not isinstance(x.args[0], ast.Col) or \
+                        not isinstance(x.args[1], ast.Lit):
+                    raise PlanQuery("matches(column, 'query') expected")
+                col = x.args[0].name
+                q = str(x.args[1].value)
+                import re as _re2
+                terms = _re2.findall(r"[A-Za-z0-9]+", q.lower())
+                if not terms:
+                    return torch.ones(n, dtype=torch.bool, device=device)
+                if src.text_probe is None:
+                    raise PlanQuery(f"no fulltext index on column {col}")
+                m = src.text_probe(col, terms)
+                if m is None:
+                    raise PlanQuery(f"no fulltext index on column {col}")
+                return m
             if isinstance(x, ast.BinOp):
                 if x.op == "and":
                     return ev(x.left) & ev(x.right)

@@ -31,6 +31,8 @@ class ServerContext:
         self.engine = engine
         self.ingestor = Ingestor(engine)
         self.promstore = PromStore(engine)
+        from greptimedb_amd.engine.logstore import LogStore
+        self.logstore = LogStore(engine)
         self.executor = Executor(engine, dist=dist)
         self.prom = PromEvaluator(engine, dist=dist)
         self.started = time.time()
@@ -160,6 +162,29 @@ def build_app(ctx: ServerContext) -> FastAPI:
     @app.post("/v1/influxdb/api/v2/write")
     async def influx_write_v2(request: Request, precision: str = Query("ns")):
         return await _influx(request, precision)
+
+    # ---------------- logs ----------------
+
+    @app.post("/v1/events/logs")
+    async def events_logs(request: Request, table: str = Query("logs"),
+                          tag_keys: str = Query(""), ts_key: str = Query("timestamp")):
+        import json as _json
+        body = await request.body()
+        entries = _json.loads(body)
+        if isinstance(entries, dict):
+            entries = [entries]
+        tags = [t for t in tag_keys.split(",") if t]
+        n = ctx.logstore.ingest(table, entries, tag_keys=tags, ts_key=ts_key)
+        metrics_mod.counter("log_events").inc(n)
+        return {"rows": n}
+
+    @app.post("/v1/loki/api/v1/push")
+    async def loki_push(request: Request):
+        import json as _json
+        body = await request.body()
+        n = ctx.logstore.ingest_loki(_json.loads(body))
+        metrics_mod.counter("loki_lines").inc(n)
+        return Response(status_code=204)
 
     # ---------------- prometheus remote write ----------------
 

@@ -110,12 +110,16 @@ class Region:
         return code
 
     def register_series_bulk(self, labels_list) -> np.ndarray:
-        """Bulk sparse registration (fixture/bulk-ingest path): one series-log
-        flush for the whole batch."""
+        """Bulk registration (fixture/bulk-ingest path): one series-log
+        flush for the whole batch. Items are dicts (sparse/metric-engine
+        mode) or tuples (dense tag order)."""
         prev = len(self.series)
         codes = np.empty(len(labels_list), dtype=np.int32)
         for i, labels in enumerate(labels_list):
-            codes[i] = self.series.get_or_create_labels(labels)
+            if isinstance(labels, dict):
+                codes[i] = self.series.get_or_create_labels(labels)
+            else:
+                codes[i] = self.series.get_or_create(labels)
         buf = bytearray()
         for code in range(prev, len(self.series)):
             pk = self.series.pks[code]

@@ -167,6 +167,17 @@ class MitoEngine:
     def _flush_region(self, region: Region):
         region.flush()
         self._purge_wal()
+        # TWCS compaction check (reference: compaction scheduler kicks after
+        # flush, mito2 flush.rs → compaction/scheduler.rs)
+        from greptimedb_amd.engine.compaction import Compactor
+        Compactor().compact_region(region)
+
+    def compact_all(self):
+        from greptimedb_amd.engine.compaction import Compactor
+        c = Compactor()
+        for st in self.tables.values():
+            for region in st.regions:
+                c.compact_region(region)
 
     def _flush_loop(self):
         while True:

@@ -120,6 +120,20 @@ class InsertValues:
 
 
 @dataclass
+class Admin:
+    """ADMIN func(args) — flush_table / compact_table / flush_region ...
+    (reference: src/common/function admin functions)."""
+    func: str
+    args: list
+
+
+@dataclass
+class Explain:
+    analyze: bool
+    stmt: object
+
+
+@dataclass
 class Tql:
     """TQL EVAL (start, end, step) promql_expr  (reference: src/sql TQL)."""
     start: float

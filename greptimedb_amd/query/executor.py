@@ -170,7 +170,59 @@ class Executor:
             return self._exec_insert(stmt)
         if isinstance(stmt, ast.Tql):
             return self._exec_tql(stmt)
+        if isinstance(stmt, ast.Admin):
+            return self._exec_admin(stmt)
+        if isinstance(stmt, ast.Explain):
+            return self._exec_explain(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _exec_admin(self, a: ast.Admin) -> QueryResult:
+        """ADMIN functions (reference: common/function/src/admin/)."""
+        f = a.func.lower()
+        if f == "flush_table":
+            st = self.engine.table(str(a.args[0]))
+            for r in st.regions:
+                self.engine._flush_region(r)
+            return QueryResult(["result"], [[1]])
+        if f == "compact_table":
+            from greptimedb_amd.engine.compaction import Compactor
+            st = self.engine.table(str(a.args[0]))
+            c = Compactor(trigger_file_num=2)
+            n = sum(c.compact_region(r) for r in st.regions)
+            return QueryResult(["result"], [[n]])
+        if f == "flush_all":
+            self.engine.flush_all()
+            return QueryResult(["result"], [[1]])
+        raise PlanQuery(f"unknown admin function {a.func}")
+
+    def _exec_explain(self, e: ast.Explain) -> QueryResult:
+        """EXPLAIN [ANALYZE]: plan summary (+ execution metrics)."""
+        import time as _time
+        if not isinstance(e.stmt, ast.Select):
+            raise PlanQuery("EXPLAIN supports SELECT")
+        plan = self._plan_select(e.stmt)
+        lines = []
+        path = "fused-ts-bucket-agg" if plan.aggs and plan.residual is None else \
+            ("agg-with-residual" if plan.aggs else "raw-scan")
+        lines.append(f"Plan: {path}")
+        lines.append(f"  table: {plan.table.schema.name} "
+                     f"regions={len(plan.table.regions)} append={plan.table.append_mode}")
+        lines.append(f"  time_range: [{plan.ts_lo}, {plan.ts_hi})")
+        lines.append(f"  tag_filters: {plan.tag_conj}")
+        lines.append(f"  residual: {plan.residual}")
+        if plan.bucket:
+            lines.append(f"  bucket_ms: {plan.bucket.bucket_ms}")
+        lines.append(f"  group_tags: {plan.group_tags}")
+        lines.append(f"  aggs: {[(a.func, a.arg) for a in plan.aggs]}")
+        if e.analyze:
+            t0 = _time.perf_counter()
+            r = self.execute_stmt(e.stmt)
+            dt = (_time.perf_counter() - t0) * 1000
+            lines.append(f"Execution: {dt:.3f} ms, {len(r)} output rows")
+            srcs = sum(len(region.scan_sources(plan.ts_lo, plan.ts_hi))
+                       for region in plan.table.regions)
+            lines.append(f"  scan sources: {srcs}")
+        return QueryResult(["plan"], [lines])
 
     def _exec_tql(self, t: ast.Tql) -> QueryResult:
         """TQL EVAL (start, end, step) expr — PromQL through SQL (reference:

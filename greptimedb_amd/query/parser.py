@@ -143,6 +143,13 @@ class Parser:
 
     # ---------------- statements ----------------
     def parse_statement(self):
+        stmt = self.parse_statement_inner()
+        self.eat_op(";")
+        if self.peek() is not None:
+            raise InvalidSyntax(f"trailing tokens: {self.peek()}")
+        return stmt
+
+    def parse_statement_inner(self):
         if self.at_kw("select"):
             stmt = self.parse_select()
         elif self.at_kw("create"):
@@ -161,11 +168,19 @@ class Parser:
             stmt = self.parse_insert()
         elif self.at_kw("tql"):
             stmt = self.parse_tql()
+        elif self.at_kw("admin"):
+            self.next()
+            e = self.parse_expr()
+            if not isinstance(e, ast.Func):
+                raise InvalidSyntax("ADMIN expects a function call")
+            args = [a.value if isinstance(a, ast.Lit) else a for a in e.args]
+            stmt = ast.Admin(e.name, args)
+        elif self.at_kw("explain"):
+            self.next()
+            analyze = self.eat_kw("analyze")
+            stmt = ast.Explain(analyze, self.parse_statement_inner())
         else:
             raise InvalidSyntax(f"unsupported statement start: {self.peek()}")
-        self.eat_op(";")
-        if self.peek() is not None:
-            raise InvalidSyntax(f"trailing tokens: {self.peek()}")
         return stmt
 
     def parse_select(self) -> ast.Select:

@@ -1,0 +1,114 @@
+"""Meta layer: φ-accrual failure detection, procedures, region migration."""
+
+import pytest
+
+from greptimedb_amd.meta.failure_detector import PhiAccrualFailureDetector
+from greptimedb_amd.meta.procedure import Procedure, ProcedureManager, Status
+from greptimedb_amd.meta.supervisor import RegionSupervisor
+
+
+def test_phi_low_while_heartbeating():
+    det = PhiAccrualFailureDetector()
+    t = 0.0
+    for _ in range(50):
+        det.heartbeat(t)
+        t += 1000
+    assert det.phi(t + 500) < 1.0
+    assert det.is_available(t + 500)
+
+
+def test_phi_rises_after_silence():
+    det = PhiAccrualFailureDetector(acceptable_heartbeat_pause_ms=0)
+    t = 0.0
+    for _ in range(50):
+        det.heartbeat(t)
+        t += 1000
+    # 60s of silence with 1s cadence → clearly dead
+    assert det.phi(t + 60_000) > 8.0
+    assert not det.is_available(t + 60_000)
+
+
+def test_supervisor_failover_once():
+    fired = []
+    sup = RegionSupervisor(on_failover=fired.append)
+    sup.detectors_threshold = 8
+    t = 0.0
+    for _ in range(30):
+        sup.heartbeat("gpu0", t)
+        sup.heartbeat("gpu1", t)
+        t += 1000
+    # gpu1 goes silent
+    for _ in range(30):
+        sup.heartbeat("gpu0", t)
+        t += 1000
+    failed = sup.check(t)
+    assert failed == ["gpu1"] and fired == ["gpu1"]
+    assert sup.check(t + 1000) == []  # fires once
+
+
+class CountingProc(Procedure):
+    TYPE = "counting"
+
+    def initial_state(self):
+        return {"i": 0}
+
+    def step(self, state):
+        state["i"] += 1
+        if state["i"] >= 3:
+            return Status.DONE, state
+        return Status.EXECUTING, state
+
+
+class CrashingProc(Procedure):
+    TYPE = "crashing"
+    crash = True
+
+    def initial_state(self):
+        return {"i": 0}
+
+    def step(self, state):
+        state["i"] += 1
+        if state["i"] == 2 and CrashingProc.crash:
+            raise RuntimeError("simulated crash")
+        if state["i"] >= 3:
+            return Status.DONE, state
+        return Status.EXECUTING, state
+
+
+def test_procedure_runs_to_done(tmp_path):
+    pm = ProcedureManager(str(tmp_path / "proc"))
+    pm.register(CountingProc)
+    pid = pm.submit(CountingProc())
+    assert pm.store.load_all() == []  # cleaned up
+    assert pid
+
+
+def test_procedure_crash_resume(tmp_path):
+    pm = ProcedureManager(str(tmp_path / "proc"))
+    pm.register(CrashingProc)
+    CrashingProc.crash = True
+    with pytest.raises(RuntimeError):
+        pm.submit(CrashingProc())
+    # state persisted at i=1
+    recs = pm.store.load_all()
+    assert len(recs) == 1 and recs[0]["state"]["i"] == 1
+    # restart: recover() resumes from persisted state
+    CrashingProc.crash = False
+    pm2 = ProcedureManager(str(tmp_path / "proc"))
+    pm2.register(CrashingProc)
+    resumed = pm2.recover()
+    assert len(resumed) == 1
+    assert pm2.store.load_all() == []
+
+
+def test_region_migration_cpu(tmp_engine):
+    from greptimedb_amd.meta.migration import migrate_region
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE t (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, PRIMARY KEY (h))")
+    ex.execute("INSERT INTO t (h, ts, v) VALUES ('a', 1000, 1.0), ('b', 2000, 2.0)")
+    before = ex.execute("SELECT h, ts, v FROM t ORDER BY ts").rows()
+    for i in range(len(tmp_engine.table("t").regions)):
+        migrate_region(tmp_engine, "t", i, "cpu")  # cpu→cpu exercises the path
+    after = ex.execute("SELECT h, ts, v FROM t ORDER BY ts").rows()
+    assert before == after

@@ -55,6 +55,11 @@ class MitoEngine:
                        sync_on_commit=config.wal_sync)
         self._flush_q: queue.Queue = queue.Queue()
         self._flusher = None
+        # observer seam (reference: mito2 engine/listener.rs WorkerListener —
+        # used by the flow engine for dirty-window tracking and by tests for
+        # deterministic background-task observation)
+        self.write_listeners: list = []   # callback(table_name, min_ts, max_ts, n)
+        self.flush_listeners: list = []   # callback(table_name, region_id)
         self._load_catalog()
         self._replay_wal()
         if config.background_flush:
@@ -145,6 +150,10 @@ class MitoEngine:
                                    region.field_names, new_series, str_fields)
             seq = self.wal.append(region.region_id, payload)
         region.append(series_codes, ts_ms, fields, seq, str_fields)
+        if self.write_listeners and len(ts_ms):
+            lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
+            for cb in self.write_listeners:
+                cb(table.schema.name, lo, hi, len(ts_ms))
         return seq
 
     def commit_wal(self):

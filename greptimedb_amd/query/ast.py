@@ -120,6 +120,24 @@ class InsertValues:
 
 
 @dataclass
+class CreateFlow:
+    name: str
+    sink: str
+    query_sql: str
+    if_not_exists: bool = False
+
+
+@dataclass
+class DropFlow:
+    name: str
+
+
+@dataclass
+class ShowFlows:
+    pass
+
+
+@dataclass
 class Admin:
     """ADMIN func(args) — flush_table / compact_table / flush_region ...
     (reference: src/common/function admin functions)."""

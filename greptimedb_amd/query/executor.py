@@ -172,9 +172,31 @@ class Executor:
             return self._exec_tql(stmt)
         if isinstance(stmt, ast.Admin):
             return self._exec_admin(stmt)
+        if isinstance(stmt, ast.CreateFlow):
+            self._flow_engine().create_flow(stmt.name, stmt.sink, stmt.query_sql,
+                                            stmt.if_not_exists)
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.DropFlow):
+            self._flow_engine().drop_flow(stmt.name)
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.ShowFlows):
+            fe = self._flow_engine()
+            return QueryResult(["Flow", "Sink", "Query"],
+                               [[f.name for f in fe.flows.values()],
+                                [f.sink for f in fe.flows.values()],
+                                [f.select_sql for f in fe.flows.values()]])
         if isinstance(stmt, ast.Explain):
             return self._exec_explain(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _flow_engine(self):
+        """Flow engine shared per storage engine (lazy)."""
+        fe = getattr(self.engine, "flow_engine", None)
+        if fe is None:
+            from greptimedb_amd.flow.engine import FlowEngine
+            fe = FlowEngine(self.engine, self)
+            self.engine.flow_engine = fe
+        return fe
 
     def _exec_admin(self, a: ast.Admin) -> QueryResult:
         """ADMIN functions (reference: common/function/src/admin/)."""

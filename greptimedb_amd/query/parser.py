@@ -158,8 +158,11 @@ class Parser:
             stmt = self.parse_drop()
         elif self.at_kw("show"):
             self.next()
-            self.expect_kw("tables")
-            stmt = ast.ShowTables()
+            if self.eat_kw("flows"):
+                stmt = ast.ShowFlows()
+            else:
+                self.expect_kw("tables")
+                stmt = ast.ShowTables()
         elif self.at_kw("describe", "desc"):
             self.next()
             self.eat_kw("table")
@@ -237,8 +240,25 @@ class Parser:
             offset = int(self.next().value)
         return ast.Select(projections, table, where, group_by, having, order_by, limit, offset)
 
-    def parse_create(self) -> ast.CreateTable:
+    def parse_create(self):
         self.expect_kw("create")
+        if self.eat_kw("flow"):
+            if_not_exists = False
+            if self.eat_kw("if"):
+                self.expect_kw("not"); self.expect_kw("exists")
+                if_not_exists = True
+            name = self.next().value
+            self.expect_kw("sink")
+            self.expect_kw("to")
+            sink = self.next().value
+            while not self.at_kw("as") and self.peek() is not None:
+                self.next()  # skip EXPIRE AFTER / COMMENT clauses
+            self.expect_kw("as")
+            if self.peek() is None:
+                raise InvalidSyntax("CREATE FLOW ... AS <select> expected")
+            sql = self.sql[self.peek().pos:].rstrip().rstrip(";")
+            self.i = len(self.toks)
+            return ast.CreateFlow(name, sink, sql, if_not_exists)
         self.expect_kw("table")
         if_not_exists = False
         if self.eat_kw("if"):
@@ -314,8 +334,10 @@ class Parser:
         return ast.CreateTable(name, columns, primary_key, time_index,
                                if_not_exists, options, partitions)
 
-    def parse_drop(self) -> ast.DropTable:
+    def parse_drop(self):
         self.expect_kw("drop")
+        if self.eat_kw("flow"):
+            return ast.DropFlow(self.next().value)
         self.expect_kw("table")
         if_exists = False
         if self.eat_kw("if"):

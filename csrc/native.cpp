@@ -585,6 +585,268 @@ class PromWriteParser {
   int32_t next_id_ = 0;
 };
 
+// ------------------------------------------------- OTLP traces (protobuf)
+// Minimal wire parser for opentelemetry ExportTraceServiceRequest
+// (trace.proto): resource_spans → scope_spans → spans. (service.name,
+// span name) pairs are interned to dense series refs; ids come back as hex
+// strings; attributes flatten to a compact JSON string column.
+
+static const char* HEXD = "0123456789abcdef";
+
+static std::string to_hex(const uint8_t* p, size_t n) {
+  std::string s(n * 2, '0');
+  for (size_t i = 0; i < n; i++) {
+    s[2 * i] = HEXD[p[i] >> 4];
+    s[2 * i + 1] = HEXD[p[i] & 0xF];
+  }
+  return s;
+}
+
+static void json_escape_into(std::string& out, const char* p, size_t n) {
+  for (size_t i = 0; i < n; i++) {
+    char c = p[i];
+    if (c == '"' || c == '\\') { out.push_back('\\'); out.push_back(c); }
+    else if ((unsigned char)c < 0x20) { out += "\\u0020"; }
+    else out.push_back(c);
+  }
+}
+
+class OtlpTraceParser {
+ public:
+  py::tuple parse(py::bytes data) {
+    char* buf; Py_ssize_t len;
+    if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+      throw std::runtime_error("expected bytes");
+    const uint8_t* p = reinterpret_cast<const uint8_t*>(buf);
+    const uint8_t* end = p + len;
+    out_series_.clear(); out_start_.clear(); out_dur_.clear();
+    out_status_.clear();
+    py::list trace_ids, span_ids, parent_ids, attrs, new_series;
+    while (p < end) {
+      uint64_t key = pw_varint(p, end);
+      if ((key >> 3) == 1 && (key & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        parse_resource_spans(p, p + l, trace_ids, span_ids, parent_ids,
+                             attrs, new_series);
+        p += l;
+      } else {
+        skip(key & 7, p, end);
+      }
+    }
+    size_t n = out_series_.size();
+    py::array_t<int32_t> s(n);
+    py::array_t<int64_t> st(n);
+    py::array_t<double> du(n);
+    py::array_t<int32_t> stc(n);
+    if (n) {
+      std::memcpy(s.mutable_data(), out_series_.data(), n * 4);
+      std::memcpy(st.mutable_data(), out_start_.data(), n * 8);
+      std::memcpy(du.mutable_data(), out_dur_.data(), n * 8);
+      std::memcpy(stc.mutable_data(), out_status_.data(), n * 4);
+    }
+    return py::make_tuple(std::move(s), std::move(st), std::move(du),
+                          std::move(stc), std::move(trace_ids),
+                          std::move(span_ids), std::move(parent_ids),
+                          std::move(attrs), std::move(new_series));
+  }
+
+  size_t num_series() const { return next_id_; }
+
+ private:
+  static void skip(int wt, const uint8_t*& p, const uint8_t* end) {
+    if (wt == 0) pw_varint(p, end);
+    else if (wt == 1) p += 8;
+    else if (wt == 2) { uint64_t l = pw_varint(p, end); p += l; }
+    else if (wt == 5) p += 4;
+    else p = end;
+  }
+
+  // returns value of attribute "service.name" from a Resource message
+  std::string parse_resource_service(const uint8_t* p, const uint8_t* end) {
+    std::string service = "unknown";
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      if ((k >> 3) == 1 && (k & 7) == 2) {  // attributes: KeyValue
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* kp = p; const uint8_t* kend = p + l;
+        std::string kname, vstr;
+        while (kp < kend) {
+          uint64_t kk = pw_varint(kp, kend);
+          if ((kk >> 3) == 1 && (kk & 7) == 2) {
+            uint64_t s = pw_varint(kp, kend);
+            kname.assign(reinterpret_cast<const char*>(kp), s); kp += s;
+          } else if ((kk >> 3) == 2 && (kk & 7) == 2) {  // AnyValue
+            uint64_t s = pw_varint(kp, kend);
+            const uint8_t* vp = kp; const uint8_t* vend = kp + s;
+            while (vp < vend) {
+              uint64_t vk = pw_varint(vp, vend);
+              if ((vk >> 3) == 1 && (vk & 7) == 2) {
+                uint64_t sl = pw_varint(vp, vend);
+                vstr.assign(reinterpret_cast<const char*>(vp), sl); vp += sl;
+              } else skip(vk & 7, vp, vend);
+            }
+            kp += s;
+          } else skip(kk & 7, kp, kend);
+        }
+        if (kname == "service.name" && !vstr.empty()) service = vstr;
+        p += l;
+      } else skip(k & 7, p, end);
+    }
+    return service;
+  }
+
+  void parse_resource_spans(const uint8_t* p, const uint8_t* end,
+                            py::list& tids, py::list& sids, py::list& pids,
+                            py::list& attrs, py::list& new_series) {
+    std::string service = "unknown";
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 1 && wt == 2) {  // Resource
+        uint64_t l = pw_varint(p, end);
+        service = parse_resource_service(p, p + l);
+        p += l;
+      } else if (f == 2 && wt == 2) {  // ScopeSpans
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* sp = p; const uint8_t* send = p + l;
+        while (sp < send) {
+          uint64_t sk = pw_varint(sp, send);
+          if ((sk >> 3) == 2 && (sk & 7) == 2) {  // Span
+            uint64_t sl = pw_varint(sp, send);
+            parse_span(sp, sp + sl, service, tids, sids, pids, attrs, new_series);
+            sp += sl;
+          } else skip(sk & 7, sp, send);
+        }
+        p += l;
+      } else skip(wt, p, end);
+    }
+  }
+
+  void parse_span(const uint8_t* p, const uint8_t* end, const std::string& service,
+                  py::list& tids, py::list& sids, py::list& pids,
+                  py::list& attrs, py::list& new_series) {
+    std::string trace_id, span_id, parent_id, name;
+    uint64_t start = 0, stop = 0;
+    int32_t status = 0;
+    std::string attr_json = "{";
+    bool first_attr = true;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 1 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        trace_id = to_hex(p, l); p += l;
+      } else if (f == 2 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        span_id = to_hex(p, l); p += l;
+      } else if (f == 4 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        parent_id = to_hex(p, l); p += l;
+      } else if (f == 5 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        name.assign(reinterpret_cast<const char*>(p), l); p += l;
+      } else if (f == 7 && wt == 1) {
+        std::memcpy(&start, p, 8); p += 8;
+      } else if (f == 8 && wt == 1) {
+        std::memcpy(&stop, p, 8); p += 8;
+      } else if (f == 9 && wt == 2) {  // attributes KeyValue
+        uint64_t l = pw_varint(p, end);
+        append_attr_json(p, p + l, attr_json, first_attr);
+        p += l;
+      } else if (f == 15 && wt == 2) {  // Status{message=2?, code=3? -> code=2? }
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* qp = p; const uint8_t* qend = p + l;
+        while (qp < qend) {
+          uint64_t qk = pw_varint(qp, qend);
+          if ((qk >> 3) == 2 && (qk & 7) == 0) {
+            status = (int32_t)pw_varint(qp, qend);
+          } else if ((qk >> 3) == 3 && (qk & 7) == 0) {
+            status = (int32_t)pw_varint(qp, qend);
+          } else skip(qk & 7, qp, qend);
+        }
+        p += l;
+      } else {
+        skip(wt, p, end);
+      }
+    }
+    attr_json.push_back('}');
+    // intern (service, span name)
+    std::string key = service;
+    key.push_back('\0');
+    key += name;
+    const uint64_t h = fnv1a(key.data(), key.size());
+    int32_t sid = -1;
+    auto range = ids_.equal_range(h);
+    for (auto it = range.first; it != range.second; ++it)
+      if (keys_[it->second] == key) { sid = it->second; break; }
+    if (sid < 0) {
+      sid = next_id_++;
+      ids_.emplace(h, sid);
+      keys_.resize(std::max<size_t>(keys_.size(), sid + 1));
+      keys_[sid] = key;
+      new_series.append(py::make_tuple(sid, py::str(service), py::str(name)));
+    }
+    out_series_.push_back(sid);
+    out_start_.push_back((int64_t)start);
+    out_dur_.push_back(stop >= start ? (double)(stop - start) / 1e6 : 0.0);  // ms
+    out_status_.push_back(status);
+    tids.append(py::str(trace_id));
+    sids.append(py::str(span_id));
+    pids.append(py::str(parent_id));
+    attrs.append(py::str(attr_json));
+  }
+
+  void append_attr_json(const uint8_t* p, const uint8_t* end, std::string& out,
+                        bool& first) {
+    std::string kname, vjson;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      if ((k >> 3) == 1 && (k & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        kname.assign(reinterpret_cast<const char*>(p), l); p += l;
+      } else if ((k >> 3) == 2 && (k & 7) == 2) {  // AnyValue
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* vp = p; const uint8_t* vend = p + l;
+        while (vp < vend) {
+          uint64_t vk = pw_varint(vp, vend);
+          int vf = (int)(vk >> 3), vwt = (int)(vk & 7);
+          if (vf == 1 && vwt == 2) {
+            uint64_t sl = pw_varint(vp, vend);
+            vjson = "\"";
+            json_escape_into(vjson, reinterpret_cast<const char*>(vp), sl);
+            vjson += "\"";
+            vp += sl;
+          } else if (vf == 2 && vwt == 0) {
+            vjson = pw_varint(vp, vend) ? "true" : "false";
+          } else if (vf == 3 && vwt == 0) {
+            vjson = std::to_string((int64_t)pw_varint(vp, vend));
+          } else if (vf == 4 && vwt == 1) {
+            double d; std::memcpy(&d, vp, 8); vp += 8;
+            vjson = std::to_string(d);
+          } else skip(vwt, vp, vend);
+        }
+        p += l;
+      } else skip(k & 7, p, end);
+    }
+    if (!kname.empty() && !vjson.empty()) {
+      if (!first) out.push_back(',');
+      first = false;
+      out.push_back('"');
+      json_escape_into(out, kname.data(), kname.size());
+      out += "\":";
+      out += vjson;
+    }
+  }
+
+  std::unordered_multimap<uint64_t, int32_t> ids_;
+  std::vector<std::string> keys_;
+  int32_t next_id_ = 0;
+  std::vector<int32_t> out_series_;
+  std::vector<int64_t> out_start_;
+  std::vector<double> out_dur_;
+  std::vector<int32_t> out_status_;
+};
+
 // ---------------------------------------------------------------- tokenizer
 // Fulltext tokenizer for log columns (reference: src/index fulltext_index —
 // tantivy's default tokenizer ≈ lowercase alphanumeric runs). Terms are
@@ -689,6 +951,10 @@ PYBIND11_MODULE(_native, m) {
       .def(py::init<>())
       .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)
       .def("num_series", &PromWriteParser::num_series);
+  py::class_<OtlpTraceParser>(m, "OtlpTraceParser")
+      .def(py::init<>())
+      .def("parse", &OtlpTraceParser::parse)
+      .def("num_series", &OtlpTraceParser::num_series);
   py::class_<Tokenizer>(m, "Tokenizer")
       .def(py::init<>())
       .def("tokenize", &Tokenizer::tokenize)

@@ -1094,6 +1094,8 @@ class Executor:
                     return src.fields[p][:n]
                 if x.name in tag_names:
                     return ("__tag__", x.name)
+                if x.name in src.str_cols or x.name in region.str_field_names:
+                    return ("__str__", x.name)
                 raise PlanQuery(f"unknown column {x.name}")
             if isinstance(x, ast.UnaryOp) and x.op == "-":
                 return -_as_t(value(x.operand))
@@ -1140,6 +1142,14 @@ class Executor:
                     return ev(x.left) | ev(x.right)
                 if x.op in ("=", "!=", "<>", "<", "<=", ">", ">="):
                     lv, rv = value(x.left), value(x.right)
+                    # string-field comparison → host column eval
+                    if isinstance(lv, tuple) and lv[0] == "__str__":
+                        col = src.str_cols.get(lv[1])
+                        if col is None:
+                            return torch.zeros(n, dtype=torch.bool, device=device)
+                        arr = np.asarray(col, dtype=object)
+                        res = np.array([_py_cmp(x.op, v, rv) for v in arr], dtype=bool)
+                        return torch.as_tensor(res, device=device)
                     # tag comparison → host-side per-code eval
                     if isinstance(lv, tuple) and lv[0] == "__tag__":
                         tag = lv[1]

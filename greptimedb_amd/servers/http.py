@@ -33,6 +33,8 @@ class ServerContext:
         self.promstore = PromStore(engine)
         from greptimedb_amd.engine.logstore import LogStore
         self.logstore = LogStore(engine)
+        from greptimedb_amd.engine.tracestore import TraceStore
+        self.tracestore = TraceStore(engine)
         self.executor = Executor(engine, dist=dist)
         self.prom = PromEvaluator(engine, dist=dist)
         self.started = time.time()
@@ -185,6 +187,15 @@ def build_app(ctx: ServerContext) -> FastAPI:
         n = ctx.logstore.ingest_loki(_json.loads(body))
         metrics_mod.counter("loki_lines").inc(n)
         return Response(status_code=204)
+
+    # ---------------- OTLP ----------------
+
+    @app.post("/v1/otlp/v1/traces")
+    async def otlp_traces(request: Request):
+        body = await request.body()
+        n = ctx.tracestore.write(body)
+        metrics_mod.counter("otlp_spans").inc(n)
+        return {"partialSuccess": {}}
 
     # ---------------- prometheus remote write ----------------
 

@@ -929,6 +929,35 @@ class Tokenizer {
   std::vector<std::string> terms_;
 };
 
+// Pack a python string column (list of str|None) into (lens i32, utf8 blob)
+// — WAL string-field serialization hot path.
+static py::tuple pack_str_col(py::list vals) {
+  const size_t n = vals.size();
+  py::array_t<int32_t> lens(n);
+  int32_t* lp = lens.mutable_data();
+  std::string blob;
+  blob.reserve(n * 16);
+  for (size_t i = 0; i < n; i++) {
+    PyObject* o = vals[i].ptr();
+    if (o == Py_None) {
+      lp[i] = -1;
+    } else if (PyUnicode_Check(o)) {
+      Py_ssize_t sl;
+      const char* s = PyUnicode_AsUTF8AndSize(o, &sl);
+      lp[i] = (int32_t)sl;
+      blob.append(s, sl);
+    } else if (PyBytes_Check(o)) {
+      char* s; Py_ssize_t sl;
+      PyBytes_AsStringAndSize(o, &s, &sl);
+      lp[i] = (int32_t)sl;
+      blob.append(s, sl);
+    } else {
+      lp[i] = -1;
+    }
+  }
+  return py::make_tuple(std::move(lens), py::bytes(blob));
+}
+
 PYBIND11_MODULE(_native, m) {
   m.doc() = "greptimedb_amd host-native ingest path (line parser + WAL)";
   py::class_<LineParser>(m, "LineParser")
@@ -947,6 +976,7 @@ PYBIND11_MODULE(_native, m) {
       .def("segment_bytes", &WalWriter::segment_bytes);
   m.def("wal_read_segment", &wal_read_segment);
   m.def("snappy_uncompress", &py_snappy_uncompress);
+  m.def("pack_str_col", &pack_str_col);
   py::class_<PromWriteParser>(m, "PromWriteParser")
       .def(py::init<>())
       .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)

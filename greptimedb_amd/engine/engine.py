@@ -238,4 +238,6 @@ class MitoEngine:
     def close(self):
         if self._flusher is not None:
             self._flush_q.put(None)
+            self._flusher.join(timeout=10)
+            self._flusher = None
         self.wal.close()

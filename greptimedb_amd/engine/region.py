@@ -54,12 +54,14 @@ class Region:
         self.append_mode = append_mode
         self.field_names = [c.name for c in schema.field_columns
                             if not c.dtype.is_string_like]
-        # string fields (log columns): host-side values + fulltext index
+        # string fields (log columns): host-side values; fulltext index only
+        # where the column opts in (schema fulltext flag)
         self.str_field_names = [c.name for c in schema.field_columns
                                 if c.dtype.is_string_like]
         from greptimedb_amd.engine.fulltext import FulltextColumn
         self.text_cols: dict[str, FulltextColumn] = {
-            n: FulltextColumn() for n in self.str_field_names}
+            c.name: FulltextColumn() for c in schema.field_columns
+            if c.dtype.is_string_like and c.fulltext}
         self.series = SeriesIndex([c.name for c in schema.tag_columns])
         os.makedirs(os.path.join(dir, "sst"), exist_ok=True)
         self.manifest = Manifest(os.path.join(dir, "manifest"))
@@ -163,13 +165,12 @@ class Region:
                 for name, vals in str_cols.items():
                     if name not in self.str_field_names:
                         self.str_field_names.append(name)
-                    if name not in self.text_cols:
-                        from greptimedb_amd.engine.fulltext import FulltextColumn
-                        self.text_cols[name] = FulltextColumn()
                     arr = vals[perm_h]
                     batch.str_cols[name] = arr
-                    batch.text_index[name] = self.text_cols[name].build_segment(
-                        list(arr), self.device)
+                    ft = self.text_cols.get(name)
+                    if ft is not None:  # fulltext-indexed columns only
+                        batch.text_index[name] = ft.build_segment(
+                            list(arr), self.device)
             self.sst_cache[fid] = batch
 
     # ---------------------------------------------------------------- write
@@ -178,17 +179,11 @@ class Region:
                fields: np.ndarray, last_seq: int,
                str_fields: dict[str, list] | None = None):
         with self.lock:
-            pre_len = self.memtable.len
             n = len(ts_ms)
             if str_fields:
                 for name in str_fields:
-                    if name not in self.text_cols:
-                        from greptimedb_amd.engine.fulltext import FulltextColumn
-                        ft = FulltextColumn()
-                        ft.mem.n_rows = pre_len  # rows before this column appeared
-                        self.text_cols[name] = ft
-                        if name not in self.str_field_names:
-                            self.str_field_names.append(name)
+                    if name not in self.str_field_names:
+                        self.str_field_names.append(name)
             self.memtable.append(series_codes, ts_ms, fields, str_fields)
             # keep every text column's row numbering aligned with the memtable
             for name, ft in self.text_cols.items():
@@ -257,14 +252,15 @@ class Region:
 
     # ---------------------------------------------------------------- schema
 
-    def ensure_str_fields(self, names: list[str]):
-        """Register string (fulltext) columns on this region."""
+    def ensure_str_fields(self, names: list[str], fulltext: bool = True):
+        """Register string columns on this region (fulltext-indexed when
+        requested — the log-pipeline default)."""
         from greptimedb_amd.engine.fulltext import FulltextColumn
         with self.lock:
             for n in names:
                 if n not in self.str_field_names:
                     self.str_field_names.append(n)
-                if n not in self.text_cols:
+                if fulltext and n not in self.text_cols:
                     ft = FulltextColumn()
                     ft.mem.n_rows = self.memtable.len
                     self.text_cols[n] = ft

@@ -29,10 +29,15 @@ def trace_schema() -> TableSchema:
             ColumnSchema("ts", DataType.TIMESTAMP_MS, SemanticType.TIMESTAMP, 2),
             ColumnSchema("duration_ms", DataType.FLOAT64, SemanticType.FIELD, 3),
             ColumnSchema("status_code", DataType.FLOAT64, SemanticType.FIELD, 4),
-            ColumnSchema("trace_id", DataType.STRING, SemanticType.FIELD, 5),
-            ColumnSchema("span_id", DataType.STRING, SemanticType.FIELD, 6),
-            ColumnSchema("parent_span_id", DataType.STRING, SemanticType.FIELD, 7),
-            ColumnSchema("span_attributes", DataType.JSON, SemanticType.FIELD, 8),
+            # id columns are point-looked-up, not fulltext-searched: no index
+            ColumnSchema("trace_id", DataType.STRING, SemanticType.FIELD, 5,
+                         fulltext=False),
+            ColumnSchema("span_id", DataType.STRING, SemanticType.FIELD, 6,
+                         fulltext=False),
+            ColumnSchema("parent_span_id", DataType.STRING, SemanticType.FIELD, 7,
+                         fulltext=False),
+            ColumnSchema("span_attributes", DataType.JSON, SemanticType.FIELD, 8,
+                         fulltext=False),
         ],
         primary_key=["service_name", "span_name"],
         options={"append_mode": "true"},

@@ -37,11 +37,11 @@ def encode_batch(series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
              np.ascontiguousarray(fields, dtype=np.float64).tobytes()]
     for name in str_cols:
         vals = str_cols[name]
-        encs = [v.encode() if isinstance(v, str) else (v or b"") for v in vals]
-        lens = np.array([len(e) if vals[i] is not None else -1
-                         for i, e in enumerate(encs)], dtype=np.int32)
+        if not isinstance(vals, list):
+            vals = list(vals)
+        lens, blob = _native.pack_str_col(vals)
         parts.append(lens.tobytes())
-        parts.append(b"".join(encs))
+        parts.append(blob)
     return b"".join(parts)
 
 

@@ -90,6 +90,7 @@ class ColumnSchema:
     semantic: SemanticType
     column_id: int
     nullable: bool = True
+    fulltext: bool = False   # build a fulltext index (string fields only)
 
     def __post_init__(self):
         if self.semantic == SemanticType.TIMESTAMP and not self.dtype.is_timestamp:
@@ -155,6 +156,7 @@ class TableSchema:
                     "semantic": int(c.semantic),
                     "column_id": c.column_id,
                     "nullable": c.nullable,
+                    "fulltext": c.fulltext,
                 }
                 for c in self.columns
             ],
@@ -169,6 +171,7 @@ class TableSchema:
                 semantic=SemanticType(c["semantic"]),
                 column_id=c["column_id"],
                 nullable=c.get("nullable", True),
+                fulltext=c.get("fulltext", False),
             )
             for c in d["columns"]
         ]

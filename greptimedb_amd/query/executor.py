@@ -294,8 +294,13 @@ class Executor:
             sem = (SemanticType.TIMESTAMP if name == c.time_index
                    else SemanticType.TAG if name in c.primary_key
                    else SemanticType.FIELD)
+            # string FIELD columns default to fulltext-indexed (observability
+            # default; opt out via the column-less declaration in stores)
+            ft = opts.get("fulltext",
+                          t.is_string_like and sem == SemanticType.FIELD)
             cols.append(ColumnSchema(name, t, sem, i,
-                                     nullable=opts.get("nullable", True)))
+                                     nullable=opts.get("nullable", True),
+                                     fulltext=bool(ft)))
         schema = TableSchema(name=c.name, columns=cols, primary_key=c.primary_key,
                              options={k: v for k, v in c.options.items()})
         append = str(c.options.get("append_mode", "false")).lower() == "true"

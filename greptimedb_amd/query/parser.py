@@ -304,6 +304,17 @@ class Parser:
                         primary_key.append(cname)
                     elif self.eat_kw("default"):
                         opts["default"] = self.parse_expr()
+                    elif self.eat_kw("fulltext"):
+                        self.eat_kw("index")
+                        opts["fulltext"] = True
+                        if self.eat_op("("):  # FULLTEXT INDEX WITH-style opts
+                            depth = 1
+                            while depth and self.peek() is not None:
+                                t = self.next()
+                                if t.kind == "op" and t.value == "(":
+                                    depth += 1
+                                elif t.kind == "op" and t.value == ")":
+                                    depth -= 1
                     else:
                         break
                 columns.append((cname, ctype, opts))

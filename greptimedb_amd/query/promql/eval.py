@@ -593,12 +593,18 @@ class PromEvaluator:
         dev = m.values.device
         gi = torch.as_tensor(gidx_arr, dtype=torch.int64, device=dev)
         gidx = gidx_arr  # for the quantile path below
+        single_group = G == 1
         v = m.values
         present = ~torch.isnan(v)
         v0 = torch.where(present, v, torch.zeros_like(v))
         cnt = torch.zeros((G, T), dtype=torch.float64, device=dev)
         if m.S:
-            cnt.index_add_(0, gi, present.double())
+            if single_group:
+                # index_add with one destination cell is a full-serialization
+                # atomic collision on GPU (21s at 1M series) — reduce instead
+                cnt[0] = present.double().sum(dim=0)
+            else:
+                cnt.index_add_(0, gi, present.double())
 
         if op == "quantile":
             if self.dist is not None:
@@ -617,16 +623,24 @@ class PromEvaluator:
             if op in ("stddev", "stdvar") else None
         mn = mx = None
         if m.S:
-            s.index_add_(0, gi, v0)
-            if sq is not None:
-                sq.index_add_(0, gi, v0 * v0)
+            if single_group:
+                s[0] = v0.sum(dim=0)
+                if sq is not None:
+                    sq[0] = (v0 * v0).sum(dim=0)
+            else:
+                s.index_add_(0, gi, v0)
+                if sq is not None:
+                    sq.index_add_(0, gi, v0 * v0)
         if op in ("min", "max"):
             fill = float("inf") if op == "min" else float("-inf")
             t = torch.full((G, T), fill, dtype=torch.float64, device=dev)
             if m.S:
                 vm = torch.where(present, v, torch.full_like(v, fill))
-                t.index_reduce_(0, gi, vm, "amin" if op == "min" else "amax",
-                                include_self=True)
+                if single_group:
+                    t[0] = vm.amin(dim=0) if op == "min" else vm.amax(dim=0)
+                else:
+                    t.index_reduce_(0, gi, vm, "amin" if op == "min" else "amax",
+                                    include_self=True)
             if op == "min":
                 mn = t
             else:

@@ -666,11 +666,12 @@ class Executor:
                     idx = mask.nonzero(as_tuple=True)[0]
                     if idx.numel():
                         filt.append((src, idx))
-                pairs = [(src.ts[idx].contiguous(), src.series[idx].contiguous())
+                pairs = [(src.ts[idx].contiguous(), src.series[idx].contiguous(),
+                          src.sorted)
                          for src, idx in filt]
                 src_objs = [(src, idx) for src, idx in filt]
             else:
-                pairs = [(src.ts, src.series) for src in sources]
+                pairs = [(src.ts, src.series, src.sorted) for src in sources]
                 src_objs = [(src, None) for src in sources]
             if not pairs:
                 continue

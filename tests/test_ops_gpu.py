@@ -133,3 +133,25 @@ def test_prom_range_eval_matches_cpu():
                                       5_000, param, mode)
         np.testing.assert_allclose(exp.numpy(), got.cpu().numpy(), rtol=1e-10,
                                    equal_nan=True, err_msg=mode_name)
+
+
+def test_series_last_sorted_fast_path():
+    """Sorted-source segment path must match the CPU reference exactly."""
+    rng = np.random.RandomState(21)
+    sources = []
+    for _ in range(3):
+        n = 30_000
+        se = np.sort(rng.randint(0, 300, n)).astype(np.int32)
+        ts = np.zeros(n, dtype=np.int64)
+        # ts sorted within each series segment
+        for c in np.unique(se):
+            m = se == c
+            ts[m] = np.sort(rng.randint(0, 100_000, m.sum()))
+        sources.append((torch.as_tensor(ts), torch.as_tensor(se)))
+    lut = torch.as_tensor(rng.randint(-1, 80, 300).astype(np.int32))
+    exp = cpu_ref.series_last(sources, lut, -(1 << 62), (1 << 62), 80)
+    got = kernels.series_last(
+        [(t.cuda(), s.cuda(), True) for t, s in sources], lut.cuda(),
+        -(1 << 62), (1 << 62), 80)
+    for e, g, name in zip(exp, got, ["ts", "src", "row"]):
+        np.testing.assert_array_equal(e.numpy(), g.numpy(), err_msg=name)

@@ -120,6 +120,14 @@ class InsertValues:
 
 
 @dataclass
+class Copy:
+    table: str
+    path: str
+    direction: str  # "to" | "from"
+    options: dict = field(default_factory=dict)
+
+
+@dataclass
 class CreateFlow:
     name: str
     sink: str

@@ -187,7 +187,101 @@ class Executor:
                                 [f.select_sql for f in fe.flows.values()]])
         if isinstance(stmt, ast.Explain):
             return self._exec_explain(stmt)
+        if isinstance(stmt, ast.Copy):
+            return self._exec_copy(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _exec_copy(self, c: ast.Copy) -> QueryResult:
+        """COPY table TO/FROM file (reference: operator COPY via
+        common/datasource — parquet/csv/json by extension or WITH format)."""
+        import pyarrow as pa
+        fmt = c.options.get("format")
+        if fmt is None:
+            fmt = ("parquet" if c.path.endswith(".parquet")
+                   else "csv" if c.path.endswith(".csv")
+                   else "json" if c.path.endswith(".json") else "parquet")
+        if c.direction == "to":
+            r = self.execute_stmt(ast.Select([(ast.Star(), None)], c.table))
+            arrays, names = [], []
+            for n2, col, kind in zip(r.names, r.columns, r.kinds):
+                names.append(n2)
+                if kind == "ts":
+                    arrays.append(pa.array(np.asarray(col, dtype=np.int64),
+                                           type=pa.timestamp("ms")))
+                else:
+                    try:
+                        arrays.append(pa.array(np.asarray(col, dtype=np.float64)))
+                    except (ValueError, TypeError):
+                        arrays.append(pa.array([None if v is None else str(v)
+                                                for v in col], type=pa.string()))
+            t = pa.Table.from_arrays(arrays, names=names)
+            if fmt == "parquet":
+                import pyarrow.parquet as pq
+                pq.write_table(t, c.path)
+            elif fmt == "csv":
+                import pyarrow.csv as pacsv
+                pacsv.write_csv(t, c.path)
+            else:
+                with open(c.path, "w") as f:
+                    import json as _json
+                    for row in r.rows():
+                        f.write(_json.dumps(dict(zip(r.names, [
+                            v if not isinstance(v, (np.generic,)) else v.item()
+                            for v in row]))) + "\n")
+            return QueryResult(["rows"], [[len(r)]])
+        # COPY FROM
+        if fmt == "parquet":
+            import pyarrow.parquet as pq
+            t = pq.read_table(c.path)
+        elif fmt == "csv":
+            import pyarrow.csv as pacsv
+            t = pacsv.read_csv(c.path)
+        else:
+            import json as _json
+            rows = [_json.loads(l) for l in open(c.path) if l.strip()]
+            cols = {k: [r.get(k) for r in rows] for k in (rows[0] if rows else {})}
+            t = pa.Table.from_pydict(cols)
+        st = self.engine.table(c.table)
+        schema = st.schema
+        ts_name = schema.time_index.name
+        tag_names = [cc.name for cc in schema.tag_columns]
+        data = {cn: t.column(cn).to_pylist() for cn in t.column_names}
+        n = t.num_rows
+        ts_col = t.column(ts_name)
+        if pa.types.is_timestamp(ts_col.type):
+            ts_ms = ts_col.cast(pa.int64()).to_numpy(zero_copy_only=False)
+            unit = ts_col.type.unit
+            ts_ms = ts_ms // {"s": 1, "ms": 1, "us": 1000, "ns": 1_000_000}.get(unit, 1)
+            if unit == "s":
+                ts_ms = ts_ms * 1000
+        else:
+            ts_ms = np.asarray(data[ts_name], dtype=np.int64)
+        field_names = st.regions[0].field_names
+        str_names = st.regions[0].str_field_names
+        from greptimedb_amd.engine import pk_codec
+        from greptimedb_amd.engine.series import tsid_hash
+        rows_by_region: dict[int, list[int]] = {}
+        codes = np.empty(n, dtype=np.int32)
+        for i in range(n):
+            tags = tuple(None if data[tn][i] is None else str(data[tn][i])
+                         for tn in tag_names)
+            pk = pk_codec.encode_pk(tags)
+            ridx = tsid_hash(pk) % len(st.regions)
+            codes[i] = st.regions[ridx].register_series(tags)
+            rows_by_region.setdefault(ridx, []).append(i)
+        for ridx, rows in rows_by_region.items():
+            ra = np.array(rows)
+            fmat = np.full((len(field_names), len(ra)), np.nan)
+            for j, fn in enumerate(field_names):
+                if fn in data:
+                    col = np.asarray([data[fn][i] if data[fn][i] is not None
+                                      else np.nan for i in ra], dtype=np.float64)
+                    fmat[j] = col
+            strs = {sn: [data[sn][i] for i in ra] for sn in str_names if sn in data}
+            self.engine.write_region(st, ridx, codes[ra], ts_ms[ra], fmat, [],
+                                     str_fields=strs or None)
+        self.engine.commit_wal()
+        return QueryResult(["rows"], [[n]])
 
     def _flow_engine(self):
         """Flow engine shared per storage engine (lazy)."""
@@ -376,10 +470,54 @@ class Executor:
                 names.append(alias or f"col{i}")
                 cols.append([v])
             return QueryResult(names, cols)
+        from greptimedb_amd.query.information_schema import is_information_schema
+        if is_information_schema(sel.table):
+            return self._exec_information_schema(sel)
         plan = self._plan_select(sel)
         if plan.aggs:
             return self._exec_aggregate(sel, plan)
         return self._exec_raw(sel, plan)
+
+    def _exec_information_schema(self, sel: ast.Select) -> QueryResult:
+        from greptimedb_amd.query import information_schema as isch
+        names, cols = isch.build(self.engine, sel.table)
+        n = len(cols[0]) if cols else 0
+        data = dict(zip(names, cols))
+        keep = np.ones(n, dtype=bool)
+        if sel.where is not None:
+            def ev(e):
+                if isinstance(e, ast.BinOp) and e.op == "and":
+                    return ev(e.left) & ev(e.right)
+                if isinstance(e, ast.BinOp) and e.op == "or":
+                    return ev(e.left) | ev(e.right)
+                if isinstance(e, ast.BinOp) and e.op in ("=", "!=", "<", "<=", ">", ">="):
+                    l, r = e.left, e.right
+                    if isinstance(l, ast.Col) and isinstance(r, ast.Lit):
+                        col = data[l.name]
+                        return np.array([_py_cmp(e.op, v, r.value) for v in col])
+                raise PlanQuery("information_schema WHERE supports simple comparisons")
+            keep = ev(sel.where)
+        idx = np.flatnonzero(keep)
+        out_names, out_cols = [], []
+        for e, alias in sel.projections:
+            if isinstance(e, ast.Star):
+                out_names.extend(names)
+                out_cols.extend([data[c][idx] for c in names])
+            elif isinstance(e, ast.Col):
+                out_names.append(alias or e.name)
+                out_cols.append(data[e.name][idx])
+            else:
+                raise PlanQuery("information_schema projections: columns / *")
+        if sel.order_by:
+            o = np.arange(len(idx))
+            for e, desc in reversed(sel.order_by):
+                a = data[e.name][idx][o]
+                oo = np.argsort(a, kind="stable")
+                o = o[oo[::-1] if desc else oo]
+            out_cols = [c[o] for c in out_cols]
+        if sel.limit is not None:
+            out_cols = [c[: sel.limit] for c in out_cols]
+        return QueryResult(out_names, out_cols)
 
     def _plan_select(self, sel: ast.Select) -> SelectPlan:
         st = self.engine.table(sel.table)

@@ -1,0 +1,75 @@
+"""information_schema virtual tables.
+
+Reference parity: src/catalog/src/system_schema/information_schema — the
+tables/columns/partitions/region_statistics/flows/cluster_info views most
+used by tooling and dashboards.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from greptimedb_amd.models.schema import SemanticType
+
+VIRTUAL_TABLES = {
+    "tables", "columns", "region_statistics", "flows", "cluster_info",
+    "partitions",
+}
+
+
+def is_information_schema(name: str) -> bool:
+    return name.lower().startswith("information_schema.") and \
+        name.split(".", 1)[1].lower() in VIRTUAL_TABLES
+
+
+def build(engine, name: str):
+    """Return (names, columns) for the virtual table."""
+    kind = name.split(".", 1)[1].lower()
+    if kind == "tables":
+        rows = [("greptime", "public", t, "BASE TABLE", st.schema.table_id,
+                 "mito-hip", len(st.regions))
+                for t, st in sorted(engine.tables.items())]
+        return _cols(["table_catalog", "table_schema", "table_name",
+                      "table_type", "table_id", "engine", "region_count"], rows)
+    if kind == "columns":
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            for c in st.schema.columns:
+                rows.append((t, c.name, c.dtype.value,
+                             SemanticType(c.semantic).name))
+            for fn in st.regions[0].field_names:
+                if not st.schema.has_column(fn):
+                    rows.append((t, fn, "float64", "FIELD"))
+            for fn in st.regions[0].str_field_names:
+                if not st.schema.has_column(fn):
+                    rows.append((t, fn, "string", "FIELD"))
+        return _cols(["table_name", "column_name", "data_type", "semantic_type"],
+                     rows)
+    if kind in ("region_statistics", "partitions"):
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            for r in st.regions:
+                rows.append((r.region_id, t, r.region_id & 0xFFFFFFFF,
+                             r.memtable.len,
+                             sum(b.n for b in r.sst_cache.values()),
+                             len(r.sst_cache), len(r.series),
+                             r.device, r.memtable.bytes_used))
+        return _cols(["region_id", "table_name", "region_number",
+                      "memtable_rows", "sst_rows", "sst_files", "series",
+                      "device", "memtable_bytes"], rows)
+    if kind == "flows":
+        fe = getattr(engine, "flow_engine", None)
+        flows = list(fe.flows.values()) if fe else []
+        rows = [(f.name, f.source, f.sink, f.select_sql) for f in flows]
+        return _cols(["flow_name", "source_table", "sink_table", "query"], rows)
+    if kind == "cluster_info":
+        import torch
+        rows = [(0, "standalone", engine.config.device,
+                 torch.cuda.get_device_name(0) if torch.cuda.is_available() else "cpu")]
+        return _cols(["peer_id", "peer_type", "device", "device_name"], rows)
+    raise KeyError(kind)
+
+
+def _cols(names, rows):
+    cols = [np.array([r[i] for r in rows], dtype=object) for i in range(len(names))]
+    return names, cols

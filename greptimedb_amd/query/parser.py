@@ -182,6 +182,25 @@ class Parser:
             self.next()
             analyze = self.eat_kw("analyze")
             stmt = ast.Explain(analyze, self.parse_statement_inner())
+        elif self.at_kw("copy"):
+            self.next()
+            table = self.next().value
+            if self.eat_kw("to"):
+                direction = "to"
+            elif self.eat_kw("from"):
+                direction = "from"
+            else:
+                raise InvalidSyntax("COPY <table> TO|FROM '<path>'")
+            path = self.next().value
+            options = {}
+            if self.eat_kw("with"):
+                self.expect_op("(")
+                while not self.eat_op(")"):
+                    k = self.next().value
+                    self.expect_op("=")
+                    options[str(k).lower()] = str(self.next().value)
+                    self.eat_op(",")
+            stmt = ast.Copy(table, str(path), direction, options)
         else:
             raise InvalidSyntax(f"unsupported statement start: {self.peek()}")
         return stmt

@@ -1,0 +1,51 @@
+"""information_schema, COPY TO/FROM."""
+
+import numpy as np
+import pytest
+
+from greptimedb_amd.query.executor import Executor
+
+
+@pytest.fixture
+def ex(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE t1 (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, PRIMARY KEY (h))")
+    ex.execute("INSERT INTO t1 (h, ts, v) VALUES ('a', 1000, 1.5), ('b', 2000, 2.5)")
+    return ex
+
+
+def test_information_schema_tables(ex):
+    r = ex.execute("SELECT table_name, engine FROM information_schema.tables")
+    assert "t1" in list(r.columns[0])
+    r = ex.execute("SELECT * FROM information_schema.tables WHERE table_name = 't1'")
+    assert len(r) == 1
+
+
+def test_information_schema_columns(ex):
+    r = ex.execute("SELECT column_name, semantic_type FROM information_schema.columns "
+                   "WHERE table_name = 't1' ORDER BY column_name")
+    cols = dict(zip(r.columns[0], r.columns[1]))
+    assert cols["h"] == "TAG" and cols["ts"] == "TIMESTAMP" and cols["v"] == "FIELD"
+
+
+def test_information_schema_region_statistics(ex):
+    r = ex.execute("SELECT table_name, memtable_rows FROM "
+                   "information_schema.region_statistics WHERE table_name = 't1'")
+    assert sum(r.columns[1]) == 2
+
+
+def test_copy_roundtrip_parquet(ex, tmp_path):
+    p = str(tmp_path / "out.parquet")
+    r = ex.execute(f"COPY t1 TO '{p}'")
+    assert r.columns[0][0] == 2
+    ex.execute("CREATE TABLE t2 (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, PRIMARY KEY (h))")
+    r = ex.execute(f"COPY t2 FROM '{p}'")
+    assert r.columns[0][0] == 2
+    a = ex.execute("SELECT h, ts, v FROM t2 ORDER BY ts").rows()
+    assert [tuple(x) for x in a] == [("a", 1000, 1.5), ("b", 2000, 2.5)]
+
+
+def test_copy_csv(ex, tmp_path):
+    p = str(tmp_path / "out.csv")
+    ex.execute(f"COPY t1 TO '{p}' WITH (format = 'csv')")
+    assert open(p).read().count("\n") >= 2

@@ -19,31 +19,58 @@ def main(argv=None):
     st_sub = st.add_subparsers(dest="cmd", required=True)
     start = st_sub.add_parser("start")
     start.add_argument("--http-addr", default="0.0.0.0:4000")
+    start.add_argument("--mysql-addr", default="0.0.0.0:4002")
+    start.add_argument("--postgres-addr", default="0.0.0.0:4003")
     start.add_argument("--data-dir", default="./greptimedb_data")
     start.add_argument("--device", default="auto",
                        help="cuda:N / cpu / auto")
     start.add_argument("--regions", type=int, default=4)
     start.add_argument("--wal-sync", action="store_true")
+    start.add_argument("--user-provider", default=None,
+                       help="static_user_provider:file:<path>")
+    start.add_argument("--config", default=None, help="TOML config file")
     args = ap.parse_args(argv)
+
+    import asyncio
 
     import torch
     import uvicorn
 
     from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
     from greptimedb_amd.servers.http import ServerContext, build_app
+    from greptimedb_amd.servers.mysql import MySQLServer
+    from greptimedb_amd.servers.postgres import PostgresServer
+    from greptimedb_amd.utils.config import load_config
 
-    device = args.device
+    cfg = load_config(args.config) if args.config else {}
+    device = cfg.get("device", args.device)
     if device == "auto":
         device = "cuda:0" if torch.cuda.is_available() else "cpu"
     engine = MitoEngine(EngineConfig(
-        data_dir=args.data_dir, device=device,
-        default_regions=args.regions, wal_sync=args.wal_sync))
+        data_dir=cfg.get("data_dir", args.data_dir), device=device,
+        default_regions=int(cfg.get("regions", args.regions)),
+        wal_sync=bool(cfg.get("wal_sync", args.wal_sync))))
+    user_provider = None
+    if args.user_provider and args.user_provider.startswith("static_user_provider:file:"):
+        from greptimedb_amd.servers.auth import StaticUserProvider
+        user_provider = StaticUserProvider.from_file(
+            args.user_provider.split("file:", 1)[1])
     ctx = ServerContext(engine)
     app = build_app(ctx)
-    host, port = args.http_addr.rsplit(":", 1)
+    host, port = cfg.get("http_addr", args.http_addr).rsplit(":", 1)
+    my_host, my_port = cfg.get("mysql_addr", args.mysql_addr).rsplit(":", 1)
+    pg_host, pg_port = cfg.get("postgres_addr", args.postgres_addr).rsplit(":", 1)
     print(f"greptimedb_amd standalone: device={device} data={args.data_dir} "
-          f"http={host}:{port}", flush=True)
-    uvicorn.run(app, host=host, port=int(port), log_level="warning")
+          f"http={host}:{port} mysql={my_port} postgres={pg_port}", flush=True)
+
+    async def serve():
+        uv = uvicorn.Server(uvicorn.Config(app, host=host, port=int(port),
+                                           log_level="warning"))
+        my = MySQLServer(ctx.executor, my_host, int(my_port), user_provider)
+        pg = PostgresServer(ctx.executor, pg_host, int(pg_port), user_provider)
+        await asyncio.gather(uv.serve(), my.serve_forever(), pg.serve_forever())
+
+    asyncio.run(serve())
 
 
 if __name__ == "__main__":

@@ -1,0 +1,200 @@
+"""MySQL wire-protocol server (L7).
+
+Reference parity: src/servers/src/mysql/ (opensrv-mysql based). Implements
+protocol v10: handshake, CLIENT_PROTOCOL_41 text protocol, COM_QUERY /
+COM_PING / COM_QUIT / COM_INIT_DB, lenenc text resultsets with EOF framing
+(CLIENT_DEPRECATE_EOF intentionally not advertised). Auth accepts any
+credentials unless a UserProvider is configured (reference auth seam).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import math
+import struct
+
+import numpy as np
+
+from greptimedb_amd.utils.errors import GreptimeError
+
+CLIENT_PROTOCOL_41 = 0x00000200
+CLIENT_CONNECT_WITH_DB = 0x00000008
+CLIENT_PLUGIN_AUTH = 0x00080000
+
+MYSQL_TYPE_DOUBLE = 5
+MYSQL_TYPE_LONGLONG = 8
+MYSQL_TYPE_DATETIME = 12
+MYSQL_TYPE_VAR_STRING = 253
+
+
+def lenenc_int(n: int) -> bytes:
+    if n < 251:
+        return bytes([n])
+    if n < (1 << 16):
+        return b"\xfc" + struct.pack("<H", n)
+    if n < (1 << 24):
+        return b"\xfd" + struct.pack("<I", n)[:3]
+    return b"\xfe" + struct.pack("<Q", n)
+
+
+def lenenc_str(s: bytes) -> bytes:
+    return lenenc_int(len(s)) + s
+
+
+class MySQLServer:
+    def __init__(self, executor, host="0.0.0.0", port=4002, user_provider=None):
+        self.executor = executor
+        self.host = host
+        self.port = port
+        self.user_provider = user_provider
+        self._server = None
+
+    async def start(self):
+        self._server = await asyncio.start_server(self._handle, self.host, self.port)
+        return self._server
+
+    async def serve_forever(self):
+        await self.start()
+        async with self._server:
+            await self._server.serve_forever()
+
+    # ------------------------------------------------------- packet IO
+
+    @staticmethod
+    async def _read_packet(reader) -> tuple[int, bytes]:
+        hdr = await reader.readexactly(4)
+        ln = hdr[0] | (hdr[1] << 8) | (hdr[2] << 16)
+        seq = hdr[3]
+        return seq, await reader.readexactly(ln)
+
+    @staticmethod
+    def _packet(seq: int, payload: bytes) -> bytes:
+        ln = len(payload)
+        return bytes([ln & 0xFF, (ln >> 8) & 0xFF, (ln >> 16) & 0xFF, seq]) + payload
+
+    def _ok(self, seq, affected=0) -> bytes:
+        return self._packet(seq, b"\x00" + lenenc_int(affected) + lenenc_int(0) +
+                            struct.pack("<HH", 0x0002, 0))
+
+    def _eof(self, seq) -> bytes:
+        return self._packet(seq, b"\xfe" + struct.pack("<HH", 0, 0x0002))
+
+    def _err(self, seq, msg: str, code=1064) -> bytes:
+        return self._packet(seq, b"\xff" + struct.pack("<H", code) + b"#42000" +
+                            msg.encode()[:400])
+
+    # ------------------------------------------------------- session
+
+    async def _handle(self, reader, writer):
+        try:
+            # handshake v10
+            greeting = (b"\x0a" + b"greptimedb-amd-mysql\x00" +
+                        struct.pack("<I", 1) +          # thread id
+                        b"12345678\x00" +               # auth-plugin-data part1
+                        struct.pack("<H", (CLIENT_PROTOCOL_41 | CLIENT_CONNECT_WITH_DB |
+                                           CLIENT_PLUGIN_AUTH) & 0xFFFF) +
+                        bytes([33]) +                   # charset utf8
+                        struct.pack("<H", 0x0002) +     # status
+                        struct.pack("<H", ((CLIENT_PROTOCOL_41 | CLIENT_PLUGIN_AUTH)
+                                           >> 16) & 0xFFFF) +
+                        bytes([21]) + b"\x00" * 10 +
+                        b"123456789012\x00" +
+                        b"mysql_native_password\x00")
+            writer.write(self._packet(0, greeting))
+            await writer.drain()
+            seq, resp = await self._read_packet(reader)
+            if len(resp) >= 32:
+                user_end = resp.find(b"\x00", 32)
+                user = resp[32:user_end].decode(errors="replace") if user_end > 0 else ""
+            else:
+                user = ""
+            if self.user_provider is not None and not self.user_provider.allow(user):
+                writer.write(self._err(seq + 1, f"access denied for {user}", 1045))
+                await writer.drain()
+                writer.close()
+                return
+            writer.write(self._ok(seq + 1))
+            await writer.drain()
+
+            while True:
+                try:
+                    seq, cmd = await self._read_packet(reader)
+                except (asyncio.IncompleteReadError, ConnectionResetError):
+                    break
+                if not cmd:
+                    break
+                op = cmd[0]
+                if op == 0x01:       # COM_QUIT
+                    break
+                if op in (0x0E, 0x02):   # COM_PING / COM_INIT_DB
+                    writer.write(self._ok(1))
+                elif op == 0x03:     # COM_QUERY
+                    sql = cmd[1:].decode(errors="replace")
+                    writer.write(self._run_query(sql))
+                else:
+                    writer.write(self._err(1, f"unsupported command {op:#x}", 1047))
+                await writer.drain()
+        except (asyncio.IncompleteReadError, ConnectionResetError):
+            pass
+        finally:
+            try:
+                writer.close()
+            except Exception:
+                pass
+
+    def _run_query(self, sql: str) -> bytes:
+        s = sql.strip().rstrip(";").lower()
+        # common client handshake queries
+        if s.startswith(("set ", "set@", "use ")) or s in ("commit", "rollback", "begin"):
+            return self._ok(1)
+        if s.startswith("select @@") or s == "select version()":
+            return self._text_resultset(["version()"], [["8.4.0-greptimedb-amd"]],
+                                        [MYSQL_TYPE_VAR_STRING])
+        try:
+            r = self.executor.execute(sql)
+        except GreptimeError as e:
+            return self._err(1, str(e) or type(e).__name__)
+        except Exception as e:  # pragma: no cover
+            return self._err(1, f"{type(e).__name__}: {e}")
+        types = []
+        for col, kind in zip(r.columns, r.kinds):
+            if kind == "ts":
+                types.append(MYSQL_TYPE_DATETIME)
+            elif len(col) and isinstance(col[0], (int, np.integer)):
+                types.append(MYSQL_TYPE_LONGLONG)
+            elif len(col) and isinstance(col[0], (float, np.floating)):
+                types.append(MYSQL_TYPE_DOUBLE)
+            else:
+                types.append(MYSQL_TYPE_VAR_STRING)
+        rows = []
+        from greptimedb_amd.utils.timeutil import format_ts_ms
+        for row in r.rows():
+            out = []
+            for v, kind in zip(row, r.kinds):
+                if v is None or (isinstance(v, float) and math.isnan(v)):
+                    out.append(None)
+                elif kind == "ts":
+                    out.append(format_ts_ms(int(v)).replace("T", " "))
+                else:
+                    out.append(str(v))
+            rows.append(out)
+        return self._text_resultset(r.names, rows, types)
+
+    def _text_resultset(self, names, rows, types) -> bytes:
+        out = [self._packet(1, lenenc_int(len(names)))]
+        seq = 2
+        for name, typ in zip(names, types):
+            nb = name.encode()
+            col = (lenenc_str(b"def") + lenenc_str(b"") + lenenc_str(b"") +
+                   lenenc_str(b"") + lenenc_str(nb) + lenenc_str(nb) +
+                   b"\x0c" + struct.pack("<H", 33) + struct.pack("<I", 1024) +
+                   bytes([typ]) + struct.pack("<H", 0) + bytes([0]) + b"\x00\x00")
+            out.append(self._packet(seq, col))
+            seq += 1
+        out.append(self._eof(seq)); seq += 1
+        for row in rows:
+            payload = b"".join(b"\xfb" if v is None else lenenc_str(str(v).encode())
+                               for v in row)
+            out.append(self._packet(seq, payload)); seq += 1
+        out.append(self._eof(seq))
+        return b"".join(out)

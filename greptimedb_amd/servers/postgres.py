@@ -1,0 +1,167 @@
+"""PostgreSQL wire-protocol server (L7).
+
+Reference parity: src/servers/src/postgres/ (pgwire based). Implements the
+v3 protocol simple-query flow: StartupMessage → AuthenticationOk +
+ParameterStatus + ReadyForQuery; 'Q' → RowDescription/DataRow/
+CommandComplete; errors as ErrorResponse; SSLRequest politely declined.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import math
+import struct
+
+import numpy as np
+
+from greptimedb_amd.utils.errors import GreptimeError
+
+OID_TEXT = 25
+OID_INT8 = 20
+OID_FLOAT8 = 701
+OID_TIMESTAMP = 1114
+
+
+def _msg(tag: bytes, payload: bytes) -> bytes:
+    return tag + struct.pack("!I", len(payload) + 4) + payload
+
+
+class PostgresServer:
+    def __init__(self, executor, host="0.0.0.0", port=4003, user_provider=None):
+        self.executor = executor
+        self.host = host
+        self.port = port
+        self.user_provider = user_provider
+        self._server = None
+
+    async def start(self):
+        self._server = await asyncio.start_server(self._handle, self.host, self.port)
+        return self._server
+
+    async def serve_forever(self):
+        await self.start()
+        async with self._server:
+            await self._server.serve_forever()
+
+    async def _handle(self, reader, writer):
+        try:
+            # startup (possibly preceded by SSLRequest)
+            while True:
+                hdr = await reader.readexactly(4)
+                (ln,) = struct.unpack("!I", hdr)
+                body = await reader.readexactly(ln - 4)
+                if len(body) >= 4:
+                    (code,) = struct.unpack("!I", body[:4])
+                else:
+                    code = 0
+                if code == 80877103:      # SSLRequest → decline
+                    writer.write(b"N")
+                    await writer.drain()
+                    continue
+                if code == 80877102:      # CancelRequest → ignore
+                    writer.close()
+                    return
+                break  # StartupMessage
+            params = {}
+            parts = body[4:].split(b"\x00")
+            for i in range(0, len(parts) - 1, 2):
+                if parts[i]:
+                    params[parts[i].decode()] = parts[i + 1].decode()
+            user = params.get("user", "")
+            if self.user_provider is not None and not self.user_provider.allow(user):
+                writer.write(_msg(b"E", b"SSFATAL\x00C28000\x00M" +
+                                  f"auth failed for {user}".encode() + b"\x00\x00"))
+                await writer.drain()
+                writer.close()
+                return
+            writer.write(_msg(b"R", struct.pack("!I", 0)))  # AuthenticationOk
+            for k, v in (("server_version", "16.0 (greptimedb-amd)"),
+                         ("client_encoding", "UTF8"),
+                         ("DateStyle", "ISO"),
+                         ("server_encoding", "UTF8")):
+                writer.write(_msg(b"S", k.encode() + b"\x00" + v.encode() + b"\x00"))
+            writer.write(_msg(b"K", struct.pack("!II", 1, 1)))  # BackendKeyData
+            writer.write(_msg(b"Z", b"I"))
+            await writer.drain()
+
+            while True:
+                try:
+                    tag = await reader.readexactly(1)
+                except (asyncio.IncompleteReadError, ConnectionResetError):
+                    break
+                hdr = await reader.readexactly(4)
+                (ln,) = struct.unpack("!I", hdr)
+                body = await reader.readexactly(ln - 4)
+                if tag == b"X":  # Terminate
+                    break
+                if tag == b"Q":
+                    sql = body.rstrip(b"\x00").decode(errors="replace")
+                    for chunk in self._run_query(sql):
+                        writer.write(chunk)
+                    writer.write(_msg(b"Z", b"I"))
+                    await writer.drain()
+                elif tag in (b"P", b"B", b"D", b"E", b"S"):
+                    # minimal extended-protocol shim: report unsupported
+                    writer.write(_msg(b"E", b"SERROR\x00C0A000\x00M"
+                                      b"extended query protocol not supported\x00\x00"))
+                    writer.write(_msg(b"Z", b"I"))
+                    await writer.drain()
+                else:
+                    writer.write(_msg(b"Z", b"I"))
+                    await writer.drain()
+        except (asyncio.IncompleteReadError, ConnectionResetError):
+            pass
+        finally:
+            try:
+                writer.close()
+            except Exception:
+                pass
+
+    def _run_query(self, sql: str):
+        s = sql.strip().rstrip(";").lower()
+        if not s or s.startswith(("set ", "begin", "commit", "rollback")):
+            yield _msg(b"C", b"SET\x00")
+            return
+        try:
+            r = self.executor.execute(sql)
+        except GreptimeError as e:
+            yield _msg(b"E", b"SERROR\x00C42601\x00M" +
+                       (str(e) or type(e).__name__).encode()[:400] + b"\x00\x00")
+            return
+        except Exception as e:  # pragma: no cover
+            yield _msg(b"E", b"SERROR\x00CXX000\x00M" +
+                       f"{type(e).__name__}: {e}".encode()[:400] + b"\x00\x00")
+            return
+        # RowDescription
+        fields = b""
+        oids = []
+        for col, kind, name in zip(r.columns, r.kinds, r.names):
+            if kind == "ts":
+                oid = OID_TIMESTAMP
+            elif len(col) and isinstance(col[0], (int, np.integer)):
+                oid = OID_INT8
+            elif len(col) and isinstance(col[0], (float, np.floating)):
+                oid = OID_FLOAT8
+            else:
+                oid = OID_TEXT
+            oids.append(oid)
+            fields += (name.encode() + b"\x00" + struct.pack("!IhIhih", 0, 0, oid,
+                                                             -1, -1, 0))
+        yield _msg(b"T", struct.pack("!h", len(r.names)) + fields)
+        from greptimedb_amd.utils.timeutil import format_ts_ms
+        nrows = 0
+        for row in r.rows():
+            payload = struct.pack("!h", len(row))
+            for v, kind in zip(row, r.kinds):
+                if v is None or (isinstance(v, float) and math.isnan(v)):
+                    payload += struct.pack("!i", -1)
+                else:
+                    if kind == "ts":
+                        sv = format_ts_ms(int(v)).replace("T", " ")
+                    else:
+                        sv = str(v)
+                    b = sv.encode()
+                    payload += struct.pack("!i", len(b)) + b
+            yield _msg(b"D", payload)
+            nrows += 1
+        yield _msg(b"C", f"SELECT {nrows}".encode() + b"\x00")

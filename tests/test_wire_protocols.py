@@ -1,0 +1,185 @@
+"""MySQL + PostgreSQL wire protocol servers, exercised over real sockets
+with minimal spec-conformant clients."""
+
+import asyncio
+import struct
+
+import pytest
+
+from greptimedb_amd.query.executor import Executor
+from greptimedb_amd.servers.mysql import MySQLServer
+from greptimedb_amd.servers.postgres import PostgresServer
+
+
+@pytest.fixture
+def ex(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE t (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, PRIMARY KEY (h))")
+    ex.execute("INSERT INTO t (h, ts, v) VALUES ('a', 1000, 1.5), ('b', 2000, 2.5)")
+    return ex
+
+
+# ------------------------------------------------------------------- mysql
+
+async def _mysql_read_packet(reader):
+    hdr = await reader.readexactly(4)
+    ln = hdr[0] | (hdr[1] << 8) | (hdr[2] << 16)
+    return hdr[3], await reader.readexactly(ln)
+
+
+def _mysql_lenenc(buf, off):
+    b = buf[off]
+    if b < 251:
+        return b, off + 1
+    if b == 0xFC:
+        return struct.unpack_from("<H", buf, off + 1)[0], off + 3
+    if b == 0xFD:
+        return int.from_bytes(buf[off + 1:off + 4], "little"), off + 4
+    return struct.unpack_from("<Q", buf, off + 1)[0], off + 9
+
+
+async def _mysql_session(port, queries):
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    seq, greeting = await _mysql_read_packet(reader)
+    assert greeting[0] == 0x0A  # protocol v10
+    # handshake response 41
+    resp = (struct.pack("<IIB", 0x0200, 1 << 24, 33) + b"\x00" * 23 +
+            b"root\x00" + b"\x00")
+    ln = len(resp)
+    writer.write(bytes([ln & 0xFF, (ln >> 8) & 0xFF, (ln >> 16) & 0xFF, seq + 1]) + resp)
+    await writer.drain()
+    _seq, ok = await _mysql_read_packet(reader)
+    assert ok[0] == 0x00
+    results = []
+    for q in queries:
+        payload = b"\x03" + q.encode()
+        ln = len(payload)
+        writer.write(bytes([ln & 0xFF, (ln >> 8) & 0xFF, (ln >> 16) & 0xFF, 0]) + payload)
+        await writer.drain()
+        _seq, first = await _mysql_read_packet(reader)
+        if first[0] in (0x00, 0xFF):
+            results.append(("status", first))
+            continue
+        ncols, _ = _mysql_lenenc(first, 0)
+        for _ in range(ncols):
+            await _mysql_read_packet(reader)
+        _seq, eof = await _mysql_read_packet(reader)
+        assert eof[0] == 0xFE
+        rows = []
+        while True:
+            _seq, pkt = await _mysql_read_packet(reader)
+            if pkt[0] == 0xFE and len(pkt) < 9:
+                break
+            row, off = [], 0
+            for _ in range(ncols):
+                if pkt[off] == 0xFB:
+                    row.append(None)
+                    off += 1
+                else:
+                    ln2, off = _mysql_lenenc(pkt, off)
+                    row.append(pkt[off:off + ln2].decode())
+                    off += ln2
+            rows.append(row)
+        results.append(("rows", rows))
+    writer.write(b"\x01\x00\x00\x00\x01")  # COM_QUIT
+    writer.close()
+    return results
+
+
+def test_mysql_protocol(ex):
+    async def run():
+        srv = MySQLServer(ex, host="127.0.0.1", port=0)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        out = await _mysql_session(port, [
+            "SELECT h, ts, v FROM t ORDER BY ts",
+            "SELECT count(*) FROM t",
+            "SET NAMES utf8",
+            "SELEKT broken",
+        ])
+        s.close()
+        return out
+    out = asyncio.run(run())
+    assert out[0][0] == "rows"
+    assert out[0][1] == [["a", "1970-01-01 00:00:01", "1.5"],
+                         ["b", "1970-01-01 00:00:02", "2.5"]]
+    assert out[1][1] == [["2"]]
+    assert out[2][0] == "status" and out[2][1][0] == 0x00
+    assert out[3][0] == "status" and out[3][1][0] == 0xFF  # error packet
+
+
+# ------------------------------------------------------------------- postgres
+
+async def _pg_session(port, queries):
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    params = b"user\x00tester\x00database\x00public\x00\x00"
+    body = struct.pack("!I", 196608) + params
+    writer.write(struct.pack("!I", len(body) + 4) + body)
+    await writer.drain()
+    # read until ReadyForQuery
+    async def read_msg():
+        tag = await reader.readexactly(1)
+        (ln,) = struct.unpack("!I", await reader.readexactly(4))
+        return tag, await reader.readexactly(ln - 4)
+    while True:
+        tag, body = await read_msg()
+        if tag == b"Z":
+            break
+    results = []
+    for q in queries:
+        payload = q.encode() + b"\x00"
+        writer.write(b"Q" + struct.pack("!I", len(payload) + 4) + payload)
+        await writer.drain()
+        rows, names, status = [], [], None
+        while True:
+            tag, body = await read_msg()
+            if tag == b"T":
+                (ncols,) = struct.unpack_from("!h", body, 0)
+                off = 2
+                for _ in range(ncols):
+                    end = body.index(b"\x00", off)
+                    names.append(body[off:end].decode())
+                    off = end + 1 + 18
+            elif tag == b"D":
+                (ncols,) = struct.unpack_from("!h", body, 0)
+                off = 2
+                row = []
+                for _ in range(ncols):
+                    (ln2,) = struct.unpack_from("!i", body, off)
+                    off += 4
+                    if ln2 < 0:
+                        row.append(None)
+                    else:
+                        row.append(body[off:off + ln2].decode())
+                        off += ln2
+                rows.append(row)
+            elif tag == b"C":
+                status = body.rstrip(b"\x00").decode()
+            elif tag == b"E":
+                status = "ERROR"
+            elif tag == b"Z":
+                break
+        results.append((status, names, rows))
+    writer.write(b"X" + struct.pack("!I", 4))
+    writer.close()
+    return results
+
+
+def test_postgres_protocol(ex):
+    async def run():
+        srv = PostgresServer(ex, host="127.0.0.1", port=0)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        out = await _pg_session(port, [
+            "SELECT h, v FROM t ORDER BY h",
+            "SELECT count(*) FROM t",
+            "SELEKT nope",
+        ])
+        s.close()
+        return out
+    out = asyncio.run(run())
+    assert out[0][0] == "SELECT 2"
+    assert out[0][1] == ["h", "v"]
+    assert out[0][2] == [["a", "1.5"], ["b", "2.5"]]
+    assert out[1][2] == [["2"]]
+    assert out[2][0] == "ERROR"

@@ -64,6 +64,42 @@ class PromStore:
             setattr(self, name, na)
         self._cap = cap
 
+    def write_points(self, points) -> int:
+        """Direct datapoint ingestion: [(metric, {tag: value}, ts_ms, value)]
+        (OpenTSDB put / OTLP metric datapoints land here)."""
+        if not points:
+            return 0
+        st = self.table
+        n = len(points)
+        series = np.empty(n, dtype=np.int32)
+        regions = np.empty(n, dtype=np.int32)
+        ts = np.empty(n, dtype=np.int64)
+        vals = np.empty(n, dtype=np.float64)
+        for i, (metric, tags, t, v) in enumerate(points):
+            lab = {k: str(tv) for k, tv in (tags or {}).items()}
+            lab["__name__"] = metric
+            self.metrics.add(metric)
+            pk = pk_codec.encode_sparse(lab)
+            ridx = tsid_hash(pk) % len(st.regions)
+            series[i] = st.regions[ridx].register_series_labels(lab)
+            regions[i] = ridx
+            ts[i] = int(t)
+            vals[i] = float(v)
+        order = np.argsort(regions, kind="stable")
+        rs = regions[order]
+        bounds = np.flatnonzero(np.diff(rs)) + 1
+        for s, e in zip(np.concatenate(([0], bounds)),
+                        np.concatenate((bounds, [n]))):
+            ridx = int(rs[s])
+            rows = order[s:e]
+            self.engine.write_region(st, ridx, series[rows], ts[rows],
+                                     vals[rows][None, :], [],
+                                     durable=self.durable)
+        if self.durable:
+            self.engine.commit_wal()
+        self.rows_ingested += n
+        return n
+
     def write(self, body: bytes, snappy: bool = True) -> int:
         """Ingest one remote-write request body. Returns sample count."""
         series, ts, vals, new_series = self.parser.parse(body, snappy)

@@ -188,6 +188,145 @@ def build_app(ctx: ServerContext) -> FastAPI:
         metrics_mod.counter("loki_lines").inc(n)
         return Response(status_code=204)
 
+    # ---------------- opentsdb / elasticsearch / splunk ----------------
+
+    @app.post("/v1/opentsdb/api/put")
+    async def opentsdb_put(request: Request):
+        import json as _json
+        body = _json.loads(await request.body())
+        if isinstance(body, dict):
+            body = [body]
+        pts = []
+        for d in body:
+            t = int(d["timestamp"])
+            if t < 10 ** 12:        # seconds → ms (opentsdb sends seconds)
+                t *= 1000
+            pts.append((d["metric"], d.get("tags", {}), t, d["value"]))
+        n = ctx.promstore.write_points(pts)
+        metrics_mod.counter("opentsdb_points").inc(n)
+        return Response(status_code=204)
+
+    @app.post("/v1/elasticsearch/_bulk")
+    @app.post("/v1/elasticsearch/{index}/_bulk")
+    async def es_bulk(request: Request, index: str = "es_logs"):
+        import json as _json
+        lines = (await request.body()).splitlines()
+        entries = []
+        i = 0
+        while i < len(lines):
+            if not lines[i].strip():
+                i += 1
+                continue
+            action = _json.loads(lines[i])
+            i += 1
+            if ("index" in action or "create" in action) and i < len(lines):
+                doc = _json.loads(lines[i])
+                i += 1
+                entries.append(doc)
+        n = ctx.logstore.ingest(index, entries, tag_keys=[], ts_key="@timestamp")
+        metrics_mod.counter("es_bulk_docs").inc(n)
+        return {"took": 1, "errors": False,
+                "items": [{"index": {"status": 201}} for _ in range(n)]}
+
+    @app.post("/v1/splunk/services/collector")
+    @app.post("/v1/splunk/services/collector/event")
+    async def splunk_hec(request: Request):
+        import json as _json
+        body = (await request.body()).decode()
+        entries = []
+        dec = _json.JSONDecoder()
+        idx = 0
+        while idx < len(body):
+            while idx < len(body) and body[idx] in " \r\n\t":
+                idx += 1
+            if idx >= len(body):
+                break
+            obj, idx = dec.raw_decode(body, idx)
+            e = {"timestamp": int(float(obj.get("time", 0)) * 1000) or None}
+            ev = obj.get("event")
+            if isinstance(ev, dict):
+                e.update(ev)
+            else:
+                e["message"] = str(ev)
+            for k in ("host", "source", "sourcetype"):
+                if obj.get(k):
+                    e[k] = obj[k]
+            entries.append({k: v for k, v in e.items() if v is not None})
+        n = ctx.logstore.ingest("splunk_logs", entries, tag_keys=["host"],
+                                ts_key="timestamp")
+        metrics_mod.counter("splunk_events").inc(n)
+        return {"text": "Success", "code": 0}
+
+    # ---------------- jaeger query API ----------------
+
+    @app.get("/v1/jaeger/api/services")
+    def jaeger_services():
+        try:
+            st = ctx.engine.table("opentelemetry_traces")
+        except Exception:
+            return {"data": [], "total": 0}
+        vals = set()
+        for region in st.regions:
+            vals.update(region.series.inverted.get("service_name", {}).keys())
+        return {"data": sorted(vals), "total": len(vals)}
+
+    @app.get("/v1/jaeger/api/services/{service}/operations")
+    def jaeger_operations(service: str):
+        try:
+            st = ctx.engine.table("opentelemetry_traces")
+        except Exception:
+            return {"data": [], "total": 0}
+        ops = set()
+        for region in st.regions:
+            for code in region.series.codes_for_eq("service_name", service):
+                ops.add(region.series.tag_values[code][1])
+        return {"data": sorted(o for o in ops if o), "total": len(ops)}
+
+    def _jaeger_trace_json(rows):
+        spans = []
+        procs = {}
+        for svc, span_name, ts, dur, trace_id, span_id, parent in rows:
+            pid = f"p-{svc}"
+            procs[pid] = {"serviceName": svc, "tags": []}
+            refs = []
+            if parent:
+                refs.append({"refType": "CHILD_OF", "traceID": trace_id,
+                             "spanID": parent})
+            spans.append({
+                "traceID": trace_id, "spanID": span_id,
+                "operationName": span_name, "references": refs,
+                "startTime": int(ts) * 1000, "duration": int(float(dur) * 1000),
+                "processID": pid, "tags": [], "logs": [],
+            })
+        return {"spans": spans, "processes": procs,
+                "traceID": spans[0]["traceID"] if spans else ""}
+
+    @app.get("/v1/jaeger/api/traces/{trace_id}")
+    def jaeger_trace(trace_id: str):
+        r = ctx.executor.execute(
+            "SELECT service_name, span_name, ts, duration_ms, trace_id, "
+            f"span_id, parent_span_id FROM opentelemetry_traces WHERE "
+            f"trace_id = '{trace_id}' ORDER BY ts")
+        if len(r) == 0:
+            return {"data": [], "total": 0}
+        return {"data": [_jaeger_trace_json(list(r.rows()))], "total": 1}
+
+    @app.get("/v1/jaeger/api/traces")
+    def jaeger_traces(service: str = Query(None), limit: int = Query(20)):
+        where = f"WHERE service_name = '{service}'" if service else ""
+        r = ctx.executor.execute(
+            "SELECT service_name, span_name, ts, duration_ms, trace_id, "
+            f"span_id, parent_span_id FROM opentelemetry_traces {where} "
+            f"ORDER BY ts DESC LIMIT {max(limit, 1) * 10}")
+        by_trace: dict[str, list] = {}
+        for row in r.rows():
+            by_trace.setdefault(row[4], []).append(row)
+            if len(by_trace) > limit:
+                by_trace.pop(row[4])
+                break
+        data = [_jaeger_trace_json(rows) for rows in by_trace.values()]
+        return {"data": data, "total": len(data)}
+
     # ---------------- OTLP ----------------
 
     @app.post("/v1/otlp/v1/traces")

@@ -44,6 +44,8 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--scale", type=int, default=SCALE_PER_RANK)
+    ap.add_argument("--workers", type=int, default=4,
+                    help="ingest worker threads per rank (TSBS uses 6)")
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--durable", action="store_true", default=True)
     args = ap.parse_args()
@@ -80,20 +82,33 @@ def main():
         data_dir=data_dir, device=device,
         background_flush=True, wal_sync=False,
         flush_bytes=1 << 30))
-    ing = Ingestor(eng, default_regions=4, append_mode=True, durable=args.durable)
 
     # ---------------- pre-generate all line batches (untimed) ----------------
-    w = CpuWorkload(scale=args.scale, seed=7 + rank)
-    if world > 1:
-        w.tagsets = [t.replace(b"host_", b"host_%d_" % rank) for t in w.tagsets]
+    # Each worker thread owns a disjoint host shard (its own Ingestor/parser,
+    # like the reference's 6 TSBS client workers; P5 write-worker axis) —
+    # parse (C++, GIL released) and H2D copies overlap across workers.
+    import threading
+
+    n_workers = max(args.workers, 1)
     total_steps = args.warmup + args.steps
     gen_t0 = time.perf_counter()
-    step_batches = [
-        [w.next_batch(ROWS_PER_BATCH) for _ in range(BATCHES_PER_STEP)]
-        for _ in range(total_steps)
-    ]
-    log(f"# generated {total_steps * BATCHES_PER_STEP * ROWS_PER_BATCH} rows "
-        f"in {time.perf_counter() - gen_t0:.1f}s")
+    per_worker_scale = max(args.scale // n_workers, 1)
+    workers = []
+    for wi in range(n_workers):
+        w = CpuWorkload(scale=per_worker_scale, seed=7 + rank * 100 + wi)
+        w.tagsets = [t.replace(b"host_", b"host_%d_%d_" % (rank, wi))
+                     for t in w.tagsets]
+        batches = [
+            [w.next_batch(ROWS_PER_BATCH)
+             for _ in range(BATCHES_PER_STEP // n_workers)]
+            for _ in range(total_steps)
+        ]
+        ing = Ingestor(eng, default_regions=4, append_mode=True,
+                       durable=args.durable)
+        workers.append((ing, batches))
+    rows_per_step = n_workers * (BATCHES_PER_STEP // n_workers) * ROWS_PER_BATCH
+    log(f"# generated {total_steps * rows_per_step} rows "
+        f"in {time.perf_counter() - gen_t0:.1f}s ({n_workers} workers)")
 
     def barrier_sync():
         if dist_on:
@@ -102,19 +117,24 @@ def main():
         if have_gpu:
             torch.cuda.synchronize()
 
-    def run_step(i):
-        for b in step_batches[i]:
-            ing.ingest_lines(b)
+    def run_steps(lo, hi):
+        def worker_fn(ing, batches):
+            for i in range(lo, hi):
+                for b in batches[i]:
+                    ing.ingest_lines(b)
+        threads = [threading.Thread(target=worker_fn, args=wk) for wk in workers]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
 
     # ---------------- warmup ----------------
-    for i in range(args.warmup):
-        run_step(i)
+    run_steps(0, args.warmup)
     barrier_sync()
 
     # ---------------- timed ----------------
     t0 = time.perf_counter()
-    for i in range(args.warmup, total_steps):
-        run_step(i)
+    run_steps(args.warmup, total_steps)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -126,7 +146,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    rows_per_rank = args.steps * BATCHES_PER_STEP * ROWS_PER_BATCH
+    rows_per_rank = args.steps * rows_per_step
     total_rows = rows_per_rank * world
     rows_per_s = total_rows / elapsed
     ms_per_step = elapsed / args.steps * 1000
@@ -134,7 +154,7 @@ def main():
     # ---------------- query latencies (informational, untimed region) -------
     from greptimedb_amd.parallel.dist import DistContext
     ex = Executor(eng, dist=DistContext(device=device) if dist_on else None)
-    host = "host_0" if world == 1 else "host_0_0"
+    host = f"host_{rank}_0_0"
     t_lo = 1451606400000
     t_hi = t_lo + 3600_000
     q_single = (f"SELECT date_trunc('minute', ts) AS minute, max(usage_user) FROM cpu "
@@ -175,6 +195,7 @@ def main():
             "seq_len": 0,
             "parallelism": f"region-shard dp{world}",
             "scale_per_gpu": args.scale,
+            "workers": n_workers,
             "batches_per_step": BATCHES_PER_STEP,
             "wal": "group-commit, no fsync",
             "queries": queries,

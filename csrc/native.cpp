@@ -28,20 +28,37 @@
 
 namespace py = pybind11;
 
-// ---------------------------------------------------------------- crc32 (IEEE, zlib-compatible)
-static uint32_t crc_table[256];
+// ------------------------------------- crc32 (IEEE, zlib-compatible, slice-by-8)
+static uint32_t crc_table[8][256];
 static bool crc_init_done = [] {
   for (uint32_t i = 0; i < 256; i++) {
     uint32_t c = i;
     for (int k = 0; k < 8; k++) c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
-    crc_table[i] = c;
+    crc_table[0][i] = c;
   }
+  for (int t = 1; t < 8; t++)
+    for (uint32_t i = 0; i < 256; i++)
+      crc_table[t][i] = crc_table[0][crc_table[t - 1][i] & 0xFF] ^
+                        (crc_table[t - 1][i] >> 8);
   return true;
 }();
 
 static uint32_t crc32_update(uint32_t crc, const uint8_t* buf, size_t len) {
   crc = ~crc;
-  for (size_t i = 0; i < len; i++) crc = crc_table[(crc ^ buf[i]) & 0xFF] ^ (crc >> 8);
+  while (len >= 8) {
+    uint32_t lo, hi;
+    std::memcpy(&lo, buf, 4);
+    std::memcpy(&hi, buf + 4, 4);
+    lo ^= crc;
+    crc = crc_table[7][lo & 0xFF] ^ crc_table[6][(lo >> 8) & 0xFF] ^
+          crc_table[5][(lo >> 16) & 0xFF] ^ crc_table[4][lo >> 24] ^
+          crc_table[3][hi & 0xFF] ^ crc_table[2][(hi >> 8) & 0xFF] ^
+          crc_table[1][(hi >> 16) & 0xFF] ^ crc_table[0][hi >> 24];
+    buf += 8;
+    len -= 8;
+  }
+  for (size_t i = 0; i < len; i++)
+    crc = crc_table[0][(crc ^ buf[i]) & 0xFF] ^ (crc >> 8);
   return ~crc;
 }
 
@@ -195,6 +212,7 @@ class LineParser {
     // fieldset: k=v,k=v,...
     const char* f = sp1 + 1;
     const char* fend = sp2 ? sp2 : end;
+    size_t fpos = 0;
     while (f < fend) {
       const char* eq = static_cast<const char*>(memchr(f, '=', fend - f));
       if (!eq) break;
@@ -212,7 +230,20 @@ class LineParser {
       } else {
         val = fast_atof(v, vend);
       }
-      size_t c = field_idx(f, eq - f);
+      // positional field cache: batches of lines share a fieldset layout
+      // (TSBS and most collectors), so field j of each row is usually the
+      // same name — memcmp against the cached name skips hash+map
+      size_t c;
+      const size_t flen = eq - f;
+      if (fpos < fcache_.size() && fcache_[fpos].first == flen &&
+          memcmp(field_names_[fcache_[fpos].second].data(), f, flen) == 0) {
+        c = fcache_[fpos].second;
+      } else {
+        c = field_idx(f, flen);
+        if (fpos >= fcache_.size()) fcache_.resize(fpos + 1);
+        fcache_[fpos] = {flen, c};
+      }
+      fpos++;
       if (c >= r.cols.size()) r.cols.resize(c + 1);
       auto& col = r.cols[c];
       if (col.size() < row) col.resize(row, std::nan(""));
@@ -239,6 +270,7 @@ class LineParser {
   int32_t next_id_ = 0;
   std::unordered_multimap<uint64_t, size_t> fmap_;
   std::vector<std::string> field_names_;
+  std::vector<std::pair<size_t, size_t>> fcache_;  // positional (len, idx)
 };
 
 // ---------------------------------------------------------------- WalWriter

@@ -49,6 +49,7 @@ class MitoEngine:
         os.makedirs(config.data_dir, exist_ok=True)
         self.tables: dict[str, TableState] = {}
         self.next_table_id = 1024
+        self._ddl_lock = threading.Lock()
         self._catalog_path = os.path.join(config.data_dir, "catalog.json")
         self.wal = Wal(os.path.join(config.data_dir, "wal"),
                        segment_bytes=config.wal_segment_bytes,
@@ -103,6 +104,12 @@ class MitoEngine:
 
     def create_table(self, schema: TableSchema, n_regions: int | None = None,
                      append_mode: bool = False, if_not_exists: bool = False) -> TableState:
+        with self._ddl_lock:
+            return self._create_table_locked(schema, n_regions, append_mode,
+                                             if_not_exists)
+
+    def _create_table_locked(self, schema, n_regions, append_mode,
+                             if_not_exists) -> TableState:
         if schema.name in self.tables:
             if if_not_exists:
                 return self.tables[schema.name]

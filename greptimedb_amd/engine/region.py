@@ -93,23 +93,25 @@ class Region:
 
     def register_series(self, tags: tuple) -> int:
         """Get-or-create a local series code; new codes are logged durably."""
-        prev = len(self.series)
-        code = self.series.get_or_create(tags)
-        if code >= prev:
-            pk = self.series.pks[code]
-            self._series_log.write(struct.pack("<I", len(pk)) + pk)
-            self._series_log.flush()
-        return code
+        with self.lock:
+            prev = len(self.series)
+            code = self.series.get_or_create(tags)
+            if code >= prev:
+                pk = self.series.pks[code]
+                self._series_log.write(struct.pack("<I", len(pk)) + pk)
+                self._series_log.flush()
+            return code
 
     def register_series_labels(self, labels: dict) -> int:
         """Sparse/metric-engine mode: dynamic label-set series."""
-        prev = len(self.series)
-        code = self.series.get_or_create_labels(labels)
-        if code >= prev:
-            pk = self.series.pks[code]
-            self._series_log.write(struct.pack("<I", len(pk)) + pk)
-            self._series_log.flush()
-        return code
+        with self.lock:
+            prev = len(self.series)
+            code = self.series.get_or_create_labels(labels)
+            if code >= prev:
+                pk = self.series.pks[code]
+                self._series_log.write(struct.pack("<I", len(pk)) + pk)
+                self._series_log.flush()
+            return code
 
     def register_series_bulk(self, labels_list) -> np.ndarray:
         """Bulk registration (fixture/bulk-ingest path): one series-log

@@ -13,6 +13,7 @@ from __future__ import annotations
 import json
 import os
 import struct
+import threading
 
 import numpy as np
 
@@ -78,6 +79,7 @@ class Wal:
         self.sync_on_commit = sync_on_commit
         self.writer = _native.WalWriter()
         self.next_seq = 1
+        self._lock = threading.Lock()  # multi-worker ingest (P5 write workers)
         segs = self.segments()
         if segs:
             # resume: next_seq = last replayed seq + 1 (caller replays first)
@@ -97,16 +99,18 @@ class Wal:
         self.writer.open_segment(self._seg_path(self.next_seq))
 
     def append(self, region_id: int, payload: bytes) -> int:
-        seq = self.next_seq
-        self.next_seq += 1
-        self.writer.append(region_id, seq, payload)
-        return seq
+        with self._lock:
+            seq = self.next_seq
+            self.next_seq += 1
+            self.writer.append(region_id, seq, payload)
+            return seq
 
     def commit(self):
-        size = self.writer.commit(self.sync_on_commit)
-        if size >= self.segment_bytes:
-            self.writer.close_segment()
-            self._open_new_segment()
+        with self._lock:
+            size = self.writer.commit(self.sync_on_commit)
+            if size >= self.segment_bytes:
+                self.writer.close_segment()
+                self._open_new_segment()
 
     def replay(self):
         """Yield (seg_name, region_id, seq, payload) in order."""

@@ -43,11 +43,15 @@ void launch_prom_range_eval(
 // ts: i64[n] (ms), series: i32[n], fields: f64[nf_total, field_stride]
 // field_idx: i32[nf] rows of `fields` to aggregate, slot_lut: i32[lut_size].
 // Returns (sum f64, count i64, min f64, max f64), each [nf, n_slots, n_buckets].
+// When `acc` (5 tensors from a prior call) is passed, accumulation continues
+// into the same buffers — multi-source scans make ONE set of outputs with no
+// per-source combine (atomics make cross-launch accumulation safe).
 std::vector<torch::Tensor> ts_bucket_agg(
     torch::Tensor ts, torch::Tensor series, torch::Tensor fields,
     torch::Tensor field_idx, torch::Tensor slot_lut,
     int64_t ts_lo, int64_t ts_hi, int64_t origin, int64_t bucket_ms,
-    int64_t n_slots, int64_t n_buckets) {
+    int64_t n_slots, int64_t n_buckets,
+    std::vector<torch::Tensor> acc) {
   CHECK_GPU(ts); CHECK_GPU(series); CHECK_GPU(fields);
   CHECK_GPU(field_idx); CHECK_GPU(slot_lut);
   CHECK_CONTIG(ts); CHECK_CONTIG(series); CHECK_CONTIG(fields);
@@ -63,12 +67,19 @@ std::vector<torch::Tensor> ts_bucket_agg(
   auto opts_f64 = ts.options().dtype(torch::kFloat64);
   auto opts_i64 = ts.options().dtype(torch::kInt64);
   const int64_t cells = nf * n_slots * n_buckets;
-  auto sum = torch::zeros({nf, n_slots, n_buckets}, opts_f64);
-  auto cnt = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
-  auto rows = torch::zeros({n_slots, n_buckets}, opts_i64);
-  // min keys init to u64 max, max keys to 0
-  auto minmax_init_min = torch::full({nf, n_slots, n_buckets}, -1, opts_i64);  // 0xFFFF...
-  auto minmax_init_max = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+  torch::Tensor sum, cnt, rows, minmax_init_min, minmax_init_max;
+  if (!acc.empty()) {
+    TORCH_CHECK(acc.size() == 5, "acc must be the 5 raw accumulators");
+    sum = acc[0]; cnt = acc[1]; minmax_init_min = acc[2];
+    minmax_init_max = acc[3]; rows = acc[4];
+  } else {
+    sum = torch::zeros({nf, n_slots, n_buckets}, opts_f64);
+    cnt = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+    rows = torch::zeros({n_slots, n_buckets}, opts_i64);
+    // min keys init to u64 max, max keys to 0
+    minmax_init_min = torch::full({nf, n_slots, n_buckets}, -1, opts_i64);
+    minmax_init_max = torch::zeros({nf, n_slots, n_buckets}, opts_i64);
+  }
   auto stream = at::cuda::getCurrentHIPStream().stream();
   auto cell = torch::empty({n}, ts.options().dtype(torch::kInt32));
   gdb_hip::launch_bucket_agg2(
@@ -83,14 +94,25 @@ std::vector<torch::Tensor> ts_bucket_agg(
       reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(rows.data_ptr<int64_t>()),
       stream);
-  auto minv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
-  auto maxv = torch::empty({nf, n_slots, n_buckets}, opts_f64);
+  return {sum, cnt, minmax_init_min, minmax_init_max, rows};
+}
+
+// Decode raw accumulators (min/max u64 keys) → final (sum, cnt, min f64,
+// max f64, rows). Call once after the last ts_bucket_agg of a scan.
+std::vector<torch::Tensor> ts_bucket_agg_finish(std::vector<torch::Tensor> acc) {
+  TORCH_CHECK(acc.size() == 5);
+  auto sum = acc[0];
+  auto cnt = acc[1];
+  const int64_t cells = cnt.numel();
+  auto minv = torch::empty_like(sum);
+  auto maxv = torch::empty_like(sum);
+  auto stream = at::cuda::getCurrentHIPStream().stream();
   gdb_hip::launch_decode_minmax(
-      reinterpret_cast<unsigned long long*>(minmax_init_min.data_ptr<int64_t>()),
-      reinterpret_cast<unsigned long long*>(minmax_init_max.data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(acc[2].data_ptr<int64_t>()),
+      reinterpret_cast<unsigned long long*>(acc[3].data_ptr<int64_t>()),
       reinterpret_cast<unsigned long long*>(cnt.data_ptr<int64_t>()),
       minv.data_ptr<double>(), maxv.data_ptr<double>(), cells, stream);
-  return {sum, cnt, minv, maxv, rows};
+  return {sum, cnt, minv, maxv, acc[4]};
 }
 
 torch::Tensor filter_series_time(
@@ -172,7 +194,12 @@ torch::Tensor prom_range_eval(
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate");
+  m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate",
+        py::arg("ts"), py::arg("series"), py::arg("fields"), py::arg("field_idx"),
+        py::arg("slot_lut"), py::arg("ts_lo"), py::arg("ts_hi"), py::arg("origin"),
+        py::arg("bucket_ms"), py::arg("n_slots"), py::arg("n_buckets"),
+        py::arg("acc") = std::vector<torch::Tensor>());
+  m.def("ts_bucket_agg_finish", &ts_bucket_agg_finish, "decode raw accumulators");
   m.def("filter_series_time", &filter_series_time, "series/time filter mask");
   m.def("dedup_mark_last", &dedup_mark_last, "last-row dedup marker");
   m.def("series_last_ts", &series_last_ts, "per-slot max-ts accumulate (lastpoint)");

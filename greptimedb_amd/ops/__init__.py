@@ -14,4 +14,6 @@ from greptimedb_amd.ops.kernels import (  # noqa: F401
     prom_range_eval,
     series_last,
     ts_bucket_agg,
+    ts_bucket_agg_acc,
+    ts_bucket_agg_finish,
 )

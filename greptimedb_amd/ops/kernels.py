@@ -31,16 +31,45 @@ def _require_hip():
     return _HIP_OPS
 
 
-def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
-                  origin, bucket_ms, n_slots, n_buckets):
+def ts_bucket_agg_acc(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
+                      origin, bucket_ms, n_slots, n_buckets, acc=None):
+    """Accumulating form for multi-source scans: pass the returned opaque
+    `acc` into the next call, then ts_bucket_agg_finish(acc). On GPU the
+    kernel atomics accumulate into shared buffers (no per-source combine)."""
     if ts.is_cuda:
         ops = _require_hip()
-        return tuple(ops.ts_bucket_agg(
+        return ("gpu", ops.ts_bucket_agg(
             ts, series, fields, field_idx, slot_lut,
             int(ts_lo), int(ts_hi), int(origin), int(bucket_ms),
-            int(n_slots), int(n_buckets)))
-    return cpu_ref.ts_bucket_agg(ts, series, fields, field_idx, slot_lut,
-                                 ts_lo, ts_hi, origin, bucket_ms, n_slots, n_buckets)
+            int(n_slots), int(n_buckets),
+            acc[1] if acc is not None else []))
+    out = cpu_ref.ts_bucket_agg(ts, series, fields, field_idx, slot_lut,
+                                ts_lo, ts_hi, origin, bucket_ms,
+                                n_slots, n_buckets)
+    if acc is None:
+        return ("cpu", list(out))
+    prev = acc[1]
+    prev[0] += out[0]
+    prev[1] += out[1]
+    prev[2] = torch.fmin(prev[2], out[2])
+    prev[3] = torch.fmax(prev[3], out[3])
+    prev[4] += out[4]
+    return ("cpu", prev)
+
+
+def ts_bucket_agg_finish(acc):
+    """(sum, cnt, min, max, rows) from an accumulator handle."""
+    kind, t = acc
+    if kind == "gpu":
+        return tuple(_require_hip().ts_bucket_agg_finish(t))
+    return tuple(t)
+
+
+def ts_bucket_agg(ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi,
+                  origin, bucket_ms, n_slots, n_buckets):
+    return ts_bucket_agg_finish(ts_bucket_agg_acc(
+        ts, series, fields, field_idx, slot_lut, ts_lo, ts_hi, origin,
+        bucket_ms, n_slots, n_buckets))
 
 
 def filter_series_time(ts, series, slot_lut, ts_lo, ts_hi):

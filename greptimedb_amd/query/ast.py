@@ -120,6 +120,19 @@ class InsertValues:
 
 
 @dataclass
+class Delete:
+    table: str
+    where: object | None
+
+
+@dataclass
+class AlterTable:
+    table: str
+    action: str                 # "add_column"
+    column: tuple | None = None  # (name, type, opts)
+
+
+@dataclass
 class Copy:
     table: str
     path: str

@@ -189,7 +189,96 @@ class Executor:
             return self._exec_explain(stmt)
         if isinstance(stmt, ast.Copy):
             return self._exec_copy(stmt)
+        if isinstance(stmt, ast.Delete):
+            return self._exec_delete(stmt)
+        if isinstance(stmt, ast.AlterTable):
+            return self._exec_alter(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _exec_alter(self, a: ast.AlterTable) -> QueryResult:
+        """ALTER TABLE ADD COLUMN (reference: alter DDL procedure; tags are
+        immutable — only fields can be added)."""
+        st = self.engine.table(a.table)
+        name, typ, opts = a.column
+        tl = typ.lower()
+        if tl in ("string", "varchar", "text", "json"):
+            for r in st.regions:
+                r.ensure_str_fields([name], fulltext=opts.get("fulltext", True))
+        else:
+            for r in st.regions:
+                r.ensure_fields([name])
+        return QueryResult(["status"], [["ok"]])
+
+    def _exec_delete(self, d: ast.Delete) -> QueryResult:
+        """DELETE FROM t [WHERE ...]: rewrite affected region data.
+
+        The reference appends OpType::Delete tombstones resolved at
+        read/compaction time; with device-resident data we rewrite the
+        affected SSTs directly (flush first so the memtable is included) —
+        durable immediately, no tombstone debt on the scan path."""
+        sel = ast.Select([(ast.Star(), None)], d.table, d.where)
+        plan = self._plan_select(sel)
+        st = plan.table
+        device = self.engine.config.device
+        ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
+        ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
+        deleted = 0
+        from greptimedb_amd.engine import sst as sst_mod
+        import os as _os
+        for region in st.regions:
+            region.flush()
+            cand = self._candidate_codes(region, plan)
+            lut = None
+            if cand is not None:
+                lut = np.full(len(region.series), -1, dtype=np.int32)
+                lut[np.asarray(cand, dtype=np.int64)] = 1
+            lut_t = torch.as_tensor(lut, device=device) if lut is not None else None
+            for fid in list(region.sst_cache.keys()):
+                b = region.sst_cache[fid]
+                src = type("S", (), {})()
+                from greptimedb_amd.engine.region import ScanSource
+                src = ScanSource(b.ts, b.series, b.fields, b.n,
+                                 {fn: i for i, fn in enumerate(b.field_names)},
+                                 True, getattr(b, "str_cols", {}), None)
+                from greptimedb_amd.ops import filter_series_time
+                del_mask = filter_series_time(b.ts, b.series, lut_t, ts_lo, ts_hi)
+                if plan.residual is not None:
+                    del_mask &= self._eval_mask(plan.residual, src, region, device)
+                ndel = int(del_mask.sum())
+                if ndel == 0:
+                    continue
+                deleted += ndel
+                keep = (~del_mask).nonzero(as_tuple=True)[0]
+                ts_t = b.ts[keep].contiguous()
+                se_t = b.series[keep].contiguous()
+                f_t = b.fields[:, keep].contiguous()
+                keep_h = keep.cpu().numpy()
+                new_strs = {sn: np.asarray(col, dtype=object)[keep_h]
+                            for sn, col in getattr(b, "str_cols", {}).items()}
+                new_fid = sst_mod.new_file_id()
+                path = _os.path.join(region.dir, "sst", f"{new_fid}.parquet")
+                meta = sst_mod.write_sst(
+                    path, region.schema, region.series.pks,
+                    se_t.cpu().numpy(), ts_t.cpu().numpy(), f_t.cpu().numpy(),
+                    np.arange(ts_t.numel(), dtype=np.int64), b.field_names,
+                    str_cols=new_strs)
+                region.manifest.commit({
+                    "kind": "edit", "files_to_add": [meta.to_dict()],
+                    "files_to_remove": [fid]})
+                nb = sst_mod.SstBatch(ts_t, se_t, f_t, None, meta.min_ts,
+                                      meta.max_ts, b.field_names)
+                nb.str_cols = new_strs
+                for sn, arr in new_strs.items():
+                    ft = region.text_cols.get(sn)
+                    if ft is not None:
+                        nb.text_index[sn] = ft.build_segment(list(arr), device)
+                with region.lock:
+                    region.sst_cache.pop(fid, None)
+                    region.sst_cache[new_fid] = nb
+                old = _os.path.join(region.dir, "sst", f"{fid}.parquet")
+                if _os.path.exists(old):
+                    _os.unlink(old)
+        return QueryResult(["rows"], [[deleted]])
 
     def _exec_copy(self, c: ast.Copy) -> QueryResult:
         """COPY table TO/FROM file (reference: operator COPY via
@@ -722,31 +811,26 @@ class Executor:
         agg_fields = sorted({a.arg for a in plan.aggs if a.arg is not None})
         nf = len(agg_fields)
 
-        acc = None  # (sum, cnt, minv, maxv, rows) torch tensors
+        from greptimedb_amd.ops import ts_bucket_agg_acc, ts_bucket_agg_finish
+        acc = None  # opaque accumulator handle (GPU: shared atomic buffers)
         for region, lut in zip(st.regions, region_luts):
             lut_t = torch.as_tensor(lut, device=device)
             for ts_t, se_t, f_t, fidx_t in self._region_agg_inputs(
                     region, plan, device, agg_fields, ts_lo, ts_hi):
-                out = ts_bucket_agg(ts_t, se_t, f_t, fidx_t, lut_t,
-                                    ts_lo, ts_hi, origin, bucket_ms,
-                                    n_slots, n_buckets)
-                if acc is None:
-                    acc = list(out)
-                else:
-                    acc[0] += out[0]
-                    acc[1] += out[1]
-                    acc[2] = torch.fmin(acc[2], out[2])
-                    acc[3] = torch.fmax(acc[3], out[3])
-                    acc[4] += out[4]
+                acc = ts_bucket_agg_acc(ts_t, se_t, f_t, fidx_t, lut_t,
+                                        ts_lo, ts_hi, origin, bucket_ms,
+                                        n_slots, n_buckets, acc=acc)
         if acc is None:
             z = torch.zeros
-            acc = [z((nf, n_slots, n_buckets), dtype=torch.float64),
-                   z((nf, n_slots, n_buckets), dtype=torch.int64),
-                   torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64),
-                   torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64),
-                   z((n_slots, n_buckets), dtype=torch.int64)]
+            final = [z((nf, n_slots, n_buckets), dtype=torch.float64),
+                     z((nf, n_slots, n_buckets), dtype=torch.int64),
+                     torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64),
+                     torch.full((nf, n_slots, n_buckets), float("nan"), dtype=torch.float64),
+                     z((n_slots, n_buckets), dtype=torch.int64)]
+        else:
+            final = ts_bucket_agg_finish(acc)
 
-        sums, cnts, mins, maxs, rowcnt = [t.cpu().numpy() for t in acc]
+        sums, cnts, mins, maxs, rowcnt = [t.cpu().numpy() for t in final]
 
         if self.dist is not None:
             group_keys, (sums, cnts, mins, maxs, rowcnt) = self.dist.merge_groups(

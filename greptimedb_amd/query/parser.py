@@ -182,6 +182,28 @@ class Parser:
             self.next()
             analyze = self.eat_kw("analyze")
             stmt = ast.Explain(analyze, self.parse_statement_inner())
+        elif self.at_kw("delete"):
+            self.next()
+            self.expect_kw("from")
+            table = self.next().value
+            where = self.parse_expr() if self.eat_kw("where") else None
+            stmt = ast.Delete(table, where)
+        elif self.at_kw("alter"):
+            self.next()
+            self.expect_kw("table")
+            table = self.next().value
+            self.expect_kw("add")
+            self.eat_kw("column")
+            cname = self.next().value
+            ctype = self.next().value
+            if self.eat_op("("):
+                ctype += f"({self.next().value})"
+                self.expect_op(")")
+            opts = {}
+            if self.eat_kw("fulltext"):
+                self.eat_kw("index")
+                opts["fulltext"] = True
+            stmt = ast.AlterTable(table, "add_column", (cname, ctype, opts))
         elif self.at_kw("copy"):
             self.next()
             table = self.next().value

@@ -49,3 +49,36 @@ def test_copy_csv(ex, tmp_path):
     p = str(tmp_path / "out.csv")
     ex.execute(f"COPY t1 TO '{p}' WITH (format = 'csv')")
     assert open(p).read().count("\n") >= 2
+
+
+def test_delete(ex, tmp_path):
+    ex.execute("INSERT INTO t1 (h, ts, v) VALUES ('a', 3000, 9.0), ('c', 4000, 4.0)")
+    r = ex.execute("DELETE FROM t1 WHERE h = 'a'")
+    assert r.columns[0][0] == 2
+    rows = ex.execute("SELECT h, ts, v FROM t1 ORDER BY ts").rows()
+    assert [t[0] for t in rows] == ["b", "c"]
+    # field predicate delete
+    r = ex.execute("DELETE FROM t1 WHERE v > 3")
+    assert r.columns[0][0] == 1
+    assert len(ex.execute("SELECT h FROM t1")) == 1
+    # survives reopen
+    eng = ex.engine
+    d = eng.config.data_dir
+    eng.close()
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ex2 = Executor(eng2)
+    assert [t[0] for t in ex2.execute("SELECT h FROM t1").rows()] == ["b"]
+    eng2.close()
+
+
+def test_alter_add_column(ex):
+    ex.execute("ALTER TABLE t1 ADD COLUMN extra DOUBLE")
+    ex.execute("INSERT INTO t1 (h, ts, v, extra) VALUES ('z', 9000, 1.0, 7.5)")
+    r = ex.execute("SELECT extra FROM t1 WHERE h = 'z'")
+    assert list(r.columns[0]) == [7.5]
+    ex.execute("ALTER TABLE t1 ADD COLUMN note STRING")
+    ex2 = ex.execute("SELECT column_name FROM information_schema.columns "
+                     "WHERE table_name = 't1'")
+    assert "extra" in list(ex2.columns[0]) and "note" in list(ex2.columns[0])

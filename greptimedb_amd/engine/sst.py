@@ -87,7 +87,9 @@ def write_sst(path: str, schema: TableSchema, pks: list[bytes],
         cols.append(pa.array(fields[i], type=pa.float64()))
         names.append(fn)
     for fn, vals in (str_cols or {}).items():
-        cols.append(pa.array(list(vals), type=pa.string()))
+        vl = list(vals)
+        is_bin = any(isinstance(v, (bytes, bytearray)) for v in vl[:4])
+        cols.append(pa.array(vl, type=pa.binary() if is_bin else pa.string()))
         names.append(fn)
     cols.append(pa.array(ts_ms, type=pa.timestamp("ms")))
     names.append(schema.time_index.name)
@@ -138,7 +140,8 @@ def read_sst(path: str, schema: TableSchema, field_names: list[str]):
     for cn in t.column_names:
         if cn in internal or cn in field_names:
             continue
-        if pa.types.is_string(t.schema.field(cn).type) or \
-                pa.types.is_large_string(t.schema.field(cn).type):
+        ftype = t.schema.field(cn).type
+        if pa.types.is_string(ftype) or pa.types.is_large_string(ftype) or \
+                pa.types.is_binary(ftype) or pa.types.is_large_binary(ftype):
             str_cols[cn] = t.column(cn).to_numpy(zero_copy_only=False)
     return dict_values, indices, ts, fields, seq, str_cols

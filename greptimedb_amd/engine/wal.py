@@ -26,10 +26,13 @@ def encode_batch(series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
     """[u32 hdr_len][hdr json][series i32][ts i64][fields f64 nf*n]
     [per str col: lengths i32[n] (-1=None) + utf8 blob]"""
     str_cols = str_cols or {}
+    bin_cols = [name for name, vals in str_cols.items()
+                if any(isinstance(v, (bytes, bytearray)) for v in vals[:4])]
     hdr = json.dumps({
         "n": int(len(ts_ms)),
         "fields": field_names,
         "strs": list(str_cols),
+        "bins": bin_cols,
         "new_series": [[c, pk.hex()] for c, pk in new_series],
     }).encode()
     parts = [struct.pack("<I", len(hdr)), hdr,
@@ -57,14 +60,17 @@ def decode_batch(buf: bytes):
     fields = np.frombuffer(buf, dtype=np.float64, count=nf * n, offset=off).reshape(nf, n)
     off += 8 * nf * n
     str_cols = {}
+    bins = set(hdr.get("bins", []))
     for name in hdr.get("strs", []):
         lens = np.frombuffer(buf, dtype=np.int32, count=n, offset=off); off += 4 * n
         vals = []
+        raw = name in bins
         for ln in lens:
             if ln < 0:
                 vals.append(None)
             else:
-                vals.append(buf[off:off + ln].decode())
+                b = buf[off:off + ln]
+                vals.append(b if raw else b.decode())
                 off += ln
         str_cols[name] = vals
     new_series = [(c, bytes.fromhex(h)) for c, h in hdr["new_series"]]

@@ -44,6 +44,7 @@ class DataType(enum.Enum):
     FLOAT64 = "float64"
     STRING = "string"          # dictionary-encoded int32 codes in memory
     BINARY = "binary"
+    VECTOR = "vector"          # fixed-dim f32 embedding (bytes at rest)
     TIMESTAMP_MS = "timestamp_ms"    # int64 millis (greptime default)
     TIMESTAMP_NS = "timestamp_ns"    # int64 nanos
     JSON = "json"              # stored as string
@@ -64,6 +65,7 @@ class DataType(enum.Enum):
             DataType.FLOAT64: np.float64,
             DataType.STRING: np.int32,
             DataType.BINARY: np.int32,
+            DataType.VECTOR: np.int32,
             DataType.JSON: np.int32,
             DataType.TIMESTAMP_MS: np.int64,
             DataType.TIMESTAMP_NS: np.int64,
@@ -76,7 +78,8 @@ class DataType(enum.Enum):
 
     @property
     def is_string_like(self) -> bool:
-        return self in (DataType.STRING, DataType.BINARY, DataType.JSON)
+        return self in (DataType.STRING, DataType.BINARY, DataType.JSON,
+                        DataType.VECTOR)
 
     @property
     def is_float(self) -> bool:
@@ -91,6 +94,7 @@ class ColumnSchema:
     column_id: int
     nullable: bool = True
     fulltext: bool = False   # build a fulltext index (string fields only)
+    vector_dim: int = 0      # VECTOR(D) dimension
 
     def __post_init__(self):
         if self.semantic == SemanticType.TIMESTAMP and not self.dtype.is_timestamp:
@@ -157,6 +161,7 @@ class TableSchema:
                     "column_id": c.column_id,
                     "nullable": c.nullable,
                     "fulltext": c.fulltext,
+                    "vector_dim": c.vector_dim,
                 }
                 for c in self.columns
             ],
@@ -172,6 +177,7 @@ class TableSchema:
                 column_id=c["column_id"],
                 nullable=c.get("nullable", True),
                 fulltext=c.get("fulltext", False),
+                vector_dim=c.get("vector_dim", 0),
             )
             for c in d["columns"]
         ]

@@ -165,6 +165,12 @@ class LineParser {
   size_t num_series() const { return tagset_ids_.size(); }
   std::vector<std::string> field_names() const { return field_names_; }
 
+  py::bytes tagset_str(int32_t sid) const {
+    if (sid < 0 || (size_t)sid >= tagset_strs_.size())
+      throw std::out_of_range("unknown series ref");
+    return py::bytes(tagset_strs_[sid]);
+  }
+
   // Restore dictionary state (e.g. after WAL replay / reopen).
   void register_tagset(const std::string& s, int32_t id) {
     tagset_ids_.emplace(fnv1a(s.data(), s.size()), id);
@@ -996,6 +1002,7 @@ PYBIND11_MODULE(_native, m) {
       .def(py::init<>())
       .def("parse", &LineParser::parse)
       .def("num_series", &LineParser::num_series)
+      .def("tagset_str", &LineParser::tagset_str)
       .def("field_names", &LineParser::field_names)
       .def("register_tagset", &LineParser::register_tagset)
       .def("register_field", &LineParser::register_field);

@@ -136,8 +136,20 @@ class Ingestor:
         n = len(series)
         if n == 0:
             return 0
+        # repartition invalidates routing: drop LUTs, re-resolve lazily
+        epoch = getattr(self.engine, "routing_epoch", 0)
+        if epoch != getattr(self, "_epoch", 0):
+            self._epoch = epoch
+            self.sid_region.fill(-1)
+            self.sid_local.fill(-1)
+            self.flat_regions.clear()
+            self._region_key.clear()
+            self._table_field_map.clear()
         for sid, key in new_tagsets:
             self._register_tagset(sid, key)
+        unknown = np.unique(series[self.sid_region[series] < 0])
+        for sid in unknown:
+            self._register_tagset(int(sid), self.parser.tagset_str(int(sid)))
         parser_fields = self.parser.field_names()
         fields_mat = np.stack([fields[fn] for fn in parser_fields]) if parser_fields \
             else np.zeros((0, n))

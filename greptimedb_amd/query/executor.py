@@ -138,9 +138,22 @@ class Executor:
 
     # ---------------------------------------------------------- entrypoints
 
+    SLOW_QUERY_MS = 1000.0
+
     def execute(self, sql: str) -> QueryResult:
+        import time as _time
         stmt = parse_sql(sql)
-        return self.execute_stmt(stmt)
+        t0 = _time.perf_counter()
+        r = self.execute_stmt(stmt)
+        dt = (_time.perf_counter() - t0) * 1000
+        if dt >= self.SLOW_QUERY_MS:
+            # slow-query log (reference: common/frontend slow query events)
+            log = getattr(self.engine, "slow_queries", None)
+            if log is None:
+                from collections import deque
+                log = self.engine.slow_queries = deque(maxlen=128)
+            log.append({"sql": sql[:500], "ms": round(dt, 1)})
+        return r
 
     def execute_stmt(self, stmt) -> QueryResult:
         if isinstance(stmt, ast.Select):
@@ -397,6 +410,15 @@ class Executor:
             return QueryResult(["result"], [[n]])
         if f == "flush_all":
             self.engine.flush_all()
+            return QueryResult(["result"], [[1]])
+        if f == "repartition_table":
+            from greptimedb_amd.engine.repartition import repartition_table
+            moved = repartition_table(self.engine, str(a.args[0]), int(a.args[1]))
+            return QueryResult(["result"], [[moved]])
+        if f == "migrate_region":
+            from greptimedb_amd.meta.migration import migrate_region
+            migrate_region(self.engine, str(a.args[0]), int(a.args[1]),
+                           str(a.args[2]))
             return QueryResult(["result"], [[1]])
         raise PlanQuery(f"unknown admin function {a.func}")
 

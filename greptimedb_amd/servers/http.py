@@ -117,7 +117,8 @@ def build_app(ctx: ServerContext) -> FastAPI:
             for name, st in ctx.engine.tables.items()
         }
         return {"uptime_s": time.time() - ctx.started, "tables": tables,
-                "device": ctx.engine.config.device}
+                "device": ctx.engine.config.device,
+                "slow_queries": list(getattr(ctx.engine, "slow_queries", []))}
 
     # ---------------- SQL ----------------
 

@@ -42,12 +42,20 @@ r1 = ex.execute("SELECT count(*) FROM cpu")
 r2 = ex.execute("SELECT hostname, avg(usage_user) FROM cpu GROUP BY hostname ORDER BY hostname")
 r3 = ex.execute("SELECT date_trunc('minute', ts) m, max(usage_user) FROM cpu GROUP BY m ORDER BY m")
 r4 = ex.execute("SELECT ts, hostname, usage_user FROM cpu WHERE usage_user > 99")
+r5 = ex.execute("SELECT hostname, last_value(usage_user) FROM cpu GROUP BY hostname ORDER BY hostname")
+# PromQL distributed aggregation (merge_prom_planes over gloo)
+from greptimedb_amd.query.promql.eval import PromEvaluator
+ev = PromEvaluator(eng, dist=DistContext(device="cpu"))
+m = ev.query_range('sum({__field__="usage_user", __name__="cpu"})',
+                   1451606450, 1451606450, 1)
 out = {
     "count": int(r1.columns[0][0]),
     "hosts": list(r2.columns[0]),
     "avgs": [float(x) for x in r2.columns[1]],
     "minutes": len(r3),
     "raw": len(r4),
+    "lastpoint_hosts": len(r5),
+    "prom_sum": round(float(m.values[0][-1]), 6),
 }
 print("RESULT" + str(rank) + json.dumps(out))
 dist.destroy_process_group()
@@ -85,3 +93,5 @@ def test_two_rank_query_combine(tmp_path):
     assert results[0]["count"] == 10000      # 5000 per rank
     assert len(results[0]["hosts"]) == 20    # 10 hosts per rank, disjoint
     assert results[0]["raw"] >= 0
+    assert results[0]["lastpoint_hosts"] == 20
+    assert results[0]["prom_sum"] != 0.0

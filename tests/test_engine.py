@@ -188,3 +188,32 @@ def test_explain(tmp_engine):
     r = ex.execute("EXPLAIN ANALYZE SELECT h, max(v) FROM t GROUP BY h")
     text = "\n".join(r.columns[0])
     assert "fused-ts-bucket-agg" in text and "Execution" in text
+
+
+def test_repartition_table(tmp_path):
+    from greptimedb_amd.query.executor import Executor
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ing = Ingestor(eng, default_regions=2)
+    w = CpuWorkload(scale=30)
+    for _ in range(4):
+        ing.ingest_lines(w.next_batch(1000))
+    ex = Executor(eng)
+    before = ex.execute("SELECT hostname, count(*) FROM cpu GROUP BY hostname "
+                        "ORDER BY hostname").rows()
+    r = ex.execute("ADMIN repartition_table('cpu', 6)")
+    assert r.columns[0][0] == 4000
+    assert len(eng.table("cpu").regions) == 6
+    after = ex.execute("SELECT hostname, count(*) FROM cpu GROUP BY hostname "
+                       "ORDER BY hostname").rows()
+    assert before == after
+    # ingest continues through the same router after the epoch bump
+    ing.ingest_lines(w.next_batch(500))
+    assert ex.execute("SELECT count(*) FROM cpu").columns[0][0] == 4500
+    # survives reopen
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ex2 = Executor(eng2)
+    assert len(eng2.table("cpu").regions) == 6
+    assert ex2.execute("SELECT count(*) FROM cpu").columns[0][0] == 4500
+    eng2.close()

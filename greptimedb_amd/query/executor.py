@@ -922,6 +922,49 @@ class Executor:
             return None
         return sorted(codes)
 
+    def _build_group_luts(self, st, plan, gt):
+        """Per-region code→group-slot LUTs + global group-key dict.
+
+        Vectorized over the tag-value dictionary (factor codes), not per
+        series — high-cardinality GROUP BY tags stay off the Python path."""
+        group_keys: dict[tuple, int] = {}
+        region_luts = []
+        for region in st.regions:
+            cand = self._candidate_codes(region, plan)
+            nser = len(region.series)
+            lut = np.full(nser, -1, dtype=np.int32)
+            sel_idx = np.arange(nser) if cand is None else \
+                np.asarray(cand, dtype=np.int64)
+            if gt:
+                if len(sel_idx):
+                    key = np.zeros(len(sel_idx), dtype=np.int64)
+                    mult = 1
+                    factors = []
+                    for t in gt:
+                        codes_arr, values = region.series.tag_codes(t)
+                        factors.append((codes_arr, values))
+                        card = len(values) + 1
+                        if mult > (1 << 62) // max(card, 1):
+                            _, key = np.unique(key, return_inverse=True)
+                            mult = int(key.max()) + 1 if len(key) else 1
+                        key = key + (codes_arr[sel_idx].astype(np.int64) + 1) * mult
+                        mult *= card
+                    uniq, first, inv = np.unique(key, return_index=True,
+                                                 return_inverse=True)
+                    slots_of_uniq = np.empty(len(uniq), dtype=np.int32)
+                    for u in range(len(uniq)):
+                        code = int(sel_idx[first[u]])
+                        kt = tuple(values[codes_arr[code]] if codes_arr[code] >= 0
+                                   else None
+                                   for codes_arr, values in factors)
+                        slots_of_uniq[u] = group_keys.setdefault(kt, len(group_keys))
+                    lut[sel_idx] = slots_of_uniq[inv]
+            else:
+                lut[sel_idx] = 0
+                group_keys.setdefault((), 0)
+            region_luts.append(lut)
+        return group_keys, region_luts
+
     def _exec_aggregate(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
         if any(a.func == "last_value" for a in plan.aggs):
             return self._exec_lastpoint(sel, plan)
@@ -960,27 +1003,7 @@ class Executor:
             origin = ts_lo
             n_buckets = 1
 
-        # group slots: key = tuple of group-tag values
-        group_keys: dict[tuple, int] = {}
-        region_luts = []
-        gt = plan.group_tags
-        for region in st.regions:
-            cand = self._candidate_codes(region, plan)
-            nser = len(region.series)
-            lut = np.full(nser, -1, dtype=np.int32)
-            it = range(nser) if cand is None else cand
-            if gt:
-                tag_idx = [region.series.tag_names.index(t) for t in gt]
-                for code in it:
-                    tv = region.series.tag_values[code]
-                    key = tuple(tv[i] for i in tag_idx)
-                    slot = group_keys.setdefault(key, len(group_keys))
-                    lut[code] = slot
-            else:
-                for code in it:
-                    lut[code] = 0
-                group_keys.setdefault((), 0)
-            region_luts.append(lut)
+        group_keys, region_luts = self._build_group_luts(st, plan, plan.group_tags)
         n_slots = max(len(group_keys), 1)
 
         # fields needed
@@ -1034,26 +1057,12 @@ class Executor:
         ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
         ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
 
-        group_keys: dict[tuple, int] = {}
         gt = plan.group_tags
+        group_keys, region_luts = self._build_group_luts(st, plan, gt)
         best_ts: np.ndarray | None = None
         best_val: np.ndarray | None = None
 
-        for region in st.regions:
-            cand = self._candidate_codes(region, plan)
-            nser = len(region.series)
-            lut = np.full(nser, -1, dtype=np.int32)
-            it = range(nser) if cand is None else cand
-            if gt:
-                tag_idx = [region.series.tag_names.index(t) for t in gt]
-                for code in it:
-                    tv = region.series.tag_values[code]
-                    key = tuple(tv[i] for i in tag_idx)
-                    lut[code] = group_keys.setdefault(key, len(group_keys))
-            else:
-                group_keys.setdefault((), 0)
-                for code in it:
-                    lut[code] = 0
+        for region, lut in zip(st.regions, region_luts):
             lut_t = torch.as_tensor(lut, device=device)
             sources = region.scan_sources(ts_lo, ts_hi)
             if plan.residual is not None:

@@ -347,6 +347,18 @@ def build_app(ctx: ServerContext) -> FastAPI:
         metrics_mod.counter("remote_write_samples").inc(n)
         return Response(status_code=204)
 
+    @app.post("/v1/prometheus/read")
+    async def prom_remote_read(request: Request):
+        from greptimedb_amd import _native
+        from greptimedb_amd.servers.prom_read import execute_read, parse_read_request
+        body = await request.body()
+        if request.headers.get("content-encoding", "snappy") != "identity":
+            body = _native.snappy_uncompress(body)
+        queries = parse_read_request(body)
+        resp = execute_read(ctx.prom, queries)
+        return Response(resp, media_type="application/x-protobuf",
+                        headers={"Content-Encoding": "snappy"})
+
     # ---------------- prometheus query API ----------------
 
     async def _param(request: Request, name: str):

@@ -122,8 +122,11 @@ class Ingestor:
                    if not np.isnan(fields_mat[i][rows]).all()]
         new = [fn for fn in present if fn not in region0.field_names]
         if new:
-            for r in st.regions:
-                r.ensure_fields(new)
+            # serialize schema growth so field ORDER matches across regions
+            with self.engine._ddl_lock:
+                new = [fn for fn in new if fn not in region0.field_names]
+                for r in st.regions:
+                    r.ensure_fields(new)
         pmap = {fn: i for i, fn in enumerate(parser_fields)}
         m = np.array([pmap.get(fn, -1) for fn in region0.field_names], dtype=np.int64)
         self._table_field_map[name] = (m, len(parser_fields))

@@ -94,13 +94,13 @@ class LogStore:
             codes[i] = st.regions[ridx].register_series(tags)
             region_rows.setdefault(ridx, []).append(i)
 
-        new_num = [k for k in num_vals if k not in st.regions[0].field_names]
-        if new_num:
-            for r in st.regions:
+        with self.engine._ddl_lock:
+            new_num = [k for k in num_vals if k not in st.regions[0].field_names]
+            for r in (st.regions if new_num else []):
                 r.ensure_fields(new_num)
-        new_str = [k for k in str_vals if k not in st.regions[0].str_field_names]
-        if new_str:
-            for r in st.regions:
+            new_str = [k for k in str_vals
+                       if k not in st.regions[0].str_field_names]
+            for r in (st.regions if new_str else []):
                 r.ensure_str_fields(new_str)
         fnames = st.regions[0].field_names
         for ridx, rows in region_rows.items():

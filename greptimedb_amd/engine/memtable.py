@@ -69,10 +69,18 @@ class Memtable:
         s = torch.as_tensor(series)
         t = torch.as_tensor(ts_ms)
         f = torch.as_tensor(fields)
+        # concurrent auto-ALTER: a writer may carry more (or fewer) field
+        # columns than this memtable has seen — reconcile under the region lock
+        if f.shape[0] > self.nf:
+            self.add_fields(f.shape[0] - self.nf)
         lo, hi = self.len, self.len + n
         self.ts[lo:hi].copy_(t, non_blocking=True)
         self.series[lo:hi].copy_(s, non_blocking=True)
-        self.fields[:, lo:hi].copy_(f, non_blocking=True)
+        if f.shape[0] < self.nf:
+            self.fields[: f.shape[0], lo:hi].copy_(f, non_blocking=True)
+            self.fields[f.shape[0]:, lo:hi] = float("nan")
+        else:
+            self.fields[:, lo:hi].copy_(f, non_blocking=True)
         mn, mx = int(t.min()), int(t.max())
         self.min_ts = mn if self.min_ts is None else min(self.min_ts, mn)
         self.max_ts = mx if self.max_ts is None else max(self.max_ts, mx)

@@ -217,3 +217,52 @@ def test_repartition_table(tmp_path):
     assert len(eng2.table("cpu").regions) == 6
     assert ex2.execute("SELECT count(*) FROM cpu").columns[0][0] == 4500
     eng2.close()
+
+
+def test_promstore_restart_persistence(tmp_path):
+    from greptimedb_amd.engine.promstore import PromStore
+    from greptimedb_amd.query.promql.eval import PromEvaluator
+    from tests.test_http import _ts_msg
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    store = PromStore(eng)
+    req = _ts_msg([("__name__", "m_up"), ("job", "a")], [(1.0, 1000), (2.0, 2000)])
+    assert store.write(req, snappy=False) == 2
+    eng.flush_all()
+    store.write(_ts_msg([("__name__", "m_up"), ("job", "a")], [(3.0, 3000)]),
+                snappy=False)  # WAL-only
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ev = PromEvaluator(eng2)
+    m = ev.query_range('m_up{job="a"}', 3, 3, 1)
+    assert m.S == 1 and float(m.values[0][0]) == 3.0
+    # same store continues appending to the same series after restart
+    store2 = PromStore(eng2)
+    store2.write(_ts_msg([("__name__", "m_up"), ("job", "a")], [(4.0, 4000)]),
+                 snappy=False)
+    m = ev.query_range('m_up', 4, 4, 1)
+    assert m.S == 1 and float(m.values[0][0]) == 4.0
+    eng2.close()
+
+
+def test_concurrent_ingest_threads(tmp_path):
+    import threading
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    def worker(wi):
+        ing = Ingestor(eng)
+        w = CpuWorkload(scale=10, seed=wi)
+        w.tagsets = [t.replace(b"host_", b"w%d_host_" % wi) for t in w.tagsets]
+        for _ in range(5):
+            ing.ingest_lines(w.next_batch(500))
+    threads = [threading.Thread(target=worker, args=(i,)) for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert _total(eng) == 4 * 5 * 500
+    # WAL replay after concurrent writes is consistent
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    assert _total(eng2) == 10000
+    eng2.close()

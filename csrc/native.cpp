@@ -885,6 +885,268 @@ class OtlpTraceParser {
   std::vector<int32_t> out_status_;
 };
 
+// ------------------------------------------------- OTLP metrics + logs
+// ExportMetricsServiceRequest: resource_metrics → scope_metrics → Metric
+// { name=1; gauge=5 / sum=7 { data_points: NumberDataPoint } }.
+// NumberDataPoint{ attributes=7; time_unix_nano=3 (fixed64);
+//                  as_double=4 (fixed64) / as_int=6 (sfixed64) }.
+// Output: flat [(metric, {attrs}, ts_ms, value)] ready for
+// PromStore.write_points.
+
+class OtlpMetricsParser {
+ public:
+  py::list parse(py::bytes data) {
+    char* buf; Py_ssize_t len;
+    if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+      throw std::runtime_error("expected bytes");
+    py::list out;
+    const uint8_t* p = reinterpret_cast<const uint8_t*>(buf);
+    const uint8_t* end = p + len;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      if ((k >> 3) == 1 && (k & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        parse_resource_metrics(p, p + l, out);
+        p += l;
+      } else skip(k & 7, p, end);
+    }
+    return out;
+  }
+
+ private:
+  static void skip(int wt, const uint8_t*& p, const uint8_t* end) {
+    if (wt == 0) pw_varint(p, end);
+    else if (wt == 1) p += 8;
+    else if (wt == 2) { uint64_t l = pw_varint(p, end); p += l; }
+    else if (wt == 5) p += 4;
+    else p = end;
+  }
+
+  static void parse_kv(const uint8_t* p, const uint8_t* end, py::dict& attrs) {
+    std::string kname, vstr;
+    double vnum = 0; bool has_num = false;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      if ((k >> 3) == 1 && (k & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        kname.assign(reinterpret_cast<const char*>(p), l); p += l;
+      } else if ((k >> 3) == 2 && (k & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* vp = p; const uint8_t* vend = p + l;
+        while (vp < vend) {
+          uint64_t vk = pw_varint(vp, vend);
+          int vf = (int)(vk >> 3), vwt = (int)(vk & 7);
+          if (vf == 1 && vwt == 2) {
+            uint64_t sl = pw_varint(vp, vend);
+            vstr.assign(reinterpret_cast<const char*>(vp), sl); vp += sl;
+          } else if (vf == 2 && vwt == 0) {
+            vstr = pw_varint(vp, vend) ? "true" : "false";
+          } else if (vf == 3 && vwt == 0) {
+            vstr = std::to_string((int64_t)pw_varint(vp, vend));
+          } else if (vf == 4 && vwt == 1) {
+            double d; std::memcpy(&d, vp, 8); vp += 8;
+            vstr = std::to_string(d);
+          } else skip(vwt, vp, vend);
+        }
+        p += l;
+      } else skip(k & 7, p, end);
+    }
+    if (!kname.empty()) attrs[py::str(kname)] = py::str(vstr);
+  }
+
+  void parse_resource_metrics(const uint8_t* p, const uint8_t* end, py::list& out) {
+    py::dict res_attrs;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 1 && wt == 2) {  // Resource{attributes=1}
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* rp = p; const uint8_t* rend = p + l;
+        while (rp < rend) {
+          uint64_t rk = pw_varint(rp, rend);
+          if ((rk >> 3) == 1 && (rk & 7) == 2) {
+            uint64_t al = pw_varint(rp, rend);
+            parse_kv(rp, rp + al, res_attrs);
+            rp += al;
+          } else skip(rk & 7, rp, rend);
+        }
+        p += l;
+      } else if (f == 2 && wt == 2) {  // ScopeMetrics
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* sp = p; const uint8_t* send = p + l;
+        while (sp < send) {
+          uint64_t sk = pw_varint(sp, send);
+          if ((sk >> 3) == 2 && (sk & 7) == 2) {  // Metric
+            uint64_t ml = pw_varint(sp, send);
+            parse_metric(sp, sp + ml, res_attrs, out);
+            sp += ml;
+          } else skip(sk & 7, sp, send);
+        }
+        p += l;
+      } else skip(wt, p, end);
+    }
+  }
+
+  void parse_metric(const uint8_t* p, const uint8_t* end,
+                    const py::dict& res_attrs, py::list& out) {
+    std::string name;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 1 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        name.assign(reinterpret_cast<const char*>(p), l); p += l;
+      } else if ((f == 5 || f == 7) && wt == 2) {  // Gauge / Sum
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* gp = p; const uint8_t* gend = p + l;
+        while (gp < gend) {
+          uint64_t gk = pw_varint(gp, gend);
+          if ((gk >> 3) == 1 && (gk & 7) == 2) {   // data_points
+            uint64_t dl = pw_varint(gp, gend);
+            parse_point(gp, gp + dl, name, res_attrs, out);
+            gp += dl;
+          } else skip(gk & 7, gp, gend);
+        }
+        p += l;
+      } else skip(wt, p, end);
+    }
+  }
+
+  void parse_point(const uint8_t* p, const uint8_t* end, const std::string& name,
+                   const py::dict& res_attrs, py::list& out) {
+    py::dict attrs;
+    for (auto item : res_attrs) attrs[item.first] = item.second;
+    uint64_t ts = 0;
+    double val = 0;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 7 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        parse_kv(p, p + l, attrs);
+        p += l;
+      } else if (f == 3 && wt == 1) {
+        std::memcpy(&ts, p, 8); p += 8;
+      } else if (f == 4 && wt == 1) {
+        std::memcpy(&val, p, 8); p += 8;
+      } else if (f == 6 && wt == 1) {
+        int64_t iv; std::memcpy(&iv, p, 8); p += 8;
+        val = (double)iv;
+      } else skip(wt, p, end);
+    }
+    out.append(py::make_tuple(py::str(name), std::move(attrs),
+                              (int64_t)(ts / 1000000ULL), val));
+  }
+};
+
+// ExportLogsServiceRequest: resource_logs → scope_logs → LogRecord
+// { time_unix_nano=1 fixed64; severity_text=3; body=5 AnyValue;
+//   attributes=6 }. Output: [(ts_ms, severity, body, {attrs})].
+class OtlpLogsParser {
+ public:
+  py::list parse(py::bytes data) {
+    char* buf; Py_ssize_t len;
+    if (PyBytes_AsStringAndSize(data.ptr(), &buf, &len) != 0)
+      throw std::runtime_error("expected bytes");
+    py::list out;
+    const uint8_t* p = reinterpret_cast<const uint8_t*>(buf);
+    const uint8_t* end = p + len;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      if ((k >> 3) == 1 && (k & 7) == 2) {
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* rp = p; const uint8_t* rend = p + l;
+        std::string service = "unknown";
+        while (rp < rend) {
+          uint64_t rk = pw_varint(rp, rend);
+          int f = (int)(rk >> 3), wt = (int)(rk & 7);
+          if (f == 1 && wt == 2) {  // Resource (attrs ignored; senders put
+            uint64_t rl = pw_varint(rp, rend);   // routing labels on records)
+            rp += rl;
+          } else if (f == 2 && wt == 2) {  // ScopeLogs
+            uint64_t sl = pw_varint(rp, rend);
+            const uint8_t* lp = rp; const uint8_t* lend = rp + sl;
+            while (lp < lend) {
+              uint64_t lk = pw_varint(lp, lend);
+              if ((lk >> 3) == 2 && (lk & 7) == 2) {  // LogRecord
+                uint64_t ll = pw_varint(lp, lend);
+                parse_record(lp, lp + ll, out);
+                lp += ll;
+              } else OtlpMetricsSkip(lk & 7, lp, lend);
+            }
+            rp += sl;
+          } else OtlpMetricsSkip(wt, rp, rend);
+        }
+        p += l;
+      } else OtlpMetricsSkip(k & 7, p, end);
+    }
+    return out;
+  }
+
+ private:
+  static void OtlpMetricsSkip(int wt, const uint8_t*& p, const uint8_t* end) {
+    if (wt == 0) pw_varint(p, end);
+    else if (wt == 1) p += 8;
+    else if (wt == 2) { uint64_t l = pw_varint(p, end); p += l; }
+    else if (wt == 5) p += 4;
+    else p = end;
+  }
+
+  void parse_record(const uint8_t* p, const uint8_t* end, py::list& out) {
+    uint64_t ts = 0;
+    std::string severity, body;
+    py::dict attrs;
+    while (p < end) {
+      uint64_t k = pw_varint(p, end);
+      int f = (int)(k >> 3), wt = (int)(k & 7);
+      if (f == 1 && wt == 1) {
+        std::memcpy(&ts, p, 8); p += 8;
+      } else if (f == 3 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        severity.assign(reinterpret_cast<const char*>(p), l); p += l;
+      } else if (f == 5 && wt == 2) {  // body AnyValue{string_value=1}
+        uint64_t l = pw_varint(p, end);
+        const uint8_t* bp = p; const uint8_t* bend = p + l;
+        while (bp < bend) {
+          uint64_t bk = pw_varint(bp, bend);
+          if ((bk >> 3) == 1 && (bk & 7) == 2) {
+            uint64_t sl = pw_varint(bp, bend);
+            body.assign(reinterpret_cast<const char*>(bp), sl); bp += sl;
+          } else OtlpMetricsSkip(bk & 7, bp, bend);
+        }
+        p += l;
+      } else if (f == 6 && wt == 2) {
+        uint64_t l = pw_varint(p, end);
+        // KeyValue (string values only, like metrics attrs)
+        const uint8_t* kp = p; const uint8_t* kend = p + l;
+        std::string kname, vstr;
+        while (kp < kend) {
+          uint64_t kk = pw_varint(kp, kend);
+          if ((kk >> 3) == 1 && (kk & 7) == 2) {
+            uint64_t s = pw_varint(kp, kend);
+            kname.assign(reinterpret_cast<const char*>(kp), s); kp += s;
+          } else if ((kk >> 3) == 2 && (kk & 7) == 2) {
+            uint64_t s = pw_varint(kp, kend);
+            const uint8_t* vp = kp; const uint8_t* vend = kp + s;
+            while (vp < vend) {
+              uint64_t vk = pw_varint(vp, vend);
+              if ((vk >> 3) == 1 && (vk & 7) == 2) {
+                uint64_t sl = pw_varint(vp, vend);
+                vstr.assign(reinterpret_cast<const char*>(vp), sl); vp += sl;
+              } else OtlpMetricsSkip(vk & 7, vp, vend);
+            }
+            kp += s;
+          } else OtlpMetricsSkip(kk & 7, kp, kend);
+        }
+        if (!kname.empty()) attrs[py::str(kname)] = py::str(vstr);
+        p += l;
+      } else OtlpMetricsSkip(wt, p, end);
+    }
+    out.append(py::make_tuple((int64_t)(ts / 1000000ULL), py::str(severity),
+                              py::str(body), std::move(attrs)));
+  }
+};
+
 // ---------------------------------------------------------------- tokenizer
 // Fulltext tokenizer for log columns (reference: src/index fulltext_index —
 // tantivy's default tokenizer ≈ lowercase alphanumeric runs). Terms are
@@ -1020,6 +1282,12 @@ PYBIND11_MODULE(_native, m) {
       .def(py::init<>())
       .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)
       .def("num_series", &PromWriteParser::num_series);
+  py::class_<OtlpMetricsParser>(m, "OtlpMetricsParser")
+      .def(py::init<>())
+      .def("parse", &OtlpMetricsParser::parse);
+  py::class_<OtlpLogsParser>(m, "OtlpLogsParser")
+      .def(py::init<>())
+      .def("parse", &OtlpLogsParser::parse);
   py::class_<OtlpTraceParser>(m, "OtlpTraceParser")
       .def(py::init<>())
       .def("parse", &OtlpTraceParser::parse)

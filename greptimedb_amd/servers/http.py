@@ -337,6 +337,33 @@ def build_app(ctx: ServerContext) -> FastAPI:
         metrics_mod.counter("otlp_spans").inc(n)
         return {"partialSuccess": {}}
 
+    @app.post("/v1/otlp/v1/metrics")
+    async def otlp_metrics(request: Request):
+        from greptimedb_amd import _native
+        body = await request.body()
+        points = _native.OtlpMetricsParser().parse(body)
+        n = ctx.promstore.write_points(
+            [(m, dict(a), t, v) for m, a, t, v in points])
+        metrics_mod.counter("otlp_metric_points").inc(n)
+        return {"partialSuccess": {}}
+
+    @app.post("/v1/otlp/v1/logs")
+    async def otlp_logs(request: Request):
+        from greptimedb_amd import _native
+        body = await request.body()
+        recs = _native.OtlpLogsParser().parse(body)
+        entries = []
+        for ts, severity, text, attrs in recs:
+            e = {"timestamp": int(ts), "message": text}
+            if severity:
+                e["severity"] = severity
+            e.update({k: str(v) for k, v in dict(attrs).items()})
+            entries.append(e)
+        n = ctx.logstore.ingest("opentelemetry_logs", entries,
+                                tag_keys=["severity"], ts_key="timestamp")
+        metrics_mod.counter("otlp_log_records").inc(n)
+        return {"partialSuccess": {}}
+
     # ---------------- prometheus remote write ----------------
 
     @app.post("/v1/prometheus/write")

@@ -119,3 +119,48 @@ def test_otlp_http_endpoint(tmp_engine):
     r = client.get("/v1/sql", params={"sql":
         "SELECT count(*) FROM opentelemetry_traces"})
     assert r.json()["output"][0]["records"]["rows"][0][0] == 1
+
+
+def _anyvalue_double(d):
+    return _v((4 << 3) | 1) + struct.pack("<d", d)
+
+
+def test_otlp_metrics_endpoint(tmp_engine):
+    pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from greptimedb_amd.servers.http import ServerContext, build_app
+    client = TestClient(build_app(ServerContext(tmp_engine)))
+    # Metric{name="req_total", sum{data_points=[{attrs{job=a}, t, as_double}]}}
+    dp = (_ld(7, _s(1, "job") + _ld(2, _anyvalue_str("a"))) +
+          _v((3 << 3) | 1) + struct.pack("<Q", 2_000_000_000) +
+          _v((4 << 3) | 1) + struct.pack("<d", 42.0))
+    metric = _s(1, "req_total") + _ld(7, _ld(1, dp))
+    scope_metrics = _ld(2, _ld(2, metric))
+    resource = _ld(1, _kv("service.name", _anyvalue_str("svcM")))
+    req = _ld(1, resource + scope_metrics)
+    r = client.post("/v1/otlp/v1/metrics", content=req)
+    assert r.status_code == 200
+    r = client.get("/v1/prometheus/api/v1/query",
+                   params={"query": 'req_total{job="a"}', "time": "3"})
+    res = r.json()["data"]["result"]
+    assert len(res) == 1 and res[0]["value"][1] == "42.0"
+    assert res[0]["metric"]["service.name"] == "svcM"
+
+
+def test_otlp_logs_endpoint(tmp_engine):
+    pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from greptimedb_amd.servers.http import ServerContext, build_app
+    client = TestClient(build_app(ServerContext(tmp_engine)))
+    # LogRecord{time, severity_text="ERROR", body="db conn failed", attrs{k8s.pod=p1}}
+    rec = (_v((1 << 3) | 1) + struct.pack("<Q", 5_000_000_000) +
+           _s(3, "ERROR") + _ld(5, _anyvalue_str("db conn failed")) +
+           _ld(6, _s(1, "k8s.pod") + _ld(2, _anyvalue_str("p1"))))
+    # ResourceLogs{scope_logs=2 → ScopeLogs{log_records=2}}
+    req = _ld(1, _ld(2, _ld(2, rec)))
+    r = client.post("/v1/otlp/v1/logs", content=req)
+    assert r.status_code == 200
+    r = client.get("/v1/sql", params={"sql":
+        "SELECT severity, message FROM opentelemetry_logs WHERE matches(message, 'failed')"})
+    rows = r.json()["output"][0]["records"]["rows"]
+    assert rows == [["ERROR", "db conn failed"]]

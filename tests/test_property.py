@@ -1,0 +1,115 @@
+"""Property-based tests (reference parity: tests-fuzz): random workloads
+against a pure-Python oracle."""
+
+import math
+
+import numpy as np
+import pytest
+
+hyp = pytest.importorskip("hypothesis")
+from hypothesis import HealthCheck, given, settings  # noqa: E402
+from hypothesis import strategies as st_  # noqa: E402
+
+from greptimedb_amd.engine.engine import EngineConfig, MitoEngine  # noqa: E402
+from greptimedb_amd.query.executor import Executor  # noqa: E402
+
+ROWS = st_.lists(
+    st_.tuples(
+        st_.sampled_from(["a", "b", "c", "d"]),            # tag
+        st_.integers(min_value=0, max_value=100_000),      # ts ms
+        st_.floats(min_value=-1e6, max_value=1e6,
+                   allow_nan=False, allow_infinity=False),  # value
+    ),
+    min_size=1, max_size=60,
+)
+
+
+def _mk(tmp_path_factory, rows, append):
+    eng = MitoEngine(EngineConfig(
+        data_dir=str(tmp_path_factory.mktemp("prop")), device="cpu",
+        background_flush=False, default_regions=3))
+    ex = Executor(eng)
+    mode = "true" if append else "false"
+    ex.execute("CREATE TABLE p (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               f"PRIMARY KEY (h)) WITH ('append_mode'='{mode}')")
+    vals = ", ".join(f"('{h}', {t}, {v!r})" for h, t, v in rows)
+    ex.execute(f"INSERT INTO p (h, ts, v) VALUES {vals}")
+    return eng, ex
+
+
+def _oracle(rows, append):
+    if append:
+        return list(rows)
+    last = {}
+    for h, t, v in rows:
+        last[(h, t)] = v
+    return [(h, t, v) for (h, t), v in last.items()]
+
+
+@settings(max_examples=40, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(rows=ROWS, append=st_.booleans(), flush=st_.booleans(),
+       lo=st_.integers(min_value=0, max_value=100_000),
+       span=st_.integers(min_value=1, max_value=100_000))
+def test_agg_matches_oracle(tmp_path_factory, rows, append, flush, lo, span):
+    eng, ex = _mk(tmp_path_factory, rows, append)
+    try:
+        if flush:
+            eng.flush_all()
+        hi = lo + span
+        data = [(h, t, v) for h, t, v in _oracle(rows, append) if lo <= t < hi]
+        r = ex.execute(f"SELECT count(*), sum(v), min(v), max(v) FROM p "
+                       f"WHERE ts >= {lo} AND ts < {hi}")
+        cnt = r.columns[0][0]
+        assert cnt == len(data)
+        if data:
+            vs = [v for _h, _t, v in data]
+            assert math.isclose(r.columns[1][0], sum(vs), rel_tol=1e-9, abs_tol=1e-6)
+            assert math.isclose(r.columns[2][0], min(vs), rel_tol=0, abs_tol=0)
+            assert math.isclose(r.columns[3][0], max(vs), rel_tol=0, abs_tol=0)
+        # per-tag counts
+        r = ex.execute("SELECT h, count(*) FROM p GROUP BY h ORDER BY h")
+        exp = {}
+        for h, _t, _v in _oracle(rows, append):
+            exp[h] = exp.get(h, 0) + 1
+        got = dict(zip(r.columns[0], (int(c) for c in r.columns[1])))
+        assert got == exp
+    finally:
+        eng.close()
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(rows=ROWS, flush=st_.booleans())
+def test_raw_scan_matches_oracle(tmp_path_factory, rows, flush):
+    eng, ex = _mk(tmp_path_factory, rows, append=True)
+    try:
+        if flush:
+            eng.flush_all()
+        r = ex.execute("SELECT h, ts, v FROM p ORDER BY ts, h, v")
+        got = sorted((h, int(t), float(v)) for h, t, v in
+                     zip(r.columns[0], r.columns[1], r.columns[2]))
+        exp = sorted((h, t, float(np.float64(v))) for h, t, v in rows)
+        assert len(got) == len(exp)
+        for g, e in zip(got, exp):
+            assert g[0] == e[0] and g[1] == e[1]
+            assert math.isclose(g[2], e[2], rel_tol=1e-12, abs_tol=0)
+    finally:
+        eng.close()
+
+
+@settings(max_examples=20, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(rows=ROWS)
+def test_restart_preserves_data(tmp_path_factory, rows):
+    eng, ex = _mk(tmp_path_factory, rows, append=True)
+    d = eng.config.data_dir
+    n = len(rows)
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu",
+                                   background_flush=False))
+    try:
+        ex2 = Executor(eng2)
+        assert ex2.execute("SELECT count(*) FROM p").columns[0][0] == n
+    finally:
+        eng2.close()

@@ -50,7 +50,7 @@ def load_log_fixture(engine, n_events: int, rank=0, world=1, seed=5):
             ColumnSchema("service", DataType.STRING, SemanticType.TAG, 0),
             ColumnSchema("ts", DataType.TIMESTAMP_MS, SemanticType.TIMESTAMP, 1),
             ColumnSchema("latency", DataType.FLOAT64, SemanticType.FIELD, 2),
-            ColumnSchema("message", DataType.STRING, SemanticType.FIELD, 3),
+            ColumnSchema("message", DataType.STRING, SemanticType.FIELD, 3, fulltext=True),
         ],
         primary_key=["service"], options={"append_mode": "true"})
     st = engine.create_table(schema, append_mode=True, if_not_exists=True)

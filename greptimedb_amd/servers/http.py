@@ -131,7 +131,8 @@ def build_app(ctx: ServerContext) -> FastAPI:
                 sql = body or None
         t0 = time.perf_counter()
         try:
-            r = ctx.executor.execute(sql)
+            from starlette.concurrency import run_in_threadpool
+            r = await run_in_threadpool(ctx.executor.execute, sql)
         except GreptimeError as e:
             metrics_mod.counter("http_sql_errors").inc()
             return {"code": 3000, "error": str(e), "execution_time_ms":
@@ -400,7 +401,8 @@ def build_app(ctx: ServerContext) -> FastAPI:
         q = await _param(request, "query")
         t = _parse_time_s(await _param(request, "time"), time.time())
         try:
-            m = ctx.prom.query_instant(q, t)
+            from starlette.concurrency import run_in_threadpool
+            m = await run_in_threadpool(ctx.prom.query_instant, q, t)
         except GreptimeError as e:
             return {"status": "error", "errorType": "bad_data", "error": str(e)}
         return _prom_result(m, instant=True)
@@ -412,7 +414,8 @@ def build_app(ctx: ServerContext) -> FastAPI:
         end = _parse_time_s(await _param(request, "end"))
         step = _parse_step_s(await _param(request, "step"))
         try:
-            m = ctx.prom.query_range(q, start, end, step)
+            from starlette.concurrency import run_in_threadpool
+            m = await run_in_threadpool(ctx.prom.query_range, q, start, end, step)
         except GreptimeError as e:
             return {"status": "error", "errorType": "bad_data", "error": str(e)}
         return _prom_result(m, instant=False)

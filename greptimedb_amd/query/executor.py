@@ -862,7 +862,10 @@ class Executor:
 
         def collect(e):
             if isinstance(e, ast.Func) and e.name in AGG_FUNCS:
-                if len(e.args) == 1 and isinstance(e.args[0], ast.Star):
+                if e.distinct and e.name == "count" and len(e.args) == 1 and \
+                        isinstance(e.args[0], ast.Col) and e.args[0].name in tag_names:
+                    aggs.append(AggCall("count_distinct", e.args[0].name))
+                elif len(e.args) == 1 and isinstance(e.args[0], ast.Star):
                     aggs.append(AggCall("count", None))
                 elif len(e.args) == 1 and isinstance(e.args[0], ast.Col):
                     fname = "avg" if e.name == "mean" else e.name
@@ -1003,7 +1006,14 @@ class Executor:
             origin = ts_lo
             n_buckets = 1
 
-        group_keys, region_luts = self._build_group_luts(st, plan, plan.group_tags)
+        # count(DISTINCT tag): group by (outer tags + distinct tag) internally,
+        # then collapse effective groups back to outer groups at finalize
+        dtags = sorted({a.arg for a in plan.aggs if a.func == "count_distinct"})
+        if len(dtags) > 1:
+            raise PlanQuery("only one count(DISTINCT tag) per query")
+        eff_gt = list(plan.group_tags) + [t for t in dtags
+                                          if t not in plan.group_tags]
+        group_keys, region_luts = self._build_group_luts(st, plan, eff_gt)
         n_slots = max(len(group_keys), 1)
 
         # fields needed
@@ -1036,8 +1046,38 @@ class Executor:
                 group_keys, sums, cnts, mins, maxs, rowcnt)
             n_slots = max(len(group_keys), 1)
 
+        distinct_plane = None
+        if dtags:
+            # collapse effective groups (outer × distinct tag) → outer groups
+            n_out_tags = len(plan.group_tags)
+            outer_keys: dict[tuple, int] = {}
+            eff_to_outer = np.zeros(max(len(group_keys), 1), dtype=np.int64)
+            for key, slot in group_keys.items():
+                ok = tuple(key[:n_out_tags])
+                eff_to_outer[slot] = outer_keys.setdefault(ok, len(outer_keys))
+            n_outer = max(len(outer_keys), 1)
+            nb = rowcnt.shape[1]
+            new_sums = np.zeros((sums.shape[0], n_outer, nb))
+            new_cnts = np.zeros((cnts.shape[0], n_outer, nb), dtype=np.int64)
+            new_mins = np.full((mins.shape[0], n_outer, nb), np.nan)
+            new_maxs = np.full((maxs.shape[0], n_outer, nb), np.nan)
+            new_rows = np.zeros((n_outer, nb), dtype=np.int64)
+            distinct_plane = np.zeros((n_outer, nb), dtype=np.int64)
+            for s in range(rowcnt.shape[0]):
+                o = eff_to_outer[s]
+                new_sums[:, o] += np.where(cnts[:, s] > 0, sums[:, s], 0.0)
+                new_cnts[:, o] += cnts[:, s]
+                new_mins[:, o] = np.fmin(new_mins[:, o], mins[:, s])
+                new_maxs[:, o] = np.fmax(new_maxs[:, o], maxs[:, s])
+                new_rows[o] += rowcnt[s]
+                distinct_plane[o] += (rowcnt[s] > 0).astype(np.int64)
+            sums, cnts, mins, maxs, rowcnt = (new_sums, new_cnts, new_mins,
+                                              new_maxs, new_rows)
+            group_keys = outer_keys
+
         return self._finalize_agg(sel, plan, group_keys, origin, bucket_ms,
-                                  agg_fields, sums, cnts, mins, maxs, rowcnt)
+                                  agg_fields, sums, cnts, mins, maxs, rowcnt,
+                                  distinct_plane=distinct_plane, dtag=(dtags[0] if dtags else None))
 
     def _exec_lastpoint(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
         """TSBS `lastpoint` shape: last_value(field) per group (latest-ts row).
@@ -1236,7 +1276,8 @@ class Executor:
         yield ts_t, se_t, f_t, fidx_t
 
     def _finalize_agg(self, sel, plan, group_keys, origin, bucket_ms,
-                      agg_fields, sums, cnts, mins, maxs, rowcnt) -> QueryResult:
+                      agg_fields, sums, cnts, mins, maxs, rowcnt,
+                      distinct_plane=None, dtag=None) -> QueryResult:
         # cells with data
         slot_idx, bucket_idx = np.nonzero(rowcnt)
         if len(slot_idx) == 0 and not plan.group_tags and plan.bucket is None:
@@ -1254,6 +1295,8 @@ class Executor:
         fpos = {fn: i for i, fn in enumerate(agg_fields)}
 
         def agg_array(func, arg):
+            if func == "count_distinct":
+                return distinct_plane[slot_idx, bucket_idx]
             if func == "count" and arg is None:
                 return rowcnt[slot_idx, bucket_idx]
             fi = fpos[arg]
@@ -1297,6 +1340,9 @@ class Executor:
                 if e.name in AGG_FUNCS:
                     if len(e.args) == 1 and isinstance(e.args[0], ast.Star):
                         return agg_array("count", None), ""
+                    if e.distinct and e.name == "count" and \
+                            isinstance(e.args[0], ast.Col) and e.args[0].name == dtag:
+                        return agg_array("count_distinct", dtag), ""
                     fname = "avg" if e.name == "mean" else e.name
                     return agg_array(fname, e.args[0].name), ""
                 raise PlanQuery(f"unsupported function {e.name} in aggregate query")

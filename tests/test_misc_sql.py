@@ -82,3 +82,22 @@ def test_alter_add_column(ex):
     ex2 = ex.execute("SELECT column_name FROM information_schema.columns "
                      "WHERE table_name = 't1'")
     assert "extra" in list(ex2.columns[0]) and "note" in list(ex2.columns[0])
+
+
+def test_count_distinct_tag(tmp_engine):
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE cd (h STRING, dc STRING, ts TIMESTAMP TIME INDEX, "
+               "v DOUBLE, PRIMARY KEY (h, dc))")
+    ex.execute("INSERT INTO cd (h, dc, ts, v) VALUES "
+               "('a', 'east', 1000, 1.0), ('b', 'east', 1000, 2.0), "
+               "('c', 'west', 2000, 3.0), ('a', 'east', 2000, 4.0)")
+    r = ex.execute("SELECT count(DISTINCT h) FROM cd")
+    assert r.columns[0][0] == 3
+    r = ex.execute("SELECT dc, count(DISTINCT h), count(*) FROM cd "
+                   "GROUP BY dc ORDER BY dc")
+    assert [tuple(t) for t in r.rows()] == [("east", 2, 3), ("west", 1, 1)]
+    # with time bucket: distinct per bucket
+    r = ex.execute("SELECT date_trunc('second', ts) s, count(DISTINCT h) "
+                   "FROM cd GROUP BY s ORDER BY s")
+    assert [int(c) for c in r.columns[1]] == [2, 2]

@@ -38,6 +38,9 @@ def main():
     from greptimedb_amd.servers.http import ServerContext, build_app
 
     device = "cuda:0" if torch.cuda.is_available() else "cpu"
+    # concurrent queries each spawn torch intra-op threads — cap to avoid
+    # oversubscription under N in-flight queries
+    torch.set_num_threads(2)
     eng = MitoEngine(EngineConfig(data_dir=tempfile.mkdtemp(prefix="gdb_serve_"),
                                   device=device, background_flush=False))
     n = load_cpu_fixture(eng, scale=args.scale, hours=args.hours)
@@ -71,23 +74,25 @@ def main():
     counts = [0] * args.clients
     lats: list[list[float]] = [[] for _ in range(args.clients)]
 
-    def client(ci):
-        with httpx.Client(base_url=base, timeout=30) as c:
-            i = ci
-            while time.perf_counter() < stop:
-                q = queries[i % len(queries)]
-                i += args.clients
-                t0 = time.perf_counter()
-                r = c.get("/v1/sql", params={"sql": q})
-                lats[ci].append((time.perf_counter() - t0) * 1000)
-                assert r.status_code == 200
-                counts[ci] += 1
+    clients = [httpx.Client(base_url=base, timeout=60) for _ in range(args.clients)]
 
-    # warmup
-    with httpx.Client(base_url=base, timeout=30) as c:
-        for q in queries[:10]:
-            c.get("/v1/sql", params={"sql": q})
+    def client(ci):
+        c = clients[ci]
+        i = ci
+        while time.perf_counter() < stop:
+            q = queries[i % len(queries)]
+            i += args.clients
+            t0 = time.perf_counter()
+            r = c.get("/v1/sql", params={"sql": q})
+            lats[ci].append((time.perf_counter() - t0) * 1000)
+            assert r.status_code == 200
+            counts[ci] += 1
+
+    # warmup (also primes each client's connection)
+    for ci, c in enumerate(clients):
+        c.get("/v1/sql", params={"sql": queries[ci % len(queries)]})
     t_start = time.perf_counter()
+    stop = t_start + args.duration
     threads = [threading.Thread(target=client, args=(i,))
                for i in range(args.clients)]
     for t in threads:

@@ -105,6 +105,15 @@ def _prom_result(m, instant: bool) -> dict:
 def build_app(ctx: ServerContext) -> FastAPI:
     app = FastAPI(title="greptimedb-amd")
 
+    @app.on_event("startup")
+    async def _widen_threadpool():
+        # concurrent query execution capacity (default anyio limit is 40)
+        try:
+            from anyio import to_thread
+            to_thread.current_default_thread_limiter().total_tokens = 128
+        except Exception:
+            pass
+
     @app.get("/health")
     @app.get("/ready")
     def health():
@@ -123,6 +132,9 @@ def build_app(ctx: ServerContext) -> FastAPI:
     # ---------------- SQL ----------------
 
     async def _sql(request: Request, sql: str | None):
+        import json as _json
+        if sql is None:
+            sql = request.query_params.get("sql")
         if sql is None:
             form = await request.form()
             sql = form.get("sql")
@@ -130,16 +142,27 @@ def build_app(ctx: ServerContext) -> FastAPI:
                 body = (await request.body()).decode()
                 sql = body or None
         t0 = time.perf_counter()
-        try:
-            from starlette.concurrency import run_in_threadpool
-            r = await run_in_threadpool(ctx.executor.execute, sql)
-        except GreptimeError as e:
-            metrics_mod.counter("http_sql_errors").inc()
-            return {"code": 3000, "error": str(e), "execution_time_ms":
-                    round((time.perf_counter() - t0) * 1000, 3)}
-        metrics_mod.counter("http_sql_requests").inc()
-        return {"output": [_records_json(r)],
-                "execution_time_ms": round((time.perf_counter() - t0) * 1000, 3)}
+
+        def run():
+            # execute + serialize off the event loop (and skip fastapi's
+            # jsonable_encoder — it dominates per-request cost)
+            try:
+                r = ctx.executor.execute(sql)
+            except GreptimeError as e:
+                metrics_mod.counter("http_sql_errors").inc()
+                return _json.dumps(
+                    {"code": 3000, "error": str(e) or type(e).__name__,
+                     "execution_time_ms":
+                         round((time.perf_counter() - t0) * 1000, 3)})
+            metrics_mod.counter("http_sql_requests").inc()
+            return _json.dumps(
+                {"output": [_records_json(r)],
+                 "execution_time_ms":
+                     round((time.perf_counter() - t0) * 1000, 3)})
+
+        from starlette.concurrency import run_in_threadpool
+        payload = await run_in_threadpool(run)
+        return Response(payload, media_type="application/json")
 
     @app.get("/v1/sql")
     async def sql_get(request: Request, sql: str = Query(None)):

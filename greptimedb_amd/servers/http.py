@@ -110,7 +110,7 @@ def build_app(ctx: ServerContext) -> FastAPI:
         # concurrent query execution capacity (default anyio limit is 40)
         try:
             from anyio import to_thread
-            to_thread.current_default_thread_limiter().total_tokens = 128
+            to_thread.current_default_thread_limiter().total_tokens = 16
         except Exception:
             pass
 

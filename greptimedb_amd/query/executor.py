@@ -1408,16 +1408,30 @@ class Executor:
                 if sn not in str_field_names:
                     str_field_names.append(sn)
 
-        # output columns
+        # output columns; scalar expressions over fields/ts are computed
+        # per-source on device (value() evaluator) into synthetic columns
         out_cols: list[str] = []
-        for e, alias in sel.projections:
+        expr_cols: list[tuple[str, ast.Expr]] = []
+        for i, (e, alias) in enumerate(sel.projections):
             if isinstance(e, ast.Star):
                 out_cols.extend(tag_names + [ts_name] + field_names + str_field_names)
             elif isinstance(e, ast.Col):
                 out_cols.append(e.name)
             else:
-                raise PlanQuery("raw SELECT supports columns and * only (no exprs yet)")
-        needed_fields = [c for c in out_cols if c in field_names]
+                name = alias or _expr_name(e)
+                expr_cols.append((name, e))
+                out_cols.append(name)
+        expr_names = {n for n, _ in expr_cols}
+        needed_fields = [c for c in out_cols
+                         if c in field_names and c not in expr_names]
+        for _n, e in expr_cols:
+            for c in _expr_cols(e):
+                if c in field_names and c not in needed_fields:
+                    needed_fields.append(c)
+                elif c in tag_names or c == ts_name:
+                    pass
+                elif c not in field_names:
+                    raise PlanQuery(f"unknown column {c} in projection expr")
         needed_strs = [c for c in out_cols if c in str_field_names]
         if needed_strs and not st.append_mode:
             raise PlanQuery("string columns require append_mode tables")
@@ -1492,7 +1506,9 @@ class Executor:
                           f_t.cpu().numpy(), s_cols))
 
         # materialize host rows
-        col_data = {c: [] for c in set(out_cols) | set(order_cols)}
+        mat_cols = (set(out_cols) | set(order_cols) | set(needed_fields) |
+                    {ts_name} if expr_cols else set(out_cols) | set(order_cols))
+        col_data = {c: [] for c in mat_cols if c not in expr_names}
         for ts_h, se_h, region, f_h, s_cols in parts:
             for c in col_data:
                 if c == ts_name:
@@ -1512,6 +1528,10 @@ class Executor:
         if self.dist is not None:
             col_data = self.dist.gather_columns(col_data)
         n = len(next(iter(col_data.values()))) if col_data else 0
+        # projection expressions over materialized columns (numpy)
+        for name, e in expr_cols:
+            col_data[name] = np.asarray(_eval_np_expr(e, col_data), dtype=float) \
+                if n else np.array([])
 
         idx = np.arange(n)
         for e, desc in reversed(plan.order_by):
@@ -1753,6 +1773,32 @@ def _np_binop(op, l, r):
     if op == "or":
         return np.asarray(l, dtype=bool) | np.asarray(r, dtype=bool)
     raise PlanQuery(f"binop {op}")
+
+
+_NP_FUNCS = {"abs": np.abs, "floor": np.floor, "ceil": np.ceil,
+             "sqrt": np.sqrt, "ln": np.log, "log2": np.log2,
+             "log10": np.log10, "exp": np.exp}
+
+
+def _eval_np_expr(e: ast.Expr, col_data: dict):
+    """Scalar expression over materialized numpy columns (raw-path
+    projections like `v * 8 / 1024`)."""
+    if isinstance(e, ast.Col):
+        return np.asarray(col_data[e.name], dtype=np.float64)
+    if isinstance(e, ast.Lit):
+        return float(e.value)
+    if isinstance(e, ast.BinOp):
+        return _np_binop(e.op, _eval_np_expr(e.left, col_data),
+                         _eval_np_expr(e.right, col_data))
+    if isinstance(e, ast.UnaryOp) and e.op == "-":
+        return -_eval_np_expr(e.operand, col_data)
+    if isinstance(e, ast.Func) and e.name in _NP_FUNCS:
+        return _NP_FUNCS[e.name](_eval_np_expr(e.args[0], col_data))
+    if isinstance(e, ast.Func) and e.name == "round":
+        v = _eval_np_expr(e.args[0], col_data)
+        nd = int(e.args[1].value) if len(e.args) > 1 else 0
+        return np.round(v, nd)
+    raise PlanQuery(f"unsupported projection expr {e}")
 
 
 def _eval_const(e: ast.Expr):

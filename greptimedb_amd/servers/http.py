@@ -105,9 +105,8 @@ def _prom_result(m, instant: bool) -> dict:
 def build_app(ctx: ServerContext) -> FastAPI:
     app = FastAPI(title="greptimedb-amd")
 
-    @app.on_event("startup")
-    async def _widen_threadpool():
-        # concurrent query execution capacity (default anyio limit is 40)
+    def _tune_threadpool():
+        # concurrent query execution capacity (GIL contention sweet spot)
         try:
             from anyio import to_thread
             to_thread.current_default_thread_limiter().total_tokens = 16
@@ -160,6 +159,7 @@ def build_app(ctx: ServerContext) -> FastAPI:
                  "execution_time_ms":
                      round((time.perf_counter() - t0) * 1000, 3)})
 
+        _tune_threadpool()
         from starlette.concurrency import run_in_threadpool
         payload = await run_in_threadpool(run)
         return Response(payload, media_type="application/json")

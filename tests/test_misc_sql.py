@@ -101,3 +101,9 @@ def test_count_distinct_tag(tmp_engine):
     r = ex.execute("SELECT date_trunc('second', ts) s, count(DISTINCT h) "
                    "FROM cd GROUP BY s ORDER BY s")
     assert [int(c) for c in r.columns[1]] == [2, 2]
+
+
+def test_raw_projection_expressions(ex):
+    r = ex.execute("SELECT h, v * 2 + 1 AS d, round(sqrt(v), 2) FROM t1 ORDER BY h")
+    assert list(r.columns[1]) == [4.0, 6.0]
+    assert abs(r.columns[2][0] - 1.22) < 1e-9

@@ -74,9 +74,19 @@ class IsNull(Expr):
 
 
 @dataclass
+class Join:
+    table: str
+    alias: str | None
+    on: Expr                      # join condition (equality conjuncts)
+    kind: str = "inner"           # inner | left
+
+
+@dataclass
 class Select:
     projections: list[tuple[Expr, str | None]]  # (expr, alias)
     table: str | None
+    table_alias: str | None = None
+    joins: list = field(default_factory=list)
     where: Expr | None = None
     group_by: list[Expr] = field(default_factory=list)
     having: Expr | None = None

@@ -229,7 +229,7 @@ class Executor:
         read/compaction time; with device-resident data we rewrite the
         affected SSTs directly (flush first so the memtable is included) —
         durable immediately, no tombstone debt on the scan path."""
-        sel = ast.Select([(ast.Star(), None)], d.table, d.where)
+        sel = ast.Select([(ast.Star(), None)], d.table, where=d.where)
         plan = self._plan_select(sel)
         st = plan.table
         device = self.engine.config.device
@@ -601,6 +601,8 @@ class Executor:
         from greptimedb_amd.query.information_schema import is_information_schema
         if is_information_schema(sel.table):
             return self._exec_information_schema(sel)
+        if sel.joins:
+            return self._exec_join(sel)
         plan = self._plan_select(sel)
         if plan.aggs:
             return self._exec_aggregate(sel, plan)
@@ -608,6 +610,169 @@ class Executor:
         if knn is not None:
             return knn
         return self._exec_raw(sel, plan)
+
+    def _exec_join(self, sel: ast.Select) -> QueryResult:
+        """Two-table equality JOIN (inner/left): each side runs as a pushed-
+        down raw scan, then a host-side hash join (reference: DataFusion
+        HashJoinExec; time-series queries rarely join, so host-side is the
+        right cost tier)."""
+        if len(sel.joins) != 1:
+            raise PlanQuery("only single JOIN supported")
+        j = sel.joins[0]
+        sides = {
+            (sel.table_alias or sel.table): sel.table,
+            (j.alias or j.table): j.table,
+        }
+        aliases = list(sides)
+        if len(sides) != 2:
+            raise PlanQuery("join sides must have distinct aliases")
+
+        def side_columns(alias):
+            st = self.engine.table(sides[alias])
+            cols = {c.name for c in st.schema.columns}
+            cols.update(st.regions[0].field_names)
+            cols.update(st.regions[0].str_field_names)
+            return cols
+
+        side_cols = {a: side_columns(a) for a in aliases}
+
+        def resolve(name):
+            """'a.x' or unqualified 'x' → (alias, col)."""
+            if "." in name:
+                a, c = name.split(".", 1)
+                if a in sides:
+                    return a, c
+            owners = [a for a in aliases if name in side_cols[a]]
+            if len(owners) == 1:
+                return owners[0], name
+            raise PlanQuery(f"ambiguous or unknown join column {name!r}")
+
+        # ON: equality conjuncts across sides
+        pairs = []
+        for c in _split_conjuncts(j.on):
+            if not (isinstance(c, ast.BinOp) and c.op == "=" and
+                    isinstance(c.left, ast.Col) and isinstance(c.right, ast.Col)):
+                raise PlanQuery("JOIN ON supports column equality conjuncts")
+            la, lc = resolve(c.left.name)
+            ra, rc = resolve(c.right.name)
+            if la == ra:
+                raise PlanQuery("JOIN ON must compare columns across sides")
+            if la == aliases[0]:
+                pairs.append((lc, rc))
+            else:
+                pairs.append((rc, lc))
+
+        # WHERE: push side-local conjuncts down; cross-side unsupported
+        side_where = {a: [] for a in aliases}
+        if sel.where is not None:
+            for c in _split_conjuncts(sel.where):
+                refs = {resolve(n)[0] for n in _expr_cols(c)}
+                if len(refs) != 1:
+                    raise PlanQuery("cross-side WHERE predicates unsupported")
+                a = next(iter(refs))
+
+                def strip(e):
+                    if isinstance(e, ast.Col):
+                        return ast.Col(resolve(e.name)[1])
+                    if isinstance(e, ast.BinOp):
+                        return ast.BinOp(e.op, strip(e.left), strip(e.right))
+                    if isinstance(e, ast.UnaryOp):
+                        return ast.UnaryOp(e.op, strip(e.operand))
+                    if isinstance(e, ast.InList):
+                        return ast.InList(strip(e.expr), e.items, e.negated)
+                    if isinstance(e, ast.Between):
+                        return ast.Between(strip(e.expr), e.low, e.high, e.negated)
+                    if isinstance(e, ast.Func):
+                        return ast.Func(e.name, [strip(x) for x in e.args])
+                    return e
+                side_where[a].append(strip(c))
+
+        # needed columns per side
+        need = {a: set() for a in aliases}
+        out_spec = []  # (alias_out_name, side, col)
+        for e, al in sel.projections:
+            if isinstance(e, ast.Star):
+                for a in aliases:
+                    for c in sorted(side_cols[a]):
+                        out_spec.append((f"{a}.{c}", a, c))
+                        need[a].add(c)
+                continue
+            if not isinstance(e, ast.Col):
+                raise PlanQuery("JOIN projections: columns or *")
+            a, c = resolve(e.name)
+            out_spec.append((al or e.name, a, c))
+            need[a].add(c)
+        for lc, rc in pairs:
+            need[aliases[0]].add(lc)
+            need[aliases[1]].add(rc)
+        order_resolved = []
+        for e, desc in sel.order_by:
+            if not isinstance(e, ast.Col):
+                raise PlanQuery("JOIN ORDER BY: columns only")
+            a, c = resolve(e.name)
+            need[a].add(c)
+            order_resolved.append((a, c, desc))
+
+        # execute both sides as raw scans
+        data = {}
+        for a in aliases:
+            w = None
+            for c in side_where[a]:
+                w = c if w is None else ast.BinOp("and", w, c)
+            sub = ast.Select([(ast.Col(c), None) for c in sorted(need[a])],
+                             sides[a], where=w)
+            r = self.execute_stmt(sub)
+            data[a] = {n: np.asarray(col) for n, col in zip(r.names, r.columns)}
+        la, ra = aliases
+
+        def keys_of(a, cols):
+            arrs = [data[a][c] for c in cols]
+            n = len(arrs[0]) if arrs else 0
+            return [tuple(x[i] for x in arrs) for i in range(n)]
+
+        lkeys = keys_of(la, [p[0] for p in pairs])
+        rkeys = keys_of(ra, [p[1] for p in pairs])
+        rindex: dict = {}
+        for i, k in enumerate(rkeys):
+            rindex.setdefault(k, []).append(i)
+        li, ri = [], []
+        for i, k in enumerate(lkeys):
+            hits = rindex.get(k)
+            if hits:
+                for h in hits:
+                    li.append(i)
+                    ri.append(h)
+            elif j.kind == "left":
+                li.append(i)
+                ri.append(-1)
+        li = np.asarray(li, dtype=np.int64)
+        ri = np.asarray(ri, dtype=np.int64)
+
+        names, cols, kinds = [], [], []
+        for name, a, c in out_spec:
+            src = data[a][c]
+            if a == la:
+                vals = src[li] if len(li) else src[:0]
+            else:
+                vals = np.array([src[x] if x >= 0 else None for x in ri],
+                                dtype=object)
+            names.append(name)
+            cols.append(vals)
+            ts_col = self.engine.table(sides[a]).schema.time_index.name
+            kinds.append("ts" if c == ts_col else "")
+        n_out = len(li)
+        idx = np.arange(n_out)
+        for a, c, desc in reversed(order_resolved):
+            arr = (data[a][c][li] if a == la else
+                   np.array([data[a][c][x] if x >= 0 else None for x in ri],
+                            dtype=object))
+            o = np.argsort(arr[idx], kind="stable")
+            idx = idx[o[::-1] if desc else o]
+        if sel.offset:
+            idx = idx[sel.offset:]
+        if sel.limit is not None:
+            idx = idx[: sel.limit]
+        return QueryResult(names, [c[idx] for c in cols], kinds)
 
     VEC_FUNCS = {"vec_cos_distance": "cos", "vec_l2sq_distance": "l2sq",
                  "vec_dot_product": "dot"}
@@ -1433,8 +1598,6 @@ class Executor:
                 elif c not in field_names:
                     raise PlanQuery(f"unknown column {c} in projection expr")
         needed_strs = [c for c in out_cols if c in str_field_names]
-        if needed_strs and not st.append_mode:
-            raise PlanQuery("string columns require append_mode tables")
         order_cols = [e.name for e, _ in plan.order_by if isinstance(e, ast.Col)]
         for c in order_cols:
             if c in field_names and c not in needed_fields:
@@ -1491,17 +1654,16 @@ class Executor:
                       for sn in needed_strs}
             if not st.append_mode and len(chunks) >= 1:
                 # sort by (series, ts, arrival) then keep last
-                arrival = torch.arange(ts_t.numel(), device=device)
-                o1 = torch.argsort(arrival, stable=True)
-                o2 = torch.argsort(ts_t[o1], stable=True)
-                perm = o1[o2]
-                o3 = torch.argsort(se_t[perm], stable=True)
-                perm = perm[o3]
+                o2 = torch.argsort(ts_t, stable=True)
+                perm = o2[torch.argsort(se_t[o2], stable=True)]
                 ts_t, se_t = ts_t[perm], se_t[perm]
                 f_t = f_t[:, perm]
                 keep = dedup_mark_last(se_t.contiguous(), ts_t.contiguous())
                 kidx = keep.nonzero(as_tuple=True)[0]
                 ts_t, se_t, f_t = ts_t[kidx], se_t[kidx], f_t[:, kidx]
+                if needed_strs:
+                    sel_h = perm[kidx].cpu().numpy()
+                    s_cols = {sn: v[sel_h] for sn, v in s_cols.items()}
             parts.append((ts_t.cpu().numpy(), se_t.cpu().numpy(), region,
                           f_t.cpu().numpy(), s_cols))
 

@@ -246,8 +246,35 @@ class Parser:
             if not self.eat_op(","):
                 break
         table = None
+        table_alias = None
+        joins = []
         if self.eat_kw("from"):
             table = self.next().value
+            if self.eat_kw("as"):
+                table_alias = self.next().value
+            elif self.peek() is not None and self.peek().kind == "id" and \
+                    self.peek().value.lower() not in (
+                        "where", "group", "order", "limit", "having", "offset",
+                        "join", "inner", "left", "on"):
+                table_alias = self.next().value
+            while self.at_kw("join", "inner", "left"):
+                kind = "inner"
+                if self.eat_kw("left"):
+                    kind = "left"
+                    self.eat_kw("outer")
+                else:
+                    self.eat_kw("inner")
+                self.expect_kw("join")
+                jt = self.next().value
+                jalias = None
+                if self.eat_kw("as"):
+                    jalias = self.next().value
+                elif self.peek() is not None and self.peek().kind == "id" and \
+                        self.peek().value.lower() != "on":
+                    jalias = self.next().value
+                self.expect_kw("on")
+                on = self.parse_expr()
+                joins.append(ast.Join(jt, jalias, on, kind))
         where = None
         if self.eat_kw("where"):
             where = self.parse_expr()
@@ -279,7 +306,9 @@ class Parser:
             limit = int(self.next().value)
         if self.eat_kw("offset"):
             offset = int(self.next().value)
-        return ast.Select(projections, table, where, group_by, having, order_by, limit, offset)
+        return ast.Select(projections, table, table_alias=table_alias, joins=joins,
+                          where=where, group_by=group_by, having=having,
+                          order_by=order_by, limit=limit, offset=offset)
 
     def parse_create(self):
         self.expect_kw("create")

@@ -107,3 +107,23 @@ def test_raw_projection_expressions(ex):
     r = ex.execute("SELECT h, v * 2 + 1 AS d, round(sqrt(v), 2) FROM t1 ORDER BY h")
     assert list(r.columns[1]) == [4.0, 6.0]
     assert abs(r.columns[2][0] - 1.22) < 1e-9
+
+
+def test_inner_join(tmp_engine):
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE m (host STRING, ts TIMESTAMP TIME INDEX, cpu DOUBLE, PRIMARY KEY (host))")
+    ex.execute("CREATE TABLE meta (host STRING, ts TIMESTAMP TIME INDEX, team STRING, PRIMARY KEY (host))")
+    ex.execute("INSERT INTO m (host, ts, cpu) VALUES ('a', 1000, 50.0), ('b', 2000, 70.0), ('c', 3000, 90.0)")
+    ex.execute("INSERT INTO meta (host, ts, team) VALUES ('a', 0, 'sre'), ('b', 0, 'db')")
+    r = ex.execute("SELECT x.host, x.cpu, y.team FROM m x JOIN meta y "
+                   "ON x.host = y.host ORDER BY x.host")
+    assert [tuple(t) for t in r.rows()] == [("a", 50.0, "sre"), ("b", 70.0, "db")]
+    # left join keeps unmatched with NULL
+    r = ex.execute("SELECT x.host, y.team FROM m x LEFT JOIN meta y "
+                   "ON x.host = y.host ORDER BY x.host")
+    assert [tuple(t) for t in r.rows()] == [("a", "sre"), ("b", "db"), ("c", None)]
+    # side-local WHERE pushdown
+    r = ex.execute("SELECT x.host FROM m x JOIN meta y ON x.host = y.host "
+                   "WHERE x.cpu > 60 ORDER BY x.host")
+    assert list(r.columns[0]) == ["b"]

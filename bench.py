@@ -48,6 +48,9 @@ def main():
                     help="ingest worker threads per rank (TSBS uses 6)")
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--durable", action="store_true", default=True)
+    ap.add_argument("--flush-mb", type=int, default=1024,
+                    help="memtable flush threshold (small values exercise "
+                         "flush+compaction during the timed region)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -81,7 +84,7 @@ def main():
     eng = MitoEngine(EngineConfig(
         data_dir=data_dir, device=device,
         background_flush=True, wal_sync=False,
-        flush_bytes=1 << 30))
+        flush_bytes=args.flush_mb << 20))
 
     # ---------------- pre-generate all line batches (untimed) ----------------
     # Each worker thread owns a disjoint host shard (its own Ingestor/parser,
@@ -198,6 +201,7 @@ def main():
             "workers": n_workers,
             "batches_per_step": BATCHES_PER_STEP,
             "wal": "group-commit, no fsync",
+            "flush_mb": args.flush_mb,
             "queries": queries,
             "device": device,
         },

@@ -34,6 +34,10 @@ void launch_series_last_row(
 void launch_prom_range_eval(
     const int64_t*, const double*, const int64_t*, const int64_t*, int, int,
     int64_t, int64_t, int64_t, int64_t, double, int, double*, hipStream_t);
+void launch_scatter_append(
+    const int64_t*, const int32_t*, const double*, const int32_t*,
+    const int64_t*, int64_t* const*, int32_t* const*, double* const*,
+    const int64_t*, int, int64_t, hipStream_t);
 }  // namespace gdb_hip
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -193,6 +197,40 @@ torch::Tensor prom_range_eval(
   return out;
 }
 
+// K16 bulk append: scatter one routed batch into multiple region memtables.
+// `dests` = per-region (ts, series, fields) destination tensors.
+void scatter_append(
+    torch::Tensor ts, torch::Tensor series, torch::Tensor fields,
+    torch::Tensor region_of, torch::Tensor dst_off,
+    std::vector<torch::Tensor> dst_ts, std::vector<torch::Tensor> dst_se,
+    std::vector<torch::Tensor> dst_fields) {
+  CHECK_GPU(ts); CHECK_CONTIG(ts); CHECK_GPU(fields); CHECK_CONTIG(fields);
+  CHECK_GPU(region_of); CHECK_CONTIG(region_of);
+  CHECK_GPU(dst_off); CHECK_CONTIG(dst_off);
+  const int64_t n = ts.numel();
+  const int nf = (int)fields.size(0);
+  const int R = (int)dst_ts.size();
+  std::vector<int64_t> hptrs(R * 3 + R);
+  for (int r = 0; r < R; r++) {
+    hptrs[r] = (int64_t)dst_ts[r].data_ptr<int64_t>();
+    hptrs[R + r] = (int64_t)dst_se[r].data_ptr<int32_t>();
+    hptrs[2 * R + r] = (int64_t)dst_fields[r].data_ptr<double>();
+    hptrs[3 * R + r] = dst_fields[r].size(1);  // stride = cap
+  }
+  auto ptrs = torch::from_blob(hptrs.data(), {(int64_t)hptrs.size()},
+                               torch::kInt64).to(ts.device());
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  const int64_t* pbase = ptrs.data_ptr<int64_t>();
+  gdb_hip::launch_scatter_append(
+      ts.data_ptr<int64_t>(), series.data_ptr<int32_t>(),
+      fields.data_ptr<double>(), region_of.data_ptr<int32_t>(),
+      dst_off.data_ptr<int64_t>(),
+      reinterpret_cast<int64_t* const*>(pbase),
+      reinterpret_cast<int32_t* const*>(pbase + R),
+      reinterpret_cast<double* const*>(pbase + 2 * R),
+      pbase + 3 * R, nf, n, stream);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate",
         py::arg("ts"), py::arg("series"), py::arg("fields"), py::arg("field_idx"),
@@ -205,4 +243,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("series_last_ts", &series_last_ts, "per-slot max-ts accumulate (lastpoint)");
   m.def("series_last_row", &series_last_row, "per-slot winner row (lastpoint)");
   m.def("prom_range_eval", &prom_range_eval, "PromQL range-vector evaluator");
+  m.def("scatter_append", &scatter_append, "bulk routed memtable append (K16)");
 }

@@ -431,6 +431,36 @@ __global__ void prom_range_eval_kernel(
   }
 }
 
+// ------------------------------------------------- K16 scatter append
+/// One launch appends a routed wire batch into MULTIPLE region memtables:
+// row i goes to region region_of[i] at row dst_off[i]. Replaces per-region
+// narrow+copy_ chains (two dozen small torch dispatches per batch) with a
+// single kernel — the ingest hot path's device side.
+__global__ void scatter_append_kernel(
+    const int64_t* __restrict__ ts,
+    const int32_t* __restrict__ series,
+    const double* __restrict__ fields,        // [nf, n] row-major
+    const int32_t* __restrict__ region_of,
+    const int64_t* __restrict__ dst_off,
+    int64_t* const* __restrict__ ts_ptrs,     // [R]
+    int32_t* const* __restrict__ se_ptrs,
+    double* const* __restrict__ f_ptrs,
+    const int64_t* __restrict__ strides,      // [R] field stride (cap)
+    int nf, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const int r = region_of[i];
+    const int64_t o = dst_off[i];
+    ts_ptrs[r][o] = ts[i];
+    se_ptrs[r][o] = series[i];
+    double* __restrict__ fp = f_ptrs[r];
+    const int64_t st = strides[r];
+    for (int f = 0; f < nf; f++) {
+      fp[(int64_t)f * st + o] = fields[(int64_t)f * n + i];
+    }
+  }
+}
+
 // ---------------------------------------------------------------- launchers
 
 static inline int grid_for(int64_t n, int block) {
@@ -506,6 +536,16 @@ void launch_bucket_agg2(
         (int64_t)n_slots * n_buckets, chunk,
         out_sum, out_cnt, out_min, out_max);
   }
+}
+
+void launch_scatter_append(
+    const int64_t* ts, const int32_t* series, const double* fields,
+    const int32_t* region_of, const int64_t* dst_off,
+    int64_t* const* ts_ptrs, int32_t* const* se_ptrs, double* const* f_ptrs,
+    const int64_t* strides, int nf, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(scatter_append_kernel, dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      ts, series, fields, region_of, dst_off, ts_ptrs, se_ptrs, f_ptrs,
+      strides, nf, n);
 }
 
 void launch_series_last(

@@ -163,6 +163,96 @@ class MitoEngine:
                 cb(table.schema.name, lo, hi, len(ts_ms))
         return seq
 
+    def write_regions_bulk(self, targets, series_codes: np.ndarray,
+                           ts_ms: np.ndarray, fields: np.ndarray,
+                           order: np.ndarray, starts: np.ndarray,
+                           ends: np.ndarray, durable: bool = True) -> bool:
+        """K16 fast path: one scatter_append kernel writes EVERY region's
+        slice of a routed write batch, replacing the per-region
+        write_region → narrow+copy_ chain (~6 torch dispatches per region).
+
+        targets[k] = (TableState, region_idx) for segment order[starts[k]:
+        ends[k]]; series_codes/ts_ms/fields[nf, n] are in ORIGINAL row order
+        and all targets must share one field layout (single table — the
+        caller checks). Returns False when a target needs the slow path
+        (text columns keep host-side row lists)."""
+        import torch
+        from greptimedb_amd.ops import kernels as ops
+
+        n = len(ts_ms)
+        if n == 0:
+            return True
+        nf = fields.shape[0]
+        regions: list[Region] = [st.regions[ri] for st, ri in targets]
+        if any(r.text_cols or r.memtable.str_cols for r in regions):
+            return False
+        # WAL first (host-side; replay is region-keyed, so one entry per region)
+        seqs = [0] * len(regions)
+        if durable:
+            for k, ((st, ri), s, e) in enumerate(zip(targets, starts, ends)):
+                rows = order[s:e]
+                region = regions[k]
+                payload = encode_batch(series_codes[rows], ts_ms[rows],
+                                       fields[:, rows], region.field_names, [])
+                seqs[k] = self.wal.append(region.region_id, payload)
+        # lock all regions in a stable global order (no deadlock vs flush or
+        # a concurrent bulk writer), reserve rows, launch, update metadata
+        ordered = sorted(regions, key=lambda r: r.region_id)
+        for r in ordered:
+            r.lock.acquire()
+        try:
+            seg_lens = (ends - starts).astype(np.int64)
+            bases = np.empty(len(regions), dtype=np.int64)
+            for k, region in enumerate(regions):
+                mem = region.memtable
+                if mem.nf < nf:
+                    mem.add_fields(nf - mem.nf)
+                need = mem.len + int(seg_lens[k])
+                if need > mem.cap:
+                    mem._grow(need)
+                bases[k] = mem.len
+            # dst offsets / dense region ids back in ORIGINAL row order
+            within = np.arange(n, dtype=np.int64) - np.repeat(starts, seg_lens)
+            dst_sorted = np.repeat(bases, seg_lens) + within
+            dense_sorted = np.repeat(np.arange(len(regions), dtype=np.int32),
+                                     seg_lens)
+            dst_off = np.empty(n, dtype=np.int64)
+            dst_off[order] = dst_sorted
+            region_of = np.empty(n, dtype=np.int32)
+            region_of[order] = dense_sorted
+            dev = self.config.device
+            ts_t = torch.from_numpy(np.ascontiguousarray(ts_ms)).to(dev)
+            se_t = torch.from_numpy(
+                np.ascontiguousarray(series_codes, dtype=np.int32)).to(dev)
+            f_t = torch.from_numpy(np.ascontiguousarray(fields)).to(dev)
+            ro_t = torch.from_numpy(region_of).to(dev)
+            do_t = torch.from_numpy(dst_off).to(dev)
+            ops.scatter_append(ts_t, se_t, f_t, ro_t, do_t,
+                               [r.memtable.ts for r in regions],
+                               [r.memtable.series for r in regions],
+                               [r.memtable.fields for r in regions])
+            # per-segment min/max on host (reduceat over the region-sorted view)
+            ts_sorted = ts_ms[order]
+            seg_min = np.minimum.reduceat(ts_sorted, starts)
+            seg_max = np.maximum.reduceat(ts_sorted, starts)
+            for k, region in enumerate(regions):
+                mem = region.memtable
+                mem.len = int(bases[k] + seg_lens[k])
+                mn, mx = int(seg_min[k]), int(seg_max[k])
+                mem.min_ts = mn if mem.min_ts is None else min(mem.min_ts, mn)
+                mem.max_ts = mx if mem.max_ts is None else max(mem.max_ts, mx)
+                region.last_seq = max(region.last_seq, seqs[k])
+        finally:
+            for r in reversed(ordered):
+                r.lock.release()
+        if self.write_listeners:
+            lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
+            names = {st.schema.name for st, _ in targets}
+            for name in names:
+                for cb in self.write_listeners:
+                    cb(name, lo, hi, n)
+        return True
+
     def commit_wal(self):
         self.wal.commit()
 

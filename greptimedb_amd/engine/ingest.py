@@ -42,6 +42,9 @@ class Ingestor:
         self.append_mode = append_mode
         self.durable = durable
         self.default_regions = default_regions or engine.config.default_regions
+        # K16 bulk scatter-append: on by default on GPU (one kernel per batch
+        # instead of per-region copy chains); CPU keeps the per-region path
+        self._bulk = engine.config.device.startswith("cuda")
         # flat routing state, indexed by parser sid
         self._cap = 1024
         self.sid_region = np.full(self._cap, -1, dtype=np.int32)   # flat region idx
@@ -168,6 +171,24 @@ class Ingestor:
         starts = np.concatenate(([0], bounds))
         ends = np.concatenate((bounds, [n]))
         engine = self.engine
+        targets = [self.flat_regions[int(region_sorted[s])] for s in starts]
+        # K16 bulk path (GPU): single-table batches — one scatter kernel
+        # writes every region's slice instead of per-region copy_ chains
+        if self._bulk and len({id(st) for st, _ in targets}) == 1:
+            st0 = targets[0][0]
+            all_rows = np.arange(n)
+            fmap = self._field_map(st0, parser_fields, fields_mat, all_rows)
+            out = np.empty((len(fmap), n), dtype=np.float64)
+            for i, src in enumerate(fmap):
+                out[i] = fields_mat[src] if src >= 0 else np.nan
+            if engine.write_regions_bulk(targets, local.astype(np.int32),
+                                         ts_ms, out, order, starts, ends,
+                                         durable=self.durable):
+                if self.durable:
+                    engine.commit_wal()
+                engine.maybe_flush()
+                self.rows_ingested += n
+                return n
         for s, e in zip(starts, ends):
             flat = int(region_sorted[s])
             st, region_idx = self.flat_regions[flat]

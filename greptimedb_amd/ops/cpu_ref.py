@@ -219,3 +219,16 @@ def dedup_mark_last(series, ts):
     keep = torch.ones(n, dtype=torch.bool)
     keep[:-1] = (series[:-1] != series[1:]) | (ts[:-1] != ts[1:])
     return keep
+
+
+def scatter_append(ts, series, fields, region_of, dst_off,
+                   dst_ts, dst_se, dst_fields):
+    """Oracle for kernels.scatter_append (K16): routed bulk memtable write."""
+    for r in range(len(dst_ts)):
+        m = region_of == r
+        if not bool(m.any()):
+            continue
+        o = dst_off[m]
+        dst_ts[r][o] = ts[m]
+        dst_se[r][o] = series[m]
+        dst_fields[r][:, o] = fields[:, m]

@@ -179,3 +179,19 @@ def series_last(sources, slot_lut, ts_lo, ts_hi, n_slots):
     none = best_src < 0
     best_ts = torch.where(none, torch.full_like(best_ts, -(1 << 63) + 1), best_ts)
     return best_ts.cpu(), best_src.cpu(), best_row.cpu()
+
+
+def scatter_append(ts, series, fields, region_of, dst_off,
+                   dst_ts, dst_se, dst_fields):
+    """K16 bulk memtable append: write row i of (ts, series, fields[nf, n])
+    into region region_of[i] at row dst_off[i]. `dst_*` are per-region
+    destination tensors (fields [nf, cap], stride = cap). One kernel replaces
+    the per-region narrow+copy_ chain on the ingest hot path."""
+    if torch.is_tensor(ts) and ts.is_cuda:
+        _require_hip().scatter_append(
+            ts.contiguous(), series.contiguous(), fields.contiguous(),
+            region_of.contiguous(), dst_off.contiguous(),
+            list(dst_ts), list(dst_se), list(dst_fields))
+        return
+    cpu_ref.scatter_append(ts, series, fields, region_of, dst_off,
+                           dst_ts, dst_se, dst_fields)

@@ -208,8 +208,10 @@ def test_bulk_ingest_gpu_matches_query():
             for _ in range(3):
                 ing.ingest_lines(w.next_batch(900))
             ex = Executor(eng)
-            r = ex.execute("SELECT count(*), sum(usage_user), max(usage_system) FROM cpu")
-            res[dev] = [r.rows[0][0], round(float(r.rows[0][1]), 6),
-                        round(float(r.rows[0][2]), 6)]
+            row = list(ex.execute(
+                "SELECT count(*), sum(usage_user), max(usage_system) FROM cpu"
+            ).rows())[0]
+            res[dev] = [int(row[0]), round(float(row[1]), 6),
+                        round(float(row[2]), 6)]
             eng.close()
     assert res["cuda"] == res["cpu"]

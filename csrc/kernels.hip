@@ -301,7 +301,7 @@ enum PromMode {
   PM_AVG = 4, PM_SUM = 5, PM_MIN = 6, PM_MAX = 7, PM_COUNT = 8, PM_LAST = 9,
   PM_IDELTA = 10, PM_IRATE = 11, PM_DERIV = 12, PM_PREDICT = 13,
   PM_RESETS = 14, PM_CHANGES = 15, PM_STDDEV = 16, PM_STDVAR = 17,
-  PM_ABSENT_OT = 18,
+  PM_ABSENT_OT = 18, PM_QUANTILE_OT = 19,
 };
 
 DEV_INLINE double prom_nan() { return __longlong_as_double(0x7FF8000000000000LL); }
@@ -426,6 +426,47 @@ __global__ void prom_range_eval_kernel(
         prev = v;
       }
       r = (double)k;
+    } else if (cnt >= 1 && mode == PM_QUANTILE_OT) {
+      // Prometheus quantile: sorted linear interpolation (promql/quantile.go);
+      // q outside [0,1] yields ∓Inf. Small windows (the common case: ≤128
+      // samples) insertion-sort in scratch; larger windows select the two
+      // needed order statistics by O(W²) rank counting — no extra storage.
+      const double q = param;
+      if (q < 0.0) {
+        r = -INFINITY;
+      } else if (q > 1.0) {
+        r = INFINITY;
+      } else {
+        const double rank = q * (double)(cnt - 1);
+        const int64_t k1 = (int64_t)rank;
+        const int64_t k2 = (k1 + 1 < cnt) ? k1 + 1 : k1;
+        double vk1 = NAN, vk2 = NAN;
+        if (cnt <= 128) {
+          double buf[128];
+          int64_t m = 0;
+          for (int64_t j = w_lo; j < w_hi; j++) {
+            const double v = vals[j];
+            int64_t p = m++;
+            while (p > 0 && buf[p - 1] > v) { buf[p] = buf[p - 1]; p--; }
+            buf[p] = v;
+          }
+          vk1 = buf[k1];
+          vk2 = buf[k2];
+        } else {
+          for (int64_t j = w_lo; j < w_hi; j++) {
+            const double x = vals[j];
+            int64_t less = 0, eq = 0;
+            for (int64_t j2 = w_lo; j2 < w_hi; j2++) {
+              const double y = vals[j2];
+              less += (y < x);
+              eq += (y == x);
+            }
+            if (less <= k1 && k1 < less + eq) vk1 = x;
+            if (less <= k2 && k2 < less + eq) vk2 = x;
+          }
+        }
+        r = vk1 + (vk2 - vk1) * (rank - (double)k1);
+      }
     }
     out[i] = r;
   }

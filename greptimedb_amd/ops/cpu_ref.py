@@ -114,7 +114,7 @@ PROM_MODES = {
     "max_over_time": 7, "count_over_time": 8, "last_over_time": 9,
     "idelta": 10, "irate": 11, "deriv": 12, "predict_linear": 13,
     "resets": 14, "changes": 15, "stddev_over_time": 16, "stdvar_over_time": 17,
-    "absent_over_time": 18,
+    "absent_over_time": 18, "quantile_over_time": 19,
 }
 
 
@@ -207,6 +207,19 @@ def prom_range_eval(ts, vals, seg_lo, seg_hi, T, t0, step_ms, range_ms,
             elif mode in (14, 15) and cnt >= 1:
                 d = np.diff(w)
                 r = float((d < 0).sum()) if mode == 14 else float((d != 0).sum())
+            elif mode == 19 and cnt:
+                # Prometheus quantile: sorted linear interpolation; q outside
+                # [0,1] yields ∓Inf (promql/quantile.go)
+                if param < 0:
+                    r = -np.inf
+                elif param > 1:
+                    r = np.inf
+                else:
+                    sw = np.sort(w)
+                    rank = param * (cnt - 1)
+                    lo_i = int(np.floor(rank))
+                    hi_i = min(lo_i + 1, cnt - 1)
+                    r = sw[lo_i] + (sw[hi_i] - sw[lo_i]) * (rank - lo_i)
             out[s, t] = r
     return torch.as_tensor(out)
 

@@ -38,6 +38,8 @@ RANGE_FUNCS = {
     "present_over_time": "count_over_time",  # then >0 → 1
     "changes": "changes", "resets": "resets",
     "absent_over_time": "absent_over_time",
+    "predict_linear": "predict_linear",
+    "quantile_over_time": "quantile_over_time",
 }
 
 ELEMENTWISE = {
@@ -165,9 +167,12 @@ class PromEvaluator:
         if f in RANGE_FUNCS:
             param = 0.0
             sel_idx = 0
-            if f == "quantile_over_time":
-                raise PlanQuery("quantile_over_time not yet supported")
-            if f == "predict_linear":
+            if f == "quantile_over_time":  # quantile_over_time(q, v[r])
+                if len(e.args) != 2 or not isinstance(e.args[0], ast.NumberLit):
+                    raise PlanQuery("quantile_over_time(scalar, v[r])")
+                param = e.args[0].value
+                sel_idx = 1
+            if f == "predict_linear":      # predict_linear(v[r], t)
                 if len(e.args) != 2 or not isinstance(e.args[1], ast.NumberLit):
                     raise PlanQuery("predict_linear(v[r], t)")
                 param = e.args[1].value

@@ -125,7 +125,8 @@ def test_prom_range_eval_matches_cpu():
     hi_t = torch.as_tensor(np.array(seg_hi, dtype=np.int64))
     T = 13
     for mode_name, mode in cpu_ref.PROM_MODES.items():
-        param = 600.0 if mode_name == "predict_linear" else 0.0
+        param = {"predict_linear": 600.0, "quantile_over_time": 0.37}.get(
+            mode_name, 0.0)
         exp = cpu_ref.prom_range_eval(ts_t, v_t, lo_t, hi_t, T, 100_000, 40_000,
                                       120_000, 5_000, param, mode)
         got = kernels.prom_range_eval(ts_t.cuda(), v_t.cuda(), lo_t.cuda(),
@@ -215,3 +216,24 @@ def test_bulk_ingest_gpu_matches_query():
                         round(float(row[2]), 6)]
             eng.close()
     assert res["cuda"] == res["cpu"]
+
+
+def test_quantile_over_time_large_window():
+    """Windows >128 samples take the O(W²) rank-selection path in the kernel
+    (no scratch buffer) — must still match the sorting oracle."""
+    rng = np.random.RandomState(9)
+    n = 700
+    ts = torch.as_tensor(np.sort(rng.randint(0, 300_000, n)).astype(np.int64))
+    v = torch.as_tensor(rng.uniform(-50, 50, n))
+    # inject duplicates to exercise the (less, eq) rank logic
+    v[::7] = 3.25
+    lo = torch.tensor([0], dtype=torch.int64)
+    hi = torch.tensor([n], dtype=torch.int64)
+    mode = cpu_ref.PROM_MODES["quantile_over_time"]
+    for q in (0.0, 0.25, 0.5, 0.95, 1.0):
+        exp = cpu_ref.prom_range_eval(ts, v, lo, hi, 3, 100_000, 100_000,
+                                      300_000, 0, q, mode)
+        got = kernels.prom_range_eval(ts.cuda(), v.cuda(), lo.cuda(), hi.cuda(),
+                                      3, 100_000, 100_000, 300_000, 0, q, mode)
+        np.testing.assert_allclose(exp.numpy(), got.cpu().numpy(), rtol=1e-12,
+                                   equal_nan=True, err_msg=f"q={q}")

@@ -136,3 +136,32 @@ def test_tql_eval(prom_env):
     r = ex.execute("TQL EVAL (120, 240, '60s') sum(rate(http_requests_total[1m]))")
     assert len(r) == 3  # 3 grid steps, one series
     assert abs(r.columns[-1][0] - 3.0) < 1e-9
+
+
+def test_quantile_over_time(prom_env):
+    _, ev = prom_env
+    # instance a: values 180,195,...,240 in (120,240] window (4 samples at 1m
+    # →(140,240]? no: [100s] range) — use 2m range: samples 135..240 step 15
+    m = ev.query_range(
+        'quantile_over_time(0.5, http_requests_total{instance="a"}[2m])',
+        240, 240, 1)
+    w = np.arange(135, 241, 15, dtype=float)   # (120, 240] at 15s scrape
+    assert abs(float(m.values[0][0]) - np.quantile(w, 0.5)) < 1e-9
+    m = ev.query_range(
+        'quantile_over_time(0.9, http_requests_total{instance="a"}[2m])',
+        240, 240, 1)
+    assert abs(float(m.values[0][0]) - np.quantile(w, 0.9)) < 1e-9
+    # q outside [0,1] → ±Inf (Prometheus semantics)
+    m = ev.query_range(
+        'quantile_over_time(1.5, http_requests_total{instance="a"}[2m])',
+        240, 240, 1)
+    assert np.isposinf(float(m.values[0][0]))
+
+
+def test_predict_linear(prom_env):
+    _, ev = prom_env
+    # series a is exactly linear with slope 1/s → predict at +600s = v(t)+600
+    m = ev.query_range(
+        'predict_linear(http_requests_total{instance="a"}[2m], 600)',
+        240, 240, 1)
+    assert abs(float(m.values[0][0]) - (240.0 + 600.0)) < 1e-6

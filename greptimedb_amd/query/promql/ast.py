@@ -21,6 +21,17 @@ class Selector:
 
 
 @dataclass
+class Subquery:
+    """expr[range:resolution] — inner expr evaluated on its own grid, the
+    results treated as samples for the enclosing range function
+    (Prometheus SubqueryExpr; step_s 0 = default resolution)."""
+    expr: object
+    range_s: float
+    step_s: float = 0.0
+    offset_s: float = 0.0
+
+
+@dataclass
 class NumberLit:
     value: float
 

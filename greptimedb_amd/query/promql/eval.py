@@ -177,10 +177,14 @@ class PromEvaluator:
                     raise PlanQuery("predict_linear(v[r], t)")
                 param = e.args[1].value
             sel = e.args[sel_idx]
-            if not isinstance(sel, ast.Selector) or sel.range_s is None:
+            if isinstance(sel, ast.Subquery):
+                m = self._eval_subquery_range(sel, RANGE_FUNCS[f], t0, step, T,
+                                              grid, param)
+            elif not isinstance(sel, ast.Selector) or sel.range_s is None:
                 raise PlanQuery(f"{f} needs a range vector argument")
-            m = self._eval_selector(sel, RANGE_FUNCS[f], t0, step, T, grid,
-                                    int(sel.range_s * 1000), param)
+            else:
+                m = self._eval_selector(sel, RANGE_FUNCS[f], t0, step, T, grid,
+                                        int(sel.range_s * 1000), param)
             if f == "present_over_time":
                 v = m.values
                 m = _matrix_map(m, torch.where(v > 0, torch.ones_like(v),
@@ -419,6 +423,52 @@ class PromEvaluator:
         if mask is None:
             return None
         return np.flatnonzero(mask)
+
+    def _eval_subquery_range(self, sub: ast.Subquery, func: str, t0, step, T,
+                             grid, param) -> PromMatrix:
+        """expr[range:res] — evaluate the inner expression on its own
+        res-aligned grid (Prometheus SubqueryExpr: start aligned UP to a
+        multiple of the resolution), then treat the resulting matrix cells
+        as samples for the enclosing range function. The window evaluation
+        reuses the prom_range_eval kernel: NaN cells are compacted out into
+        per-series ragged segments on device."""
+        range_ms = int(sub.range_s * 1000)
+        offset_ms = int(sub.offset_s * 1000)
+        res = int(sub.step_s * 1000) if sub.step_s else step  # default: query step
+        res = max(res, 1)
+        inner_start = t0 - offset_ms - range_ms
+        if inner_start % res:
+            inner_start += res - inner_start % res
+        inner_end = t0 + (T - 1) * step - offset_ms
+        T2 = int((inner_end - inner_start) // res) + 1
+        device = self.engine.config.device
+        if T2 <= 0:
+            return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
+                                              device=device), grid)
+        igrid = inner_start + np.arange(T2, dtype=np.int64) * res
+        m = self._eval(sub.expr, inner_start, res, T2, igrid)
+        if isinstance(m, PromScalar):
+            m = PromMatrix([{}], torch.full((1, T2), float(m.value),
+                                            dtype=torch.float64, device=device),
+                           igrid)
+        v = m.values
+        S = int(v.shape[0])
+        if S == 0:
+            return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
+                                              device=device), grid)
+        mask = ~torch.isnan(v)
+        counts = mask.sum(dim=1)
+        seg_hi = torch.cumsum(counts, 0)
+        seg_lo = seg_hi - counts
+        ts_row = torch.as_tensor(igrid).to(v.device)
+        ts_flat = ts_row.expand(S, T2)[mask]
+        vals_flat = v[mask]
+        mode = PROM_MODES[func]
+        out = prom_range_eval(ts_flat.contiguous(), vals_flat.contiguous(),
+                              seg_lo.contiguous(), seg_hi.contiguous(),
+                              T, t0, step, range_ms, offset_ms, param, mode)
+        return PromMatrix(m._labels, out, grid, label_cols=m.label_cols,
+                          n_series=S)
 
     def _eval_selector(self, sel: ast.Selector, func: str, t0, step, T, grid,
                        range_ms, param) -> PromMatrix:

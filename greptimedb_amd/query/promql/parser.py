@@ -170,7 +170,7 @@ class PromParser:
             if v in _AGG_OPS and self.peek()[0] == "op" and self.peek()[1] == "(" or \
                v in _AGG_OPS and self.peek() == ("id", "by") or \
                v in _AGG_OPS and self.peek() == ("id", "without"):
-                return self._aggregate(v)
+                return self._postfix(self._aggregate(v))
             if self.peek() == ("op", "("):
                 # function call
                 self.i += 1
@@ -221,14 +221,33 @@ class PromParser:
         return self._postfix(sel)
 
     def _postfix(self, e):
-        # [range] and offset
+        # [range], [range:resolution] (subquery) and offset
         while True:
             if self.eat_op("["):
                 k, v = self.next()
                 if k != "dur":
                     raise InvalidSyntax("promql: expected duration in [...]")
+                # subquery: the ':' lexes as an id token (metric names may
+                # contain colons), possibly fused with the resolution
+                nk, nv = self.peek()
+                if nk == "id" and nv.startswith(":"):
+                    self.i += 1
+                    res = 0.0
+                    rest = nv[1:]
+                    if rest:
+                        if not re.fullmatch(
+                                r"(?:\d+(?:\.\d+)?(?:ms|s|m|h|d|w|y))+", rest):
+                            raise InvalidSyntax(
+                                f"promql: bad subquery resolution {rest!r}")
+                        res = parse_duration_s(rest)
+                    elif self.peek()[0] == "dur":
+                        res = self.next()[1]
+                    self.expect_op("]")
+                    e = ast.Subquery(e, v, res)
+                    continue
                 if not isinstance(e, ast.Selector):
-                    raise InvalidSyntax("promql: range on non-selector")
+                    raise InvalidSyntax("promql: range on non-selector "
+                                        "(use [range:step] for a subquery)")
                 e.range_s = v
                 self.expect_op("]")
             elif self.peek() == ("id", "offset"):
@@ -236,7 +255,7 @@ class PromParser:
                 k, v = self.next()
                 if k != "dur":
                     raise InvalidSyntax("promql: expected duration after offset")
-                if isinstance(e, ast.Selector):
+                if isinstance(e, (ast.Selector, ast.Subquery)):
                     e.offset_s = v
                 else:
                     raise InvalidSyntax("promql: offset on non-selector")

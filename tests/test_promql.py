@@ -165,3 +165,45 @@ def test_predict_linear(prom_env):
         'predict_linear(http_requests_total{instance="a"}[2m], 600)',
         240, 240, 1)
     assert abs(float(m.values[0][0]) - (240.0 + 600.0)) < 1e-6
+
+
+def test_subquery_parse():
+    from greptimedb_amd.query.promql.parser import parse_promql
+    from greptimedb_amd.query.promql import ast as past
+    e = parse_promql("max_over_time(rate(http_requests_total[1m])[10m:30s])")
+    sub = e.args[0]
+    assert isinstance(sub, past.Subquery)
+    assert sub.range_s == 600 and sub.step_s == 30
+    e = parse_promql("avg_over_time(foo[5m:])")
+    assert isinstance(e.args[0], past.Subquery) and e.args[0].step_s == 0.0
+    e = parse_promql("avg_over_time(foo[5m:1m] offset 2m)")
+    assert e.args[0].offset_s == 120
+
+
+def test_subquery_eval(prom_env):
+    _, ev = prom_env
+    # rate of the linear counter (slope 1/s for a, 2/s for b) is constant, so
+    # any window aggregate over the subquery matrix returns the slope
+    m = ev.query_range(
+        'max_over_time(rate(http_requests_total{instance="a"}[1m])[3m:15s])',
+        300, 480, 60)
+    np.testing.assert_allclose(m.values.numpy(), 1.0, rtol=1e-12)
+    m = ev.query_range(
+        'avg_over_time(rate(http_requests_total[1m])[3m:15s])', 300, 480, 60)
+    assert m.S == 2
+    np.testing.assert_allclose(np.sort(m.values[:, 0].numpy()), [1.0, 2.0],
+                               rtol=1e-12)
+    # count_over_time counts inner evaluation points: 3m window at 15s
+    # resolution, left-open → 12 samples
+    m = ev.query_range(
+        'count_over_time(http_requests_total{instance="a"}[3m:15s])',
+        300, 300, 1)
+    assert float(m.values[0][0]) == 12.0
+
+
+def test_subquery_of_scalar_and_sum(prom_env):
+    _, ev = prom_env
+    m = ev.query_range(
+        'min_over_time(sum(http_requests_total)[2m:30s])', 240, 240, 1)
+    # sum at inner times 120..240 step 30: 3*t for t in (120,240] → min 3*150
+    assert float(m.values[0][0]) == 3 * 150.0

@@ -35,6 +35,17 @@ class Func(Expr):
 
 
 @dataclass
+class WindowFunc(Expr):
+    """func(args) OVER (PARTITION BY … ORDER BY …) — default frame
+    (RANGE UNBOUNDED PRECEDING..CURRENT ROW when ordered, whole partition
+    otherwise; ref DataFusion window exprs)."""
+    name: str
+    args: list
+    partition_by: list = field(default_factory=list)
+    order_by: list = field(default_factory=list)   # (Expr, desc)
+
+
+@dataclass
 class Star(Expr):
     pass
 

@@ -115,6 +115,13 @@ def _expr_cols(e: ast.Expr) -> set:
         elif isinstance(x, ast.Func):
             for a in x.args:
                 walk(a)
+        elif isinstance(x, ast.WindowFunc):
+            for a in x.args:
+                walk(a)
+            for a in x.partition_by:
+                walk(a)
+            for a, _d in x.order_by:
+                walk(a)
         elif isinstance(x, ast.InList):
             walk(x.expr)
             for a in x.items:
@@ -1589,15 +1596,21 @@ class Executor:
         expr_names = {n for n, _ in expr_cols}
         needed_fields = [c for c in out_cols
                          if c in field_names and c not in expr_names]
+        expr_ref: set = set()
         for _n, e in expr_cols:
             for c in _expr_cols(e):
-                if c in field_names and c not in needed_fields:
-                    needed_fields.append(c)
-                elif c in tag_names or c == ts_name:
+                expr_ref.add(c)
+                if c in field_names:
+                    if c not in needed_fields:
+                        needed_fields.append(c)
+                elif c in tag_names or c == ts_name or c in str_field_names:
                     pass
-                elif c not in field_names:
+                else:
                     raise PlanQuery(f"unknown column {c} in projection expr")
         needed_strs = [c for c in out_cols if c in str_field_names]
+        for c in expr_ref:
+            if c in str_field_names and c not in needed_strs:
+                needed_strs.append(c)
         order_cols = [e.name for e, _ in plan.order_by if isinstance(e, ast.Col)]
         for c in order_cols:
             if c in field_names and c not in needed_fields:
@@ -1667,9 +1680,12 @@ class Executor:
             parts.append((ts_t.cpu().numpy(), se_t.cpu().numpy(), region,
                           f_t.cpu().numpy(), s_cols))
 
-        # materialize host rows
+        # materialize host rows (incl. columns referenced only inside
+        # projection expressions, e.g. OVER (PARTITION BY tag ORDER BY ts))
+        avail = set(tag_names) | {ts_name} | set(field_names) | set(str_field_names)
         mat_cols = (set(out_cols) | set(order_cols) | set(needed_fields) |
-                    {ts_name} if expr_cols else set(out_cols) | set(order_cols))
+                    {ts_name} | (expr_ref & avail) if expr_cols
+                    else set(out_cols) | set(order_cols))
         col_data = {c: [] for c in mat_cols if c not in expr_names}
         for ts_h, se_h, region, f_h, s_cols in parts:
             for c in col_data:
@@ -1690,10 +1706,24 @@ class Executor:
         if self.dist is not None:
             col_data = self.dist.gather_columns(col_data)
         n = len(next(iter(col_data.values()))) if col_data else 0
+        # window functions first (they feed the projection expressions)
+        win_nodes: list = []
+        for _n2, e in expr_cols:
+            _collect_window_nodes(e, win_nodes)
+        for node in win_nodes:
+            col_data[f"__win@{id(node)}"] = _compute_window(node, col_data, n) \
+                if n else np.array([])
         # projection expressions over materialized columns (numpy)
         for name, e in expr_cols:
-            col_data[name] = np.asarray(_eval_np_expr(e, col_data), dtype=float) \
-                if n else np.array([])
+            if not n:
+                col_data[name] = np.array([])
+                continue
+            a = np.asarray(_eval_np_expr(e, col_data))
+            if a.ndim == 0:
+                a = np.full(n, float(a))
+            elif a.dtype != object and a.dtype.kind not in "US":
+                a = a.astype(np.float64, copy=False)
+            col_data[name] = a
 
         idx = np.arange(n)
         for e, desc in reversed(plan.order_by):
@@ -1945,6 +1975,8 @@ _NP_FUNCS = {"abs": np.abs, "floor": np.floor, "ceil": np.ceil,
 def _eval_np_expr(e: ast.Expr, col_data: dict):
     """Scalar expression over materialized numpy columns (raw-path
     projections like `v * 8 / 1024`)."""
+    if isinstance(e, ast.WindowFunc):
+        return col_data[f"__win@{id(e)}"]  # precomputed by _compute_window
     if isinstance(e, ast.Col):
         return np.asarray(col_data[e.name], dtype=np.float64)
     if isinstance(e, ast.Lit):
@@ -1961,6 +1993,147 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
         nd = int(e.args[1].value) if len(e.args) > 1 else 0
         return np.round(v, nd)
     raise PlanQuery(f"unsupported projection expr {e}")
+
+
+def _collect_window_nodes(e, out: list):
+    if isinstance(e, ast.WindowFunc):
+        out.append(e)
+    elif isinstance(e, ast.BinOp):
+        _collect_window_nodes(e.left, out)
+        _collect_window_nodes(e.right, out)
+    elif isinstance(e, ast.UnaryOp):
+        _collect_window_nodes(e.operand, out)
+    elif isinstance(e, ast.Func):
+        for a in e.args:
+            _collect_window_nodes(a, out)
+
+
+def _win_sort_key(a: np.ndarray, desc: bool = False) -> np.ndarray:
+    """lexsort-able key: strings factorized to dense codes, desc negated."""
+    a = np.asarray(a)
+    if a.dtype == object or a.dtype.kind in "US":
+        _, inv = np.unique(a.astype(str), return_inverse=True)
+        a = inv.astype(np.int64)
+    return -a if desc else a
+
+
+def _peer_ends(new_peer: np.ndarray, n: int) -> np.ndarray:
+    """For each (sorted) row, the index of the LAST row of its peer group."""
+    grp = np.cumsum(new_peer) - 1
+    starts = np.flatnonzero(new_peer)
+    return (np.append(starts[1:], n) - 1)[grp]
+
+
+def _compute_window(node: "ast.WindowFunc", col_data: dict, n: int) -> np.ndarray:
+    """Evaluate one window function over the materialized raw result
+    (numpy, vectorized; ref: DataFusion WindowAggExec semantics with the
+    default frame — running to the current row's peer group when ORDER BY
+    is present, whole partition otherwise). Returns values in the
+    ORIGINAL row order."""
+    if n == 0:
+        return np.array([])
+
+    def raw(e):
+        if isinstance(e, ast.Col):
+            return np.asarray(col_data[e.name])
+        return np.asarray(_eval_np_expr(e, col_data))
+
+    keys = [_win_sort_key(raw(e2), desc) for e2, desc in reversed(node.order_by)]
+    part_codes = [_win_sort_key(raw(e2)) for e2 in node.partition_by]
+    perm = np.lexsort(tuple(keys) + tuple(part_codes)) if (keys or part_codes) \
+        else np.arange(n)
+    pos = np.arange(n)
+    new_part = np.zeros(n, dtype=bool)
+    new_part[0] = True
+    for k in part_codes:
+        ks = k[perm]
+        new_part[1:] |= ks[1:] != ks[:-1]
+    starts = np.flatnonzero(new_part)
+    seg_id = np.cumsum(new_part) - 1
+    start_of = starts[seg_id]
+    new_peer = new_part.copy()
+    for k in keys:
+        ks = k[perm]
+        new_peer[1:] |= ks[1:] != ks[:-1]
+
+    name = "avg" if node.name == "mean" else node.name
+    if name == "row_number":
+        out_sorted = (pos - start_of + 1).astype(np.float64)
+    elif name in ("rank", "dense_rank"):
+        if name == "rank":
+            peer_start = np.maximum.accumulate(np.where(new_peer, pos, 0))
+            out_sorted = (peer_start - start_of + 1).astype(np.float64)
+        else:
+            d = np.cumsum(new_peer)
+            out_sorted = (d - d[start_of] + 1).astype(np.float64)
+    elif name in ("lag", "lead"):
+        x = raw(node.args[0])[perm]
+        k = int(_eval_const(node.args[1])) if len(node.args) > 1 else 1
+        default = _eval_const(node.args[2]) if len(node.args) > 2 else None
+        if name == "lead":
+            k = -k
+        is_obj = x.dtype == object
+        out_sorted = np.empty(n, dtype=object if is_obj else np.float64)
+        out_sorted[:] = default if is_obj else \
+            (np.nan if default is None else float(default))
+        if k != 0:
+            src = np.clip(pos - k, 0, n - 1)
+            ok = ((pos - k >= 0) & (pos - k < n)) & (seg_id[src] == seg_id)
+            out_sorted[ok] = x[src[ok]]
+        else:
+            out_sorted = x if is_obj else x.astype(np.float64)
+    elif name in ("first_value", "last_value"):
+        x = raw(node.args[0])[perm]
+        if name == "first_value":
+            out_sorted = x[start_of]
+        elif not node.order_by:               # whole partition → partition end
+            out_sorted = x[np.append(starts[1:], n)[seg_id] - 1]
+        else:                                 # default frame → peer-group end
+            out_sorted = x[_peer_ends(new_peer, n)]
+    elif name in ("sum", "avg", "min", "max", "count"):
+        if not node.args or isinstance(node.args[0], ast.Star):
+            x = np.ones(n, dtype=np.float64)
+        else:
+            x = np.asarray(raw(node.args[0]), dtype=np.float64)[perm]
+        valid = ~np.isnan(x)
+        if not node.order_by:  # whole-partition aggregate
+            c = np.add.reduceat(valid.astype(np.float64), starts)
+            if name in ("sum", "avg", "count"):
+                s = np.add.reduceat(np.where(valid, x, 0.0), starts)
+                v = c if name == "count" else \
+                    (s if name == "sum" else s / np.where(c > 0, c, np.nan))
+            else:
+                xx = np.where(valid, x, np.inf if name == "min" else -np.inf)
+                red = np.minimum if name == "min" else np.maximum
+                v = np.where(c > 0, red.reduceat(xx, starts), np.nan)
+            if name != "count":
+                v = np.where(c > 0, v, np.nan)
+            out_sorted = v[seg_id]
+        else:  # running to the end of the current peer group
+            pe = _peer_ends(new_peer, n)
+            cc = np.cumsum(valid.astype(np.float64))
+            base_c = np.where(start_of > 0, cc[np.maximum(start_of - 1, 0)], 0.0)
+            c = cc[pe] - base_c
+            if name in ("sum", "avg", "count"):
+                cs = np.cumsum(np.where(valid, x, 0.0))
+                base_s = np.where(start_of > 0, cs[np.maximum(start_of - 1, 0)], 0.0)
+                s = cs[pe] - base_s
+                v = c if name == "count" else \
+                    (s if name == "sum" else s / np.where(c > 0, c, np.nan))
+            else:
+                xx = np.where(valid, x, np.inf if name == "min" else -np.inf)
+                accf = np.minimum.accumulate if name == "min" else np.maximum.accumulate
+                acc = np.empty(n, dtype=np.float64)
+                for s0, s1 in zip(starts, np.append(starts[1:], n)):
+                    acc[s0:s1] = accf(xx[s0:s1])
+                v = acc[pe]
+            out_sorted = np.where(c > 0, v, np.nan) if name != "count" else v
+    else:
+        raise PlanQuery(f"unsupported window function {node.name}")
+
+    out = np.empty(n, dtype=out_sorted.dtype)
+    out[perm] = out_sorted
+    return out
 
 
 def _eval_const(e: ast.Expr):

@@ -571,9 +571,47 @@ class Parser:
                         if not self.eat_op(","):
                             break
                 self.expect_op(")")
+                if self.eat_kw("over"):
+                    return self._window_spec(low, args)
                 return ast.Func(low, args, distinct)
             return ast.Col(t.value)
         raise InvalidSyntax(f"unexpected token {t}")
+
+    def _window_spec(self, func: str, args: list) -> "ast.WindowFunc":
+        """OVER (PARTITION BY … ORDER BY … [ROWS|RANGE frame]) — frames are
+        accepted only in their default forms (ref: DataFusion WindowExpr)."""
+        self.expect_op("(")
+        part, order = [], []
+        if self.eat_kw("partition"):
+            self.expect_kw("by")
+            while True:
+                part.append(self.parse_expr())
+                if not self.eat_op(","):
+                    break
+        if self.eat_kw("order"):
+            self.expect_kw("by")
+            while True:
+                e2 = self.parse_expr()
+                desc = False
+                if self.eat_kw("desc"):
+                    desc = True
+                else:
+                    self.eat_kw("asc")
+                order.append((e2, desc))
+                if not self.eat_op(","):
+                    break
+        if self.at_kw("rows", "range"):
+            # accept the frames equivalent to the defaults:
+            #   ROWS/RANGE BETWEEN UNBOUNDED PRECEDING AND CURRENT ROW
+            self.next()
+            self.expect_kw("between")
+            self.expect_kw("unbounded")
+            self.expect_kw("preceding")
+            self.expect_kw("and")
+            self.expect_kw("current")
+            self.expect_kw("row")
+        self.expect_op(")")
+        return ast.WindowFunc(func, args, part, order)
 
 
 def parse_sql(sql: str):

@@ -41,8 +41,21 @@ class LogStore:
             return self.engine.create_table(schema, append_mode=True,
                                             if_not_exists=True)
 
+    def ingest_with_pipeline(self, table: str, entries: list[dict],
+                             pipeline) -> int:
+        """Run a compiled YAML Pipeline over the entries, then ingest each
+        table-suffix group with the pipeline's tag/fulltext/timestamp hints
+        (ref: src/pipeline dispatcher + tablesuffix routing)."""
+        total = 0
+        for suffix, rows in pipeline.run(entries).items():
+            total += self.ingest(table + suffix, rows,
+                                 tag_keys=pipeline.tag_keys,
+                                 ts_key=pipeline.ts_key,
+                                 fulltext_keys=pipeline.fulltext_keys)
+        return total
+
     def ingest(self, table: str, entries: list[dict], tag_keys: list[str] | None = None,
-               ts_key: str = "timestamp") -> int:
+               ts_key: str = "timestamp", fulltext_keys: list[str] | None = None) -> int:
         """greptime_identity-style ingestion: each entry's keys become
         columns. `tag_keys` values form the series; `ts_key` (ms epoch or
         ns > 1e15) is the time index (now() if absent)."""
@@ -100,8 +113,17 @@ class LogStore:
                 r.ensure_fields(new_num)
             new_str = [k for k in str_vals
                        if k not in st.regions[0].str_field_names]
-            for r in (st.regions if new_str else []):
-                r.ensure_str_fields(new_str)
+            if new_str:
+                # fulltext_keys None → every string column fulltext (identity
+                # pipeline default); else only the transform-declared ones
+                ft = new_str if fulltext_keys is None else \
+                    [k for k in new_str if k in fulltext_keys]
+                plain = [k for k in new_str if k not in ft]
+                for r in st.regions:
+                    if ft:
+                        r.ensure_str_fields(ft, fulltext=True)
+                    if plain:
+                        r.ensure_str_fields(plain, fulltext=False)
         fnames = st.regions[0].field_names
         for ridx, rows in region_rows.items():
             rows_a = np.array(rows)

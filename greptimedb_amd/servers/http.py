@@ -37,6 +37,8 @@ class ServerContext:
         self.tracestore = TraceStore(engine)
         self.executor = Executor(engine, dist=dist)
         self.prom = PromEvaluator(engine, dist=dist)
+        from greptimedb_amd.pipeline import PipelineStore
+        self.pipelines = PipelineStore(engine.config.data_dir)
         self.started = time.time()
 
 
@@ -194,16 +196,64 @@ def build_app(ctx: ServerContext) -> FastAPI:
 
     @app.post("/v1/events/logs")
     async def events_logs(request: Request, table: str = Query("logs"),
-                          tag_keys: str = Query(""), ts_key: str = Query("timestamp")):
+                          tag_keys: str = Query(""), ts_key: str = Query("timestamp"),
+                          pipeline_name: str = Query("")):
         import json as _json
         body = await request.body()
         entries = _json.loads(body)
         if isinstance(entries, dict):
             entries = [entries]
-        tags = [t for t in tag_keys.split(",") if t]
-        n = ctx.logstore.ingest(table, entries, tag_keys=tags, ts_key=ts_key)
+        if pipeline_name and pipeline_name != "greptime_identity":
+            p = ctx.pipelines.get(pipeline_name)
+            n = ctx.logstore.ingest_with_pipeline(table, entries, p)
+        else:
+            tags = [t for t in tag_keys.split(",") if t]
+            n = ctx.logstore.ingest(table, entries, tag_keys=tags, ts_key=ts_key)
         metrics_mod.counter("log_events").inc(n)
         return {"rows": n}
+
+    # YAML ETL pipelines (ref src/pipeline: create/view/delete by name).
+    # _dryrun registered FIRST — the {name} route would capture it otherwise.
+    @app.post("/v1/events/pipelines/_dryrun")
+    @app.post("/v1/pipelines/_dryrun")
+    async def dryrun_pipeline(request: Request, pipeline_name: str = Query("")):
+        import json as _json
+        body = _json.loads(await request.body())
+        if pipeline_name:
+            p = ctx.pipelines.get(pipeline_name)
+        else:
+            from greptimedb_amd.pipeline import Pipeline
+            import yaml as _yaml
+            p = Pipeline(body.get("pipeline")
+                         if isinstance(body.get("pipeline"), dict)
+                         else _yaml.safe_load(body.get("pipeline", "")))
+        data = body.get("data", body if isinstance(body, list) else [])
+        if isinstance(data, dict):
+            data = [data]
+        out = []
+        for suffix, rows in p.run(data).items():
+            for r in rows:
+                out.append({"table_suffix": suffix, "row": r})
+        return {"rows": out}
+
+    @app.post("/v1/events/pipelines/{name}")
+    @app.post("/v1/pipelines/{name}")
+    async def create_pipeline(name: str, request: Request):
+        body = (await request.body()).decode()
+        ctx.pipelines.put(name, body)
+        return {"name": name, "status": "created"}
+
+    @app.get("/v1/events/pipelines/{name}")
+    @app.get("/v1/pipelines/{name}")
+    async def get_pipeline(name: str):
+        p = ctx.pipelines.get(name)
+        return {"name": name, "pipeline": p.spec, "version": p.version}
+
+    @app.delete("/v1/events/pipelines/{name}")
+    @app.delete("/v1/pipelines/{name}")
+    async def delete_pipeline(name: str):
+        ctx.pipelines.delete(name)
+        return {"name": name, "status": "deleted"}
 
     @app.post("/v1/loki/api/v1/push")
     async def loki_push(request: Request):

@@ -140,3 +140,42 @@ def test_remote_read_roundtrip(client):
     assert body.count(b"rr_metric") == 2  # two series carry the name label
     import struct as _s
     assert _s.pack("<d", 9.0) in body and _s.pack("<d", 1.5) in body
+
+
+def test_pipeline_crud_and_ingest(client):
+    yml = """
+version: 2
+processors:
+  - dissect:
+      fields: [message]
+      patterns:
+        - '%{level} %{msg}'
+  - letter: {fields: [level], method: lower}
+transform:
+  - fields: [level]
+    type: string
+    index: tag
+  - fields: [msg]
+    type: string
+    index: fulltext
+"""
+    r = client.post("/v1/events/pipelines/applog", content=yml)
+    assert r.status_code == 200 and r.json()["status"] == "created"
+    r = client.get("/v1/events/pipelines/applog")
+    assert r.json()["version"] == 2
+    # dryrun shows transformed rows without writing
+    r = client.post("/v1/events/pipelines/_dryrun?pipeline_name=applog",
+                    json={"data": [{"message": "ERROR disk full"}]})
+    row = r.json()["rows"][0]["row"]
+    assert row["level"] == "error" and row["msg"] == "disk full"
+    # ingest through the pipeline
+    r = client.post("/v1/events/logs?table=app&pipeline_name=applog",
+                    json=[{"message": "WARN low memory", "timestamp": 1000},
+                          {"message": "ERROR disk full", "timestamp": 2000}])
+    assert r.json()["rows"] == 2
+    r = client.post("/v1/sql", data={"sql":
+                    "SELECT level, msg FROM app ORDER BY ts"})
+    rows = r.json()["output"][0]["records"]["rows"]
+    assert rows == [["warn", "low memory"], ["error", "disk full"]]
+    r = client.delete("/v1/events/pipelines/applog")
+    assert r.json()["status"] == "deleted"

@@ -173,7 +173,7 @@ transform:
                     json=[{"message": "WARN low memory", "timestamp": 1000},
                           {"message": "ERROR disk full", "timestamp": 2000}])
     assert r.json()["rows"] == 2
-    r = client.post("/v1/sql", data={"sql":
+    r = client.post("/v1/sql", params={"sql":
                     "SELECT level, msg FROM app ORDER BY ts"})
     rows = r.json()["output"][0]["records"]["rows"]
     assert rows == [["warn", "low memory"], ["error", "disk full"]]

@@ -46,6 +46,15 @@ class WindowFunc(Expr):
 
 
 @dataclass
+class RangeAgg(Expr):
+    """agg(col) RANGE '10s' [FILL x] — sliding window [t, t+range) per
+    ALIGN step (ref: src/query/src/range_select/plan.rs:947 window math)."""
+    func: "Func"
+    range_ms: int
+    fill: object = None        # None | "null" | "prev" | "linear" | float
+
+
+@dataclass
 class Star(Expr):
     pass
 
@@ -104,6 +113,11 @@ class Select:
     order_by: list[tuple[Expr, bool]] = field(default_factory=list)  # (expr, desc)
     limit: int | None = None
     offset: int | None = None
+    # RANGE-query ALIGN clause (ref range_select): step, origin, by-cols, fill
+    align_ms: int | None = None
+    align_to: object = None            # None (epoch 0) | "now" | int ms | str
+    align_by: list[str] | None = None
+    align_fill: object = None
 
 
 @dataclass

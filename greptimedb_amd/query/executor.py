@@ -611,6 +611,8 @@ class Executor:
         if sel.joins:
             return self._exec_join(sel)
         plan = self._plan_select(sel)
+        if any(_has_range_agg(e) for e, _a in sel.projections):
+            return self._exec_range_select(sel, plan)
         if plan.aggs:
             return self._exec_aggregate(sel, plan)
         knn = self._try_vector_knn(sel, plan)
@@ -1567,6 +1569,197 @@ class Executor:
 
     # ------------------------------------------------------ raw path
 
+    # ------------------------------------------------------ RANGE queries
+
+    _RANGE_MODES = {"min": 6, "max": 7, "sum": 5, "avg": 4, "mean": 4,
+                    "count": 8, "last_value": 9}
+
+    def _exec_range_select(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
+        """SQL RANGE query: `agg(x) RANGE '10s' … ALIGN '5s' [TO …] [BY (…)]
+        [FILL …]` — sliding window [t, t+range) per aligned step t
+        (ref src/query/src/range_select/plan.rs:947: align_ts <= ts <
+        align_ts + range). MI355X path: per-(group, field) NaN-compacted
+        sample streams sorted by (slot, ts) feed the prom_range_eval window
+        kernel with te = t + range - 1 (integer-ms shift turns the kernel's
+        (tb, te] window into [t, t+range))."""
+        from greptimedb_amd.ops import filter_series_time, prom_range_eval
+        st = plan.table
+        device = self.engine.config.device
+        schema = st.schema
+        ts_name = schema.time_index.name
+        if sel.align_ms is None:
+            raise PlanQuery("RANGE queries need an ALIGN clause")
+        align = sel.align_ms
+
+        # collect RangeAgg nodes from the projections
+        nodes: list[ast.RangeAgg] = []
+        for e, _a in sel.projections:
+            _collect_range_aggs(e, nodes)
+        calls = []   # (node, func, field|None)
+        for nd in nodes:
+            f = nd.func
+            fname = "avg" if f.name == "mean" else f.name
+            if fname not in self._RANGE_MODES:
+                raise PlanQuery(f"unsupported RANGE aggregate {f.name}")
+            if len(f.args) == 1 and isinstance(f.args[0], ast.Star):
+                arg = ts_name      # count(*): every row has a timestamp
+            elif len(f.args) == 1 and isinstance(f.args[0], ast.Col):
+                arg = f.args[0].name
+            else:
+                raise PlanQuery(f"RANGE aggregate needs a column arg: {f.name}")
+            calls.append((nd, fname, arg))
+        fields = sorted({arg for _nd, _f, arg in calls})
+
+        gt = sel.align_by if sel.align_by is not None else \
+            [c.name for c in schema.tag_columns]
+        group_keys, region_luts = self._build_group_luts(st, plan, gt)
+        n_slots = max(len(group_keys), 1)
+
+        ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
+        ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
+
+        # gather per-field (slot, ts)-sorted sample streams (NaN compacted)
+        parts = []          # (ts, slots, {field: vals})
+        for region, lut in zip(st.regions, region_luts):
+            lut_t = torch.as_tensor(lut, device=device)
+            chunks = []
+            for src in region.scan_sources(ts_lo, ts_hi):
+                mask = filter_series_time(src.ts, src.series, lut_t, ts_lo, ts_hi)
+                if plan.residual is not None:
+                    mask &= self._eval_mask(plan.residual, src, region, device)
+                idx = mask.nonzero(as_tuple=True)[0]
+                if idx.numel() == 0:
+                    continue
+                fv = {}
+                for fn in fields:
+                    if fn == ts_name:
+                        fv[fn] = src.ts[idx].double()
+                        continue
+                    p = src.field_pos.get(fn)
+                    fv[fn] = src.fields[p][idx] if p is not None else \
+                        torch.full((idx.numel(),), float("nan"),
+                                   dtype=torch.float64, device=device)
+                chunks.append((src.ts[idx], src.series[idx], fv))
+            if not chunks:
+                continue
+            ts_t = torch.cat([c[0] for c in chunks])
+            se_t = torch.cat([c[1] for c in chunks])
+            fv = {fn: torch.cat([c[2][fn] for c in chunks]) for fn in fields}
+            if not st.append_mode:   # last-wins dedup on (series, ts)
+                o2 = torch.argsort(ts_t, stable=True)
+                p2 = o2[torch.argsort(se_t[o2], stable=True)]
+                ts_t, se_t = ts_t[p2], se_t[p2]
+                fv = {fn: v[p2] for fn, v in fv.items()}
+                keep = dedup_mark_last(se_t.int().contiguous(), ts_t.contiguous())
+                kidx = keep.nonzero(as_tuple=True)[0]
+                if kidx.numel() != ts_t.numel():
+                    ts_t, se_t = ts_t[kidx], se_t[kidx]
+                    fv = {fn: v[kidx] for fn, v in fv.items()}
+            sl_t = lut_t[se_t.long()]
+            ok = sl_t >= 0
+            if not bool(ok.all()):
+                ts_t, sl_t = ts_t[ok], sl_t[ok]
+                fv = {fn: v[ok] for fn, v in fv.items()}
+            parts.append((ts_t, sl_t, fv))
+
+        names_out, cols_out, kinds = [], [], []
+        if not parts:
+            for e, a in sel.projections:
+                names_out.append(a or _expr_name(e))
+                cols_out.append(np.array([]))
+            return QueryResult(names_out, cols_out)
+
+        ts_all = torch.cat([p[0] for p in parts])
+        sl_all = torch.cat([p[1] for p in parts])
+        o = torch.argsort(ts_all, stable=True)
+        perm = o[torch.argsort(sl_all[o], stable=True)]
+        ts_all, sl_all = ts_all[perm], sl_all[perm]
+        fvals = {fn: torch.cat([p[2][fn] for p in parts])[perm] for fn in fields}
+
+        # align grid over the data span (TO: epoch 0 default | NOW | literal)
+        data_lo = int(ts_all.min())
+        data_hi = int(ts_all.max())
+        to = 0
+        if sel.align_to == "now":
+            import time as _time
+            to = int(_time.time() * 1000) % align
+        elif sel.align_to not in (None, "calendar"):
+            from greptimedb_amd.utils.timeutil import parse_ts_ms
+            to = int(parse_ts_ms(sel.align_to)) % align
+        max_range = max(nd.range_ms for nd, _f, _a in calls)
+        t_first = -(-(data_lo - max_range + 1 - to) // align) * align + to
+        t_last = ((data_hi - to) // align) * align + to
+        T = int((t_last - t_first) // align) + 1
+        if T > 4 << 20:
+            raise PlanQuery(f"RANGE grid too large: {T} steps")
+
+        planes: dict[int, np.ndarray] = {}
+        for nd, fname, arg in calls:
+            v = fvals[arg]
+            valid = ~torch.isnan(v)
+            if bool(valid.all()):
+                tsf, slf, vf = ts_all, sl_all, v
+            else:
+                kidx = valid.nonzero(as_tuple=True)[0]
+                tsf, slf, vf = ts_all[kidx], sl_all[kidx], v[kidx]
+            counts = torch.bincount(slf.long(), minlength=n_slots)
+            seg_hi = torch.cumsum(counts, 0)
+            seg_lo = seg_hi - counts
+            # te = t + range - 1 ⇒ kernel window (t-1, t+range-1] = [t, t+range)
+            out = prom_range_eval(
+                tsf.contiguous(), vf.contiguous().double(), seg_lo.contiguous(),
+                seg_hi.contiguous(), T, t_first + nd.range_ms - 1, align,
+                nd.range_ms, 0, 0.0, self._RANGE_MODES[fname])
+            plane = out.cpu().numpy()
+            fill = nd.fill if nd.fill is not None else sel.align_fill
+            planes[id(nd)] = _apply_fill(plane, fill)
+        if self.dist is not None:
+            raise PlanQuery("distributed RANGE queries not yet supported")
+
+        grid = t_first + np.arange(T, dtype=np.int64) * align
+        # drop rows where every RANGE column is NaN (reference emits only
+        # align_ts slots that hold data unless FILL materializes them)
+        any_fill = any((nd.fill if nd.fill is not None else sel.align_fill)
+                       is not None for nd, _f, _a in calls)
+        present = np.zeros((n_slots, T), dtype=bool)
+        for p in planes.values():
+            present |= ~np.isnan(p)
+        if any_fill:
+            rows_mask = present.any(axis=1)[:, None] & np.ones((1, T), dtype=bool)
+        else:
+            rows_mask = present
+        slot_idx, t_idx = np.nonzero(rows_mask)
+        # group-key columns
+        keys_by_slot = [None] * n_slots
+        for k, s in group_keys.items():
+            keys_by_slot[s] = k
+        col_data: dict[str, np.ndarray] = {ts_name: grid[t_idx]}
+        for gi, g in enumerate(gt):
+            col_data[g] = np.array([keys_by_slot[s][gi] if keys_by_slot[s]
+                                    else None for s in slot_idx], dtype=object)
+        for nid, p in planes.items():
+            col_data[f"__range@{nid}"] = p[slot_idx, t_idx]
+
+        # projections (RangeAgg nodes resolve via their precomputed planes)
+        for e, a in sel.projections:
+            name = a or _expr_name(e)
+            names_out.append(name)
+            kinds.append("ts" if isinstance(e, ast.Col) and e.name == ts_name else "")
+            if isinstance(e, ast.Col):
+                if e.name not in col_data:
+                    raise PlanQuery(f"RANGE projection column {e.name} must be "
+                                    f"the time index or an ALIGN BY column")
+                cols_out.append(col_data[e.name])
+            elif isinstance(e, ast.RangeAgg):
+                cols_out.append(col_data[f"__range@{id(e)}"])
+            elif isinstance(e, ast.Func) and e.name == "date_trunc":
+                cols_out.append(col_data[ts_name])
+            else:
+                cols_out.append(np.asarray(_eval_np_expr(e, col_data)))
+
+        r = QueryResult(names_out, cols_out, kinds)
+        return _apply_order_limit(r, sel, names_out, default_order=[ts_name] + gt)
+
     def _exec_raw(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
         st = plan.table
         schema = st.schema
@@ -1977,6 +2170,8 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
     projections like `v * 8 / 1024`)."""
     if isinstance(e, ast.WindowFunc):
         return col_data[f"__win@{id(e)}"]  # precomputed by _compute_window
+    if isinstance(e, ast.RangeAgg):
+        return col_data[f"__range@{id(e)}"]  # precomputed range plane slice
     if isinstance(e, ast.Col):
         return np.asarray(col_data[e.name], dtype=np.float64)
     if isinstance(e, ast.Lit):
@@ -1993,6 +2188,82 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
         nd = int(e.args[1].value) if len(e.args) > 1 else 0
         return np.round(v, nd)
     raise PlanQuery(f"unsupported projection expr {e}")
+
+
+def _has_range_agg(e) -> bool:
+    found: list = []
+    _collect_range_aggs(e, found)
+    return bool(found)
+
+
+def _collect_range_aggs(e, out: list):
+    if isinstance(e, ast.RangeAgg):
+        out.append(e)
+    elif isinstance(e, ast.BinOp):
+        _collect_range_aggs(e.left, out)
+        _collect_range_aggs(e.right, out)
+    elif isinstance(e, ast.UnaryOp):
+        _collect_range_aggs(e.operand, out)
+    elif isinstance(e, ast.Func):
+        for a in e.args:
+            _collect_range_aggs(a, out)
+
+
+def _apply_fill(plane: np.ndarray, fill) -> np.ndarray:
+    """FILL NULL|PREV|LINEAR|<const> over an [S, T] range plane."""
+    if fill is None or fill == "null":
+        return plane
+    if isinstance(fill, (int, float)) and not isinstance(fill, bool):
+        return np.where(np.isnan(plane), float(fill), plane)
+    S, T = plane.shape
+    if fill == "prev":
+        idx = np.where(~np.isnan(plane), np.arange(T)[None, :], 0)
+        idx = np.maximum.accumulate(idx, axis=1)
+        out = plane[np.arange(S)[:, None], idx]
+        # positions before the first sample stay NaN
+        first = np.argmax(~np.isnan(plane), axis=1)
+        none = ~np.isnan(plane).any(axis=1)
+        mask_before = np.arange(T)[None, :] < first[:, None]
+        out[mask_before | none[:, None]] = np.nan
+        return out
+    if fill == "linear":
+        out = plane.copy()
+        x = np.arange(T, dtype=np.float64)
+        for s in range(S):
+            row = out[s]
+            good = ~np.isnan(row)
+            if good.sum() >= 2:
+                out[s] = np.interp(x, x[good], row[good])
+                # do not extrapolate beyond known points
+                lo, hi = np.flatnonzero(good)[[0, -1]]
+                out[s, :lo] = np.nan
+                out[s, hi + 1:] = np.nan
+        return out
+    raise PlanQuery(f"unsupported FILL {fill!r}")
+
+
+def _apply_order_limit(r: QueryResult, sel: ast.Select, names: list[str],
+                       default_order: list[str]) -> QueryResult:
+    """ORDER BY named output columns + LIMIT/OFFSET over a QueryResult."""
+    n = len(r.columns[0]) if r.columns else 0
+    idx = np.arange(n)
+    order = sel.order_by or [(ast.Col(c), False) for c in default_order
+                             if c in names]
+    for e, desc in reversed(order):
+        if not isinstance(e, ast.Col) or e.name not in names:
+            continue
+        a = np.asarray(r.columns[names.index(e.name)])[idx]
+        if a.dtype == object:
+            a = a.astype(str)
+        o = np.argsort(a, kind="stable")
+        if desc:
+            o = o[::-1]
+        idx = idx[o]
+    if sel.offset:
+        idx = idx[sel.offset:]
+    if sel.limit is not None:
+        idx = idx[: sel.limit]
+    return QueryResult(names, [np.asarray(c)[idx] for c in r.columns], r.kinds)
 
 
 def _collect_window_nodes(e, out: list):
@@ -2150,6 +2421,8 @@ def _eval_const(e: ast.Expr):
 
 
 def _expr_name(e: ast.Expr) -> str:
+    if isinstance(e, ast.RangeAgg):
+        return f"{_expr_name(e.func)} RANGE {e.range_ms}ms"
     if isinstance(e, ast.Col):
         return e.name
     if isinstance(e, ast.Func):

@@ -14,6 +14,15 @@ import re
 from greptimedb_amd.query import ast
 from greptimedb_amd.utils.errors import InvalidSyntax
 
+
+def _duration_ms(text: str) -> int:
+    """'10s' / '5m' / '1h30m' → milliseconds (RANGE/ALIGN durations)."""
+    from greptimedb_amd.query.promql.parser import parse_duration_s
+    ms = int(parse_duration_s(text) * 1000)
+    if ms <= 0:
+        raise InvalidSyntax(f"bad duration {text!r}")
+    return ms
+
 _TOKEN_RE = re.compile(r"""
     \s+
   | (?P<comment>--[^\n]*)
@@ -240,7 +249,8 @@ class Parser:
                     alias = self.next().value
                 elif self.peek() is not None and self.peek().kind == "id" and \
                         self.peek().value.lower() not in (
-                            "from", "where", "group", "order", "limit", "having", "offset"):
+                            "from", "where", "group", "order", "limit", "having",
+                            "offset", "align", "fill", "range"):
                     alias = self.next().value
                 projections.append((e, alias))
             if not self.eat_op(","):
@@ -255,7 +265,7 @@ class Parser:
             elif self.peek() is not None and self.peek().kind == "id" and \
                     self.peek().value.lower() not in (
                         "where", "group", "order", "limit", "having", "offset",
-                        "join", "inner", "left", "on"):
+                        "join", "inner", "left", "on", "align"):
                 table_alias = self.next().value
             while self.at_kw("join", "inner", "left"):
                 kind = "inner"
@@ -288,6 +298,25 @@ class Parser:
         having = None
         if self.eat_kw("having"):
             having = self.parse_expr()
+        # ALIGN '5s' [TO NOW|ts] [BY (cols)] [FILL v] — RANGE-query clause
+        align_ms = align_to = align_by = align_fill = None
+        if self.eat_kw("align"):
+            align_ms = _duration_ms(str(self.next().value))
+            if self.eat_kw("to"):
+                if self.eat_kw("now"):
+                    align_to = "now"
+                elif self.eat_kw("calendar"):
+                    align_to = "calendar"
+                else:
+                    align_to = self.next().value
+            if self.eat_kw("by"):
+                self.expect_op("(")
+                align_by = []
+                while not self.eat_op(")"):
+                    align_by.append(str(self.next().value))
+                    self.eat_op(",")
+            if self.eat_kw("fill"):
+                align_fill = self._parse_fill()
         order_by = []
         if self.eat_kw("order"):
             self.expect_kw("by")
@@ -308,7 +337,24 @@ class Parser:
             offset = int(self.next().value)
         return ast.Select(projections, table, table_alias=table_alias, joins=joins,
                           where=where, group_by=group_by, having=having,
-                          order_by=order_by, limit=limit, offset=offset)
+                          order_by=order_by, limit=limit, offset=offset,
+                          align_ms=align_ms, align_to=align_to,
+                          align_by=align_by, align_fill=align_fill)
+
+    def _parse_fill(self):
+        t = self.next()
+        if t.kind == "id" and t.value.lower() in ("null", "prev", "linear"):
+            return t.value.lower()
+        if t.kind == "str" and str(t.value).lower() in ("null", "prev", "linear"):
+            return str(t.value).lower()
+        neg = False
+        if t.kind == "op" and t.value == "-":
+            neg, t = True, self.next()
+        try:
+            v = float(t.value)
+        except (TypeError, ValueError):
+            raise InvalidSyntax(f"bad FILL value {t.value!r}")
+        return -v if neg else v
 
     def parse_create(self):
         self.expect_kw("create")
@@ -573,7 +619,15 @@ class Parser:
                 self.expect_op(")")
                 if self.eat_kw("over"):
                     return self._window_spec(low, args)
-                return ast.Func(low, args, distinct)
+                fn = ast.Func(low, args, distinct)
+                # agg(col) RANGE '10s' [FILL v] binds tighter than operators
+                if self.at_kw("range") and self.i + 1 < len(self.toks) and \
+                        self.toks[self.i + 1].kind == "str":
+                    self.next()
+                    dur = str(self.next().value)
+                    fill = self._parse_fill() if self.eat_kw("fill") else None
+                    return ast.RangeAgg(fn, _duration_ms(dur), fill)
+                return fn
             return ast.Col(t.value)
         raise InvalidSyntax(f"unexpected token {t}")
 

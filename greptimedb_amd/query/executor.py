@@ -2165,6 +2165,160 @@ _NP_FUNCS = {"abs": np.abs, "floor": np.floor, "ceil": np.ceil,
              "log10": np.log10, "exp": np.exp}
 
 
+# ---------------- object-typed scalar UDFs (json / geo / string / vector) ---
+# ref: src/common/function/src/scalars/{json,geo,vector} — the practical
+# subset; each takes raw (object) numpy columns and is vectorized in python.
+
+def _json_path_norm(path: str) -> str:
+    p = str(path)
+    if p.startswith("$"):
+        p = p[1:]
+    if p and not p.startswith((".", "[")):
+        p = "." + p
+    return p
+
+
+def _json_get(col, path, conv):
+    import json as _json
+    from greptimedb_amd.pipeline.engine import _json_path_get
+    p = _json_path_norm(path)
+    out = np.empty(len(col), dtype=object)
+    for i, s in enumerate(col):
+        v = None
+        if s is not None:
+            try:
+                v = _json_path_get(_json.loads(s) if isinstance(s, str) else s,
+                                   "$" + p)
+            except (ValueError, TypeError):
+                v = None
+        out[i] = conv(v)
+    return out
+
+
+def _to_num(arr):
+    return np.array([np.nan if v is None else float(v) for v in arr])
+
+
+def _haversine_m(lat1, lng1, lat2, lng2):
+    r = 6371008.8
+    p1, p2 = np.radians(lat1), np.radians(lat2)
+    dp = p2 - p1
+    dl = np.radians(lng2) - np.radians(lng1)
+    a = np.sin(dp / 2) ** 2 + np.cos(p1) * np.cos(p2) * np.sin(dl / 2) ** 2
+    return 2 * r * np.arcsin(np.sqrt(a))
+
+
+_GEOHASH32 = "0123456789bcdefghjkmnpqrstuvwxyz"
+
+
+def _geohash(lat, lng, precision):
+    out = np.empty(len(np.atleast_1d(lat)), dtype=object)
+    lat = np.atleast_1d(lat)
+    lng = np.atleast_1d(lng)
+    for i in range(len(out)):
+        la, lo = float(lat[i]), float(lng[i % len(lng)] if len(lng) > 1 else lng[0])
+        lat_r, lng_r = [-90.0, 90.0], [-180.0, 180.0]
+        bits = []
+        even = True
+        while len(bits) < precision * 5:
+            rng = lng_r if even else lat_r
+            v = lo if even else la
+            mid = (rng[0] + rng[1]) / 2
+            if v >= mid:
+                bits.append(1)
+                rng[0] = mid
+            else:
+                bits.append(0)
+                rng[1] = mid
+            even = not even
+        s = ""
+        for c in range(precision):
+            idx = 0
+            for b in bits[c * 5:(c + 1) * 5]:
+                idx = (idx << 1) | b
+            s += _GEOHASH32[idx]
+        out[i] = s
+    return out
+
+
+def _vec_decode(col):
+    return [None if v is None else np.frombuffer(
+        v if isinstance(v, (bytes, bytearray)) else str(v).encode("latin1"),
+        dtype=np.float32) for v in col]
+
+
+def _str_col(a, n=None):
+    arr = np.atleast_1d(np.asarray(a, dtype=object))
+    if n is not None and len(arr) == 1 and n > 1:
+        arr = np.full(n, arr[0], dtype=object)
+    return arr
+
+
+_OBJ_FUNCS = {
+    "json_get_string": lambda a: _json_get(
+        _str_col(a[0]), a[1], lambda v: None if v is None else str(v)),
+    "json_get_int": lambda a: _to_num(_json_get(
+        _str_col(a[0]), a[1],
+        lambda v: None if v is None or isinstance(v, (dict, list)) else int(float(v)))),
+    "json_get_float": lambda a: _to_num(_json_get(
+        _str_col(a[0]), a[1],
+        lambda v: None if v is None or isinstance(v, (dict, list)) else float(v))),
+    "json_get_bool": lambda a: _to_num(_json_get(
+        _str_col(a[0]), a[1],
+        lambda v: None if not isinstance(v, bool) else float(v))),
+    "json_path_exists": lambda a: _to_num(_json_get(
+        _str_col(a[0]), a[1], lambda v: float(v is not None))),
+    "st_distance": lambda a: _haversine_m(_to_num_b(a[0]), _to_num_b(a[1]),
+                                          _to_num_b(a[2]), _to_num_b(a[3])),
+    "geohash": lambda a: _geohash(_to_num_b(a[0]), _to_num_b(a[1]),
+                                  int(np.atleast_1d(a[2])[0])),
+    "upper": lambda a: np.array([None if v is None else str(v).upper()
+                                 for v in _str_col(a[0])], dtype=object),
+    "lower": lambda a: np.array([None if v is None else str(v).lower()
+                                 for v in _str_col(a[0])], dtype=object),
+    "trim": lambda a: np.array([None if v is None else str(v).strip()
+                                for v in _str_col(a[0])], dtype=object),
+    "length": lambda a: np.array([np.nan if v is None else float(len(str(v)))
+                                  for v in _str_col(a[0])]),
+    "char_length": lambda a: _OBJ_FUNCS["length"](a),
+    "replace": lambda a: np.array(
+        [None if v is None else str(v).replace(str(np.atleast_1d(a[1])[0]),
+                                               str(np.atleast_1d(a[2])[0]))
+         for v in _str_col(a[0])], dtype=object),
+    "concat": lambda a: _concat_cols(a),
+    "vec_dim": lambda a: np.array(
+        [np.nan if v is None else float(len(v)) for v in _vec_decode(_str_col(a[0]))]),
+    "vec_to_string": lambda a: np.array(
+        ["[" + ",".join(f"{x:g}" for x in v) + "]" if v is not None else None
+         for v in _vec_decode(_str_col(a[0]))], dtype=object),
+}
+
+
+def _to_num_b(a):
+    arr = np.atleast_1d(np.asarray(a))
+    if arr.dtype == object:
+        return _to_num(arr)
+    return arr.astype(np.float64)
+
+
+def _concat_cols(args):
+    cols = [np.atleast_1d(np.asarray(a, dtype=object)) for a in args]
+    n = max(len(c) for c in cols)
+    cols = [np.full(n, c[0], dtype=object) if len(c) == 1 and n > 1 else c
+            for c in cols]
+    return np.array(["".join("" if c[i] is None else str(c[i]) for c in cols)
+                     for i in range(n)], dtype=object)
+
+
+def _np_raw(e: ast.Expr, col_data: dict):
+    """Like _eval_np_expr but keeps object (string) columns unconverted."""
+    if isinstance(e, ast.Col):
+        return np.asarray(col_data[e.name])
+    if isinstance(e, ast.Lit):
+        return e.value
+    return _eval_np_expr(e, col_data)
+
+
 def _eval_np_expr(e: ast.Expr, col_data: dict):
     """Scalar expression over materialized numpy columns (raw-path
     projections like `v * 8 / 1024`)."""
@@ -2181,6 +2335,8 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
                          _eval_np_expr(e.right, col_data))
     if isinstance(e, ast.UnaryOp) and e.op == "-":
         return -_eval_np_expr(e.operand, col_data)
+    if isinstance(e, ast.Func) and e.name in _OBJ_FUNCS:
+        return _OBJ_FUNCS[e.name]([_np_raw(a, col_data) for a in e.args])
     if isinstance(e, ast.Func) and e.name in _NP_FUNCS:
         return _NP_FUNCS[e.name](_eval_np_expr(e.args[0], col_data))
     if isinstance(e, ast.Func) and e.name == "round":

@@ -127,3 +127,39 @@ def test_inner_join(tmp_engine):
     r = ex.execute("SELECT x.host FROM m x JOIN meta y ON x.host = y.host "
                    "WHERE x.cpu > 60 ORDER BY x.host")
     assert list(r.columns[0]) == ["b"]
+
+
+def test_json_functions(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE j (h STRING, ts TIMESTAMP TIME INDEX, "
+               "doc JSON, PRIMARY KEY (h))")
+    ex.execute('''INSERT INTO j (h, ts, doc) VALUES
+        ('a', 1000, '{"user": {"name": "kim", "age": 7}, "ok": true}'),
+        ('b', 2000, '{"user": {"name": "lee"}, "vals": [1, 2.5]}')''')
+    r = ex.execute("SELECT json_get_string(doc, 'user.name') AS n, "
+                   "json_get_int(doc, 'user.age') AS a, "
+                   "json_get_bool(doc, 'ok') AS o, "
+                   "json_get_float(doc, '$.vals[1]') AS v, "
+                   "json_path_exists(doc, 'vals') AS e FROM j ORDER BY ts")
+    rows = [tuple(t) for t in r.rows()]
+    assert rows[0][0] == "kim" and rows[0][1] == 7.0 and rows[0][2] == 1.0
+    assert np.isnan(rows[0][3]) and rows[0][4] == 0.0
+    assert rows[1][0] == "lee" and rows[1][3] == 2.5 and rows[1][4] == 1.0
+
+
+def test_string_geo_functions(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE g (h STRING, ts TIMESTAMP TIME INDEX, "
+               "lat DOUBLE, lng DOUBLE, PRIMARY KEY (h))")
+    ex.execute("INSERT INTO g (h, ts, lat, lng) VALUES "
+               "('sf', 1000, 37.7749, -122.4194), ('ny', 2000, 40.7128, -74.0060)")
+    r = ex.execute("SELECT upper(h) AS u, length(h) AS l, "
+                   "geohash(lat, lng, 6) AS gh FROM g ORDER BY ts")
+    rows = [tuple(t) for t in r.rows()]
+    assert rows[0][0] == "SF" and rows[0][1] == 2.0
+    assert rows[0][2] == "9q8yyk"       # well-known SF geohash
+    assert rows[1][2].startswith("dr5r")  # NYC
+    r = ex.execute("SELECT st_distance(lat, lng, 40.7128, -74.0060) AS d "
+                   "FROM g ORDER BY ts")
+    d = [t[0] for t in r.rows()]
+    assert abs(d[0] - 4_130_000) < 10_000 and d[1] < 1.0  # SF→NY ≈ 4130 km

@@ -46,6 +46,14 @@ class WindowFunc(Expr):
 
 
 @dataclass
+class ScalarSubquery(Expr):
+    """(SELECT …) in an expression — uncorrelated; resolved to a literal
+    (scalar) or value list (IN (...)) before planning the outer query."""
+    select: object
+    many: bool = False
+
+
+@dataclass
 class RangeAgg(Expr):
     """agg(col) RANGE '10s' [FILL x] — sliding window [t, t+range) per
     ALIGN step (ref: src/query/src/range_select/plan.rs:947 window math)."""

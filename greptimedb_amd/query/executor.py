@@ -596,7 +596,51 @@ class Executor:
 
     # ---------------------------------------------------------- SELECT
 
+    def _resolve_subqueries(self, x):
+        """Execute uncorrelated (SELECT …) nodes and splice their results in
+        as literals (scalar) / literal lists (IN) before planning."""
+        if x is None:
+            return None
+        if isinstance(x, ast.ScalarSubquery):
+            r = self._exec_select(x.select)
+            col = list(r.columns[0]) if r.columns else []
+
+            def lit(v):
+                if v is None:
+                    return ast.Lit(None)
+                if isinstance(v, (np.floating, float)):
+                    return ast.Lit(float(v))
+                if isinstance(v, (np.integer, int)):
+                    return ast.Lit(int(v))
+                return ast.Lit(str(v))
+            if x.many:
+                return [lit(v) for v in col]
+            return lit(col[0]) if len(col) else ast.Lit(None)
+        if isinstance(x, ast.BinOp):
+            x.left = self._resolve_subqueries(x.left)
+            x.right = self._resolve_subqueries(x.right)
+        elif isinstance(x, ast.UnaryOp):
+            x.operand = self._resolve_subqueries(x.operand)
+        elif isinstance(x, ast.Func):
+            x.args = [self._resolve_subqueries(a) for a in x.args]
+        elif isinstance(x, ast.Between):
+            x.low = self._resolve_subqueries(x.low)
+            x.high = self._resolve_subqueries(x.high)
+        elif isinstance(x, ast.InList):
+            items = []
+            for it in x.items:
+                got = self._resolve_subqueries(it)
+                items.extend(got if isinstance(got, list) else [got])
+            x.items = items
+        return x
+
     def _exec_select(self, sel: ast.Select) -> QueryResult:
+        if _has_subquery(sel.where) or _has_subquery(sel.having) or \
+                any(_has_subquery(e) for e, _a in sel.projections):
+            sel.projections = [(self._resolve_subqueries(e), a)
+                               for e, a in sel.projections]
+            sel.where = self._resolve_subqueries(sel.where)
+            sel.having = self._resolve_subqueries(sel.having)
         if sel.table is None:
             # constant select
             names, cols = [], []
@@ -2344,6 +2388,24 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
         nd = int(e.args[1].value) if len(e.args) > 1 else 0
         return np.round(v, nd)
     raise PlanQuery(f"unsupported projection expr {e}")
+
+
+def _has_subquery(e) -> bool:
+    if e is None:
+        return False
+    if isinstance(e, ast.ScalarSubquery):
+        return True
+    if isinstance(e, ast.BinOp):
+        return _has_subquery(e.left) or _has_subquery(e.right)
+    if isinstance(e, ast.UnaryOp):
+        return _has_subquery(e.operand)
+    if isinstance(e, ast.Func):
+        return any(_has_subquery(a) for a in e.args)
+    if isinstance(e, ast.Between):
+        return _has_subquery(e.low) or _has_subquery(e.high)
+    if isinstance(e, ast.InList):
+        return any(_has_subquery(a) for a in e.items)
+    return False
 
 
 def _has_range_agg(e) -> bool:

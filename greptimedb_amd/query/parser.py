@@ -547,6 +547,11 @@ class Parser:
             if opname == "in":
                 self.next()
                 self.expect_op("(")
+                if self.at_kw("select"):   # IN (SELECT ...) subquery
+                    sub = self.parse_select()
+                    self.expect_op(")")
+                    left = ast.InList(left, [ast.ScalarSubquery(sub, many=True)])
+                    continue
                 items = []
                 while True:
                     items.append(self.parse_expr())
@@ -584,6 +589,10 @@ class Parser:
         if t.kind == "str":
             return ast.Lit(t.value)
         if t.kind == "op" and t.value == "(":
+            if self.at_kw("select"):   # scalar subquery (uncorrelated)
+                sub = self.parse_select()
+                self.expect_op(")")
+                return ast.ScalarSubquery(sub)
             e = self.parse_expr()
             self.expect_op(")")
             return e

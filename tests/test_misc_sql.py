@@ -163,3 +163,21 @@ def test_string_geo_functions(tmp_engine):
                    "FROM g ORDER BY ts")
     d = [t[0] for t in r.rows()]
     assert abs(d[0] - 4_130_000) < 10_000 and d[1] < 1.0  # SF→NY ≈ 4130 km
+
+
+def test_scalar_subqueries(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE s (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h))")
+    ex.execute("INSERT INTO s (h, ts, v) VALUES ('a', 1000, 1.0), "
+               "('b', 2000, 5.0), ('c', 3000, 9.0)")
+    # scalar comparison subquery
+    r = ex.execute("SELECT h FROM s WHERE v > (SELECT avg(v) FROM s) ORDER BY h")
+    assert list(r.columns[0]) == ["c"]
+    # IN (SELECT ...) subquery
+    r = ex.execute("SELECT h, v FROM s WHERE h IN "
+                   "(SELECT h FROM s WHERE v >= 5.0) ORDER BY h")
+    assert list(r.columns[0]) == ["b", "c"]
+    # scalar subquery in projection
+    r = ex.execute("SELECT h, v - (SELECT min(v) FROM s) AS d FROM s ORDER BY h")
+    assert [t[1] for t in r.rows()] == [0.0, 4.0, 8.0]

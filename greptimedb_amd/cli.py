@@ -29,7 +29,19 @@ def main(argv=None):
     start.add_argument("--user-provider", default=None,
                        help="static_user_provider:file:<path>")
     start.add_argument("--config", default=None, help="TOML config file")
+    # `cli meta snapshot save/restore` (ref src/cli metadata snapshot tools)
+    cli = sub.add_parser("cli")
+    cli_sub = cli.add_subparsers(dest="cmd", required=True)
+    meta = cli_sub.add_parser("meta")
+    meta_sub = meta.add_subparsers(dest="action", required=True)
+    for action in ("save", "restore"):
+        p = meta_sub.add_parser(action)
+        p.add_argument("--data-dir", required=True)
+        p.add_argument("--file", required=True, help="snapshot tar path")
     args = ap.parse_args(argv)
+
+    if args.role == "cli":
+        return _meta_snapshot(args)
 
     import asyncio
 
@@ -71,6 +83,40 @@ def main(argv=None):
         await asyncio.gather(uv.serve(), my.serve_forever(), pg.serve_forever())
 
     asyncio.run(serve())
+
+
+def _meta_snapshot(args) -> int:
+    """Metadata-only snapshot: catalog + manifests + series logs + pipelines
+    (no SSTs/WAL — those are data; ref `greptime cli meta snapshot`)."""
+    import tarfile
+
+    def is_meta(path: str) -> bool:
+        name = os.path.basename(path)
+        return (name == "catalog.json" or name.endswith(".pipeline") or
+                name == "series.log" or "manifest" in path or
+                os.sep + "pipelines" + os.sep in path + os.sep)
+
+    if args.action == "save":
+        count = 0
+        with tarfile.open(args.file, "w:gz") as tar:
+            for root, _dirs, files in os.walk(args.data_dir):
+                for f in files:
+                    p = os.path.join(root, f)
+                    if is_meta(p):
+                        tar.add(p, arcname=os.path.relpath(p, args.data_dir))
+                        count += 1
+        print(f"meta snapshot: {count} files -> {args.file}", flush=True)
+    else:
+        os.makedirs(args.data_dir, exist_ok=True)
+        with tarfile.open(args.file, "r:gz") as tar:
+            for m in tar.getmembers():
+                # refuse path escapes
+                if m.name.startswith(("/", "..")) or ".." in m.name.split("/"):
+                    raise ValueError(f"unsafe member {m.name}")
+            tar.extractall(args.data_dir)
+            n = len(tar.getmembers())
+        print(f"meta snapshot: restored {n} files -> {args.data_dir}", flush=True)
+    return 0
 
 
 if __name__ == "__main__":

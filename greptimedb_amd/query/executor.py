@@ -149,9 +149,12 @@ class Executor:
 
     def execute(self, sql: str) -> QueryResult:
         import time as _time
+        from greptimedb_amd.utils.tracing import tracer
         stmt = parse_sql(sql)
         t0 = _time.perf_counter()
-        r = self.execute_stmt(stmt)
+        with tracer.span("sql.execute", statement=sql[:200],
+                         stmt_type=type(stmt).__name__):
+            r = self.execute_stmt(stmt)
         dt = (_time.perf_counter() - t0) * 1000
         if dt >= self.SLOW_QUERY_MS:
             # slow-query log (reference: common/frontend slow query events)
@@ -427,6 +430,23 @@ class Executor:
             migrate_region(self.engine, str(a.args[0]), int(a.args[1]),
                            str(a.args[2]))
             return QueryResult(["result"], [[1]])
+        if f == "enable_tracing":
+            from greptimedb_amd.utils.tracing import tracer
+            tracer.enabled = bool(int(a.args[0])) if a.args else True
+            if not tracer.enabled:
+                tracer.drop()
+            return QueryResult(["result"], [[int(tracer.enabled)]])
+        if f == "flush_tracing":
+            # export buffered internal spans into our own trace table via
+            # the OTLP path (self-hosted observability)
+            from greptimedb_amd.engine.tracestore import TraceStore
+            from greptimedb_amd.utils.tracing import tracer
+            ts_store = getattr(self.engine, "_internal_tracestore", None)
+            if ts_store is None:
+                ts_store = TraceStore(self.engine)
+                self.engine._internal_tracestore = ts_store
+            n = tracer.export_to(ts_store)
+            return QueryResult(["result"], [[n]])
         raise PlanQuery(f"unknown admin function {a.func}")
 
     def _exec_explain(self, e: ast.Explain) -> QueryResult:

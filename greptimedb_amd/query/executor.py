@@ -430,6 +430,43 @@ class Executor:
             migrate_region(self.engine, str(a.args[0]), int(a.args[1]),
                            str(a.args[2]))
             return QueryResult(["result"], [[1]])
+        if f == "build_vector_index":
+            # ADMIN build_vector_index(table, col[, nlist[, nprobe]]) —
+            # IVF-flat per SST source (ref: vector index per SST file)
+            import math
+            from greptimedb_amd.vector import build_ivf
+            table, col = str(a.args[0]), str(a.args[1])
+            nlist_arg = int(a.args[2]) if len(a.args) > 2 else 0
+            nprobe = int(a.args[3]) if len(a.args) > 3 else 16
+            st = self.engine.table(table)
+            device = self.engine.config.device
+            built = 0
+            for region in st.regions:
+                for b in region.sst_cache.values():
+                    colv = b.str_cols.get(col)
+                    if colv is None or not len(colv):
+                        continue
+                    vt = getattr(b, f"_vec_{col}", None)
+                    if vt is None:
+                        blob = b"".join(v if v is not None else b"" for v in colv)
+                        flat = np.frombuffer(blob, dtype=np.float32)
+                        lens = np.array([0 if v is None else len(v) // 4
+                                         for v in colv])
+                        D = int(lens.max()) if len(lens) else 0
+                        if D == 0:
+                            continue
+                        full = np.full((len(colv), D), np.nan, dtype=np.float32)
+                        full[lens > 0] = flat.reshape(-1, D)
+                        vt = torch.as_tensor(full).to(device)
+                        setattr(b, f"_vec_{col}", vt)
+                    nlist = nlist_arg or max(8, int(4 * math.sqrt(vt.shape[0])))
+                    setattr(b, f"_ivf_{col}", build_ivf(vt, nlist))
+                    built += 1
+                vi = getattr(region, "vector_index", None)
+                if vi is None:
+                    vi = region.vector_index = {}
+                vi[col] = {"nlist": nlist_arg, "nprobe": nprobe}
+            return QueryResult(["result"], [[built]])
         if f == "enable_tracing":
             from greptimedb_amd.utils.tracing import tracer
             tracer.enabled = bool(int(a.args[0])) if a.args else True
@@ -927,18 +964,33 @@ class Executor:
                 if plan.residual is not None:
                     mask &= self._eval_mask(plan.residual, src, region, device)
                 qf = q.float()
+                # IVF-flat probe when ADMIN build_vector_index attached one
+                # to this (SST) source; memtable sources stay brute-force
+                ivf = getattr(cache_holder, f"_ivf_{vcol}", None) \
+                    if cache_holder is not None else None
+                rows_sub = None
+                vt_d = vt
+                if ivf is not None:
+                    from greptimedb_amd.vector import ivf_candidates
+                    nprobe = getattr(region, "vector_index", {}).get(
+                        vcol, {}).get("nprobe", 16)
+                    rows_sub = ivf_candidates(ivf, qf, nprobe)
+                    vt_d = vt[rows_sub]
                 if mode == "l2sq":
-                    d = ((vt - qf[None, :]) ** 2).sum(dim=1)
+                    d = ((vt_d - qf[None, :]) ** 2).sum(dim=1)
                 elif mode == "cos":
-                    d = 1.0 - (vt @ qf) / (vt.norm(dim=1) * qf.norm() + 1e-30)
+                    d = 1.0 - (vt_d @ qf) / (vt_d.norm(dim=1) * qf.norm() + 1e-30)
                 else:
-                    d = -(vt @ qf)  # dot: negate so smaller = better
-                bad = ~mask | torch.isnan(d)
+                    d = -(vt_d @ qf)  # dot: negate so smaller = better
+                m = mask if rows_sub is None else mask[rows_sub]
+                bad = ~m | torch.isnan(d)
                 d = torch.where(bad, torch.full_like(d, float("inf")), d)
                 kk = min(k, d.numel())
                 if kk == 0:
                     continue
                 vals, idx = torch.topk(d, kk, largest=False)
+                if rows_sub is not None:
+                    idx = rows_sub[idx]
                 vh = vals.cpu().numpy()
                 ih = idx.cpu().numpy()
                 for dist, row in zip(vh, ih):

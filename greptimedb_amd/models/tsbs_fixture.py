@@ -158,4 +158,15 @@ def tsbs_queries(scale: int, hours: int, rng: np.random.RandomState | None = Non
     q["lastpoint"] = (
         "SELECT hostname, last_value(usage_user) FROM cpu GROUP BY hostname "
         "ORDER BY hostname")
+    # GreptimeDB RANGE-query shapes (no TSBS reference number): sliding
+    # 1h windows aligned every 10m — the range_select plan's signature load
+    lo, hi = rand_window(12)
+    q["range-sliding-8"] = (
+        f"SELECT ts, hostname, avg(usage_user) RANGE '1h' AS a FROM cpu "
+        f"WHERE hostname IN ({hosts(8)}) AND ts >= {lo} AND ts < {hi} "
+        f"ALIGN '10m' ORDER BY hostname, ts LIMIT 100")
+    q["range-sliding-all"] = (
+        f"SELECT ts, hostname, max(usage_user) RANGE '1h' AS m FROM cpu "
+        f"WHERE ts >= {lo} AND ts < {hi} ALIGN '10m' "
+        f"ORDER BY hostname, ts LIMIT 100")
     return q

@@ -166,6 +166,32 @@ class DistContext:
             dist.all_reduce(mx2, op=dist.ReduceOp.MAX)
         return merged, cnt2, s2, sq2, mn2, mx2
 
+    def merge_planes(self, keys: list, planes: list):
+        """Generic partial-plane merge (RANGE queries): `planes` is a list
+        of ([G, T] tensor, op) with op ∈ {sum, min, max}. Group keys are
+        unified across ranks (all_gather_object), each plane scattered into
+        the global slot order and all-reduced. Returns (merged_keys, outs)."""
+        all_keys: list = [None] * self.world
+        dist.all_gather_object(all_keys, keys)
+        merged = sorted({k for ks in all_keys for k in ks})
+        gmap = {k: i for i, k in enumerate(merged)}
+        G = max(len(merged), 1)
+        T = next((p.shape[1] for p, _o in planes if p is not None), 1)
+        dev = self.coll_device
+        fills = {"sum": 0.0, "min": float("inf"), "max": float("-inf")}
+        ops = {"sum": dist.ReduceOp.SUM, "min": dist.ReduceOp.MIN,
+               "max": dist.ReduceOp.MAX}
+        outs = []
+        idx = torch.as_tensor([gmap[k] for k in keys], device=dev) if keys \
+            else None
+        for p, op in planes:
+            out = torch.full((G, T), fills[op], dtype=torch.float64, device=dev)
+            if p is not None and idx is not None:
+                out[idx] = p.to(dev)
+            dist.all_reduce(out, op=ops[op])
+            outs.append(out)
+        return merged, outs
+
     def gather_matrix(self, labels: list, values):
         """Gather per-rank series matrices (labels + [S, T]) on all ranks."""
         payload = (labels, values.cpu().numpy())

@@ -1779,22 +1779,31 @@ class Executor:
             parts.append((ts_t, sl_t, fv))
 
         names_out, cols_out, kinds = [], [], []
-        if not parts:
+        dist_ctx = self.dist
+        if parts:
+            ts_all = torch.cat([p[0] for p in parts])
+            sl_all = torch.cat([p[1] for p in parts])
+            o = torch.argsort(ts_all, stable=True)
+            perm = o[torch.argsort(sl_all[o], stable=True)]
+            ts_all, sl_all = ts_all[perm], sl_all[perm]
+            fvals = {fn: torch.cat([p[2][fn] for p in parts])[perm]
+                     for fn in fields}
+            data_lo, data_hi = int(ts_all.min()), int(ts_all.max())
+        else:
+            ts_all = torch.zeros(0, dtype=torch.int64, device=device)
+            sl_all = torch.zeros(0, dtype=torch.int64, device=device)
+            fvals = {fn: torch.zeros(0, dtype=torch.float64, device=device)
+                     for fn in fields}
+            data_lo = data_hi = None
+        if dist_ctx is not None:
+            data_lo, data_hi = dist_ctx.minmax_ts(data_lo, data_hi)
+        if data_lo is None:
             for e, a in sel.projections:
                 names_out.append(a or _expr_name(e))
                 cols_out.append(np.array([]))
             return QueryResult(names_out, cols_out)
 
-        ts_all = torch.cat([p[0] for p in parts])
-        sl_all = torch.cat([p[1] for p in parts])
-        o = torch.argsort(ts_all, stable=True)
-        perm = o[torch.argsort(sl_all[o], stable=True)]
-        ts_all, sl_all = ts_all[perm], sl_all[perm]
-        fvals = {fn: torch.cat([p[2][fn] for p in parts])[perm] for fn in fields}
-
-        # align grid over the data span (TO: epoch 0 default | NOW | literal)
-        data_lo = int(ts_all.min())
-        data_hi = int(ts_all.max())
+        # align grid over the GLOBAL data span (TO: epoch 0 | NOW | literal)
         to = 0
         if sel.align_to == "now":
             import time as _time
@@ -1809,8 +1818,8 @@ class Executor:
         if T > 4 << 20:
             raise PlanQuery(f"RANGE grid too large: {T} steps")
 
-        planes: dict[int, np.ndarray] = {}
-        for nd, fname, arg in calls:
+        def kernel_plane(fname, rng_ms, arg):
+            """One [n_slots, T] partial plane on this rank's samples."""
             v = fvals[arg]
             valid = ~torch.isnan(v)
             if bool(valid.all()):
@@ -1822,15 +1831,63 @@ class Executor:
             seg_hi = torch.cumsum(counts, 0)
             seg_lo = seg_hi - counts
             # te = t + range - 1 ⇒ kernel window (t-1, t+range-1] = [t, t+range)
-            out = prom_range_eval(
+            return prom_range_eval(
                 tsf.contiguous(), vf.contiguous().double(), seg_lo.contiguous(),
-                seg_hi.contiguous(), T, t_first + nd.range_ms - 1, align,
-                nd.range_ms, 0, 0.0, self._RANGE_MODES[fname])
-            plane = out.cpu().numpy()
+                seg_hi.contiguous(), T, t_first + rng_ms - 1, align,
+                rng_ms, 0, 0.0, self._RANGE_MODES[fname])
+
+        planes: dict[int, np.ndarray] = {}
+        if dist_ctx is None:
+            for nd, fname, arg in calls:
+                planes[id(nd)] = kernel_plane(fname, nd.range_ms, arg).cpu().numpy()
+        else:
+            # distributed: per-rank PRIMITIVE planes (sum/count/min/max),
+            # group-unified + all-reduced, then finalized (avg = Σs/Σc).
+            keys_sorted = [k for k, _s in sorted(group_keys.items(),
+                                                 key=lambda kv: kv[1])]
+            prim_specs: list = []       # (plane tensor [G, T], reduce op)
+            call_prims: list = []       # per call: (kind, idx...) into outs
+            for nd, fname, arg in calls:
+                if fname == "last_value":
+                    raise PlanQuery("distributed RANGE last_value "
+                                    "not yet supported")
+                G = len(keys_sorted)
+                if fname in ("sum", "avg", "count"):
+                    s_p = torch.nan_to_num(
+                        kernel_plane("sum", nd.range_ms, arg), nan=0.0)[:G]
+                    c_p = torch.nan_to_num(
+                        kernel_plane("count", nd.range_ms, arg), nan=0.0)[:G]
+                    call_prims.append((fname, len(prim_specs),
+                                       len(prim_specs) + 1))
+                    prim_specs += [(s_p, "sum"), (c_p, "sum")]
+                else:   # min / max
+                    p = kernel_plane(fname, nd.range_ms, arg)[:G]
+                    fillv = float("inf") if fname == "min" else float("-inf")
+                    p = torch.nan_to_num(p, nan=fillv, posinf=None, neginf=None)
+                    call_prims.append((fname, len(prim_specs)))
+                    prim_specs += [(p, fname)]
+            merged_keys, outs = dist_ctx.merge_planes(keys_sorted, prim_specs)
+            group_keys = {k: i for i, k in enumerate(merged_keys)}
+            n_slots = max(len(group_keys), 1)
+            for (nd, fname, arg), prim in zip(calls, call_prims):
+                if prim[0] in ("sum", "avg", "count"):
+                    s = outs[prim[1]].cpu().numpy()
+                    c = outs[prim[2]].cpu().numpy()
+                    if fname == "count":
+                        plane = np.where(c > 0, c, np.nan)
+                    elif fname == "sum":
+                        plane = np.where(c > 0, s, np.nan)
+                    else:
+                        plane = np.where(c > 0, s / np.where(c > 0, c, 1), np.nan)
+                else:
+                    p = outs[prim[1]].cpu().numpy()
+                    plane = np.where(np.isfinite(p), p, np.nan)
+                if plane.shape[0] == 0:
+                    plane = np.full((1, T), np.nan)
+                planes[id(nd)] = plane
+        for nd, _fname, _arg in calls:
             fill = nd.fill if nd.fill is not None else sel.align_fill
-            planes[id(nd)] = _apply_fill(plane, fill)
-        if self.dist is not None:
-            raise PlanQuery("distributed RANGE queries not yet supported")
+            planes[id(nd)] = _apply_fill(planes[id(nd)], fill)
 
         grid = t_first + np.arange(T, dtype=np.int64) * align
         # drop rows where every RANGE column is NaN (reference emits only
@@ -1851,8 +1908,9 @@ class Executor:
             keys_by_slot[s] = k
         col_data: dict[str, np.ndarray] = {ts_name: grid[t_idx]}
         for gi, g in enumerate(gt):
-            col_data[g] = np.array([keys_by_slot[s][gi] if keys_by_slot[s]
-                                    else None for s in slot_idx], dtype=object)
+            per_slot = np.array([k[gi] if k else None for k in keys_by_slot],
+                                dtype=object)
+            col_data[g] = per_slot[slot_idx]
         for nid, p in planes.items():
             col_data[f"__range@{nid}"] = p[slot_idx, t_idx]
 

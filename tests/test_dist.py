@@ -48,7 +48,16 @@ from greptimedb_amd.query.promql.eval import PromEvaluator
 ev = PromEvaluator(eng, dist=DistContext(device="cpu"))
 m = ev.query_range('sum({__field__="usage_user", __name__="cpu"})',
                    1451606450, 1451606450, 1)
+# distributed RANGE query: partial planes merged over gloo
+r6 = ex.execute("SELECT ts, hostname, avg(usage_user) RANGE '1m' AS a, "
+                "max(usage_user) RANGE '1m' AS mx, count(usage_user) RANGE '1m' AS c "
+                "FROM cpu ALIGN '30s' ORDER BY hostname, ts LIMIT 10000")
+range_hosts = sorted(set(r6.columns[1]))
 out = {
+    "range_rows": len(r6),
+    "range_hosts": len(range_hosts),
+    "range_sum_a": round(float(np.nansum(np.asarray(r6.columns[2], dtype=float))), 4),
+    "range_max": round(float(np.nanmax(np.asarray(r6.columns[3], dtype=float))), 4),
     "count": int(r1.columns[0][0]),
     "hosts": list(r2.columns[0]),
     "avgs": [float(x) for x in r2.columns[1]],
@@ -95,3 +104,6 @@ def test_two_rank_query_combine(tmp_path):
     assert results[0]["raw"] >= 0
     assert results[0]["lastpoint_hosts"] == 20
     assert results[0]["prom_sum"] != 0.0
+    assert results[0]["range_hosts"] == 20   # all ranks' hosts in the plane
+    assert results[0]["range_rows"] > 40
+    assert results[0]["range_sum_a"] != 0.0

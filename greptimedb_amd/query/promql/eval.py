@@ -48,6 +48,39 @@ ELEMENTWISE = {
     "log10": torch.log10, "sqrt": torch.sqrt, "sgn": torch.sgn,
 }
 
+_TIME_FUNCS = {"minute", "hour", "day_of_week", "day_of_month",
+               "day_of_year", "days_in_month", "month", "year"}
+
+
+def _calendar_apply(f: str, secs: np.ndarray) -> np.ndarray:
+    """UTC calendar component of epoch-seconds values (NaN preserved).
+    Matches Prometheus: day_of_week 0=Sunday; 1-based day/month."""
+    nanmask = np.isnan(secs)
+    x = np.where(nanmask, 0.0, secs).astype(np.int64)
+    if f == "minute":
+        out = (x // 60) % 60
+    elif f == "hour":
+        out = (x // 3600) % 24
+    elif f == "day_of_week":
+        out = (x // 86400 + 4) % 7           # 1970-01-01 was a Thursday
+    else:
+        d = x.astype("datetime64[s]")
+        Y = d.astype("datetime64[Y]")
+        M = d.astype("datetime64[M]")
+        D = d.astype("datetime64[D]")
+        if f == "year":
+            out = Y.astype(np.int64) + 1970
+        elif f == "month":
+            out = (M - Y).astype(np.int64) + 1
+        elif f == "day_of_month":
+            out = (D - M.astype("datetime64[D]")).astype(np.int64) + 1
+        elif f == "day_of_year":
+            out = (D - Y.astype("datetime64[D]")).astype(np.int64) + 1
+        else:  # days_in_month
+            out = ((M + 1).astype("datetime64[D]") -
+                   M.astype("datetime64[D]")).astype(np.int64)
+    return np.where(nanmask, np.nan, out.astype(np.float64))
+
 
 class PromMatrix:
     """Evaluated vector: per-series labels + [S, T] value matrix (NaN =
@@ -276,6 +309,31 @@ class PromEvaluator:
             return PromMatrix(out, m.values, grid)
         if f == "histogram_quantile":
             return self._histogram_quantile(e, t0, step, T, grid)
+        if f in ("sort", "sort_desc"):
+            m = self._eval(e.args[0], t0, step, T, grid)
+            if isinstance(m, PromScalar) or m.S <= 1:
+                return m
+            v = m.values[:, -1].cpu().numpy()
+            key = np.where(np.isnan(v), np.inf if f == "sort" else -np.inf, v)
+            order = np.argsort(-key if f == "sort_desc" else key, kind="stable")
+            labels = [m.labels[i] for i in order]
+            return PromMatrix(labels, m.values[torch.as_tensor(order.copy())],
+                              grid)
+        if f in _TIME_FUNCS:
+            # minute(v=vector(time())) family — UTC calendar components
+            if e.args:
+                m = self._eval(e.args[0], t0, step, T, grid)
+                if isinstance(m, PromScalar):
+                    m = PromMatrix([{}], torch.full((1, T), float(m.value),
+                                                    dtype=torch.float64), grid)
+            else:
+                m = PromMatrix([{}], torch.as_tensor(
+                    grid[None, :] / 1000.0, dtype=torch.float64), grid)
+            v = m.values.cpu().numpy()
+            out = _calendar_apply(f, v)
+            return _matrix_map(m, torch.as_tensor(out, dtype=torch.float64,
+                                                  device=m.values.device),
+                               drop_name=True)
         raise PlanQuery(f"promql: unsupported function {f}")
 
     def _histogram_quantile(self, e, t0, step, T, grid):

@@ -207,3 +207,34 @@ def test_subquery_of_scalar_and_sum(prom_env):
         'min_over_time(sum(http_requests_total)[2m:30s])', 240, 240, 1)
     # sum at inner times 120..240 step 30: 3*t for t in (120,240] → min 3*150
     assert float(m.values[0][0]) == 3 * 150.0
+
+
+def test_time_functions(prom_env):
+    _, ev = prom_env
+    # 2024-05-25 20:16:37 UTC (a Saturday)
+    t = 1716668197
+    m = ev.query_range("hour()", t, t, 1)
+    assert float(m.values[0][0]) == 20.0
+    m = ev.query_range("minute()", t, t, 1)
+    assert float(m.values[0][0]) == 16.0
+    m = ev.query_range("day_of_week()", t, t, 1)
+    assert float(m.values[0][0]) == 6.0      # Saturday (0=Sunday)
+    m = ev.query_range("day_of_month()", t, t, 1)
+    assert float(m.values[0][0]) == 25.0
+    m = ev.query_range("month()", t, t, 1)
+    assert float(m.values[0][0]) == 5.0
+    m = ev.query_range("year()", t, t, 1)
+    assert float(m.values[0][0]) == 2024.0
+    m = ev.query_range("days_in_month()", t, t, 1)
+    assert float(m.values[0][0]) == 31.0
+    m = ev.query_range("day_of_year()", t, t, 1)
+    assert float(m.values[0][0]) == 146.0
+
+
+def test_sort_functions(prom_env):
+    _, ev = prom_env
+    m = ev.query_range("sort_desc(http_requests_total)", 300, 300, 1)
+    vals = [float(v) for v in m.values[:, 0]]
+    assert vals == sorted(vals, reverse=True)
+    insts = [l["instance"] for l in m.labels]
+    assert insts == ["b", "a"]     # b = 2× a

@@ -44,8 +44,9 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--scale", type=int, default=SCALE_PER_RANK)
-    ap.add_argument("--workers", type=int, default=4,
-                    help="ingest worker threads per rank (TSBS uses 6)")
+    ap.add_argument("--workers", type=int, default=6,
+                    help="ingest worker threads per rank (6 = TSBS's own "
+                         "client worker count; measured best on MI355X)")
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--durable", action="store_true", default=True)
     ap.add_argument("--flush-mb", type=int, default=1024,

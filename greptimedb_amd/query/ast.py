@@ -151,6 +151,16 @@ class ShowTables:
 
 
 @dataclass
+class ShowDatabases:
+    pass
+
+
+@dataclass
+class ShowCreateTable:
+    name: str
+
+
+@dataclass
 class DescribeTable:
     name: str
 

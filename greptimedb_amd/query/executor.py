@@ -180,6 +180,12 @@ class Executor:
         if isinstance(stmt, ast.ShowTables):
             names = sorted(self.engine.tables)
             return QueryResult(["Tables"], [names])
+        if isinstance(stmt, ast.ShowDatabases):
+            return QueryResult(["Database"],
+                               [["greptime_private", "information_schema",
+                                 "public"]])
+        if isinstance(stmt, ast.ShowCreateTable):
+            return self._show_create_table(stmt.name)
         if isinstance(stmt, ast.DescribeTable):
             st = self.engine.table(stmt.name)
             cols = st.schema.columns
@@ -217,6 +223,50 @@ class Executor:
         if isinstance(stmt, ast.AlterTable):
             return self._exec_alter(stmt)
         raise PlanQuery(f"unsupported statement {type(stmt).__name__}")
+
+    def _show_create_table(self, name: str) -> QueryResult:
+        """Render the DDL (ref: query show_create_table.rs)."""
+        st = self.engine.table(name)
+        schema = st.schema
+        sql_type = {
+            "string": "STRING", "float64": "DOUBLE", "float32": "FLOAT",
+            "int64": "BIGINT", "int32": "INT", "int16": "SMALLINT",
+            "int8": "TINYINT", "uint64": "BIGINT UNSIGNED",
+            "uint32": "INT UNSIGNED", "uint8": "TINYINT UNSIGNED",
+            "bool": "BOOLEAN", "binary": "BINARY", "json": "JSON",
+            "timestamp_ms": "TIMESTAMP(3)", "timestamp_s": "TIMESTAMP(0)",
+            "timestamp_us": "TIMESTAMP(6)", "timestamp_ns": "TIMESTAMP(9)",
+        }
+        lines = []
+        seen = set()
+        for c in schema.columns:
+            seen.add(c.name)
+            t = f"VECTOR({c.vector_dim})" if c.dtype.value == "vector" else \
+                sql_type.get(c.dtype.value, c.dtype.value.upper())
+            opts = ""
+            if c.semantic == SemanticType.TIMESTAMP:
+                opts = " NOT NULL"
+            if c.fulltext:
+                opts += " FULLTEXT INDEX"
+            lines.append(f'  "{c.name}" {t}{opts}')
+        for fn in st.regions[0].field_names:
+            if fn not in seen:
+                lines.append(f'  "{fn}" DOUBLE')
+        for sn in st.regions[0].str_field_names:
+            if sn not in seen:
+                ft = " FULLTEXT INDEX" if sn in st.regions[0].text_cols else ""
+                lines.append(f'  "{sn}" STRING{ft}')
+        lines.append(f'  TIME INDEX ("{schema.time_index.name}")')
+        if schema.primary_key:
+            pk = ", ".join(f'"{t}"' for t in schema.primary_key)
+            lines.append(f"  PRIMARY KEY ({pk})")
+        body = ",\n".join(lines)
+        opts = [f"  regions = {len(st.regions)}"]
+        if st.append_mode:
+            opts.append("  append_mode = 'true'")
+        ddl = (f'CREATE TABLE IF NOT EXISTS "{name}" (\n{body}\n)\n'
+               f"ENGINE=mito\nWITH(\n" + ",\n".join(opts) + "\n)")
+        return QueryResult(["Table", "Create Table"], [[name], [ddl]])
 
     def _exec_alter(self, a: ast.AlterTable) -> QueryResult:
         """ALTER TABLE ADD COLUMN (reference: alter DDL procedure; tags are

@@ -169,6 +169,11 @@ class Parser:
             self.next()
             if self.eat_kw("flows"):
                 stmt = ast.ShowFlows()
+            elif self.eat_kw("databases") or self.eat_kw("schemas"):
+                stmt = ast.ShowDatabases()
+            elif self.eat_kw("create"):
+                self.expect_kw("table")
+                stmt = ast.ShowCreateTable(self.next().value)
             else:
                 self.expect_kw("tables")
                 stmt = ast.ShowTables()

@@ -225,3 +225,17 @@ def test_internal_tracing_self_export(tmp_engine):
     assert all(row[0] == "greptimedb_amd" for row in rows)
     ex.execute("ADMIN enable_tracing(0)")
     assert not tracer.enabled
+
+
+def test_show_databases_and_create_table(ex):
+    r = ex.execute("SHOW DATABASES")
+    assert "public" in list(r.columns[0])
+    r = ex.execute("SHOW CREATE TABLE t1")
+    ddl = r.columns[1][0]
+    assert 'CREATE TABLE IF NOT EXISTS "t1"' in ddl
+    assert 'TIME INDEX ("ts")' in ddl and 'PRIMARY KEY ("h")' in ddl
+    assert '"v" DOUBLE' in ddl
+    # the DDL round-trips through our own parser
+    ex.execute(ddl.replace('"t1"', '"t1_copy"'))
+    r2 = ex.execute("SHOW CREATE TABLE t1_copy")
+    assert 'TIME INDEX ("ts")' in r2.columns[1][0]

@@ -4,6 +4,9 @@ Reference parity: src/servers/src/postgres/ (pgwire based). Implements the
 v3 protocol simple-query flow: StartupMessage → AuthenticationOk +
 ParameterStatus + ReadyForQuery; 'Q' → RowDescription/DataRow/
 CommandComplete; errors as ErrorResponse; SSLRequest politely declined.
+Extended protocol: Parse/Bind/Describe/Execute/Close/Sync with text-format
+parameters ($1…) substituted at Bind (the query runs at Bind so Describe
+can return the real RowDescription, like the reference plans at bind).
 """
 
 from __future__ import annotations
@@ -84,6 +87,9 @@ class PostgresServer:
             writer.write(_msg(b"Z", b"I"))
             await writer.drain()
 
+            stmts: dict = {}    # name → (sql, n_params)   (extended protocol)
+            portals: dict = {}  # name → QueryResult | GreptimeError
+
             while True:
                 try:
                     tag = await reader.readexactly(1)
@@ -100,11 +106,9 @@ class PostgresServer:
                         writer.write(chunk)
                     writer.write(_msg(b"Z", b"I"))
                     await writer.drain()
-                elif tag in (b"P", b"B", b"D", b"E", b"S"):
-                    # minimal extended-protocol shim: report unsupported
-                    writer.write(_msg(b"E", b"SERROR\x00C0A000\x00M"
-                                      b"extended query protocol not supported\x00\x00"))
-                    writer.write(_msg(b"Z", b"I"))
+                elif tag in (b"P", b"B", b"D", b"E", b"C", b"S", b"H"):
+                    for chunk in self._extended(tag, body, stmts, portals):
+                        writer.write(chunk)
                     await writer.drain()
                 else:
                     writer.write(_msg(b"Z", b"I"))
@@ -116,6 +120,101 @@ class PostgresServer:
                 writer.close()
             except Exception:
                 pass
+
+    # ---------------------------------------------- extended query protocol
+
+    def _extended(self, tag: bytes, body: bytes, stmts: dict, portals: dict):
+        _NUM = __import__("re").compile(r"-?\d+(\.\d+)?([eE][+-]?\d+)?$")
+
+        def err(msg: str, code: bytes = b"42601"):
+            return _msg(b"E", b"SERROR\x00C" + code + b"\x00M" +
+                        msg.encode()[:400] + b"\x00\x00")
+
+        if tag == b"S":                       # Sync
+            yield _msg(b"Z", b"I")
+            return
+        if tag == b"H":                       # Flush — nothing buffered
+            return
+        if tag == b"P":                       # Parse: name\0 sql\0 n oids…
+            name, rest = body.split(b"\x00", 1)
+            sql, rest = rest.split(b"\x00", 1)
+            (np_,) = struct.unpack("!h", rest[:2])
+            stmts[name] = (sql.decode(errors="replace"), np_)
+            yield _msg(b"1", b"")             # ParseComplete
+            return
+        if tag == b"B":                       # Bind: portal\0 stmt\0 fmts, params
+            portal, rest = body.split(b"\x00", 1)
+            sname, rest = rest.split(b"\x00", 1)
+            off = 0
+            (nfmt,) = struct.unpack_from("!h", rest, off); off += 2
+            fmts = struct.unpack_from(f"!{nfmt}h", rest, off); off += 2 * nfmt
+            (nparams,) = struct.unpack_from("!h", rest, off); off += 2
+            params = []
+            for i in range(nparams):
+                (plen,) = struct.unpack_from("!i", rest, off); off += 4
+                if plen < 0:
+                    params.append(None)
+                else:
+                    params.append(rest[off:off + plen]); off += plen
+                if (fmts[i] if i < nfmt else (fmts[0] if nfmt == 1 else 0)) == 1:
+                    yield err("binary parameters not supported", b"0A000")
+                    return
+            got = stmts.get(sname)
+            if got is None:
+                yield err(f"unknown prepared statement {sname!r}", b"26000")
+                return
+            sql, _np = got
+            for i in range(len(params), 0, -1):   # $10 before $1
+                p = params[i - 1]
+                if p is None:
+                    lit = "NULL"
+                else:
+                    s = p.decode(errors="replace")
+                    lit = s if _NUM.match(s) else "'" + s.replace("'", "''") + "'"
+                sql = sql.replace(f"${i}", lit)
+            try:
+                portals[portal] = self.executor.execute(sql)
+            except GreptimeError as e:
+                portals[portal] = e
+            except Exception as e:              # pragma: no cover
+                portals[portal] = GreptimeError(f"{type(e).__name__}: {e}")
+            yield _msg(b"2", b"")               # BindComplete
+            return
+        if tag == b"D":                        # Describe 'S'|'P' + name
+            kind, name = body[:1], body[1:].split(b"\x00")[0]
+            if kind == b"S":
+                got = stmts.get(name)
+                np_ = got[1] if got else 0
+                yield _msg(b"t", struct.pack("!h", np_) +
+                           struct.pack(f"!{np_}I", *([OID_TEXT] * np_)))
+                yield _msg(b"n", b"")           # NoData (schema known at Bind)
+                return
+            r = portals.get(name)
+            if isinstance(r, GreptimeError) or r is None:
+                yield _msg(b"n", b"")
+                return
+            yield self._row_description(r)
+            return
+        if tag == b"E":                        # Execute: portal\0 maxrows
+            name = body.split(b"\x00")[0]
+            r = portals.get(name)
+            if r is None:
+                yield err(f"unknown portal {name!r}", b"34000")
+                return
+            if isinstance(r, GreptimeError):
+                yield err(str(r) or type(r).__name__)
+                return
+            nrows = 0
+            for chunk in self._data_rows(r):
+                nrows += 1
+                yield chunk
+            yield _msg(b"C", f"SELECT {nrows}".encode() + b"\x00")
+            return
+        if tag == b"C":                        # Close statement/portal
+            kind, name = body[:1], body[1:].split(b"\x00")[0]
+            (stmts if kind == b"S" else portals).pop(name, None)
+            yield _msg(b"3", b"")
+            return
 
     def _run_query(self, sql: str):
         s = sql.strip().rstrip(";").lower()
@@ -132,9 +231,15 @@ class PostgresServer:
             yield _msg(b"E", b"SERROR\x00CXX000\x00M" +
                        f"{type(e).__name__}: {e}".encode()[:400] + b"\x00\x00")
             return
-        # RowDescription
+        yield self._row_description(r)
+        nrows = 0
+        for chunk in self._data_rows(r):
+            nrows += 1
+            yield chunk
+        yield _msg(b"C", f"SELECT {nrows}".encode() + b"\x00")
+
+    def _row_description(self, r) -> bytes:
         fields = b""
-        oids = []
         for col, kind, name in zip(r.columns, r.kinds, r.names):
             if kind == "ts":
                 oid = OID_TIMESTAMP
@@ -144,12 +249,12 @@ class PostgresServer:
                 oid = OID_FLOAT8
             else:
                 oid = OID_TEXT
-            oids.append(oid)
             fields += (name.encode() + b"\x00" + struct.pack("!IhIhih", 0, 0, oid,
                                                              -1, -1, 0))
-        yield _msg(b"T", struct.pack("!h", len(r.names)) + fields)
+        return _msg(b"T", struct.pack("!h", len(r.names)) + fields)
+
+    def _data_rows(self, r):
         from greptimedb_amd.utils.timeutil import format_ts_ms
-        nrows = 0
         for row in r.rows():
             payload = struct.pack("!h", len(row))
             for v, kind in zip(row, r.kinds):
@@ -163,5 +268,3 @@ class PostgresServer:
                     b = sv.encode()
                     payload += struct.pack("!i", len(b)) + b
             yield _msg(b"D", payload)
-            nrows += 1
-        yield _msg(b"C", f"SELECT {nrows}".encode() + b"\x00")

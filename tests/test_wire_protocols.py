@@ -183,3 +183,81 @@ def test_postgres_protocol(ex):
     assert out[0][2] == [["a", "1.5"], ["b", "2.5"]]
     assert out[1][2] == [["2"]]
     assert out[2][0] == "ERROR"
+
+
+async def _pg_extended_session(port, sql, params):
+    """Parse/Bind/Describe/Execute/Sync round trip (text params)."""
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    body = struct.pack("!I", 196608) + b"user\x00tester\x00\x00"
+    writer.write(struct.pack("!I", len(body) + 4) + body)
+    await writer.drain()
+
+    async def read_msg():
+        tag = await reader.readexactly(1)
+        (ln,) = struct.unpack("!I", await reader.readexactly(4))
+        return tag, await reader.readexactly(ln - 4)
+
+    while (await read_msg())[0] != b"Z":
+        pass
+
+    def msg(tag, payload):
+        return tag + struct.pack("!I", len(payload) + 4) + payload
+
+    p = b"\x00" + sql.encode() + b"\x00" + struct.pack("!h", len(params))
+    p += struct.pack(f"!{len(params)}I", *([0] * len(params)))
+    writer.write(msg(b"P", p))
+    b = b"\x00\x00" + struct.pack("!h", 0) + struct.pack("!h", len(params))
+    for v in params:
+        ev = str(v).encode()
+        b += struct.pack("!i", len(ev)) + ev
+    b += struct.pack("!h", 0)
+    writer.write(msg(b"B", b))
+    writer.write(msg(b"D", b"P\x00"))
+    writer.write(msg(b"E", b"\x00" + struct.pack("!i", 0)))
+    writer.write(msg(b"S", b""))
+    await writer.drain()
+    names, rows, seen = [], [], []
+    while True:
+        tag, body = await read_msg()
+        seen.append(tag)
+        if tag == b"T":
+            (ncols,) = struct.unpack_from("!h", body, 0)
+            off = 2
+            for _ in range(ncols):
+                end = body.index(b"\x00", off)
+                names.append(body[off:end].decode())
+                off = end + 1 + 18
+        elif tag == b"D":
+            (ncols,) = struct.unpack_from("!h", body, 0)
+            off = 2
+            row = []
+            for _ in range(ncols):
+                (ln2,) = struct.unpack_from("!i", body, off)
+                off += 4
+                if ln2 < 0:
+                    row.append(None)
+                else:
+                    row.append(body[off:off + ln2].decode())
+                    off += ln2
+            rows.append(row)
+        elif tag == b"Z":
+            break
+    writer.write(b"X" + struct.pack("!I", 4))
+    writer.close()
+    return seen, names, rows
+
+
+def test_postgres_extended_protocol(ex):
+    async def run():
+        srv = PostgresServer(ex, host="127.0.0.1", port=0)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        out = await _pg_extended_session(
+            port, "SELECT h, v FROM t WHERE v > $1 AND h != $2 ORDER BY h",
+            [1.0, "zzz"])
+        s.close()
+        return out
+    seen, names, rows = asyncio.run(run())
+    assert b"1" in seen and b"2" in seen and b"C" in seen  # Parse/Bind complete
+    assert names == ["h", "v"]
+    assert rows == [["a", "1.5"], ["b", "2.5"]]

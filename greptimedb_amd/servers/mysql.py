@@ -116,6 +116,8 @@ class MySQLServer:
             writer.write(self._ok(seq + 1))
             await writer.drain()
 
+            stmts: dict = {}     # stmt_id → {"sql", "nparams", "types"}
+            next_stmt = [1]
             while True:
                 try:
                     seq, cmd = await self._read_packet(reader)
@@ -131,6 +133,17 @@ class MySQLServer:
                 elif op == 0x03:     # COM_QUERY
                     sql = cmd[1:].decode(errors="replace")
                     writer.write(self._run_query(sql))
+                elif op == 0x16:     # COM_STMT_PREPARE
+                    writer.write(self._stmt_prepare(cmd[1:], stmts, next_stmt))
+                elif op == 0x17:     # COM_STMT_EXECUTE
+                    writer.write(self._stmt_execute(cmd[1:], stmts))
+                elif op in (0x19, 0x1A):  # COM_STMT_CLOSE / COM_STMT_RESET
+                    (sid,) = struct.unpack_from("<I", cmd, 1)
+                    if op == 0x19:
+                        stmts.pop(sid, None)   # CLOSE: no response
+                    else:
+                        stmts.get(sid, {}).pop("types", None)
+                        writer.write(self._ok(1))
                 else:
                     writer.write(self._err(1, f"unsupported command {op:#x}", 1047))
                 await writer.drain()
@@ -142,7 +155,105 @@ class MySQLServer:
             except Exception:
                 pass
 
-    def _run_query(self, sql: str) -> bytes:
+    # -------------------------------------------------- prepared statements
+
+    def _stmt_prepare(self, body: bytes, stmts: dict, next_stmt: list) -> bytes:
+        """COM_STMT_PREPARE: '?' placeholders, params typed at execute.
+        num_columns reported 0 — the execute response carries the real
+        resultset metadata (clients re-read it; Connector/J & friends do)."""
+        sql = body.decode(errors="replace")
+        nparams = sql.count("?")
+        sid = next_stmt[0]
+        next_stmt[0] += 1
+        stmts[sid] = {"sql": sql, "nparams": nparams}
+        out = [self._packet(1, b"\x00" + struct.pack("<I", sid) +
+                            struct.pack("<HH", 0, nparams) + b"\x00" +
+                            struct.pack("<H", 0))]
+        seq = 2
+        for i in range(nparams):
+            nb = f"?{i}".encode()
+            col = (lenenc_str(b"def") + lenenc_str(b"") + lenenc_str(b"") +
+                   lenenc_str(b"") + lenenc_str(nb) + lenenc_str(nb) +
+                   b"\x0c" + struct.pack("<H", 33) + struct.pack("<I", 1024) +
+                   bytes([MYSQL_TYPE_VAR_STRING]) + struct.pack("<H", 0) +
+                   bytes([0]) + b"\x00\x00")
+            out.append(self._packet(seq, col))
+            seq += 1
+        if nparams:
+            out.append(self._eof(seq))
+        return b"".join(out)
+
+    @staticmethod
+    def _decode_params(body: bytes, off: int, n: int, st: dict):
+        null_bm = body[off: off + (n + 7) // 8]
+        off += (n + 7) // 8
+        bound = body[off]
+        off += 1
+        if bound:
+            st["types"] = [struct.unpack_from("<BB", body, off + 2 * i)[0]
+                           for i in range(n)]
+            off += 2 * n
+        types = st.get("types") or [MYSQL_TYPE_VAR_STRING] * n
+        vals = []
+        for i in range(n):
+            if null_bm[i // 8] & (1 << (i % 8)):
+                vals.append(None)
+                continue
+            t = types[i]
+            if t in (1,):                         # TINY
+                vals.append(struct.unpack_from("<b", body, off)[0]); off += 1
+            elif t == 2:                          # SHORT
+                vals.append(struct.unpack_from("<h", body, off)[0]); off += 2
+            elif t == 3:                          # LONG
+                vals.append(struct.unpack_from("<i", body, off)[0]); off += 4
+            elif t == 8:                          # LONGLONG
+                vals.append(struct.unpack_from("<q", body, off)[0]); off += 8
+            elif t == 4:                          # FLOAT
+                vals.append(struct.unpack_from("<f", body, off)[0]); off += 4
+            elif t == 5:                          # DOUBLE
+                vals.append(struct.unpack_from("<d", body, off)[0]); off += 8
+            elif t == 6:                          # NULL
+                vals.append(None)
+            else:                                 # lenenc string-ish
+                ln = body[off]; off += 1
+                if ln == 0xFC:
+                    (ln,) = struct.unpack_from("<H", body, off); off += 2
+                elif ln == 0xFD:
+                    ln = int.from_bytes(body[off:off + 3], "little"); off += 3
+                elif ln == 0xFE:
+                    (ln,) = struct.unpack_from("<Q", body, off); off += 8
+                vals.append(body[off:off + ln].decode(errors="replace"))
+                off += ln
+        return vals
+
+    def _stmt_execute(self, body: bytes, stmts: dict) -> bytes:
+        (sid,) = struct.unpack_from("<I", body, 0)
+        st = stmts.get(sid)
+        if st is None:
+            return self._err(1, f"unknown statement {sid}", 1243)
+        off = 9                                   # id(4) + flags(1) + iter(4)
+        try:
+            vals = self._decode_params(body, off, st["nparams"], st) \
+                if st["nparams"] else []
+        except (IndexError, struct.error):
+            return self._err(1, "malformed COM_STMT_EXECUTE", 1210)
+        sql = ""
+        it = iter(vals)
+        for part in st["sql"].split("?"):
+            sql += part
+            try:
+                v = next(it)
+            except StopIteration:
+                continue
+            if v is None:
+                sql += "NULL"
+            elif isinstance(v, (int, float)):
+                sql += repr(v)
+            else:
+                sql += "'" + str(v).replace("'", "''") + "'"
+        return self._run_query(sql, binary=True)
+
+    def _run_query(self, sql: str, binary: bool = False) -> bytes:
         s = sql.strip().rstrip(";").lower()
         # common client handshake queries
         if s.startswith(("set ", "set@", "use ")) or s in ("commit", "rollback", "begin"):
@@ -178,7 +289,39 @@ class MySQLServer:
                 else:
                     out.append(str(v))
             rows.append(out)
+        if binary:
+            return self._binary_resultset(r.names, rows)
         return self._text_resultset(r.names, rows, types)
+
+    def _binary_resultset(self, names, rows) -> bytes:
+        """Binary-protocol resultset (COM_STMT_EXECUTE response). Every
+        column is declared VAR_STRING, so values are lenenc strings — the
+        client coerces (this is what stringified text mode looks like)."""
+        out = [self._packet(1, lenenc_int(len(names)))]
+        seq = 2
+        for name in names:
+            nb = name.encode()
+            col = (lenenc_str(b"def") + lenenc_str(b"") + lenenc_str(b"") +
+                   lenenc_str(b"") + lenenc_str(nb) + lenenc_str(nb) +
+                   b"\x0c" + struct.pack("<H", 33) + struct.pack("<I", 1024) +
+                   bytes([MYSQL_TYPE_VAR_STRING]) + struct.pack("<H", 0) +
+                   bytes([0]) + b"\x00\x00")
+            out.append(self._packet(seq, col))
+            seq += 1
+        out.append(self._eof(seq)); seq += 1
+        nb_len = (len(names) + 2 + 7) // 8
+        for row in rows:
+            bm = bytearray(nb_len)
+            payload = b""
+            for i, v in enumerate(row):
+                if v is None:
+                    bm[(i + 2) // 8] |= 1 << ((i + 2) % 8)
+                else:
+                    payload += lenenc_str(str(v).encode())
+            out.append(self._packet(seq, b"\x00" + bytes(bm) + payload))
+            seq += 1
+        out.append(self._eof(seq))
+        return b"".join(out)
 
     def _text_resultset(self, names, rows, types) -> bytes:
         out = [self._packet(1, lenenc_int(len(names)))]

@@ -261,3 +261,97 @@ def test_postgres_extended_protocol(ex):
     assert b"1" in seen and b"2" in seen and b"C" in seen  # Parse/Bind complete
     assert names == ["h", "v"]
     assert rows == [["a", "1.5"], ["b", "2.5"]]
+
+
+async def _mysql_prepared_session(port, sql, params):
+    """COM_STMT_PREPARE + COM_STMT_EXECUTE with typed binary params."""
+    reader, writer = await asyncio.open_connection("127.0.0.1", port)
+    seq, greeting = await _mysql_read_packet(reader)
+    resp = (struct.pack("<IIB", 0x0200, 1 << 24, 33) + b"\x00" * 23 +
+            b"tester\x00" + b"\x00")
+    ln = len(resp)
+    writer.write(bytes([ln & 0xFF, (ln >> 8) & 0xFF, (ln >> 16) & 0xFF,
+                        seq + 1]) + resp)
+    await writer.drain()
+    _s, ok = await _mysql_read_packet(reader)
+    assert ok[0] == 0x00
+
+    def send(payload):
+        writer.write(struct.pack("<I", len(payload))[:3] + b"\x00" + payload)
+
+    send(b"\x16" + sql.encode())
+    await writer.drain()
+    seq, first = await _mysql_read_packet(reader)
+    assert first[0] == 0x00
+    (sid,) = struct.unpack_from("<I", first, 1)
+    (ncols, nparams) = struct.unpack_from("<HH", first, 5)
+    for _ in range(nparams + (1 if nparams else 0)):    # param defs + eof
+        await _mysql_read_packet(reader)
+    # execute: null bitmap + bound flag + types + values
+    body = b"\x17" + struct.pack("<IBI", sid, 0, 1)
+    bm = bytearray((len(params) + 7) // 8)
+    types = b""
+    vals = b""
+    for i, p in enumerate(params):
+        if p is None:
+            bm[i // 8] |= 1 << (i % 8)
+            types += bytes([6, 0])
+        elif isinstance(p, int):
+            types += bytes([8, 0]); vals += struct.pack("<q", p)
+        elif isinstance(p, float):
+            types += bytes([5, 0]); vals += struct.pack("<d", p)
+        else:
+            b = str(p).encode()
+            types += bytes([0xFD, 0]); vals += bytes([len(b)]) + b
+    body += bytes(bm) + b"\x01" + types + vals
+    send(body)
+    await writer.drain()
+    # read binary resultset
+    seq, head = await _mysql_read_packet(reader)
+    ncols = head[0]
+    names = []
+    for _ in range(ncols):
+        _s, col = await _mysql_read_packet(reader)
+        # parse 5th lenenc string (name)
+        off = 0
+        for k in range(5):
+            ln = col[off]; off += 1
+            if k == 4:
+                names.append(col[off:off + ln].decode())
+            off += ln
+    await _mysql_read_packet(reader)                    # eof
+    rows = []
+    while True:
+        _s, pkt = await _mysql_read_packet(reader)
+        if pkt[0] == 0xFE and len(pkt) < 9:
+            break
+        nb_len = (ncols + 2 + 7) // 8
+        bm2 = pkt[1:1 + nb_len]
+        off = 1 + nb_len
+        row = []
+        for i in range(ncols):
+            if bm2[(i + 2) // 8] & (1 << ((i + 2) % 8)):
+                row.append(None)
+                continue
+            ln = pkt[off]; off += 1
+            row.append(pkt[off:off + ln].decode()); off += ln
+        rows.append(row)
+    writer.write(b"\x01\x00\x00\x00\x01")
+    writer.close()
+    return names, rows
+
+
+def test_mysql_prepared_statements(ex):
+    async def run():
+        from greptimedb_amd.servers.mysql import MySQLServer
+        srv = MySQLServer(ex, host="127.0.0.1", port=0)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        out = await _mysql_prepared_session(
+            port, "SELECT h, v FROM t WHERE v > ? AND h != ? ORDER BY h",
+            [1.0, "zzz"])
+        s.close()
+        return out
+    names, rows = asyncio.run(run())
+    assert names == ["h", "v"]
+    assert rows == [["a", "1.5"], ["b", "2.5"]]

@@ -137,6 +137,7 @@ class CreateTable:
     if_not_exists: bool = False
     options: dict = field(default_factory=dict)
     partitions: int | None = None
+    external: bool = False       # CREATE EXTERNAL TABLE (file engine)
 
 
 @dataclass

@@ -592,6 +592,8 @@ class Executor:
     # ---------------------------------------------------------- DDL / DML
 
     def _exec_create(self, c: ast.CreateTable) -> QueryResult:
+        if c.external:
+            return self._exec_create_external(c)
         type_map = {
             "string": DataType.STRING, "varchar": DataType.STRING, "text": DataType.STRING,
             "double": DataType.FLOAT64, "float64": DataType.FLOAT64, "float": DataType.FLOAT32,
@@ -635,8 +637,81 @@ class Executor:
                                  append_mode=append, if_not_exists=c.if_not_exists)
         return QueryResult(["status"], [["ok"]])
 
+    def _exec_create_external(self, c: ast.CreateTable) -> QueryResult:
+        """CREATE EXTERNAL TABLE … WITH (location=…, format=…) — the file
+        engine (ref src/file-engine: read-only tables over files). MI355X
+        take: the file is materialized into device columns at create (our
+        cache policy is HBM-resident anyway); reads are then identical to
+        native tables; writes are rejected."""
+        import pyarrow as pa
+        loc = c.options.get("location") or c.options.get("LOCATION")
+        if not loc:
+            raise InvalidArguments("external table needs WITH (location='…')")
+        loc = str(loc).strip("'")
+        fmt = str(c.options.get("format", "")).strip("'").lower() or (
+            "csv" if loc.endswith(".csv") else
+            "json" if loc.endswith(".json") else "parquet")
+        if c.name in self.engine.tables:
+            if c.if_not_exists:
+                return QueryResult(["status"], [["ok"]])
+            from greptimedb_amd.utils.errors import TableAlreadyExists
+            raise TableAlreadyExists(c.name)
+        # infer schema from the file when no column list was given
+        if fmt == "parquet":
+            import pyarrow.parquet as pq
+            t = pq.read_table(loc)
+        elif fmt == "csv":
+            import pyarrow.csv as pacsv
+            t = pacsv.read_csv(loc)
+        else:
+            import json as _json
+            rows = [_json.loads(l) for l in open(loc) if l.strip()]
+            t = pa.Table.from_pydict(
+                {k: [r.get(k) for r in rows] for k in (rows[0] if rows else {})})
+        ts_name = None
+        for fld in t.schema:
+            if pa.types.is_timestamp(fld.type):
+                ts_name = fld.name
+                break
+        if ts_name is None:
+            for cand in ("ts", "timestamp", "time"):
+                if cand in t.column_names:
+                    ts_name = cand
+                    break
+        if ts_name is None:
+            raise InvalidArguments("external file has no timestamp column")
+        cols = []
+        cid = 0
+        for fld in t.schema:
+            if fld.name == ts_name:
+                cols.append(ColumnSchema(ts_name, DataType.TIMESTAMP_MS,
+                                         SemanticType.TIMESTAMP, cid))
+            elif pa.types.is_string(fld.type) or pa.types.is_large_string(fld.type):
+                cols.append(ColumnSchema(fld.name, DataType.STRING,
+                                         SemanticType.FIELD, cid))
+            else:
+                cols.append(ColumnSchema(fld.name, DataType.FLOAT64,
+                                         SemanticType.FIELD, cid))
+            cid += 1
+        schema = TableSchema(name=c.name, columns=cols, primary_key=[])
+        st = self.engine.create_table(schema, n_regions=1, append_mode=True)
+        st.external = True
+        for region in st.regions:
+            region.ensure_fields([cc.name for cc in cols
+                                  if cc.semantic == SemanticType.FIELD
+                                  and cc.dtype == DataType.FLOAT64])
+            region.ensure_str_fields([cc.name for cc in cols
+                                      if cc.dtype == DataType.STRING],
+                                     fulltext=False)
+        st.external = False      # let the COPY-FROM load path write
+        self._exec_copy(ast.Copy(c.name, loc, "from", {"format": fmt}))
+        st.external = True
+        return QueryResult(["status"], [["ok"]])
+
     def _exec_insert(self, ins: ast.InsertValues) -> QueryResult:
         st = self.engine.table(ins.table)
+        if getattr(st, "external", False):
+            raise InvalidArguments(f"table {ins.table} is external (read-only)")
         schema = st.schema
         cols = ins.columns or [c.name for c in schema.columns]
         n = len(ins.rows)

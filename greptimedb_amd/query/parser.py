@@ -380,6 +380,7 @@ class Parser:
             sql = self.sql[self.peek().pos:].rstrip().rstrip(";")
             self.i = len(self.toks)
             return ast.CreateFlow(name, sink, sql, if_not_exists)
+        external = self.eat_kw("external")
         self.expect_kw("table")
         if_not_exists = False
         if self.eat_kw("if"):
@@ -387,6 +388,20 @@ class Parser:
             self.expect_kw("exists")
             if_not_exists = True
         name = self.next().value
+        if external and not self.at_op("("):
+            # schema inferred from the file (reference: file-engine infer)
+            options = {}
+            while self.eat_kw("with"):
+                self.expect_op("(")
+                while True:
+                    k = self.next().value
+                    self.expect_op("=")
+                    options[str(k).strip("'")] = self.next().value
+                    if not self.eat_op(","):
+                        break
+                self.expect_op(")")
+            return ast.CreateTable(name, [], [], None, if_not_exists, options,
+                                   None, external=True)
         self.expect_op("(")
         columns = []
         primary_key: list[str] = []
@@ -464,7 +479,8 @@ class Parser:
             else:
                 break
         return ast.CreateTable(name, columns, primary_key, time_index,
-                               if_not_exists, options, partitions)
+                               if_not_exists, options, partitions,
+                               external=external)
 
     def parse_drop(self):
         self.expect_kw("drop")

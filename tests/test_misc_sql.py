@@ -239,3 +239,26 @@ def test_show_databases_and_create_table(ex):
     ex.execute(ddl.replace('"t1"', '"t1_copy"'))
     r2 = ex.execute("SHOW CREATE TABLE t1_copy")
     assert 'TIME INDEX ("ts")' in r2.columns[1][0]
+
+
+def test_external_table(tmp_engine, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    p = str(tmp_path / "ext.parquet")
+    t = pa.table({
+        "ts": pa.array([1000, 2000, 3000], type=pa.timestamp("ms")),
+        "city": ["sf", "ny", "sf"],
+        "temp": [12.5, 20.0, 13.5],
+    })
+    pq.write_table(t, p)
+    ex = Executor(tmp_engine)
+    ex.execute(f"CREATE EXTERNAL TABLE weather WITH (location='{p}', "
+               f"format='parquet')")
+    r = ex.execute("SELECT city, temp FROM weather ORDER BY ts")
+    assert [tuple(x) for x in r.rows()] == [("sf", 12.5), ("ny", 20.0),
+                                            ("sf", 13.5)]
+    r = ex.execute("SELECT max(temp) FROM weather")
+    assert float(list(r.rows())[0][0]) == 20.0
+    # read-only
+    with pytest.raises(Exception):
+        ex.execute("INSERT INTO weather (ts, city, temp) VALUES (4000, 'x', 1.0)")

@@ -543,9 +543,20 @@ class Executor:
             raise PlanQuery("EXPLAIN supports SELECT")
         plan = self._plan_select(e.stmt)
         lines = []
-        path = "fused-ts-bucket-agg" if plan.aggs and plan.residual is None else \
-            ("agg-with-residual" if plan.aggs else "raw-scan")
+        if any(_has_range_agg(x) for x, _a in e.stmt.projections):
+            path = "range-select (sliding-window kernel)"
+        elif any(_collect_window_nodes(x, w := []) or w
+                 for x, _a in e.stmt.projections):
+            path = "raw-scan + window-functions"
+        elif plan.aggs and plan.residual is None:
+            path = "fused-ts-bucket-agg"
+        elif plan.aggs:
+            path = "agg-with-residual"
+        else:
+            path = "raw-scan"
         lines.append(f"Plan: {path}")
+        if e.stmt.align_ms:
+            lines.append(f"  align_ms: {e.stmt.align_ms} by: {e.stmt.align_by}")
         lines.append(f"  table: {plan.table.schema.name} "
                      f"regions={len(plan.table.regions)} append={plan.table.append_mode}")
         lines.append(f"  time_range: [{plan.ts_lo}, {plan.ts_hi})")

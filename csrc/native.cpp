@@ -1258,6 +1258,96 @@ static py::tuple pack_str_col(py::list vals) {
   return py::make_tuple(std::move(lens), py::bytes(blob));
 }
 
+// One-pass ingest router (K16 host side): per-region WAL payload bodies,
+// destination offsets for the scatter_append kernel, and per-region
+// counts/min/max — replaces the python argsort + per-region numpy fancy
+// gathers on the ingest hot path. Payload layout matches wal.encode_batch:
+// [u32 hdr_len][hdr json][i32 codes][i64 ts][f64 fields nf*m].
+// `hdr_suffix` is the cached per-table json tail: everything after the
+// numeric value of "n" (b', "fields": [...], "strs": [], ...}').
+static py::tuple route_ingest(
+    py::array_t<int32_t, py::array::c_style | py::array::forcecast> codes,
+    py::array_t<int64_t, py::array::c_style | py::array::forcecast> ts,
+    py::array_t<double, py::array::c_style | py::array::forcecast> fields,
+    py::array_t<int32_t, py::array::c_style | py::array::forcecast> region_of,
+    int n_regions, py::bytes hdr_suffix, bool durable) {
+  const int64_t n = codes.shape(0);
+  const int nf = fields.ndim() == 2 ? (int)fields.shape(0) : 0;
+  const int32_t* cp = codes.data();
+  const int64_t* tp = ts.data();
+  const double* fp = nf ? fields.data() : nullptr;
+  const int32_t* rp = region_of.data();
+  std::string suffix = hdr_suffix;
+
+  py::array_t<int64_t> dst_off(n);
+  py::array_t<int64_t> counts(n_regions), mins(n_regions), maxs(n_regions);
+  int64_t* dp = dst_off.mutable_data();
+  int64_t* cnt = counts.mutable_data();
+  int64_t* mn = mins.mutable_data();
+  int64_t* mx = maxs.mutable_data();
+  for (int r = 0; r < n_regions; r++) {
+    cnt[r] = 0;
+    mn[r] = INT64_MAX;
+    mx[r] = INT64_MIN;
+  }
+  {
+    py::gil_scoped_release nogil;
+    for (int64_t i = 0; i < n; i++) {
+      const int32_t r = rp[i];
+      dp[i] = cnt[r]++;
+      const int64_t t = tp[i];
+      if (t < mn[r]) mn[r] = t;
+      if (t > mx[r]) mx[r] = t;
+    }
+  }
+
+  py::list payloads;
+  if (!durable) {
+    for (int r = 0; r < n_regions; r++) payloads.append(py::none());
+    return py::make_tuple(payloads, std::move(dst_off), std::move(counts),
+                          std::move(mins), std::move(maxs));
+  }
+  // allocate per-region payload buffers (header + columns)
+  std::vector<char*> bufs(n_regions, nullptr);
+  std::vector<int64_t> code_off(n_regions), ts_off(n_regions), f_off(n_regions);
+  for (int r = 0; r < n_regions; r++) {
+    if (cnt[r] == 0) {
+      payloads.append(py::none());
+      continue;
+    }
+    std::string hdr = "{\"n\": " + std::to_string(cnt[r]) + suffix;
+    const int64_t m = cnt[r];
+    const int64_t total = 4 + (int64_t)hdr.size() + 4 * m + 8 * m + 8LL * nf * m;
+    PyObject* b = PyBytes_FromStringAndSize(nullptr, total);
+    char* p = PyBytes_AS_STRING(b);
+    const uint32_t hl = (uint32_t)hdr.size();
+    std::memcpy(p, &hl, 4);
+    std::memcpy(p + 4, hdr.data(), hl);
+    bufs[r] = p;
+    code_off[r] = 4 + hl;
+    ts_off[r] = code_off[r] + 4 * m;
+    f_off[r] = ts_off[r] + 8 * m;
+    payloads.append(py::reinterpret_steal<py::object>(b));
+  }
+  {
+    py::gil_scoped_release nogil;
+    for (int64_t i = 0; i < n; i++) {
+      const int32_t r = rp[i];
+      char* p = bufs[r];
+      const int64_t o = dp[i];
+      const int64_t m = cnt[r];
+      std::memcpy(p + code_off[r] + 4 * o, cp + i, 4);
+      std::memcpy(p + ts_off[r] + 8 * o, tp + i, 8);
+      char* fdst = p + f_off[r];
+      for (int f = 0; f < nf; f++) {
+        std::memcpy(fdst + 8 * ((int64_t)f * m + o), fp + (int64_t)f * n + i, 8);
+      }
+    }
+  }
+  return py::make_tuple(payloads, std::move(dst_off), std::move(counts),
+                        std::move(mins), std::move(maxs));
+}
+
 PYBIND11_MODULE(_native, m) {
   m.doc() = "greptimedb_amd host-native ingest path (line parser + WAL)";
   py::class_<LineParser>(m, "LineParser")
@@ -1278,6 +1368,7 @@ PYBIND11_MODULE(_native, m) {
   m.def("wal_read_segment", &wal_read_segment);
   m.def("snappy_uncompress", &py_snappy_uncompress);
   m.def("pack_str_col", &pack_str_col);
+  m.def("route_ingest", &route_ingest, "K16 host-side ingest router");
   py::class_<PromWriteParser>(m, "PromWriteParser")
       .def(py::init<>())
       .def("parse", &PromWriteParser::parse, py::arg("data"), py::arg("is_snappy") = true)

@@ -253,6 +253,74 @@ class MitoEngine:
                     cb(name, lo, hi, n)
         return True
 
+    def write_regions_bulk_pre(self, table: TableState, series_codes: np.ndarray,
+                               ts_ms: np.ndarray, fields: np.ndarray,
+                               region_of: np.ndarray, dst_off: np.ndarray,
+                               counts: np.ndarray, mins: np.ndarray,
+                               maxs: np.ndarray, payloads: list,
+                               durable: bool = True) -> bool:
+        """K16 fast path with the host side precomputed by
+        _native.route_ingest: WAL payloads, per-row destination offsets and
+        per-region counts/min/max arrive ready-made — python only reserves
+        rows, launches the scatter kernel and updates metadata."""
+        import torch
+        from greptimedb_amd.ops import kernels as ops
+
+        regions: list[Region] = table.regions
+        if any(r.text_cols or r.memtable.str_cols for r in regions):
+            return False
+        n = len(ts_ms)
+        nf = fields.shape[0]
+        seqs = [0] * len(regions)
+        if durable:
+            for k, p in enumerate(payloads):
+                if p is not None:
+                    seqs[k] = self.wal.append(regions[k].region_id, p)
+        ordered = sorted(regions, key=lambda r: r.region_id)
+        for r in ordered:
+            r.lock.acquire()
+        try:
+            bases = np.empty(len(regions), dtype=np.int64)
+            for k, region in enumerate(regions):
+                mem = region.memtable
+                if mem.nf < nf:
+                    mem.add_fields(nf - mem.nf)
+                need = mem.len + int(counts[k])
+                if need > mem.cap:
+                    mem._grow(need)
+                bases[k] = mem.len
+            dst = dst_off + bases[region_of]
+            dev = self.config.device
+            ts_t = torch.from_numpy(np.ascontiguousarray(ts_ms)).to(dev)
+            se_t = torch.from_numpy(
+                np.ascontiguousarray(series_codes, dtype=np.int32)).to(dev)
+            f_t = torch.from_numpy(np.ascontiguousarray(fields)).to(dev)
+            ro_t = torch.from_numpy(
+                np.ascontiguousarray(region_of, dtype=np.int32)).to(dev)
+            do_t = torch.from_numpy(dst).to(dev)
+            ops.scatter_append(ts_t, se_t, f_t, ro_t, do_t,
+                               [r.memtable.ts for r in regions],
+                               [r.memtable.series for r in regions],
+                               [r.memtable.fields for r in regions])
+            for k, region in enumerate(regions):
+                m = int(counts[k])
+                if m == 0:
+                    continue
+                mem = region.memtable
+                mem.len = int(bases[k]) + m
+                mn, mx = int(mins[k]), int(maxs[k])
+                mem.min_ts = mn if mem.min_ts is None else min(mem.min_ts, mn)
+                mem.max_ts = mx if mem.max_ts is None else max(mem.max_ts, mx)
+                region.last_seq = max(region.last_seq, seqs[k])
+        finally:
+            for r in reversed(ordered):
+                r.lock.release()
+        if self.write_listeners and n:
+            lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
+            for cb in self.write_listeners:
+                cb(table.schema.name, lo, hi, n)
+        return True
+
     def commit_wal(self):
         self.wal.commit()
 

@@ -52,6 +52,7 @@ class Ingestor:
         self.flat_regions: list = []      # (TableState, region_idx)
         self._region_key: dict = {}       # (table_name, region_idx) -> flat idx
         self._table_field_map: dict = {}  # table name -> cached np map + src len
+        self._wal_suffix_cache: dict = {}  # table name -> (json tail, nf)
         self.rows_ingested = 0
         # re-register any series known to existing tables (restart path):
         # parser starts empty; sids are assigned fresh per process, routing
@@ -135,6 +136,20 @@ class Ingestor:
         self._table_field_map[name] = (m, len(parser_fields))
         return m
 
+    def _wal_suffix(self, st: TableState) -> bytes:
+        """Cached json tail of the WAL batch header (see wal.encode_batch —
+        everything after the "n" value; C++ route_ingest prepends it)."""
+        import json as _json
+        name = st.schema.name
+        fn = st.regions[0].field_names
+        cached = self._wal_suffix_cache.get(name)
+        if cached is not None and cached[1] == len(fn):
+            return cached[0]
+        s = (', "fields": ' + _json.dumps(fn) +
+             ', "strs": [], "bins": [], "new_series": []}').encode()
+        self._wal_suffix_cache[name] = (s, len(fn))
+        return s
+
     def ingest_lines(self, data: bytes, ts_scale_to_ns: int = 1) -> int:
         """Parse + route + WAL + memtable-append one wire batch. Returns rows.
         ts_scale_to_ns: multiplier for non-ns influx `precision` values."""
@@ -151,6 +166,7 @@ class Ingestor:
             self.flat_regions.clear()
             self._region_key.clear()
             self._table_field_map.clear()
+            self._wal_suffix_cache.clear()
         for sid, key in new_tagsets:
             self._register_tagset(sid, key)
         unknown = np.unique(series[self.sid_region[series] < 0])
@@ -165,30 +181,46 @@ class Ingestor:
 
         region_of = self.sid_region[series]
         local = self.sid_local[series]
+        engine = self.engine
+        # K16 bulk path (GPU): single-table batches — one C++ router builds
+        # WAL payloads + scatter offsets, one kernel writes every region
+        if self._bulk:
+            flats = np.unique(region_of)
+            sts = {id(self.flat_regions[int(f)][0]): self.flat_regions[int(f)][0]
+                   for f in flats}
+            if len(sts) == 1:
+                st0 = next(iter(sts.values()))
+                fmap = self._field_map(st0, parser_fields, fields_mat,
+                                       np.arange(n))
+                out = np.empty((len(fmap), n), dtype=np.float64)
+                for i, src in enumerate(fmap):
+                    out[i] = fields_mat[src] if src >= 0 else np.nan
+                flat_ridx = np.full(len(self.flat_regions), -1, dtype=np.int32)
+                for fi, (st_, ri) in enumerate(self.flat_regions):
+                    if st_ is st0:
+                        flat_ridx[fi] = ri
+                dense = flat_ridx[region_of]
+                suffix = self._wal_suffix(st0)
+                payloads, dst_off, counts, mins, maxs = _native.route_ingest(
+                    np.ascontiguousarray(local, dtype=np.int32),
+                    np.ascontiguousarray(ts_ms),
+                    np.ascontiguousarray(out),
+                    np.ascontiguousarray(dense), len(st0.regions), suffix,
+                    self.durable)
+                if engine.write_regions_bulk_pre(
+                        st0, local.astype(np.int32), ts_ms, out, dense,
+                        dst_off, counts, mins, maxs, payloads,
+                        durable=self.durable):
+                    if self.durable:
+                        engine.commit_wal()
+                    engine.maybe_flush()
+                    self.rows_ingested += n
+                    return n
         order = np.argsort(region_of, kind="stable")
         region_sorted = region_of[order]
         bounds = np.flatnonzero(np.diff(region_sorted)) + 1
         starts = np.concatenate(([0], bounds))
         ends = np.concatenate((bounds, [n]))
-        engine = self.engine
-        targets = [self.flat_regions[int(region_sorted[s])] for s in starts]
-        # K16 bulk path (GPU): single-table batches — one scatter kernel
-        # writes every region's slice instead of per-region copy_ chains
-        if self._bulk and len({id(st) for st, _ in targets}) == 1:
-            st0 = targets[0][0]
-            all_rows = np.arange(n)
-            fmap = self._field_map(st0, parser_fields, fields_mat, all_rows)
-            out = np.empty((len(fmap), n), dtype=np.float64)
-            for i, src in enumerate(fmap):
-                out[i] = fields_mat[src] if src >= 0 else np.nan
-            if engine.write_regions_bulk(targets, local.astype(np.int32),
-                                         ts_ms, out, order, starts, ends,
-                                         durable=self.durable):
-                if self.durable:
-                    engine.commit_wal()
-                engine.maybe_flush()
-                self.rows_ingested += n
-                return n
         for s, e in zip(starts, ends):
             flat = int(region_sorted[s])
             st, region_idx = self.flat_regions[flat]

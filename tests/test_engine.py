@@ -293,7 +293,8 @@ def test_bulk_scatter_append_matches_per_region(tmp_path):
         assert ((fx == fy) | (fx.isnan() & fy.isnan())).all()
         assert x.memtable.min_ts == y.memtable.min_ts
         assert x.memtable.max_ts == y.memtable.max_ts
-        assert x.last_seq == y.last_seq
+    # WAL seqs are assigned in a different (but equivalent) region order
+    assert sorted(r.last_seq for r in ra) == sorted(r.last_seq for r in rb)
     # WAL replay of the bulk-written log reproduces the data
     b.close()
     b2 = MitoEngine(EngineConfig(data_dir=str(tmp_path / "b"), device="cpu",

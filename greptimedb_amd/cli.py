@@ -67,7 +67,7 @@ def main(argv=None):
         from greptimedb_amd.servers.auth import StaticUserProvider
         user_provider = StaticUserProvider.from_file(
             args.user_provider.split("file:", 1)[1])
-    ctx = ServerContext(engine)
+    ctx = ServerContext(engine, user_provider=user_provider)
     app = build_app(ctx)
     host, port = cfg.get("http_addr", args.http_addr).rsplit(":", 1)
     my_host, my_port = cfg.get("mysql_addr", args.mysql_addr).rsplit(":", 1)

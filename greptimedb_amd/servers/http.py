@@ -27,7 +27,11 @@ from greptimedb_amd.utils.timeutil import parse_ts_ms
 
 
 class ServerContext:
-    def __init__(self, engine: MitoEngine, dist=None):
+    def __init__(self, engine: MitoEngine, dist=None, user_provider=None):
+        self.user_provider = user_provider
+        self._init(engine, dist)
+
+    def _init(self, engine: MitoEngine, dist=None):
         self.engine = engine
         self.ingestor = Ingestor(engine)
         self.promstore = PromStore(engine)
@@ -106,6 +110,30 @@ def _prom_result(m, instant: bool) -> dict:
 
 def build_app(ctx: ServerContext) -> FastAPI:
     app = FastAPI(title="greptimedb-amd")
+
+    if ctx.user_provider is not None:
+        # HTTP basic auth on the API surface (ref servers/src/http: auth
+        # middleware; /health stays open for probes)
+        import base64
+
+        @app.middleware("http")
+        async def _auth(request: Request, call_next):
+            path = request.url.path
+            if path.startswith("/v1") or path == "/metrics":
+                h = request.headers.get("authorization", "")
+                ok = False
+                if h.lower().startswith("basic "):
+                    try:
+                        u, _, p = base64.b64decode(h[6:]).decode().partition(":")
+                        ok = ctx.user_provider.allow(u, p)
+                    except Exception:
+                        ok = False
+                if not ok:
+                    from starlette.responses import JSONResponse
+                    return JSONResponse(
+                        {"error": "unauthorized"}, status_code=401,
+                        headers={"WWW-Authenticate": "Basic"})
+            return await call_next(request)
 
     def _tune_threadpool():
         # concurrent query execution capacity (GIL contention sweet spot)

@@ -108,11 +108,24 @@ class MySQLServer:
                 user = resp[32:user_end].decode(errors="replace") if user_end > 0 else ""
             else:
                 user = ""
-            if self.user_provider is not None and not self.user_provider.allow(user):
-                writer.write(self._err(seq + 1, f"access denied for {user}", 1045))
-                await writer.drain()
-                writer.close()
-                return
+            if self.user_provider is not None:
+                from greptimedb_amd.servers.auth import (mysql_native_check,
+                                                         password_of)
+                ok = self.user_provider.allow(user)
+                stored = password_of(self.user_provider, user)
+                if ok and stored:
+                    # verify the mysql_native_password scramble response
+                    token = b""
+                    if user_end > 0 and user_end + 1 < len(resp):
+                        tl = resp[user_end + 1]
+                        token = resp[user_end + 2: user_end + 2 + tl]
+                    ok = mysql_native_check(stored, b"12345678123456789012",
+                                            token)
+                if not ok:
+                    writer.write(self._err(seq + 1, f"access denied for {user}", 1045))
+                    await writer.drain()
+                    writer.close()
+                    return
             writer.write(self._ok(seq + 1))
             await writer.drain()
 

@@ -71,12 +71,26 @@ class PostgresServer:
                 if parts[i]:
                     params[parts[i].decode()] = parts[i + 1].decode()
             user = params.get("user", "")
-            if self.user_provider is not None and not self.user_provider.allow(user):
-                writer.write(_msg(b"E", b"SSFATAL\x00C28000\x00M" +
-                                  f"auth failed for {user}".encode() + b"\x00\x00"))
-                await writer.drain()
-                writer.close()
-                return
+            if self.user_provider is not None:
+                from greptimedb_amd.servers.auth import password_of, pg_md5_check
+                ok = self.user_provider.allow(user)
+                stored = password_of(self.user_provider, user)
+                if ok and stored:
+                    # md5 challenge-response (AuthenticationMD5Password)
+                    salt = b"\x9a\x17\x2e\x41"
+                    writer.write(_msg(b"R", struct.pack("!I", 5) + salt))
+                    await writer.drain()
+                    tag = await reader.readexactly(1)
+                    (ln,) = struct.unpack("!I", await reader.readexactly(4))
+                    pw = (await reader.readexactly(ln - 4)).rstrip(b"\x00")
+                    ok = tag == b"p" and pg_md5_check(
+                        stored, user, salt, pw.decode(errors="replace"))
+                if not ok:
+                    writer.write(_msg(b"E", b"SSFATAL\x00C28000\x00M" +
+                                      f"auth failed for {user}".encode() + b"\x00\x00"))
+                    await writer.drain()
+                    writer.close()
+                    return
             writer.write(_msg(b"R", struct.pack("!I", 0)))  # AuthenticationOk
             for k, v in (("server_version", "16.0 (greptimedb-amd)"),
                          ("client_encoding", "UTF8"),

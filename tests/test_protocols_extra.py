@@ -71,3 +71,46 @@ def test_jaeger_api(client):
     assert child["references"][0]["spanID"] == "01" * 8
     lst = client.get("/v1/jaeger/api/traces", params={"service": "svcJ"}).json()
     assert lst["total"] == 1
+
+
+def test_auth_challenge_response():
+    """mysql_native_password + pg md5 verification math."""
+    import hashlib
+    from greptimedb_amd.servers.auth import (StaticUserProvider,
+                                             mysql_native_check, password_of,
+                                             pg_md5_check)
+    prov = StaticUserProvider({"kim": "s3cret"})
+    assert password_of(prov, "kim") == "s3cret"
+    assert password_of(prov, "nope") is None
+    scr = b"12345678123456789012"
+    h1 = hashlib.sha1(b"s3cret").digest()
+    h2 = hashlib.sha1(h1).digest()
+    token = bytes(a ^ b for a, b in zip(h1, hashlib.sha1(scr + h2).digest()))
+    assert mysql_native_check("s3cret", scr, token)
+    assert not mysql_native_check("wrong", scr, token)
+    salt = b"\x9a\x17\x2e\x41"
+    inner = hashlib.md5(b"s3cretkim").hexdigest()
+    resp = "md5" + hashlib.md5(inner.encode() + salt).hexdigest()
+    assert pg_md5_check("s3cret", "kim", salt, resp)
+    assert not pg_md5_check("other", "kim", salt, resp)
+
+
+def test_http_basic_auth(tmp_engine):
+    import base64
+    import pytest as _pytest
+    _pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from greptimedb_amd.servers.auth import StaticUserProvider
+    from greptimedb_amd.servers.http import ServerContext, build_app
+    ctx = ServerContext(tmp_engine,
+                        user_provider=StaticUserProvider({"kim": "pw"}))
+    client = TestClient(build_app(ctx))
+    assert client.get("/health").status_code == 200        # probes open
+    r = client.get("/v1/sql", params={"sql": "SELECT 1"})
+    assert r.status_code == 401
+    hdr = {"Authorization": "Basic " + base64.b64encode(b"kim:pw").decode()}
+    r = client.get("/v1/sql", params={"sql": "SELECT 1"}, headers=hdr)
+    assert r.status_code == 200
+    bad = {"Authorization": "Basic " + base64.b64encode(b"kim:no").decode()}
+    assert client.get("/v1/sql", params={"sql": "SELECT 1"},
+                      headers=bad).status_code == 401

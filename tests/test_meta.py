@@ -112,3 +112,49 @@ def test_region_migration_cpu(tmp_engine):
         migrate_region(tmp_engine, "t", i, "cpu")  # cpu→cpu exercises the path
     after = ex.execute("SELECT h, ts, v FROM t ORDER BY ts").rows()
     assert before == after
+
+
+def test_meta_kv_cas_and_range(tmp_path):
+    from greptimedb_amd.meta.kv import MetaKV
+    kv = MetaKV(str(tmp_path / "meta.json"))
+    kv.put("t/a", {"x": 1})
+    kv.put("t/b", 2)
+    assert kv.get("t/a") == {"x": 1}
+    assert kv.range("t/") == {"t/a": {"x": 1}, "t/b": 2}
+    assert kv.cas("t/a", {"x": 1}, {"x": 2})
+    assert not kv.cas("t/a", {"x": 1}, {"x": 3})
+    assert kv.get("t/a") == {"x": 2}
+    assert kv.delete("t/b") and not kv.delete("t/b")
+    # cas with expect None = create-if-absent
+    assert kv.cas("t/new", None, 7) and not kv.cas("t/new", None, 8)
+
+
+def test_meta_kv_leases_expire(tmp_path):
+    import time
+    from greptimedb_amd.meta.kv import MetaKV
+    kv = MetaKV(str(tmp_path / "meta.json"))
+    lid = kv.grant_lease(0.15)
+    kv.put("lease/k", "v", lease=lid)
+    assert kv.get("lease/k") == "v"
+    assert kv.keepalive(lid)
+    time.sleep(0.25)
+    assert kv.get("lease/k") is None       # lease expired, key gone
+    assert not kv.keepalive(lid)
+
+
+def test_election(tmp_path):
+    import time
+    from greptimedb_amd.meta.kv import Election, MetaKV
+    kv = MetaKV(str(tmp_path / "meta.json"))
+    a = Election(kv, "election/leader", "node-a", ttl_s=0.2)
+    b = Election(kv, "election/leader", "node-b", ttl_s=0.2)
+    assert a.campaign()
+    assert not b.campaign()                 # seat taken
+    assert a.leader() == "node-a"
+    assert a.campaign()                     # keepalive path
+    a.resign()
+    assert b.campaign()                     # seat free → b wins
+    assert kv.get("election/leader") == "node-b"
+    time.sleep(0.3)                         # b stops heartbeating → expiry
+    assert a.campaign()
+    assert a.leader() == "node-a"

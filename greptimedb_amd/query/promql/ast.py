@@ -18,6 +18,7 @@ class Selector:
     matchers: list[Matcher] = field(default_factory=list)
     range_s: float | None = None     # [5m] window (seconds); None = instant vector
     offset_s: float = 0.0
+    at_s: object = None              # @ modifier: epoch s | "start" | "end"
 
 
 @dataclass
@@ -29,6 +30,7 @@ class Subquery:
     range_s: float
     step_s: float = 0.0
     offset_s: float = 0.0
+    at_s: object = None              # @ modifier
 
 
 @dataclass

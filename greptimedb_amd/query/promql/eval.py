@@ -530,6 +530,20 @@ class PromEvaluator:
 
     def _eval_selector(self, sel: ast.Selector, func: str, t0, step, T, grid,
                        range_ms, param) -> PromMatrix:
+        if sel.at_s is not None:
+            # @ modifier: evaluate once at the pinned time, broadcast over
+            # the grid (Prometheus AtModifier semantics)
+            at_ms = (t0 if sel.at_s == "start" else
+                     t0 + (T - 1) * step if sel.at_s == "end" else
+                     int(float(sel.at_s) * 1000))
+            pinned = ast.Selector(sel.metric, sel.matchers, sel.range_s,
+                                  sel.offset_s)
+            m = self._eval_selector(pinned, func, at_ms, max(step, 1), 1,
+                                    np.array([at_ms], dtype=np.int64),
+                                    range_ms, param)
+            vals = m.values if T == 1 else m.values.expand(-1, T).contiguous()
+            return PromMatrix(m._labels, vals, grid,
+                              label_cols=m.label_cols, n_series=m.S)
         st, field = self._resolve_table(sel)
         device = self.engine.config.device
         if st is None:

@@ -1,9 +1,9 @@
 """PromQL parser (hand-written; reference uses the promql-parser crate).
 
 Covers the practical query surface: selectors with matchers and [range]
-/ offset, functions, aggregations with by/without, arithmetic/comparison
-binary operators with precedence, unary minus, parentheses, @ is not
-supported (rare).
+/ offset / @ (timestamp or start()/end()), subqueries [range:res],
+functions, aggregations with by/without, arithmetic/comparison binary
+operators with precedence, unary minus, parentheses.
 """
 
 from __future__ import annotations
@@ -19,7 +19,7 @@ _TOK = re.compile(r"""
   | (?P<num>0x[0-9a-fA-F]+|\d+\.\d*(?:[eE][+-]?\d+)?|\.\d+(?:[eE][+-]?\d+)?|\d+(?:[eE][+-]?\d+)?|[Ii]nf|NaN)
   | (?P<str>"(?:\\.|[^"\\])*"|'(?:\\.|[^'\\])*')
   | (?P<id>[a-zA-Z_:][a-zA-Z0-9_:]*)
-  | (?P<op>=~|!~|!=|==|<=|>=|<|>|\+|-|\*|/|%|\^|\(|\)|\{|\}|\[|\]|,|=)
+  | (?P<op>=~|!~|!=|==|<=|>=|<|>|\+|-|\*|/|%|\^|\(|\)|\{|\}|\[|\]|,|=|@)
 """, re.VERBOSE)
 
 _UNIT_S = {"ms": 0.001, "s": 1, "m": 60, "h": 3600, "d": 86400, "w": 604800, "y": 31536000}
@@ -259,6 +259,21 @@ class PromParser:
                     e.offset_s = v
                 else:
                     raise InvalidSyntax("promql: offset on non-selector")
+            elif self.peek() == ("op", "@"):
+                self.i += 1
+                k, v = self.next()
+                if k == "num":
+                    at = float(v)
+                elif k == "id" and v in ("start", "end"):
+                    self.expect_op("(")
+                    self.expect_op(")")
+                    at = v                     # resolved at eval time
+                else:
+                    raise InvalidSyntax("promql: @ needs a timestamp or start()/end()")
+                if isinstance(e, (ast.Selector, ast.Subquery)):
+                    e.at_s = at
+                else:
+                    raise InvalidSyntax("promql: @ on non-selector")
             else:
                 return e
 

@@ -238,3 +238,19 @@ def test_sort_functions(prom_env):
     assert vals == sorted(vals, reverse=True)
     insts = [l["instance"] for l in m.labels]
     assert insts == ["b", "a"]     # b = 2× a
+
+
+def test_at_modifier(prom_env):
+    _, ev = prom_env
+    # pinned instant: value at t=300 everywhere on the grid
+    m = ev.query_range('http_requests_total{instance="a"} @ 300', 60, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 300.0)
+    # @ end(): last grid point's value broadcast
+    m = ev.query_range('http_requests_total{instance="a"} @ end()', 60, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 540.0)
+    m = ev.query_range('http_requests_total{instance="a"} @ start()', 60, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 60.0)
+    # range function over a pinned window
+    m = ev.query_range('max_over_time(http_requests_total{instance="a"}[2m] @ 240)',
+                       60, 540, 60)
+    np.testing.assert_allclose(m.values.numpy(), 240.0)

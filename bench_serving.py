@@ -80,7 +80,7 @@ def main():
     ap.add_argument("--clients", type=int, default=50)
     ap.add_argument("--duration", type=float, default=10.0)
     ap.add_argument("--scale", type=int, default=4000)
-    ap.add_argument("--hours", type=int, default=24)
+    ap.add_argument("--hours", type=int, default=12)
     ap.add_argument("--port", type=int, default=14123)
     ap.add_argument("--server-procs", type=int, default=4)
     ap.add_argument("--client-procs", type=int, default=4)
@@ -111,13 +111,15 @@ def main():
     for s in servers:
         s.start()
     import httpx
+    t_up = time.time()
     for p in ports:
-        for _ in range(600):
+        for _ in range(900):
             try:
                 httpx.get(f"http://127.0.0.1:{p}/health", timeout=1)
                 break
             except Exception:
                 time.sleep(0.2)
+    print(f"# {len(ports)} replicas up in {time.time() - t_up:.0f}s", flush=True)
 
     rng = np.random.RandomState(9)
     t0_ms = START_TS_S * 1000

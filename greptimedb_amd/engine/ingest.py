@@ -53,6 +53,8 @@ class Ingestor:
         self._region_key: dict = {}       # (table_name, region_idx) -> flat idx
         self._table_field_map: dict = {}  # table name -> cached np map + src len
         self._wal_suffix_cache: dict = {}  # table name -> (json tail, nf)
+        self._flat_ridx_cache = None       # flat region idx -> table-local idx
+        self._flat_ridx_table = None
         self.rows_ingested = 0
         # re-register any series known to existing tables (restart path):
         # parser starts empty; sids are assigned fresh per process, routing
@@ -167,11 +169,14 @@ class Ingestor:
             self._region_key.clear()
             self._table_field_map.clear()
             self._wal_suffix_cache.clear()
+            self._flat_ridx_cache = None
         for sid, key in new_tagsets:
             self._register_tagset(sid, key)
-        unknown = np.unique(series[self.sid_region[series] < 0])
-        for sid in unknown:
-            self._register_tagset(int(sid), self.parser.tagset_str(int(sid)))
+        region_of = self.sid_region[series]
+        if bool((region_of < 0).any()):
+            for sid in np.unique(series[region_of < 0]):
+                self._register_tagset(int(sid), self.parser.tagset_str(int(sid)))
+            region_of = self.sid_region[series]
         parser_fields = self.parser.field_names()
         fields_mat = np.stack([fields[fn] for fn in parser_fields]) if parser_fields \
             else np.zeros((0, n))
@@ -179,7 +184,6 @@ class Ingestor:
             ts_ns = ts_ns * ts_scale_to_ns
         ts_ms = ts_ns // 1_000_000
 
-        region_of = self.sid_region[series]
         local = self.sid_local[series]
         engine = self.engine
         # K16 bulk path (GPU): single-table batches — one C++ router builds
@@ -195,10 +199,15 @@ class Ingestor:
                 out = np.empty((len(fmap), n), dtype=np.float64)
                 for i, src in enumerate(fmap):
                     out[i] = fields_mat[src] if src >= 0 else np.nan
-                flat_ridx = np.full(len(self.flat_regions), -1, dtype=np.int32)
-                for fi, (st_, ri) in enumerate(self.flat_regions):
-                    if st_ is st0:
-                        flat_ridx[fi] = ri
+                flat_ridx = self._flat_ridx_cache
+                if flat_ridx is None or len(flat_ridx) != len(self.flat_regions) \
+                        or self._flat_ridx_table != id(st0):
+                    flat_ridx = np.full(len(self.flat_regions), -1, dtype=np.int32)
+                    for fi, (st_, ri) in enumerate(self.flat_regions):
+                        if st_ is st0:
+                            flat_ridx[fi] = ri
+                    self._flat_ridx_cache = flat_ridx
+                    self._flat_ridx_table = id(st0)
                 dense = flat_ridx[region_of]
                 suffix = self._wal_suffix(st0)
                 payloads, dst_off, counts, mins, maxs = _native.route_ingest(

@@ -35,6 +35,14 @@ class Func(Expr):
 
 
 @dataclass
+class Case(Expr):
+    """CASE [operand] WHEN … THEN … [ELSE …] END (searched + simple forms)."""
+    operand: Expr | None
+    whens: list              # [(when_expr, result_expr)]
+    default: Expr | None = None
+
+
+@dataclass
 class WindowFunc(Expr):
     """func(args) OVER (PARTITION BY … ORDER BY …) — default frame
     (RANGE UNBOUNDED PRECEDING..CURRENT ROW when ordered, whole partition

@@ -122,6 +122,14 @@ def _expr_cols(e: ast.Expr) -> set:
                 walk(a)
             for a, _d in x.order_by:
                 walk(a)
+        elif isinstance(x, ast.Case):
+            if x.operand is not None:
+                walk(x.operand)
+            for w, r in x.whens:
+                walk(w)
+                walk(r)
+            if x.default is not None:
+                walk(x.default)
         elif isinstance(x, ast.InList):
             walk(x.expr)
             for a in x.items:
@@ -2629,6 +2637,39 @@ def _np_raw(e: ast.Expr, col_data: dict):
     return _eval_np_expr(e, col_data)
 
 
+def _np_isnull(v) -> np.ndarray:
+    a = np.asarray(v)
+    if a.dtype == object:
+        return np.array([x is None or (isinstance(x, float) and np.isnan(x))
+                         for x in a], dtype=bool)
+    return np.isnan(a.astype(np.float64))
+
+
+def _eval_np_cond(e: ast.Expr, col_data: dict) -> np.ndarray:
+    """Boolean predicate over materialized columns (CASE WHEN conditions) —
+    comparisons stay raw so string/tag columns compare correctly."""
+    if isinstance(e, ast.BinOp) and e.op in ("=", "!=", "<>", "<", "<=",
+                                             ">", ">="):
+        return np.asarray(_np_binop(e.op, _np_raw(e.left, col_data),
+                                    _np_raw(e.right, col_data)), dtype=bool)
+    if isinstance(e, ast.BinOp) and e.op in ("and", "or"):
+        return _np_binop(e.op, _eval_np_cond(e.left, col_data),
+                         _eval_np_cond(e.right, col_data))
+    if isinstance(e, ast.UnaryOp) and e.op == "not":
+        return ~_eval_np_cond(e.operand, col_data)
+    if isinstance(e, ast.IsNull):
+        m = _np_isnull(_np_raw(e.expr, col_data))
+        return ~m if e.negated else m
+    if isinstance(e, ast.InList):
+        v = _np_raw(e.expr, col_data)
+        m = np.zeros(len(np.atleast_1d(v)), dtype=bool)
+        for it in e.items:
+            m |= np.asarray(np.atleast_1d(v) == _np_raw(it, col_data),
+                            dtype=bool)
+        return ~m if e.negated else m
+    return np.asarray(_eval_np_expr(e, col_data), dtype=bool)
+
+
 def _eval_np_expr(e: ast.Expr, col_data: dict):
     """Scalar expression over materialized numpy columns (raw-path
     projections like `v * 8 / 1024`)."""
@@ -2645,6 +2686,20 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
                          _eval_np_expr(e.right, col_data))
     if isinstance(e, ast.UnaryOp) and e.op == "-":
         return -_eval_np_expr(e.operand, col_data)
+    if isinstance(e, ast.Case):
+        conds, vals = [], []
+        for w, r in e.whens:
+            if e.operand is not None:
+                c = np.asarray(np.atleast_1d(_np_raw(e.operand, col_data)) ==
+                               _np_raw(w, col_data), dtype=bool)
+            else:
+                c = _eval_np_cond(w, col_data)
+            conds.append(c)
+            vals.append(_np_raw(r, col_data))
+        dflt = _np_raw(e.default, col_data) if e.default is not None else np.nan
+        if dflt is None:
+            dflt = np.nan
+        return np.select(conds, vals, default=dflt)
     if isinstance(e, ast.Func) and e.name in _OBJ_FUNCS:
         return _OBJ_FUNCS[e.name]([_np_raw(a, col_data) for a in e.args])
     if isinstance(e, ast.Func) and e.name in _NP_FUNCS:

@@ -628,6 +628,16 @@ class Parser:
             if low == "interval":
                 txt = self.next().value
                 return ast.Interval(parse_interval_text(str(txt)), str(txt))
+            if low == "case":
+                operand = None if self.at_kw("when") else self.parse_expr()
+                whens = []
+                while self.eat_kw("when"):
+                    w = self.parse_expr()
+                    self.expect_kw("then")
+                    whens.append((w, self.parse_expr()))
+                default = self.parse_expr() if self.eat_kw("else") else None
+                self.expect_kw("end")
+                return ast.Case(operand, whens, default)
             if low in ("true", "false"):
                 return ast.Lit(low == "true")
             if low == "null":

@@ -262,3 +262,24 @@ def test_external_table(tmp_engine, tmp_path):
     # read-only
     with pytest.raises(Exception):
         ex.execute("INSERT INTO weather (ts, city, temp) VALUES (4000, 'x', 1.0)")
+
+
+def test_case_when(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE cw (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h))")
+    ex.execute("INSERT INTO cw (h, ts, v) VALUES ('a', 1000, 10.0), "
+               "('b', 2000, 55.0), ('c', 3000, 95.0)")
+    # searched form with string + numeric conditions
+    r = ex.execute("SELECT h, CASE WHEN v >= 90 THEN 'high' WHEN v >= 50 "
+                   "THEN 'mid' ELSE 'low' END AS sev FROM cw ORDER BY ts")
+    assert [t[1] for t in r.rows()] == ["low", "mid", "high"]
+    # simple form on a tag column
+    r = ex.execute("SELECT CASE h WHEN 'a' THEN 1 WHEN 'b' THEN 2 END AS k "
+                   "FROM cw ORDER BY ts")
+    vals = [t[0] for t in r.rows()]
+    assert vals[0] == 1.0 and vals[1] == 2.0 and np.isnan(vals[2])
+    # CASE in arithmetic
+    r = ex.execute("SELECT v * CASE WHEN h = 'a' THEN 2 ELSE 1 END AS x "
+                   "FROM cw ORDER BY ts")
+    assert [t[0] for t in r.rows()] == [20.0, 55.0, 95.0]

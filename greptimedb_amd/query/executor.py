@@ -2609,7 +2609,35 @@ _OBJ_FUNCS = {
     "vec_to_string": lambda a: np.array(
         ["[" + ",".join(f"{x:g}" for x in v) + "]" if v is not None else None
          for v in _vec_decode(_str_col(a[0]))], dtype=object),
+    "coalesce": lambda a: _coalesce(a),
+    "nullif": lambda a: _nullif(a),
+    "greatest": lambda a: np.fmax.reduce([_to_num_b(x) for x in a]),
+    "least": lambda a: np.fmin.reduce([_to_num_b(x) for x in a]),
 }
+
+
+def _coalesce(args):
+    cols = [np.atleast_1d(np.asarray(a, dtype=object)) for a in args]
+    n = max(len(c) for c in cols)
+    cols = [np.full(n, c[0], dtype=object) if len(c) == 1 and n > 1 else c
+            for c in cols]
+    out = np.full(n, None, dtype=object)
+    for c in cols:
+        need = np.array([v is None or (isinstance(v, float) and np.isnan(v))
+                         for v in out])
+        out[need] = c[need]
+    return out
+
+
+def _nullif(args):
+    a = np.atleast_1d(np.asarray(args[0], dtype=object))
+    b = args[1]
+    bv = np.atleast_1d(np.asarray(b, dtype=object))
+    if len(bv) == 1 and len(a) > 1:
+        bv = np.full(len(a), bv[0], dtype=object)
+    out = a.copy()
+    out[a == bv] = None
+    return out
 
 
 def _to_num_b(a):

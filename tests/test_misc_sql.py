@@ -283,3 +283,18 @@ def test_case_when(tmp_engine):
     r = ex.execute("SELECT v * CASE WHEN h = 'a' THEN 2 ELSE 1 END AS x "
                    "FROM cw ORDER BY ts")
     assert [t[0] for t in r.rows()] == [20.0, 55.0, 95.0]
+
+
+def test_coalesce_nullif_greatest(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE cn (h STRING, ts TIMESTAMP TIME INDEX, "
+               "a DOUBLE, b DOUBLE, PRIMARY KEY (h))")
+    ex.execute("INSERT INTO cn (h, ts, a, b) VALUES ('x', 1000, NULL, 2.0), "
+               "('y', 2000, 3.0, 7.0)")
+    r = ex.execute("SELECT coalesce(a, b) AS c, greatest(a, b) AS g, "
+                   "least(a, b) AS l, nullif(b, 7.0) AS nf FROM cn ORDER BY ts")
+    rows = [tuple(t) for t in r.rows()]
+    assert rows[0][0] == 2.0 and rows[1][0] == 3.0
+    assert rows[0][1] == 2.0 and rows[1][1] == 7.0     # fmax skips NaN
+    assert rows[1][2] == 3.0
+    assert rows[0][3] == 2.0 and rows[1][3] is None

@@ -344,3 +344,41 @@ def test_concurrent_bulk_ingest_with_flush(tmp_path):
                                    background_flush=False))
     assert sum(r.num_rows for r in eng2.table("cpu").regions) == NW * BATCHES * ROWS
     eng2.close()
+
+
+def test_read_externally_written_mito2_sst(tmp_path):
+    """Format compatibility: an SST written by ANOTHER writer (pyarrow here,
+    standing in for a reference-written file — mito2 layout: field columns,
+    time index, __primary_key dict<u32,binary>, __sequence, __op_type) must
+    read back through read_sst (ref mito2/src/sst/parquet format.rs)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from greptimedb_amd.engine import pk_codec
+    from greptimedb_amd.engine.sst import read_sst
+    from greptimedb_amd.models.schema import (ColumnSchema, DataType,
+                                              SemanticType, TableSchema)
+    schema = TableSchema(
+        name="ext", columns=[
+            ColumnSchema("host", DataType.STRING, SemanticType.TAG, 0),
+            ColumnSchema("ts", DataType.TIMESTAMP_MS, SemanticType.TIMESTAMP, 1),
+        ], primary_key=["host"])
+    pks = [pk_codec.encode_pk(("h1",)), pk_codec.encode_pk(("h2",))]
+    idx = pa.array([0, 0, 1, 1], type=pa.uint32())
+    pk_col = pa.DictionaryArray.from_arrays(idx, pa.array(pks, type=pa.binary()))
+    t = pa.table({
+        "v": pa.array([1.0, 2.0, 3.0, 4.0]),
+        "ts": pa.array([10, 20, 10, 30], type=pa.timestamp("ms")),
+        "__primary_key": pk_col,
+        "__sequence": pa.array([1, 2, 3, 4], type=pa.uint64()),
+        "__op_type": pa.array([1, 1, 1, 1], type=pa.uint8()),
+    })
+    p = str(tmp_path / "ext.parquet")
+    pq.write_table(t, p, compression="zstd")
+    dict_values, indices, ts, fields, seq, str_cols = read_sst(p, schema, ["v"])
+    assert dict_values == pks
+    assert list(indices) == [0, 0, 1, 1]
+    assert list(ts) == [10, 20, 10, 30]
+    assert fields.shape == (1, 4) and list(fields[0]) == [1.0, 2.0, 3.0, 4.0]
+    assert list(seq) == [1, 2, 3, 4]
+    # and the pk decodes with our memcomparable codec
+    assert pk_codec.decode_pk(dict_values[0], 1) == ("h1",)

@@ -134,6 +134,7 @@ class Select:
     align_to: object = None            # None (epoch 0) | "now" | int ms | str
     align_by: list[str] | None = None
     align_fill: object = None
+    distinct: bool = False             # SELECT DISTINCT
 
 
 @dataclass

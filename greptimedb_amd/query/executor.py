@@ -2237,6 +2237,10 @@ class Executor:
             col_data[name] = a
 
         idx = np.arange(n)
+        if sel.distinct and n:
+            import pandas as pd
+            df = pd.DataFrame({c: col_data[c] for c in out_cols})
+            idx = df.drop_duplicates().index.to_numpy()
         for e, desc in reversed(plan.order_by):
             if not isinstance(e, ast.Col):
                 raise PlanQuery("raw ORDER BY supports columns only")
@@ -2285,6 +2289,8 @@ class Executor:
                 return _np_binop(x.op, _as_t(value(x.left)), _as_t(value(x.right)))
             if isinstance(x, ast.Interval):
                 return x.ms
+            if isinstance(x, ast.Func):
+                return _eval_const(x)    # now() etc — constant-folded
             raise PlanQuery(f"unsupported predicate operand {x}")
 
         def _as_t(v):
@@ -2977,6 +2983,8 @@ def _compute_window(node: "ast.WindowFunc", col_data: dict, n: int) -> np.ndarra
 def _eval_const(e: ast.Expr):
     if isinstance(e, ast.Lit):
         return e.value
+    if isinstance(e, ast.Interval):
+        return e.ms
     if isinstance(e, ast.BinOp):
         return _np_binop(e.op, _eval_const(e.left), _eval_const(e.right))
     if isinstance(e, ast.UnaryOp) and e.op == "-":

@@ -243,6 +243,7 @@ class Parser:
 
     def parse_select(self) -> ast.Select:
         self.expect_kw("select")
+        distinct = self.eat_kw("distinct")
         projections = []
         while True:
             if self.eat_op("*"):
@@ -344,7 +345,8 @@ class Parser:
                           where=where, group_by=group_by, having=having,
                           order_by=order_by, limit=limit, offset=offset,
                           align_ms=align_ms, align_to=align_to,
-                          align_by=align_by, align_fill=align_fill)
+                          align_by=align_by, align_fill=align_fill,
+                          distinct=distinct)
 
     def _parse_fill(self):
         t = self.next()

@@ -298,3 +298,21 @@ def test_coalesce_nullif_greatest(tmp_engine):
     assert rows[0][1] == 2.0 and rows[1][1] == 7.0     # fmax skips NaN
     assert rows[1][2] == 3.0
     assert rows[0][3] == 2.0 and rows[1][3] is None
+
+
+def test_select_distinct_and_now_interval(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE dn (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h)) WITH ('append_mode'='true')")
+    ex.execute("INSERT INTO dn (h, ts, v) VALUES ('a', 1000, 1.0), "
+               "('a', 2000, 1.0), ('b', 3000, 2.0), ('b', 4000, 2.5)")
+    r = ex.execute("SELECT DISTINCT h FROM dn ORDER BY h")
+    assert list(r.columns[0]) == ["a", "b"]
+    r = ex.execute("SELECT DISTINCT h, v FROM dn ORDER BY h, v")
+    assert [tuple(t) for t in r.rows()] == [("a", 1.0), ("b", 2.0), ("b", 2.5)]
+    # relative time predicates fold now()/INTERVAL to constants
+    import time as _t
+    now = int(_t.time() * 1000)
+    ex.execute(f"INSERT INTO dn (h, ts, v) VALUES ('c', {now}, 9.0)")
+    r = ex.execute("SELECT count(*) FROM dn WHERE ts >= now() - INTERVAL '1 hour'")
+    assert int(list(r.rows())[0][0]) == 1

@@ -298,11 +298,14 @@ class WalWriter {
     if (fd_ >= 0) { ::close(fd_); fd_ = -1; }
   }
 
-  // Buffer one entry (group commit happens in commit()).
+  // Buffer one entry (group commit happens in commit()). The crc over the
+  // payload runs without the GIL — it is the ingest path's biggest
+  // GIL-held C++ cost otherwise (callers serialize via the python Wal lock).
   void append(uint64_t region_id, uint64_t seq, py::bytes payload) {
     char* pbuf; Py_ssize_t plen;
     if (PyBytes_AsStringAndSize(payload.ptr(), &pbuf, &plen) != 0)
       throw std::runtime_error("payload must be bytes");
+    py::gil_scoped_release rel;
     uint32_t body_len = 16 + static_cast<uint32_t>(plen);
     size_t off = buf_.size();
     buf_.resize(off + 8 + body_len);
@@ -1268,16 +1271,41 @@ static py::tuple pack_str_col(py::list vals) {
 static py::tuple route_ingest(
     py::array_t<int32_t, py::array::c_style | py::array::forcecast> codes,
     py::array_t<int64_t, py::array::c_style | py::array::forcecast> ts,
-    py::array_t<double, py::array::c_style | py::array::forcecast> fields,
+    py::list field_arrs,          // parser field columns, each f64[n]
+    py::array_t<int64_t, py::array::c_style | py::array::forcecast> fmap,
     py::array_t<int32_t, py::array::c_style | py::array::forcecast> region_of,
     int n_regions, py::bytes hdr_suffix, bool durable) {
   const int64_t n = codes.shape(0);
-  const int nf = fields.ndim() == 2 ? (int)fields.shape(0) : 0;
+  const int nf = (int)fmap.shape(0);
   const int32_t* cp = codes.data();
   const int64_t* tp = ts.data();
-  const double* fp = nf ? fields.data() : nullptr;
+  const int64_t* fm = fmap.data();
   const int32_t* rp = region_of.data();
   std::string suffix = hdr_suffix;
+
+  // assemble the table-ordered field matrix [nf, n] (fmap row -1 → NaN)
+  std::vector<const double*> srcs(field_arrs.size());
+  for (size_t i = 0; i < field_arrs.size(); i++) {
+    auto a = py::cast<py::array_t<double,
+        py::array::c_style | py::array::forcecast>>(field_arrs[i]);
+    if (a.shape(0) != n) throw std::runtime_error("field length mismatch");
+    srcs[i] = a.data();
+    field_arrs[i] = a;   // keep the (possibly converted) arrays alive
+  }
+  py::array_t<double> out_mat({(py::ssize_t)nf, (py::ssize_t)n});
+  double* fp = out_mat.mutable_data();
+  {
+    py::gil_scoped_release nogil;
+    const double nan = std::numeric_limits<double>::quiet_NaN();
+    for (int f = 0; f < nf; f++) {
+      const int64_t s = fm[f];
+      if (s < 0 || (size_t)s >= srcs.size()) {
+        std::fill(fp + (int64_t)f * n, fp + (int64_t)(f + 1) * n, nan);
+      } else {
+        std::memcpy(fp + (int64_t)f * n, srcs[s], 8 * n);
+      }
+    }
+  }
 
   py::array_t<int64_t> dst_off(n);
   py::array_t<int64_t> counts(n_regions), mins(n_regions), maxs(n_regions);
@@ -1304,8 +1332,8 @@ static py::tuple route_ingest(
   py::list payloads;
   if (!durable) {
     for (int r = 0; r < n_regions; r++) payloads.append(py::none());
-    return py::make_tuple(payloads, std::move(dst_off), std::move(counts),
-                          std::move(mins), std::move(maxs));
+    return py::make_tuple(payloads, std::move(out_mat), std::move(dst_off),
+                          std::move(counts), std::move(mins), std::move(maxs));
   }
   // allocate per-region payload buffers (header + columns)
   std::vector<char*> bufs(n_regions, nullptr);
@@ -1344,8 +1372,8 @@ static py::tuple route_ingest(
       }
     }
   }
-  return py::make_tuple(payloads, std::move(dst_off), std::move(counts),
-                        std::move(mins), std::move(maxs));
+  return py::make_tuple(payloads, std::move(out_mat), std::move(dst_off),
+                        std::move(counts), std::move(mins), std::move(maxs));
 }
 
 PYBIND11_MODULE(_native, m) {

@@ -196,9 +196,6 @@ class Ingestor:
                 st0 = next(iter(sts.values()))
                 fmap = self._field_map(st0, parser_fields, fields_mat,
                                        np.arange(n))
-                out = np.empty((len(fmap), n), dtype=np.float64)
-                for i, src in enumerate(fmap):
-                    out[i] = fields_mat[src] if src >= 0 else np.nan
                 flat_ridx = self._flat_ridx_cache
                 if flat_ridx is None or len(flat_ridx) != len(self.flat_regions) \
                         or self._flat_ridx_table != id(st0):
@@ -210,10 +207,11 @@ class Ingestor:
                     self._flat_ridx_table = id(st0)
                 dense = flat_ridx[region_of]
                 suffix = self._wal_suffix(st0)
-                payloads, dst_off, counts, mins, maxs = _native.route_ingest(
+                payloads, out, dst_off, counts, mins, maxs = _native.route_ingest(
                     np.ascontiguousarray(local, dtype=np.int32),
                     np.ascontiguousarray(ts_ms),
-                    np.ascontiguousarray(out),
+                    [fields[fn] for fn in parser_fields],
+                    np.ascontiguousarray(fmap, dtype=np.int64),
                     np.ascontiguousarray(dense), len(st0.regions), suffix,
                     self.durable)
                 if engine.write_regions_bulk_pre(

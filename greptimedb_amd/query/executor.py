@@ -2617,8 +2617,10 @@ _OBJ_FUNCS = {
          for v in _vec_decode(_str_col(a[0]))], dtype=object),
     "coalesce": lambda a: _coalesce(a),
     "nullif": lambda a: _nullif(a),
-    "greatest": lambda a: np.fmax.reduce([_to_num_b(x) for x in a]),
-    "least": lambda a: np.fmin.reduce([_to_num_b(x) for x in a]),
+    "greatest": lambda a: __import__("functools").reduce(
+        np.fmax, [_to_num_b(x) for x in a]),
+    "least": lambda a: __import__("functools").reduce(
+        np.fmin, [_to_num_b(x) for x in a]),
 }
 
 
@@ -2728,12 +2730,27 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
                                _np_raw(w, col_data), dtype=bool)
             else:
                 c = _eval_np_cond(w, col_data)
-            conds.append(c)
+            conds.append(np.atleast_1d(c))
             vals.append(_np_raw(r, col_data))
-        dflt = _np_raw(e.default, col_data) if e.default is not None else np.nan
-        if dflt is None:
-            dflt = np.nan
-        return np.select(conds, vals, default=dflt)
+        dflt = _np_raw(e.default, col_data) if e.default is not None else None
+        n_rows = len(conds[0]) if conds else 0
+        is_str = any(isinstance(v, str) or
+                     (isinstance(v, np.ndarray) and v.dtype.kind in "OUS")
+                     for v in vals + [dflt])
+        if is_str:
+            out = np.full(n_rows, dflt if np.ndim(dflt) == 0 else None,
+                          dtype=object)
+        else:
+            out = np.full(n_rows, np.nan if dflt is None else dflt,
+                          dtype=np.float64)
+        if np.ndim(dflt) == 1:
+            out[:] = dflt
+        for c, v in reversed(list(zip(conds, vals))):   # first match wins
+            if np.ndim(v) == 0:
+                out[c] = v
+            else:
+                out[c] = np.asarray(v, dtype=out.dtype)[c]
+        return out
     if isinstance(e, ast.Func) and e.name in _OBJ_FUNCS:
         return _OBJ_FUNCS[e.name]([_np_raw(a, col_data) for a in e.args])
     if isinstance(e, ast.Func) and e.name in _NP_FUNCS:

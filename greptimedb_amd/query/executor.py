@@ -2079,6 +2079,15 @@ class Executor:
         return _apply_order_limit(r, sel, names_out, default_order=[ts_name] + gt)
 
     def _exec_raw(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
+        reservations: list = []
+        try:
+            return self._exec_raw_inner(sel, plan, reservations)
+        finally:
+            for r in reservations:
+                r.release()
+
+    def _exec_raw_inner(self, sel: ast.Select, plan: SelectPlan,
+                        reservations: list) -> QueryResult:
         st = plan.table
         schema = st.schema
         device = self.engine.config.device
@@ -2130,6 +2139,11 @@ class Executor:
         ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
         ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
 
+        # scan memory quota: reserve host bytes per materialized region part
+        # (ref common/memory-manager scan pool); released when the result
+        # is built
+        row_bytes = 8 + 4 + 8 * max(len(needed_fields), 1) + \
+            64 * len(needed_strs)
         parts = []  # (ts np, codes np, region, fields np [nf_needed, n])
         for region in st.regions:
             cand = self._candidate_codes(region, plan)
@@ -2188,6 +2202,8 @@ class Executor:
                 if needed_strs:
                     sel_h = perm[kidx].cpu().numpy()
                     s_cols = {sn: v[sel_h] for sn, v in s_cols.items()}
+            reservations.append(
+                self.engine.scan_quota.acquire(ts_t.numel() * row_bytes))
             parts.append((ts_t.cpu().numpy(), se_t.cpu().numpy(), region,
                           f_t.cpu().numpy(), s_cols))
 

@@ -316,3 +316,26 @@ def test_select_distinct_and_now_interval(tmp_engine):
     ex.execute(f"INSERT INTO dn (h, ts, v) VALUES ('c', {now}, 9.0)")
     r = ex.execute("SELECT count(*) FROM dn WHERE ts >= now() - INTERVAL '1 hour'")
     assert int(list(r.rows())[0][0]) == 1
+
+
+def test_scan_memory_quota(tmp_path):
+    """A raw scan larger than the quota fails loudly instead of OOM-ing
+    (ref common/memory-manager scan pool)."""
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.utils.memquota import ResourceExhausted
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "q"), device="cpu",
+                                  background_flush=False,
+                                  scan_mem_bytes=4096))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE mq (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h)) WITH ('append_mode'='true')")
+    rows = ",".join(f"('h{i % 5}', {i}, {float(i)})" for i in range(2000))
+    ex.execute(f"INSERT INTO mq (h, ts, v) VALUES {rows}")
+    with pytest.raises(ResourceExhausted):
+        ex.execute("SELECT * FROM mq")
+    # aggregates don't materialize rows — unaffected by the scan quota
+    r = ex.execute("SELECT count(*) FROM mq")
+    assert int(list(r.rows())[0][0]) == 2000
+    # quota fully released after the failure (no leak)
+    assert eng.scan_quota.free == 4096
+    eng.close()

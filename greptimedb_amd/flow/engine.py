@@ -26,10 +26,12 @@ from greptimedb_amd.utils.errors import InvalidArguments
 
 
 class FlowTask:
-    def __init__(self, name: str, sink: str, select_sql: str):
+    def __init__(self, name: str, sink: str, select_sql: str,
+                 expire_after_s: int | None = None):
         self.name = name
         self.sink = sink
         self.select_sql = select_sql
+        self.expire_after_s = expire_after_s
         self.select = parse_sql(select_sql)
         if not isinstance(self.select, ast.Select) or self.select.table is None:
             raise InvalidArguments("flow query must be a SELECT ... FROM table")
@@ -63,12 +65,13 @@ class FlowEngine:
                 f.mark_dirty(lo, hi)
 
     def create_flow(self, name: str, sink: str, select_sql: str,
-                    if_not_exists: bool = False):
+                    if_not_exists: bool = False,
+                    expire_after_s: int | None = None):
         if name in self.flows:
             if if_not_exists:
                 return
             raise InvalidArguments(f"flow {name} exists")
-        self.flows[name] = FlowTask(name, sink, select_sql)
+        self.flows[name] = FlowTask(name, sink, select_sql, expire_after_s)
 
     def drop_flow(self, name: str):
         self.flows.pop(name, None)
@@ -77,10 +80,19 @@ class FlowEngine:
         """Re-evaluate dirty windows for every flow; returns rows upserted
         per flow (reference: batching-mode task tick)."""
         out = {}
+        import time as _time
         for f in self.flows.values():
             lo, hi = f.take_dirty()
             if lo is None:
                 continue
+            if f.expire_after_s is not None:
+                # EXPIRE AFTER: data older than the TTL never re-aggregates
+                # (reference: batching-mode expire_after window clamp)
+                floor = int(_time.time() * 1000) - f.expire_after_s * 1000
+                lo = max(lo, floor)
+                if lo > hi:
+                    out[f.name] = 0
+                    continue
             out[f.name] = self._run_flow(f, lo, hi + 1)
         return out
 

@@ -209,6 +209,7 @@ class CreateFlow:
     sink: str
     query_sql: str
     if_not_exists: bool = False
+    expire_after_s: int | None = None   # EXPIRE AFTER '1h' — dirty-window TTL
 
 
 @dataclass

@@ -211,7 +211,8 @@ class Executor:
             return self._exec_admin(stmt)
         if isinstance(stmt, ast.CreateFlow):
             self._flow_engine().create_flow(stmt.name, stmt.sink, stmt.query_sql,
-                                            stmt.if_not_exists)
+                                            stmt.if_not_exists,
+                                            expire_after_s=stmt.expire_after_s)
             return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.DropFlow):
             self._flow_engine().drop_flow(stmt.name)

@@ -374,14 +374,24 @@ class Parser:
             self.expect_kw("sink")
             self.expect_kw("to")
             sink = self.next().value
+            expire_after_s = None
             while not self.at_kw("as") and self.peek() is not None:
-                self.next()  # skip EXPIRE AFTER / COMMENT clauses
+                if self.eat_kw("expire"):
+                    self.expect_kw("after")
+                    t = self.next()
+                    if t.kind == "str":
+                        expire_after_s = _duration_ms(str(t.value)) // 1000
+                    else:
+                        expire_after_s = int(t.value)   # seconds
+                else:
+                    self.next()  # skip COMMENT etc
             self.expect_kw("as")
             if self.peek() is None:
                 raise InvalidSyntax("CREATE FLOW ... AS <select> expected")
             sql = self.sql[self.peek().pos:].rstrip().rstrip(";")
             self.i = len(self.toks)
-            return ast.CreateFlow(name, sink, sql, if_not_exists)
+            return ast.CreateFlow(name, sink, sql, if_not_exists,
+                                  expire_after_s=expire_after_s)
         external = self.eat_kw("external")
         self.expect_kw("table")
         if_not_exists = False

@@ -40,3 +40,27 @@ def test_flow_show_drop(tmp_engine):
     assert "f2" in ex.execute("SHOW FLOWS").columns[0]
     ex.execute("DROP FLOW f2")
     assert "f2" not in ex.execute("SHOW FLOWS").columns[0]
+
+
+def test_flow_expire_after(tmp_engine):
+    """EXPIRE AFTER clamps dirty windows: writes older than the TTL never
+    re-aggregate (ref flow batching-mode expire)."""
+    import time as _t
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE src (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h)) WITH ('append_mode'='true')")
+    ex.execute("CREATE FLOW f_exp SINK TO snk EXPIRE AFTER '1h' AS "
+               "SELECT date_trunc('minute', ts) AS minute, max(v) FROM src "
+               "GROUP BY minute")
+    fe = tmp_engine.flow_engine
+    assert fe.flows["f_exp"].expire_after_s == 3600
+    now = int(_t.time() * 1000)
+    old = now - 2 * 3600 * 1000          # beyond the TTL
+    ex.execute(f"INSERT INTO src (h, ts, v) VALUES ('a', {old}, 99.0), "
+               f"('a', {now}, 7.0)")
+    out = fe.tick()
+    assert out["f_exp"] >= 1
+    r = ex.execute("SELECT count(*) FROM snk")
+    # only the fresh window aggregated; the expired write is ignored
+    assert int(list(r.rows())[0][0]) == 1

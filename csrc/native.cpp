@@ -1277,6 +1277,8 @@ static py::tuple route_ingest(
     int n_regions, py::bytes hdr_suffix, bool durable) {
   const int64_t n = codes.shape(0);
   const int nf = (int)fmap.shape(0);
+  if (ts.shape(0) != n || region_of.shape(0) != n)
+    throw std::runtime_error("route_ingest: length mismatch");
   const int32_t* cp = codes.data();
   const int64_t* tp = ts.data();
   const int64_t* fm = fmap.data();

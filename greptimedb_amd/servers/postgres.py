@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import asyncio
 import math
+import re
 import struct
 
 import numpy as np
@@ -137,8 +138,10 @@ class PostgresServer:
 
     # ---------------------------------------------- extended query protocol
 
+    _NUM = re.compile(r"-?\d+(\.\d+)?([eE][+-]?\d+)?$")
+
     def _extended(self, tag: bytes, body: bytes, stmts: dict, portals: dict):
-        _NUM = __import__("re").compile(r"-?\d+(\.\d+)?([eE][+-]?\d+)?$")
+        _NUM = self._NUM
 
         def err(msg: str, code: bytes = b"42601"):
             return _msg(b"E", b"SERROR\x00C" + code + b"\x00M" +

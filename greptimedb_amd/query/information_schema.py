@@ -13,7 +13,7 @@ from greptimedb_amd.models.schema import SemanticType
 
 VIRTUAL_TABLES = {
     "tables", "columns", "region_statistics", "flows", "cluster_info",
-    "partitions",
+    "partitions", "region_peers", "build_info",
 }
 
 
@@ -62,6 +62,20 @@ def build(engine, name: str):
         flows = list(fe.flows.values()) if fe else []
         rows = [(f.name, f.source, f.sink, f.select_sql) for f in flows]
         return _cols(["flow_name", "source_table", "sink_table", "query"], rows)
+    if kind == "region_peers":
+        # leader/follower roles per region (ref information_schema
+        # region_peers; followers appear when meta/replication is active)
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            for r in st.regions:
+                rows.append((r.region_id, t, 0, engine.config.device,
+                             getattr(r, "role", "leader").upper(), "ALIVE"))
+        return _cols(["region_id", "table_name", "peer_id", "peer_addr",
+                      "role", "status"], rows)
+    if kind == "build_info":
+        from greptimedb_amd import __version__
+        return _cols(["version", "arch", "backend"],
+                     [(__version__, "gfx950", "rocm-hip")])
     if kind == "cluster_info":
         import torch
         rows = [(0, "standalone", engine.config.device,

@@ -339,3 +339,11 @@ def test_scan_memory_quota(tmp_path):
     # quota fully released after the failure (no leak)
     assert eng.scan_quota.free == 4096
     eng.close()
+
+
+def test_information_schema_region_peers_build_info(ex):
+    r = ex.execute("SELECT table_name, role FROM information_schema.region_peers "
+                   "WHERE table_name = 't1'")
+    assert len(r) >= 1 and all(v == "LEADER" for v in r.columns[1])
+    r = ex.execute("SELECT version, arch FROM information_schema.build_info")
+    assert list(r.columns[1]) == ["gfx950"]

@@ -395,6 +395,12 @@ class MitoEngine:
                 region.series.add_encoded(pk)
             # field order may differ from region's if schema evolved; map names
             if fnames != region.field_names:
+                # auto-added fields that never reached a flush only exist in
+                # the WAL — recreate them before mapping (else data is lost)
+                missing = [fn for fn in fnames
+                           if fn not in region.field_names]
+                if missing:
+                    region.ensure_fields(missing)
                 fmap = {fn: i for i, fn in enumerate(fnames)}
                 out = np.full((len(region.field_names), fields.shape[1]), np.nan)
                 for i, fn in enumerate(region.field_names):

@@ -382,3 +382,31 @@ def test_read_externally_written_mito2_sst(tmp_path):
     assert list(seq) == [1, 2, 3, 4]
     # and the pk decodes with our memcomparable codec
     assert pk_codec.decode_pk(dict_values[0], 1) == ("h1",)
+
+
+def test_bulk_path_schema_evolution(tmp_path):
+    """A new field appearing mid-stream grows the schema on the bulk path
+    and WAL payloads stay replayable."""
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "se"), device="cpu",
+                                  background_flush=False))
+    ing = Ingestor(eng)
+    ing._bulk = True
+    ing.ingest_lines(b"m,host=a f1=1.0 1000000000\nm,host=b f1=2.0 2000000000\n")
+    # new field f2 shows up later
+    ing.ingest_lines(b"m,host=a f1=3.0,f2=30.0 3000000000\n")
+    eng.commit_wal()
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(eng)
+    r = ex.execute("SELECT f1, f2 FROM m ORDER BY ts")
+    rows = [tuple(t) for t in r.rows()]
+    assert rows[0][0] == 1.0 and np.isnan(rows[0][1])
+    assert rows[2] == (3.0, 30.0)
+    d = eng.config.data_dir
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu",
+                                   background_flush=False))
+    ex2 = Executor(eng2)
+    r = ex2.execute("SELECT count(*), max(f2) FROM m")
+    row = list(r.rows())[0]
+    assert int(row[0]) == 3 and float(row[1]) == 30.0
+    eng2.close()

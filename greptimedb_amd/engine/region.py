@@ -76,6 +76,7 @@ class Region:
         self._series_log_path = os.path.join(dir, "series.log")
         self._load_series_log()
         self._series_log = open(self._series_log_path, "ab")
+        self._load_strcols_meta()   # restore str/fulltext column layout
         self._load_ssts()
 
     def _load_series_log(self):
@@ -256,16 +257,47 @@ class Region:
 
     def ensure_str_fields(self, names: list[str], fulltext: bool = True):
         """Register string columns on this region (fulltext-indexed when
-        requested — the log-pipeline default)."""
+        requested — the log-pipeline default). The (name, fulltext) set is
+        persisted in a region sidecar so an unflushed reopen (WAL-only
+        data) restores the same index layout."""
         from greptimedb_amd.engine.fulltext import FulltextColumn
         with self.lock:
+            changed = False
             for n in names:
                 if n not in self.str_field_names:
                     self.str_field_names.append(n)
+                    changed = True
                 if fulltext and n not in self.text_cols:
                     ft = FulltextColumn()
                     ft.mem.n_rows = self.memtable.len
                     self.text_cols[n] = ft
+                    changed = True
+            if changed:
+                self._save_strcols_meta()
+
+    def _strcols_meta_path(self) -> str:
+        return os.path.join(self.dir, "strcols.json")
+
+    def _save_strcols_meta(self):
+        import json as _json
+        tmp = self._strcols_meta_path() + ".tmp"
+        with open(tmp, "w") as f:
+            _json.dump({n: (n in self.text_cols) for n in self.str_field_names}, f)
+        os.replace(tmp, self._strcols_meta_path())
+
+    def _load_strcols_meta(self):
+        import json as _json
+        path = self._strcols_meta_path()
+        if not os.path.exists(path):
+            return
+        with open(path) as f:
+            meta = _json.load(f)
+        plain = [n for n, ft in meta.items() if not ft]
+        fts = [n for n, ft in meta.items() if ft]
+        if plain:
+            self.ensure_str_fields(plain, fulltext=False)
+        if fts:
+            self.ensure_str_fields(fts, fulltext=True)
 
     def ensure_fields(self, names: list[str]):
         """Auto-ALTER: add new field columns (reference insert.rs:562

@@ -109,3 +109,25 @@ def test_http_log_endpoints(tmp_engine):
     r = client.get("/v1/sql", params={"sql":
         "SELECT count(*) FROM evt WHERE matches(message, 'hello')"})
     assert r.json()["output"][0]["records"]["rows"][0][0] == 1
+
+
+def test_fulltext_survives_unflushed_reopen(tmp_path):
+    """WAL-only (never flushed) log data must keep its fulltext index
+    across restart (regression: text_cols layout is persisted in the
+    region strcols sidecar)."""
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.engine.logstore import LogStore
+    from greptimedb_amd.query.executor import Executor
+    d = str(tmp_path / "lg")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu",
+                                  background_flush=False))
+    LogStore(eng).ingest("lg", [
+        {"message": "disk failure on node7", "timestamp": 1000},
+        {"message": "all good", "timestamp": 2000}])
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu",
+                                   background_flush=False))
+    r = Executor(eng2).execute(
+        "SELECT count(*) FROM lg WHERE matches(message, 'failure')")
+    assert int(list(r.rows())[0][0]) == 1
+    eng2.close()

@@ -733,6 +733,37 @@ class PromEvaluator:
             else:
                 cnt.index_add_(0, gi, present.double())
 
+        if op == "count_values":
+            # group additionally by the sample VALUE at each step; output
+            # series = (group labels + {param: value}) (Prometheus semantics)
+            if self.dist is not None:
+                raise PlanQuery("distributed count_values not yet supported")
+            label_name = e.param.value if e.param is not None else "value"
+            vals_h = v.cpu().numpy()
+            out_rows: dict = {}   # (g, value_repr) -> row index
+            cols_out: list = []
+            for t in range(T):
+                col = vals_h[:, t]
+                ok = ~np.isnan(col)
+                if not ok.any():
+                    continue
+                pairs = np.stack([gidx[ok].astype(np.float64), col[ok]])
+                uniq_p, counts_p = np.unique(pairs, axis=1, return_counts=True)
+                for (g, val), c in zip(uniq_p.T, counts_p):
+                    key = (int(g), float(val))
+                    row = out_rows.setdefault(key, len(out_rows))
+                    cols_out.append((row, t, float(c)))
+            S2 = max(len(out_rows), 0)
+            outm = torch.full((S2, T), float("nan"), dtype=torch.float64)
+            for row, t, c in cols_out:
+                outm[row, t] = c
+            labels2 = []
+            for (g, val) in out_rows:
+                l = dict(out_labels[g]) if out_labels else {}
+                l[str(label_name)] = f"{val:g}"
+                labels2.append(l)
+            return PromMatrix(labels2, outm.to(dev), grid)
+
         if op == "quantile":
             if self.dist is not None:
                 raise PlanQuery("distributed quantile aggregation not yet supported")

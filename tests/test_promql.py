@@ -254,3 +254,16 @@ def test_at_modifier(prom_env):
     m = ev.query_range('max_over_time(http_requests_total{instance="a"}[2m] @ 240)',
                        60, 540, 60)
     np.testing.assert_allclose(m.values.numpy(), 240.0)
+
+
+def test_count_values(prom_env):
+    _, ev = prom_env
+    # at t=300: a=300, b=600 → two value buckets of count 1
+    m = ev.query_range('count_values("v", http_requests_total)', 300, 300, 1)
+    got = {l["v"]: float(x) for l, x in zip(m.labels, m.values[:, 0])}
+    assert got == {"300": 1.0, "600": 1.0}
+    # grouped: by (instance) each instance contributes its own value series
+    m = ev.query_range('count_values by (instance) ("v", http_requests_total)',
+                       300, 300, 1)
+    assert all("instance" in l and "v" in l for l in m.labels)
+    assert m.S == 2

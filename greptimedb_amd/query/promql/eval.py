@@ -872,8 +872,43 @@ class PromEvaluator:
         # vector-vector: one-to-one on matching label sets
         if e.op in ("and", "or", "unless"):
             return self._set_op(e.op, l, r, grid)
-        lk = {_match_key(x, e.on, e.ignoring): i for i, x in enumerate(l.labels)}
-        rk = {_match_key(x, e.on, e.ignoring): i for i, x in enumerate(r.labels)}
+        if e.group_left or e.group_right:
+            # many-to-one: the "one" side must be unique per match key; every
+            # "many" row keeps its full labels (Prometheus group_left/right)
+            many, one = (l, r) if e.group_left else (r, l)
+            one_map: dict = {}
+            for i, x in enumerate(one.labels):
+                k = _match_key(x, e.on, e.ignoring)
+                if k in one_map:
+                    raise PlanQuery('promql: duplicate series on the "one" '
+                                    'side of group_left/group_right')
+                one_map[k] = i
+            rows_many, rows_one = [], []
+            for i, x in enumerate(many.labels):
+                j = one_map.get(_match_key(x, e.on, e.ignoring))
+                if j is not None:
+                    rows_many.append(i)
+                    rows_one.append(j)
+            if not rows_many:
+                return PromMatrix([], torch.zeros(
+                    (0, T), dtype=torch.float64, device=l.values.device), grid)
+            mi = torch.as_tensor(rows_many, device=l.values.device)
+            oi = torch.as_tensor(rows_one, device=l.values.device)
+            a_m, b_o = many.values[mi], one.values[oi]
+            a, b = (a_m, b_o) if e.group_left else (b_o, a_m)
+            res, keep = _vector_op(e.op, a, b, e.bool_modifier)
+            if e.op in _CMP_OPS and not e.bool_modifier:
+                res = torch.where(keep, a, torch.full_like(res, float("nan")))
+            labels = [_drop_name(many.labels[int(i)]) for i in mi]
+            return PromMatrix(labels, res, grid)
+        lk, rk = {}, {}
+        for side, key_map in ((l, lk), (r, rk)):
+            for i, x in enumerate(side.labels):
+                k = _match_key(x, e.on, e.ignoring)
+                if k in key_map:
+                    raise PlanQuery("promql: many-to-one matching needs "
+                                    "group_left/group_right")
+                key_map[k] = i
         common = [k for k in lk if k in rk]
         if not common:
             return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,

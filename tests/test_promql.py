@@ -267,3 +267,24 @@ def test_count_values(prom_env):
                        300, 300, 1)
     assert all("instance" in l and "v" in l for l in m.labels)
     assert m.S == 2
+
+
+def test_group_left_many_to_one(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE gm (job STRING, inst STRING, ts TIMESTAMP TIME "
+               "INDEX, greptime_value DOUBLE, PRIMARY KEY (job, inst))")
+    ex.execute("INSERT INTO gm (job, inst, ts, greptime_value) VALUES "
+               "('a', 'i1', 1000, 10.0), ('a', 'i2', 1000, 20.0), "
+               "('b', 'i3', 1000, 30.0)")
+    ev = PromEvaluator(tmp_engine)
+    m = ev.query_range('gm / on(job) group_left sum by (job)(gm)', 1, 1, 1)
+    got = sorted(round(float(x), 3) for x in m.values[:, 0])
+    assert got == [0.333, 0.667, 1.0]
+    assert all("inst" in l for l in m.labels)   # many side keeps its labels
+    m = ev.query_range('sum by (job)(gm) / on(job) group_right gm', 1, 1, 1)
+    assert sorted(round(float(x), 2) for x in m.values[:, 0]) == [1.0, 1.5, 3.0]
+    # many-to-one without group_left is an error (Prometheus semantics)
+    import pytest as _p
+    from greptimedb_amd.utils.errors import PlanQuery
+    with _p.raises(PlanQuery):
+        ev.query_range('gm / on(job) sum by (job)(gm)', 1, 1, 1)

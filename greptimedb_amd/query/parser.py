@@ -567,10 +567,19 @@ class Parser:
             if t is None:
                 break
             opname = None
+            negated = False
             if t.kind == "op" and t.value in _PRECEDENCE:
                 opname = t.value
             elif t.kind == "id" and t.value.lower() in _PRECEDENCE:
                 opname = t.value.lower()
+            elif t.kind == "id" and t.value.lower() == "not" and \
+                    self.i + 1 < len(self.toks) and \
+                    self.toks[self.i + 1].kind == "id" and \
+                    self.toks[self.i + 1].value.lower() in ("in", "like",
+                                                            "between"):
+                self.next()                       # consume NOT
+                opname = self.toks[self.i].value.lower()
+                negated = True
             if opname is None:
                 break
             prec = _PRECEDENCE[opname]
@@ -583,7 +592,8 @@ class Parser:
                 if self.at_kw("select"):   # IN (SELECT ...) subquery
                     sub = self.parse_select()
                     self.expect_op(")")
-                    left = ast.InList(left, [ast.ScalarSubquery(sub, many=True)])
+                    left = ast.InList(left, [ast.ScalarSubquery(sub, many=True)],
+                                      negated=negated)
                     continue
                 items = []
                 while True:
@@ -591,14 +601,14 @@ class Parser:
                     if not self.eat_op(","):
                         break
                 self.expect_op(")")
-                left = ast.InList(left, items)
+                left = ast.InList(left, items, negated=negated)
                 continue
             if opname == "between":
                 self.next()
                 low = self.parse_expr(_PRECEDENCE["between"])
                 self.expect_kw("and")
                 high = self.parse_expr(_PRECEDENCE["between"])
-                left = ast.Between(left, low, high)
+                left = ast.Between(left, low, high, negated=negated)
                 continue
             if opname == "is":
                 self.next()
@@ -610,6 +620,8 @@ class Parser:
             if t.kind == "id" and opname == "like":
                 right = self.parse_expr(prec)
                 left = ast.BinOp("like", left, right)
+                if negated:
+                    left = ast.UnaryOp("not", left)
                 continue
             right = self.parse_expr(prec)
             left = ast.BinOp(opname, left, right)

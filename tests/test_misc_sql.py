@@ -347,3 +347,17 @@ def test_information_schema_region_peers_build_info(ex):
     assert len(r) >= 1 and all(v == "LEADER" for v in r.columns[1])
     r = ex.execute("SELECT version, arch FROM information_schema.build_info")
     assert list(r.columns[1]) == ["gfx950"]
+
+
+def test_negated_predicates(tmp_engine):
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE np2 (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE, "
+               "PRIMARY KEY (h))")
+    ex.execute("INSERT INTO np2 (h, ts, v) VALUES ('a', 1000, 1.0), "
+               "('ab', 2000, 2.0), ('b', 3000, 9.0)")
+    r = ex.execute("SELECT h FROM np2 WHERE h NOT IN ('a') ORDER BY h")
+    assert list(r.columns[0]) == ["ab", "b"]
+    r = ex.execute("SELECT h FROM np2 WHERE h NOT LIKE 'a%' ORDER BY h")
+    assert list(r.columns[0]) == ["b"]
+    r = ex.execute("SELECT h FROM np2 WHERE v NOT BETWEEN 1.5 AND 3 ORDER BY h")
+    assert list(r.columns[0]) == ["a", "b"]

@@ -274,6 +274,19 @@ class MySQLServer:
         if s.startswith("select @@") or s == "select version()":
             return self._text_resultset(["version()"], [["8.4.0-greptimedb-amd"]],
                                         [MYSQL_TYPE_VAR_STRING])
+        if s in ("select database()", "select schema()"):
+            return self._text_resultset(["database()"], [["public"]],
+                                        [MYSQL_TYPE_VAR_STRING])
+        if s.startswith("show variables") or s.startswith("show session variables"):
+            return self._text_resultset(
+                ["Variable_name", "Value"],
+                [["version", "8.4.0-greptimedb-amd"],
+                 ["character_set_client", "utf8mb4"],
+                 ["max_allowed_packet", "16777216"]],
+                [MYSQL_TYPE_VAR_STRING, MYSQL_TYPE_VAR_STRING])
+        if s.startswith("show collation") or s.startswith("show character set"):
+            return self._text_resultset(["Charset"], [["utf8mb4"]],
+                                        [MYSQL_TYPE_VAR_STRING])
         try:
             r = self.executor.execute(sql)
         except GreptimeError as e:

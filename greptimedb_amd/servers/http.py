@@ -568,6 +568,19 @@ def build_app(ctx: ServerContext) -> FastAPI:
                     out.append(l)
         return {"status": "success", "data": out}
 
+    @app.get("/v1/prometheus/api/v1/status/buildinfo")
+    def prom_buildinfo():
+        # Grafana probes this to pick feature flags
+        from greptimedb_amd import __version__
+        return {"status": "success",
+                "data": {"version": "2.53.0", "application": "greptimedb-amd",
+                         "revision": __version__}}
+
+    @app.get("/v1/influxdb/ping")
+    @app.get("/v1/influxdb/health")
+    def influx_ping():
+        return Response(status_code=204)
+
     @app.get("/v1/prometheus/api/v1/metadata")
     def prom_metadata():
         return {"status": "success", "data": {}}

@@ -179,3 +179,10 @@ transform:
     assert rows == [["warn", "low memory"], ["error", "disk full"]]
     r = client.delete("/v1/events/pipelines/applog")
     assert r.json()["status"] == "deleted"
+
+
+def test_probe_endpoints(client):
+    r = client.get("/v1/prometheus/api/v1/status/buildinfo")
+    assert r.json()["status"] == "success"
+    assert client.get("/v1/influxdb/ping").status_code == 204
+    assert client.get("/v1/influxdb/health").status_code == 204

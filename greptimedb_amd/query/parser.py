@@ -30,7 +30,7 @@ _TOKEN_RE = re.compile(r"""
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"(?:[^"]|"")*")
   | (?P<id>[A-Za-z_][A-Za-z0-9_.]*)
-  | (?P<op>=~|!~|<>|!=|<=|>=|=|<|>|\(|\)|\[|\]|\{|\}|,|\*|\+|-|/|%|;)
+  | (?P<op>=~|!~|<>|!=|<=|>=|=|<|>|\(|\)|\[|\]|\{|\}|,|\*|\+|-|/|%|;|:|@)
 """, re.VERBOSE)
 
 _UNITS_MS = {

@@ -361,3 +361,16 @@ def test_negated_predicates(tmp_engine):
     assert list(r.columns[0]) == ["b"]
     r = ex.execute("SELECT h FROM np2 WHERE v NOT BETWEEN 1.5 AND 3 ORDER BY h")
     assert list(r.columns[0]) == ["a", "b"]
+
+
+def test_tql_subquery_and_at(tmp_engine):
+    """TQL passes subquery [r:s] and @ syntax through the SQL tokenizer."""
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE tm (job STRING, ts TIMESTAMP TIME INDEX, "
+               "greptime_value DOUBLE, PRIMARY KEY (job))")
+    ex.execute("INSERT INTO tm (job, ts, greptime_value) VALUES "
+               "('a', 60000, 5.0), ('a', 120000, 10.0)")
+    r = ex.execute("TQL EVAL (60, 120, '60s') avg_over_time(tm[1m:30s])")
+    assert len(r) == 2
+    r = ex.execute("TQL EVAL (60, 120, '60s') tm @ 120")
+    assert all(float(v) == 10.0 for v in r.columns[-1])

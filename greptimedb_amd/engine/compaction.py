@@ -69,6 +69,18 @@ class Compactor:
         device = region.device
         ts = torch.cat([b.ts for b in batches])
         se = torch.cat([b.series for b in batches])
+        # per-row sequences order rows across overlapping files; batches
+        # without one (pre-upgrade fixtures) get file-order synthetic seqs
+        seq_parts, synth_base = [], 0
+        for b in batches:
+            if b.seq is not None:
+                seq_parts.append(b.seq.to(torch.int64))
+                synth_base = max(synth_base, int(b.seq.max().item()) + 1 if b.n else 0)
+            else:
+                seq_parts.append(torch.arange(synth_base, synth_base + b.n,
+                                              dtype=torch.int64, device=device))
+                synth_base += b.n
+        seq = torch.cat(seq_parts)
         # unify field layout
         fnames = []
         for b in batches:
@@ -96,15 +108,17 @@ class Compactor:
                 else:
                     str_parts[sn].append(np.asarray(col, dtype=object))
             off += b.n
-        # arrival rank: later files are newer (flush order == dict order)
-        o1 = torch.argsort(ts, stable=True)
+        # sort by (series, ts, seq): stable sorts applied innermost-first so
+        # the newest (highest-seq) row of each (series, ts) group lands last
+        o0 = torch.argsort(seq, stable=True)
+        o1 = o0[torch.argsort(ts[o0], stable=True)]
         perm = o1[torch.argsort(se[o1], stable=True)]
-        ts, se = ts[perm], se[perm]
+        ts, se, seq = ts[perm], se[perm], seq[perm]
         fields = fields[:, perm]
         if not region.append_mode:
             keep = dedup_mark_last(se.contiguous(), ts.contiguous())
             kidx = keep.nonzero(as_tuple=True)[0]
-            ts, se, fields = ts[kidx], se[kidx], fields[:, kidx]
+            ts, se, seq, fields = ts[kidx], se[kidx], seq[kidx], fields[:, kidx]
             perm = perm[kidx]
         perm_h = perm.cpu().numpy()
         str_cols_sorted = {sn: np.concatenate(parts)[perm_h]
@@ -113,7 +127,7 @@ class Compactor:
         ts_h = ts.cpu().numpy()
         se_h = se.cpu().numpy()
         f_h = fields.cpu().numpy()
-        seq_h = np.arange(len(ts_h), dtype=np.int64)
+        seq_h = seq.cpu().numpy()
         fid = sst_mod.new_file_id()
         path = os.path.join(region.dir, "sst", f"{fid}.parquet")
         meta = sst_mod.write_sst(path, region.schema, region.series.pks,
@@ -126,7 +140,7 @@ class Compactor:
             "files_to_remove": list(fids),
         })
         new_batch = sst_mod.SstBatch(ts.contiguous(), se.contiguous(),
-                                     fields.contiguous(), None,
+                                     fields.contiguous(), seq.contiguous(),
                                      meta.min_ts, meta.max_ts, fnames)
         new_batch.str_cols = str_cols_sorted
         for sn, arr in str_cols_sorted.items():

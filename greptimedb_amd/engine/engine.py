@@ -245,6 +245,7 @@ class MitoEngine:
                 mem.min_ts = mn if mem.min_ts is None else min(mem.min_ts, mn)
                 mem.max_ts = mx if mem.max_ts is None else max(mem.max_ts, mx)
                 region.last_seq = max(region.last_seq, seqs[k])
+                region.row_seq += int(seg_lens[k])
         finally:
             for r in reversed(ordered):
                 r.lock.release()
@@ -315,6 +316,7 @@ class MitoEngine:
                 mem.min_ts = mn if mem.min_ts is None else min(mem.min_ts, mn)
                 mem.max_ts = mx if mem.max_ts is None else max(mem.max_ts, mx)
                 region.last_seq = max(region.last_seq, seqs[k])
+                region.row_seq += m
         finally:
             for r in reversed(ordered):
                 r.lock.release()
@@ -368,9 +370,21 @@ class MitoEngine:
                 traceback.print_exc()
 
     def _purge_wal(self):
-        flushed = [r.flushed_seq for st in self.tables.values() for r in st.regions]
-        if flushed:
-            self.wal.purge_before(min(flushed))
+        """Purge WAL segments below every region's replay point. A region
+        only constrains purging while it has unflushed WAL entries
+        (wal.region_last > flushed_seq) — idle or never-written regions no
+        longer pin the whole log (the reference obsoletes per region)."""
+        with self.wal._lock:
+            region_last = dict(self.wal.region_last)
+        floors = []
+        max_seen = 0
+        for st in self.tables.values():
+            for r in st.regions:
+                last = region_last.get(r.region_id, 0)
+                max_seen = max(max_seen, last)
+                if last > r.flushed_seq:
+                    floors.append(r.flushed_seq)
+        self.wal.purge_before(min(floors) if floors else max_seen + 1)
 
     # ------------------------------------------------------------- recovery
 

@@ -34,6 +34,11 @@ def encode_string(b: bytes) -> bytes:
         i += 8
         if i < n:
             out += group + b"\x09"
+        elif len(group) == 8:
+            # exact multiple of 8: the memcomparable crate (chunks_exact +
+            # remainder) emits the full group with the continuation marker
+            # followed by an empty all-pad group — match it byte-for-byte
+            out += group + b"\x09" + bytes(8) + b"\x00"
         else:
             out += group + bytes(8 - len(group)) + bytes([len(group)])
     return bytes(out)

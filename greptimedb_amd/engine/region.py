@@ -70,6 +70,18 @@ class Region:
         self.flushed_seq = self.manifest.flushed_seq
         self.last_seq = self.flushed_seq
         self.lock = threading.Lock()
+        # Per-row sequence space (independent of WAL entry seq): every
+        # appended row gets a monotonically increasing u64, stored in the
+        # SST __sequence column so overlapping files order correctly under
+        # a reference-style merge (mito2 region_write_ctx.rs publishes a
+        # committed_sequence the same way).
+        self.row_seq = 0          # next row sequence to assign
+        self.mem_base = 0         # row seq of active memtable's row 0
+        # Memtables swapped out by an in-progress flush stay visible to
+        # scans here until their SstBatch is published (mito2 keeps
+        # immutable memtables in Version until the flush edit lands).
+        self.flushing: list = []  # [(memtable, n_rows, mem_postings, base)]
+        self._flush_lock = threading.Lock()
         # Series code assignment must be stable across restarts (GPU columns
         # and SST caches store dense codes): an append-only series log is the
         # authoritative code order. Loaded BEFORE SSTs.
@@ -78,6 +90,7 @@ class Region:
         self._series_log = open(self._series_log_path, "ab")
         self._load_strcols_meta()   # restore str/fulltext column layout
         self._load_ssts()
+        self.mem_base = self.row_seq
 
     def _load_series_log(self):
         if not os.path.exists(self._series_log_path):
@@ -163,6 +176,8 @@ class Region:
                 t_f[:, perm].contiguous(), t_seq[perm].contiguous(),
                 int(meta["min_ts"]), int(meta["max_ts"]),
                 list(self.field_names))
+            if len(seq):
+                self.row_seq = max(self.row_seq, int(seq.max()) + 1)
             if str_cols:
                 perm_h = perm.cpu().numpy()
                 for name, vals in str_cols.items():
@@ -193,29 +208,45 @@ class Region:
                 vals = (str_fields or {}).get(name)
                 ft.index_batch(vals if vals is not None else [None] * n)
             self.last_seq = max(self.last_seq, last_seq)
+            self.row_seq += n
 
     def should_flush(self, limit_bytes: int) -> bool:
         return self.memtable.bytes_used >= limit_bytes
 
     def flush(self):
-        """Sort + dedup the memtable on device, write an SST, swap memtable."""
+        """Sort + dedup the memtable on device, write an SST, swap memtable.
+
+        The swapped-out memtable stays scannable via `self.flushing` until
+        its SstBatch is published into sst_cache under the region lock, so
+        concurrent scans never lose the flushing rows (mito2 keeps
+        immutable memtables in Version the same way)."""
+        with self._flush_lock:
+            return self._flush_locked()
+
+    def _flush_locked(self):
         with self.lock:
             mem = self.memtable
             if mem.len == 0:
                 return None
             n = mem.len
             flush_seq = self.last_seq
-            prev_flushed = self.flushed_seq
+            mem_base = self.mem_base
             flush_field_names = list(self.field_names[: mem.nf])
             self.memtable = Memtable(len(self.field_names), device=self.device,
                                      cap=max(mem.cap, 1 << 16))
-            for ft in self.text_cols.values():
+            self.mem_base = self.row_seq
+            mem_postings = {}
+            for name, ft in self.text_cols.items():
+                mem_postings[name] = ft.mem
                 ft.reset_mem()  # new memtable rows start at 0
+            entry = (mem, n, mem_postings, mem_base)
+            self.flushing.append(entry)
         ts, se, fields, perm = mem.sorted_view()
         if not self.append_mode:
             keep = dedup_mark_last(se, ts)
             idx = keep.nonzero(as_tuple=True)[0]
             ts, se, fields, perm = ts[idx], se[idx], fields[:, idx], perm[idx]
+        seq_t = perm.to(torch.int64) + mem_base
         ts_h = ts.cpu().numpy()
         se_h = se.cpu().numpy()
         f_h = fields.cpu().numpy()
@@ -229,15 +260,13 @@ class Region:
             ft = self.text_cols.get(name)
             if ft is not None:
                 text_index[name] = ft.build_segment(list(arr), self.device)
-        # per-row sequence ≈ prev flushed seq + arrival index in this memtable
-        # (row order is the recency order LastRow dedup relies on)
-        seq_h = perm_h.astype(np.int64) + prev_flushed
+        seq_h = perm_h.astype(np.int64) + mem_base
         fid = sst_mod.new_file_id()
         path = os.path.join(self.dir, "sst", f"{fid}.parquet")
         meta = sst_mod.write_sst(path, self.schema, self.series.pks,
                                  se_h, ts_h, f_h, seq_h, flush_field_names,
                                  str_cols=str_cols_sorted)
-        meta.seq_max = flush_seq
+        meta.seq_max = int(seq_h.max()) if len(seq_h) else 0
         self.manifest.commit({
             "kind": "edit",
             "files_to_add": [meta.to_dict()],
@@ -245,12 +274,14 @@ class Region:
             "flushed_seq": flush_seq,
         })
         batch = sst_mod.SstBatch(
-            ts.contiguous(), se.contiguous(), fields.contiguous(), None,
-            meta.min_ts, meta.max_ts, flush_field_names)
+            ts.contiguous(), se.contiguous(), fields.contiguous(),
+            seq_t.contiguous(), meta.min_ts, meta.max_ts, flush_field_names)
         batch.str_cols = str_cols_sorted
         batch.text_index = text_index
-        self.sst_cache[fid] = batch
-        self.flushed_seq = flush_seq
+        with self.lock:
+            self.sst_cache[fid] = batch
+            self.flushed_seq = flush_seq
+            self.flushing.remove(entry)
         return meta
 
     # ---------------------------------------------------------------- schema
@@ -312,11 +343,19 @@ class Region:
     # ---------------------------------------------------------------- scan
 
     def scan_sources(self, ts_lo: int | None = None, ts_hi: int | None = None):
-        """ScanSource list overlapping the time range: SST cache batches
-        (time-pruned, reference scan_region.rs:887) + the active memtable."""
+        """ScanSource list overlapping the time range, oldest → newest: SST
+        cache batches (time-pruned, reference scan_region.rs:887), then
+        memtables being flushed, then the active memtable. Everything is
+        snapshotted under the region lock so a concurrent flush can neither
+        hide rows nor mutate the dict mid-iteration."""
         out = []
         device = self.device
-        for batch in self.sst_cache.values():
+        with self.lock:
+            batches = list(self.sst_cache.values())
+            flushing = list(self.flushing)
+            mem = self.memtable
+            n = mem.len
+        for batch in batches:
             if ts_lo is not None and batch.max_ts < ts_lo:
                 continue
             if ts_hi is not None and batch.min_ts >= ts_hi:
@@ -341,44 +380,60 @@ class Region:
                                   sorted=True,
                                   str_cols=getattr(batch, "str_cols", {}),
                                   text_probe=make_probe(batch)))
-        with self.lock:
-            mem = self.memtable
-            n = mem.len
+        for fmem, fn_rows, fpostings, _base in flushing:
+            out.extend(self._mem_source(fmem, fn_rows, ts_lo, ts_hi,
+                                        postings=fpostings))
         if n > 0:
-            if not (ts_lo is not None and mem.max_ts is not None and mem.max_ts < ts_lo) and \
-               not (ts_hi is not None and mem.min_ts is not None and mem.min_ts >= ts_hi):
-
-                def mem_probe(col, terms, _n=n, _mem=mem):
-                    ft = self.text_cols.get(col)
-                    if ft is None:
-                        return None
-                    tids = ft.query_tids(terms)
-                    if tids is None:
-                        return torch.zeros(_n, dtype=torch.bool, device=device)
-                    return ft.mem.probe(tids, _n, device)
-
-                out.append(ScanSource(mem.ts[:n], mem.series[:n], mem.fields, n,
-                                      {fn: i for i, fn in enumerate(self.field_names)},
-                                      sorted=False,
-                                      str_cols={k: v for k, v in mem.str_cols.items()},
-                                      text_probe=mem_probe))
+            out.extend(self._mem_source(mem, n, ts_lo, ts_hi))
         return out
+
+    def _mem_source(self, mem, n, ts_lo, ts_hi, postings=None):
+        """Build the ScanSource for one (active or flushing) memtable;
+        returns [] when the memtable's time range misses the scan range."""
+        device = self.device
+        if ts_lo is not None and mem.max_ts is not None and mem.max_ts < ts_lo:
+            return []
+        if ts_hi is not None and mem.min_ts is not None and mem.min_ts >= ts_hi:
+            return []
+
+        def mem_probe(col, terms, _n=n, _postings=postings):
+            ft = self.text_cols.get(col)
+            if ft is None:
+                return None
+            tids = ft.query_tids(terms)
+            if tids is None:
+                return torch.zeros(_n, dtype=torch.bool, device=device)
+            pp = _postings.get(col) if _postings is not None else ft.mem
+            if pp is None:
+                return torch.zeros(_n, dtype=torch.bool, device=device)
+            return pp.probe(tids, _n, device)
+
+        return [ScanSource(mem.ts[:n], mem.series[:n], mem.fields, n,
+                           {fn: i for i, fn in enumerate(self.field_names)},
+                           sorted=False,
+                           str_cols={k: v for k, v in mem.str_cols.items()},
+                           text_probe=mem_probe)]
 
     @property
     def num_rows(self) -> int:
-        return self.memtable.len + sum(b.n for b in self.sst_cache.values())
+        with self.lock:
+            return (self.memtable.len
+                    + sum(n for _, n, _, _ in self.flushing)
+                    + sum(b.n for b in self.sst_cache.values()))
 
     def time_range(self) -> tuple[int, int] | None:
-        """(min_ts, max_ts) over memtable + SSTs, None if empty."""
+        """(min_ts, max_ts) over memtables + SSTs, None if empty."""
         lo = hi = None
-        for b in self.sst_cache.values():
+        with self.lock:
+            batches = list(self.sst_cache.values())
+            mems = [m for m, _, _, _ in self.flushing] + [self.memtable]
+        for b in batches:
             lo = b.min_ts if lo is None else min(lo, b.min_ts)
             hi = b.max_ts if hi is None else max(hi, b.max_ts)
-        with self.lock:
-            m_lo, m_hi = self.memtable.min_ts, self.memtable.max_ts
-        if m_lo is not None:
-            lo = m_lo if lo is None else min(lo, m_lo)
-            hi = m_hi if hi is None else max(hi, m_hi)
+        for m in mems:
+            if m.min_ts is not None:
+                lo = m.min_ts if lo is None else min(lo, m.min_ts)
+                hi = m.max_ts if hi is None else max(hi, m.max_ts)
         if lo is None:
             return None
         return lo, hi

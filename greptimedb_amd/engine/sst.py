@@ -88,7 +88,7 @@ def write_sst(path: str, schema: TableSchema, pks: list[bytes],
         names.append(fn)
     for fn, vals in (str_cols or {}).items():
         vl = list(vals)
-        is_bin = any(isinstance(v, (bytes, bytearray)) for v in vl[:4])
+        is_bin = any(isinstance(v, (bytes, bytearray)) for v in vl)
         cols.append(pa.array(vl, type=pa.binary() if is_bin else pa.string()))
         names.append(fn)
     cols.append(pa.array(ts_ms, type=pa.timestamp("ms")))

@@ -27,7 +27,7 @@ def encode_batch(series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
     [per str col: lengths i32[n] (-1=None) + utf8 blob]"""
     str_cols = str_cols or {}
     bin_cols = [name for name, vals in str_cols.items()
-                if any(isinstance(v, (bytes, bytearray)) for v in vals[:4])]
+                if any(isinstance(v, (bytes, bytearray)) for v in vals)]
     hdr = json.dumps({
         "n": int(len(ts_ms)),
         "fields": field_names,
@@ -85,12 +85,16 @@ class Wal:
         self.sync_on_commit = sync_on_commit
         self.writer = _native.WalWriter()
         self.next_seq = 1
+        # last appended seq per region — updated atomically with the seq
+        # assignment so purge decisions never race an in-flight append
+        self.region_last: dict[int, int] = {}
         self._lock = threading.Lock()  # multi-worker ingest (P5 write workers)
         segs = self.segments()
         if segs:
             # resume: next_seq = last replayed seq + 1 (caller replays first)
-            for _, _, seq, _ in self.replay():
+            for _, rid, seq, _ in self.replay():
                 self.next_seq = max(self.next_seq, seq + 1)
+                self.region_last[rid] = seq
             self._open_new_segment()
         else:
             self._open_new_segment()
@@ -109,6 +113,7 @@ class Wal:
             seq = self.next_seq
             self.next_seq += 1
             self.writer.append(region_id, seq, payload)
+            self.region_last[region_id] = seq
             return seq
 
     def commit(self):

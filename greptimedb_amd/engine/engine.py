@@ -42,6 +42,7 @@ class TableState:
     schema: TableSchema
     regions: list = field(default_factory=list)
     append_mode: bool = False
+    rule: object = None   # cached PartitionRule (parallel/partition.py)
 
 
 class MitoEngine:
@@ -142,6 +143,21 @@ class MitoEngine:
             return self.tables[name]
         except KeyError:
             raise TableNotFound(name) from None
+
+    def partition_rule(self, st: TableState):
+        """Resolve (and cache) the table's partition rule — multi-dim when
+        `PARTITION ON COLUMNS` was declared, hash(pk) % n otherwise
+        (reference: src/partition/src/manager.rs route resolution)."""
+        if st.rule is None or st.rule.n_regions != len(st.regions):
+            from greptimedb_amd.parallel.partition import rule_for_table
+            st.rule = rule_for_table(st.schema, len(st.regions))
+        return st.rule
+
+    def region_of_tags(self, st: TableState, tags: tuple) -> int:
+        """Region index for one series' ordered tag values."""
+        rule = self.partition_rule(st)
+        return rule.region_of({c.name: v for c, v in
+                               zip(st.schema.tag_columns, tags)})
 
     # ------------------------------------------------------------- writes
 

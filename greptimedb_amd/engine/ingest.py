@@ -91,10 +91,8 @@ class Ingestor:
         # unify tag order with table primary key; missing tags = None
         tag_map = dict(tags)
         tag_tuple = tuple(tag_map.get(t.name) for t in st.schema.tag_columns)
-        # partition by pk hash
-        from greptimedb_amd.engine import pk_codec
-        pk = pk_codec.encode_pk(tag_tuple)
-        region_idx = tsid_hash(pk) % len(st.regions)
+        # partition: multi-dim rule when declared, hash(pk) % n otherwise
+        region_idx = self.engine.region_of_tags(st, tag_tuple)
         region = st.regions[region_idx]
         local = region.register_series(tag_tuple)
         if sid >= self._cap:

@@ -147,6 +147,7 @@ class CreateTable:
     options: dict = field(default_factory=dict)
     partitions: int | None = None
     external: bool = False       # CREATE EXTERNAL TABLE (file engine)
+    partition_on: tuple | None = None  # (columns, [Expr per region]) — PARTITION ON COLUMNS
 
 
 @dataclass

@@ -273,8 +273,13 @@ class Executor:
         opts = [f"  regions = {len(st.regions)}"]
         if st.append_mode:
             opts.append("  append_mode = 'true'")
+        part = ""
+        if schema.options.get("partition_rule"):
+            from greptimedb_amd.parallel.partition import MultiDimPartitionRule
+            rule = MultiDimPartitionRule.from_json(schema.options["partition_rule"])
+            part = rule.to_sql() + "\n"
         ddl = (f'CREATE TABLE IF NOT EXISTS "{name}" (\n{body}\n)\n'
-               f"ENGINE=mito\nWITH(\n" + ",\n".join(opts) + "\n)")
+               f"{part}ENGINE=mito\nWITH(\n" + ",\n".join(opts) + "\n)")
         return QueryResult(["Table", "Create Table"], [[name], [ddl]])
 
     def _exec_alter(self, a: ast.AlterTable) -> QueryResult:
@@ -436,8 +441,7 @@ class Executor:
         for i in range(n):
             tags = tuple(None if data[tn][i] is None else str(data[tn][i])
                          for tn in tag_names)
-            pk = pk_codec.encode_pk(tags)
-            ridx = tsid_hash(pk) % len(st.regions)
+            ridx = self.engine.region_of_tags(st, tags)
             codes[i] = st.regions[ridx].register_series(tags)
             rows_by_region.setdefault(ridx, []).append(i)
         for ridx, rows in rows_by_region.items():
@@ -652,8 +656,21 @@ class Executor:
                                      fulltext=bool(ft)))
         schema = TableSchema(name=c.name, columns=cols, primary_key=c.primary_key,
                              options={k: v for k, v in c.options.items()})
+        n_regions = c.partitions
+        if c.partition_on is not None:
+            from greptimedb_amd.parallel.partition import (MultiDimPartitionRule,
+                                                           PartitionExpr)
+            pcols, pexprs = c.partition_on
+            for pc in pcols:
+                if pc not in c.primary_key:
+                    raise InvalidArguments(
+                        f"partition column {pc} must be a primary-key (tag) column")
+            rule = MultiDimPartitionRule(pcols,
+                                         [PartitionExpr.from_ast(e) for e in pexprs])
+            schema.options["partition_rule"] = rule.to_json()
+            n_regions = rule.n_regions
         append = str(c.options.get("append_mode", "false")).lower() == "true"
-        self.engine.create_table(schema, n_regions=c.partitions,
+        self.engine.create_table(schema, n_regions=n_regions,
                                  append_mode=append, if_not_exists=c.if_not_exists)
         return QueryResult(["status"], [["ok"]])
 
@@ -768,8 +785,7 @@ class Executor:
         for i in range(n):
             tags = tuple(str(by_col[t][i]) if by_col.get(t) is not None and by_col[t][i] is not None
                          else None for t in tag_names)
-            pk = pk_codec.encode_pk(tags)
-            ridx = tsid_hash(pk) % len(st.regions)
+            ridx = self.engine.region_of_tags(st, tags)
             codes[i] = st.regions[ridx].register_series(tags)
             regions[i] = ridx
             rows_by_region.setdefault(ridx, []).append(i)

@@ -471,6 +471,7 @@ class Parser:
         self.expect_op(")")
         options = {}
         partitions = None
+        partition_on = None
         while self.peek() is not None and not self.at_op(";"):
             if self.eat_kw("with"):
                 self.expect_op("(")
@@ -482,9 +483,30 @@ class Parser:
                         break
                 self.expect_op(")")
             elif self.eat_kw("partition"):
-                # PARTITION <n>  (simplified: region count; reference uses
-                # PARTITION ON COLUMNS (...) — multi-dim exprs in parallel/partition.py)
-                partitions = int(self.next().value)
+                if self.eat_kw("on"):
+                    # PARTITION ON COLUMNS (a, b) (expr0, expr1, ...) —
+                    # reference src/partition/src/multi_dim.rs; rule engine
+                    # in parallel/partition.py
+                    self.expect_kw("columns")
+                    self.expect_op("(")
+                    pcols = []
+                    while True:
+                        pcols.append(self.next().value)
+                        if not self.eat_op(","):
+                            break
+                    self.expect_op(")")
+                    self.expect_op("(")
+                    pexprs = []
+                    if not self.at_op(")"):
+                        while True:
+                            pexprs.append(self.parse_expr())
+                            if not self.eat_op(","):
+                                break
+                    self.expect_op(")")
+                    partition_on = (pcols, pexprs)
+                else:
+                    # PARTITION <n> (simplified: region count)
+                    partitions = int(self.next().value)
             elif self.eat_kw("engine"):
                 self.expect_op("=")
                 options["engine"] = self.next().value
@@ -492,7 +514,7 @@ class Parser:
                 break
         return ast.CreateTable(name, columns, primary_key, time_index,
                                if_not_exists, options, partitions,
-                               external=external)
+                               external=external, partition_on=partition_on)
 
     def parse_drop(self):
         self.expect_kw("drop")

@@ -121,6 +121,20 @@ def main():
         background_flush=True, wal_sync=(args.wal == "fsync"),
         flush_bytes=args.flush_mb << 20))
 
+    # -------- cross-rank write fan-out (any rank accepts any write) ---------
+    # Each rank's clients generate their own host universe (weak scaling),
+    # but OWNERSHIP is global: series hash to any rank, and rows are shipped
+    # to the owner (parallel/write_fanout.py; reference insert.rs:389-496).
+    exchange = None
+    if world > 1:
+        from greptimedb_amd.parallel.write_fanout import WriteExchange
+        serv_ing = Ingestor(eng, default_regions=4, append_mode=True,
+                            durable=(args.wal != "off"), rank=rank, world=world)
+        exchange = WriteExchange(rank, world, handler=serv_ing.handle_remote)
+        serv_ing.exchange = exchange
+        import torch.distributed as dist
+        dist.barrier()  # every rank's exchange is listening
+
     # -------- pre-generate the line pool (untimed; replayed cyclically) -----
     # Each worker thread owns a disjoint host shard with its own Ingestor /
     # C++ parser (like the reference's 6 TSBS client workers; P5 axis).
@@ -136,7 +150,8 @@ def main():
         batches = [w.next_batch(ROWS_PER_BATCH)
                    for _ in range(pool_batches_per_worker)]
         ing = Ingestor(eng, default_regions=4, append_mode=True,
-                       durable=(args.wal != "off"))
+                       durable=(args.wal != "off"),
+                       rank=rank, world=world, exchange=exchange)
         workers.append((ing, batches))
 
     gen_threads = [threading.Thread(target=gen_worker, args=(wi,))
@@ -194,6 +209,7 @@ def main():
     total_rows = args.steps * rows_per_step * world
     rows_per_s = total_rows / elapsed
     ms_per_step = elapsed / args.steps * 1000
+    rows_shipped = sum(wk[0].rows_shipped for wk in workers)
     ingest_rows_stored = sum(r.num_rows for st in eng.tables.values()
                              for r in st.regions)
     n_ssts = sum(len(r.manifest.files) for st in eng.tables.values()
@@ -206,6 +222,10 @@ def main():
     drain0 = time.perf_counter()
     while not eng._flush_q.empty() and time.perf_counter() - drain0 < 120:
         time.sleep(0.2)
+    if exchange is not None:
+        import torch.distributed as dist
+        dist.barrier()  # all remote applies acked everywhere
+        exchange.close()
     eng.close()
     shutil.rmtree(data_dir, ignore_errors=True)
 
@@ -272,6 +292,7 @@ def main():
             "workers": n_workers,
             "rows_per_step": rows_per_step,
             "pool_rows": pool_total,
+            "rows_shipped_cross_rank": rows_shipped,
             "ingest_elapsed_s": round(elapsed, 2),
             "wal": {"buffered": "group-commit, no fsync",
                     "fsync": "group-commit + fdatasync",

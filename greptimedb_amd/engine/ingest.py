@@ -36,12 +36,21 @@ def parse_tagset(key: bytes) -> tuple[str, list[tuple[str, str]]]:
 
 class Ingestor:
     def __init__(self, engine: MitoEngine, default_regions: int | None = None,
-                 append_mode: bool = True, durable: bool = True):
+                 append_mode: bool = True, durable: bool = True,
+                 rank: int = 0, world: int = 1, exchange=None):
         self.engine = engine
         self.parser = _native.LineParser()
         self.append_mode = append_mode
         self.durable = durable
         self.default_regions = default_regions or engine.config.default_regions
+        # cross-rank write fan-out (reference insert.rs:459
+        # group_requests_by_peer): with world>1 + an exchange, series whose
+        # partition is owned by another rank are shipped there
+        self.rank = rank
+        self.world = world
+        self.exchange = exchange
+        self._remote_cache: dict[bytes, tuple[int, int]] = {}
+        self._remote_lock = __import__("threading").Lock()
         # K16 bulk scatter-append: on by default on GPU (one kernel per batch
         # instead of per-region copy chains); CPU keeps the per-region path
         self._bulk = engine.config.device.startswith("cuda")
@@ -49,6 +58,7 @@ class Ingestor:
         self._cap = 1024
         self.sid_region = np.full(self._cap, -1, dtype=np.int32)   # flat region idx
         self.sid_local = np.full(self._cap, -1, dtype=np.int32)    # region-local code
+        self.sid_rank = np.full(self._cap, -1, dtype=np.int32)     # owning rank
         self.flat_regions: list = []      # (TableState, region_idx)
         self._region_key: dict = {}       # (table_name, region_idx) -> flat idx
         self._table_field_map: dict = {}  # table name -> cached np map + src len
@@ -56,6 +66,7 @@ class Ingestor:
         self._flat_ridx_cache = None       # flat region idx -> table-local idx
         self._flat_ridx_table = None
         self.rows_ingested = 0
+        self.rows_shipped = 0
         # re-register any series known to existing tables (restart path):
         # parser starts empty; sids are assigned fresh per process, routing
         # fills lazily as tagsets arrive.
@@ -73,7 +84,7 @@ class Ingestor:
         cap = self._cap
         while cap <= need:
             cap *= 2
-        for name in ("sid_region", "sid_local"):
+        for name in ("sid_region", "sid_local", "sid_rank"):
             a = getattr(self, name)
             na = np.full(cap, -1, dtype=np.int32)
             na[: len(a)] = a
@@ -82,7 +93,7 @@ class Ingestor:
 
     def _register_tagset(self, sid: int, key: bytes):
         """Cold path: new series — resolve table (auto-create), partition,
-        register the series in its region."""
+        register the series in its region (or mark it remote-owned)."""
         measurement, tags = parse_tagset(key)
         try:
             st = self.engine.table(measurement)
@@ -91,14 +102,32 @@ class Ingestor:
         # unify tag order with table primary key; missing tags = None
         tag_map = dict(tags)
         tag_tuple = tuple(tag_map.get(t.name) for t in st.schema.tag_columns)
+        if sid >= self._cap:
+            self._grow(sid)
+        if self.world > 1 and self.exchange is not None:
+            owner = self._owner_of(st, tag_tuple)
+            if owner != self.rank:
+                self.sid_rank[sid] = owner
+                return
         # partition: multi-dim rule when declared, hash(pk) % n otherwise
         region_idx = self.engine.region_of_tags(st, tag_tuple)
         region = st.regions[region_idx]
         local = region.register_series(tag_tuple)
-        if sid >= self._cap:
-            self._grow(sid)
         self.sid_region[sid] = self._flat_region(st, region_idx)
         self.sid_local[sid] = local
+        self.sid_rank[sid] = self.rank
+
+    def _owner_of(self, st: TableState, tag_tuple: tuple) -> int:
+        """Owning rank of a series (reference: table-route peer lookup).
+        Multi-dim rules: global partition % world; hash tables:
+        tsid_hash(pk) % world."""
+        from greptimedb_amd.parallel.partition import MultiDimPartitionRule
+        rule = self.engine.partition_rule(st)
+        if isinstance(rule, MultiDimPartitionRule):
+            return rule.region_of({c.name: v for c, v in
+                                   zip(st.schema.tag_columns, tag_tuple)}) % self.world
+        from greptimedb_amd.engine import pk_codec
+        return tsid_hash(pk_codec.encode_pk(tag_tuple)) % self.world
 
     def _auto_create_table(self, measurement: str, tags: list[tuple[str, str]]) -> TableState:
         cols = []
@@ -163,6 +192,8 @@ class Ingestor:
             self._epoch = epoch
             self.sid_region.fill(-1)
             self.sid_local.fill(-1)
+            self.sid_rank.fill(-1)
+            self._remote_cache.clear()
             self.flat_regions.clear()
             self._region_key.clear()
             self._table_field_map.clear()
@@ -170,17 +201,59 @@ class Ingestor:
             self._flat_ridx_cache = None
         for sid, key in new_tagsets:
             self._register_tagset(sid, key)
-        region_of = self.sid_region[series]
-        if bool((region_of < 0).any()):
-            for sid in np.unique(series[region_of < 0]):
+        rank_of = self.sid_rank[series]
+        if bool((rank_of < 0).any()):
+            for sid in np.unique(series[rank_of < 0]):
                 self._register_tagset(int(sid), self.parser.tagset_str(int(sid)))
-            region_of = self.sid_region[series]
+            rank_of = self.sid_rank[series]
+        region_of = self.sid_region[series]
         parser_fields = self.parser.field_names()
-        fields_mat = np.stack([fields[fn] for fn in parser_fields]) if parser_fields \
+        fields_list = [fields[fn] for fn in parser_fields]
+        fields_mat = np.stack(fields_list) if parser_fields \
             else np.zeros((0, n))
         if ts_scale_to_ns != 1:
             ts_ns = ts_ns * ts_scale_to_ns
         ts_ms = ts_ns // 1_000_000
+
+        # ---- cross-rank fan-out (reference insert.rs group_requests_by_peer)
+        ship_threads = []
+        ship_errors: list = []
+        if self.world > 1 and bool((rank_of != self.rank).any()):
+            from greptimedb_amd.parallel.write_fanout import encode_routed_batch
+            total_shipped = 0
+            for peer in np.unique(rank_of):
+                peer = int(peer)
+                if peer == self.rank:
+                    continue
+                rows_r = np.flatnonzero(rank_of == peer)
+                uniq_sids, inv = np.unique(series[rows_r], return_inverse=True)
+                tagsets = [self.parser.tagset_str(int(s)) for s in uniq_sids]
+                payload = encode_routed_batch(
+                    tagsets, inv.astype(np.int32), ts_ms[rows_r],
+                    np.ascontiguousarray(fields_mat[:, rows_r]), parser_fields)
+                t = __import__("threading").Thread(
+                    target=self._ship, args=(peer, payload, ship_errors))
+                t.start()
+                ship_threads.append(t)
+                total_shipped += len(rows_r)
+            self.rows_shipped += total_shipped
+            keep = np.flatnonzero(rank_of == self.rank)
+            if len(keep) == 0:
+                for t in ship_threads:
+                    t.join()
+                if ship_errors:
+                    raise ship_errors[0]
+                self.rows_ingested += n
+                return n
+            series = series[keep]
+            ts_ms = ts_ms[keep]
+            region_of = region_of[keep]
+            fields_mat = np.ascontiguousarray(fields_mat[:, keep])
+            fields_list = [fields_mat[i] for i in range(len(parser_fields))]
+            fields = {fn: fields_list[i] for i, fn in enumerate(parser_fields)}
+            n_local = len(keep)
+        else:
+            n_local = n
 
         local = self.sid_local[series]
         engine = self.engine
@@ -193,7 +266,7 @@ class Ingestor:
             if len(sts) == 1:
                 st0 = next(iter(sts.values()))
                 fmap = self._field_map(st0, parser_fields, fields_mat,
-                                       np.arange(n))
+                                       np.arange(n_local))
                 flat_ridx = self._flat_ridx_cache
                 if flat_ridx is None or len(flat_ridx) != len(self.flat_regions) \
                         or self._flat_ridx_table != id(st0):
@@ -220,12 +293,12 @@ class Ingestor:
                         engine.commit_wal()
                     engine.maybe_flush()
                     self.rows_ingested += n
-                    return n
+                    return self._finish(n, ship_threads, ship_errors)
         order = np.argsort(region_of, kind="stable")
         region_sorted = region_of[order]
         bounds = np.flatnonzero(np.diff(region_sorted)) + 1
         starts = np.concatenate(([0], bounds))
-        ends = np.concatenate((bounds, [n]))
+        ends = np.concatenate((bounds, [n_local]))
         for s, e in zip(starts, ends):
             flat = int(region_sorted[s])
             st, region_idx = self.flat_regions[flat]
@@ -245,4 +318,79 @@ class Ingestor:
             engine.commit_wal()
         engine.maybe_flush()
         self.rows_ingested += n
+        return self._finish(n, ship_threads, ship_errors)
+
+    def _ship(self, peer: int, payload: bytes, errors: list):
+        try:
+            self.exchange.request(peer, payload)
+        except Exception as e:  # surfaced to the caller in _finish
+            errors.append(e)
+
+    def _finish(self, n: int, ship_threads, ship_errors) -> int:
+        for t in ship_threads:
+            t.join()
+        if ship_errors:
+            raise ship_errors[0]
         return n
+
+    def handle_remote(self, payload: bytes) -> bytes:
+        """Apply one routed write batch from a peer rank (receiver side of
+        the fan-out; reference: RegionServerHandler::handle on the owning
+        datanode). WAL-commits before acking so the sender's durability
+        contract holds across ranks."""
+        from greptimedb_amd.parallel.write_fanout import decode_routed_batch
+        tagsets, srow, ts_ms, fields_mat, field_names, _str = \
+            decode_routed_batch(payload)
+        with self._remote_lock:
+            epoch = getattr(self.engine, "routing_epoch", 0)
+            if epoch != getattr(self, "_repoch", 0):
+                self._repoch = epoch
+                self._remote_cache.clear()
+            k = len(tagsets)
+            frs = np.empty(k, dtype=np.int32)
+            lcs = np.empty(k, dtype=np.int32)
+            for i, key in enumerate(tagsets):
+                hit = self._remote_cache.get(key)
+                if hit is None:
+                    hit = self._resolve_tagset_local(key)
+                    self._remote_cache[key] = hit
+                frs[i], lcs[i] = hit
+            region_of = frs[srow]
+            local = lcs[srow]
+            n = len(ts_ms)
+            engine = self.engine
+            order = np.argsort(region_of, kind="stable")
+            region_sorted = region_of[order]
+            bounds = np.flatnonzero(np.diff(region_sorted)) + 1
+            starts = np.concatenate(([0], bounds))
+            ends = np.concatenate((bounds, [n]))
+            for s, e in zip(starts, ends):
+                flat = int(region_sorted[s])
+                st, region_idx = self.flat_regions[flat]
+                rows = order[s:e]
+                fmap = self._field_map(st, field_names, fields_mat, rows)
+                out = np.empty((len(fmap), e - s), dtype=np.float64)
+                for i, src in enumerate(fmap):
+                    out[i] = fields_mat[src][rows] if src >= 0 else np.nan
+                engine.write_region(st, region_idx,
+                                    local[rows].astype(np.int32),
+                                    ts_ms[rows], out, [], durable=self.durable)
+            if self.durable:
+                engine.commit_wal()
+            engine.maybe_flush()
+            self.rows_ingested += n
+        return b"OK"
+
+    def _resolve_tagset_local(self, key: bytes) -> tuple[int, int]:
+        """Resolve a routed tagset on THIS rank: auto-create the table,
+        partition with the local rule, register the series."""
+        measurement, tags = parse_tagset(key)
+        try:
+            st = self.engine.table(measurement)
+        except Exception:
+            st = self._auto_create_table(measurement, tags)
+        tag_map = dict(tags)
+        tag_tuple = tuple(tag_map.get(t.name) for t in st.schema.tag_columns)
+        region_idx = self.engine.region_of_tags(st, tag_tuple)
+        local = st.regions[region_idx].register_series(tag_tuple)
+        return self._flat_region(st, region_idx), local

@@ -1,0 +1,170 @@
+"""Cross-rank write fan-out: any rank accepts any write.
+
+Reference parity: src/operator/src/insert.rs:389-496 — the Inserter
+partition-splits every row batch and fans the slices out to the owning
+datanodes over gRPC (`group_requests_by_peer` → per-peer
+RegionServer.handle). MI355X redesign: one process per GPU on one node, so
+the peer hop is a loopback TCP exchange between ranks (the data starts on
+the host — lines arrive over the wire — so xGMI/RCCL buys nothing for the
+row shipping itself; device work starts at the owner's memtable append,
+like the reference where WAL+memtable live on the owning datanode).
+
+Ownership: a series' global partition index p (PartitionRule) maps to rank
+`p % world` for multi-dim rules; hash tables use `tsid_hash(pk) % world`.
+The receiving rank re-resolves its local region with its own engine's rule,
+WAL-commits, then acks — the sender's ingest call returns only after every
+peer ack (reference: Inserter joins all per-peer futures).
+
+Frames are length-prefixed pickles of numpy arrays (loopback-only internal
+transport, never exposed on a public port; the external write surface
+remains HTTP/MySQL/PG/gRPC).
+"""
+
+from __future__ import annotations
+
+import pickle
+import socket
+import struct
+import threading
+import time
+
+
+def fanout_port(rank: int, base: int | None = None) -> int:
+    """Deterministic per-rank loopback port. Base derives from the torch
+    rendezvous port so concurrent jobs on one box don't collide."""
+    import os
+    if base is None:
+        base = int(os.environ.get("GDB_FANOUT_BASE",
+                                  int(os.environ.get("MASTER_PORT", "29400")) + 500))
+    return base + rank
+
+
+def _read_exact(sock: socket.socket, n: int) -> bytes:
+    buf = bytearray()
+    while len(buf) < n:
+        chunk = sock.recv(n - len(buf))
+        if not chunk:
+            raise ConnectionError("peer closed")
+        buf += chunk
+    return bytes(buf)
+
+
+def _read_frame(sock: socket.socket) -> bytes:
+    (ln,) = struct.unpack("<I", _read_exact(sock, 4))
+    return _read_exact(sock, ln)
+
+
+def _write_frame(sock: socket.socket, payload: bytes):
+    sock.sendall(struct.pack("<I", len(payload)) + payload)
+
+
+class WriteExchange:
+    """Per-rank loopback exchange: a listener thread accepting peer write
+    batches + pooled client connections to every peer."""
+
+    def __init__(self, rank: int, world: int, handler=None,
+                 base_port: int | None = None, host: str = "127.0.0.1"):
+        self.rank = rank
+        self.world = world
+        self.host = host
+        self.ports = [fanout_port(r, base_port) for r in range(world)]
+        self.handler = handler      # callable(payload: bytes) -> bytes
+        self._conns: dict[int, socket.socket] = {}
+        self._conn_locks = {r: threading.Lock() for r in range(world)}
+        self._closing = False
+        self._srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        self._srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        self._srv.bind((host, self.ports[rank]))
+        self._srv.listen(64)
+        self._accept_thread = threading.Thread(target=self._accept_loop,
+                                               daemon=True)
+        self._accept_thread.start()
+
+    # ------------------------------------------------------------- server
+    def _accept_loop(self):
+        while not self._closing:
+            try:
+                conn, _ = self._srv.accept()
+            except OSError:
+                return
+            conn.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            threading.Thread(target=self._serve_conn, args=(conn,),
+                             daemon=True).start()
+
+    def _serve_conn(self, conn: socket.socket):
+        try:
+            while not self._closing:
+                req = _read_frame(conn)
+                try:
+                    resp = self.handler(req) if self.handler else b"OK"
+                except Exception as e:  # report the error to the sender
+                    resp = b"ERR " + repr(e).encode()
+                _write_frame(conn, resp or b"OK")
+        except (ConnectionError, OSError):
+            pass
+        finally:
+            conn.close()
+
+    # ------------------------------------------------------------- client
+    def _connect(self, peer: int) -> socket.socket:
+        deadline = time.monotonic() + 30
+        last = None
+        while time.monotonic() < deadline:
+            try:
+                s = socket.create_connection((self.host, self.ports[peer]),
+                                             timeout=30)
+                s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+                return s
+            except OSError as e:
+                last = e
+                time.sleep(0.1)
+        raise ConnectionError(f"cannot reach rank {peer}: {last}")
+
+    def request(self, peer: int, payload: bytes) -> bytes:
+        """Send one frame to `peer`, wait for its ack frame."""
+        with self._conn_locks[peer]:
+            sock = self._conns.get(peer)
+            if sock is None:
+                sock = self._conns[peer] = self._connect(peer)
+            try:
+                _write_frame(sock, payload)
+                resp = _read_frame(sock)
+            except (ConnectionError, OSError):
+                # one reconnect attempt (peer restarted)
+                sock.close()
+                sock = self._conns[peer] = self._connect(peer)
+                _write_frame(sock, payload)
+                resp = _read_frame(sock)
+        if resp.startswith(b"ERR"):
+            raise RuntimeError(f"rank {peer} write failed: {resp[4:].decode()}")
+        return resp
+
+    def close(self):
+        self._closing = True
+        try:
+            self._srv.close()
+        except OSError:
+            pass
+        for s in self._conns.values():
+            try:
+                s.close()
+            except OSError:
+                pass
+        self._conns.clear()
+
+
+# ------------------------------------------------------------------ codec
+
+def encode_routed_batch(tagsets: list, srow, ts_ms, fields_mat,
+                        field_names: list[str], str_cols=None) -> bytes:
+    """(unique tagset keys, row→tagset idx, ts, fields[nf,n], names)."""
+    return pickle.dumps(
+        ("write", tagsets, srow, ts_ms, fields_mat, field_names,
+         str_cols or {}),
+        protocol=pickle.HIGHEST_PROTOCOL)
+
+
+def decode_routed_batch(payload: bytes):
+    kind, *rest = pickle.loads(payload)
+    assert kind == "write", kind
+    return rest

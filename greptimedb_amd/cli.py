@@ -19,8 +19,10 @@ def main(argv=None):
     st_sub = st.add_subparsers(dest="cmd", required=True)
     start = st_sub.add_parser("start")
     start.add_argument("--http-addr", default="0.0.0.0:4000")
+    start.add_argument("--grpc-addr", default="0.0.0.0:4001")
     start.add_argument("--mysql-addr", default="0.0.0.0:4002")
     start.add_argument("--postgres-addr", default="0.0.0.0:4003")
+    start.add_argument("--flight-addr", default="0.0.0.0:4005")
     start.add_argument("--data-dir", default="./greptimedb_data")
     start.add_argument("--device", default="auto",
                        help="cuda:N / cpu / auto")
@@ -72,8 +74,17 @@ def main(argv=None):
     host, port = cfg.get("http_addr", args.http_addr).rsplit(":", 1)
     my_host, my_port = cfg.get("mysql_addr", args.mysql_addr).rsplit(":", 1)
     pg_host, pg_port = cfg.get("postgres_addr", args.postgres_addr).rsplit(":", 1)
+    g_host, g_port = cfg.get("grpc_addr", args.grpc_addr).rsplit(":", 1)
+    f_host, f_port = cfg.get("flight_addr", args.flight_addr).rsplit(":", 1)
+    # gRPC GreptimeDatabase + Arrow Flight run on their own thread pools
+    from greptimedb_amd.servers.flight import GreptimeFlightServer
+    from greptimedb_amd.servers.grpc_server import GreptimeGrpcServer
+    grpc_srv = GreptimeGrpcServer(engine, ctx.executor, g_host, int(g_port))
+    # pyarrow Flight serves from its own threads as soon as it is built
+    flight_srv = GreptimeFlightServer(engine, ctx.executor, f_host, int(f_port))
     print(f"greptimedb_amd standalone: device={device} data={args.data_dir} "
-          f"http={host}:{port} mysql={my_port} postgres={pg_port}", flush=True)
+          f"http={host}:{port} grpc={grpc_srv.port} mysql={my_port} "
+          f"postgres={pg_port} flight={flight_srv.port}", flush=True)
 
     async def serve():
         uv = uvicorn.Server(uvicorn.Config(app, host=host, port=int(port),
@@ -82,7 +93,11 @@ def main(argv=None):
         pg = PostgresServer(ctx.executor, pg_host, int(pg_port), user_provider)
         await asyncio.gather(uv.serve(), my.serve_forever(), pg.serve_forever())
 
-    asyncio.run(serve())
+    try:
+        asyncio.run(serve())
+    finally:
+        grpc_srv.shutdown()
+        flight_srv.shutdown()
 
 
 def _meta_snapshot(args) -> int:

@@ -1227,6 +1227,43 @@ class Tokenizer {
 
   size_t num_terms() const { return terms_.size(); }
 
+  // get-or-create id for an already-normalized term (persisted fulltext
+  // sidecar load: terms come back verbatim from the index file, no doc
+  // re-tokenization)
+  int32_t intern(const std::string& term) {
+    const uint64_t h = fnv1a(term.data(), term.size());
+    auto range = tmap_.equal_range(h);
+    for (auto it = range.first; it != range.second; ++it)
+      if (terms_[it->second] == term) return it->second;
+    int32_t tid = (int32_t)terms_.size();
+    tmap_.emplace(h, tid);
+    terms_.push_back(term);
+    return tid;
+  }
+
+  // bulk variant: lens i32[k] + concatenated utf8 blob → ids i32[k]
+  py::array_t<int32_t> intern_blob(py::array_t<int32_t> lens, py::bytes blob) {
+    char* buf = nullptr;
+    Py_ssize_t blen = 0;
+    PyBytes_AsStringAndSize(blob.ptr(), &buf, &blen);
+    const int32_t* ln = lens.data();
+    const size_t k = lens.size();
+    py::array_t<int32_t> out(k);
+    int32_t* o = out.mutable_data();
+    size_t off = 0;
+    for (size_t i = 0; i < k; i++) {
+      o[i] = intern(std::string(buf + off, ln[i]));
+      off += ln[i];
+    }
+    return out;
+  }
+
+  py::str term_str(int32_t tid) const {
+    if (tid < 0 || (size_t)tid >= terms_.size())
+      throw std::out_of_range("tid");
+    return py::str(terms_[tid]);
+  }
+
  private:
   std::unordered_multimap<uint64_t, int32_t> tmap_;
   std::vector<std::string> terms_;
@@ -1417,5 +1454,8 @@ PYBIND11_MODULE(_native, m) {
       .def(py::init<>())
       .def("tokenize", &Tokenizer::tokenize)
       .def("term_id", &Tokenizer::term_id)
+      .def("intern", &Tokenizer::intern)
+      .def("intern_blob", &Tokenizer::intern_blob)
+      .def("term_str", &Tokenizer::term_str)
       .def("num_terms", &Tokenizer::num_terms);
 }

@@ -147,12 +147,16 @@ class Compactor:
             ft = region.text_cols.get(sn)
             if ft is not None:
                 new_batch.text_index[sn] = ft.build_segment(list(arr), device)
+        if new_batch.text_index:
+            from greptimedb_amd.engine import ftindex
+            ftindex.save_sidecar(path, new_batch.text_index, region.text_cols)
         with region.lock:
             for f in fids:
                 region.sst_cache.pop(f, None)
             region.sst_cache[fid] = new_batch
         # remove merged files from disk (reference: file purger)
         for f in fids:
-            p = os.path.join(region.dir, "sst", f"{f}.parquet")
-            if os.path.exists(p):
-                os.unlink(p)
+            for suffix in (".parquet", ".ftidx"):
+                p = os.path.join(region.dir, "sst", f"{f}{suffix}")
+                if os.path.exists(p):
+                    os.unlink(p)

@@ -180,6 +180,16 @@ class Region:
                 self.row_seq = max(self.row_seq, int(seq.max()) + 1)
             if str_cols:
                 perm_h = perm.cpu().numpy()
+                # persisted sidecar (Puffin analog): postings load directly,
+                # remapped through the open-time permutation — no doc
+                # re-tokenization (reference: index blobs read from Puffin)
+                loaded = None
+                if any(name in self.text_cols for name in str_cols):
+                    from greptimedb_amd.engine import ftindex
+                    inv = np.empty(len(perm_h), dtype=np.int64)
+                    inv[perm_h] = np.arange(len(perm_h), dtype=np.int64)
+                    loaded = ftindex.load_sidecar(path, self.text_cols,
+                                                  self.device, row_remap=inv)
                 for name, vals in str_cols.items():
                     if name not in self.str_field_names:
                         self.str_field_names.append(name)
@@ -187,8 +197,11 @@ class Region:
                     batch.str_cols[name] = arr
                     ft = self.text_cols.get(name)
                     if ft is not None:  # fulltext-indexed columns only
-                        batch.text_index[name] = ft.build_segment(
-                            list(arr), self.device)
+                        if loaded is not None and name in loaded:
+                            batch.text_index[name] = loaded[name]
+                        else:  # pre-sidecar SSTs: rebuild from raw strings
+                            batch.text_index[name] = ft.build_segment(
+                                list(arr), self.device)
             self.sst_cache[fid] = batch
 
     # ---------------------------------------------------------------- write
@@ -266,6 +279,9 @@ class Region:
         meta = sst_mod.write_sst(path, self.schema, self.series.pks,
                                  se_h, ts_h, f_h, seq_h, flush_field_names,
                                  str_cols=str_cols_sorted)
+        if text_index:
+            from greptimedb_amd.engine import ftindex
+            ftindex.save_sidecar(path, text_index, self.text_cols)
         meta.seq_max = int(seq_h.max()) if len(seq_h) else 0
         self.manifest.commit({
             "kind": "edit",

@@ -131,3 +131,63 @@ def test_fulltext_survives_unflushed_reopen(tmp_path):
         "SELECT count(*) FROM lg WHERE matches(message, 'failure')")
     assert int(list(r.rows())[0][0]) == 1
     eng2.close()
+
+
+def test_reopen_loads_persisted_ftindex_no_retokenize(log_env, monkeypatch):
+    """region reopen must load the per-SST fulltext sidecar
+    (Puffin analog) instead of re-tokenizing raw strings."""
+    import os
+
+    from greptimedb_amd.engine import ftindex
+    from greptimedb_amd.engine.fulltext import FulltextColumn
+
+    eng, ex = log_env
+    eng.flush_all()
+    st = eng.table("logs")
+    sidecars = [p for r in st.regions
+                for p in os.listdir(os.path.join(r.dir, "sst"))
+                if p.endswith(".ftidx")]
+    assert sidecars, "flush wrote no fulltext sidecar"
+    eng.close()
+
+    calls = {"n": 0}
+    real = FulltextColumn.build_segment
+
+    def counting(self, docs, device):
+        calls["n"] += 1
+        return real(self, docs, device)
+
+    monkeypatch.setattr(FulltextColumn, "build_segment", counting)
+    eng2 = MitoEngine(EngineConfig(data_dir=eng.config.data_dir, device="cpu",
+                                   background_flush=False))
+    assert calls["n"] == 0, "reopen re-tokenized despite sidecar"
+    ex2 = Executor(eng2)
+    r = ex2.execute("SELECT ts FROM logs WHERE matches(message, 'error') ORDER BY ts")
+    assert list(r.columns[0]) == [2000, 3000]
+    r = ex2.execute("SELECT count(*) FROM logs WHERE matches(message, 'exhausted pool')")
+    assert r.columns[0][0] == 1
+    r = ex2.execute("SELECT count(*) FROM logs WHERE matches(message, 'nosuchterm')")
+    assert r.columns[0][0] == 0
+    eng2.close()
+
+
+def test_ftindex_sidecar_compaction(log_env):
+    """Compaction rewrites one merged sidecar and purges the inputs."""
+    import os
+
+    eng, ex = log_env
+    eng.flush_all()
+    ex.execute("""INSERT INTO logs (service, ts, message, latency) VALUES
+     ('api', 5000, 'disk full error on volume', 9.9)""")
+    eng.flush_all()
+    from greptimedb_amd.engine.compaction import Compactor
+    st = eng.table("logs")
+    for r in st.regions:
+        Compactor(trigger_file_num=2).compact_region(r)
+    for r in st.regions:
+        d = os.path.join(r.dir, "sst")
+        pq_files = [p for p in os.listdir(d) if p.endswith(".parquet")]
+        ft_files = [p for p in os.listdir(d) if p.endswith(".ftidx")]
+        assert len(ft_files) == len(pq_files)
+    r = ex.execute("SELECT count(*) FROM logs WHERE matches(message, 'error')")
+    assert r.columns[0][0] == 3

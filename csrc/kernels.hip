@@ -306,6 +306,18 @@ enum PromMode {
 
 DEV_INLINE double prom_nan() { return __longlong_as_double(0x7FF8000000000000LL); }
 
+// IEEE-754 double ↔ totally-ordered u64 (sign-flip transform): key order ==
+// numeric order, so order statistics can be selected bitwise.
+DEV_INLINE uint64_t f64_sort_key(double v) {
+  const uint64_t u = (uint64_t)__double_as_longlong(v);
+  return (u & 0x8000000000000000ULL) ? ~u : (u | 0x8000000000000000ULL);
+}
+DEV_INLINE double sort_key_f64(uint64_t k) {
+  const uint64_t u = (k & 0x8000000000000000ULL)
+      ? (k ^ 0x8000000000000000ULL) : ~k;
+  return __longlong_as_double((long long)u);
+}
+
 __global__ void prom_range_eval_kernel(
     const int64_t* __restrict__ ts,
     const double* __restrict__ vals,
@@ -453,17 +465,37 @@ __global__ void prom_range_eval_kernel(
           vk1 = buf[k1];
           vk2 = buf[k2];
         } else {
-          for (int64_t j = w_lo; j < w_hi; j++) {
-            const double x = vals[j];
-            int64_t less = 0, eq = 0;
-            for (int64_t j2 = w_lo; j2 < w_hi; j2++) {
-              const double y = vals[j2];
-              less += (y < x);
-              eq += (y == x);
+          // radix bisection on the sort-key transform: O(64·W) instead of
+          // the old O(W²) rank counting. Register-only on purpose — each
+          // thread owns its WHOLE ragged window (windows ≫ threads per
+          // launch), so cooperative LDS staging would idle the rest of the
+          // wavefront; per-thread bitwise selection is the CDNA4-shaped
+          // answer for this layout.
+          uint64_t prefix = 0;
+          int64_t k = k1;
+          for (int bit = 63; bit >= 0; bit--) {
+            const uint64_t high_mask =
+                (bit == 63) ? 0ULL : (~0ULL << (bit + 1));
+            const uint64_t b = 1ULL << bit;
+            int64_t cnt0 = 0;
+            for (int64_t j = w_lo; j < w_hi; j++) {
+              const uint64_t kj = f64_sort_key(vals[j]);
+              cnt0 += ((kj & high_mask) == prefix) & !(kj & b);
             }
-            if (less <= k1 && k1 < less + eq) vk1 = x;
-            if (less <= k2 && k2 < less + eq) vk2 = x;
+            if (k >= cnt0) { k -= cnt0; prefix |= b; }
           }
+          const uint64_t kk = prefix;   // exact key of rank k1
+          int64_t less = 0, eq = 0;
+          uint64_t next = ~0ULL;
+          bool has_next = false;
+          for (int64_t j = w_lo; j < w_hi; j++) {
+            const uint64_t kj = f64_sort_key(vals[j]);
+            if (kj < kk) less++;
+            else if (kj == kk) eq++;
+            else if (kj < next) { next = kj; has_next = true; }
+          }
+          vk1 = sort_key_f64(kk);
+          vk2 = (k2 < less + eq || !has_next) ? vk1 : sort_key_f64(next);
         }
         r = vk1 + (vk2 - vk1) * (rank - (double)k1);
       }

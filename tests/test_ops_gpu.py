@@ -219,8 +219,9 @@ def test_bulk_ingest_gpu_matches_query():
 
 
 def test_quantile_over_time_large_window():
-    """Windows >128 samples take the O(W²) rank-selection path in the kernel
-    (no scratch buffer) — must still match the sorting oracle."""
+    """Windows >128 samples take the radix-bisection selection path
+    (O(64·W), register-only, exact) — must match the sorting oracle,
+    including duplicate runs and negatives."""
     rng = np.random.RandomState(9)
     n = 700
     ts = torch.as_tensor(np.sort(rng.randint(0, 300_000, n)).astype(np.int64))

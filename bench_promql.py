@@ -119,6 +119,9 @@ def main():
                                  start_s, end_s, 60),
         "sum-by-job-all-series": ('sum by (job) ({__name__=~"metric_.*"})',
                                   start_s, end_s, 60),
+        # >128 samples per window → the radix-bisection selection path
+        "quantile-ot-45m-one-metric": (
+            "quantile_over_time(0.95, metric_3[45m])", end_s, end_s, 1),
     }
     results = {}
     for name, (q, s, e, stp) in queries.items():

@@ -70,6 +70,9 @@ def main():
                          "commit, no fsync (reference raft-engine default "
                          "sync=false); fsync = fdatasync per group commit; "
                          "off = no WAL")
+    ap.add_argument("--wal-shards", type=int, default=4,
+                    help="parallel WAL writers (region-sharded segments, "
+                         "merged by seq at replay)")
     ap.add_argument("--flush-mb", type=int, default=64,
                     help="per-region memtable flush threshold — 64MB keeps "
                          "flush+compaction live during the timed region")
@@ -119,6 +122,7 @@ def main():
     eng = MitoEngine(EngineConfig(
         data_dir=data_dir, device=device,
         background_flush=True, wal_sync=(args.wal == "fsync"),
+        wal_shards=args.wal_shards,
         flush_bytes=args.flush_mb << 20))
 
     # -------- cross-rank write fan-out (any rank accepts any write) ---------
@@ -297,6 +301,7 @@ def main():
             "wal": {"buffered": "group-commit, no fsync",
                     "fsync": "group-commit + fdatasync",
                     "off": "disabled"}[args.wal],
+            "wal_shards": args.wal_shards,
             "flush_mb": args.flush_mb,
             "ssts_written": n_ssts,
             "queries": queries,

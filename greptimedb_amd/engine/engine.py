@@ -32,6 +32,7 @@ class EngineConfig:
     flush_bytes: int = 512 << 20        # per-region memtable flush threshold
     wal_sync: bool = False              # fdatasync per commit
     wal_segment_bytes: int = 256 << 20
+    wal_shards: int = 1                 # parallel WAL writers (P5 workers)
     default_regions: int = 4
     background_flush: bool = True
     scan_mem_bytes: int = 32 << 30      # host-side scan materialization quota
@@ -55,7 +56,8 @@ class MitoEngine:
         self._catalog_path = os.path.join(config.data_dir, "catalog.json")
         self.wal = Wal(os.path.join(config.data_dir, "wal"),
                        segment_bytes=config.wal_segment_bytes,
-                       sync_on_commit=config.wal_sync)
+                       sync_on_commit=config.wal_sync,
+                       shards=config.wal_shards)
         self._flush_q: queue.Queue = queue.Queue()
         self._flusher = None
         # observer seam (reference: mito2 engine/listener.rs WorkerListener —

@@ -1,17 +1,26 @@
-"""WAL: segment management + batch payload codec over the native writer.
+"""WAL: sharded segment management + batch payload codec over the native writer.
 
 Reference parity: src/log-store raft_engine backend + mito2/src/wal.rs
 (WalWriter group commit, per-region entries, replay from entry id,
-obsolete/purge). Segments are `{first_seq:020d}.wal`; frames are written by
-the C++ WalWriter (crc32'd, torn-tail safe). Payloads carry a parsed
-columnar write batch (see encode_batch) so replay does not re-parse wire
-protocol.
+obsolete/purge). Segments are `{first_seq:020d}[.s{shard}].wal`; frames are
+written by the C++ WalWriter (crc32'd, torn-tail safe). Payloads carry a
+parsed columnar write batch (see encode_batch) so replay does not re-parse
+wire protocol.
+
+Sharding (P5 write workers, round-1 VERDICT #8): with `shards > 1` each
+region maps to one shard (own writer + own lock + own segment files), so
+parallel ingest workers stop serializing on a single append lock — only
+the global sequence counter stays shared (a two-instruction critical
+section). Sequences are globally monotonic, so replay heap-merges the
+per-shard streams by seq and all flushed_seq/purge logic is unchanged.
 """
 
 from __future__ import annotations
 
+import heapq
 import json
 import os
+import re
 import struct
 import threading
 
@@ -77,73 +86,124 @@ def decode_batch(buf: bytes):
     return series, ts, fields, hdr["fields"], new_series, str_cols
 
 
+_SEG_RE = re.compile(r"^(\d{20})(?:\.s(\d+))?\.wal$")
+
+
+class _Shard:
+    __slots__ = ("idx", "dir", "writer", "lock", "dirty")
+
+    def __init__(self, idx: int, dir: str):
+        self.idx = idx
+        self.dir = dir
+        self.writer = _native.WalWriter()
+        self.lock = threading.Lock()
+        self.dirty = False
+
+    def seg_path(self, first_seq: int) -> str:
+        if self.idx == 0:
+            return os.path.join(self.dir, f"{first_seq:020d}.wal")
+        return os.path.join(self.dir, f"{first_seq:020d}.s{self.idx}.wal")
+
+
 class Wal:
-    def __init__(self, dir: str, segment_bytes: int = 128 << 20, sync_on_commit: bool = False):
+    def __init__(self, dir: str, segment_bytes: int = 128 << 20,
+                 sync_on_commit: bool = False, shards: int = 1):
         self.dir = dir
         os.makedirs(dir, exist_ok=True)
         self.segment_bytes = segment_bytes
         self.sync_on_commit = sync_on_commit
-        self.writer = _native.WalWriter()
         self.next_seq = 1
         # last appended seq per region — updated atomically with the seq
         # assignment so purge decisions never race an in-flight append
         self.region_last: dict[int, int] = {}
-        self._lock = threading.Lock()  # multi-worker ingest (P5 write workers)
-        segs = self.segments()
-        if segs:
+        self._lock = threading.Lock()   # guards next_seq + region_last
+        self.shards = [_Shard(i, dir) for i in range(max(shards, 1))]
+        if self.segments():
             # resume: next_seq = last replayed seq + 1 (caller replays first)
             for _, rid, seq, _ in self.replay():
                 self.next_seq = max(self.next_seq, seq + 1)
                 self.region_last[rid] = seq
-            self._open_new_segment()
-        else:
-            self._open_new_segment()
+        for sh in self.shards:
+            sh.writer.open_segment(sh.seg_path(self.next_seq))
 
-    def _seg_path(self, first_seq: int) -> str:
-        return os.path.join(self.dir, f"{first_seq:020d}.wal")
+    # ------------------------------------------------------------- layout
+    def segments(self, shard: int | None = None) -> list[str]:
+        out = []
+        for f in os.listdir(self.dir):
+            m = _SEG_RE.match(f)
+            if not m:
+                continue
+            if shard is not None and int(m.group(2) or 0) != shard:
+                continue
+            out.append(f)
+        return sorted(out)
 
-    def segments(self) -> list[str]:
-        return sorted(f for f in os.listdir(self.dir) if f.endswith(".wal"))
+    def _shard_of(self, region_id: int) -> _Shard:
+        return self.shards[region_id % len(self.shards)]
 
-    def _open_new_segment(self):
-        self.writer.open_segment(self._seg_path(self.next_seq))
-
+    # ------------------------------------------------------------- append
     def append(self, region_id: int, payload: bytes) -> int:
         with self._lock:
             seq = self.next_seq
             self.next_seq += 1
-            self.writer.append(region_id, seq, payload)
             self.region_last[region_id] = seq
-            return seq
+        sh = self._shard_of(region_id)
+        with sh.lock:
+            sh.writer.append(region_id, seq, payload)
+            sh.dirty = True
+        return seq
 
     def commit(self):
-        with self._lock:
-            size = self.writer.commit(self.sync_on_commit)
-            if size >= self.segment_bytes:
-                self.writer.close_segment()
-                self._open_new_segment()
+        for sh in self.shards:
+            if not sh.dirty:
+                continue
+            with sh.lock:
+                if not sh.dirty:
+                    continue
+                size = sh.writer.commit(self.sync_on_commit)
+                sh.dirty = False
+                if size >= self.segment_bytes:
+                    sh.writer.close_segment()
+                    sh.writer.open_segment(sh.seg_path(self.next_seq))
 
+    # ------------------------------------------------------------- replay
     def replay(self):
-        """Yield (seg_name, region_id, seq, payload) in order."""
-        for seg in self.segments():
-            for region, seq, payload in _native.wal_read_segment(os.path.join(self.dir, seg)):
-                yield seg, region, seq, payload
+        """Yield (seg_name, region_id, seq, payload) in GLOBAL seq order
+        (heap-merge of the per-shard seq-ascending streams — a region's
+        entries replay in order even if the shard count changed)."""
+        def shard_stream(shard_idx):
+            for seg in self.segments(shard_idx):
+                for region, seq, payload in _native.wal_read_segment(
+                        os.path.join(self.dir, seg)):
+                    yield seq, seg, region, payload
+
+        streams = [shard_stream(i) for i in range(len(self.shards))]
+        # segments of shards beyond the current count (count was lowered):
+        present = {int(_SEG_RE.match(f).group(2) or 0) for f in self.segments()}
+        for extra in sorted(present - set(range(len(self.shards)))):
+            streams.append(shard_stream(extra))
+        for seq, seg, region, payload in heapq.merge(*streams):
+            yield seg, region, seq, payload
 
     def purge_before(self, seq: int):
-        """Delete whole segments whose every entry has seq < `seq`.
-        A segment named by its first seq is obsolete when the NEXT segment's
-        first seq is <= `seq` (reference: WAL truncation after flush)."""
-        segs = self.segments()
-        for i, seg in enumerate(segs[:-1]):
-            nxt_first = int(segs[i + 1].split(".")[0])
-            if nxt_first <= seq:
-                os.unlink(os.path.join(self.dir, seg))
+        """Delete whole segments whose every entry has seq < `seq` — per
+        shard, a segment is obsolete when the shard's NEXT segment's first
+        seq is <= `seq` (reference: WAL truncation after flush)."""
+        for sh in range(len(self.shards)):
+            segs = self.segments(sh)
+            for i, seg in enumerate(segs[:-1]):
+                nxt_first = int(_SEG_RE.match(segs[i + 1]).group(1))
+                if nxt_first <= seq:
+                    os.unlink(os.path.join(self.dir, seg))
 
     def close(self):
-        if self.writer is not None:
+        for sh in self.shards:
+            if sh.writer is None:
+                continue
             try:
-                self.writer.commit(self.sync_on_commit)
+                sh.writer.commit(self.sync_on_commit)
             except RuntimeError:
                 pass  # already closed
-            self.writer.close_segment()
-            self.writer = None
+            sh.writer.close_segment()
+            sh.writer = None
+        self.shards = []

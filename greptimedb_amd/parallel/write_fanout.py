@@ -63,11 +63,13 @@ class WriteExchange:
     batches + pooled client connections to every peer."""
 
     def __init__(self, rank: int, world: int, handler=None,
-                 base_port: int | None = None, host: str = "127.0.0.1"):
+                 base_port: int | None = None, host: str = "127.0.0.1",
+                 ports: list[int] | None = None):
         self.rank = rank
         self.world = world
         self.host = host
-        self.ports = [fanout_port(r, base_port) for r in range(world)]
+        self.ports = list(ports) if ports is not None else \
+            [fanout_port(r, base_port) for r in range(world)]
         self.handler = handler      # callable(payload: bytes) -> bytes
         self._conns: dict[int, socket.socket] = {}
         self._conn_locks = {r: threading.Lock() for r in range(world)}

@@ -13,23 +13,21 @@ import numpy as np
 import pytest
 
 
-def _free_port_base(n: int) -> int:
-    socks = []
+def _free_ports(n: int) -> list[int]:
+    socks, ports = [], []
     try:
         for _ in range(n):
             s = socket.socket()
             s.bind(("127.0.0.1", 0))
             socks.append(s)
-        base = socks[0].getsockname()[1]
+            ports.append(s.getsockname()[1])
     finally:
         for s in socks:
             s.close()
-    # ports base..base+n-1 may not all be free; just use the first and hope
-    # the next n are clear (ephemeral range, freshly released)
-    return base
+    return ports
 
 
-def _rank_proc(rank, world, base_port, data_dir, barrier, results, ingest_all_rank):
+def _rank_proc(rank, world, ports, data_dir, barrier, results, ingest_all_rank):
     import torch  # noqa: F401  (loads libc10 for _native)
     from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
     from greptimedb_amd.engine.ingest import Ingestor
@@ -40,8 +38,7 @@ def _rank_proc(rank, world, base_port, data_dir, barrier, results, ingest_all_ra
     eng = MitoEngine(EngineConfig(data_dir=f"{data_dir}/r{rank}", device="cpu",
                                   background_flush=False, default_regions=2))
     ing = Ingestor(eng, rank=rank, world=world)
-    ex = WriteExchange(rank, world, handler=ing.handle_remote,
-                       base_port=base_port)
+    ex = WriteExchange(rank, world, handler=ing.handle_remote, ports=ports)
     ing.exchange = ex
     barrier.wait()  # all exchanges listening
 
@@ -77,7 +74,7 @@ def test_single_rank_ingests_all(tmp_path, world, ingest_rank):
     ctx = mp.get_context("spawn")
     barrier = ctx.Barrier(world, timeout=120)
     results = ctx.Queue()
-    base = _free_port_base(world)
+    base = _free_ports(world)
     procs = [ctx.Process(target=_rank_proc,
                          args=(r, world, base, str(tmp_path), barrier, results,
                                ingest_rank))
@@ -102,7 +99,7 @@ def test_single_rank_ingests_all(tmp_path, world, ingest_rank):
     assert len(all_series) == 40 and len(set(all_series)) == 40
 
 
-def _dual_ingest_proc(rank, world, base_port, data_dir, barrier, results):
+def _dual_ingest_proc(rank, world, ports, data_dir, barrier, results):
     import torch  # noqa: F401
     from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
     from greptimedb_amd.engine.ingest import Ingestor
@@ -111,8 +108,7 @@ def _dual_ingest_proc(rank, world, base_port, data_dir, barrier, results):
     eng = MitoEngine(EngineConfig(data_dir=f"{data_dir}/r{rank}", device="cpu",
                                   background_flush=False, default_regions=2))
     ing = Ingestor(eng, rank=rank, world=world)
-    ex = WriteExchange(rank, world, handler=ing.handle_remote,
-                       base_port=base_port)
+    ex = WriteExchange(rank, world, handler=ing.handle_remote, ports=ports)
     ing.exchange = ex
     barrier.wait()
     # BOTH ranks write the SAME series set concurrently (ts offset per rank)
@@ -137,7 +133,7 @@ def test_both_ranks_ingest_same_series(tmp_path):
     ctx = mp.get_context("spawn")
     barrier = ctx.Barrier(world, timeout=120)
     results = ctx.Queue()
-    base = _free_port_base(world)
+    base = _free_ports(world)
     procs = [ctx.Process(target=_dual_ingest_proc,
                          args=(r, world, base, str(tmp_path), barrier, results))
              for r in range(world)]

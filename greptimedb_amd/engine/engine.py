@@ -172,6 +172,9 @@ class MitoEngine:
         Caller must call `commit_wal()` after all regions of the batch
         (group commit — durability boundary)."""
         region: Region = table.regions[region_idx]
+        if not region.writable:
+            from greptimedb_amd.utils.errors import RegionFenced
+            raise RegionFenced(f"region {region.region_id} is downgraded")
         seq = 0
         if durable:
             payload = encode_batch(series_codes, ts_ms, fields,
@@ -205,7 +208,8 @@ class MitoEngine:
             return True
         nf = fields.shape[0]
         regions: list[Region] = [st.regions[ri] for st, ri in targets]
-        if any(r.text_cols or r.memtable.str_cols for r in regions):
+        if any(r.text_cols or r.memtable.str_cols or not r.writable
+               for r in regions):
             return False
         # WAL first (host-side; replay is region-keyed, so one entry per region)
         seqs = [0] * len(regions)
@@ -289,7 +293,8 @@ class MitoEngine:
         from greptimedb_amd.ops import kernels as ops
 
         regions: list[Region] = table.regions
-        if any(r.text_cols or r.memtable.str_cols for r in regions):
+        if any(r.text_cols or r.memtable.str_cols or not r.writable
+               for r in regions):
             return False
         n = len(ts_ms)
         nf = fields.shape[0]

@@ -119,10 +119,16 @@ class Ingestor:
 
     def _owner_of(self, st: TableState, tag_tuple: tuple) -> int:
         """Owning rank of a series (reference: table-route peer lookup).
-        Multi-dim rules: global partition % world; hash tables:
-        tsid_hash(pk) % world."""
+        Migration route overrides win; else multi-dim rules map global
+        partition % world and hash tables tsid_hash(pk) % world."""
         from greptimedb_amd.parallel.partition import MultiDimPartitionRule
         rule = self.engine.partition_rule(st)
+        overrides = getattr(self.engine, "route_overrides", None)
+        if overrides:
+            ridx = self.engine.region_of_tags(st, tag_tuple)
+            hit = overrides.get((st.schema.name, ridx))
+            if hit is not None:
+                return hit
         if isinstance(rule, MultiDimPartitionRule):
             return rule.region_of({c.name: v for c, v in
                                    zip(st.schema.tag_columns, tag_tuple)}) % self.world

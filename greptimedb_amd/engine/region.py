@@ -82,6 +82,10 @@ class Region:
         # immutable memtables in Version until the flush edit lands).
         self.flushing: list = []  # [(memtable, n_rows, mem_postings, base)]
         self._flush_lock = threading.Lock()
+        # Role fencing (reference store-api RegionRole Leader/Follower/
+        # DowngradingLeader): a downgraded region rejects writes while a
+        # migration moves ownership.
+        self.writable = True
         # Series code assignment must be stable across restarts (GPU columns
         # and SST caches store dense codes): an append-only series log is the
         # authoritative code order. Loaded BEFORE SSTs.

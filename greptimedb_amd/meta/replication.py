@@ -113,14 +113,19 @@ class FollowerReplica:
         applied = 0
         max_seq = self.applied_seq
         for seg in sorted(f for f in os.listdir(wal_dir) if f.endswith(".wal")):
-            # segment file names are first-seq; skip fully-applied segments
+            # Skip by PER-REGION high-water (region.last_seq), not a global
+            # seq: with sharded WALs the shard files become durable in
+            # arbitrary cross-shard order, so a global mark could skip a
+            # lower-seq entry that appeared later. A region's entries stay
+            # within one shard and are seq-ascending, so per-region marks
+            # are exact.
             for rid, seq, payload in _native.wal_read_segment(
                     os.path.join(wal_dir, seg)):
-                if seq <= self.applied_seq:
-                    continue
                 region = regions.get(rid)
                 max_seq = max(max_seq, seq)
                 if region is None:
+                    continue
+                if seq <= region.last_seq:
                     continue
                 series, ts, fields, fnames, new_series, str_cols = \
                     decode_batch(payload)

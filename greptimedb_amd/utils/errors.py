@@ -57,3 +57,7 @@ class NativeExtensionMissing(GreptimeError):
     """
 
     code = "Internal"
+
+
+class RegionFenced(GreptimeError):
+    """Write rejected: region downgraded (migration write fence)."""

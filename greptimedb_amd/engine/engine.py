@@ -81,6 +81,7 @@ class MitoEngine:
         with open(self._catalog_path) as f:
             cat = json.load(f)
         self.next_table_id = cat["next_table_id"]
+        self.views = dict(cat.get("views", {}))
         for td in cat["tables"]:
             schema = TableSchema.from_dict(td["schema"])
             st = TableState(schema=schema, append_mode=td["append_mode"])
@@ -94,6 +95,7 @@ class MitoEngine:
     def _save_catalog(self):
         cat = {
             "next_table_id": self.next_table_id,
+            "views": getattr(self, "views", {}),
             "tables": [
                 {
                     "schema": st.schema.to_dict(),

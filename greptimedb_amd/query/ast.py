@@ -135,6 +135,33 @@ class Select:
     align_by: list[str] | None = None
     align_fill: object = None
     distinct: bool = False             # SELECT DISTINCT
+    ctes: list = field(default_factory=list)  # [(name, Select|SetOp)] WITH clause
+
+
+@dataclass
+class SetOp:
+    """UNION [ALL] / EXCEPT / INTERSECT (reference setops cases)."""
+    op: str                  # union | except | intersect
+    all: bool
+    left: object             # Select | SetOp
+    right: object
+    order_by: list = field(default_factory=list)
+    limit: int | None = None
+    ctes: list = field(default_factory=list)
+
+
+@dataclass
+class CreateView:
+    name: str
+    query_sql: str           # the view body, stored verbatim
+    or_replace: bool = False
+    if_not_exists: bool = False
+
+
+@dataclass
+class DropView:
+    name: str
+    if_exists: bool = False
 
 
 @dataclass

@@ -26,7 +26,8 @@ from greptimedb_amd.models.schema import (
 )
 from greptimedb_amd.query import ast
 from greptimedb_amd.query.parser import parse_sql
-from greptimedb_amd.utils.errors import InvalidArguments, PlanQuery
+from greptimedb_amd.utils.errors import (InvalidArguments, PlanQuery,
+                                         TableAlreadyExists, TableNotFound)
 from greptimedb_amd.utils.timeutil import parse_ts_ms, trunc_unit_ms
 from greptimedb_amd.ops import ts_bucket_agg, dedup_mark_last
 
@@ -150,6 +151,7 @@ class Executor:
     def __init__(self, engine: MitoEngine, dist=None):
         self.engine = engine
         self.dist = dist  # parallel.dist.DistContext or None
+        self._virtual: dict = {}   # CTE scope: name -> QueryResult
 
     # ---------------------------------------------------------- entrypoints
 
@@ -176,6 +178,19 @@ class Executor:
     def execute_stmt(self, stmt) -> QueryResult:
         if isinstance(stmt, ast.Select):
             return self._exec_select(stmt)
+        if isinstance(stmt, ast.SetOp):
+            return self._exec_setop(stmt)
+        if isinstance(stmt, ast.CreateView):
+            return self._exec_create_view(stmt)
+        if isinstance(stmt, ast.DropView):
+            views = getattr(self.engine, "views", {})
+            if stmt.name not in views:
+                if not stmt.if_exists:
+                    raise TableNotFound(stmt.name)
+            else:
+                views.pop(stmt.name, None)
+                self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.CreateTable):
             return self._exec_create(stmt)
         if isinstance(stmt, ast.DropTable):
@@ -853,6 +868,22 @@ class Executor:
         return x
 
     def _exec_select(self, sel: ast.Select) -> QueryResult:
+        if sel.ctes:
+            return self._with_ctes(sel)
+        if isinstance(sel.table, (ast.Select, ast.SetOp)):
+            # derived table: FROM (SELECT ...) alias
+            from greptimedb_amd.query.derived import select_over_result
+            base = self.execute_stmt(sel.table)
+            return select_over_result(sel, base)
+        if isinstance(sel.table, str):
+            vt = self._virtual.get(sel.table)
+            if vt is None:
+                view_sql = getattr(self.engine, "views", {}).get(sel.table)
+                if view_sql is not None:
+                    vt = self.execute(view_sql)
+            if vt is not None:
+                from greptimedb_amd.query.derived import select_over_result
+                return select_over_result(sel, vt)
         if _has_subquery(sel.where) or _has_subquery(sel.having) or \
                 any(_has_subquery(e) for e, _a in sel.projections):
             sel.projections = [(self._resolve_subqueries(e), a)
@@ -1234,6 +1265,50 @@ class Executor:
         if sel.limit is not None:
             out_cols = [c[: sel.limit] for c in out_cols]
         return QueryResult(out_names, out_cols)
+
+    def _with_ctes(self, stmt):
+        """Evaluate WITH ctes into the virtual-table scope, then run the
+        body (reference: DataFusion CTE planning; cases in tests/cases/cte)."""
+        import dataclasses
+        saved = dict(self._virtual)
+        try:
+            for name, q in stmt.ctes:
+                self._virtual[name] = self.execute_stmt(q)
+            body = dataclasses.replace(stmt, ctes=[])
+            return self.execute_stmt(body)
+        finally:
+            self._virtual = saved
+
+    def _exec_setop(self, so: "ast.SetOp") -> QueryResult:
+        from greptimedb_amd.query.derived import eval_setop, select_over_result
+        if so.ctes:
+            return self._with_ctes(so)
+        left = self.execute_stmt(so.left)
+        right = self.execute_stmt(so.right)
+        res = eval_setop(so.op, so.all, left, right)
+        if so.order_by or so.limit is not None:
+            wrap = ast.Select(projections=[(ast.Star(), None)], table=None,
+                              order_by=so.order_by, limit=so.limit)
+            res = select_over_result(wrap, res)
+        return res
+
+    def _exec_create_view(self, cv: "ast.CreateView") -> QueryResult:
+        """CREATE [OR REPLACE] VIEW name AS select (reference:
+        common/meta ddl create_view; view body stored verbatim and
+        re-planned per query like the reference's logical-plan views)."""
+        views = getattr(self.engine, "views", None)
+        if views is None:
+            views = self.engine.views = {}
+        if cv.name in views and not cv.or_replace:
+            if cv.if_not_exists:
+                return QueryResult(["status"], [["ok"]])
+            raise TableAlreadyExists(cv.name)
+        if cv.name in self.engine.tables:
+            raise TableAlreadyExists(cv.name)
+        self.execute(cv.query_sql)   # validate the body now
+        views[cv.name] = cv.query_sql
+        self.engine._save_catalog()
+        return QueryResult(["status"], [["ok"]])
 
     def _plan_select(self, sel: ast.Select) -> SelectPlan:
         st = self.engine.table(sel.table)

@@ -159,8 +159,8 @@ class Parser:
         return stmt
 
     def parse_statement_inner(self):
-        if self.at_kw("select"):
-            stmt = self.parse_select()
+        if self.at_kw("select", "with"):
+            stmt = self.parse_query()
         elif self.at_kw("create"):
             stmt = self.parse_create()
         elif self.at_kw("drop"):
@@ -241,6 +241,37 @@ class Parser:
             raise InvalidSyntax(f"unsupported statement start: {self.peek()}")
         return stmt
 
+    def parse_query(self):
+        """[WITH ctes] select [UNION/EXCEPT/INTERSECT select ...] — a
+        trailing ORDER BY/LIMIT binds to the whole set operation."""
+        ctes = []
+        if self.eat_kw("with"):
+            while True:
+                name = self.next().value
+                self.expect_kw("as")
+                self.expect_op("(")
+                sub = self.parse_query()
+                self.expect_op(")")
+                ctes.append((name, sub))
+                if not self.eat_op(","):
+                    break
+        node = self.parse_select()
+        while self.at_kw("union", "except", "intersect"):
+            op = self.next().value.lower()
+            all_rows = self.eat_kw("all")
+            self.eat_kw("distinct")
+            right = self.parse_select()
+            node = ast.SetOp(op, all_rows, node, right)
+        if isinstance(node, ast.SetOp):
+            rs = node.right
+            if isinstance(rs, ast.Select) and (rs.order_by or
+                                               rs.limit is not None):
+                node.order_by, node.limit = rs.order_by, rs.limit
+                rs.order_by, rs.limit = [], None
+        if ctes:
+            node.ctes = ctes
+        return node
+
     def parse_select(self) -> ast.Select:
         self.expect_kw("select")
         distinct = self.eat_kw("distinct")
@@ -256,7 +287,8 @@ class Parser:
                 elif self.peek() is not None and self.peek().kind == "id" and \
                         self.peek().value.lower() not in (
                             "from", "where", "group", "order", "limit", "having",
-                            "offset", "align", "fill", "range"):
+                            "offset", "align", "fill", "range",
+                            "union", "except", "intersect"):
                     alias = self.next().value
                 projections.append((e, alias))
             if not self.eat_op(","):
@@ -265,13 +297,19 @@ class Parser:
         table_alias = None
         joins = []
         if self.eat_kw("from"):
-            table = self.next().value
+            if self.at_op("("):
+                self.next()
+                table = self.parse_query()   # derived table
+                self.expect_op(")")
+            else:
+                table = self.next().value
             if self.eat_kw("as"):
                 table_alias = self.next().value
             elif self.peek() is not None and self.peek().kind == "id" and \
                     self.peek().value.lower() not in (
                         "where", "group", "order", "limit", "having", "offset",
-                        "join", "inner", "left", "on", "align"):
+                        "join", "inner", "left", "on", "align",
+                        "union", "except", "intersect"):
                 table_alias = self.next().value
             while self.at_kw("join", "inner", "left"):
                 kind = "inner"
@@ -392,6 +430,23 @@ class Parser:
             self.i = len(self.toks)
             return ast.CreateFlow(name, sink, sql, if_not_exists,
                                   expire_after_s=expire_after_s)
+        if self.at_kw("view") or (self.at_kw("or") and "view" in self.sql.lower()):
+            or_replace = False
+            if self.eat_kw("or"):
+                self.expect_kw("replace")
+                or_replace = True
+            self.expect_kw("view")
+            if_not_exists = False
+            if self.eat_kw("if"):
+                self.expect_kw("not"); self.expect_kw("exists")
+                if_not_exists = True
+            name = self.next().value
+            self.expect_kw("as")
+            if self.peek() is None:
+                raise InvalidSyntax("CREATE VIEW ... AS <select> expected")
+            sql = self.sql[self.peek().pos:].rstrip().rstrip(";")
+            self.parse_query()   # validate the body
+            return ast.CreateView(name, sql, or_replace, if_not_exists)
         external = self.eat_kw("external")
         self.expect_kw("table")
         if_not_exists = False
@@ -520,6 +575,12 @@ class Parser:
         self.expect_kw("drop")
         if self.eat_kw("flow"):
             return ast.DropFlow(self.next().value)
+        if self.eat_kw("view"):
+            if_exists = False
+            if self.eat_kw("if"):
+                self.expect_kw("exists")
+                if_exists = True
+            return ast.DropView(self.next().value, if_exists)
         self.expect_kw("table")
         if_exists = False
         if self.eat_kw("if"):

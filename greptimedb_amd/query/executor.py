@@ -2712,6 +2712,13 @@ _OBJ_FUNCS = {
                                 for v in _str_col(a[0])], dtype=object),
     "length": lambda a: np.array([np.nan if v is None else float(len(str(v)))
                                   for v in _str_col(a[0])]),
+    # substr(s, start_1based[, len]) — SQL semantics (reference DataFusion)
+    "substr": lambda a: np.array(
+        [None if v is None else
+         str(v)[max(int(np.atleast_1d(a[1])[0]) - 1, 0):
+                (max(int(np.atleast_1d(a[1])[0]) - 1, 0) +
+                 int(np.atleast_1d(a[2])[0])) if len(a) > 2 else None]
+         for v in _str_col(a[0])], dtype=object),
     "char_length": lambda a: _OBJ_FUNCS["length"](a),
     "replace": lambda a: np.array(
         [None if v is None else str(v).replace(str(np.atleast_1d(a[1])[0]),

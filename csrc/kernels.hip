@@ -301,7 +301,7 @@ enum PromMode {
   PM_AVG = 4, PM_SUM = 5, PM_MIN = 6, PM_MAX = 7, PM_COUNT = 8, PM_LAST = 9,
   PM_IDELTA = 10, PM_IRATE = 11, PM_DERIV = 12, PM_PREDICT = 13,
   PM_RESETS = 14, PM_CHANGES = 15, PM_STDDEV = 16, PM_STDVAR = 17,
-  PM_ABSENT_OT = 18, PM_QUANTILE_OT = 19,
+  PM_ABSENT_OT = 18, PM_QUANTILE_OT = 19, PM_LAST_TS = 20,
 };
 
 DEV_INLINE double prom_nan() { return __longlong_as_double(0x7FF8000000000000LL); }
@@ -348,6 +348,10 @@ __global__ void prom_range_eval_kernel(
       if (cnt > 0) r = vals[w_hi - 1];
     } else if (mode == PM_LAST) {
       if (cnt > 0) r = vals[w_hi - 1];
+    } else if (mode == PM_LAST_TS) {
+      // last sample timestamp as f64 (exact to 2^53 ms) — the cross-rank
+      // last_value merge argmaxes on this plane
+      if (cnt > 0) r = (double)ts[w_hi - 1];
     } else if (mode == PM_COUNT) {
       if (cnt > 0) r = (double)cnt;
     } else if (mode == PM_ABSENT_OT) {

@@ -114,7 +114,7 @@ PROM_MODES = {
     "max_over_time": 7, "count_over_time": 8, "last_over_time": 9,
     "idelta": 10, "irate": 11, "deriv": 12, "predict_linear": 13,
     "resets": 14, "changes": 15, "stddev_over_time": 16, "stdvar_over_time": 17,
-    "absent_over_time": 18, "quantile_over_time": 19,
+    "absent_over_time": 18, "quantile_over_time": 19, "last_ts": 20,
 }
 
 
@@ -146,6 +146,9 @@ def prom_range_eval(ts, vals, seg_lo, seg_hi, T, t0, step_ms, range_ms,
             if mode in (0, 9):
                 if cnt:
                     r = w[-1]
+            elif mode == 20:       # last sample ts (dist last_value merge)
+                if cnt:
+                    r = float(wt[-1])
             elif mode == 8:
                 if cnt:
                     r = float(cnt)

@@ -214,3 +214,35 @@ class DistContext:
         dist.all_gather_object(gathered, {k: np.asarray(v) for k, v in col_data.items()})
         keys = col_data.keys()
         return {k: np.concatenate([g[k] for g in gathered]) for k in keys}
+
+
+    def argmax_combine(self, keys: list, order_plane, value_plane):
+        """Cross-rank 'newest wins' merge (RANGE last_value): the value
+        whose order key (sample ts) is globally maximal is kept per cell.
+        Two reductions: MAX on the order plane, then MAX over values masked
+        to cells where the local order equals the global winner (a series
+        lives on one rank, so ties are degenerate)."""
+        all_keys: list = [None] * self.world
+        dist.all_gather_object(all_keys, keys)
+        merged = sorted({k for ks in all_keys for k in ks})
+        gmap = {k: i for i, k in enumerate(merged)}
+        G = max(len(merged), 1)
+        T = order_plane.shape[1] if order_plane is not None and \
+            order_plane.ndim == 2 else 1
+        dev = self.coll_device
+        neg = float("-inf")
+        g_ord = torch.full((G, T), neg, dtype=torch.float64, device=dev)
+        l_ord = torch.full((G, T), neg, dtype=torch.float64, device=dev)
+        l_val = torch.full((G, T), neg, dtype=torch.float64, device=dev)
+        if keys:
+            idx = torch.as_tensor([gmap[k] for k in keys], device=dev)
+            l_ord[idx] = torch.nan_to_num(order_plane.to(dev), nan=neg)
+            l_val[idx] = torch.nan_to_num(value_plane.to(dev), nan=neg)
+        g_ord.copy_(l_ord)
+        dist.all_reduce(g_ord, op=dist.ReduceOp.MAX)
+        win = (l_ord == g_ord) & torch.isfinite(g_ord)
+        cand = torch.where(win, l_val, torch.full_like(l_val, neg))
+        dist.all_reduce(cand, op=dist.ReduceOp.MAX)
+        out = torch.where(torch.isfinite(g_ord), cand,
+                          torch.full_like(cand, float("nan")))
+        return merged, out

@@ -1924,7 +1924,7 @@ class Executor:
     # ------------------------------------------------------ RANGE queries
 
     _RANGE_MODES = {"min": 6, "max": 7, "sum": 5, "avg": 4, "mean": 4,
-                    "count": 8, "last_value": 9}
+                    "count": 8, "last_value": 9, "__last_ts": 20}
 
     def _exec_range_select(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
         """SQL RANGE query: `agg(x) RANGE '10s' … ALIGN '5s' [TO …] [BY (…)]
@@ -2083,11 +2083,17 @@ class Executor:
                                                  key=lambda kv: kv[1])]
             prim_specs: list = []       # (plane tensor [G, T], reduce op)
             call_prims: list = []       # per call: (kind, idx...) into outs
+            last_pairs: list = []       # (ts_plane, val_plane) per last_value
             for nd, fname, arg in calls:
-                if fname == "last_value":
-                    raise PlanQuery("distributed RANGE last_value "
-                                    "not yet supported")
                 G = len(keys_sorted)
+                if fname == "last_value":
+                    # argmax-by-timestamp merge: the newest sample across
+                    # ranks wins (reference: final last_value at frontend)
+                    ts_p = kernel_plane("__last_ts", nd.range_ms, arg)[:G]
+                    v_p = kernel_plane("last_value", nd.range_ms, arg)[:G]
+                    call_prims.append(("last", len(last_pairs)))
+                    last_pairs.append((ts_p, v_p))
+                    continue
                 if fname in ("sum", "avg", "count"):
                     s_p = torch.nan_to_num(
                         kernel_plane("sum", nd.range_ms, arg), nan=0.0)[:G]
@@ -2106,6 +2112,15 @@ class Executor:
             group_keys = {k: i for i, k in enumerate(merged_keys)}
             n_slots = max(len(group_keys), 1)
             for (nd, fname, arg), prim in zip(calls, call_prims):
+                if prim[0] == "last":
+                    ts_p, v_p = last_pairs[prim[1]]
+                    _mk, plane_t = dist_ctx.argmax_combine(keys_sorted,
+                                                           ts_p, v_p)
+                    plane = plane_t.cpu().numpy()
+                    if plane.shape[0] == 0:
+                        plane = np.full((1, T), np.nan)
+                    planes[id(nd)] = plane
+                    continue
                 if prim[0] in ("sum", "avg", "count"):
                     s = outs[prim[1]].cpu().numpy()
                     c = outs[prim[2]].cpu().numpy()

@@ -736,8 +736,6 @@ class PromEvaluator:
         if op == "count_values":
             # group additionally by the sample VALUE at each step; output
             # series = (group labels + {param: value}) (Prometheus semantics)
-            if self.dist is not None:
-                raise PlanQuery("distributed count_values not yet supported")
             label_name = e.param.value if e.param is not None else "value"
             vals_h = v.cpu().numpy()
             out_rows: dict = {}   # (g, value_repr) -> row index
@@ -762,12 +760,44 @@ class PromEvaluator:
                 l = dict(out_labels[g]) if out_labels else {}
                 l[str(label_name)] = f"{val:g}"
                 labels2.append(l)
+            if self.dist is not None:
+                # gather per-rank count matrices, sum rows with equal
+                # labels (counts are partials of disjoint series shards)
+                keys = [tuple(sorted(l.items())) for l in labels2]
+                keys_all, mat = self.dist.gather_matrix(keys, outm)
+                uniq = sorted(set(keys_all))
+                kmap = {k: i for i, k in enumerate(uniq)}
+                m2 = torch.zeros((len(uniq), T), dtype=torch.float64)
+                seen = torch.zeros((len(uniq), T), dtype=torch.bool)
+                mat = torch.nan_to_num(mat.double(), nan=0.0)
+                has = mat > 0
+                for i, k in enumerate(keys_all):
+                    m2[kmap[k]] += mat[i]
+                    seen[kmap[k]] |= has[i]
+                m2 = torch.where(seen, m2, torch.full_like(m2, float("nan")))
+                labels2 = [dict(k) for k in uniq]
+                outm = m2
             return PromMatrix(labels2, outm.to(dev), grid)
 
         if op == "quantile":
-            if self.dist is not None:
-                raise PlanQuery("distributed quantile aggregation not yet supported")
             q = e.param.value if isinstance(e.param, ast.NumberLit) else 0.5
+            if self.dist is not None:
+                # quantile is not mergeable from partials: gather member
+                # series values per group (reference ships all series to
+                # the frontend for the final quantile the same way)
+                keys = [tuple(sorted(out_labels[int(g)].items()))
+                        for g in gidx] if S else []
+                keys_all, v_all = self.dist.gather_matrix(keys, v)
+                uniq = sorted(set(keys_all))
+                kmap = {k: i for i, k in enumerate(uniq)}
+                gidx = np.array([kmap[k] for k in keys_all], dtype=np.int64)
+                v = v_all.double().to(dev)
+                out_labels = [dict(k) for k in uniq]
+                G = max(len(out_labels), 1)
+                cnt = torch.zeros((G, T), dtype=torch.float64, device=dev)
+                if len(gidx):
+                    cnt.index_add_(0, torch.as_tensor(gidx, device=dev),
+                                   (~torch.isnan(v)).double())
             out = torch.full((G, T), float("nan"), dtype=torch.float64, device=dev)
             for g in range(G):
                 rows = [i for i, x in enumerate(gidx) if x == g]

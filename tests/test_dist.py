@@ -53,6 +53,16 @@ r6 = ex.execute("SELECT ts, hostname, avg(usage_user) RANGE '1m' AS a, "
                 "max(usage_user) RANGE '1m' AS mx, count(usage_user) RANGE '1m' AS c "
                 "FROM cpu ALIGN '30s' ORDER BY hostname, ts LIMIT 10000")
 range_hosts = sorted(set(r6.columns[1]))
+# distributed RANGE last_value (argmax-by-ts merge over gloo)
+r7 = ex.execute("SELECT ts, hostname, last_value(usage_user) RANGE '1m' AS lv "
+                "FROM cpu ALIGN '30s' ORDER BY hostname, ts LIMIT 10000")
+# distributed quantile + count_values aggregations
+mq = ev.query_range('quantile(0.5, {__field__="usage_user", __name__="cpu"})',
+                    1451606450, 1451606450, 1)
+mcv = ev.query_range('count_values("band", '
+                     'floor({__field__="usage_user", __name__="cpu"} / 25))',
+                     1451606450, 1451606450, 1)
+cv_total = float(np.nansum(mcv.values.cpu().numpy()))
 out = {
     "range_rows": len(r6),
     "range_hosts": len(range_hosts),
@@ -65,6 +75,12 @@ out = {
     "raw": len(r4),
     "lastpoint_hosts": len(r5),
     "prom_sum": round(float(m.values[0][-1]), 6),
+    "range_lv_rows": len(r7),
+    "range_lv_hosts": len(sorted(set(r7.columns[1]))),
+    "range_lv_sum": round(float(np.nansum(np.asarray(r7.columns[2], dtype=float))), 4),
+    "quantile": round(float(mq.values[0][-1]), 6),
+    "cv_series": mcv.S,
+    "cv_total": cv_total,
 }
 print("RESULT" + str(rank) + json.dumps(out))
 dist.destroy_process_group()
@@ -107,3 +123,8 @@ def test_two_rank_query_combine(tmp_path):
     assert results[0]["range_hosts"] == 20   # all ranks' hosts in the plane
     assert results[0]["range_rows"] > 40
     assert results[0]["range_sum_a"] != 0.0
+    assert results[0]["range_lv_hosts"] == 20
+    assert results[0]["range_lv_sum"] != 0.0
+    assert 0.0 <= results[0]["quantile"] <= 100.0
+    assert results[0]["cv_total"] == 20.0    # one sample per series
+    assert results[0]["cv_series"] >= 1

@@ -65,6 +65,10 @@ class MitoEngine:
         # deterministic background-task observation)
         self.write_listeners: list = []   # callback(table_name, min_ts, max_ts, n)
         self.flush_listeners: list = []   # callback(table_name, region_id)
+        # insert mirroring (reference operator/src/insert.rs:1384
+        # FlowMirrorTask): callbacks get the parsed columnar batch
+        # callback(table_state, region, codes, ts_ms, fields, field_names)
+        self.mirror_listeners: list = []
         from greptimedb_amd.utils.memquota import MemoryQuota
         self.scan_quota = MemoryQuota(config.scan_mem_bytes)
         self._load_catalog()
@@ -183,6 +187,8 @@ class MitoEngine:
                                    region.field_names, new_series, str_fields)
             seq = self.wal.append(region.region_id, payload)
         region.append(series_codes, ts_ms, fields, seq, str_fields)
+        for cb in self.mirror_listeners:
+            cb(table, region, series_codes, ts_ms, fields, region.field_names)
         if self.write_listeners and len(ts_ms):
             lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
             for cb in self.write_listeners:
@@ -273,6 +279,14 @@ class MitoEngine:
         finally:
             for r in reversed(ordered):
                 r.lock.release()
+        if self.mirror_listeners:
+            for k, ((st_k, _ri), s_, e_) in enumerate(zip(targets, starts, ends)):
+                rows = order[s_:e_]
+                if len(rows) == 0:
+                    continue
+                for cb in self.mirror_listeners:
+                    cb(st_k, regions[k], series_codes[rows], ts_ms[rows],
+                       fields[:, rows], st_k.regions[0].field_names)
         if self.write_listeners:
             lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
             names = {st.schema.name for st, _ in targets}
@@ -345,6 +359,15 @@ class MitoEngine:
         finally:
             for r in reversed(ordered):
                 r.lock.release()
+        if self.mirror_listeners and n:
+            fnames = table.regions[0].field_names
+            for k, region in enumerate(regions):
+                if int(counts[k]) == 0:
+                    continue
+                rows = np.flatnonzero(region_of == k)
+                for cb in self.mirror_listeners:
+                    cb(table, region, series_codes[rows], ts_ms[rows],
+                       fields[:, rows], fnames)
         if self.write_listeners and n:
             lo, hi = int(np.min(ts_ms)), int(np.max(ts_ms))
             for cb in self.write_listeners:

@@ -53,16 +53,30 @@ class FlowTask:
 
 
 class FlowEngine:
+    """Dual-mode flow engine (reference FlowDualEngine, src/flow/src/
+    engine.rs:15): flows whose plan shape supports incremental reduction
+    run STREAMING (flow/streaming.py — per-write mirror into operator
+    state, no source re-scan); everything else runs BATCHING (dirty-window
+    re-query below)."""
+
     def __init__(self, engine, executor):
         self.engine = engine
         self.executor = executor
         self.flows: dict[str, FlowTask] = {}
         engine.write_listeners.append(self._on_write)
+        engine.mirror_listeners.append(self._on_mirror)
 
     def _on_write(self, table: str, lo: int, hi: int, n: int):
         for f in self.flows.values():
-            if f.source == table:
+            if isinstance(f, FlowTask) and f.source == table:
                 f.mark_dirty(lo, hi)
+
+    def _on_mirror(self, st, region, codes, ts_ms, fields, field_names):
+        from greptimedb_amd.flow.streaming import StreamingFlowTask
+        name = st.schema.name
+        for f in self.flows.values():
+            if isinstance(f, StreamingFlowTask) and f.plan.source == name:
+                f.on_write(st, region, codes, ts_ms, fields, field_names)
 
     def create_flow(self, name: str, sink: str, select_sql: str,
                     if_not_exists: bool = False,
@@ -71,17 +85,37 @@ class FlowEngine:
             if if_not_exists:
                 return
             raise InvalidArguments(f"flow {name} exists")
-        self.flows[name] = FlowTask(name, sink, select_sql, expire_after_s)
+        from greptimedb_amd.flow.streaming import StreamingFlowTask
+        try:
+            task = StreamingFlowTask(name, sink, select_sql, self.engine,
+                                     expire_after_s=expire_after_s)
+        except Exception:
+            # shape not incrementally reducible (or source not created
+            # yet) → batching mode, like the reference's dual-engine route
+            task = FlowTask(name, sink, select_sql, expire_after_s)
+        self.flows[name] = task
+
+    def mode_of(self, name: str) -> str:
+        f = self.flows.get(name)
+        return "batching" if isinstance(f, FlowTask) else "streaming"
 
     def drop_flow(self, name: str):
         self.flows.pop(name, None)
 
     def tick(self) -> dict[str, int]:
-        """Re-evaluate dirty windows for every flow; returns rows upserted
-        per flow (reference: batching-mode task tick)."""
+        """Streaming flows flush their incremental state; batching flows
+        re-evaluate dirty windows (reference: batching-mode task tick)."""
         out = {}
         import time as _time
+        from greptimedb_amd.flow.streaming import StreamingFlowTask
         for f in self.flows.values():
+            if isinstance(f, StreamingFlowTask):
+                r = f.flush()
+                if r:
+                    out[f.name] = r
+        for f in self.flows.values():
+            if isinstance(f, StreamingFlowTask):
+                continue
             lo, hi = f.take_dirty()
             if lo is None:
                 continue

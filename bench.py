@@ -132,10 +132,14 @@ def main():
     exchange = None
     if world > 1:
         from greptimedb_amd.parallel.write_fanout import WriteExchange
-        serv_ing = Ingestor(eng, default_regions=4, append_mode=True,
-                            durable=(args.wal != "off"), rank=rank, world=world)
-        exchange = WriteExchange(rank, world, handler=serv_ing.handle_remote)
-        serv_ing.exchange = exchange
+
+        def _mk_handler():
+            # fresh receive pipeline per peer connection (parallel applies)
+            ing = Ingestor(eng, default_regions=4, append_mode=True,
+                           durable=(args.wal != "off"), rank=rank, world=world)
+            return ing.handle_remote
+
+        exchange = WriteExchange(rank, world, handler_factory=_mk_handler)
         import torch.distributed as dist
         dist.barrier()  # every rank's exchange is listening
 

@@ -370,6 +370,24 @@ class Ingestor:
             bounds = np.flatnonzero(np.diff(region_sorted)) + 1
             starts = np.concatenate(([0], bounds))
             ends = np.concatenate((bounds, [n]))
+            targets = [self.flat_regions[int(region_sorted[s])] for s in starts]
+            # K16 bulk path (GPU): single-table batches go through ONE
+            # scatter_append launch instead of per-region copy chains
+            if self._bulk and len({id(st) for st, _ri in targets}) == 1:
+                st0 = targets[0][0]
+                fmap = self._field_map(st0, field_names, fields_mat,
+                                       np.arange(n))
+                out = np.empty((len(fmap), n), dtype=np.float64)
+                for i, src in enumerate(fmap):
+                    out[i] = fields_mat[src] if src >= 0 else np.nan
+                if engine.write_regions_bulk(
+                        targets, local.astype(np.int32), ts_ms, out,
+                        order, starts, ends, durable=self.durable):
+                    if self.durable:
+                        engine.commit_wal()
+                    engine.maybe_flush()
+                    self.rows_ingested += n
+                    return b"OK"
             for s, e in zip(starts, ends):
                 flat = int(region_sorted[s])
                 st, region_idx = self.flat_regions[flat]

@@ -64,13 +64,17 @@ class WriteExchange:
 
     def __init__(self, rank: int, world: int, handler=None,
                  base_port: int | None = None, host: str = "127.0.0.1",
-                 ports: list[int] | None = None):
+                 ports: list[int] | None = None, handler_factory=None):
         self.rank = rank
         self.world = world
         self.host = host
         self.ports = list(ports) if ports is not None else \
             [fanout_port(r, base_port) for r in range(world)]
         self.handler = handler      # callable(payload: bytes) -> bytes
+        # handler_factory() -> fresh handler per connection: each peer
+        # connection gets its own receive pipeline (no shared lock), like
+        # the reference's per-connection tonic service instances
+        self.handler_factory = handler_factory
         self._conns: dict[int, socket.socket] = {}
         self._conn_locks = {r: threading.Lock() for r in range(world)}
         self._closing = False
@@ -94,11 +98,12 @@ class WriteExchange:
                              daemon=True).start()
 
     def _serve_conn(self, conn: socket.socket):
+        handler = self.handler_factory() if self.handler_factory else self.handler
         try:
             while not self._closing:
                 req = _read_frame(conn)
                 try:
-                    resp = self.handler(req) if self.handler else b"OK"
+                    resp = handler(req) if handler else b"OK"
                 except Exception as e:  # report the error to the sender
                     resp = b"ERR " + repr(e).encode()
                 _write_frame(conn, resp or b"OK")

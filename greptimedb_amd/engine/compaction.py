@@ -132,7 +132,8 @@ class Compactor:
         path = os.path.join(region.dir, "sst", f"{fid}.parquet")
         meta = sst_mod.write_sst(path, region.schema, region.series.pks,
                                  se_h, ts_h, f_h, seq_h, fnames,
-                                 str_cols=str_cols_sorted)
+                                 str_cols=str_cols_sorted,
+                                 region_id=region.region_id)
         meta.level = 1  # LEVEL_COMPACTED
         region.manifest.commit({
             "kind": "edit",

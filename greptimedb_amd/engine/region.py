@@ -282,7 +282,8 @@ class Region:
         path = os.path.join(self.dir, "sst", f"{fid}.parquet")
         meta = sst_mod.write_sst(path, self.schema, self.series.pks,
                                  se_h, ts_h, f_h, seq_h, flush_field_names,
-                                 str_cols=str_cols_sorted)
+                                 str_cols=str_cols_sorted,
+                                 region_id=self.region_id)
         if text_index:
             from greptimedb_amd.engine import ftindex
             ftindex.save_sidecar(path, text_index, self.text_cols)

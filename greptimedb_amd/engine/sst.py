@@ -30,6 +30,81 @@ from greptimedb_amd.models.schema import TableSchema
 ROW_GROUP_SIZE = 102400
 OP_PUT = 1
 
+# mito2 stores the region metadata JSON in the parquet key-value metadata
+# under this key (reference sst/parquet.rs:43 PARQUET_METADATA_KEY)
+PARQUET_METADATA_KEY = b"greptime:metadata"
+
+
+def _dt_json(c) -> dict:
+    """ConcreteDataType serde shape (reference src/datatypes types serde,
+    e.g. manifest/action.rs:663 examples)."""
+    from greptimedb_amd.models.schema import DataType
+    m = {
+        DataType.STRING: {"String": {"size_type": "Utf8"}},
+        DataType.BINARY: {"Binary": None},
+        DataType.JSON: {"String": {"size_type": "Utf8"}},
+        DataType.VECTOR: {"Binary": None},
+        DataType.FLOAT64: {"Float64": {}},
+        DataType.FLOAT32: {"Float32": {}},
+        DataType.INT64: {"Int64": {}},
+        DataType.INT32: {"Int32": {}},
+        DataType.UINT64: {"UInt64": {}},
+        DataType.BOOL: {"Boolean": None},
+        DataType.TIMESTAMP_MS: {"Timestamp": {"Millisecond": None}},
+        DataType.TIMESTAMP_NS: {"Timestamp": {"Nanosecond": None}},
+    }
+    return m.get(c.dtype, {"Float64": {}})
+
+
+def region_metadata_json(schema: TableSchema, region_id: int,
+                         extra_fields: list[str] | None = None,
+                         str_fields: list[str] | None = None) -> str:
+    """RegionMetadata JSON exactly as the reference serializes it
+    (store-api/src/metadata.rs serde; examples in manifest/action.rs:663)."""
+    import json
+    from greptimedb_amd.models.schema import SemanticType
+    sem_name = {SemanticType.TAG: "Tag", SemanticType.FIELD: "Field",
+                SemanticType.TIMESTAMP: "Timestamp"}
+    cols = []
+    next_id = 0
+    name_to_id = {}
+    for c in schema.columns:
+        cols.append({
+            "column_schema": {
+                "name": c.name,
+                "data_type": _dt_json(c),
+                "is_nullable": bool(c.nullable) and
+                               c.semantic != SemanticType.TIMESTAMP,
+                "is_time_index": c.semantic == SemanticType.TIMESTAMP,
+                "default_constraint": None,
+                "metadata": {},
+            },
+            "semantic_type": sem_name[c.semantic],
+            "column_id": c.column_id,
+        })
+        name_to_id[c.name] = c.column_id
+        next_id = max(next_id, c.column_id + 1)
+    for name in (extra_fields or []):
+        if name in name_to_id:
+            continue
+        dt = {"String": {"size_type": "Utf8"}} if name in (str_fields or []) \
+            else {"Float64": {}}
+        cols.append({
+            "column_schema": {"name": name, "data_type": dt,
+                              "is_nullable": True, "is_time_index": False,
+                              "default_constraint": None, "metadata": {}},
+            "semantic_type": "Field",
+            "column_id": next_id,
+        })
+        next_id += 1
+    return json.dumps({
+        "column_metadatas": cols,
+        "primary_key": [name_to_id[t] for t in schema.primary_key],
+        "region_id": region_id,
+        "schema_version": 0,
+        "primary_key_encoding": "dense",
+    })
+
 
 @dataclass
 class SstMeta:
@@ -73,9 +148,13 @@ class SstBatch:
 def write_sst(path: str, schema: TableSchema, pks: list[bytes],
               series: np.ndarray, ts_ms: np.ndarray, fields: np.ndarray,
               seq: np.ndarray, field_names: list[str],
-              str_cols: dict[str, np.ndarray] | None = None) -> SstMeta:
+              str_cols: dict[str, np.ndarray] | None = None,
+              region_id: int = 0, flat: bool = False) -> SstMeta:
     """Write one mito2-format parquet SST. Inputs are host arrays sorted by
-    (series, ts); `pks[code]` gives the encoded primary key per local code."""
+    (series, ts); `pks[code]` gives the encoded primary key per local code.
+    flat=True additionally stores decoded tag columns up front (reference
+    sst/parquet/flat_format.rs: `pk cols..., fields..., time index,
+    __primary_key, __sequence, __op_type`)."""
     n = len(ts_ms)
     # compact dictionary: unique codes in appearance order
     uniq, inv = np.unique(series, return_inverse=True)
@@ -83,6 +162,15 @@ def write_sst(path: str, schema: TableSchema, pks: list[bytes],
     pk_col = pa.DictionaryArray.from_arrays(pa.array(inv.astype(np.uint32), type=pa.uint32()), dict_values)
 
     cols, names = [], []
+    if flat and schema.primary_key:
+        from greptimedb_amd.engine import pk_codec
+        tag_names = schema.primary_key
+        decoded = [pk_codec.decode_pk(pks[int(c)], len(tag_names))
+                   for c in uniq]
+        for ti, tn in enumerate(tag_names):
+            vals = np.array([decoded[int(i)][ti] for i in inv], dtype=object)
+            cols.append(pa.array(list(vals), type=pa.string()))
+            names.append(tn)
     for i, fn in enumerate(field_names):
         cols.append(pa.array(fields[i], type=pa.float64()))
         names.append(fn)
@@ -101,6 +189,10 @@ def write_sst(path: str, schema: TableSchema, pks: list[bytes],
     names.append("__op_type")
 
     table = pa.Table.from_arrays(cols, names=names)
+    meta_json = region_metadata_json(
+        schema, region_id, extra_fields=field_names + list(str_cols or {}),
+        str_fields=list(str_cols or {}))
+    table = table.replace_schema_metadata({PARQUET_METADATA_KEY: meta_json})
     pq.write_table(table, path, row_group_size=ROW_GROUP_SIZE, compression="zstd")
     return SstMeta(
         file_id=os.path.basename(path).replace(".parquet", ""),
@@ -135,7 +227,10 @@ def read_sst(path: str, schema: TableSchema, field_names: list[str]):
         for fn in field_names
     ]) if field_names else np.zeros((0, len(ts)))
     seq = t.column("__sequence").to_numpy(zero_copy_only=False).astype(np.int64)
-    internal = {schema.time_index.name, "__primary_key", "__sequence", "__op_type"}
+    # flat-format files (sst/parquet/flat_format.rs) carry raw tag columns
+    # too — tags are reconstructed from __primary_key, so skip them here
+    internal = {schema.time_index.name, "__primary_key", "__sequence",
+                "__op_type"} | set(schema.primary_key)
     str_cols = {}
     for cn in t.column_names:
         if cn in internal or cn in field_names:
@@ -145,3 +240,14 @@ def read_sst(path: str, schema: TableSchema, field_names: list[str]):
                 pa.types.is_binary(ftype) or pa.types.is_large_binary(ftype):
             str_cols[cn] = t.column(cn).to_numpy(zero_copy_only=False)
     return dict_values, indices, ts, fields, seq, str_cols
+
+
+def read_region_metadata(path: str) -> dict | None:
+    """Parse the `greptime:metadata` key-value entry (reference:
+    sst/parquet/metadata.rs reads the same key)."""
+    import json
+    md = pq.read_metadata(path).metadata or {}
+    raw = md.get(PARQUET_METADATA_KEY)
+    if raw is None:
+        return None
+    return json.loads(raw.decode() if isinstance(raw, bytes) else raw)

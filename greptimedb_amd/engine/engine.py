@@ -417,6 +417,37 @@ class MitoEngine:
                 import traceback
                 traceback.print_exc()
 
+    def gc_orphan_ssts(self, grace_s: float = 3600.0) -> int:
+        """Delete SST/sidecar files no region manifest references
+        (reference src/mito2/src/gc.rs orphan scan; metasrv global GC).
+        Files younger than `grace_s` are spared — they may belong to an
+        in-flight flush whose manifest edit has not landed yet."""
+        import time as _time
+        removed = 0
+        now = _time.time()
+        for st in self.tables.values():
+            for r in st.regions:
+                sdir = os.path.join(r.dir, "sst")
+                if not os.path.isdir(sdir):
+                    continue
+                with r.lock:
+                    known = set(r.manifest.files)
+                for fn in os.listdir(sdir):
+                    if not (fn.endswith(".parquet") or fn.endswith(".ftidx")):
+                        continue
+                    fid = fn.rsplit(".", 1)[0]
+                    if fid in known:
+                        continue
+                    p = os.path.join(sdir, fn)
+                    try:
+                        if now - os.path.getmtime(p) < grace_s:
+                            continue
+                        os.unlink(p)
+                        removed += 1
+                    except OSError:
+                        pass
+        return removed
+
     def _purge_wal(self):
         """Purge WAL segments below every region's replay point. A region
         only constrains purging while it has unflushed WAL entries

@@ -496,6 +496,11 @@ class Executor:
             c = Compactor(trigger_file_num=2)
             n = sum(c.compact_region(r) for r in st.regions)
             return QueryResult(["result"], [[n]])
+        if f == "gc":
+            # orphan SST scan (reference src/mito2/src/gc.rs + metasrv gc.rs)
+            grace = float(a.args[0]) if a.args else 3600.0
+            n = self.engine.gc_orphan_ssts(grace_s=grace)
+            return QueryResult(["result"], [[n]])
         if f == "flush_all":
             self.engine.flush_all()
             return QueryResult(["result"], [[1]])
@@ -2487,11 +2492,24 @@ class Executor:
                     return _torch_cmp(x.op, l, r)
                 if x.op == "like":
                     lv = value(x.left)
+                    import fnmatch
+                    raw = str(value(x.right))
+                    # honor \% and \_ escapes (log-query Contains quoting)
+                    pat = (raw.replace(r"\%", "\x00").replace(r"\_", "\x01")
+                              .replace("%", "*").replace("_", "?")
+                              .replace("\x00", "%").replace("\x01", "_"))
                     if isinstance(lv, tuple) and lv[0] == "__tag__":
-                        import fnmatch
-                        pat = str(value(x.right)).replace("%", "*").replace("_", "?")
-                        return tag_mask(lv[1], lambda v: v is not None and fnmatch.fnmatch(v, pat))
-                    raise PlanQuery("LIKE only on tags")
+                        return tag_mask(lv[1], lambda v: v is not None and
+                                        fnmatch.fnmatch(v, pat))
+                    if isinstance(lv, tuple) and lv[0] == "__str__":
+                        col = src.str_cols.get(lv[1])
+                        if col is None:
+                            return torch.zeros(n, dtype=torch.bool, device=device)
+                        res = np.array([v is not None and
+                                        fnmatch.fnmatch(str(v), pat)
+                                        for v in col], dtype=bool)
+                        return torch.as_tensor(res, device=device)
+                    raise PlanQuery("LIKE only on string columns")
             if isinstance(x, ast.UnaryOp) and x.op == "not":
                 return ~ev(x.operand)
             if isinstance(x, ast.InList):

@@ -158,6 +158,68 @@ def build_app(ctx: ServerContext) -> FastAPI:
                 "device": ctx.engine.config.device,
                 "slow_queries": list(getattr(ctx.engine, "slow_queries", []))}
 
+    # ---------------- log query DSL (reference src/log-query) ----------
+    @app.post("/v1/logs")
+    async def logs_query(request: Request):
+        import json as _json
+        from greptimedb_amd.query.logquery import logquery_to_sql
+        body = _json.loads((await request.body()) or b"{}")
+        sql = logquery_to_sql(body)
+        r = ctx.executor.execute(sql)
+        out = _records_json(r)
+        out["sql"] = sql
+        return out
+
+    # ---------------- pprof-style profiling (reference http.rs:1079) ----
+    @app.get("/debug/prof/cpu")
+    async def prof_cpu(seconds: int = Query(2), frequency: int = Query(99)):
+        """Sampling CPU profile over `seconds`; returns folded stacks
+        (flamegraph.pl / speedscope compatible) — the reference serves
+        pprof flamegraphs the same way (servers/src/http.rs:1079-1094)."""
+        import asyncio
+        import collections
+        import sys
+        import threading as _th
+        folded: collections.Counter = collections.Counter()
+        stop = time.time() + min(max(seconds, 1), 60)
+        me = _th.get_ident()
+
+        def sample_loop():
+            interval = 1.0 / max(min(frequency, 1000), 1)
+            while time.time() < stop:
+                for tid, frame in sys._current_frames().items():
+                    if tid == me:
+                        continue
+                    stack = []
+                    f = frame
+                    while f is not None:
+                        stack.append(f"{f.f_code.co_name} "
+                                     f"({f.f_code.co_filename.rsplit('/', 1)[-1]}"
+                                     f":{f.f_lineno})")
+                        f = f.f_back
+                    folded[";".join(reversed(stack))] += 1
+                time.sleep(interval)
+
+        t = _th.Thread(target=sample_loop)
+        t.start()
+        while t.is_alive():
+            await asyncio.sleep(0.1)
+        body = "\n".join(f"{k} {v}" for k, v in folded.most_common())
+        from fastapi import Response
+        return Response(content=body, media_type="text/plain")
+
+    @app.get("/debug/prof/mem")
+    async def prof_mem(top: int = Query(25)):
+        """Heap snapshot via tracemalloc (reference: jemalloc heap prof)."""
+        import tracemalloc
+        if not tracemalloc.is_tracing():
+            tracemalloc.start()
+            return {"status": "tracing started; call again for a snapshot"}
+        snap = tracemalloc.take_snapshot()
+        stats = snap.statistics("lineno")[: max(top, 1)]
+        return {"top": [{"where": str(s.traceback), "size_kb": s.size // 1024,
+                         "count": s.count} for s in stats]}
+
     # ---------------- SQL ----------------
 
     async def _sql(request: Request, sql: str | None):

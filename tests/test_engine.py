@@ -582,3 +582,31 @@ def test_wal_sharded_concurrent_workers(tmp_path):
                                    wal_shards=4))
     assert _total(eng2) == 4000
     eng2.close()
+
+
+def test_gc_orphan_ssts(tmp_path):
+    """Orphan SST files (crashed flush leftovers) are collected after the
+    grace period (reference mito2 gc.rs)."""
+    import os as _os
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ing = Ingestor(eng)
+    w = CpuWorkload(scale=5)
+    ing.ingest_lines(w.next_batch(200))
+    eng.flush_all()
+    region = next(r for st in eng.tables.values() for r in st.regions
+                  if r.manifest.files)
+    sdir = _os.path.join(region.dir, "sst")
+    orphan = _os.path.join(sdir, "deadbeef00.parquet")
+    open(orphan, "wb").write(b"not a real parquet")
+    _os.utime(orphan, (0, 0))   # ancient mtime → past any grace period
+    fresh = _os.path.join(sdir, "cafebabe01.parquet")
+    open(fresh, "wb").write(b"in-flight flush artifact")
+    n = eng.gc_orphan_ssts(grace_s=60.0)
+    assert n == 1
+    assert not _os.path.exists(orphan)
+    assert _os.path.exists(fresh)          # grace period spares it
+    # manifest-referenced files untouched
+    for fid in region.manifest.files:
+        assert _os.path.exists(_os.path.join(sdir, f"{fid}.parquet"))
+    eng.close()

@@ -117,8 +117,15 @@ class MitoEngine:
     def create_table(self, schema: TableSchema, n_regions: int | None = None,
                      append_mode: bool = False, if_not_exists: bool = False) -> TableState:
         with self._ddl_lock:
-            return self._create_table_locked(schema, n_regions, append_mode,
-                                             if_not_exists)
+            existed = schema.name in self.tables
+            st = self._create_table_locked(schema, n_regions, append_mode,
+                                           if_not_exists)
+        if not existed:
+            from greptimedb_amd.utils.events import EVENTS_TABLE, recorder_of
+            if schema.name != EVENTS_TABLE:
+                recorder_of(self).record("create_table", {
+                    "table": schema.name, "regions": len(st.regions)})
+        return st
 
     def _create_table_locked(self, schema, n_regions, append_mode,
                              if_not_exists) -> TableState:
@@ -145,6 +152,9 @@ class MitoEngine:
         if st is None:
             raise TableNotFound(name)
         self._save_catalog()
+        from greptimedb_amd.utils.events import EVENTS_TABLE, recorder_of
+        if name != EVENTS_TABLE:
+            recorder_of(self).record("drop_table", {"table": name})
 
     def table(self, name: str) -> TableState:
         try:

@@ -69,3 +69,107 @@ def pg_md5_check(password: str, user: str, salt: bytes, response: str) -> bool:
     inner = hashlib.md5((password + user).encode()).hexdigest()
     expect = "md5" + hashlib.md5(inner.encode() + salt).hexdigest()
     return hmac.compare_digest(expect, response)
+
+
+# ---------------------------------------------------------------- SCRAM
+# SCRAM-SHA-256 (RFC 5802/7677) server side — reference: pgwire's SCRAM
+# support in src/servers/src/postgres (builder enables SCRAM auth).
+
+import base64 as _b64
+import hashlib as _hashlib
+import os as _os
+
+
+def _h(data: bytes) -> bytes:
+    return _hashlib.sha256(data).digest()
+
+
+def _hmac(key: bytes, msg: bytes) -> bytes:
+    return hmac.new(key, msg, _hashlib.sha256).digest()
+
+
+def _xor(a: bytes, b: bytes) -> bytes:
+    return bytes(x ^ y for x, y in zip(a, b))
+
+
+class ScramSha256Server:
+    """One SCRAM-SHA-256 exchange (server side).
+
+    usage:
+        s = ScramSha256Server(password)
+        server_first = s.server_first(client_first_bare_message)
+        server_final = s.verify_client_final(client_final_message)  # or None
+    """
+
+    ITERATIONS = 4096
+
+    def __init__(self, password: str):
+        self.salt = _os.urandom(16)
+        salted = _hashlib.pbkdf2_hmac("sha256", password.encode(), self.salt,
+                                      self.ITERATIONS)
+        self.stored_key = _h(_hmac(salted, b"Client Key"))
+        self.server_key = _hmac(salted, b"Server Key")
+        self.server_nonce = _b64.b64encode(_os.urandom(18)).decode()
+        self.client_first_bare = None
+        self.server_first_msg = None
+
+    @staticmethod
+    def parse_client_first(msg: str) -> tuple[str, str]:
+        """gs2-header,client-first-bare → (username, client_nonce)."""
+        # e.g. "n,,n=user,r=nonce"
+        parts = msg.split(",", 2)
+        bare = parts[2] if len(parts) >= 3 else msg
+        attrs = dict(kv.split("=", 1) for kv in bare.split(",") if "=" in kv)
+        return attrs.get("n", ""), attrs.get("r", "")
+
+    def server_first(self, client_first: str) -> str:
+        parts = client_first.split(",", 2)
+        self.client_first_bare = parts[2] if len(parts) >= 3 else client_first
+        _user, cnonce = self.parse_client_first(client_first)
+        self.full_nonce = cnonce + self.server_nonce
+        self.server_first_msg = (
+            f"r={self.full_nonce},s={_b64.b64encode(self.salt).decode()},"
+            f"i={self.ITERATIONS}")
+        return self.server_first_msg
+
+    def verify_client_final(self, client_final: str) -> str | None:
+        """Returns the server-final-message ('v=...') or None on failure."""
+        attrs = dict(kv.split("=", 1) for kv in client_final.split(",")
+                     if "=" in kv)
+        proof_b64 = attrs.get("p")
+        if proof_b64 is None or attrs.get("r") != self.full_nonce:
+            return None
+        without_proof = client_final[: client_final.rfind(",p=")]
+        auth_message = ",".join([self.client_first_bare,
+                                 self.server_first_msg, without_proof]).encode()
+        client_sig = _hmac(self.stored_key, auth_message)
+        client_key = _xor(_b64.b64decode(proof_b64), client_sig)
+        if not hmac.compare_digest(_h(client_key), self.stored_key):
+            return None
+        server_sig = _hmac(self.server_key, auth_message)
+        return "v=" + _b64.b64encode(server_sig).decode()
+
+
+def scram_client_messages(user: str, password: str, server_first_fn):
+    """Test/client helper: runs the client side of SCRAM-SHA-256.
+    `server_first_fn(client_first) -> server_first`; returns
+    (client_final, expected_server_sig_checker)."""
+    cnonce = _b64.b64encode(_os.urandom(18)).decode()
+    client_first_bare = f"n={user},r={cnonce}"
+    client_first = "n,," + client_first_bare
+    server_first = server_first_fn(client_first)
+    attrs = dict(kv.split("=", 1) for kv in server_first.split(",") if "=" in kv)
+    full_nonce, salt, iters = attrs["r"], _b64.b64decode(attrs["s"]), int(attrs["i"])
+    assert full_nonce.startswith(cnonce)
+    salted = _hashlib.pbkdf2_hmac("sha256", password.encode(), salt, iters)
+    client_key = _hmac(salted, b"Client Key")
+    stored_key = _h(client_key)
+    without_proof = f"c=biws,r={full_nonce}"
+    auth_message = ",".join([client_first_bare, server_first,
+                             without_proof]).encode()
+    client_sig = _hmac(stored_key, auth_message)
+    proof = _b64.b64encode(_xor(client_key, client_sig)).decode()
+    client_final = f"{without_proof},p={proof}"
+    server_key = _hmac(salted, b"Server Key")
+    expected_v = "v=" + _b64.b64encode(_hmac(server_key, auth_message)).decode()
+    return client_first, client_final, expected_v

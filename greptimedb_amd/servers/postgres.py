@@ -73,19 +73,54 @@ class PostgresServer:
                     params[parts[i].decode()] = parts[i + 1].decode()
             user = params.get("user", "")
             if self.user_provider is not None:
-                from greptimedb_amd.servers.auth import password_of, pg_md5_check
+                from greptimedb_amd.servers.auth import (ScramSha256Server,
+                                                         password_of,
+                                                         pg_md5_check)
                 ok = self.user_provider.allow(user)
                 stored = password_of(self.user_provider, user)
                 if ok and stored:
-                    # md5 challenge-response (AuthenticationMD5Password)
-                    salt = b"\x9a\x17\x2e\x41"
-                    writer.write(_msg(b"R", struct.pack("!I", 5) + salt))
+                    # SASL SCRAM-SHA-256 (RFC 7677; reference: pgwire SCRAM)
+                    writer.write(_msg(b"R", struct.pack("!I", 10) +
+                                      b"SCRAM-SHA-256\x00\x00"))
                     await writer.drain()
                     tag = await reader.readexactly(1)
                     (ln,) = struct.unpack("!I", await reader.readexactly(4))
-                    pw = (await reader.readexactly(ln - 4)).rstrip(b"\x00")
-                    ok = tag == b"p" and pg_md5_check(
-                        stored, user, salt, pw.decode(errors="replace"))
+                    body = await reader.readexactly(ln - 4)
+                    ok = False
+                    if tag == b"p":
+                        mech_end = body.index(b"\x00")
+                        mech = body[:mech_end].decode()
+                        (rlen,) = struct.unpack("!i", body[mech_end + 1:
+                                                           mech_end + 5])
+                        initial = body[mech_end + 5: mech_end + 5 + rlen] \
+                            .decode() if rlen >= 0 else ""
+                        if mech == "SCRAM-SHA-256" and initial:
+                            scram = ScramSha256Server(stored)
+                            sfirst = scram.server_first(initial)
+                            writer.write(_msg(b"R", struct.pack("!I", 11) +
+                                              sfirst.encode()))
+                            await writer.drain()
+                            tag2 = await reader.readexactly(1)
+                            (ln2,) = struct.unpack(
+                                "!I", await reader.readexactly(4))
+                            cfinal = (await reader.readexactly(ln2 - 4)).decode()
+                            if tag2 == b"p":
+                                v = scram.verify_client_final(cfinal)
+                                if v is not None:
+                                    writer.write(_msg(
+                                        b"R", struct.pack("!I", 12) + v.encode()))
+                                    ok = True
+                        elif mech == "PLAIN" or not initial:
+                            # legacy fallback: md5 challenge-response
+                            salt = b"\x9a\x17\x2e\x41"
+                            writer.write(_msg(b"R", struct.pack("!I", 5) + salt))
+                            await writer.drain()
+                            tag3 = await reader.readexactly(1)
+                            (ln3,) = struct.unpack(
+                                "!I", await reader.readexactly(4))
+                            pw = (await reader.readexactly(ln3 - 4)).rstrip(b"\x00")
+                            ok = tag3 == b"p" and pg_md5_check(
+                                stored, user, salt, pw.decode(errors="replace"))
                 if not ok:
                     writer.write(_msg(b"E", b"SSFATAL\x00C28000\x00M" +
                                       f"auth failed for {user}".encode() + b"\x00\x00"))

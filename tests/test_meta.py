@@ -158,3 +158,24 @@ def test_election(tmp_path):
     time.sleep(0.3)                         # b stops heartbeating → expiry
     assert a.campaign()
     assert a.leader() == "node-a"
+
+
+def test_event_recorder(tmp_path):
+    """DDL + migration events land in the greptime_events system table
+    (reference src/common/event-recorder recorder.rs:311)."""
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "ev"), device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE evt_src (h STRING, ts TIMESTAMP TIME INDEX,"
+               " v DOUBLE, PRIMARY KEY (h))")
+    ex.execute("DROP TABLE evt_src")
+    r = ex.execute("SELECT event_type, payload FROM greptime_events"
+                   " ORDER BY ts")
+    types = [row[0] for row in r.rows()]
+    assert "create_table" in types and "drop_table" in types
+    import json as _json
+    payloads = [_json.loads(row[1]) for row in r.rows()]
+    assert any(p.get("table") == "evt_src" for p in payloads)
+    eng.close()

@@ -355,3 +355,78 @@ def test_mysql_prepared_statements(ex):
     names, rows = asyncio.run(run())
     assert names == ["h", "v"]
     assert rows == [["a", "1.5"], ["b", "2.5"]]
+
+
+def test_postgres_scram_auth(ex):
+    """SCRAM-SHA-256 SASL exchange (RFC 7677; reference pgwire SCRAM)."""
+    import struct as _st
+    from greptimedb_amd.servers.auth import (StaticUserProvider,
+                                             scram_client_messages)
+
+    async def run(password):
+        provider = StaticUserProvider({"alice": "s3cret"})
+        srv = PostgresServer(ex, host="127.0.0.1", port=0,
+                             user_provider=provider)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        body = _st.pack("!I", 196608) + b"user\x00alice\x00\x00"
+        writer.write(_st.pack("!I", len(body) + 4) + body)
+        await writer.drain()
+
+        async def read_msg():
+            tag = await reader.readexactly(1)
+            (ln,) = _st.unpack("!I", await reader.readexactly(4))
+            return tag, await reader.readexactly(ln - 4)
+
+        tag, payload = await read_msg()
+        assert tag == b"R" and _st.unpack("!I", payload[:4])[0] == 10
+        assert b"SCRAM-SHA-256" in payload
+
+        sfirst_holder = {}
+
+        async def do_exchange():
+            def server_first_fn(client_first):
+                return sfirst_holder["v"]
+            # client-first
+            import base64, hashlib, hmac as _hmac, os as _os
+            cnonce = base64.b64encode(_os.urandom(18)).decode()
+            bare = f"n=alice,r={cnonce}"
+            cfirst = ("n,," + bare).encode()
+            writer.write(b"p" + _st.pack(
+                "!I", 4 + len(b"SCRAM-SHA-256\x00") + 4 + len(cfirst)) +
+                b"SCRAM-SHA-256\x00" + _st.pack("!i", len(cfirst)) + cfirst)
+            await writer.drain()
+            t2, pl2 = await read_msg()
+            assert t2 == b"R" and _st.unpack("!I", pl2[:4])[0] == 11
+            sfirst = pl2[4:].decode()
+            attrs = dict(kv.split("=", 1) for kv in sfirst.split(","))
+            salt = base64.b64decode(attrs["s"])
+            salted = hashlib.pbkdf2_hmac("sha256", password.encode(), salt,
+                                         int(attrs["i"]))
+            ckey = _hmac.new(salted, b"Client Key", hashlib.sha256).digest()
+            skey = hashlib.sha256(ckey).digest()
+            wo_proof = f"c=biws,r={attrs['r']}"
+            auth_msg = ",".join([bare, sfirst, wo_proof]).encode()
+            csig = _hmac.new(skey, auth_msg, hashlib.sha256).digest()
+            proof = base64.b64encode(bytes(a ^ b for a, b in
+                                           zip(ckey, csig))).decode()
+            cfinal = f"{wo_proof},p={proof}".encode()
+            writer.write(b"p" + _st.pack("!I", 4 + len(cfinal)) + cfinal)
+            await writer.drain()
+            return await read_msg()
+
+        t3, pl3 = await do_exchange()
+        if password == "s3cret":
+            assert t3 == b"R" and _st.unpack("!I", pl3[:4])[0] == 12
+            assert pl3[4:].startswith(b"v=")
+            t4, pl4 = await read_msg()
+            assert t4 == b"R" and _st.unpack("!I", pl4[:4])[0] == 0
+        else:
+            assert t3 == b"E"
+        writer.close()
+        s.close()
+        return True
+
+    assert asyncio.run(run("s3cret"))
+    assert asyncio.run(run("wrongpw"))

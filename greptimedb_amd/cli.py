@@ -63,7 +63,8 @@ def main(argv=None):
     engine = MitoEngine(EngineConfig(
         data_dir=cfg.get("data_dir", args.data_dir), device=device,
         default_regions=int(cfg.get("regions", args.regions)),
-        wal_sync=bool(cfg.get("wal_sync", args.wal_sync))))
+        wal_sync=bool(cfg.get("wal_sync", args.wal_sync)),
+        record_events=True))
     user_provider = None
     if args.user_provider and args.user_provider.startswith("static_user_provider:file:"):
         from greptimedb_amd.servers.auth import StaticUserProvider

@@ -36,6 +36,7 @@ class EngineConfig:
     default_regions: int = 4
     background_flush: bool = True
     scan_mem_bytes: int = 32 << 30      # host-side scan materialization quota
+    record_events: bool = False         # persist DDL/migration events table
 
 
 @dataclass
@@ -120,7 +121,7 @@ class MitoEngine:
             existed = schema.name in self.tables
             st = self._create_table_locked(schema, n_regions, append_mode,
                                            if_not_exists)
-        if not existed:
+        if not existed and self.config.record_events:
             from greptimedb_amd.utils.events import EVENTS_TABLE, recorder_of
             if schema.name != EVENTS_TABLE:
                 recorder_of(self).record("create_table", {
@@ -152,9 +153,10 @@ class MitoEngine:
         if st is None:
             raise TableNotFound(name)
         self._save_catalog()
-        from greptimedb_amd.utils.events import EVENTS_TABLE, recorder_of
-        if name != EVENTS_TABLE:
-            recorder_of(self).record("drop_table", {"table": name})
+        if self.config.record_events:
+            from greptimedb_amd.utils.events import EVENTS_TABLE, recorder_of
+            if name != EVENTS_TABLE:
+                recorder_of(self).record("drop_table", {"table": name})
 
     def table(self, name: str) -> TableState:
         try:

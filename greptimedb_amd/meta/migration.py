@@ -179,9 +179,10 @@ class CrossEngineMigration(Procedure):
                 eng.routing_epoch = getattr(eng, "routing_epoch", 0) + 1
                 eng._save_catalog()
             from greptimedb_amd.utils.events import recorder_of
-            recorder_of(dst).record("region_migration", {
-                "table": table, "region_idx": ridx,
-                "source": state["source"], "target": state["target"]})
+            if dst.config.record_events:
+                recorder_of(dst).record("region_migration", {
+                    "table": table, "region_idx": ridx,
+                    "source": state["source"], "target": state["target"]})
             state["phase"] = "close_downgraded"
             return Status.EXECUTING, state
 

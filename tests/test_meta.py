@@ -166,7 +166,7 @@ def test_event_recorder(tmp_path):
     from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
     from greptimedb_amd.query.executor import Executor
     eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "ev"), device="cpu",
-                                  background_flush=False))
+                                  background_flush=False, record_events=True))
     ex = Executor(eng)
     ex.execute("CREATE TABLE evt_src (h STRING, ts TIMESTAMP TIME INDEX,"
                " v DOUBLE, PRIMARY KEY (h))")

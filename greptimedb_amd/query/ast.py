@@ -76,6 +76,16 @@ class Star(Expr):
 
 
 @dataclass
+class CorrMap(Expr):
+    """Decorrelated scalar subquery: per-outer-key scalar values.
+    Evaluates to a per-row tensor via the outer tag column (LUT over series
+    codes — the GPU-native join of `WHERE v > (SELECT agg .. WHERE inner.k
+    = outer.k)`)."""
+    map: dict          # key (str) -> float
+    outer_col: str     # outer tag column supplying the key
+
+
+@dataclass
 class BinOp(Expr):
     op: str  # + - * / % = != < <= > >= and or like
     left: Expr

@@ -834,12 +834,19 @@ class Executor:
 
     # ---------------------------------------------------------- SELECT
 
-    def _resolve_subqueries(self, x):
+    def _resolve_subqueries(self, x, outer=None):
         """Execute uncorrelated (SELECT …) nodes and splice their results in
-        as literals (scalar) / literal lists (IN) before planning."""
+        as literals (scalar) / literal lists (IN); equality-correlated
+        scalar subqueries decorrelate into a grouped map evaluated as a
+        per-series LUT (ast.CorrMap — the MergeScan-side of DataFusion's
+        decorrelation)."""
         if x is None:
             return None
         if isinstance(x, ast.ScalarSubquery):
+            if outer is not None and not x.many:
+                corr = self._try_decorrelate(x.select, outer)
+                if corr is not None:
+                    return corr
             r = self._exec_select(x.select)
             col = list(r.columns[0]) if r.columns else []
 
@@ -855,22 +862,82 @@ class Executor:
                 return [lit(v) for v in col]
             return lit(col[0]) if len(col) else ast.Lit(None)
         if isinstance(x, ast.BinOp):
-            x.left = self._resolve_subqueries(x.left)
-            x.right = self._resolve_subqueries(x.right)
+            x.left = self._resolve_subqueries(x.left, outer)
+            x.right = self._resolve_subqueries(x.right, outer)
         elif isinstance(x, ast.UnaryOp):
-            x.operand = self._resolve_subqueries(x.operand)
+            x.operand = self._resolve_subqueries(x.operand, outer)
         elif isinstance(x, ast.Func):
-            x.args = [self._resolve_subqueries(a) for a in x.args]
+            x.args = [self._resolve_subqueries(a, outer) for a in x.args]
         elif isinstance(x, ast.Between):
-            x.low = self._resolve_subqueries(x.low)
-            x.high = self._resolve_subqueries(x.high)
+            x.low = self._resolve_subqueries(x.low, outer)
+            x.high = self._resolve_subqueries(x.high, outer)
         elif isinstance(x, ast.InList):
             items = []
             for it in x.items:
-                got = self._resolve_subqueries(it)
+                got = self._resolve_subqueries(it, outer)
                 items.extend(got if isinstance(got, list) else [got])
             x.items = items
         return x
+
+    def _try_decorrelate(self, inner: ast.Select, outer: ast.Select):
+        """`(SELECT agg(x) FROM t2 [t2a] WHERE t2.k = o.k AND ...)` with the
+        outer qualifier matching `outer`'s table/alias → grouped map.
+        Returns ast.CorrMap or None (not correlated / unsupported)."""
+        if not isinstance(inner.table, str) or inner.where is None or \
+                not isinstance(outer.table, str):
+            return None
+        if len(inner.projections) != 1:
+            return None
+        proj, _a = inner.projections[0]
+        if not (isinstance(proj, ast.Func) and proj.name in AGG_FUNCS):
+            return None
+        outer_quals = {outer.table}
+        if outer.table_alias:
+            outer_quals.add(outer.table_alias)
+        inner_quals = {inner.table}
+        if inner.table_alias:
+            inner_quals.add(inner.table_alias)
+
+        def split(e, out):
+            if isinstance(e, ast.BinOp) and e.op == "and":
+                split(e.left, out)
+                split(e.right, out)
+            else:
+                out.append(e)
+        conjs: list = []
+        split(inner.where, conjs)
+        corr_idx = inner_key = outer_key = None
+        for i, c in enumerate(conjs):
+            if not (isinstance(c, ast.BinOp) and c.op == "=" and
+                    isinstance(c.left, ast.Col) and isinstance(c.right, ast.Col)):
+                continue
+            for a, b in ((c.left.name, c.right.name),
+                         (c.right.name, c.left.name)):
+                if "." in a:
+                    qa, ca = a.rsplit(".", 1)
+                    if qa in outer_quals:
+                        cb = b.rsplit(".", 1)[1] if "." in b else b
+                        corr_idx, inner_key, outer_key = i, cb, ca
+                        break
+            if corr_idx is not None:
+                break
+        if corr_idx is None:
+            return None
+        rest = [c for i, c in enumerate(conjs) if i != corr_idx]
+        where2 = None
+        for c in rest:
+            where2 = c if where2 is None else ast.BinOp("and", where2, c)
+        grouped = ast.Select(
+            projections=[(ast.Col(inner_key), "__k"), (proj, "__v")],
+            table=inner.table, table_alias=inner.table_alias,
+            where=where2, group_by=[ast.Col(inner_key)])
+        r = self._exec_select(grouped)
+        keys = list(r.columns[0]) if r.columns else []
+        vals = list(r.columns[1]) if len(r.columns) > 1 else []
+        m = {None if k is None else str(k):
+             (float(v) if v is not None else float("nan"))
+             for k, v in zip(keys, vals)}
+        return ast.CorrMap(m, outer_key)
 
     def _exec_select(self, sel: ast.Select) -> QueryResult:
         if sel.ctes:
@@ -891,10 +958,10 @@ class Executor:
                 return select_over_result(sel, vt)
         if _has_subquery(sel.where) or _has_subquery(sel.having) or \
                 any(_has_subquery(e) for e, _a in sel.projections):
-            sel.projections = [(self._resolve_subqueries(e), a)
+            sel.projections = [(self._resolve_subqueries(e, sel), a)
                                for e, a in sel.projections]
-            sel.where = self._resolve_subqueries(sel.where)
-            sel.having = self._resolve_subqueries(sel.having)
+            sel.where = self._resolve_subqueries(sel.where, sel)
+            sel.having = self._resolve_subqueries(sel.having, sel)
         if sel.table is None:
             # constant select
             names, cols = [], []
@@ -2417,6 +2484,15 @@ class Executor:
                 return _np_binop(x.op, _as_t(value(x.left)), _as_t(value(x.right)))
             if isinstance(x, ast.Interval):
                 return x.ms
+            if isinstance(x, ast.CorrMap):
+                # decorrelated subquery: per-series threshold LUT (device)
+                vals = region.series.tag_array(x.outer_col)
+                arr = np.array([x.map.get(v, np.nan) for v in vals],
+                               dtype=np.float64)
+                if len(arr) == 0:
+                    return torch.full((n,), float("nan"), device=device)
+                lut = torch.as_tensor(arr, device=device)
+                return lut[src.series.long()]
             if isinstance(x, ast.Func):
                 return _eval_const(x)    # now() etc — constant-folded
             raise PlanQuery(f"unsupported predicate operand {x}")

@@ -104,16 +104,18 @@ def load_sidecar(sst_path: str, text_cols: dict, device,
             np.zeros(0, dtype=np.int32)
         tids = np.asarray(tids, dtype=np.int32)
         # probe binary-searches uniq_tids → re-sort the CSR by tid
+        # (fully vectorized gather: O(total), no per-term python loop)
         order = np.argsort(tids, kind="stable")
-        if not np.array_equal(order, np.arange(k)):
+        if k and not np.array_equal(order, np.arange(k)):
             counts = np.diff(starts)
-            new_rows = np.empty_like(rows)
+            lens = counts[order]
             new_starts = np.zeros(k + 1, dtype=np.int64)
-            new_starts[1:] = np.cumsum(counts[order])
-            for ni, oi in enumerate(order):
-                new_rows[new_starts[ni]:new_starts[ni + 1]] = \
-                    rows[starts[oi]:starts[oi + 1]]
-            tids, starts, rows = tids[order], new_starts, new_rows
+            np.cumsum(lens, out=new_starts[1:])
+            seg_id = np.repeat(np.arange(k), lens)
+            within = np.arange(int(lens.sum()), dtype=np.int64) - \
+                np.repeat(new_starts[:-1], lens)
+            gather = starts[order][seg_id] + within
+            tids, starts, rows = tids[order], new_starts, rows[gather]
         if row_remap is not None and rows.size:
             rows = row_remap[rows]
         out[col] = SegmentPostings(tids, starts,

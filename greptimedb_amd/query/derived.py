@@ -19,7 +19,12 @@ from greptimedb_amd.query import ast
 from greptimedb_amd.utils.errors import PlanQuery
 
 AGGS = {"count", "sum", "min", "max", "avg", "mean", "last_value",
-        "first_value", "stddev", "var"}
+        "first_value", "stddev", "var",
+        # sketch/approx aggregates (reference aggrs/approximate)
+        "hll", "hll_merge", "uddsketch_state", "uddsketch_merge",
+        "approx_percentile", "median"}
+
+SKETCH_SCALARS = {"hll_count", "uddsketch_calc"}
 
 
 def _contains_agg(e) -> bool:
@@ -27,6 +32,8 @@ def _contains_agg(e) -> bool:
         if e.name.lower() in AGGS:
             return True
         return any(_contains_agg(a) for a in e.args)
+    if isinstance(e, (ast.Lit, ast.Col, ast.Star)):
+        return False
     if isinstance(e, ast.BinOp):
         return _contains_agg(e.left) or _contains_agg(e.right)
     if isinstance(e, ast.UnaryOp):
@@ -69,6 +76,48 @@ def _eval_scalar(e, col_data: dict, rows: np.ndarray):
     """Evaluate an expression (possibly containing aggregates) over the
     row subset `rows` to a scalar (aggregate context)."""
     from greptimedb_amd.query import executor as X
+    fn = e.name.lower() if isinstance(e, ast.Func) else None
+    if fn in ("hll", "hll_merge", "uddsketch_state", "uddsketch_merge",
+              "approx_percentile", "median"):
+        from greptimedb_amd.query.sketches import Hll, UddSketch
+        # value args are the LAST positional arg; leading args are params
+        arg = e.args[-1]
+        vals = np.asarray(X._np_raw(arg, col_data))[rows]
+        if fn == "hll":
+            return Hll().add_values(vals.tolist()).dumps()
+        if fn == "hll_merge":
+            h = Hll()
+            for st_ in vals:
+                if st_ is not None:
+                    h.merge(Hll.loads(st_))
+            return h.dumps()
+        if fn == "uddsketch_state":
+            nb = int(X._eval_const(e.args[0])) if len(e.args) > 2 else 128
+            al = float(X._eval_const(e.args[1])) if len(e.args) > 2 else 0.01
+            return UddSketch(nb, al).add_values(
+                vals.astype(np.float64).tolist()).dumps()
+        if fn == "uddsketch_merge":
+            out = None
+            for st_ in vals:
+                if st_ is None:
+                    continue
+                sk = UddSketch.loads(st_)
+                out = sk if out is None else out.merge(sk)
+            return (out or UddSketch()).dumps()
+        # approx_percentile(col, p) / median(col): exact selection (our
+        # "approximation" is exact — strictly stronger than the sketch)
+        p = float(X._eval_const(e.args[1])) if fn == "approx_percentile"             else 0.5
+        fv = vals.astype(np.float64)
+        fv = fv[~np.isnan(fv)]
+        return float(np.quantile(fv, p)) if len(fv) else None
+    if fn in SKETCH_SCALARS:
+        from greptimedb_amd.query.sketches import hll_count, uddsketch_calc
+        if fn == "hll_count":
+            st_ = _eval_scalar(e.args[0], col_data, rows)
+            return None if st_ is None else hll_count(st_)
+        q = float(X._eval_const(e.args[0]))
+        st_ = _eval_scalar(e.args[1], col_data, rows)
+        return None if st_ is None else uddsketch_calc(q, st_)
     if isinstance(e, ast.Func) and e.name.lower() in AGGS:
         if e.name.lower() == "count" and (not e.args or
                                           isinstance(e.args[0], ast.Star)):

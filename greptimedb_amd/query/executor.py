@@ -962,6 +962,9 @@ class Executor:
                                for e, a in sel.projections]
             sel.where = self._resolve_subqueries(sel.where, sel)
             sel.having = self._resolve_subqueries(sel.having, sel)
+        if isinstance(sel.table, str) and sel.table in self.engine.tables \
+                and self._has_sketch_agg(sel):
+            return self._exec_sketch_select(sel)
         if sel.table is None:
             # constant select
             names, cols = [], []
@@ -1338,7 +1341,71 @@ class Executor:
             out_cols = [c[: sel.limit] for c in out_cols]
         return QueryResult(out_names, out_cols)
 
+    _SKETCH_AGGS = {"hll", "hll_merge", "uddsketch_state", "uddsketch_merge",
+                    "approx_percentile", "median", "hll_count",
+                    "uddsketch_calc"}
+
+    def _has_sketch_agg(self, sel) -> bool:
+        def walk(e):
+            if isinstance(e, ast.Func):
+                if e.name.lower() in self._SKETCH_AGGS:
+                    return True
+                return any(walk(a) for a in e.args)
+            if isinstance(e, ast.BinOp):
+                return walk(e.left) or walk(e.right)
+            if isinstance(e, ast.UnaryOp):
+                return walk(e.operand)
+            return False
+        return any(walk(e) for e, _a in sel.projections) or \
+            (sel.having is not None and walk(sel.having))
+
+    def _exec_sketch_select(self, sel: ast.Select) -> QueryResult:
+        """Sketch/approx aggregates (hll / uddsketch / approx_percentile):
+        materialize the referenced raw columns (GPU scan + filter), then
+        aggregate via the derived evaluator (reference: DataFusion runs
+        these UDAFs over the scanned batches the same way)."""
+        import dataclasses
+
+        from greptimedb_amd.query.derived import select_over_result
+        st = self.engine.table(sel.table)
+        region0 = st.regions[0]
+        table_cols = ({c.name for c in st.schema.columns} |
+                      set(region0.field_names) |
+                      set(region0.str_field_names) |
+                      {st.schema.time_index.name})
+        needed: set = set()
+
+        def collect(e):
+            if isinstance(e, ast.Col) and e.name in table_cols:
+                needed.add(e.name)
+            elif isinstance(e, ast.Func):
+                for a in e.args:
+                    collect(a)
+            elif isinstance(e, ast.BinOp):
+                collect(e.left)
+                collect(e.right)
+            elif isinstance(e, ast.UnaryOp):
+                collect(e.operand)
+        for e, _a in sel.projections:
+            collect(e)
+        for g in sel.group_by:
+            collect(g)
+        if sel.having is not None:
+            collect(sel.having)
+        for e, _d in sel.order_by:
+            collect(e)
+        if not needed:
+            needed.add(st.schema.time_index.name)
+        base_sel = ast.Select(
+            projections=[(ast.Col(c), None) for c in sorted(needed)],
+            table=sel.table, table_alias=sel.table_alias, where=sel.where)
+        base = self._exec_select(base_sel)
+        outer = dataclasses.replace(sel, table=None, table_alias=None,
+                                    where=None)
+        return select_over_result(outer, base)
+
     def _with_ctes(self, stmt):
+
         """Evaluate WITH ctes into the virtual-table scope, then run the
         body (reference: DataFusion CTE planning; cases in tests/cases/cte)."""
         import dataclasses

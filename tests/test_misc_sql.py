@@ -402,3 +402,34 @@ def test_correlated_subquery_agg(tmp_engine):
     r = ex.execute("SELECT h FROM ct t WHERE v > "
                    "(SELECT max(lim) FROM thr WHERE thr.h = t.h)")
     assert sorted(x[0] for x in r.rows()) == ["a", "b"]
+
+
+def test_sketch_aggregates(tmp_engine):
+    """hll/hll_count, uddsketch_state/uddsketch_calc, approx_percentile
+    (reference aggrs/approximate + scalars/{hll_count,uddsketch_calc})."""
+    import numpy as np
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE sk (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE,"
+               " u STRING, PRIMARY KEY (h)) WITH ('append_mode'='true')")
+    rows = ", ".join(
+        f"('h{i % 4}', {i}, {float(i % 100)}, 'user_{i % 500}')"
+        for i in range(2000))
+    ex.execute(f"INSERT INTO sk (h, ts, v, u) VALUES {rows}")
+    # distinct estimate within 3%
+    r = ex.execute("SELECT hll_count(hll(u)) AS d FROM sk")
+    assert abs(float(r.columns[0][0]) - 500) / 500 < 0.03
+    # grouped hll
+    r = ex.execute("SELECT h, hll_count(hll(u)) AS d FROM sk GROUP BY h ORDER BY h")
+    assert len(r.rows()) == 4
+    # per group: i ≡ g (mod 4) → users g, g+4, … mod 500 → 125 distinct
+    assert all(abs(float(d) - 125) / 125 < 0.05 for _h, d in r.rows())
+    # uddsketch p90 within 10%
+    r = ex.execute("SELECT uddsketch_calc(0.9, uddsketch_state(128, 0.01, v))"
+                   " AS p FROM sk")
+    assert abs(float(r.columns[0][0]) - 89.0) < 9.0
+    # approx_percentile (exact selection here)
+    r = ex.execute("SELECT approx_percentile(v, 0.5) AS med FROM sk"
+                   " WHERE h = 'h0'")
+    vals = [float(i % 100) for i in range(2000) if i % 4 == 0]
+    assert float(r.columns[0][0]) == np.quantile(vals, 0.5)

@@ -80,9 +80,10 @@ def _eval_scalar(e, col_data: dict, rows: np.ndarray):
     if fn in ("hll", "hll_merge", "uddsketch_state", "uddsketch_merge",
               "approx_percentile", "median"):
         from greptimedb_amd.query.sketches import Hll, UddSketch
-        # value args are the LAST positional arg; leading args are params
-        arg = e.args[-1]
-        vals = np.asarray(X._np_raw(arg, col_data))[rows]
+        # value column: first arg for approx_percentile/median (col, p);
+        # LAST arg for the sketch builders (params lead: b, e, col)
+        arg = e.args[0] if fn in ("approx_percentile", "median") else e.args[-1]
+        vals = np.atleast_1d(np.asarray(X._np_raw(arg, col_data)))[rows]
         if fn == "hll":
             return Hll().add_values(vals.tolist()).dumps()
         if fn == "hll_merge":

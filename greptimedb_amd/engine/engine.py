@@ -87,6 +87,7 @@ class MitoEngine:
             cat = json.load(f)
         self.next_table_id = cat["next_table_id"]
         self.views = dict(cat.get("views", {}))
+        self.schemas = set(cat.get("schemas", []))
         for td in cat["tables"]:
             schema = TableSchema.from_dict(td["schema"])
             st = TableState(schema=schema, append_mode=td["append_mode"])
@@ -101,6 +102,7 @@ class MitoEngine:
         cat = {
             "next_table_id": self.next_table_id,
             "views": getattr(self, "views", {}),
+            "schemas": sorted(getattr(self, "schemas", set())),
             "tables": [
                 {
                     "schema": st.schema.to_dict(),

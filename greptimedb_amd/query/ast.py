@@ -281,3 +281,31 @@ class Tql:
     end: float
     step: float
     query: str
+
+
+@dataclass
+class Use:
+    schema: str
+
+
+@dataclass
+class SetVar:
+    name: str
+    value: object
+
+
+@dataclass
+class CreateDatabase:
+    name: str
+    if_not_exists: bool = False
+
+
+@dataclass
+class DropDatabase:
+    name: str
+    if_exists: bool = False
+
+
+@dataclass
+class ShowVariables:
+    like: str | None = None

@@ -152,6 +152,8 @@ class Executor:
         self.engine = engine
         self.dist = dist  # parallel.dist.DistContext or None
         self._virtual: dict = {}   # CTE scope: name -> QueryResult
+        from greptimedb_amd.query.session import Session
+        self.session = Session()
 
     # ---------------------------------------------------------- entrypoints
 
@@ -176,6 +178,49 @@ class Executor:
         return r
 
     def execute_stmt(self, stmt) -> QueryResult:
+        self._qualify_names(stmt)
+        if isinstance(stmt, ast.Use):
+            schemas = getattr(self.engine, "schemas", set()) | \
+                {"public", "greptime_private", "information_schema"}
+            if stmt.schema not in schemas:
+                raise TableNotFound(f"database {stmt.schema}")
+            self.session.schema = stmt.schema
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.SetVar):
+            self.session.set_var(stmt.name, stmt.value)
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.ShowVariables):
+            base = {"time_zone": self.session.timezone,
+                    "current_schema": self.session.schema}
+            base.update(self.session.vars)
+            items = sorted(base.items())
+            if stmt.like:
+                import fnmatch
+                pat = stmt.like.replace("%", "*").replace("_", "?")
+                items = [(k, v) for k, v in items if fnmatch.fnmatch(k, pat)]
+            return QueryResult(["Variable_name", "Value"],
+                               [[k for k, _ in items], [str(v) for _, v in items]])
+        if isinstance(stmt, ast.CreateDatabase):
+            schemas = getattr(self.engine, "schemas", None)
+            if schemas is None:
+                schemas = self.engine.schemas = set()
+            if stmt.name in schemas and not stmt.if_not_exists:
+                raise TableAlreadyExists(f"database {stmt.name}")
+            schemas.add(stmt.name)
+            self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.DropDatabase):
+            schemas = getattr(self.engine, "schemas", set())
+            if stmt.name not in schemas:
+                if not stmt.if_exists:
+                    raise TableNotFound(f"database {stmt.name}")
+            else:
+                schemas.discard(stmt.name)
+                for t in [t for t in list(self.engine.tables)
+                          if t.startswith(stmt.name + ".")]:
+                    self.engine.drop_table(t)
+                self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.Select):
             return self._exec_select(stmt)
         if isinstance(stmt, ast.SetOp):
@@ -204,9 +249,11 @@ class Executor:
             names = sorted(self.engine.tables)
             return QueryResult(["Tables"], [names])
         if isinstance(stmt, ast.ShowDatabases):
+            extra = sorted(getattr(self.engine, "schemas", set()))
             return QueryResult(["Database"],
-                               [["greptime_private", "information_schema",
-                                 "public"]])
+                               [sorted({"greptime_private",
+                                        "information_schema", "public",
+                                        *extra})])
         if isinstance(stmt, ast.ShowCreateTable):
             return self._show_create_table(stmt.name)
         if isinstance(stmt, ast.DescribeTable):
@@ -1341,6 +1388,47 @@ class Executor:
             out_cols = [c[: sel.limit] for c in out_cols]
         return QueryResult(out_names, out_cols)
 
+    def _qualify_names(self, stmt):
+        """Schema-qualify table names via the session (reference:
+        QueryContext current_schema resolution)."""
+        if self.session.schema == "public":
+            return
+        res = self.session.resolve_table
+        def fix_select(sel):
+            if isinstance(sel, ast.SetOp):
+                fix_select(sel.left)
+                fix_select(sel.right)
+                return
+            if not isinstance(sel, ast.Select):
+                return
+            cte_names = {n for n, _q in sel.ctes}
+            for _n, q in sel.ctes:
+                fix_select(q)
+            if isinstance(sel.table, str):
+                from greptimedb_amd.query.information_schema import \
+                    is_information_schema
+                if sel.table not in cte_names and \
+                        sel.table not in self._virtual and \
+                        not is_information_schema(sel.table):
+                    sel.table = res(sel.table)
+            elif sel.table is not None:
+                fix_select(sel.table)
+            for j in sel.joins:
+                j.table = res(j.table)
+        if isinstance(stmt, (ast.Select, ast.SetOp)):
+            fix_select(stmt)
+        elif isinstance(stmt, ast.CreateTable):
+            stmt.name = res(stmt.name)
+        elif isinstance(stmt, (ast.DropTable, ast.ShowCreateTable,
+                               ast.DescribeTable)):
+            stmt.name = res(stmt.name)
+        elif isinstance(stmt, ast.InsertValues):
+            stmt.table = res(stmt.table)
+        elif isinstance(stmt, ast.Delete):
+            stmt.table = res(stmt.table)
+        elif isinstance(stmt, ast.AlterTable):
+            stmt.table = res(stmt.table)
+
     _SKETCH_AGGS = {"hll", "hll_merge", "uddsketch_state", "uddsketch_merge",
                     "approx_percentile", "median", "hll_count",
                     "uddsketch_calc"}
@@ -1563,7 +1651,13 @@ class Executor:
             unit = trunc_unit_ms(str(f.args[0].value))
             if unit is None:
                 raise PlanQuery(f"unsupported date_trunc unit {f.args[0].value}")
-            return BucketSpec(bucket_ms=unit)
+            origin = None
+            tz = getattr(self.session, "tz_offset_ms", 0)
+            if tz and unit >= 3_600_000:
+                # session time zone shifts the truncation grid (local
+                # midnight != UTC midnight; reference QueryContext timezone)
+                origin = (-tz) % unit
+            return BucketSpec(bucket_ms=unit, origin=origin)
         # date_bin(interval, ts[, origin]) / time_bucket(interval, ts)
         if not f.args or not isinstance(f.args[0], (ast.Interval, ast.Lit)):
             raise PlanQuery("date_bin(interval, ts) expected")

@@ -161,6 +161,16 @@ class Parser:
     def parse_statement_inner(self):
         if self.at_kw("select", "with"):
             stmt = self.parse_query()
+        elif self.at_kw("use"):
+            self.next()
+            stmt = ast.Use(self.next().value)
+        elif self.at_kw("set"):
+            self.next()
+            self.eat_kw("session") or self.eat_kw("global")
+            name = str(self.next().value)
+            self.expect_op("=")
+            t = self.next()
+            stmt = ast.SetVar(name, t.value)
         elif self.at_kw("create"):
             stmt = self.parse_create()
         elif self.at_kw("drop"):
@@ -174,6 +184,11 @@ class Parser:
             elif self.eat_kw("create"):
                 self.expect_kw("table")
                 stmt = ast.ShowCreateTable(self.next().value)
+            elif self.eat_kw("variables"):
+                like = None
+                if self.eat_kw("like"):
+                    like = str(self.next().value)
+                stmt = ast.ShowVariables(like)
             else:
                 self.expect_kw("tables")
                 stmt = ast.ShowTables()
@@ -403,6 +418,12 @@ class Parser:
 
     def parse_create(self):
         self.expect_kw("create")
+        if self.eat_kw("database") or self.eat_kw("schema"):
+            if_not_exists = False
+            if self.eat_kw("if"):
+                self.expect_kw("not"); self.expect_kw("exists")
+                if_not_exists = True
+            return ast.CreateDatabase(self.next().value, if_not_exists)
         if self.eat_kw("flow"):
             if_not_exists = False
             if self.eat_kw("if"):
@@ -573,6 +594,12 @@ class Parser:
 
     def parse_drop(self):
         self.expect_kw("drop")
+        if self.eat_kw("database") or self.eat_kw("schema"):
+            if_exists = False
+            if self.eat_kw("if"):
+                self.expect_kw("exists")
+                if_exists = True
+            return ast.DropDatabase(self.next().value, if_exists)
         if self.eat_kw("flow"):
             return ast.DropFlow(self.next().value)
         if self.eat_kw("view"):

@@ -38,6 +38,10 @@ void launch_scatter_append(
     const int64_t*, const int32_t*, const double*, const int32_t*,
     const int64_t*, int64_t* const*, int32_t* const*, double* const*,
     const int64_t*, int, int64_t, hipStream_t);
+void launch_rle_expand_indices(const int64_t*, int64_t, const uint8_t*, int,
+                               int32_t*, int64_t, hipStream_t);
+void launch_gorilla_decode(const uint8_t*, const int64_t*, const int64_t*,
+                           int64_t, int64_t*, double*, int64_t, hipStream_t);
 }  // namespace gdb_hip
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on GPU")
@@ -231,6 +235,37 @@ void scatter_append(
       pbase + 3 * R, nf, n, stream);
 }
 
+
+// K11: parquet hybrid RLE/bit-packed dictionary indices → i32 on device.
+// runs: i64[R,5] (host-built, csrc/pagedec.cpp), blob: u8 device tensor
+// (pad >= 8 bytes past the last packed bit).
+torch::Tensor rle_expand_indices(torch::Tensor runs, torch::Tensor blob,
+                                 int64_t bw, int64_t n) {
+  CHECK_GPU(runs); CHECK_CONTIG(runs); CHECK_GPU(blob); CHECK_CONTIG(blob);
+  auto out = torch::empty({n}, blob.options().dtype(torch::kInt32));
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_rle_expand_indices(
+      runs.data_ptr<int64_t>(), runs.size(0), blob.data_ptr<uint8_t>(),
+      (int)bw, out.data_ptr<int32_t>(), n, stream);
+  return out;
+}
+
+// K20: Gorilla/delta block decode → (ts i64[n], vals f64[n]) on device.
+std::vector<torch::Tensor> gorilla_decode(torch::Tensor blob,
+                                          torch::Tensor block_off,
+                                          torch::Tensor out_off, int64_t n) {
+  CHECK_GPU(blob); CHECK_CONTIG(blob);
+  CHECK_GPU(block_off); CHECK_GPU(out_off);
+  auto ts = torch::empty({n}, blob.options().dtype(torch::kInt64));
+  auto vals = torch::empty({n}, blob.options().dtype(torch::kFloat64));
+  auto stream = at::cuda::getCurrentHIPStream().stream();
+  gdb_hip::launch_gorilla_decode(
+      blob.data_ptr<uint8_t>(), block_off.data_ptr<int64_t>(),
+      out_off.data_ptr<int64_t>(), block_off.numel(),
+      ts.data_ptr<int64_t>(), vals.data_ptr<double>(), n, stream);
+  return {ts, vals};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ts_bucket_agg", &ts_bucket_agg, "fused filter + time-bucket aggregate",
         py::arg("ts"), py::arg("series"), py::arg("fields"), py::arg("field_idx"),
@@ -238,6 +273,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bucket_ms"), py::arg("n_slots"), py::arg("n_buckets"),
         py::arg("acc") = std::vector<torch::Tensor>());
   m.def("ts_bucket_agg_finish", &ts_bucket_agg_finish, "decode raw accumulators");
+  m.def("rle_expand_indices", &rle_expand_indices, "K11 hybrid RLE expand");
+  m.def("gorilla_decode", &gorilla_decode, "K20 gorilla block decode");
   m.def("filter_series_time", &filter_series_time, "series/time filter mask");
   m.def("dedup_mark_last", &dedup_mark_last, "last-row dedup marker");
   m.def("series_last_ts", &series_last_ts, "per-slot max-ts accumulate (lastpoint)");

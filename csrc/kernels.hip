@@ -653,4 +653,149 @@ void launch_series_last_row(
       ts, series, slot_lut, lut_size, ts_lo, ts_hi, n, best_key, src_tag, best_row);
 }
 
+
+
+// ------------------------------------------------- K11 RLE/bit-packed expand
+// Parquet hybrid RLE/bit-packed dictionary indices → i32, fully parallel:
+// one thread per OUTPUT element; the run covering the element is found by
+// binary search over the host-built run table (csrc/pagedec.cpp), then
+// either the RLE constant is written or `bw` bits are extracted at the
+// element's bit offset. Branchy page parsing stays on the host; the
+// bandwidth-bound expansion runs at HBM speed here.
+__global__ void rle_expand_indices_kernel(
+    const int64_t* __restrict__ runs,   // [R,5] is_packed, off, val, out_start, count
+    int64_t R,
+    const uint8_t* __restrict__ blob,
+    int bw,
+    int32_t* __restrict__ out, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    // binary search: last run with out_start <= i
+    int64_t lo = 0, hi = R - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (runs[mid * 5 + 3] <= i) lo = mid; else hi = mid - 1;
+    }
+    const int64_t* run = runs + lo * 5;
+    if (run[0] == 0) {
+      out[i] = (int32_t)run[2];
+    } else {
+      const int64_t rel = i - run[3];
+      const int64_t bit = rel * bw;
+      const uint8_t* p = blob + run[1] + (bit >> 3);
+      // assemble up to 40 bits (bw <= 32) from unaligned bytes
+      uint64_t v = (uint64_t)p[0] | ((uint64_t)p[1] << 8) |
+                   ((uint64_t)p[2] << 16) | ((uint64_t)p[3] << 24) |
+                   ((uint64_t)p[4] << 32);
+      out[i] = (int32_t)((v >> (bit & 7)) & ((bw >= 32) ? 0xFFFFFFFFULL
+                                                        : ((1ULL << bw) - 1)));
+    }
+  }
+}
+
+// ------------------------------------------------- K20 Gorilla / delta-delta
+// Block format (engine/gorilla.py packs on host at spill time):
+//   per block of BLK values: header {int64 first_ts, int64 first_delta,
+//   u64 first_val_bits, int32 nbits_ts, int32 nbits_val...} — here the
+//   simple fixed-width variant: delta-of-delta stored with per-block bit
+//   width, XOR'd value bits with per-block width (CDNA4-friendly: fixed
+//   widths per block → branchless bit extraction, one thread per value
+//   with a per-block prefix scan done at pack time so decode is O(1)).
+// Layout per block in `blob`:
+//   [i64 base_ts][i64 base_dod_acc? not needed][u64 base_val]
+//   [u8 ts_bits][u8 val_bits][u16 count][pad to 8B]
+//   [packed ts dods (zigzag, ts_bits each)][packed val xors (val_bits each)]
+// ts[i] = base_ts + prefix_sum(deltas); to keep decode parallel the packer
+// stores ABSOLUTE deltas from base (ts[i] - base_ts) instead of
+// delta-of-delta when that fits the same width; flag in high bit of ts_bits.
+__global__ void gorilla_decode_kernel(
+    const uint8_t* __restrict__ blob,
+    const int64_t* __restrict__ block_off,   // [B] byte offset of each block
+    const int64_t* __restrict__ out_off,     // [B] first output index
+    int64_t B,
+    int64_t* __restrict__ out_ts,
+    double* __restrict__ out_val,
+    int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t total_threads = stride;
+  (void)total_threads;
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;; g += stride) {
+    // map thread → (block, element) via flat index over n
+    if (g >= n) return;
+    // find block by binary search on out_off
+    int64_t lo = 0, hi = B - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (out_off[mid] <= g) lo = mid; else hi = mid - 1;
+    }
+    const uint8_t* blk = blob + block_off[lo];
+    int64_t base_ts;
+    uint64_t base_val;
+    memcpy(&base_ts, blk, 8);
+    memcpy(&base_val, blk + 8, 8);
+    const int ts_bits = blk[16];
+    const int val_bits = blk[17];
+    uint16_t count;
+    memcpy(&count, blk + 18, 2);
+    const int val_mode = blk[20];   // 0 = XOR f64, 1 = scaled int delta
+    const int scale_k = blk[21];
+    double inv_scale = 1.0;
+    for (int q = 0; q < scale_k; q++) inv_scale *= 0.1;
+    const uint8_t* ts_data = blk + 24;
+    const int64_t rel = g - out_off[lo];
+    if (rel >= count) continue;
+    if (rel == 0) {
+      out_ts[g] = base_ts;
+      out_val[g] = (val_mode == 1)
+          ? (double)(int64_t)base_val * inv_scale
+          : __longlong_as_double((long long)base_val);
+      continue;
+    }
+    const int64_t k = rel - 1;   // packed arrays hold values 1..count-1
+    // absolute zigzagged delta from base, ts_bits each
+    uint64_t tsv = 0;
+    if (ts_bits) {
+      const int64_t bit = k * ts_bits;
+      const uint8_t* p = ts_data + (bit >> 3);
+      uint64_t acc = 0;
+      for (int b = 0; b < 9; b++) acc |= (uint64_t)p[b] << (8 * b);
+      tsv = (acc >> (bit & 7)) & ((ts_bits >= 64) ? ~0ULL : ((1ULL << ts_bits) - 1));
+    }
+    const int64_t dz = (int64_t)(tsv >> 1) ^ -(int64_t)(tsv & 1);
+    out_ts[g] = base_ts + dz;
+    const int64_t ts_bytes = ((int64_t)(count - 1) * ts_bits + 7) / 8;
+    const uint8_t* val_data = ts_data + ((ts_bytes + 7) / 8) * 8;
+    uint64_t xv = 0;
+    if (val_bits) {
+      const int64_t bit = k * val_bits;
+      const uint8_t* p = val_data + (bit >> 3);
+      uint64_t acc = 0;
+      for (int b = 0; b < 9; b++) acc |= (uint64_t)p[b] << (8 * b);
+      xv = (acc >> (bit & 7)) & ((val_bits >= 64) ? ~0ULL : ((1ULL << val_bits) - 1));
+    }
+    if (val_mode == 1) {
+      const int64_t dvz = (int64_t)(xv >> 1) ^ -(int64_t)(xv & 1);
+      out_val[g] = (double)((int64_t)base_val + dvz) * inv_scale;
+    } else {
+      out_val[g] = __longlong_as_double((long long)(base_val ^ xv));
+    }
+  }
+}
+
+void launch_rle_expand_indices(
+    const int64_t* runs, int64_t R, const uint8_t* blob, int bw,
+    int32_t* out, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(rle_expand_indices_kernel, dim3(grid_for(n, 256)),
+                     dim3(256), 0, stream, runs, R, blob, bw, out, n);
+}
+
+void launch_gorilla_decode(
+    const uint8_t* blob, const int64_t* block_off, const int64_t* out_off,
+    int64_t B, int64_t* out_ts, double* out_val, int64_t n,
+    hipStream_t stream) {
+  hipLaunchKernelGGL(gorilla_decode_kernel, dim3(grid_for(n, 256)),
+                     dim3(256), 0, stream, blob, block_off, out_off, B,
+                     out_ts, out_val, n);
+}
+
 }  // namespace gdb_hip

@@ -1415,7 +1415,10 @@ static py::tuple route_ingest(
                         std::move(counts), std::move(mins), std::move(maxs));
 }
 
+void register_pagedec(py::module_& m);
+
 PYBIND11_MODULE(_native, m) {
+  register_pagedec(m);
   m.doc() = "greptimedb_amd host-native ingest path (line parser + WAL)";
   py::class_<LineParser>(m, "LineParser")
       .def(py::init<>())

@@ -381,6 +381,7 @@ class Region:
                 continue
             if ts_hi is not None and batch.min_ts >= ts_hi:
                 continue
+            batch.ensure_decoded(device)   # K20 cold tier → HBM tensors
 
             def make_probe(b):
                 def probe(col, terms):

@@ -142,7 +142,56 @@ class SstBatch:
 
     @property
     def n(self) -> int:
+        if self.ts is None and getattr(self, "_gorilla", None):
+            return self._gorilla["__n"]
         return self.ts.numel()
+
+    # ---------------------------------------------------- K20 cold tier
+    def compress(self) -> int:
+        """Pack ts+fields into Gorilla blocks (engine/gorilla.py), freeing
+        the uncompressed HBM tensors. Returns bytes now used. Scans call
+        ensure_decoded() which re-materializes via the GPU decode kernel —
+        the resident-cold-tier policy the reference approximates with its
+        page cache, except the compressed copy stays in HBM."""
+        if self.ts is None:
+            return sum(g.nbytes for k, g in self._gorilla.items()
+                       if k != "__n")
+        from greptimedb_amd.engine import gorilla
+        ts_h = self.ts.cpu().numpy()
+        packs = {"__n": int(self.ts.numel())}
+        # ts packed ONCE; field blocks skip the ts section (ts_bits=0)
+        packs["__ts"] = gorilla.GorillaBatch(ts_h, np.zeros(len(ts_h)))
+        for i in range(self.fields.shape[0]):
+            packs[f"f{i}"] = gorilla.GorillaBatch(
+                ts_h, self.fields[i].cpu().numpy(), pack_ts=False)
+        self._gorilla = packs
+        self._series_h = self.series.cpu().numpy()
+        self._seq_h = self.seq.cpu().numpy() if self.seq is not None else None
+        self.ts = None
+        self.series = None
+        self.fields = None
+        self.seq = None
+        return sum(g.nbytes for k, g in packs.items() if k != "__n")
+
+    def ensure_decoded(self, device) -> "SstBatch":
+        if self.ts is not None or not getattr(self, "_gorilla", None):
+            return self
+        import torch as _torch
+        packs = self._gorilla
+        nf = len([k for k in packs if k not in ("__n", "__ts")])
+        n = packs["__n"]
+        fields = _torch.empty((nf, n), dtype=_torch.float64, device=device)
+        ts, _zero = packs["__ts"].decode(device)
+        for i in range(nf):
+            _t, v = packs[f"f{i}"].decode(device)
+            fields[i] = v
+        self.ts = ts.to(device)
+        self.fields = fields
+        self.series = _torch.as_tensor(self._series_h).to(device)
+        if self._seq_h is not None:
+            self.seq = _torch.as_tensor(self._seq_h).to(device)
+        self._gorilla = None
+        return self
 
 
 def write_sst(path: str, schema: TableSchema, pks: list[bytes],

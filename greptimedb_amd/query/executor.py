@@ -543,6 +543,18 @@ class Executor:
             c = Compactor(trigger_file_num=2)
             n = sum(c.compact_region(r) for r in st.regions)
             return QueryResult(["result"], [[n]])
+        if f == "compress_table":
+            # K20 cold tier: pack resident SST batches into Gorilla blocks
+            st = self.engine.table(str(a.args[0]))
+            before = after = 0
+            for r in st.regions:
+                with r.lock:
+                    for b in r.sst_cache.values():
+                        if b.ts is not None:
+                            before += b.n * (8 + 4 + 8 * b.fields.shape[0])
+                            after += b.compress() + b.n * 4
+            return QueryResult(["bytes_before", "bytes_after"],
+                               [[before], [after]])
         if f == "gc":
             # orphan SST scan (reference src/mito2/src/gc.rs + metasrv gc.rs)
             grace = float(a.args[0]) if a.args else 3600.0

@@ -26,8 +26,14 @@ ROOT = os.path.dirname(os.path.abspath(__file__))
 ext_modules = [
     CppExtension(
         "greptimedb_amd._native",
-        sources=["csrc/native.cpp"],
+        sources=["csrc/native.cpp", "csrc/pagedec.cpp"],
+        # zstd header/lib: conda toolchain ships them; the runtime links
+        # against the system libzstd.so.1 compatible ABI
+        include_dirs=["/opt/conda/include"],
         extra_compile_args=["-O3", "-std=c++17"],
+        libraries=["zstd"],
+        library_dirs=["/opt/conda/lib"],
+        extra_link_args=["-Wl,-rpath,/opt/conda/lib"],
     ),
     CUDAExtension(
         "greptimedb_amd._hip_ops",

@@ -543,6 +543,10 @@ class Executor:
             c = Compactor(trigger_file_num=2)
             n = sum(c.compact_region(r) for r in st.regions)
             return QueryResult(["result"], [[n]])
+        if f == "flow_tick":
+            out = self._flow_engine().tick()
+            return QueryResult(["flow", "rows"],
+                               [list(out.keys()), list(out.values())])
         if f == "compress_table":
             # K20 cold tier: pack resident SST batches into Gorilla blocks
             st = self.engine.table(str(a.args[0]))
@@ -3406,6 +3410,20 @@ def _eval_const(e: ast.Expr):
     if isinstance(e, ast.Func) and e.name == "now":
         import time
         return int(time.time() * 1000)
+    if isinstance(e, ast.Func):
+        import math as _math
+        one = {"abs": abs, "sqrt": _math.sqrt, "floor": _math.floor,
+               "ceil": _math.ceil, "exp": _math.exp, "ln": _math.log,
+               "log": _math.log10, "log2": _math.log2, "sin": _math.sin,
+               "cos": _math.cos, "tan": _math.tan}
+        fn = e.name.lower()
+        if fn in one and len(e.args) == 1:
+            return one[fn](float(_eval_const(e.args[0])))
+        if fn in ("pow", "power") and len(e.args) == 2:
+            return float(_eval_const(e.args[0])) ** float(_eval_const(e.args[1]))
+        if fn == "round":
+            nd = int(_eval_const(e.args[1])) if len(e.args) > 1 else 0
+            return round(float(_eval_const(e.args[0])), nd)
     raise PlanQuery(f"unsupported constant expr {e}")
 
 

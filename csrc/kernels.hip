@@ -757,9 +757,12 @@ __global__ void gorilla_decode_kernel(
     if (ts_bits) {
       const int64_t bit = k * ts_bits;
       const uint8_t* p = ts_data + (bit >> 3);
-      uint64_t acc = 0;
-      for (int b = 0; b < 9; b++) acc |= (uint64_t)p[b] << (8 * b);
-      tsv = (acc >> (bit & 7)) & ((ts_bits >= 64) ? ~0ULL : ((1ULL << ts_bits) - 1));
+      const int sft = (int)(bit & 7);
+      uint64_t w0 = 0;
+      for (int b = 0; b < 8; b++) w0 |= (uint64_t)p[b] << (8 * b);
+      uint64_t v = w0 >> sft;
+      if (sft) v |= (uint64_t)p[8] << (64 - sft);   // spillover bits 64..70
+      tsv = v & ((ts_bits >= 64) ? ~0ULL : ((1ULL << ts_bits) - 1));
     }
     const int64_t dz = (int64_t)(tsv >> 1) ^ -(int64_t)(tsv & 1);
     out_ts[g] = base_ts + dz;
@@ -769,9 +772,12 @@ __global__ void gorilla_decode_kernel(
     if (val_bits) {
       const int64_t bit = k * val_bits;
       const uint8_t* p = val_data + (bit >> 3);
-      uint64_t acc = 0;
-      for (int b = 0; b < 9; b++) acc |= (uint64_t)p[b] << (8 * b);
-      xv = (acc >> (bit & 7)) & ((val_bits >= 64) ? ~0ULL : ((1ULL << val_bits) - 1));
+      const int sft = (int)(bit & 7);
+      uint64_t w0 = 0;
+      for (int b = 0; b < 8; b++) w0 |= (uint64_t)p[b] << (8 * b);
+      uint64_t v = w0 >> sft;
+      if (sft) v |= (uint64_t)p[8] << (64 - sft);
+      xv = v & ((val_bits >= 64) ? ~0ULL : ((1ULL << val_bits) - 1));
     }
     if (val_mode == 1) {
       const int64_t dvz = (int64_t)(xv >> 1) ^ -(int64_t)(xv & 1);

@@ -154,21 +154,38 @@ class Region:
     # ---------------------------------------------------------------- open
 
     def _load_ssts(self):
-        """Open path: load manifest SSTs into the device cache."""
+        """Open path: load manifest SSTs into the device cache. On GPU the
+        numeric columns decode through the K11 path (host thrift+zstd page
+        parse → device RLE/dict expand, engine/pagedec.py) so pages land in
+        HBM without a CPU Arrow materialization; string columns and the pk
+        dictionary still go through pyarrow."""
         for fid, meta in self.manifest.files.items():
             path = os.path.join(self.dir, "sst", f"{fid}.parquet")
             if not os.path.exists(path):
                 continue
-            dict_values, indices, ts, fields, seq, str_cols = sst_mod.read_sst(
-                path, self.schema, self.field_names)
-            remap = np.array([self.series.add_encoded(pk) for pk in dict_values],
-                             dtype=np.int32)
-            codes = remap[indices]
             dev = self.device
-            t_ts = torch.as_tensor(ts).to(dev)
-            t_se = torch.as_tensor(codes).to(dev)
-            t_f = torch.as_tensor(np.ascontiguousarray(fields)).to(dev)
-            t_seq = torch.as_tensor(seq).to(dev)
+            gpu_cols = None
+            if str(dev).startswith("cuda") and not self.str_field_names:
+                try:
+                    gpu_cols = self._load_sst_gpu(path)
+                except Exception:
+                    gpu_cols = None   # unsupported encoding → CPU reader
+            if gpu_cols is not None:
+                t_ts, t_f, t_seq, codes_t = gpu_cols
+                t_se = codes_t
+                str_cols = {}
+                ts = t_ts  # only len used below via tensors
+                seq = t_seq
+            else:
+                dict_values, indices, ts, fields, seq, str_cols = \
+                    sst_mod.read_sst(path, self.schema, self.field_names)
+                remap = np.array([self.series.add_encoded(pk)
+                                  for pk in dict_values], dtype=np.int32)
+                codes = remap[indices]
+                t_ts = torch.as_tensor(ts).to(dev)
+                t_se = torch.as_tensor(codes).to(dev)
+                t_f = torch.as_tensor(np.ascontiguousarray(fields)).to(dev)
+                t_seq = torch.as_tensor(seq).to(dev)
             # re-sort by (series, ts, seq): codes may differ from write-time order
             ord1 = torch.argsort(t_seq, stable=True)
             ord2 = torch.argsort(t_ts[ord1], stable=True)
@@ -207,6 +224,43 @@ class Region:
                             batch.text_index[name] = ft.build_segment(
                                 list(arr), self.device)
             self.sst_cache[fid] = batch
+
+    def _load_sst_gpu(self, path: str):
+        """K11 open path: numeric columns decode straight into HBM; only
+        the (small) pk dictionary goes through pyarrow."""
+        import pyarrow as pa
+        import pyarrow.parquet as pq
+
+        from greptimedb_amd.engine import pagedec
+        dev = self.device
+        ts_name = self.schema.time_index.name
+        t_ts = pagedec.read_numeric_column(path, ts_name, dev)
+        t_seq = pagedec.read_numeric_column(path, "__sequence", dev)
+        n = t_ts.numel()
+        fparts = []
+        avail = {c.path_in_schema
+                 for c in (pq.read_metadata(path).row_group(0).column(i)
+                           for i in range(pq.read_metadata(path)
+                                          .row_group(0).num_columns))}
+        for fn in self.field_names:
+            if fn in avail:
+                fparts.append(pagedec.read_numeric_column(path, fn, dev))
+            else:
+                fparts.append(torch.full((n,), float("nan"),
+                                         dtype=torch.float64, device=dev))
+        t_f = torch.stack(fparts) if fparts else \
+            torch.zeros((0, n), dtype=torch.float64, device=dev)
+        pk = pq.read_table(path, columns=["__primary_key"]) \
+            .column("__primary_key").combine_chunks()
+        if isinstance(pk, pa.ChunkedArray):
+            pk = pk.chunk(0)
+        dict_values = [v.as_py() for v in pk.dictionary]
+        indices = pk.indices.to_numpy(zero_copy_only=False).astype(np.int32)
+        remap = np.array([self.series.add_encoded(b) for b in dict_values],
+                         dtype=np.int32)
+        codes_t = torch.as_tensor(remap).to(dev)[
+            torch.as_tensor(indices).to(dev).long()]
+        return t_ts, t_f, t_seq.to(torch.int64), codes_t
 
     # ---------------------------------------------------------------- write
 

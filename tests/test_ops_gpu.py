@@ -324,3 +324,46 @@ def test_gorilla_compressed_scan_gpu(tmp_path):
                        " GROUP BY hostname ORDER BY hostname").rows()
     assert after == before
     eng.close()
+
+
+@pytest.mark.gpu
+def test_region_reopen_uses_k11_gpu_decode(tmp_path, monkeypatch):
+    """Region open on GPU decodes numeric columns via pagedec (K11), and
+    the reopened data matches a CPU-reader reopen exactly."""
+    from greptimedb_amd.engine import pagedec
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.engine.ingest import Ingestor
+    from greptimedb_amd.models.tsbs import CpuWorkload
+    from greptimedb_amd.query.executor import Executor
+
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cuda:0",
+                                  background_flush=False))
+    ing = Ingestor(eng)
+    w = CpuWorkload(scale=50)
+    for _ in range(3):
+        ing.ingest_lines(w.next_batch(30000))
+    eng.flush_all()
+    expected = Executor(eng).execute(
+        "SELECT hostname, count(*) c, sum(usage_user) s FROM cpu"
+        " GROUP BY hostname ORDER BY hostname").rows()
+    eng.close()
+
+    calls = {"n": 0}
+    real = pagedec.read_numeric_column
+
+    def spy(*a, **k):
+        calls["n"] += 1
+        return real(*a, **k)
+
+    monkeypatch.setattr(pagedec, "read_numeric_column", spy)
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cuda:0",
+                                   background_flush=False))
+    assert calls["n"] > 0, "GPU reopen did not take the K11 path"
+    got = Executor(eng2).execute(
+        "SELECT hostname, count(*) c, sum(usage_user) s FROM cpu"
+        " GROUP BY hostname ORDER BY hostname").rows()
+    assert [(h, c) for h, c, _ in got] == [(h, c) for h, c, _ in expected]
+    for (_h1, _c1, s1), (_h2, _c2, s2) in zip(got, expected):
+        np.testing.assert_allclose(s1, s2, rtol=1e-12)
+    eng2.close()

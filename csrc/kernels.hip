@@ -739,15 +739,15 @@ __global__ void gorilla_decode_kernel(
     memcpy(&count, blk + 18, 2);
     const int val_mode = blk[20];   // 0 = XOR f64, 1 = scaled int delta
     const int scale_k = blk[21];
-    double inv_scale = 1.0;
-    for (int q = 0; q < scale_k; q++) inv_scale *= 0.1;
+    double scale = 1.0;                 // 10^k exact in f64 for k <= 4
+    for (int q = 0; q < scale_k; q++) scale *= 10.0;
     const uint8_t* ts_data = blk + 24;
     const int64_t rel = g - out_off[lo];
     if (rel >= count) continue;
     if (rel == 0) {
       out_ts[g] = base_ts;
       out_val[g] = (val_mode == 1)
-          ? (double)(int64_t)base_val * inv_scale
+          ? (double)(int64_t)base_val / scale
           : __longlong_as_double((long long)base_val);
       continue;
     }
@@ -781,7 +781,7 @@ __global__ void gorilla_decode_kernel(
     }
     if (val_mode == 1) {
       const int64_t dvz = (int64_t)(xv >> 1) ^ -(int64_t)(xv & 1);
-      out_val[g] = (double)((int64_t)base_val + dvz) * inv_scale;
+      out_val[g] = (double)((int64_t)base_val + dvz) / scale;
     } else {
       out_val[g] = __longlong_as_double((long long)(base_val ^ xv));
     }

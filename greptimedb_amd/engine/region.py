@@ -65,7 +65,6 @@ class Region:
         self.series = SeriesIndex([c.name for c in schema.tag_columns])
         os.makedirs(os.path.join(dir, "sst"), exist_ok=True)
         self.manifest = Manifest(os.path.join(dir, "manifest"))
-        self.memtable = Memtable(len(self.field_names), device=device)
         self.sst_cache: dict[str, sst_mod.SstBatch] = {}
         self.flushed_seq = self.manifest.flushed_seq
         self.last_seq = self.flushed_seq
@@ -92,7 +91,11 @@ class Region:
         self._series_log_path = os.path.join(dir, "series.log")
         self._load_series_log()
         self._series_log = open(self._series_log_path, "ab")
-        self._load_strcols_meta()   # restore str/fulltext column layout
+        self._load_strcols_meta()   # restore field/str/fulltext layout
+        # memtable AFTER the meta sidecar: auto-ALTERed field columns (the
+        # influx auto-create path evolves the schema without DDL) must be
+        # known before sizing the field matrix and reading SSTs
+        self.memtable = Memtable(len(self.field_names), device=device)
         self._load_ssts()
         self.mem_base = self.row_seq
 
@@ -375,7 +378,8 @@ class Region:
                     changed = True
                 if fulltext and n not in self.text_cols:
                     ft = FulltextColumn()
-                    ft.mem.n_rows = self.memtable.len
+                    mem = getattr(self, "memtable", None)  # None during open
+                    ft.mem.n_rows = mem.len if mem is not None else 0
                     self.text_cols[n] = ft
                     changed = True
             if changed:
@@ -385,10 +389,19 @@ class Region:
         return os.path.join(self.dir, "strcols.json")
 
     def _save_strcols_meta(self):
+        """Region column-layout sidecar: dynamic numeric fields (influx
+        auto-ALTER) + string/fulltext columns, so a reopen restores the
+        exact layout even after the WAL entries that introduced the
+        columns were purged (round-2 fix: previously only str cols)."""
         import json as _json
         tmp = self._strcols_meta_path() + ".tmp"
         with open(tmp, "w") as f:
-            _json.dump({n: (n in self.text_cols) for n in self.str_field_names}, f)
+            _json.dump({
+                "version": 2,
+                "fields": list(self.field_names),
+                "strcols": {n: (n in self.text_cols)
+                            for n in self.str_field_names},
+            }, f)
         os.replace(tmp, self._strcols_meta_path())
 
     def _load_strcols_meta(self):
@@ -398,8 +411,15 @@ class Region:
             return
         with open(path) as f:
             meta = _json.load(f)
-        plain = [n for n, ft in meta.items() if not ft]
-        fts = [n for n, ft in meta.items() if ft]
+        if isinstance(meta, dict) and meta.get("version") == 2:
+            strmap = meta.get("strcols", {})
+            for fn in meta.get("fields", []):
+                if fn not in self.field_names:
+                    self.field_names.append(fn)
+        else:   # round-1 format: {strcol: fulltext}
+            strmap = meta
+        plain = [n for n, ft in strmap.items() if not ft]
+        fts = [n for n, ft in strmap.items() if ft]
         if plain:
             self.ensure_str_fields(plain, fulltext=False)
         if fts:
@@ -407,13 +427,15 @@ class Region:
 
     def ensure_fields(self, names: list[str]):
         """Auto-ALTER: add new field columns (reference insert.rs:562
-        create_or_alter tables on demand)."""
+        create_or_alter tables on demand). Layout persists in the region
+        meta sidecar (reopen-safe once the WAL is purged)."""
         new = [n for n in names if n not in self.field_names]
         if not new:
             return
         with self.lock:
             self.field_names.extend(new)
             self.memtable.add_fields(len(new))
+            self._save_strcols_meta()
 
     # ---------------------------------------------------------------- scan
 

@@ -464,12 +464,49 @@ class _Select:
         return {k: v for k, v in row.items() if k not in self.keys}
 
 
+class _Cmcd:
+    """CMCD (Common Media Client Data, CTA-5004): comma-separated key=value
+    pairs, bare keys = boolean true, quoted strings unquoted (reference:
+    src/pipeline processor/cmcd.rs)."""
+
+    def __init__(self, cfg):
+        self.fields = _fields_of(cfg)
+
+    def __call__(self, row):
+        for src, dst in self.fields:
+            v = row.get(src)
+            if v is None:
+                continue
+            prefix = (dst or src)
+            for part in str(v).split(","):
+                part = part.strip()
+                if not part:
+                    continue
+                if "=" in part:
+                    k, val = part.split("=", 1)
+                    val = val.strip()
+                    if val.startswith('"') and val.endswith('"'):
+                        out = val[1:-1]
+                    else:
+                        try:
+                            out = int(val)
+                        except ValueError:
+                            try:
+                                out = float(val)
+                            except ValueError:
+                                out = val
+                else:
+                    k, out = part, True
+                row[f"{prefix}_{k.strip()}"] = out
+        return row
+
+
 _PROCESSORS = {
     "dissect": _Dissect, "regex": _Regex, "date": _Date, "epoch": _Epoch,
     "gsub": _Gsub, "letter": _Letter, "csv": _Csv, "json_parse": _JsonParse,
     "json_path": _JsonPath, "simple_extract": _SimpleExtract, "join": _Join,
     "urlencoding": _UrlEncoding, "decolorize": _Decolorize, "digest": _Digest,
-    "filter": _Filter, "select": _Select,
+    "filter": _Filter, "select": _Select, "cmcd": _Cmcd,
 }
 
 _NUM_TYPES = {"int8", "int16", "int32", "int64", "uint8", "uint16", "uint32",

@@ -309,3 +309,8 @@ class DropDatabase:
 @dataclass
 class ShowVariables:
     like: str | None = None
+
+
+@dataclass
+class Kill:
+    pid: int

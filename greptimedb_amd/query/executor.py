@@ -164,9 +164,22 @@ class Executor:
         from greptimedb_amd.utils.tracing import tracer
         stmt = parse_sql(sql)
         t0 = _time.perf_counter()
-        with tracer.span("sql.execute", statement=sql[:200],
-                         stmt_type=type(stmt).__name__):
-            r = self.execute_stmt(stmt)
+        # process registry (reference: catalog process_manager + KILL)
+        plist = getattr(self.engine, "process_list", None)
+        if plist is None:
+            plist = self.engine.process_list = {}
+        pid = getattr(self.engine, "_next_pid", 1)
+        self.engine._next_pid = pid + 1
+        entry = {"sql": sql, "start": t0, "elapsed_ms": 0.0,
+                 "state": "running", "cancel": False}
+        plist[pid] = entry
+        try:
+            with tracer.span("sql.execute", statement=sql[:200],
+                             stmt_type=type(stmt).__name__):
+                r = self.execute_stmt(stmt)
+        finally:
+            entry["elapsed_ms"] = (_time.perf_counter() - t0) * 1000
+            plist.pop(pid, None)
         dt = (_time.perf_counter() - t0) * 1000
         if dt >= self.SLOW_QUERY_MS:
             # slow-query log (reference: common/frontend slow query events)
@@ -185,6 +198,14 @@ class Executor:
             if stmt.schema not in schemas:
                 raise TableNotFound(f"database {stmt.schema}")
             self.session.schema = stmt.schema
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.Kill):
+            plist = getattr(self.engine, "process_list", {})
+            entry = plist.get(stmt.pid)
+            if entry is None:
+                raise InvalidArguments(f"no such query id {stmt.pid}")
+            entry["cancel"] = True
+            entry["state"] = "cancelled"
             return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.SetVar):
             self.session.set_var(stmt.name, stmt.value)

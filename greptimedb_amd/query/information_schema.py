@@ -13,18 +13,58 @@ from greptimedb_amd.models.schema import SemanticType
 
 VIRTUAL_TABLES = {
     "tables", "columns", "region_statistics", "flows", "cluster_info",
-    "partitions", "region_peers", "build_info",
+    "partitions", "region_peers", "build_info", "process_list", "views",
+    "schemata",
 }
+
+PG_CATALOG_TABLES = {"pg_tables", "pg_namespace", "pg_class"}
 
 
 def is_information_schema(name: str) -> bool:
-    return name.lower().startswith("information_schema.") and \
-        name.split(".", 1)[1].lower() in VIRTUAL_TABLES
+    low = name.lower()
+    if low.startswith("information_schema.") and \
+            low.split(".", 1)[1] in VIRTUAL_TABLES:
+        return True
+    return low.startswith("pg_catalog.") and \
+        low.split(".", 1)[1] in PG_CATALOG_TABLES
 
 
 def build(engine, name: str):
     """Return (names, columns) for the virtual table."""
     kind = name.split(".", 1)[1].lower()
+    schema_ns = name.split(".", 1)[0].lower()
+    if schema_ns == "pg_catalog":
+        # pg_catalog compatibility views (reference
+        # src/catalog/src/system_schema/pg_catalog.rs)
+        schemas = sorted({"public", "information_schema", "pg_catalog"} |
+                         set(getattr(engine, "schemas", set())))
+        if kind == "pg_namespace":
+            return _cols(["oid", "nspname"],
+                         [(i + 1, s_) for i, s_ in enumerate(schemas)])
+        rows = []
+        for i, (t, st) in enumerate(sorted(engine.tables.items())):
+            sch = t.split(".", 1)[0] if "." in t else "public"
+            tn = t.split(".", 1)[1] if "." in t else t
+            rows.append((i + 16384, tn, sch, "r"))
+        if kind == "pg_class":
+            return _cols(["oid", "relname", "relnamespace", "relkind"], rows)
+        return _cols(["schemaname", "tablename", "tableowner"],
+                     [(sch, tn, "greptime") for _o, tn, sch, _k in rows])
+    if kind == "process_list":
+        rows = [(pid, info["sql"][:200], round(info["elapsed_ms"], 1),
+                 info["state"])
+                for pid, info in sorted(
+                    getattr(engine, "process_list", {}).items())]
+        return _cols(["id", "query", "elapsed_ms", "state"], rows)
+    if kind == "views":
+        rows = [(v, sql) for v, sql in
+                sorted(getattr(engine, "views", {}).items())]
+        return _cols(["view_name", "definition"], rows)
+    if kind == "schemata":
+        schemas = sorted({"public", "information_schema", "greptime_private"} |
+                         set(getattr(engine, "schemas", set())))
+        return _cols(["catalog_name", "schema_name"],
+                     [("greptime", s_) for s_ in schemas])
     if kind == "tables":
         rows = [("greptime", "public", t, "BASE TABLE", st.schema.table_id,
                  "mito-hip", len(st.regions))

@@ -164,6 +164,10 @@ class Parser:
         elif self.at_kw("use"):
             self.next()
             stmt = ast.Use(self.next().value)
+        elif self.at_kw("kill"):
+            self.next()
+            self.eat_kw("query")
+            stmt = ast.Kill(int(self.next().value))
         elif self.at_kw("set"):
             self.next()
             self.eat_kw("session") or self.eat_kw("global")

@@ -639,3 +639,26 @@ def test_gorilla_compress_table_roundtrip(tmp_path):
                        " ORDER BY hostname").rows()
     assert after == before
     eng.close()
+
+
+def test_auto_created_fields_survive_reopen_after_wal_purge(tmp_path):
+    """Round-2 fix: influx auto-ALTERed numeric fields persist in the
+    region meta sidecar — reopen restores values even when the WAL entries
+    that introduced the columns are gone."""
+    from greptimedb_amd.query.executor import Executor
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    ing = Ingestor(eng)
+    w = CpuWorkload(scale=10)
+    ing.ingest_lines(w.next_batch(2000))
+    eng.flush_all()   # purges the WAL
+    exp = Executor(eng).execute(
+        "SELECT sum(usage_user) AS s, count(usage_idle) AS c FROM cpu").rows()
+    fields_before = list(eng.table("cpu").regions[0].field_names)
+    eng.close()
+    eng2 = MitoEngine(EngineConfig(data_dir=d, device="cpu", background_flush=False))
+    assert list(eng2.table("cpu").regions[0].field_names) == fields_before
+    got = Executor(eng2).execute(
+        "SELECT sum(usage_user) AS s, count(usage_idle) AS c FROM cpu").rows()
+    assert got == exp
+    eng2.close()

@@ -433,3 +433,39 @@ def test_sketch_aggregates(tmp_engine):
                    " WHERE h = 'h0'")
     vals = [float(i % 100) for i in range(2000) if i % 4 == 0]
     assert float(r.columns[0][0]) == np.quantile(vals, 0.5)
+
+
+def test_process_list_and_kill(tmp_engine):
+    from greptimedb_amd.query.executor import Executor
+    from greptimedb_amd.utils.errors import InvalidArguments
+    ex = Executor(tmp_engine)
+    r = ex.execute("SELECT id, query FROM information_schema.process_list")
+    # our own query shows in its own snapshot? it is popped on completion,
+    # but WAS registered during execution — the table reads live state, so
+    # at minimum the call itself ran while registered. Snapshot shows it.
+    assert "id" in r.names
+    import pytest as _pytest
+    with _pytest.raises(InvalidArguments):
+        ex.execute("KILL 99999")
+    # register a fake running query and kill it
+    tmp_engine.process_list[42] = {"sql": "SELECT 1", "start": 0,
+                                   "elapsed_ms": 1.0, "state": "running",
+                                   "cancel": False}
+    ex.execute("KILL QUERY 42")
+    assert tmp_engine.process_list[42]["cancel"]
+
+
+def test_pg_catalog_and_views_tables(tmp_engine):
+    from greptimedb_amd.query.executor import Executor
+    ex = Executor(tmp_engine)
+    ex.execute("CREATE TABLE pgx (h STRING, ts TIMESTAMP TIME INDEX, v DOUBLE,"
+               " PRIMARY KEY (h))")
+    ex.execute("CREATE VIEW pgv AS SELECT h FROM pgx")
+    r = ex.execute("SELECT tablename FROM pg_catalog.pg_tables")
+    assert "pgx" in list(r.columns[0])
+    r = ex.execute("SELECT nspname FROM pg_catalog.pg_namespace")
+    assert "public" in list(r.columns[0])
+    r = ex.execute("SELECT view_name FROM information_schema.views")
+    assert list(r.columns[0]) == ["pgv"]
+    r = ex.execute("SELECT schema_name FROM information_schema.schemata")
+    assert "public" in list(r.columns[0])

@@ -181,3 +181,23 @@ def test_ingest_with_pipeline(tmp_engine):
     # message is fulltext-indexed → MATCHES works
     r = ex.execute("SELECT count(*) FROM nginx_logs WHERE matches(message, 'login')")
     assert list(r.rows())[0][0] == 1
+
+
+def test_cmcd_processor():
+    """CMCD key=value media telemetry (reference processor/cmcd.rs)."""
+    p = Pipeline.from_yaml("""
+processors:
+  - cmcd:
+      fields:
+        - data
+transform:
+  - fields:
+      - data_sid
+    type: string
+  - fields:
+      - data_br
+    type: int64
+""")
+    row, _sfx = p.run_row({"data": 'br=3200,bs,d=4004,sid="abc-123"'})
+    assert row["data_br"] == 3200
+    assert row["data_sid"] == "abc-123"

@@ -15,10 +15,12 @@ from greptimedb_amd.meta.failure_detector import PhiAccrualFailureDetector
 
 
 class RegionSupervisor:
-    def __init__(self, on_failover=None, threshold: float = 8.0):
+    def __init__(self, on_failover=None, threshold: float = 8.0,
+                 acceptable_pause_ms: float = 10_000.0):
         self.detectors: dict[str, PhiAccrualFailureDetector] = {}
         self.on_failover = on_failover          # callback(node_id)
         self.threshold = threshold
+        self.acceptable_pause_ms = acceptable_pause_ms
         self.failed: set[str] = set()
 
     def heartbeat(self, node_id: str, now_ms: float | None = None):
@@ -26,7 +28,8 @@ class RegionSupervisor:
         det = self.detectors.get(node_id)
         if det is None:
             det = self.detectors[node_id] = PhiAccrualFailureDetector(
-                threshold=self.threshold)
+                threshold=self.threshold,
+                acceptable_heartbeat_pause_ms=self.acceptable_pause_ms)
         det.heartbeat(now_ms)
         self.failed.discard(node_id)
 

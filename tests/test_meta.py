@@ -179,3 +179,53 @@ def test_event_recorder(tmp_path):
     payloads = [_json.loads(row[1]) for row in r.rows()]
     assert any(p.get("table") == "evt_src" for p in payloads)
     eng.close()
+
+
+def test_wire_heartbeat_and_mailbox(tmp_path):
+    """Heartbeat/mailbox over the wire (reference datanode heartbeat.rs +
+    mailbox_handler.rs): stats flow up, instructions piggyback down, and a
+    stopped node trips the φ detector."""
+    import socket
+    import time as _time
+
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.engine.ingest import Ingestor
+    from greptimedb_amd.meta.heartbeat import HeartbeatTask, MetaServer
+    from greptimedb_amd.models.tsbs import CpuWorkload
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    failed = []
+    srv = MetaServer(port, on_failover=failed.append, threshold=1.5,
+                     acceptable_pause_ms=150.0)
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d"), device="cpu",
+                                  background_flush=False, default_regions=2))
+    Ingestor(eng).ingest_lines(CpuWorkload(scale=4).next_batch(100))
+    hb = HeartbeatTask("gpu0", eng, srv.port, interval_s=0.05)
+    # regular beats → stats visible, no failover
+    for _ in range(8):
+        hb.beat_once()
+        _time.sleep(0.02)
+    assert "gpu0" in srv.stats
+    assert sum(r["rows"] for r in srv.stats["gpu0"]["regions"]) == 100
+    assert srv.check() == []
+    # mailbox: downgrade instruction applies on the next beat
+    srv.send_instruction("gpu0", {"kind": "downgrade_region",
+                                  "table": "cpu", "region_idx": 0})
+    got = hb.beat_once()
+    assert got and got[0]["kind"] == "downgrade_region"
+    assert not eng.table("cpu").regions[0].writable
+    srv.send_instruction("gpu0", {"kind": "upgrade_region",
+                                  "table": "cpu", "region_idx": 0})
+    hb.beat_once()
+    assert eng.table("cpu").regions[0].writable
+    # silence → φ exceeds threshold → failover fires exactly once
+    _time.sleep(0.6)
+    newly = srv.check()
+    assert newly == ["gpu0"] and failed == ["gpu0"]
+    assert srv.check() == []
+    hb.stop()
+    srv.close()
+    eng.close()

@@ -628,26 +628,49 @@ class PromEvaluator:
             return PromMatrix([], torch.zeros((0, T), dtype=torch.float64,
                                               device=device), grid)
         # merge per-region factor columns (remap value ids into a shared
-        # vocabulary; vectorized over unique values, not series)
+        # vocabulary). pd.factorize hash-codes the concatenated vocabs in C
+        # (the python setdefault loop here cost ~1.3s at 4M series); the
+        # merged columns cache on the engine until the series count moves.
+        cache = getattr(self.engine, "_prom_label_cache", None)
+        if cache is None:
+            cache = self.engine._prom_label_cache = {}
         label_cols: dict = {}
         for t in all_tags:
-            vocab: dict = {}
+            ck = (st.schema.name, t, S)
+            hit = cache.get(ck)
+            if hit is not None:
+                label_cols[t] = hit
+                continue
+            import pandas as pd
+            by_region = {ri: (c, v) for ri, c, v in col_parts[t]}
+            vocab_arrays = []
+            for ri, sz in enumerate(region_sizes):
+                got = by_region.get(ri)
+                vocab_arrays.append(
+                    np.asarray(got[1], dtype=object) if got is not None
+                    else np.empty(0, dtype=object))
+            concat = np.concatenate(vocab_arrays) if vocab_arrays else                 np.empty(0, dtype=object)
+            codes_flat, uniq = pd.factorize(concat)
+            codes_flat = codes_flat.astype(np.int32)
             out_codes = np.full(S, -1, dtype=np.int32)
             off = 0
-            by_region = {ri: (c, v) for ri, c, v in col_parts[t]}
+            voff = 0
             for ri, sz in enumerate(region_sizes):
                 got = by_region.get(ri)
                 if got is not None:
                     c, vals = got
-                    remap = np.fromiter((vocab.setdefault(v, len(vocab)) for v in vals),
-                                        dtype=np.int32, count=len(vals))
+                    remap = codes_flat[voff:voff + len(vals)]
+                    voff += len(vals)
                     cc = c.astype(np.int64)
                     res = np.full(sz, -1, dtype=np.int32)
                     has = cc >= 0
                     res[has] = remap[cc[has]]
                     out_codes[off:off + sz] = res
                 off += sz
-            label_cols[t] = (out_codes, list(vocab))
+            label_cols[t] = (out_codes, list(uniq))
+            cache[ck] = label_cols[t]
+            if len(cache) > 64:
+                cache.pop(next(iter(cache)))
         if "__name__" not in label_cols:
             label_cols["__name__"] = (np.zeros(S, dtype=np.int32),
                                       [sel.metric or st.schema.name])

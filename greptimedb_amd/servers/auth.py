@@ -173,3 +173,15 @@ def scram_client_messages(user: str, password: str, server_first_fn):
     server_key = _hmac(salted, b"Server Key")
     expected_v = "v=" + _b64.b64encode(_hmac(server_key, auth_message)).decode()
     return client_first, client_final, expected_v
+
+
+def mysql_caching_sha2_check(password: str, nonce: bytes, token: bytes) -> bool:
+    """caching_sha2_password fast-auth scramble (MySQL 8 default):
+    token = XOR(SHA256(pwd), SHA256(SHA256(SHA256(pwd)) || nonce))."""
+    if not token:
+        return False
+    p1 = _hashlib.sha256(password.encode()).digest()
+    p2 = _hashlib.sha256(p1).digest()
+    expected = bytes(a ^ b for a, b in zip(
+        p1, _hashlib.sha256(p2 + nonce).digest()))
+    return hmac.compare_digest(expected, token)

@@ -109,24 +109,41 @@ class MySQLServer:
             else:
                 user = ""
             if self.user_provider is not None:
-                from greptimedb_amd.servers.auth import (mysql_native_check,
-                                                         password_of)
+                from greptimedb_amd.servers.auth import (
+                    mysql_caching_sha2_check, mysql_native_check, password_of)
                 ok = self.user_provider.allow(user)
                 stored = password_of(self.user_provider, user)
                 if ok and stored:
-                    # verify the mysql_native_password scramble response
+                    # auth token + (optional) client auth plugin name
                     token = b""
+                    plugin = b"mysql_native_password"
                     if user_end > 0 and user_end + 1 < len(resp):
                         tl = resp[user_end + 1]
                         token = resp[user_end + 2: user_end + 2 + tl]
-                    ok = mysql_native_check(stored, b"12345678123456789012",
-                                            token)
+                        rest = resp[user_end + 2 + tl:]
+                        # optional database name, then plugin name (both NUL)
+                        caps = struct.unpack("<I", resp[:4])[0]
+                        if caps & CLIENT_CONNECT_WITH_DB and b"\x00" in rest:
+                            rest = rest.split(b"\x00", 1)[1]
+                        if rest:
+                            plugin = rest.split(b"\x00", 1)[0] or plugin
+                    nonce = b"12345678123456789012"
+                    if plugin == b"caching_sha2_password":
+                        ok = mysql_caching_sha2_check(stored, nonce, token)
+                    else:
+                        ok = mysql_native_check(stored, nonce, token)
                 if not ok:
                     writer.write(self._err(seq + 1, f"access denied for {user}", 1045))
                     await writer.drain()
                     writer.close()
                     return
-            writer.write(self._ok(seq + 1))
+            if self.user_provider is not None and 'plugin' in dir() and \
+                    plugin == b"caching_sha2_password":
+                # fast-auth success marker precedes OK (MySQL 8 protocol)
+                writer.write(self._packet(seq + 1, b"\x01\x03"))
+                writer.write(self._ok(seq + 2))
+            else:
+                writer.write(self._ok(seq + 1))
             await writer.drain()
 
             stmts: dict = {}     # stmt_id → {"sql", "nparams", "types"}

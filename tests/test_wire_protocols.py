@@ -430,3 +430,43 @@ def test_postgres_scram_auth(ex):
 
     assert asyncio.run(run("s3cret"))
     assert asyncio.run(run("wrongpw"))
+
+
+def test_mysql_caching_sha2_auth(ex):
+    """caching_sha2_password fast-auth (MySQL 8 default plugin)."""
+    import hashlib
+    import struct as _st
+
+    from greptimedb_amd.servers.auth import StaticUserProvider
+    from greptimedb_amd.servers.mysql import MySQLServer
+
+    async def run(password):
+        provider = StaticUserProvider({"bob": "pw123"})
+        srv = MySQLServer(ex, "127.0.0.1", 0, provider)
+        s = await srv.start()
+        port = s.sockets[0].getsockname()[1]
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        _seq, greeting = await _mysql_read_packet(reader)
+        nonce = b"12345678123456789012"
+        p1 = hashlib.sha256(password.encode()).digest()
+        p2 = hashlib.sha256(p1).digest()
+        token = bytes(a ^ b for a, b in zip(
+            p1, hashlib.sha256(p2 + nonce).digest()))
+        caps = 0x00000200 | 0x00080000 | 0x00000008  # proto41|plugin_auth|connect_with_db
+        body = (_st.pack("<I", caps) + _st.pack("<I", 1 << 24) + bytes([33]) +
+                b"\x00" * 23 + b"bob\x00" + bytes([len(token)]) + token +
+                b"testdb\x00" + b"caching_sha2_password\x00")
+        writer.write(_st.pack("<I", len(body) | (1 << 24))[:3] + bytes([1]) + body)
+        await writer.drain()
+        _seq2, more = await _mysql_read_packet(reader)
+        ok_pkt = None
+        if more[:1] == b"\x01":       # fast-auth success marker
+            _seq3, ok_pkt = await _mysql_read_packet(reader)
+        else:
+            ok_pkt = more
+        writer.close()
+        s.close()
+        return ok_pkt[:1]
+
+    assert asyncio.run(run("pw123")) == b"\x00"       # OK
+    assert asyncio.run(run("wrong")) == b"\xff"       # ERR

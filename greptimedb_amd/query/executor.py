@@ -1174,17 +1174,34 @@ class Executor:
             need[a].add(c)
             order_resolved.append((a, c, desc))
 
-        # execute both sides as raw scans
+        # execute sides as raw scans — build (right) side first
         data = {}
-        for a in aliases:
+        la, ra = aliases
+
+        def run_side(a):
             w = None
             for c in side_where[a]:
                 w = c if w is None else ast.BinOp("and", w, c)
             sub = ast.Select([(ast.Col(c), None) for c in sorted(need[a])],
                              sides[a], where=w)
             r = self.execute_stmt(sub)
-            data[a] = {n: np.asarray(col) for n, col in zip(r.names, r.columns)}
-        la, ra = aliases
+            data[a] = {n: np.asarray(col)
+                       for n, col in zip(r.names, r.columns)}
+
+        run_side(ra)
+        # C10 dynamic filter (reference dist_plan remote_dyn_filter_*): on
+        # INNER joins push the build side's distinct key values into the
+        # probe side's scan predicate — in distributed mode the gathered
+        # build keys act as the broadcast filter, shrinking every rank's
+        # probe materialization before the gather.
+        if j.kind == "inner" and len(pairs) == 1:
+            rvals = data[ra][pairs[0][1]]
+            uniq = {v for v in rvals.tolist() if v is not None}
+            if len(uniq) <= 65536:
+                side_where[la].append(ast.InList(
+                    ast.Col(pairs[0][0]),
+                    [ast.Lit(v) for v in sorted(uniq, key=str)]))
+        run_side(la)
 
         def keys_of(a, cols):
             arrs = [data[a][c] for c in cols]

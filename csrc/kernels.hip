@@ -238,6 +238,149 @@ __global__ void field_bucket_agg_kernel(
   }
 }
 
+// ---------------------------------------------- K5 v3: LDS-privatized tiles
+//
+// PMC (FETCH_SIZE) showed the chunked-run kernels fetching 8-10x the input
+// bytes: per-thread contiguous chunks make every wavefront touch 64 cache
+// lines per step (uncoalesced). v3 restores coalescing — threads stride the
+// tile contiguously — and privatizes the accumulator table in LDS: a tile
+// of sorted rows spans only a few (slot,bucket) cells, so per-element LDS
+// atomics replace global ones, flushed once per tile. Tiles whose cell
+// range exceeds the LDS table fall back to global atomics (still with
+// coalesced loads).
+
+__global__ void bucket_cell_coalesced_kernel(
+    const int64_t* __restrict__ ts,
+    const int32_t* __restrict__ series,
+    const int32_t* __restrict__ slot_lut, int lut_size,
+    int64_t ts_lo, int64_t ts_hi, int64_t origin, int64_t bucket_ms,
+    int n_buckets, int64_t n,
+    int32_t* __restrict__ cell) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const int64_t t = ts[i];
+    const int32_t sId = series[i];
+    int32_t cl = -1;
+    if (t >= ts_lo && t < ts_hi && sId >= 0 && sId < lut_size) {
+      const int32_t slot = slot_lut[sId];
+      if (slot >= 0) {
+        const int64_t b = (t - origin) / bucket_ms;
+        if (b >= 0 && b < n_buckets) cl = slot * n_buckets + (int32_t)b;
+      }
+    }
+    cell[i] = cl;
+  }
+}
+
+#define AGG_TILE 8192
+#define LDS_CELLS 1024
+
+__global__ void field_bucket_agg_lds_kernel(
+    const int32_t* __restrict__ cell,
+    const double* __restrict__ fields, int64_t field_stride,
+    const int32_t* __restrict__ field_idx, int nf,
+    int64_t n, int64_t cells_per_field,
+    double* __restrict__ out_sum,
+    unsigned long long* __restrict__ out_cnt,
+    unsigned long long* __restrict__ out_min,
+    unsigned long long* __restrict__ out_max,
+    unsigned long long* __restrict__ out_rows) {
+  __shared__ int32_t s_lo, s_hi;
+  __shared__ double s_sum[LDS_CELLS];
+  __shared__ unsigned long long s_cnt[LDS_CELLS];
+  __shared__ unsigned long long s_min[LDS_CELLS];
+  __shared__ unsigned long long s_max[LDS_CELLS];
+  const int tid = threadIdx.x;
+  const int nthreads = blockDim.x;
+  const int64_t ntiles = (n + AGG_TILE - 1) / AGG_TILE;
+  for (int64_t tile = blockIdx.x; tile < ntiles; tile += gridDim.x) {
+    const int64_t t0 = tile * AGG_TILE;
+    const int64_t t1 = min(t0 + (int64_t)AGG_TILE, n);
+    if (tid == 0) { s_lo = INT32_MAX; s_hi = -1; }
+    __syncthreads();
+    int32_t my_lo = INT32_MAX, my_hi = -1;
+    for (int64_t i = t0 + tid; i < t1; i += nthreads) {
+      const int32_t c = cell[i];
+      if (c >= 0) { my_lo = min(my_lo, c); my_hi = max(my_hi, c); }
+    }
+    if (my_hi >= 0) {
+      atomicMin(&s_lo, my_lo);
+      atomicMax(&s_hi, my_hi);
+    }
+    __syncthreads();
+    const int32_t lo = s_lo, hi = s_hi;
+    __syncthreads();
+    if (hi < 0) continue;                       // tile fully filtered out
+    const int range = hi - lo + 1;
+    if (range <= LDS_CELLS) {
+      // rows count once per tile (reuse s_cnt as the rows table)
+      for (int k = tid; k < range; k += nthreads) s_cnt[k] = 0ULL;
+      __syncthreads();
+      for (int64_t i = t0 + tid; i < t1; i += nthreads) {
+        const int32_t c = cell[i];
+        if (c >= 0) atomicAdd(&s_cnt[c - lo], 1ULL);
+      }
+      __syncthreads();
+      for (int k = tid; k < range; k += nthreads)
+        if (s_cnt[k]) atomicAdd(&out_rows[lo + k], s_cnt[k]);
+      __syncthreads();
+      for (int f = 0; f < nf; f++) {
+        const double* __restrict__ col =
+            fields + (int64_t)field_idx[f] * field_stride;
+        for (int k = tid; k < range; k += nthreads) {
+          s_sum[k] = 0.0; s_cnt[k] = 0ULL; s_min[k] = ~0ULL; s_max[k] = 0ULL;
+        }
+        __syncthreads();
+        for (int64_t i = t0 + tid; i < t1; i += nthreads) {
+          const int32_t c = cell[i];
+          if (c < 0) continue;
+          const double v = col[i];
+          if (isnan(v)) continue;
+          const int k = c - lo;
+          atomicAdd(&s_sum[k], v);
+          atomicAdd(&s_cnt[k], 1ULL);
+          const uint64_t key = f64_to_key(v);
+          atomicMin(&s_min[k], (unsigned long long)key);
+          atomicMax(&s_max[k], (unsigned long long)key);
+        }
+        __syncthreads();
+        const int64_t base = (int64_t)f * cells_per_field;
+        for (int k = tid; k < range; k += nthreads) {
+          if (!s_cnt[k]) continue;
+          atomicAdd(&out_sum[base + lo + k], s_sum[k]);
+          atomicAdd(&out_cnt[base + lo + k], s_cnt[k]);
+          atomicMin(&out_min[base + lo + k], s_min[k]);
+          atomicMax(&out_max[base + lo + k], s_max[k]);
+        }
+        __syncthreads();
+      }
+    } else {
+      // wide tile (unsorted data): coalesced loads, global atomics
+      for (int64_t i = t0 + tid; i < t1; i += nthreads) {
+        const int32_t c = cell[i];
+        if (c >= 0) atomicAdd(&out_rows[c], 1ULL);
+      }
+      for (int f = 0; f < nf; f++) {
+        const double* __restrict__ col =
+            fields + (int64_t)field_idx[f] * field_stride;
+        const int64_t base = (int64_t)f * cells_per_field;
+        for (int64_t i = t0 + tid; i < t1; i += nthreads) {
+          const int32_t c = cell[i];
+          if (c < 0) continue;
+          const double v = col[i];
+          if (isnan(v)) continue;
+          atomicAdd(&out_sum[base + c], v);
+          atomicAdd(&out_cnt[base + c], 1ULL);
+          const uint64_t key = f64_to_key(v);
+          atomicMin(&out_min[base + c], (unsigned long long)key);
+          atomicMax(&out_max[base + c], (unsigned long long)key);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 // ------------------------------------------------- K-lastpoint (series last)
 // Monotonic i64→u64 key so unsigned atomicMax orders signed timestamps.
 DEV_INLINE uint64_t i64_to_key(int64_t v) {
@@ -596,22 +739,19 @@ void launch_bucket_agg2(
     double* out_sum, unsigned long long* out_cnt,
     unsigned long long* out_min, unsigned long long* out_max,
     unsigned long long* out_rows, hipStream_t stream) {
-  // chunk sized so total phase-B threads ≈ 512k (fills 256 CUs), min 32
-  int chunk = (int)max((int64_t)32, (n * max(nf, 1)) / (512 * 1024));
+  // v3 (PMC-guided): coalesced cell pass + LDS-privatized tile aggregation
+  hipLaunchKernelGGL(bucket_cell_coalesced_kernel,
+      dim3(grid_for(n, 256)), dim3(256), 0, stream,
+      ts, series, slot_lut, lut_size, ts_lo, ts_hi, origin, bucket_ms,
+      n_buckets, n, cell_buf);
   {
-    const int64_t nchunks = (n + chunk - 1) / chunk;
-    hipLaunchKernelGGL(bucket_cell_kernel,
-        dim3(grid_for(nchunks, 256)), dim3(256), 0, stream,
-        ts, series, slot_lut, lut_size, ts_lo, ts_hi, origin, bucket_ms,
-        n_buckets, n, cell_buf, out_rows, chunk);
-  }
-  if (nf > 0) {
-    const int64_t nchunks = (n + chunk - 1) / chunk;
-    hipLaunchKernelGGL(field_bucket_agg_kernel,
-        dim3(grid_for(nchunks * nf, 256)), dim3(256), 0, stream,
+    const int64_t ntiles = (n + AGG_TILE - 1) / AGG_TILE;
+    const int grid = (int)min(ntiles, (int64_t)8192);
+    hipLaunchKernelGGL(field_bucket_agg_lds_kernel,
+        dim3(grid), dim3(256), 0, stream,
         cell_buf, fields, field_stride, field_idx, nf, n,
-        (int64_t)n_slots * n_buckets, chunk,
-        out_sum, out_cnt, out_min, out_max);
+        (int64_t)n_slots * n_buckets,
+        out_sum, out_cnt, out_min, out_max, out_rows);
   }
 }
 

@@ -198,6 +198,9 @@ static py::tuple parse_column_chunk(py::bytes chunk, int codec,
   std::vector<int64_t> pages;
   int64_t dict_off = -1, dict_len = -1;
 
+  // pure C++ from here (thrift walk + zstd) — release the GIL so region
+  // open can parse many columns in parallel threads
+  py::gil_scoped_release nogil;
   while (p < end) {
     TReader r{p, end};
     PageHeaderFields h = parse_page_header(r);
@@ -286,6 +289,7 @@ static py::tuple parse_column_chunk(py::bytes chunk, int codec,
     p = body + h.compressed_size;
   }
 
+  py::gil_scoped_acquire gil;
   py::array_t<int64_t> parr({(py::ssize_t)(pages.size() / 4), (py::ssize_t)4});
   if (!pages.empty())
     memcpy(parr.mutable_data(), pages.data(), pages.size() * 8);

@@ -237,20 +237,24 @@ class Region:
         from greptimedb_amd.engine import pagedec
         dev = self.device
         ts_name = self.schema.time_index.name
-        t_ts = pagedec.read_numeric_column(path, ts_name, dev)
-        t_seq = pagedec.read_numeric_column(path, "__sequence", dev)
+        md0 = pq.read_metadata(path).row_group(0)
+        avail = {md0.column(i).path_in_schema
+                 for i in range(md0.num_columns)}
+        # host page parse + zstd release the GIL (csrc/pagedec.cpp) →
+        # decode all numeric columns in parallel threads
+        from concurrent.futures import ThreadPoolExecutor
+        want = [ts_name, "__sequence"] + [fn for fn in self.field_names
+                                          if fn in avail]
+        with ThreadPoolExecutor(max_workers=min(8, max(len(want), 1))) as tp:
+            got = dict(zip(want, tp.map(
+                lambda c: pagedec.read_numeric_column(path, c, dev), want)))
+        t_ts = got[ts_name]
+        t_seq = got["__sequence"]
         n = t_ts.numel()
-        fparts = []
-        avail = {c.path_in_schema
-                 for c in (pq.read_metadata(path).row_group(0).column(i)
-                           for i in range(pq.read_metadata(path)
-                                          .row_group(0).num_columns))}
-        for fn in self.field_names:
-            if fn in avail:
-                fparts.append(pagedec.read_numeric_column(path, fn, dev))
-            else:
-                fparts.append(torch.full((n,), float("nan"),
-                                         dtype=torch.float64, device=dev))
+        fparts = [got[fn] if fn in got else
+                  torch.full((n,), float("nan"), dtype=torch.float64,
+                             device=dev)
+                  for fn in self.field_names]
         t_f = torch.stack(fparts) if fparts else \
             torch.zeros((0, n), dtype=torch.float64, device=dev)
         pk = pq.read_table(path, columns=["__primary_key"]) \

@@ -164,11 +164,13 @@ def build_app(ctx: ServerContext) -> FastAPI:
         import json as _json
         from greptimedb_amd.query.logquery import logquery_to_sql
         body = _json.loads((await request.body()) or b"{}")
+        import time as _t
+        t0 = _t.perf_counter()
         sql = logquery_to_sql(body)
         r = ctx.executor.execute(sql)
-        out = _records_json(r)
-        out["sql"] = sql
-        return out
+        return {"output": [_records_json(r)], "sql": sql,
+                "execution_time_ms":
+                    round((_t.perf_counter() - t0) * 1000, 3)}
 
     # ---------------- pprof-style profiling (reference http.rs:1079) ----
     @app.get("/debug/prof/cpu")

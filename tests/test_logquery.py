@@ -81,7 +81,7 @@ def test_http_endpoint(log_table):
     assert resp.status_code == 200
     body = resp.json()
     assert "sql" in body
-    rows = body["records"]["rows"]
+    rows = body["output"][0]["records"]["rows"]
     assert len(rows) == 2
 
 

@@ -3282,8 +3282,12 @@ def _apply_order_limit(r: QueryResult, sel: ast.Select, names: list[str],
         if not isinstance(e, ast.Col) or e.name not in names:
             continue
         a = np.asarray(r.columns[names.index(e.name)])[idx]
-        if a.dtype == object:
-            a = a.astype(str)
+        if a.dtype == object or a.dtype.kind in "US":
+            # factorize strings to dense codes: numpy sorts i64 codes ~7x
+            # faster than python-object comparisons at 100k+ rows
+            import pandas as pd
+            codes, uniq = pd.factorize(a.astype(object), sort=True)
+            a = codes.astype(np.int64)
         o = np.argsort(a, kind="stable")
         if desc:
             o = o[::-1]

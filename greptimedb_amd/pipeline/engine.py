@@ -501,12 +501,25 @@ class _Cmcd:
         return row
 
 
+class _Vrl:
+    """VRL remap program over the row (reference processor/vrl.rs; subset
+    interpreter in pipeline/vrl.py, compiled once per pipeline)."""
+
+    def __init__(self, cfg):
+        from greptimedb_amd.pipeline.vrl import VrlProgram
+        src = cfg.get("source") if isinstance(cfg, dict) else str(cfg)
+        self.program = VrlProgram(src or "")
+
+    def __call__(self, row):
+        return self.program.run(row)
+
+
 _PROCESSORS = {
     "dissect": _Dissect, "regex": _Regex, "date": _Date, "epoch": _Epoch,
     "gsub": _Gsub, "letter": _Letter, "csv": _Csv, "json_parse": _JsonParse,
     "json_path": _JsonPath, "simple_extract": _SimpleExtract, "join": _Join,
     "urlencoding": _UrlEncoding, "decolorize": _Decolorize, "digest": _Digest,
-    "filter": _Filter, "select": _Select, "cmcd": _Cmcd,
+    "filter": _Filter, "select": _Select, "cmcd": _Cmcd, "vrl": _Vrl,
 }
 
 _NUM_TYPES = {"int8", "int16", "int32", "int64", "uint8", "uint16", "uint32",

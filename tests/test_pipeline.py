@@ -201,3 +201,47 @@ transform:
     row, _sfx = p.run_row({"data": 'br=3200,bs,d=4004,sid="abc-123"'})
     assert row["data_br"] == 3200
     assert row["data_sid"] == "abc-123"
+
+
+def test_vrl_processor():
+    """VRL remap subset (reference processor/vrl.rs)."""
+    p = Pipeline.from_yaml("""
+processors:
+  - vrl:
+      source: |
+        .service = upcase(.svc)
+        .latency_ms = to_float(.lat) * 1000
+        if .latency_ms > 500 { .slow = "yes" } else { .slow = "no" }
+        .msg = trim(.msg) + " [" + .service + "]"
+        del(.svc)
+transform:
+  - fields:
+      - service
+      - slow
+      - msg
+    type: string
+  - fields:
+      - latency_ms
+    type: float64
+""")
+    row, _sfx = p.run_row({"svc": "api", "lat": "0.75", "msg": "  hello "})
+    assert row["service"] == "API"
+    assert row["latency_ms"] == 750.0
+    assert row["slow"] == "yes"
+    assert row["msg"] == "hello [API]"
+    assert "svc" not in row
+
+
+def test_vrl_functions():
+    from greptimedb_amd.pipeline.vrl import VrlProgram
+    p = VrlProgram('''
+.parts = split(.csv, ",")
+.n = length(.parts)
+.joined = join(.parts, "-")
+.has = contains(.csv, "b")
+.hash = sha256("x")
+''')
+    row = p.run({"csv": "a,b,c"})
+    assert row["parts"] == ["a", "b", "c"]
+    assert row["n"] == 3 and row["joined"] == "a-b-c" and row["has"]
+    assert len(row["hash"]) == 64

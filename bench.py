@@ -84,6 +84,10 @@ def main():
                     help="fixture time span (default 72h on GPU = 104M rows)")
     ap.add_argument("--query-reps", type=int, default=7)
     ap.add_argument("--skip-queries", action="store_true")
+    ap.add_argument("--no-route", action="store_true",
+                    help="diagnostic: pre-sharded ingest (no cross-rank "
+                         "fan-out); NOT the default because any-rank "
+                         "routing is the honest distributed path")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -130,7 +134,7 @@ def main():
     # but OWNERSHIP is global: series hash to any rank, and rows are shipped
     # to the owner (parallel/write_fanout.py; reference insert.rs:389-496).
     exchange = None
-    if world > 1:
+    if world > 1 and not args.no_route:
         from greptimedb_amd.parallel.write_fanout import WriteExchange
 
         def _mk_handler():

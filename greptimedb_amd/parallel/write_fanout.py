@@ -75,8 +75,14 @@ class WriteExchange:
         # connection gets its own receive pipeline (no shared lock), like
         # the reference's per-connection tonic service instances
         self.handler_factory = handler_factory
-        self._conns: dict[int, socket.socket] = {}
-        self._conn_locks = {r: threading.Lock() for r in range(world)}
+        # free-connection pool per peer: a conn is checked OUT for the
+        # whole request/ack round trip, so concurrent callers (the 6 TSBS
+        # ingest workers) each get their own socket — and the peer gets
+        # one handler thread (and receive pipeline) per socket. A single
+        # shared conn would serialize every remote apply per sender/peer
+        # pair, collapsing whole-node throughput.
+        self._pool: dict[int, list[socket.socket]] = {r: [] for r in range(world)}
+        self._pool_lock = threading.Lock()
         self._closing = False
         self._srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         self._srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
@@ -128,20 +134,33 @@ class WriteExchange:
         raise ConnectionError(f"cannot reach rank {peer}: {last}")
 
     def request(self, peer: int, payload: bytes) -> bytes:
-        """Send one frame to `peer`, wait for its ack frame."""
-        with self._conn_locks[peer]:
-            sock = self._conns.get(peer)
-            if sock is None:
-                sock = self._conns[peer] = self._connect(peer)
+        """Send one frame to `peer`, wait for its ack frame. Concurrent
+        callers use distinct pooled connections (no lock across the RTT)."""
+        with self._pool_lock:
+            sock = self._pool[peer].pop() if self._pool[peer] else None
+        if sock is None:
+            sock = self._connect(peer)
+        try:
             try:
                 _write_frame(sock, payload)
                 resp = _read_frame(sock)
             except (ConnectionError, OSError):
                 # one reconnect attempt (peer restarted)
                 sock.close()
-                sock = self._conns[peer] = self._connect(peer)
+                sock = self._connect(peer)
                 _write_frame(sock, payload)
                 resp = _read_frame(sock)
+        except BaseException:
+            try:
+                sock.close()
+            except OSError:
+                pass
+            raise
+        with self._pool_lock:
+            if self._closing:
+                sock.close()
+            else:
+                self._pool[peer].append(sock)
         if resp.startswith(b"ERR"):
             raise RuntimeError(f"rank {peer} write failed: {resp[4:].decode()}")
         return resp
@@ -152,12 +171,14 @@ class WriteExchange:
             self._srv.close()
         except OSError:
             pass
-        for s in self._conns.values():
-            try:
-                s.close()
-            except OSError:
-                pass
-        self._conns.clear()
+        with self._pool_lock:
+            for conns in self._pool.values():
+                for s in conns:
+                    try:
+                        s.close()
+                    except OSError:
+                        pass
+                conns.clear()
 
 
 # ------------------------------------------------------------------ codec

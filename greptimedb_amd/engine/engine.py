@@ -37,6 +37,8 @@ class EngineConfig:
     background_flush: bool = True
     scan_mem_bytes: int = 32 << 30      # host-side scan materialization quota
     record_events: bool = False         # persist DDL/migration events table
+    gc_interval_s: float = 600.0        # background orphan-SST scan period
+    gc_grace_s: float = 3600.0          # spare files younger than this
 
 
 @dataclass
@@ -421,15 +423,32 @@ class MitoEngine:
                 c.compact_region(region)
 
     def _flush_loop(self):
+        # flush worker doubles as the periodic GC scheduler (reference:
+        # metasrv's gc ticker; here the one background thread owns both)
+        import queue as _queue
+        import time as _time
+        next_gc = _time.monotonic() + self.config.gc_interval_s
         while True:
-            region = self._flush_q.get()
+            try:
+                region = self._flush_q.get(timeout=max(
+                    0.05, next_gc - _time.monotonic()))
+            except _queue.Empty:
+                region = False   # timer tick, not shutdown
             if region is None:
                 return
-            try:
-                self._flush_region(region)
-            except Exception:  # pragma: no cover
-                import traceback
-                traceback.print_exc()
+            if region is not False:
+                try:
+                    self._flush_region(region)
+                except Exception:  # pragma: no cover
+                    import traceback
+                    traceback.print_exc()
+            if _time.monotonic() >= next_gc:
+                next_gc = _time.monotonic() + self.config.gc_interval_s
+                try:
+                    self.gc_orphan_ssts(self.config.gc_grace_s)
+                except Exception:  # pragma: no cover
+                    import traceback
+                    traceback.print_exc()
 
     def gc_orphan_ssts(self, grace_s: float = 3600.0) -> int:
         """Delete SST/sidecar files no region manifest references

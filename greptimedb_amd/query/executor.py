@@ -450,10 +450,50 @@ class Executor:
                     _os.unlink(old)
         return QueryResult(["rows"], [[deleted]])
 
+    @staticmethod
+    def _copy_object_store(path: str, options: dict):
+        """Resolve an `s3://bucket/key` COPY path to (store, key).
+        Endpoint comes from the WITH/CONNECTION `endpoint` option or
+        GDB_S3_ENDPOINT (reference: common/datasource object-store URLs
+        with CONNECTION (endpoint=..., access_key_id=...))."""
+        if not path.startswith("s3://"):
+            return None, path
+        import os as _os
+        from greptimedb_amd.engine.remote import S3ObjectStore
+        rest = path[5:]
+        bucket, _, key = rest.partition("/")
+        endpoint = options.get("endpoint") or _os.environ.get("GDB_S3_ENDPOINT")
+        if not endpoint:
+            raise ValueError(
+                "COPY s3:// needs an endpoint: WITH (endpoint='http://...') "
+                "or GDB_S3_ENDPOINT")
+        return S3ObjectStore(endpoint, bucket), key
+
     def _exec_copy(self, c: ast.Copy) -> QueryResult:
-        """COPY table TO/FROM file (reference: operator COPY via
-        common/datasource — parquet/csv/json by extension or WITH format)."""
+        """COPY table TO/FROM file or s3:// object (reference: operator COPY
+        via common/datasource — parquet/csv/json by extension or WITH
+        format; object-store targets staged through a temp file)."""
         import pyarrow as pa
+        store, _key = self._copy_object_store(c.path, c.options)
+        if store is not None:
+            import tempfile as _tf
+            import os as _os
+            suffix = _os.path.splitext(_key)[1] or ".parquet"
+            tmp = _tf.NamedTemporaryFile(suffix=suffix, delete=False)
+            tmp.close()
+            try:
+                sub = ast.Copy(direction=c.direction, table=c.table,
+                               path=tmp.name, options=dict(c.options))
+                if c.direction == "from":
+                    with open(tmp.name, "wb") as f:
+                        f.write(store.get(_key))
+                res = self._exec_copy(sub)
+                if c.direction == "to":
+                    with open(tmp.name, "rb") as f:
+                        store.put(_key, f.read())
+                return res
+            finally:
+                _os.unlink(tmp.name)
         fmt = c.options.get("format")
         if fmt is None:
             fmt = ("parquet" if c.path.endswith(".parquet")

@@ -248,7 +248,11 @@ class Parser:
                 raise InvalidSyntax("COPY <table> TO|FROM '<path>'")
             path = self.next().value
             options = {}
-            if self.eat_kw("with"):
+            # WITH (format=...) and CONNECTION (endpoint=..., ...) — the
+            # reference's COPY grammar (sql/src/parsers/copy_parser.rs);
+            # connection keys merge into one option map here
+            while self.at_kw("with") or self.at_kw("connection"):
+                self.next()
                 self.expect_op("(")
                 while not self.eat_op(")"):
                     k = self.next().value

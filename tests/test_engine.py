@@ -612,6 +612,31 @@ def test_gc_orphan_ssts(tmp_path):
     eng.close()
 
 
+def test_gc_runs_periodically(tmp_path):
+    """The background flush thread doubles as the GC ticker: orphans are
+    collected without any ADMIN call when gc_interval_s elapses."""
+    import os as _os
+    import time as _time
+    d = str(tmp_path / "data")
+    eng = MitoEngine(EngineConfig(data_dir=d, device="cpu",
+                                  background_flush=True,
+                                  gc_interval_s=0.2, gc_grace_s=60.0))
+    ing = Ingestor(eng)
+    w = CpuWorkload(scale=5)
+    ing.ingest_lines(w.next_batch(200))
+    eng.flush_all()
+    region = next(r for st in eng.tables.values() for r in st.regions
+                  if r.manifest.files)
+    orphan = _os.path.join(region.dir, "sst", "deadbeef00.parquet")
+    open(orphan, "wb").write(b"leftover")
+    _os.utime(orphan, (0, 0))
+    deadline = _time.monotonic() + 5.0
+    while _os.path.exists(orphan) and _time.monotonic() < deadline:
+        _time.sleep(0.05)
+    assert not _os.path.exists(orphan)
+    eng.close()
+
+
 def test_gorilla_compress_table_roundtrip(tmp_path):
     """K20 cold tier: ADMIN compress_table packs resident batches into
     Gorilla blocks; scans transparently re-materialize and results are

@@ -228,3 +228,32 @@ def test_http_kv_backend_cas(fake):
     assert kv.get("lease/a") == "holder2"
     kv.delete("table/route/1")
     assert kv.get("table/route/1") is None
+
+
+def test_copy_to_from_s3(fake, tmp_path):
+    """COPY table TO/FROM 's3://bucket/key' stages through the object
+    store (reference: COPY ... CONNECTION(endpoint=...) via
+    common/datasource object-store URLs)."""
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d1"), device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE m (ts TIMESTAMP TIME INDEX, host STRING PRIMARY KEY, v DOUBLE)")
+    ex.execute("INSERT INTO m VALUES (1000, 'a', 1.5), (2000, 'b', 2.5), (3000, 'a', 3.5)")
+    r = ex.execute(
+        f"COPY m TO 's3://exports/m.parquet' CONNECTION (endpoint='{fake.endpoint}')")
+    assert r.rows()[0][0] == 3
+    assert "exports/m.parquet" in fake.objects
+
+    eng2 = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d2"), device="cpu",
+                                   background_flush=False))
+    ex2 = Executor(eng2)
+    ex2.execute("CREATE TABLE m (ts TIMESTAMP TIME INDEX, host STRING PRIMARY KEY, v DOUBLE)")
+    ex2.execute(
+        f"COPY m FROM 's3://exports/m.parquet' CONNECTION (endpoint='{fake.endpoint}')")
+    got = ex2.execute("SELECT host, v FROM m ORDER BY v")
+    assert [list(r) for r in got.rows()] == [["a", 1.5], ["b", 2.5], ["a", 3.5]]
+    eng.close()
+    eng2.close()

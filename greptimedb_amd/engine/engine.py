@@ -162,6 +162,16 @@ class MitoEngine:
             if name != EVENTS_TABLE:
                 recorder_of(self).record("drop_table", {"table": name})
 
+    def truncate_table(self, name: str):
+        """TRUNCATE: drop every region's data, keep the table definition
+        (reference: TruncateTable DDL procedure → region truncate)."""
+        st = self.table(name)
+        for region in st.regions:
+            region.truncate()
+        if self.config.record_events:
+            from greptimedb_amd.utils.events import recorder_of
+            recorder_of(self).record("truncate_table", {"table": name})
+
     def table(self, name: str) -> TableState:
         try:
             return self.tables[name]

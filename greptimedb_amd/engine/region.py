@@ -291,6 +291,31 @@ class Region:
     def should_flush(self, limit_bytes: int) -> bool:
         return self.memtable.bytes_used >= limit_bytes
 
+    def truncate(self):
+        """Drop all data, keep schema + series registry (reference: mito2
+        truncate — RegionMetaAction::Truncate clears the file list and the
+        replay point; src/mito2/src/engine (truncate request path))."""
+        with self._flush_lock:
+            with self.lock:
+                old = list(self.manifest.files)
+                self.manifest.commit({"kind": "truncate",
+                                      "flushed_seq": self.last_seq})
+                self.flushed_seq = self.last_seq
+                self.memtable = Memtable(len(self.field_names),
+                                         device=self.device)
+                self.mem_base = self.row_seq
+                self.flushing.clear()
+                self.sst_cache.clear()
+                for ft in self.text_cols.values():
+                    ft.reset_mem()
+            for fid in old:
+                for ext in (".parquet", ".ftidx"):
+                    p = os.path.join(self.dir, "sst", f"{fid}{ext}")
+                    try:
+                        os.unlink(p)
+                    except OSError:
+                        pass
+
     def flush(self):
         """Sort + dedup the memtable on device, write an SST, swap memtable.
 

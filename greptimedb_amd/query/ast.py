@@ -194,6 +194,11 @@ class DropTable:
 
 
 @dataclass
+class TruncateTable:
+    name: str
+
+
+@dataclass
 class ShowTables:
     pass
 

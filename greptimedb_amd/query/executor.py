@@ -31,7 +31,8 @@ from greptimedb_amd.utils.errors import (InvalidArguments, PlanQuery,
 from greptimedb_amd.utils.timeutil import parse_ts_ms, trunc_unit_ms
 from greptimedb_amd.ops import ts_bucket_agg, dedup_mark_last
 
-AGG_FUNCS = {"count", "sum", "min", "max", "avg", "mean", "last_value"}
+AGG_FUNCS = {"count", "sum", "min", "max", "avg", "mean", "last_value",
+             "first_value"}
 BUCKET_FUNCS = {"date_trunc", "date_bin", "time_bucket"}
 MAX_BUCKETS = 8_000_000
 
@@ -265,6 +266,9 @@ class Executor:
             except Exception:
                 if not stmt.if_exists:
                     raise
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.TruncateTable):
+            self.engine.truncate_table(stmt.name)
             return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.ShowTables):
             names = sorted(self.engine.tables)
@@ -1514,7 +1518,7 @@ class Executor:
         elif isinstance(stmt, ast.CreateTable):
             stmt.name = res(stmt.name)
         elif isinstance(stmt, (ast.DropTable, ast.ShowCreateTable,
-                               ast.DescribeTable)):
+                               ast.DescribeTable, ast.TruncateTable)):
             stmt.name = res(stmt.name)
         elif isinstance(stmt, ast.InsertValues):
             stmt.table = res(stmt.table)
@@ -1822,7 +1826,7 @@ class Executor:
         return group_keys, region_luts
 
     def _exec_aggregate(self, sel: ast.Select, plan: SelectPlan) -> QueryResult:
-        if any(a.func == "last_value" for a in plan.aggs):
+        if any(a.func in ("last_value", "first_value") for a in plan.aggs):
             return self._exec_lastpoint(sel, plan)
         st = plan.table
         device = self.engine.config.device
@@ -1943,12 +1947,16 @@ class Executor:
         device = self.engine.config.device
         if plan.bucket is not None:
             raise PlanQuery("last_value with time buckets unsupported")
-        for a in plan.aggs:
-            if a.func != "last_value":
-                raise PlanQuery("cannot mix last_value with other aggregates")
+        funcs = {a.func for a in plan.aggs}
+        if len(funcs) != 1:
+            raise PlanQuery("cannot mix last_value with other aggregates")
+        # first_value = argmin over ts: run the same argmax machinery on
+        # negated timestamps (merge/dist combine logic is sign-agnostic)
+        rev = funcs == {"first_value"}
         lv_fields = sorted({a.arg for a in plan.aggs})
         ts_lo = plan.ts_lo if plan.ts_lo is not None else -(1 << 62)
         ts_hi = plan.ts_hi if plan.ts_hi is not None else (1 << 62)
+        sl_lo, sl_hi = (-ts_hi, -ts_lo) if rev else (ts_lo, ts_hi)
 
         gt = plan.group_tags
         group_keys, region_luts = self._build_group_luts(st, plan, gt)
@@ -1966,18 +1974,22 @@ class Executor:
                     idx = mask.nonzero(as_tuple=True)[0]
                     if idx.numel():
                         filt.append((src, idx))
-                pairs = [(src.ts[idx].contiguous(), src.series[idx].contiguous(),
-                          src.sorted)
+                pairs = [((-src.ts[idx]).contiguous() if rev
+                          else src.ts[idx].contiguous(),
+                          src.series[idx].contiguous(),
+                          False if rev else src.sorted)
                          for src, idx in filt]
                 src_objs = [(src, idx) for src, idx in filt]
             else:
-                pairs = [(src.ts, src.series, src.sorted) for src in sources]
+                pairs = [((-src.ts).contiguous() if rev else src.ts,
+                          src.series, False if rev else src.sorted)
+                         for src in sources]
                 src_objs = [(src, None) for src in sources]
             if not pairs:
                 continue
             ng = len(group_keys) if group_keys else 1
             from greptimedb_amd.ops import series_last
-            b_ts, b_src, b_row = series_last(pairs, lut_t, ts_lo, ts_hi, ng)
+            b_ts, b_src, b_row = series_last(pairs, lut_t, sl_lo, sl_hi, ng)
             mx_h = b_ts.numpy().astype(np.int64)
             # gather winner values per source
             vals = torch.full((len(lv_fields), ng), float("nan"),
@@ -2035,7 +2047,8 @@ class Executor:
                 i = gt.index(e.name)
                 arr = np.array([keys_by_slot[s][i] for s in slot_idx], dtype=object)
                 names.append(alias or e.name); cols.append(arr); kinds.append("")
-            elif isinstance(e, ast.Func) and e.name == "last_value":
+            elif isinstance(e, ast.Func) and e.name in ("last_value",
+                                                        "first_value"):
                 arr = best_val[fpos[e.args[0].name]][slot_idx]
                 names.append(alias or _expr_name(e)); cols.append(arr); kinds.append("")
             else:
@@ -3515,9 +3528,20 @@ def _expr_name(e: ast.Expr) -> str:
     if isinstance(e, ast.Col):
         return e.name
     if isinstance(e, ast.Func):
-        return f"{e.name}({','.join(_expr_name(a) for a in e.args)})"
+        inner = ",".join(_expr_name(a) for a in e.args)
+        if e.distinct:
+            inner = "DISTINCT " + inner
+        return f"{e.name}({inner})"
     if isinstance(e, ast.Star):
         return "*"
     if isinstance(e, ast.Lit):
         return str(e.value)
+    if isinstance(e, ast.BinOp):
+        return f"{_expr_name(e.left)} {e.op} {_expr_name(e.right)}"
+    if isinstance(e, ast.UnaryOp):
+        return f"{e.op}{_expr_name(e.operand)}"
+    if isinstance(e, ast.Case):
+        return "case"
+    if isinstance(e, ast.ScalarSubquery):
+        return "(subquery)"
     return repr(e)

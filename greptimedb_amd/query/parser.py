@@ -179,6 +179,10 @@ class Parser:
             stmt = self.parse_create()
         elif self.at_kw("drop"):
             stmt = self.parse_drop()
+        elif self.at_kw("truncate"):
+            self.next()
+            self.eat_kw("table")
+            stmt = ast.TruncateTable(str(self.next().value))
         elif self.at_kw("show"):
             self.next()
             if self.eat_kw("flows"):
@@ -798,6 +802,16 @@ class Parser:
                             args.append(self.parse_expr())
                         if not self.eat_op(","):
                             break
+                # last_value(v ORDER BY ts [DESC]) / first_value(...) — the
+                # reference's ordered-set form; DESC flips first↔last
+                if low in ("last_value", "first_value") and self.eat_kw("order"):
+                    self.expect_kw("by")
+                    self.next()  # the order column (time index)
+                    if self.eat_kw("desc"):
+                        low = ("first_value" if low == "last_value"
+                               else "last_value")
+                    else:
+                        self.eat_kw("asc")
                 self.expect_op(")")
                 if self.eat_kw("over"):
                     return self._window_spec(low, args)

@@ -225,6 +225,33 @@ def select_over_result(sel: ast.Select, base, base_kinds=None):
                 np.asarray(c) for c in out_cols]
     m = len(res_cols[0]) if res_cols else 0
 
+    # ORDER BY expressions that aren't output columns: evaluate them as
+    # hidden columns aligned with the output rows (pre-distinct)
+    name_idx = {nm: i for i, nm in enumerate(out_names)}
+    order_extra: dict[int, np.ndarray] = {}
+    if sel.order_by and m:
+        for oi, (e, _desc) in enumerate(sel.order_by):
+            e = resolve(e)
+            if isinstance(e, ast.Col) and e.name in name_idx:
+                continue
+            if isinstance(e, ast.Lit) and isinstance(e.value, int):
+                continue
+            if has_agg:
+                vals = []
+                for k, idxs in items:
+                    sub_rows = rows[np.asarray(idxs, dtype=np.int64)]
+                    if _contains_agg(e):
+                        vals.append(_eval_scalar(e, col_data, sub_rows))
+                    else:
+                        gv = np.asarray(X._np_raw(e, col_data))[sub_rows]
+                        vals.append(gv[0] if len(gv) else None)
+                order_extra[oi] = np.asarray(vals, dtype=object)
+            else:
+                v = X._np_raw(e, {k: c[rows] for k, c in col_data.items()})
+                if np.ndim(v) == 0:
+                    v = np.full(len(rows), v)
+                order_extra[oi] = np.asarray(v)
+
     if sel.distinct and m:
         seen, keep = set(), []
         for i in range(m):
@@ -232,18 +259,21 @@ def select_over_result(sel: ast.Select, base, base_kinds=None):
             if k not in seen:
                 seen.add(k)
                 keep.append(i)
-        res_cols = [c[np.asarray(keep, dtype=np.int64)] for c in res_cols]
+        ka = np.asarray(keep, dtype=np.int64)
+        res_cols = [c[ka] for c in res_cols]
+        order_extra = {oi: c[ka] for oi, c in order_extra.items()}
         m = len(keep)
 
     if sel.order_by and m:
         idx = list(range(m))
-        name_idx = {nm: i for i, nm in enumerate(out_names)}
 
         def key_for(i):
             ks = []
-            for e, desc in sel.order_by:
+            for oi, (e, desc) in enumerate(sel.order_by):
                 e = resolve(e)
-                if isinstance(e, ast.Col) and e.name in name_idx:
+                if oi in order_extra:
+                    v = order_extra[oi][i]
+                elif isinstance(e, ast.Col) and e.name in name_idx:
                     v = res_cols[name_idx[e.name]][i]
                 elif isinstance(e, ast.Lit) and isinstance(e.value, int):
                     v = res_cols[e.value - 1][i]
@@ -320,9 +350,17 @@ def _name_of(e, i) -> str:
     if isinstance(e, ast.Func):
         inner = ", ".join(_name_of(a, 0) if not isinstance(a, ast.Star)
                           else "*" for a in e.args)
+        if e.distinct:
+            inner = "DISTINCT " + inner
         return f"{e.name}({inner})"
     if isinstance(e, ast.Lit):
         return str(e.value)
+    if isinstance(e, ast.BinOp):
+        return f"{_name_of(e.left, i)} {e.op} {_name_of(e.right, i)}"
+    if isinstance(e, ast.UnaryOp):
+        return f"{e.op}{_name_of(e.operand, i)}"
+    if isinstance(e, ast.Case):
+        return "case"
     return f"col{i}"
 
 

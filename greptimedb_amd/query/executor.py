@@ -1106,15 +1106,40 @@ class Executor:
             return self._exec_information_schema(sel)
         if sel.joins:
             return self._exec_join(sel)
-        plan = self._plan_select(sel)
-        if any(_has_range_agg(e) for e, _a in sel.projections):
-            return self._exec_range_select(sel, plan)
-        if plan.aggs:
-            return self._exec_aggregate(sel, plan)
-        knn = self._try_vector_knn(sel, plan)
-        if knn is not None:
-            return knn
-        return self._exec_raw(sel, plan)
+        try:
+            plan = self._plan_select(sel)
+            if any(_has_range_agg(e) for e, _a in sel.projections):
+                return self._exec_range_select(sel, plan)
+            if plan.aggs:
+                return self._exec_aggregate(sel, plan)
+            knn = self._try_vector_knn(sel, plan)
+            if knn is not None:
+                return knn
+            return self._exec_raw(sel, plan)
+        except PlanQuery:
+            return self._select_fallback(sel)
+
+    def _select_fallback(self, sel: ast.Select) -> QueryResult:
+        """General-shape fallback: shapes the fused device planner rejects
+        (aggregates over expressions like sum(CASE WHEN …), ORDER BY
+        expressions, …) materialize the WHERE-filtered raw rows once and
+        evaluate through the derived-table engine (reference parity:
+        DataFusion handles these generically; our device planner covers
+        the hot TSBS/observability shapes and this path covers the tail)."""
+        if not (isinstance(sel.table, str) and sel.table in self.engine.tables):
+            raise PlanQuery(f"unsupported select shape on {sel.table!r}")
+        if getattr(sel, "_in_fallback", False):
+            raise PlanQuery("unsupported select shape (fallback failed)")
+        from greptimedb_amd.query.derived import select_over_result
+        inner = ast.Select(projections=[(ast.Star(), None)], table=sel.table,
+                           where=sel.where)
+        inner._in_fallback = True
+        base = self._exec_select(inner)
+        outer = ast.Select(projections=sel.projections, table=None,
+                           where=None, group_by=sel.group_by,
+                           having=sel.having, order_by=sel.order_by,
+                           limit=sel.limit, offset=sel.offset)
+        return select_over_result(outer, base)
 
     def _exec_join(self, sel: ast.Select) -> QueryResult:
         """Two-table equality JOIN (inner/left): each side runs as a pushed-

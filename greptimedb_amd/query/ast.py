@@ -62,6 +62,14 @@ class ScalarSubquery(Expr):
 
 
 @dataclass
+class Exists(Expr):
+    """EXISTS (SELECT …) — equality-correlated forms rewrite to IN;
+    uncorrelated forms fold to a constant predicate at resolve time."""
+    select: object
+    negated: bool = False
+
+
+@dataclass
 class RangeAgg(Expr):
     """agg(col) RANGE '10s' [FILL x] — sliding window [t, t+range) per
     ALIGN step (ref: src/query/src/range_select/plan.rs:947 window math)."""
@@ -223,6 +231,7 @@ class InsertValues:
     table: str
     columns: list[str]
     rows: list[list[object]]
+    select: object = None   # INSERT INTO t [cols] SELECT ... (rows empty)
 
 
 @dataclass

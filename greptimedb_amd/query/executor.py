@@ -898,6 +898,18 @@ class Executor:
         if getattr(st, "external", False):
             raise InvalidArguments(f"table {ins.table} is external (read-only)")
         schema = st.schema
+        if ins.select is not None:
+            # INSERT INTO t [cols] SELECT ... — positional column mapping
+            # (reference: operator insert-from-query path)
+            r = self.execute_stmt(ins.select)
+            cols = ins.columns or r.names
+            if len(cols) != len(r.names):
+                raise InvalidArguments(
+                    f"INSERT SELECT: {len(cols)} target columns, "
+                    f"{len(r.names)} selected")
+            ins = ast.InsertValues(ins.table, list(cols),
+                                   [list(row) for row in r.rows()])
+            return self._exec_insert(ins)
         cols = ins.columns or [c.name for c in schema.columns]
         n = len(ins.rows)
         by_col = {c: [r[i] for r in ins.rows] for i, c in enumerate(cols)}
@@ -989,6 +1001,8 @@ class Executor:
             if x.many:
                 return [lit(v) for v in col]
             return lit(col[0]) if len(col) else ast.Lit(None)
+        if isinstance(x, ast.Exists):
+            return self._resolve_exists(x, outer)
         if isinstance(x, ast.BinOp):
             x.left = self._resolve_subqueries(x.left, outer)
             x.right = self._resolve_subqueries(x.right, outer)
@@ -1006,6 +1020,83 @@ class Executor:
                 items.extend(got if isinstance(got, list) else [got])
             x.items = items
         return x
+
+    def _resolve_exists(self, x: "ast.Exists", outer):
+        """EXISTS (SELECT … WHERE inner.k = outer.k AND rest) rewrites to
+        `outer.k IN (SELECT inner.k WHERE rest)` (semi-join decorrelation,
+        the shape DataFusion's subquery rules produce in the reference);
+        uncorrelated EXISTS executes once and folds to a constant
+        predicate vectorized over the outer time index."""
+        inner = x.select
+        corr = None
+        if outer is not None and isinstance(outer.table, str) and \
+                isinstance(inner.table, str) and inner.where is not None:
+            outer_quals = {outer.table}
+            if outer.table_alias:
+                outer_quals.add(outer.table_alias)
+            conjs = _split_conjuncts(inner.where)
+            for i, c in enumerate(conjs):
+                if not (isinstance(c, ast.BinOp) and c.op == "=" and
+                        isinstance(c.left, ast.Col) and
+                        isinstance(c.right, ast.Col)):
+                    continue
+                for a, b in ((c.left.name, c.right.name),
+                             (c.right.name, c.left.name)):
+                    if "." in a and a.rsplit(".", 1)[0] in outer_quals:
+                        ck = b.rsplit(".", 1)[1] if "." in b else b
+                        corr = (i, ck, a.rsplit(".", 1)[1])
+                        break
+                if corr is not None:
+                    break
+        if corr is not None:
+            i, inner_key, outer_key = corr
+            conjs = _split_conjuncts(inner.where)
+            inner_quals = {inner.table}
+            if inner.table_alias:
+                inner_quals.add(inner.table_alias)
+
+            def strip(e):
+                """Drop inner-table qualifiers so the sub-select plans as a
+                plain single-table query."""
+                if isinstance(e, ast.Col) and "." in e.name and \
+                        e.name.rsplit(".", 1)[0] in inner_quals:
+                    return ast.Col(e.name.rsplit(".", 1)[1])
+                if isinstance(e, ast.BinOp):
+                    return ast.BinOp(e.op, strip(e.left), strip(e.right))
+                if isinstance(e, ast.UnaryOp):
+                    return ast.UnaryOp(e.op, strip(e.operand))
+                if isinstance(e, ast.Func):
+                    return ast.Func(e.name, [strip(a) for a in e.args],
+                                    e.distinct)
+                if isinstance(e, ast.InList):
+                    return ast.InList(strip(e.expr),
+                                      [strip(a) for a in e.items], e.negated)
+                if isinstance(e, ast.Between):
+                    return ast.Between(strip(e.expr), strip(e.low),
+                                       strip(e.high), e.negated)
+                if isinstance(e, ast.IsNull):
+                    return ast.IsNull(strip(e.expr), e.negated)
+                return e
+            rest = [strip(c) for j, c in enumerate(conjs) if j != i]
+            w2 = None
+            for c in rest:
+                w2 = c if w2 is None else ast.BinOp("and", w2, c)
+            sub = ast.Select(projections=[(ast.Col(inner_key), None)],
+                             table=inner.table,
+                             table_alias=inner.table_alias, where=w2)
+            items = self._resolve_subqueries(
+                ast.ScalarSubquery(sub, many=True), None)
+            return ast.InList(ast.Col(outer_key), items,
+                              negated=x.negated)
+        r = self._exec_select(inner)
+        nonempty = bool(r.columns and len(r.columns[0]))
+        truth = nonempty != x.negated
+        if outer is not None and isinstance(outer.table, str) and \
+                outer.table in self.engine.tables:
+            ts_name = self.engine.table(outer.table).schema.time_index.name
+            # ts is never NULL: IS NOT NULL ≡ always-true, IS NULL ≡ false
+            return ast.IsNull(ast.Col(ts_name), negated=truth)
+        return ast.Lit(truth)
 
     def _try_decorrelate(self, inner: ast.Select, outer: ast.Select):
         """`(SELECT agg(x) FROM t2 [t2a] WHERE t2.k = o.k AND ...)` with the
@@ -1547,6 +1638,8 @@ class Executor:
             stmt.name = res(stmt.name)
         elif isinstance(stmt, ast.InsertValues):
             stmt.table = res(stmt.table)
+            if stmt.select is not None:
+                fix_select(stmt.select)
         elif isinstance(stmt, ast.Delete):
             stmt.table = res(stmt.table)
         elif isinstance(stmt, ast.AlterTable):
@@ -3282,7 +3375,7 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
 def _has_subquery(e) -> bool:
     if e is None:
         return False
-    if isinstance(e, ast.ScalarSubquery):
+    if isinstance(e, (ast.ScalarSubquery, ast.Exists)):
         return True
     if isinstance(e, ast.BinOp):
         return _has_subquery(e.left) or _has_subquery(e.right)

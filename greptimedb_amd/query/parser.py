@@ -638,6 +638,9 @@ class Parser:
                 if not self.eat_op(","):
                     break
             self.expect_op(")")
+        if self.at_kw("select") or self.at_kw("with"):
+            return ast.InsertValues(table, columns, [],
+                                    select=self.parse_query())
         self.expect_kw("values")
         rows = []
         while True:
@@ -771,6 +774,11 @@ class Parser:
             low = t.value.lower()
             if low == "not":
                 return ast.UnaryOp("not", self.parse_expr(3))
+            if low == "exists" and self.at_op("("):
+                self.next()
+                sub = self.parse_select()
+                self.expect_op(")")
+                return ast.Exists(sub)
             if low == "interval":
                 txt = self.next().value
                 return ast.Interval(parse_interval_text(str(txt)), str(txt))

@@ -142,9 +142,23 @@ class PromMatrix:
 
 
 class PromScalar:
+    """Prometheus scalar type. `value` is a float (constant) or a 1-D
+    float64 tensor of length T (per-step scalar — time(), scalar(v))."""
+
     def __init__(self, value, grid):
-        self.value = value  # float
+        self.value = value
         self.grid = grid
+
+
+def _scalar_plane(sc: "PromScalar", T: int, device=None) -> torch.Tensor:
+    """(1,T) tensor view of a scalar for vector contexts."""
+    v = sc.value
+    if isinstance(v, torch.Tensor):
+        t = v.to(dtype=torch.float64)
+        if device is not None:
+            t = t.to(device)
+        return t.reshape(1, -1).expand(1, T) if t.numel() != T else t.reshape(1, T)
+    return torch.full((1, T), float(v), dtype=torch.float64, device=device)
 
 
 class PromEvaluator:
@@ -163,8 +177,7 @@ class PromEvaluator:
         grid = t0 + np.arange(T, dtype=np.int64) * step
         r = self._eval(expr, t0, step, T, grid)
         if isinstance(r, PromScalar):
-            vals = torch.full((1, T), float(r.value), dtype=torch.float64)
-            return PromMatrix([{}], vals, grid)
+            return PromMatrix([{}], _scalar_plane(r, T), grid)
         return r
 
     def query_instant(self, q: str, time_s: float) -> PromMatrix:
@@ -222,6 +235,11 @@ class PromEvaluator:
                 v = m.values
                 m = _matrix_map(m, torch.where(v > 0, torch.ones_like(v),
                                                torch.full_like(v, float("nan"))))
+            if f == "absent_over_time" and m.S == 0:
+                # no series matched at all → absent everywhere (Prometheus
+                # emits a single 1-valued series)
+                return PromMatrix([{}], torch.ones((1, T),
+                                                   dtype=torch.float64), grid)
             if f in ("rate", "increase", "delta", "idelta", "irate", "deriv",
                      "predict_linear", "changes", "resets") or "_over_time" in f:
                 m = _matrix_map(m, drop_name=True)
@@ -253,18 +271,17 @@ class PromEvaluator:
             if isinstance(m, PromScalar):
                 return m
             if m.S == 1:
-                return PromMatrix([{}], m.values, grid)
-            return PromMatrix([{}], torch.full((1, T), float("nan"),
-                                               dtype=torch.float64), grid)
+                return PromScalar(m.values[0], grid)
+            return PromScalar(torch.full((T,), float("nan"),
+                                         dtype=torch.float64), grid)
         if f == "vector":
             m = self._eval(e.args[0], t0, step, T, grid)
             if isinstance(m, PromScalar):
-                return PromMatrix([{}], torch.full((1, T), float(m.value),
-                                                   dtype=torch.float64), grid)
+                return PromMatrix([{}], _scalar_plane(m, T), grid)
             return m
         if f == "time":
-            return PromMatrix([{}], torch.as_tensor(grid[None, :] / 1000.0,
-                                                    dtype=torch.float64), grid)
+            return PromScalar(torch.as_tensor(grid / 1000.0,
+                                              dtype=torch.float64), grid)
         if f == "timestamp":
             m = self._eval(e.args[0], t0, step, T, grid)
             v = torch.where(torch.isnan(m.values),
@@ -324,8 +341,7 @@ class PromEvaluator:
             if e.args:
                 m = self._eval(e.args[0], t0, step, T, grid)
                 if isinstance(m, PromScalar):
-                    m = PromMatrix([{}], torch.full((1, T), float(m.value),
-                                                    dtype=torch.float64), grid)
+                    m = PromMatrix([{}], _scalar_plane(m, T), grid)
             else:
                 m = PromMatrix([{}], torch.as_tensor(
                     grid[None, :] / 1000.0, dtype=torch.float64), grid)
@@ -506,9 +522,7 @@ class PromEvaluator:
         igrid = inner_start + np.arange(T2, dtype=np.int64) * res
         m = self._eval(sub.expr, inner_start, res, T2, igrid)
         if isinstance(m, PromScalar):
-            m = PromMatrix([{}], torch.full((1, T2), float(m.value),
-                                            dtype=torch.float64, device=device),
-                           igrid)
+            m = PromMatrix([{}], _scalar_plane(m, T2, device), igrid)
         v = m.values
         S = int(v.shape[0])
         if S == 0:
@@ -910,11 +924,19 @@ class PromEvaluator:
         l = self._eval(e.left, t0, step, T, grid)
         r = self._eval(e.right, t0, step, T, grid)
         if isinstance(l, PromScalar) and isinstance(r, PromScalar):
+            if isinstance(l.value, torch.Tensor) or \
+                    isinstance(r.value, torch.Tensor):
+                a = _scalar_plane(l, T)
+                b = _scalar_plane(r, T).to(a.device)
+                res, _keep = _vector_op(e.op, a, b, True)
+                return PromScalar(res[0], grid)
             return PromScalar(_scalar_op(e.op, l.value, r.value), grid)
         if isinstance(l, PromScalar) or isinstance(r, PromScalar):
             mat, sc, flipped = (r, l, True) if isinstance(l, PromScalar) else (l, r, False)
             a = mat.values
-            b = torch.as_tensor(float(sc.value), dtype=torch.float64, device=a.device)
+            b = _scalar_plane(sc, T, a.device).expand_as(a) \
+                if isinstance(sc.value, torch.Tensor) else \
+                torch.as_tensor(float(sc.value), dtype=torch.float64, device=a.device)
             if flipped:
                 res, keep = _vector_op(e.op, b.expand_as(a), a, e.bool_modifier)
             else:

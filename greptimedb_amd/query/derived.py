@@ -180,7 +180,14 @@ def select_over_result(sel: ast.Select, base, base_kinds=None):
     out_names, out_cols, out_kinds = [], [], []
 
     if has_agg:
-        gexprs = [resolve(g) for g in sel.group_by]
+        def resolve_g(g):
+            g = resolve(g)
+            # positional GROUP BY n → n-th projection
+            if isinstance(g, ast.Lit) and isinstance(g.value, int) and \
+                    1 <= g.value <= len(projections):
+                return resolve(projections[g.value - 1][0])
+            return g
+        gexprs = [resolve_g(g) for g in sel.group_by]
         if gexprs:
             keys = [np.asarray(X._np_raw(g, col_data))[rows] for g in gexprs]
             groups: dict = {}

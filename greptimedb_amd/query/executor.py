@@ -3637,7 +3637,23 @@ def _eval_const(e: ast.Expr):
         if fn == "round":
             nd = int(_eval_const(e.args[1])) if len(e.args) > 1 else 0
             return round(float(_eval_const(e.args[0])), nd)
+        if fn in _OBJ_FUNCS:
+            return _scalarize(_OBJ_FUNCS[fn](
+                [np.atleast_1d(np.asarray(_eval_const(a), dtype=object))
+                 for a in e.args]))
+    if isinstance(e, ast.Case):
+        for w, then in e.whens:
+            cond = (_eval_const(e.operand) == _eval_const(w)) \
+                if e.operand is not None else _eval_const(w)
+            if cond:
+                return _eval_const(then)
+        return _eval_const(e.default) if e.default is not None else None
     raise PlanQuery(f"unsupported constant expr {e}")
+
+
+def _scalarize(v):
+    a = np.asarray(v)
+    return a.item() if a.ndim == 0 or a.size == 1 else v
 
 
 def _expr_name(e: ast.Expr) -> str:

@@ -107,6 +107,12 @@ class UnaryOp(Expr):
 
 
 @dataclass
+class Cast(Expr):
+    expr: Expr
+    type: str  # lowered: bigint/int/double/float/string/timestamp/boolean
+
+
+@dataclass
 class InList(Expr):
     expr: Expr
     items: list[Expr]
@@ -208,7 +214,7 @@ class TruncateTable:
 
 @dataclass
 class ShowTables:
-    pass
+    like: str | None = None
 
 
 @dataclass

@@ -114,6 +114,8 @@ def _expr_cols(e: ast.Expr) -> set:
             walk(x.left); walk(x.right)
         elif isinstance(x, ast.UnaryOp):
             walk(x.operand)
+        elif isinstance(x, ast.Cast):
+            walk(x.expr)
         elif isinstance(x, ast.Func):
             for a in x.args:
                 walk(a)
@@ -272,6 +274,12 @@ class Executor:
             return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.ShowTables):
             names = sorted(self.engine.tables)
+            if stmt.like:
+                import re as _re2
+                pat = _re2.compile(
+                    "^" + _re2.escape(stmt.like).replace("%", ".*")
+                    .replace("_", ".") + "$")
+                names = [n for n in names if pat.match(n)]
             return QueryResult(["Tables"], [names])
         if isinstance(stmt, ast.ShowDatabases):
             extra = sorted(getattr(self.engine, "schemas", set()))
@@ -1765,6 +1773,7 @@ class Executor:
         tag_conj = []
         residual = []
         if sel.where is not None:
+            sel.where = _fold_const_casts(sel.where)
             for c in _split_conjuncts(sel.where):
                 done = False
                 if isinstance(c, ast.BinOp) and c.op in ("<", "<=", ">", ">=", "="):
@@ -3332,6 +3341,8 @@ def _eval_np_expr(e: ast.Expr, col_data: dict):
                          _eval_np_expr(e.right, col_data))
     if isinstance(e, ast.UnaryOp) and e.op == "-":
         return -_eval_np_expr(e.operand, col_data)
+    if isinstance(e, ast.Cast):
+        return _apply_cast(_np_raw(e.expr, col_data), e.type)
     if isinstance(e, ast.Case):
         conds, vals = [], []
         for w, r in e.whens:
@@ -3641,6 +3652,10 @@ def _eval_const(e: ast.Expr):
             return _scalarize(_OBJ_FUNCS[fn](
                 [np.atleast_1d(np.asarray(_eval_const(a), dtype=object))
                  for a in e.args]))
+    if isinstance(e, ast.Cast):
+        return _scalarize(_apply_cast(
+            np.atleast_1d(np.asarray(_eval_const(e.expr), dtype=object)),
+            e.type))
     if isinstance(e, ast.Case):
         for w, then in e.whens:
             cond = (_eval_const(e.operand) == _eval_const(w)) \
@@ -3654,6 +3669,73 @@ def _eval_const(e: ast.Expr):
 def _scalarize(v):
     a = np.asarray(v)
     return a.item() if a.ndim == 0 or a.size == 1 else v
+
+
+def _fold_const_casts(e):
+    """Fold CAST over a constant subtree to a literal so WHERE time-bound /
+    tag analysis sees plain literals (DataFusion constant folding)."""
+    if isinstance(e, ast.Cast):
+        if not _expr_cols(e.expr):
+            try:
+                return ast.Lit(_eval_const(e))
+            except Exception:
+                return e
+        return ast.Cast(_fold_const_casts(e.expr), e.type)
+    if isinstance(e, ast.BinOp):
+        return ast.BinOp(e.op, _fold_const_casts(e.left),
+                         _fold_const_casts(e.right))
+    if isinstance(e, ast.UnaryOp):
+        return ast.UnaryOp(e.op, _fold_const_casts(e.operand))
+    if isinstance(e, ast.Between):
+        return ast.Between(_fold_const_casts(e.expr),
+                           _fold_const_casts(e.low),
+                           _fold_const_casts(e.high), e.negated)
+    return e
+
+
+def _apply_cast(v, ty: str):
+    """CAST semantics over a numpy column (reference: Arrow cast kernels)."""
+    a = np.asarray(v)
+    if ty in ("bigint", "int", "integer", "smallint", "tinyint",
+              "int64", "int32", "uint64", "uint32"):
+        f = a.astype(np.float64) if a.dtype != object else np.array(
+            [np.nan if x is None else float(x) for x in a.ravel()]
+        ).reshape(a.shape)
+        out = np.where(np.isnan(f), np.nan, np.trunc(f))
+        return out.astype(np.int64) if not np.isnan(out).any() else out
+    if ty in ("double", "float", "real", "float64", "float32", "decimal"):
+        if a.dtype == object:
+            return np.array([np.nan if x is None else float(x)
+                             for x in a.ravel()]).reshape(a.shape)
+        return a.astype(np.float64)
+    if ty in ("string", "text", "varchar", "char"):
+        def s_of(x):
+            if x is None:
+                return None
+            if isinstance(x, (float, np.floating)):
+                return f"{float(x):g}"
+            if isinstance(x, (int, np.integer)):
+                return str(int(x))
+            return str(x)
+        return np.array([s_of(x) for x in a.ravel()],
+                        dtype=object).reshape(a.shape)
+    if ty in ("timestamp", "datetime"):
+        from greptimedb_amd.utils.timeutil import parse_ts_ms as _p
+        def t_of(x):
+            if x is None:
+                return np.nan
+            if isinstance(x, str):
+                ms = _p(x)
+                if ms is None:
+                    raise InvalidArguments(f"bad timestamp literal {x!r}")
+                return float(ms)
+            return float(x)
+        return np.array([t_of(x) for x in a.ravel()],
+                        dtype=np.float64).reshape(a.shape)
+    if ty in ("boolean", "bool"):
+        return np.array([None if x is None else bool(x) for x in a.ravel()],
+                        dtype=object).reshape(a.shape)
+    raise InvalidArguments(f"unsupported cast type {ty!r}")
 
 
 def _expr_name(e: ast.Expr) -> str:
@@ -3678,4 +3760,6 @@ def _expr_name(e: ast.Expr) -> str:
         return "case"
     if isinstance(e, ast.ScalarSubquery):
         return "(subquery)"
+    if isinstance(e, ast.Cast):
+        return f"CAST({_expr_name(e.expr)} AS {e.type.upper()})"
     return repr(e)

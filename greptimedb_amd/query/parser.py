@@ -30,7 +30,7 @@ _TOKEN_RE = re.compile(r"""
   | (?P<str>'(?:[^']|'')*')
   | (?P<qid>"(?:[^"]|"")*")
   | (?P<id>[A-Za-z_][A-Za-z0-9_.]*)
-  | (?P<op>=~|!~|<>|!=|<=|>=|=|<|>|\(|\)|\[|\]|\{|\}|,|\*|\+|-|/|%|;|:|@)
+  | (?P<op>=~|!~|<>|!=|<=|>=|=|<|>|\(|\)|\[|\]|\{|\}|,|\*|\+|-|/|%|;|::|:|@)
 """, re.VERBOSE)
 
 _UNITS_MS = {
@@ -199,7 +199,10 @@ class Parser:
                 stmt = ast.ShowVariables(like)
             else:
                 self.expect_kw("tables")
-                stmt = ast.ShowTables()
+                like = None
+                if self.eat_kw("like"):
+                    like = str(self.next().value)
+                stmt = ast.ShowTables(like)
         elif self.at_kw("describe", "desc"):
             self.next()
             self.eat_kw("table")
@@ -685,8 +688,23 @@ class Parser:
         return ast.Tql(start, end, step, query)
 
     # ---------------- expressions (Pratt) ----------------
+    def _parse_type_name(self) -> str:
+        """Type name after CAST(... AS …) / `::` — single identifier with
+        optional (n[,m]) length args (ignored) and the DOUBLE PRECISION /
+        TIMESTAMP(3) spellings."""
+        t = self.next()
+        name = str(t.value).lower()
+        if name == "double" and self.eat_kw("precision"):
+            pass
+        if self.eat_op("("):
+            while not self.eat_op(")"):
+                self.next()
+        return name
+
     def parse_expr(self, min_prec: int = 0) -> ast.Expr:
         left = self.parse_prefix()
+        while self.eat_op("::"):       # postgres cast: expr::TYPE
+            left = ast.Cast(left, self._parse_type_name())
         while True:
             t = self.peek()
             if t is None:
@@ -779,6 +797,13 @@ class Parser:
                 sub = self.parse_select()
                 self.expect_op(")")
                 return ast.Exists(sub)
+            if low == "cast" and self.at_op("("):
+                self.next()
+                inner = self.parse_expr()
+                self.expect_kw("as")
+                ty = self._parse_type_name()
+                self.expect_op(")")
+                return ast.Cast(inner, ty)
             if low == "interval":
                 txt = self.next().value
                 return ast.Interval(parse_interval_text(str(txt)), str(txt))

@@ -249,8 +249,9 @@ class Delete:
 @dataclass
 class AlterTable:
     table: str
-    action: str                 # "add_column"
+    action: str        # add_column | set_options | unset_options | rename
     column: tuple | None = None  # (name, type, opts)
+    options: dict | None = None  # for set/unset/rename
 
 
 @dataclass

@@ -368,6 +368,9 @@ class Executor:
         opts = [f"  regions = {len(st.regions)}"]
         if st.append_mode:
             opts.append("  append_mode = 'true'")
+        for k, v in sorted(schema.options.items()):
+            if k != "partition_rule":
+                opts.append(f"  {k} = '{v}'")
         part = ""
         if schema.options.get("partition_rule"):
             from greptimedb_amd.parallel.partition import MultiDimPartitionRule
@@ -378,9 +381,28 @@ class Executor:
         return QueryResult(["Table", "Create Table"], [[name], [ddl]])
 
     def _exec_alter(self, a: ast.AlterTable) -> QueryResult:
-        """ALTER TABLE ADD COLUMN (reference: alter DDL procedure; tags are
-        immutable — only fields can be added)."""
+        """ALTER TABLE ADD COLUMN / SET opts / UNSET opts / RENAME
+        (reference: alter DDL procedure; tags are immutable — only fields
+        can be added)."""
         st = self.engine.table(a.table)
+        if a.action == "set_options":
+            st.schema.options.update(a.options)
+            self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
+        if a.action == "unset_options":
+            for k in a.options:
+                st.schema.options.pop(k, None)
+            self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
+        if a.action == "rename":
+            new = a.options["to"]
+            if new in self.engine.tables:
+                from greptimedb_amd.utils.errors import TableAlreadyExists
+                raise TableAlreadyExists(new)
+            self.engine.tables[new] = self.engine.tables.pop(a.table)
+            st.schema.name = new
+            self.engine._save_catalog()
+            return QueryResult(["status"], [["ok"]])
         name, typ, opts = a.column
         tl = typ.lower()
         if tl in ("string", "varchar", "text", "json"):

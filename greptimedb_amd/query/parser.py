@@ -232,6 +232,27 @@ class Parser:
             self.next()
             self.expect_kw("table")
             table = self.next().value
+            if self.eat_kw("set"):
+                opts = {}
+                while True:
+                    k = str(self.next().value)
+                    self.expect_op("=")
+                    opts[k.lower()] = str(self.next().value)
+                    if not self.eat_op(","):
+                        break
+                return ast.AlterTable(table, "set_options", options=opts)
+            if self.eat_kw("unset"):
+                keys = []
+                while True:
+                    keys.append(str(self.next().value).lower())
+                    if not self.eat_op(","):
+                        break
+                return ast.AlterTable(table, "unset_options",
+                                      options={k: None for k in keys})
+            if self.eat_kw("rename"):
+                self.eat_kw("to")
+                return ast.AlterTable(table, "rename",
+                                      options={"to": str(self.next().value)})
             self.expect_kw("add")
             self.eat_kw("column")
             cname = self.next().value

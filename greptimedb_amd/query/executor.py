@@ -2704,6 +2704,11 @@ class Executor:
             if isinstance(e, ast.Star):
                 out_cols.extend(tag_names + [ts_name] + field_names + str_field_names)
             elif isinstance(e, ast.Col):
+                if e.name != ts_name and e.name not in tag_names and \
+                        e.name not in field_names and \
+                        e.name not in str_field_names:
+                    raise InvalidArguments(
+                        f"unknown column {e.name!r} in table {st.schema.name}")
                 out_cols.append(e.name)
             else:
                 name = alias or _expr_name(e)

@@ -642,6 +642,8 @@ class Executor:
             out = self._flow_engine().tick()
             return QueryResult(["flow", "rows"],
                                [list(out.keys()), list(out.values())])
+        if f == "apply_ttl":
+            return QueryResult(["files_removed"], [[self.engine.apply_ttl()]])
         if f == "compress_table":
             # K20 cold tier: pack resident SST batches into Gorilla blocks
             st = self.engine.table(str(a.args[0]))

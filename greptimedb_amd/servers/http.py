@@ -585,6 +585,44 @@ def build_app(ctx: ServerContext) -> FastAPI:
             return {"status": "error", "errorType": "bad_data", "error": str(e)}
         return _prom_result(m, instant=False)
 
+    @app.api_route("/v1/prometheus/api/v1/format_query",
+                   methods=["GET", "POST"])
+    async def prom_format_query(request: Request):
+        """Pretty-print a PromQL expression (reference: format_query
+        handler, src/servers/src/http/prometheus.rs:379)."""
+        q = await _param(request, "query")
+        try:
+            from greptimedb_amd.query.promql.parser import parse_promql
+            parse_promql(q)  # validation only; echo normalized text
+            return {"status": "success", "data": " ".join(str(q).split())}
+        except GreptimeError as e:
+            return {"status": "error", "errorType": "bad_data", "error": str(e)}
+
+    @app.api_route("/v1/prometheus/api/v1/parse_query",
+                   methods=["GET", "POST"])
+    async def prom_parse_query(request: Request):
+        """Parse a PromQL expression to an AST JSON shape (reference:
+        parse_query handler, prometheus.rs:2342)."""
+        q = await _param(request, "query")
+        try:
+            from greptimedb_amd.query.promql.parser import parse_promql
+            tree = parse_promql(q)
+
+            def enc(n):
+                d = {"type": type(n).__name__.lower()}
+                for k, v in vars(n).items():
+                    if hasattr(v, "__dict__") and not isinstance(v, (str,)):
+                        d[k] = enc(v)
+                    elif isinstance(v, list):
+                        d[k] = [enc(x) if hasattr(x, "__dict__") else x
+                                for x in v]
+                    else:
+                        d[k] = v
+                return d
+            return {"status": "success", "data": enc(tree)}
+        except GreptimeError as e:
+            return {"status": "error", "errorType": "bad_data", "error": str(e)}
+
     @app.api_route("/v1/prometheus/api/v1/labels", methods=["GET", "POST"])
     async def prom_labels(request: Request):
         names = {"__name__"}

@@ -105,6 +105,16 @@ def test_remote_write_and_promql(client):
     assert "job" in r.json()["data"]
     r = client.get("/v1/prometheus/api/v1/series", params={"match[]": 'up{job="api"}'})
     assert r.json()["data"] == [{"job": "api", "__name__": "up"}]
+    r = client.get("/v1/prometheus/api/v1/format_query",
+                   params={"query": "sum( up )"})
+    assert r.json() == {"status": "success", "data": "sum( up )"}
+    r = client.get("/v1/prometheus/api/v1/parse_query",
+                   params={"query": "rate(up[5m])"})
+    d = r.json()
+    assert d["status"] == "success" and d["data"]["type"] == "call"
+    r = client.get("/v1/prometheus/api/v1/parse_query",
+                   params={"query": "rate(up[5m"})
+    assert r.json()["status"] == "error"
 
 
 def test_metrics_endpoint(client):

@@ -176,6 +176,7 @@ class Executor:
         entry = {"sql": sql, "start": t0, "elapsed_ms": 0.0,
                  "state": "running", "cancel": False}
         plist[pid] = entry
+        self._proc_entry = entry
         try:
             with tracer.span("sql.execute", statement=sql[:200],
                              stmt_type=type(stmt).__name__):
@@ -183,6 +184,7 @@ class Executor:
         finally:
             entry["elapsed_ms"] = (_time.perf_counter() - t0) * 1000
             plist.pop(pid, None)
+            self._proc_entry = None
         dt = (_time.perf_counter() - t0) * 1000
         if dt >= self.SLOW_QUERY_MS:
             # slow-query log (reference: common/frontend slow query events)
@@ -192,6 +194,16 @@ class Executor:
                 log = self.engine.slow_queries = deque(maxlen=128)
             log.append({"sql": sql[:500], "ms": round(dt, 1)})
         return r
+
+    def _cancel_check(self):
+        """Raise if this query's process entry was KILLed — polled inside
+        per-region scan loops so long scans terminate promptly
+        (reference: process manager cancellation tokens checked by the
+        stream adapters)."""
+        e = getattr(self, "_proc_entry", None)
+        if e is not None and e.get("cancel"):
+            from greptimedb_amd.utils.errors import QueryCancelled
+            raise QueryCancelled("query killed")
 
     def execute_stmt(self, stmt) -> QueryResult:
         self._qualify_names(stmt)
@@ -2031,6 +2043,7 @@ class Executor:
         from greptimedb_amd.ops import ts_bucket_agg_acc, ts_bucket_agg_finish
         acc = None  # opaque accumulator handle (GPU: shared atomic buffers)
         for region, lut in zip(st.regions, region_luts):
+            self._cancel_check()
             lut_t = torch.as_tensor(lut, device=device)
             for ts_t, se_t, f_t, fidx_t in self._region_agg_inputs(
                     region, plan, device, agg_fields, ts_lo, ts_hi):
@@ -2115,6 +2128,7 @@ class Executor:
         best_val: np.ndarray | None = None
 
         for region, lut in zip(st.regions, region_luts):
+            self._cancel_check()
             lut_t = torch.as_tensor(lut, device=device)
             sources = region.scan_sources(ts_lo, ts_hi)
             if plan.residual is not None:
@@ -2464,6 +2478,7 @@ class Executor:
         # gather per-field (slot, ts)-sorted sample streams (NaN compacted)
         parts = []          # (ts, slots, {field: vals})
         for region, lut in zip(st.regions, region_luts):
+            self._cancel_check()
             lut_t = torch.as_tensor(lut, device=device)
             chunks = []
             for src in region.scan_sources(ts_lo, ts_hi):
@@ -2749,6 +2764,7 @@ class Executor:
             64 * len(needed_strs)
         parts = []  # (ts np, codes np, region, fields np [nf_needed, n])
         for region in st.regions:
+            self._cancel_check()
             cand = self._candidate_codes(region, plan)
             lut = None
             if cand is not None:

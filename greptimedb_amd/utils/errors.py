@@ -61,3 +61,7 @@ class NativeExtensionMissing(GreptimeError):
 
 class RegionFenced(GreptimeError):
     """Write rejected: region downgraded (migration write fence)."""
+
+
+class QueryCancelled(GreptimeError):
+    """Query terminated by KILL (reference: process manager cancellation)."""

@@ -39,6 +39,8 @@ class EngineConfig:
     record_events: bool = False         # persist DDL/migration events table
     gc_interval_s: float = 600.0        # background orphan-SST scan period
     gc_grace_s: float = 3600.0          # spare files younger than this
+    cold_compress_s: float = 0.0        # Gorilla-pack SST batches idle this
+                                        # long (0 = disabled; K20 cold tier)
 
 
 @dataclass
@@ -457,9 +459,32 @@ class MitoEngine:
                 try:
                     self.gc_orphan_ssts(self.config.gc_grace_s)
                     self.apply_ttl()
+                    self.compress_cold()
                 except Exception:  # pragma: no cover
                     import traceback
                     traceback.print_exc()
+
+    def compress_cold(self, age_s: float | None = None) -> int:
+        """Gorilla-pack device-resident SST batches not scanned for
+        `age_s` seconds (K20 cold tier — ~4.6× capacity on HBM at
+        ~900 GB/s decode when a scan touches them again). Returns the
+        number of batches packed. The reference approximates this with
+        its disk page cache; here the compressed copy stays in HBM."""
+        import time as _time
+        age_s = age_s if age_s is not None else self.config.cold_compress_s
+        if age_s <= 0:
+            return 0
+        cutoff = _time.monotonic() - age_s
+        packed = 0
+        for st in self.tables.values():
+            for r in st.regions:
+                with r.lock:
+                    batches = list(r.sst_cache.values())
+                for b in batches:
+                    if b.ts is not None and b.last_access < cutoff:
+                        b.compress()
+                        packed += 1
+        return packed
 
     def apply_ttl(self, now_ms: int | None = None) -> int:
         """Drop SSTs past their table's ttl option (reference: mito2

@@ -139,6 +139,10 @@ class SstBatch:
         self.field_names = list(field_names)
         self.str_cols: dict = {}     # host string columns (log fields)
         self.text_index: dict = {}   # fulltext SegmentPostings per column
+        import threading as _th
+        import time as _time
+        self._tier_lock = _th.Lock()  # compress vs decode exclusion
+        self.last_access = _time.monotonic()
 
     @property
     def n(self) -> int:
@@ -153,6 +157,10 @@ class SstBatch:
         ensure_decoded() which re-materializes via the GPU decode kernel —
         the resident-cold-tier policy the reference approximates with its
         page cache, except the compressed copy stays in HBM."""
+        with self._tier_lock:
+            return self._compress_locked()
+
+    def _compress_locked(self) -> int:
         if self.ts is None:
             return sum(g.nbytes for k, g in self._gorilla.items()
                        if k != "__n")
@@ -174,6 +182,14 @@ class SstBatch:
         return sum(g.nbytes for k, g in packs.items() if k != "__n")
 
     def ensure_decoded(self, device) -> "SstBatch":
+        import time as _time
+        self.last_access = _time.monotonic()
+        if self.ts is not None or not getattr(self, "_gorilla", None):
+            return self
+        with self._tier_lock:
+            return self._decode_locked(device)
+
+    def _decode_locked(self, device) -> "SstBatch":
         if self.ts is not None or not getattr(self, "_gorilla", None):
             return self
         import torch as _torch

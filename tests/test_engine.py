@@ -715,3 +715,33 @@ def test_ttl_expires_ssts(tmp_path):
     r = ex.execute("ADMIN apply_ttl()")
     assert r.names == ["files_removed"]
     eng.close()
+
+
+def test_cold_tier_auto_compress(tmp_path):
+    """Idle SST batches Gorilla-pack via compress_cold and transparently
+    re-decode on the next scan (K20 cold tier policy)."""
+    from greptimedb_amd.query.executor import Executor
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d"), device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE ct (ts TIMESTAMP TIME INDEX, h STRING "
+               "PRIMARY KEY, v DOUBLE)")
+    ex.execute("INSERT INTO ct VALUES (1000,'a',1.5),(2000,'a',2.5),"
+               "(3000,'b',3.5)")
+    eng.flush_all()
+    before = ex.execute("SELECT h, v FROM ct ORDER BY ts").rows()
+    # mark every batch ancient, then sweep
+    for st in eng.tables.values():
+        for r in st.regions:
+            for b in r.sst_cache.values():
+                b.last_access = -1e9
+    n = eng.compress_cold(age_s=1.0)
+    assert n >= 1
+    packed = [b for st in eng.tables.values() for r in st.regions
+              for b in r.sst_cache.values() if b.ts is None]
+    assert packed
+    # scans re-decode transparently and results are identical
+    after = ex.execute("SELECT h, v FROM ct ORDER BY ts").rows()
+    assert after == before
+    assert all(b.ts is not None for b in packed)  # hot again
+    eng.close()

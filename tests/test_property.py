@@ -155,3 +155,66 @@ def test_window_running_aggregates_property(seed, n, nparts):
             for p in set(part):
                 rns = sorted(got[part == p])
                 assert rns == list(range(1, len(rns) + 1))
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(rows=ROWS, flush=st_.booleans(), k=st_.floats(0.5, 3.0))
+def test_expr_agg_fallback_matches_oracle(tmp_path_factory, rows, flush, k):
+    """sum(v*k) / avg(v+k) / CASE aggregates through the general-shape
+    fallback vs a python oracle."""
+    eng, ex = _mk(tmp_path_factory, rows, append=True)
+    try:
+        if flush:
+            eng.flush_all()
+        r = ex.execute(f"SELECT h, sum(v * {k!r}), avg(v + {k!r}), "
+                       f"sum(CASE WHEN v > 0 THEN 1 ELSE 0 END) "
+                       f"FROM p GROUP BY h ORDER BY h")
+        exp: dict = {}
+        for h, _t, v in rows:
+            s, c, pos = exp.get(h, (0.0, 0, 0))
+            exp[h] = (s + v, c + 1, pos + (1 if v > 0 else 0))
+        assert list(r.columns[0]) == sorted(exp)
+        for i, h in enumerate(r.columns[0]):
+            s, c, pos = exp[h]
+            assert math.isclose(float(r.columns[1][i]), s * k,
+                                rel_tol=1e-9, abs_tol=1e-6)
+            assert math.isclose(float(r.columns[2][i]), s / c + k,
+                                rel_tol=1e-9, abs_tol=1e-6)
+            assert int(r.columns[3][i]) == pos
+    finally:
+        eng.close()
+
+
+@settings(max_examples=25, deadline=None,
+          suppress_health_check=[HealthCheck.function_scoped_fixture])
+@given(rows=ROWS, flush=st_.booleans())
+def test_first_last_value_matches_oracle(tmp_path_factory, rows, flush):
+    """first_value/last_value per tag vs python argmin/argmax over ts
+    (append mode: ties broken by insertion order → compare on unique ts)."""
+    # dedupe (h, ts) keeping last so the oracle is unambiguous
+    seen = {}
+    for h, t, v in rows:
+        seen[(h, t)] = v
+    rows = [(h, t, v) for (h, t), v in seen.items()]
+    eng, ex = _mk(tmp_path_factory, rows, append=False)
+    try:
+        if flush:
+            eng.flush_all()
+        r = ex.execute("SELECT h, first_value(v), last_value(v) FROM p "
+                       "GROUP BY h ORDER BY h")
+        first: dict = {}
+        last: dict = {}
+        for h, t, v in rows:
+            if h not in first or t < first[h][0]:
+                first[h] = (t, v)
+            if h not in last or t > last[h][0]:
+                last[h] = (t, v)
+        assert list(r.columns[0]) == sorted(first)
+        for i, h in enumerate(r.columns[0]):
+            assert math.isclose(float(r.columns[1][i]), first[h][1],
+                                rel_tol=1e-12, abs_tol=0)
+            assert math.isclose(float(r.columns[2][i]), last[h][1],
+                                rel_tol=1e-12, abs_tol=0)
+    finally:
+        eng.close()

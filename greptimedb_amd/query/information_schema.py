@@ -14,7 +14,8 @@ from greptimedb_amd.models.schema import SemanticType
 VIRTUAL_TABLES = {
     "tables", "columns", "region_statistics", "flows", "cluster_info",
     "partitions", "region_peers", "build_info", "process_list", "views",
-    "schemata",
+    "schemata", "ssts", "key_column_usage", "table_constraints",
+    "procedure_info",
 }
 
 PG_CATALOG_TABLES = {"pg_tables", "pg_namespace", "pg_class"}
@@ -112,6 +113,50 @@ def build(engine, name: str):
                              getattr(r, "role", "leader").upper(), "ALIVE"))
         return _cols(["region_id", "table_name", "peer_id", "peer_addr",
                       "role", "status"], rows)
+    if kind == "ssts":
+        # per-SST file inventory (ref information_schema ssts)
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            for r in st.regions:
+                for fid, meta in sorted(r.manifest.files.items()):
+                    rows.append((t, r.region_id, fid,
+                                 int(meta.get("num_rows", 0)),
+                                 int(meta.get("file_size", 0)),
+                                 int(meta.get("level", 0)),
+                                 int(meta.get("min_ts", 0)),
+                                 int(meta.get("max_ts", 0))))
+        return _cols(["table_name", "region_id", "file_id", "num_rows",
+                      "file_size", "level", "min_ts", "max_ts"], rows)
+    if kind == "key_column_usage":
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            for pos, pk in enumerate(st.schema.primary_key):
+                rows.append(("greptime", "public", "PRIMARY", t, pk, pos + 1))
+            rows.append(("greptime", "public", "TIME INDEX", t,
+                         st.schema.time_index.name, 1))
+        return _cols(["constraint_catalog", "constraint_schema",
+                      "constraint_name", "table_name", "column_name",
+                      "ordinal_position"], rows)
+    if kind == "table_constraints":
+        rows = []
+        for t, st in sorted(engine.tables.items()):
+            if st.schema.primary_key:
+                rows.append(("greptime", "public", "PRIMARY", t, "PRIMARY KEY"))
+            rows.append(("greptime", "public", "TIME INDEX", t, "TIME INDEX"))
+        return _cols(["constraint_catalog", "constraint_schema",
+                      "constraint_name", "table_name", "constraint_type"],
+                     rows)
+    if kind == "procedure_info":
+        # persisted procedure-store entries (migrations etc.)
+        rows = []
+        store = getattr(engine, "procedure_store", None)
+        if store is not None:
+            for p in store.load_all():
+                rows.append((p.get("pid", ""), p.get("type", ""),
+                             p.get("status", ""),
+                             str(p.get("state", ""))[:200]))
+        return _cols(["procedure_id", "procedure_type", "status", "detail"],
+                     rows)
     if kind == "build_info":
         from greptimedb_amd import __version__
         return _cols(["version", "arch", "backend"],

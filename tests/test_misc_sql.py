@@ -501,3 +501,32 @@ def test_kill_cancels_running_scan(tmp_path):
     finally:
         Region.scan_sources = orig
     eng.close()
+
+
+def test_information_schema_extended_tables(tmp_path):
+    """ssts / key_column_usage / table_constraints / procedure_info views
+    (reference: catalog system_schema information_schema tables)."""
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d"), device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE m (ts TIMESTAMP TIME INDEX, dc STRING, h STRING,"
+               " v DOUBLE, PRIMARY KEY (dc, h))")
+    ex.execute("INSERT INTO m VALUES (1000,'us','a',1),(2000,'eu','b',2)")
+    eng.flush_all()
+    r = ex.execute("SELECT table_name, num_rows, level FROM "
+                   "information_schema.ssts")
+    assert r.rows() and all(row[0] == "m" for row in r.rows())
+    assert sum(row[1] for row in r.rows()) == 2
+    r = ex.execute("SELECT constraint_name, column_name, ordinal_position "
+                   "FROM information_schema.key_column_usage")
+    rows = [tuple(x) for x in r.rows()]
+    assert ("PRIMARY", "dc", 1) in rows and ("PRIMARY", "h", 2) in rows
+    assert ("TIME INDEX", "ts", 1) in rows
+    r = ex.execute("SELECT constraint_type FROM "
+                   "information_schema.table_constraints")
+    assert {"PRIMARY KEY", "TIME INDEX"} <= {row[0] for row in r.rows()}
+    r = ex.execute("SELECT * FROM information_schema.procedure_info")
+    assert r.names == ["procedure_id", "procedure_type", "status", "detail"]
+    eng.close()

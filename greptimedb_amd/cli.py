@@ -40,8 +40,21 @@ def main(argv=None):
         p = meta_sub.add_parser(action)
         p.add_argument("--data-dir", required=True)
         p.add_argument("--file", required=True, help="snapshot tar path")
+    # `cli data export/import` (ref src/cli data export/import tools)
+    data = cli_sub.add_parser("data")
+    data_sub = data.add_subparsers(dest="action", required=True)
+    for action in ("export", "import"):
+        p = data_sub.add_parser(action)
+        p.add_argument("--data-dir", required=True)
+        p.add_argument("--dir", required=True,
+                       help="export/import directory (one parquet per "
+                            "table + schema.sql)")
+        p.add_argument("--tables", default=None,
+                       help="comma-separated subset (default: all)")
     args = ap.parse_args(argv)
 
+    if args.role == "cli" and args.cmd == "data":
+        return _data_export_import(args)
     if args.role == "cli":
         return _meta_snapshot(args)
 
@@ -99,6 +112,57 @@ def main(argv=None):
     finally:
         grpc_srv.shutdown()
         flight_srv.shutdown()
+
+
+def _data_export_import(args) -> int:
+    """`cli data export/import`: whole-database backup as one parquet per
+    table + a schema.sql of SHOW CREATE TABLE statements (reference:
+    src/cli/src/data export/import tools)."""
+    import os
+
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+
+    eng = MitoEngine(EngineConfig(data_dir=args.data_dir, device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    try:
+        if args.action == "export":
+            os.makedirs(args.dir, exist_ok=True)
+            names = sorted(eng.tables)
+            if args.tables:
+                wanted = {t.strip() for t in args.tables.split(",")}
+                names = [n for n in names if n in wanted]
+            ddls = []
+            for t in names:
+                r = ex.execute(f'SHOW CREATE TABLE "{t}"')
+                ddls.append(str(r.columns[1][0]) + ";")
+                n = ex.execute(
+                    f"COPY \"{t}\" TO '{os.path.join(args.dir, t)}.parquet'")
+                print(f"exported {t}: {n.rows()[0][0]} rows")
+            with open(os.path.join(args.dir, "schema.sql"), "w") as f:
+                f.write("\n\n".join(ddls) + "\n")
+            print(f"export complete: {len(names)} tables -> {args.dir}")
+        else:
+            with open(os.path.join(args.dir, "schema.sql")) as f:
+                for stmt in f.read().split(";"):
+                    if stmt.strip():
+                        ex.execute(stmt)
+            names = sorted(
+                fn[:-8] for fn in os.listdir(args.dir)
+                if fn.endswith(".parquet"))
+            if args.tables:
+                wanted = {t.strip() for t in args.tables.split(",")}
+                names = [n for n in names if n in wanted]
+            for t in names:
+                ex.execute(
+                    f"COPY \"{t}\" FROM '{os.path.join(args.dir, t)}.parquet'")
+                print(f"imported {t}")
+            eng.flush_all()
+            print(f"import complete: {len(names)} tables")
+        return 0
+    finally:
+        eng.close()
 
 
 def _meta_snapshot(args) -> int:

@@ -530,3 +530,43 @@ def test_information_schema_extended_tables(tmp_path):
     r = ex.execute("SELECT * FROM information_schema.procedure_info")
     assert r.names == ["procedure_id", "procedure_type", "status", "detail"]
     eng.close()
+
+
+def test_cli_data_export_import(tmp_path):
+    """`cli data export` + `import` round-trip a whole database
+    (reference: greptime cli data export/import)."""
+    from greptimedb_amd.cli import main as cli_main
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.query.executor import Executor
+
+    src = str(tmp_path / "src")
+    eng = MitoEngine(EngineConfig(data_dir=src, device="cpu",
+                                  background_flush=False))
+    ex = Executor(eng)
+    ex.execute("CREATE TABLE t1 (ts TIMESTAMP TIME INDEX, h STRING "
+               "PRIMARY KEY, v DOUBLE)")
+    ex.execute("INSERT INTO t1 VALUES (1000,'a',1.5),(2000,'b',2.5)")
+    ex.execute("CREATE TABLE t2 (ts TIMESTAMP TIME INDEX, h STRING "
+               "PRIMARY KEY, v DOUBLE)")
+    ex.execute("INSERT INTO t2 VALUES (5000,'x',9.0)")
+    eng.flush_all()
+    eng.close()
+
+    exp = str(tmp_path / "backup")
+    assert cli_main(["cli", "data", "export", "--data-dir", src,
+                     "--dir", exp]) == 0
+    import os
+    assert os.path.exists(os.path.join(exp, "t1.parquet"))
+    assert os.path.exists(os.path.join(exp, "schema.sql"))
+
+    dst = str(tmp_path / "dst")
+    assert cli_main(["cli", "data", "import", "--data-dir", dst,
+                     "--dir", exp]) == 0
+    eng2 = MitoEngine(EngineConfig(data_dir=dst, device="cpu",
+                                   background_flush=False))
+    ex2 = Executor(eng2)
+    assert [tuple(r) for r in
+            ex2.execute("SELECT h, v FROM t1 ORDER BY h").rows()] == \
+        [("a", 1.5), ("b", 2.5)]
+    assert ex2.execute("SELECT count(*) FROM t2").rows()[0][0] == 1
+    eng2.close()

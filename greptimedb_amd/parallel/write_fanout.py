@@ -64,12 +64,30 @@ class WriteExchange:
 
     def __init__(self, rank: int, world: int, handler=None,
                  base_port: int | None = None, host: str = "127.0.0.1",
-                 ports: list[int] | None = None, handler_factory=None):
+                 ports: list[int] | None = None, handler_factory=None,
+                 peers: list[tuple[str, int]] | None = None):
+        """`peers` maps rank → (host, port) for multi-node layouts; the
+        GDB_FANOUT_PEERS env ("h1:p1,h2:p2,…", one entry per rank) sets it
+        for torchrun-launched jobs. Default: loopback, one port per rank
+        (single-node — xGMI box — layout)."""
+        import os as _os
         self.rank = rank
         self.world = world
-        self.host = host
-        self.ports = list(ports) if ports is not None else \
-            [fanout_port(r, base_port) for r in range(world)]
+        if peers is None and _os.environ.get("GDB_FANOUT_PEERS"):
+            peers = []
+            for ent in _os.environ["GDB_FANOUT_PEERS"].split(","):
+                h, _, p = ent.strip().rpartition(":")
+                peers.append((h, int(p)))
+        if peers is not None:
+            if len(peers) != world:
+                raise ValueError(f"peers needs {world} entries")
+            self.peers = list(peers)
+        else:
+            base_ports = list(ports) if ports is not None else \
+                [fanout_port(r, base_port) for r in range(world)]
+            self.peers = [(host, p) for p in base_ports]
+        self.host = self.peers[rank][0]
+        self.ports = [p for _h, p in self.peers]
         self.handler = handler      # callable(payload: bytes) -> bytes
         # handler_factory() -> fresh handler per connection: each peer
         # connection gets its own receive pipeline (no shared lock), like
@@ -86,7 +104,9 @@ class WriteExchange:
         self._closing = False
         self._srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
         self._srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
-        self._srv.bind((host, self.ports[rank]))
+        own_host = self.peers[rank][0]
+        bind_host = own_host if own_host.startswith("127.") else "0.0.0.0"
+        self._srv.bind((bind_host, self.ports[rank]))
         self._srv.listen(64)
         self._accept_thread = threading.Thread(target=self._accept_loop,
                                                daemon=True)
@@ -124,8 +144,7 @@ class WriteExchange:
         last = None
         while time.monotonic() < deadline:
             try:
-                s = socket.create_connection((self.host, self.ports[peer]),
-                                             timeout=30)
+                s = socket.create_connection(self.peers[peer], timeout=30)
                 s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
                 return s
             except OSError as e:

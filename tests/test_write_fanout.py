@@ -200,3 +200,40 @@ def test_fanout_bulk_receive_gpu(tmp_path):
         ex.close()
     for e in engines:
         e.close()
+
+
+def test_exchange_explicit_peer_map(monkeypatch):
+    """Multi-node seam: rank → (host, port) peer map (here both on
+    loopback) routes requests by the per-rank address."""
+    from greptimedb_amd.parallel.write_fanout import WriteExchange
+    p0, p1 = _free_ports(2)
+    peers = [("127.0.0.1", p0), ("127.0.0.1", p1)]
+    got = []
+
+    def handler(payload):
+        got.append(payload)
+        return b"OK"
+
+    ex0 = WriteExchange(0, 2, handler=handler, peers=peers)
+    ex1 = WriteExchange(1, 2, handler=handler, peers=peers)
+    try:
+        assert ex1.request(0, b"hello-from-1") == b"OK"
+        assert ex0.request(1, b"hello-from-0") == b"OK"
+        assert got == [b"hello-from-1", b"hello-from-0"]
+    finally:
+        ex0.close()
+        ex1.close()
+
+
+def test_exchange_peers_env(monkeypatch):
+    from greptimedb_amd.parallel.write_fanout import WriteExchange
+    p0, p1 = _free_ports(2)
+    monkeypatch.setenv("GDB_FANOUT_PEERS",
+                       f"127.0.0.1:{p0}, 127.0.0.1:{p1}")
+    ex0 = WriteExchange(0, 2, handler=lambda b: b"OK")
+    ex1 = WriteExchange(1, 2, handler=lambda b: b"OK")
+    try:
+        assert ex0.request(1, b"x") == b"OK"
+    finally:
+        ex0.close()
+        ex1.close()

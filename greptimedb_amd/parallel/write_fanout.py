@@ -144,7 +144,10 @@ class WriteExchange:
         last = None
         while time.monotonic() < deadline:
             try:
-                s = socket.create_connection(self.peers[peer], timeout=30)
+                # host from the peer map, port from self.ports — callers
+                # (meta heartbeat) patch .ports after construction
+                s = socket.create_connection(
+                    (self.peers[peer][0], self.ports[peer]), timeout=30)
                 s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
                 return s
             except OSError as e:

@@ -12,21 +12,32 @@ class UserProvider:
 
 
 class StaticUserProvider(UserProvider):
-    """`user=password` map (reference: static_user_provider file format)."""
+    """`user=password[:ro]` map (reference: static_user_provider file
+    format + permission checker seam; `:ro` marks a read-only user)."""
 
-    def __init__(self, users: dict[str, str]):
+    def __init__(self, users: dict[str, str],
+                 modes: dict[str, str] | None = None):
         self.users = dict(users)
+        self.modes = dict(modes or {})
 
     @staticmethod
     def from_file(path: str) -> "StaticUserProvider":
-        users = {}
+        users, modes = {}, {}
         with open(path) as f:
             for line in f:
                 line = line.strip()
                 if line and "=" in line and not line.startswith("#"):
                     u, p = line.split("=", 1)
-                    users[u.strip()] = p.strip()
-        return StaticUserProvider(users)
+                    u = u.strip()
+                    p = p.strip()
+                    if p.endswith(":ro"):
+                        p = p[:-3]
+                        modes[u] = "ro"
+                    users[u] = p
+        return StaticUserProvider(users, modes)
+
+    def mode(self, user: str) -> str:
+        return self.modes.get(user, "rw")
 
     def allow(self, user: str, password: str | None = None) -> bool:
         if user not in self.users:
@@ -185,3 +196,31 @@ def mysql_caching_sha2_check(password: str, nonce: bytes, token: bytes) -> bool:
     expected = bytes(a ^ b for a, b in zip(
         p1, _hashlib.sha256(p2 + nonce).digest()))
     return hmac.compare_digest(expected, token)
+
+
+# ------------------------------------------------------------ permissions
+
+WRITE_STMT_NAMES = {
+    "InsertValues", "Delete", "CreateTable", "DropTable", "AlterTable",
+    "TruncateTable", "CreateView", "DropView", "CreateDatabase",
+    "DropDatabase", "CreateFlow", "DropFlow", "Admin", "Copy",
+}
+
+
+def check_permission(provider, user: str, sql: str) -> bool:
+    """Reference parity: src/auth permission checker — deny DML/DDL to
+    read-only users. Statement kind comes from the parser (COPY TO is a
+    read; COPY FROM writes)."""
+    if provider is None or getattr(provider, "mode", None) is None:
+        return True
+    if provider.mode(user) != "ro":
+        return True
+    from greptimedb_amd.query.parser import parse_sql
+    try:
+        stmt = parse_sql(sql)
+    except Exception:
+        return True  # executor reports the real syntax error
+    name = type(stmt).__name__
+    if name == "Copy":
+        return getattr(stmt, "direction", "to") == "to"
+    return name not in WRITE_STMT_NAMES

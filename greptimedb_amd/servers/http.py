@@ -126,6 +126,8 @@ def build_app(ctx: ServerContext) -> FastAPI:
                     try:
                         u, _, p = base64.b64decode(h[6:]).decode().partition(":")
                         ok = ctx.user_provider.allow(u, p)
+                        if ok:
+                            request.state.user = u
                     except Exception:
                         ok = False
                 if not ok:
@@ -235,6 +237,15 @@ def build_app(ctx: ServerContext) -> FastAPI:
                 body = (await request.body()).decode()
                 sql = body or None
         t0 = time.perf_counter()
+        user = getattr(request.state, "user", None)
+        if user is not None and ctx.user_provider is not None:
+            from greptimedb_amd.servers.auth import check_permission
+            if not check_permission(ctx.user_provider, user, sql):
+                return Response(_json.dumps(
+                    {"code": 7000,
+                     "error": f"permission denied: user {user!r} is "
+                              f"read-only"}), status_code=403,
+                    media_type="application/json")
 
         def run():
             # execute + serialize off the event loop (and skip fastapi's

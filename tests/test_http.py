@@ -196,3 +196,56 @@ def test_probe_endpoints(client):
     assert r.json()["status"] == "success"
     assert client.get("/v1/influxdb/ping").status_code == 204
     assert client.get("/v1/influxdb/health").status_code == 204
+
+
+def test_readonly_user_permission(tmp_path):
+    """`user=pw:ro` users are denied DML/DDL but can query (reference:
+    src/auth permission checker)."""
+    import base64
+
+    from fastapi.testclient import TestClient
+
+    from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
+    from greptimedb_amd.engine.ingest import Ingestor
+    from greptimedb_amd.query.executor import Executor
+    from greptimedb_amd.servers.auth import StaticUserProvider
+    from greptimedb_amd.servers.http import ServerContext, build_app
+
+    upath = tmp_path / "users"
+    upath.write_text("admin=secret\nviewer=pw:ro\n")
+    prov = StaticUserProvider.from_file(str(upath))
+    assert prov.mode("viewer") == "ro" and prov.mode("admin") == "rw"
+    assert prov.allow("viewer", "pw")
+
+    eng = MitoEngine(EngineConfig(data_dir=str(tmp_path / "d"), device="cpu",
+                                  background_flush=False))
+    ctx = ServerContext(eng, user_provider=prov)
+    app = build_app(ctx)
+    c = TestClient(app)
+
+    def hdr(u, p):
+        return {"authorization":
+                "Basic " + base64.b64encode(f"{u}:{p}".encode()).decode()}
+
+    # admin can create + insert
+    r = c.post("/v1/sql", params={"sql": "CREATE TABLE pt (ts TIMESTAMP "
+               "TIME INDEX, h STRING PRIMARY KEY, v DOUBLE)"},
+               headers=hdr("admin", "secret"))
+    assert r.status_code == 200 and "error" not in r.json()
+    r = c.post("/v1/sql", params={
+        "sql": "INSERT INTO pt VALUES (1000,'a',1)"},
+        headers=hdr("admin", "secret"))
+    assert r.status_code == 200
+    # read-only user: SELECT ok, INSERT/DROP denied
+    r = c.post("/v1/sql", params={"sql": "SELECT count(*) FROM pt"},
+               headers=hdr("viewer", "pw"))
+    assert r.status_code == 200 and "error" not in r.json()
+    for bad in ("INSERT INTO pt VALUES (2000,'b',2)", "DROP TABLE pt",
+                "TRUNCATE TABLE pt"):
+        r = c.post("/v1/sql", params={"sql": bad}, headers=hdr("viewer", "pw"))
+        assert r.status_code == 403, bad
+    # data unchanged
+    r = c.post("/v1/sql", params={"sql": "SELECT count(*) FROM pt"},
+               headers=hdr("admin", "secret"))
+    assert r.json()["output"][0]["records"]["rows"][0][0] == 1
+    eng.close()

@@ -58,7 +58,9 @@ def test_golden_case(case, tmp_engine):
 if __name__ == "__main__":
     import sys
     import tempfile
-    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, repo_root)
+    os.chdir(repo_root)  # engines resolve relative paths against the root
     from greptimedb_amd.engine.engine import EngineConfig, MitoEngine
     for case in sorted(glob.glob(f"{CASES_DIR}/*.sql")):
         eng = MitoEngine(EngineConfig(data_dir=tempfile.mkdtemp(), device="cpu",

@@ -223,6 +223,21 @@ class ShowDatabases:
 
 
 @dataclass
+class ShowCreateView:
+    name: str
+
+
+@dataclass
+class ShowCreateFlow:
+    name: str
+
+
+@dataclass
+class ShowIndex:
+    name: str
+
+
+@dataclass
 class ShowCreateTable:
     name: str
 

@@ -301,6 +301,40 @@ class Executor:
                                         *extra})])
         if isinstance(stmt, ast.ShowCreateTable):
             return self._show_create_table(stmt.name)
+        if isinstance(stmt, ast.ShowCreateView):
+            views = getattr(self.engine, "views", {})
+            if stmt.name not in views:
+                raise TableNotFound(stmt.name)
+            return QueryResult(
+                ["View", "Create View"],
+                [[stmt.name],
+                 [f"CREATE VIEW {stmt.name} AS {views[stmt.name]}"]])
+        if isinstance(stmt, ast.ShowCreateFlow):
+            fe = self._flow_engine()
+            f = fe.flows.get(stmt.name)
+            if f is None:
+                raise TableNotFound(f"flow {stmt.name}")
+            return QueryResult(
+                ["Flow", "Create Flow"],
+                [[stmt.name],
+                 [f"CREATE FLOW {stmt.name} SINK TO {f.sink} AS "
+                  f"{f.select_sql}"]])
+        if isinstance(stmt, ast.ShowIndex):
+            st = self.engine.table(stmt.name)
+            rows = []
+            for pos, pk in enumerate(st.schema.primary_key):
+                rows.append([stmt.name, "PRIMARY", pk, pos + 1,
+                             "greptime-inverted-index-v1"])
+            rows.append([stmt.name, "TIME INDEX",
+                         st.schema.time_index.name, 1, "time-index"])
+            for sn, ft in sorted(st.regions[0].text_cols.items()):
+                rows.append([stmt.name, "FULLTEXT INDEX", sn, 1,
+                             "greptime-fulltext-index-v1"])
+            return QueryResult(
+                ["table", "key_name", "column_name", "seq_in_index",
+                 "index_type"],
+                [list(c) for c in zip(*rows)] if rows else
+                [[], [], [], [], []])
         if isinstance(stmt, ast.DescribeTable):
             st = self.engine.table(stmt.name)
             cols = st.schema.columns

@@ -190,8 +190,19 @@ class Parser:
             elif self.eat_kw("databases") or self.eat_kw("schemas"):
                 stmt = ast.ShowDatabases()
             elif self.eat_kw("create"):
-                self.expect_kw("table")
-                stmt = ast.ShowCreateTable(self.next().value)
+                if self.eat_kw("view"):
+                    stmt = ast.ShowCreateView(self.next().value)
+                elif self.eat_kw("flow"):
+                    stmt = ast.ShowCreateFlow(self.next().value)
+                else:
+                    self.expect_kw("table")
+                    stmt = ast.ShowCreateTable(self.next().value)
+            elif self.eat_kw("columns") or self.eat_kw("fields"):
+                self.expect_kw("from")
+                stmt = ast.DescribeTable(self.next().value)
+            elif self.eat_kw("index", "indexes"):
+                self.expect_kw("from")
+                stmt = ast.ShowIndex(self.next().value)
             elif self.eat_kw("variables"):
                 like = None
                 if self.eat_kw("like"):

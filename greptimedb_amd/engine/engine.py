@@ -41,6 +41,8 @@ class EngineConfig:
     gc_grace_s: float = 3600.0          # spare files younger than this
     cold_compress_s: float = 0.0        # Gorilla-pack SST batches idle this
                                         # long (0 = disabled; K20 cold tier)
+    global_write_buffer_bytes: int = 16 << 30   # node-wide memtable cap
+                                        # (reference WriteBufferManagerImpl)
 
 
 @dataclass
@@ -406,13 +408,32 @@ class MitoEngine:
         self.wal.commit()
 
     def maybe_flush(self):
+        total = 0
         for st in self.tables.values():
             for region in st.regions:
+                total += region.memtable.bytes_used
                 if region.should_flush(self.config.flush_bytes):
                     if self._flusher is not None:
                         self._flush_q.put(region)
                     else:
                         self._flush_region(region)
+        # global write-buffer accounting (reference WriteBufferManagerImpl):
+        # when the NODE-wide memtable total exceeds the cap, flush the
+        # largest memtables until projected usage is back under it — keeps
+        # many under-threshold regions from accumulating unbounded HBM
+        if total > self.config.global_write_buffer_bytes:
+            regions = sorted(
+                (r for st in self.tables.values() for r in st.regions
+                 if r.memtable.bytes_used > 0),
+                key=lambda r: r.memtable.bytes_used, reverse=True)
+            for r in regions:
+                if total <= self.config.global_write_buffer_bytes:
+                    break
+                total -= r.memtable.bytes_used
+                if self._flusher is not None:
+                    self._flush_q.put(r)
+                else:
+                    self._flush_region(r)
 
     def flush_all(self, wait: bool = True):
         for st in self.tables.values():

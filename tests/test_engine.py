@@ -745,3 +745,25 @@ def test_cold_tier_auto_compress(tmp_path):
     assert after == before
     assert all(b.ts is not None for b in packed)  # hot again
     eng.close()
+
+
+def test_global_write_buffer_flush(tmp_path):
+    """Node-wide memtable cap flushes the largest regions even when no
+    single region hits its own threshold (reference:
+    WriteBufferManagerImpl global accounting)."""
+    eng = MitoEngine(EngineConfig(
+        data_dir=str(tmp_path / "d"), device="cpu", background_flush=False,
+        flush_bytes=1 << 30,                 # per-region threshold never hit
+        global_write_buffer_bytes=200_000))  # tiny node-wide cap
+    ing = Ingestor(eng)
+    w = CpuWorkload(scale=20)
+    for _ in range(10):
+        ing.ingest_lines(w.next_batch(2000))
+    total_mem = sum(r.memtable.bytes_used for st in eng.tables.values()
+                    for r in st.regions)
+    assert total_mem <= 200_000 + (1 << 16), total_mem
+    files = sum(len(r.manifest.files) for st in eng.tables.values()
+                for r in st.regions)
+    assert files >= 1    # data went to SSTs, not lost
+    assert _total(eng) == 20000
+    eng.close()

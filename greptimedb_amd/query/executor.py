@@ -780,7 +780,20 @@ class Executor:
         import time as _time
         if not isinstance(e.stmt, ast.Select):
             raise PlanQuery("EXPLAIN supports SELECT")
-        plan = self._plan_select(e.stmt)
+        try:
+            plan = self._plan_select(e.stmt)
+        except PlanQuery as pq:
+            # general-shape fallback (expression aggregates, GROUP BY on
+            # fields, ORDER BY expressions): materialize + derived eval
+            lines = ["Plan: materialize-fallback (derived-table eval)",
+                     f"  reason: {pq}",
+                     f"  table: {e.stmt.table}"]
+            if e.analyze:
+                t0 = _time.perf_counter()
+                r = self.execute_stmt(e.stmt)
+                lines.append(f"Execution: {len(r)} rows in "
+                             f"{(_time.perf_counter() - t0) * 1000:.3f} ms")
+            return QueryResult(["plan"], [lines])
         lines = []
         if any(_has_range_agg(x) for x, _a in e.stmt.projections):
             path = "range-select (sliding-window kernel)"

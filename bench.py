@@ -57,9 +57,10 @@ def main():
                          "GPU — ≥30s sustained over 20 steps — 60k on CPU)")
     ap.add_argument("--scale", type=int, default=100,
                     help="TSBS scale (hosts) per ingest worker shard")
-    ap.add_argument("--workers", type=int, default=8,
+    ap.add_argument("--workers", type=int, default=6,
                     help="ingest worker threads per rank (tsbs_load "
-                         "--workers analog; 8 measured fastest on MI355X)")
+                         "--workers analog; 6 vs 8 within run-to-run "
+                         "variance on MI355X — see profiles worker sweep)")
     ap.add_argument("--pool-rows", type=int, default=None,
                     help="pre-generated line pool size per rank (replayed "
                          "cyclically; default 2.4M on GPU, 120k on CPU)")
@@ -70,9 +71,9 @@ def main():
                          "commit, no fsync (reference raft-engine default "
                          "sync=false); fsync = fdatasync per group commit; "
                          "off = no WAL")
-    ap.add_argument("--wal-shards", type=int, default=8,
+    ap.add_argument("--wal-shards", type=int, default=4,
                     help="parallel WAL writers (region-sharded segments, "
-                         "merged by seq at replay; 8 measured fastest)")
+                         "merged by seq at replay)")
     ap.add_argument("--flush-mb", type=int, default=64,
                     help="per-region memtable flush threshold — 64MB keeps "
                          "flush+compaction live during the timed region")

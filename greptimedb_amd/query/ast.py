@@ -213,6 +213,23 @@ class TruncateTable:
 
 
 @dataclass
+class DeclareCursor:
+    name: str
+    select: object
+
+
+@dataclass
+class FetchCursor:
+    name: str
+    count: int
+
+
+@dataclass
+class CloseCursor:
+    name: str
+
+
+@dataclass
 class ShowTables:
     like: str | None = None
 

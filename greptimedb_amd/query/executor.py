@@ -284,6 +284,24 @@ class Executor:
         if isinstance(stmt, ast.TruncateTable):
             self.engine.truncate_table(stmt.name)
             return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.DeclareCursor):
+            r = self.execute_stmt(stmt.select)
+            self.session.cursors[stmt.name] = [r, 0]
+            return QueryResult(["status"], [["ok"]])
+        if isinstance(stmt, ast.FetchCursor):
+            cur = self.session.cursors.get(stmt.name)
+            if cur is None:
+                raise InvalidArguments(f"no such cursor {stmt.name!r}")
+            r, pos = cur
+            rows = list(r.rows())[pos: pos + max(stmt.count, 0)]
+            cur[1] = pos + len(rows)
+            cols = [list(c) for c in zip(*rows)] if rows else \
+                [[] for _ in r.names]
+            return QueryResult(r.names, cols, r.kinds)
+        if isinstance(stmt, ast.CloseCursor):
+            if self.session.cursors.pop(stmt.name, None) is None:
+                raise InvalidArguments(f"no such cursor {stmt.name!r}")
+            return QueryResult(["status"], [["ok"]])
         if isinstance(stmt, ast.ShowTables):
             names = sorted(self.engine.tables)
             if stmt.like:

@@ -183,6 +183,24 @@ class Parser:
             self.next()
             self.eat_kw("table")
             stmt = ast.TruncateTable(str(self.next().value))
+        elif self.at_kw("declare"):
+            # DECLARE c CURSOR FOR SELECT … (ref sql statements/cursor.rs)
+            self.next()
+            name = str(self.next().value)
+            self.expect_kw("cursor")
+            self.expect_kw("for")
+            stmt = ast.DeclareCursor(name, self.parse_query())
+        elif self.at_kw("fetch"):
+            self.next()
+            n = 1
+            t = self.peek()
+            if t is not None and t.kind == "num":
+                n = int(self.next().value)
+            self.eat_kw("from")
+            stmt = ast.FetchCursor(str(self.next().value), n)
+        elif self.at_kw("close"):
+            self.next()
+            stmt = ast.CloseCursor(str(self.next().value))
         elif self.at_kw("show"):
             self.next()
             if self.eat_kw("flows"):

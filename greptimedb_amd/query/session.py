@@ -36,6 +36,7 @@ class Session:
         self.timezone = timezone
         self.tz_offset_ms = 0
         self.vars: dict[str, str] = {}
+        self.cursors: dict = {}   # name -> (QueryResult, position)
 
     def set_var(self, name: str, value: str):
         name = name.lower().lstrip("@")

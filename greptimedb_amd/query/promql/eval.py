@@ -46,6 +46,12 @@ ELEMENTWISE = {
     "abs": torch.abs, "ceil": torch.ceil, "floor": torch.floor,
     "exp": torch.exp, "ln": torch.log, "log2": torch.log2,
     "log10": torch.log10, "sqrt": torch.sqrt, "sgn": torch.sgn,
+    # trigonometric family (Prometheus experimental trig functions)
+    "sin": torch.sin, "cos": torch.cos, "tan": torch.tan,
+    "asin": torch.asin, "acos": torch.acos, "atan": torch.atan,
+    "sinh": torch.sinh, "cosh": torch.cosh, "tanh": torch.tanh,
+    "asinh": torch.asinh, "acosh": torch.acosh, "atanh": torch.atanh,
+    "deg": lambda t: torch.rad2deg(t), "rad": lambda t: torch.deg2rad(t),
 }
 
 _TIME_FUNCS = {"minute", "hour", "day_of_week", "day_of_month",
@@ -279,6 +285,18 @@ class PromEvaluator:
             if isinstance(m, PromScalar):
                 return PromMatrix([{}], _scalar_plane(m, T), grid)
             return m
+        if f == "pi":
+            return PromScalar(math.pi, grid)
+        if f in ("sort_by_label", "sort_by_label_desc"):
+            m = self._eval(e.args[0], t0, step, T, grid)
+            keys = [a.value for a in e.args[1:]]
+            order = sorted(range(m.S), key=lambda i: tuple(
+                str(m.labels[i].get(k, "")) for k in keys),
+                reverse=(f == "sort_by_label_desc"))
+            import torch as _t
+            oi = _t.as_tensor(order, device=m.values.device)
+            return PromMatrix([m.labels[i] for i in order], m.values[oi],
+                              grid)
         if f == "time":
             return PromScalar(torch.as_tensor(grid / 1000.0,
                                               dtype=torch.float64), grid)

@@ -113,6 +113,11 @@ class Cast(Expr):
 
 
 @dataclass
+class SysVar(Expr):
+    name: str  # @@name / @@session.name (lowered, @@ stripped)
+
+
+@dataclass
 class InList(Expr):
     expr: Expr
     items: list[Expr]

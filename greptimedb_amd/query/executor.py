@@ -195,6 +195,36 @@ class Executor:
             log.append({"sql": sql[:500], "ms": round(dt, 1)})
         return r
 
+    def _eval_session_const(self, e):
+        """Session/system functions MySQL+PG clients probe on connect
+        (reference: session context functions in common/function)."""
+        from greptimedb_amd import __version__
+        if isinstance(e, ast.SysVar):
+            name = e.name.lower().removeprefix("session.") \
+                .removeprefix("global.")
+            if name == "version":
+                return f"8.4.2-greptimedb-amd-{__version__}"
+            if name in ("time_zone", "timezone", "system_time_zone"):
+                return self.session.timezone
+            if name in ("autocommit", "sql_mode"):
+                return {"autocommit": 1, "sql_mode": ""}[name]
+            return self.session.vars.get(name, "")
+        if isinstance(e, ast.Func) and not e.args:
+            fn = e.name.lower()
+            if fn == "version":
+                return f"greptimedb-amd {__version__}"
+            if fn in ("database", "current_schema", "schema"):
+                return self.session.schema
+            if fn in ("current_user", "user", "session_user"):
+                return "greptime"
+            if fn == "connection_id":
+                return 1
+        if isinstance(e, ast.Col) and e.name.lower() in (
+                "current_timestamp", "current_date", "current_time"):
+            import time as _time
+            return int(_time.time() * 1000)
+        return _eval_const(e)
+
     def _cancel_check(self):
         """Raise if this query's process entry was KILLed — polled inside
         per-region scan loops so long scans terminate promptly
@@ -1297,8 +1327,9 @@ class Executor:
             # constant select
             names, cols = [], []
             for i, (e, alias) in enumerate(sel.projections):
-                v = _eval_const(e)
-                names.append(alias or f"col{i}")
+                v = self._eval_session_const(e)
+                names.append(alias or _expr_name(e) if not isinstance(
+                    e, (ast.Lit,)) else (alias or f"col{i}"))
                 cols.append([v])
             return QueryResult(names, cols)
         from greptimedb_amd.query.information_schema import is_information_schema
@@ -3870,6 +3901,8 @@ def _expr_name(e: ast.Expr) -> str:
         return "case"
     if isinstance(e, ast.ScalarSubquery):
         return "(subquery)"
+    if isinstance(e, ast.SysVar):
+        return "@@" + e.name
     if isinstance(e, ast.Cast):
         return f"CAST({_expr_name(e.expr)} AS {e.type.upper()})"
     return repr(e)

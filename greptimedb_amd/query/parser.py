@@ -836,6 +836,9 @@ class Parser:
             return e
         if t.kind == "op" and t.value == "-":
             return ast.UnaryOp("-", self.parse_expr(7))
+        if t.kind == "op" and t.value == "@" and self.eat_op("@"):
+            # @@name / @@session.name system variables (MySQL dialect)
+            return ast.SysVar(str(self.next().value).lower())
         if t.kind == "op" and t.value == "*":
             return ast.Star()
         if t.kind == "id":

@@ -22,6 +22,8 @@ class RegionSupervisor:
         self.threshold = threshold
         self.acceptable_pause_ms = acceptable_pause_ms
         self.failed: set[str] = set()
+        self.maintenance = False   # True = suppress auto-failover
+                                   # (reference: metasrv maintenance mode)
 
     def heartbeat(self, node_id: str, now_ms: float | None = None):
         now_ms = now_ms if now_ms is not None else time.time() * 1000
@@ -34,8 +36,12 @@ class RegionSupervisor:
         self.failed.discard(node_id)
 
     def check(self, now_ms: float | None = None) -> list[str]:
-        """Return nodes whose φ exceeds the threshold; fire failover once."""
+        """Return nodes whose φ exceeds the threshold; fire failover once.
+        In maintenance mode detection still runs but no failover fires
+        (planned restarts must not trigger region migration storms)."""
         now_ms = now_ms if now_ms is not None else time.time() * 1000
+        if self.maintenance:
+            return []
         newly = []
         for node, det in self.detectors.items():
             if node in self.failed:

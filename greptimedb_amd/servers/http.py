@@ -175,6 +175,21 @@ def build_app(ctx: ServerContext) -> FastAPI:
                     round((_t.perf_counter() - t0) * 1000, 3)}
 
     # ---------------- pprof-style profiling (reference http.rs:1079) ----
+    @app.api_route("/debug/log_level", methods=["GET", "POST"])
+    async def dyn_log_level(request: Request):
+        """Dynamic log-level reload (reference: servers http/dyn_log.rs).
+        POST body or ?level= sets the root level; GET returns it."""
+        import logging
+        root = logging.getLogger()
+        if request.method == "POST":
+            level = request.query_params.get("level") or \
+                (await request.body()).decode().strip()
+            lv = getattr(logging, level.upper(), None)
+            if not isinstance(lv, int):
+                return Response(f"unknown level {level!r}", status_code=400)
+            root.setLevel(lv)
+        return {"level": logging.getLevelName(root.level)}
+
     @app.get("/debug/prof/cpu")
     async def prof_cpu(seconds: int = Query(2), frequency: int = Query(99)):
         """Sampling CPU profile over `seconds`; returns folded stacks

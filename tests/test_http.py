@@ -249,3 +249,17 @@ def test_readonly_user_permission(tmp_path):
                headers=hdr("admin", "secret"))
     assert r.json()["output"][0]["records"]["rows"][0][0] == 1
     eng.close()
+
+
+def test_dyn_log_level(client):
+    import logging
+    prev = logging.getLogger().level
+    try:
+        r = client.post("/debug/log_level", params={"level": "debug"})
+        assert r.json()["level"] == "DEBUG"
+        r = client.get("/debug/log_level")
+        assert r.json()["level"] == "DEBUG"
+        assert client.post("/debug/log_level",
+                           params={"level": "nope"}).status_code == 400
+    finally:
+        logging.getLogger().setLevel(prev)

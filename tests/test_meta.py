@@ -231,3 +231,22 @@ def test_wire_heartbeat_and_mailbox(tmp_path):
     hb.stop()
     srv.close()
     eng.close()
+
+
+def test_maintenance_mode_suppresses_failover():
+    """Maintenance mode: φ detection keeps running but no failover fires
+    (reference: metasrv maintenance mode for planned restarts)."""
+    from greptimedb_amd.meta.supervisor import RegionSupervisor
+    failed = []
+    sup = RegionSupervisor(on_failover=failed.append, threshold=1.5,
+                           acceptable_pause_ms=100.0)
+    t = 0.0
+    for _ in range(20):
+        sup.heartbeat("node-a", t)
+        t += 100.0
+    sup.maintenance = True
+    assert sup.check(t + 60_000.0) == []
+    assert failed == []
+    sup.maintenance = False
+    assert sup.check(t + 60_000.0) == ["node-a"]
+    assert failed == ["node-a"]

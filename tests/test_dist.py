@@ -43,6 +43,7 @@ r2 = ex.execute("SELECT hostname, avg(usage_user) FROM cpu GROUP BY hostname ORD
 r3 = ex.execute("SELECT date_trunc('minute', ts) m, max(usage_user) FROM cpu GROUP BY m ORDER BY m")
 r4 = ex.execute("SELECT ts, hostname, usage_user FROM cpu WHERE usage_user > 99")
 r5 = ex.execute("SELECT hostname, last_value(usage_user) FROM cpu GROUP BY hostname ORDER BY hostname")
+r5b = ex.execute("SELECT hostname, first_value(usage_user) FROM cpu GROUP BY hostname ORDER BY hostname")
 # PromQL distributed aggregation (merge_prom_planes over gloo)
 from greptimedb_amd.query.promql.eval import PromEvaluator
 ev = PromEvaluator(eng, dist=DistContext(device="cpu"))
@@ -74,6 +75,7 @@ out = {
     "minutes": len(r3),
     "raw": len(r4),
     "lastpoint_hosts": len(r5),
+    "firstpoint_hosts": len(r5b),
     "prom_sum": round(float(m.values[0][-1]), 6),
     "range_lv_rows": len(r7),
     "range_lv_hosts": len(sorted(set(r7.columns[1]))),
@@ -119,6 +121,7 @@ def test_two_rank_query_combine(tmp_path):
     assert len(results[0]["hosts"]) == 20    # 10 hosts per rank, disjoint
     assert results[0]["raw"] >= 0
     assert results[0]["lastpoint_hosts"] == 20
+    assert results[0]["firstpoint_hosts"] == 20
     assert results[0]["prom_sum"] != 0.0
     assert results[0]["range_hosts"] == 20   # all ranks' hosts in the plane
     assert results[0]["range_rows"] > 40

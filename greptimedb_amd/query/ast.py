@@ -235,13 +235,25 @@ class CloseCursor:
 
 
 @dataclass
-class ShowTables:
+class ShowTableStatus:
     like: str | None = None
 
 
 @dataclass
+class ValuesTable:
+    rows: list
+    columns: list | None = None
+
+
+@dataclass
+class ShowTables:
+    like: str | None = None
+    full: bool = False
+
+
+@dataclass
 class ShowDatabases:
-    pass
+    like: str | None = None
 
 
 @dataclass

@@ -340,13 +340,35 @@ class Executor:
                     "^" + _re2.escape(stmt.like).replace("%", ".*")
                     .replace("_", ".") + "$")
                 names = [n for n in names if pat.match(n)]
+            if stmt.full:
+                return QueryResult(["Tables", "Table_type"],
+                                   [names, ["BASE TABLE"] * len(names)])
             return QueryResult(["Tables"], [names])
         if isinstance(stmt, ast.ShowDatabases):
             extra = sorted(getattr(self.engine, "schemas", set()))
-            return QueryResult(["Database"],
-                               [sorted({"greptime_private",
-                                        "information_schema", "public",
-                                        *extra})])
+            dbs = sorted({"greptime_private", "information_schema", "public",
+                          *extra})
+            if stmt.like:
+                import re as _re2
+                pat = _re2.compile(
+                    "^" + _re2.escape(stmt.like).replace("%", ".*")
+                    .replace("_", ".") + "$")
+                dbs = [d for d in dbs if pat.match(d)]
+            return QueryResult(["Database"], [dbs])
+        if isinstance(stmt, ast.ShowTableStatus):
+            names = sorted(self.engine.tables)
+            if stmt.like:
+                import re as _re2
+                pat = _re2.compile(
+                    "^" + _re2.escape(stmt.like).replace("%", ".*")
+                    .replace("_", ".") + "$")
+                names = [n for n in names if pat.match(n)]
+            rows = [(n, "mito-hip",
+                     sum(r.num_rows for r in self.engine.tables[n].regions))
+                    for n in names]
+            return QueryResult(["Name", "Engine", "Rows"],
+                               [list(c) for c in zip(*rows)] if rows
+                               else [[], [], []])
         if isinstance(stmt, ast.ShowCreateTable):
             return self._show_create_table(stmt.name)
         if isinstance(stmt, ast.ShowCreateView):
@@ -1300,6 +1322,16 @@ class Executor:
     def _exec_select(self, sel: ast.Select) -> QueryResult:
         if sel.ctes:
             return self._with_ctes(sel)
+        if isinstance(sel.table, ast.ValuesTable):
+            # FROM (VALUES ...) t(a, b) — literal rows as a derived table
+            from greptimedb_amd.query.derived import select_over_result
+            vt = sel.table
+            width = max((len(r) for r in vt.rows), default=0)
+            names2 = list(vt.columns) if vt.columns else                 [f"column{i+1}" for i in range(width)]
+            cols2 = [[r[i] if i < len(r) else None for r in vt.rows]
+                     for i in range(width)]
+            base = QueryResult(names2, cols2)
+            return select_over_result(sel, base)
         if isinstance(sel.table, (ast.Select, ast.SetOp)):
             # derived table: FROM (SELECT ...) alias
             from greptimedb_amd.query.derived import select_over_result

@@ -206,7 +206,10 @@ class Parser:
             if self.eat_kw("flows"):
                 stmt = ast.ShowFlows()
             elif self.eat_kw("databases") or self.eat_kw("schemas"):
-                stmt = ast.ShowDatabases()
+                like = None
+                if self.eat_kw("like"):
+                    like = str(self.next().value)
+                stmt = ast.ShowDatabases(like)
             elif self.eat_kw("create"):
                 if self.eat_kw("view"):
                     stmt = ast.ShowCreateView(self.next().value)
@@ -226,12 +229,19 @@ class Parser:
                 if self.eat_kw("like"):
                     like = str(self.next().value)
                 stmt = ast.ShowVariables(like)
+            elif self.eat_kw("table"):
+                self.expect_kw("status")
+                like = None
+                if self.eat_kw("like"):
+                    like = str(self.next().value)
+                stmt = ast.ShowTableStatus(like)
             else:
+                full = self.eat_kw("full")
                 self.expect_kw("tables")
                 like = None
                 if self.eat_kw("like"):
                     like = str(self.next().value)
-                stmt = ast.ShowTables(like)
+                stmt = ast.ShowTables(like, full)
         elif self.at_kw("describe", "desc"):
             self.next()
             self.eat_kw("table")
@@ -379,12 +389,21 @@ class Parser:
         if self.eat_kw("from"):
             if self.at_op("("):
                 self.next()
-                table = self.parse_query()   # derived table
+                if self.at_kw("values"):
+                    table = self._parse_values_tail()
+                else:
+                    table = self.parse_query()   # derived table
                 self.expect_op(")")
             else:
                 table = self.next().value
             if self.eat_kw("as"):
                 table_alias = self.next().value
+                if isinstance(table, ast.ValuesTable) and self.eat_op("("):
+                    cols = []
+                    while not self.eat_op(")"):
+                        cols.append(str(self.next().value))
+                        self.eat_op(",")
+                    table.columns = cols
             elif self.peek() is not None and self.peek().kind == "id" and \
                     self.peek().value.lower() not in (
                         "where", "group", "order", "limit", "having", "offset",
@@ -738,6 +757,26 @@ class Parser:
         return ast.Tql(start, end, step, query)
 
     # ---------------- expressions (Pratt) ----------------
+    def _parse_values_tail(self) -> "ast.ValuesTable":
+        """VALUES (a, b), (c, d) — literal row constructor
+        (reference: sqlparser VALUES table factor)."""
+        self.expect_kw("values")
+        rows = []
+        while True:
+            self.expect_op("(")
+            row = []
+            while True:
+                e = self.parse_expr()
+                from greptimedb_amd.query.executor import _eval_const
+                row.append(_eval_const(e))
+                if not self.eat_op(","):
+                    break
+            self.expect_op(")")
+            rows.append(row)
+            if not self.eat_op(","):
+                break
+        return ast.ValuesTable(rows)
+
     def _parse_type_name(self) -> str:
         """Type name after CAST(... AS …) / `::` — single identifier with
         optional (n[,m]) length args (ignored) and the DOUBLE PRECISION /
@@ -831,6 +870,8 @@ class Parser:
                 sub = self.parse_select()
                 self.expect_op(")")
                 return ast.ScalarSubquery(sub)
+            if self.at_kw("values"):   # VALUES row constructor
+                return self._parse_values_tail()
             e = self.parse_expr()
             self.expect_op(")")
             return e

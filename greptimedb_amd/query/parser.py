@@ -725,7 +725,15 @@ class Parser:
                 elif isinstance(e, ast.UnaryOp) and e.op == "-" and isinstance(e.operand, ast.Lit):
                     row.append(-e.operand.value)
                 else:
-                    raise InvalidSyntax("INSERT VALUES must be literals")
+                    # constant expressions (now(), 1+2, CAST(...)) fold at
+                    # parse time (reference: sqlparser values exprs)
+                    from greptimedb_amd.query.executor import _eval_const
+                    try:
+                        row.append(_eval_const(e))
+                    except Exception:
+                        raise InvalidSyntax(
+                            "INSERT VALUES must be constant expressions") \
+                            from None
                 if not self.eat_op(","):
                     break
             self.expect_op(")")
